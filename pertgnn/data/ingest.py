@@ -1,0 +1,201 @@
+"""Offline ingest: raw Alibaba-shaped CSVs -> processed/ artifacts.
+
+Re-implements the reference preprocessing pipeline (preprocess.py:191-381)
+with the same observable behavior and artifact formats,
+vectorized where the reference loops row-by-row in pandas:
+
+  processed/processed_df.csv            — filtered + factorized call rows
+  processed/processed_resource_df.csv   — (ts,ms) resource feature table
+  processed/tr2ts_map.joblib            — traceid -> 30 s ts bucket
+  processed/tr2data.pt                  — traceid -> {entry_id, runtime_id, timestamp, y}
+  processed/entry2runtimes.joblib       — entry -> {runtime_id: prob}
+  processed/runtime2spangraph_map.pt    — runtime_id -> span graph dict
+  processed/runtime2pertgraph_map.pt    — runtime_id -> PERT graph dict
+"""
+from __future__ import annotations
+
+import os
+
+import joblib
+import numpy as np
+import pandas as pd
+import torch
+
+from .graphs import build_pert_graph, build_span_graph
+from .schema import TS_BUCKET_MS
+
+
+def map_consecutive(df: pd.DataFrame, cols: list[str]):
+    """Factorize string ids to consecutive ints across ``cols`` jointly
+    (reference preprocess.py:80-96)."""
+    stacked = df[cols].stack()
+    codes, uniques = stacked.factorize()
+    df.loc[:, cols] = pd.Series(codes, index=stacked.index).unstack()
+    return df, uniques
+
+
+def detect_entries(df: pd.DataFrame) -> pd.DataFrame:
+    """Entry detection + trace filter (preprocess.py:99-149), vectorized.
+
+    entry row = rpctype=='http' & ts==trace-min & |rt|==trace-max; if several,
+    tie-break um=='(?)'; still ambiguous or none -> drop the trace.
+    entryid = dm + '_' + interface of the entry row.
+    """
+    g = df.groupby("traceid")
+    ts_min = g["timestamp"].transform("min")
+    rt_abs = df["rt"].abs()
+    rt_max = rt_abs.groupby(df["traceid"]).transform("max")
+    cand = df[(df["rpctype"] == "http") & (df["timestamp"] == ts_min) & (rt_abs == rt_max)]
+
+    n_cand = cand.groupby("traceid").size()
+    unique_tr = n_cand[n_cand == 1].index
+    multi_tr = n_cand[n_cand > 1].index
+    # tie-break on um == '(?)' (preprocess.py:121-123)
+    tie = cand[cand["traceid"].isin(multi_tr) & (cand["um"] == "(?)")]
+    n_tie = tie.groupby("traceid").size()
+    tie_ok = n_tie[n_tie == 1].index
+
+    entry_rows = pd.concat(
+        [cand[cand["traceid"].isin(unique_tr)], tie[tie["traceid"].isin(tie_ok)]]
+    )
+    entry_str = entry_rows["dm"].astype(str) + "_" + entry_rows["interface"].astype(str)
+    tr2entry = dict(zip(entry_rows["traceid"], entry_str))
+    df = df[df["traceid"].isin(tr2entry.keys())].copy()
+    df["entryid"] = df["traceid"].map(tr2entry)
+    return df
+
+
+def filter_by_resource_coverage(df: pd.DataFrame, resource_df: pd.DataFrame, min_frac: float = 0.6) -> pd.DataFrame:
+    """Keep traces where >= min_frac of their microservices have resource rows
+    (preprocess.py:155-177)."""
+    ms_with_res = set(resource_df["msname"].values)
+    long = pd.concat(
+        [df[["traceid", "um"]].rename(columns={"um": "ms"}),
+         df[["traceid", "dm"]].rename(columns={"dm": "ms"})]
+    ).drop_duplicates()
+    long["has"] = long["ms"].isin(ms_with_res)
+    frac = long.groupby("traceid")["has"].mean()
+    keep = frac[frac >= min_frac].index
+    return df[df["traceid"].isin(keep)]
+
+
+def filter_by_entry_occurrence(df: pd.DataFrame, min_occurence: int = 100) -> pd.DataFrame:
+    """Keep entries with > min_occurence distinct traces (preprocess.py:180-188)."""
+    occ = df.groupby("entryid")["traceid"].nunique()
+    keep = occ[occ > min_occurence].index
+    return df[df["entryid"].isin(keep)]
+
+
+def build_resource_table(resource_raw: pd.DataFrame) -> pd.DataFrame:
+    """(ts,ms) group-agg of instance cpu/mem into 8 feature columns
+    (preprocess.py:227-242)."""
+    rs = resource_raw.loc[:, ["timestamp", "msname", "instance_cpu_usage", "instance_memory_usage"]]
+    rs = rs.groupby(["timestamp", "msname"]).agg(["max", "min", "mean", "median"])
+    rs.columns = ["_".join(c) for c in rs.columns]
+    return rs.reset_index()
+
+
+def get_df(data_root: str, processed_dir: str, min_occurence: int = 100):
+    """Reference get_df (preprocess.py:191-266): read or build the processed
+    call + resource frames."""
+    p_df = os.path.join(processed_dir, "processed_df.csv")
+    p_rs = os.path.join(processed_dir, "processed_resource_df.csv")
+    if os.path.isfile(p_df) and os.path.isfile(p_rs):
+        df = pd.read_csv(p_df)
+        resource_df = pd.read_csv(p_rs)
+    else:
+        cg_dir = os.path.join(data_root, "MSCallGraph")
+        df = pd.concat(
+            (pd.read_csv(os.path.join(cg_dir, f), index_col=0).replace(np.nan, "nan")
+             for f in sorted(os.listdir(cg_dir)) if f.endswith(".csv")),
+            ignore_index=True,
+        ).drop_duplicates()
+        df = df.sort_values(by=["timestamp"])
+
+        df, _ = map_consecutive(df, ["traceid"])
+        df, _ = map_consecutive(df, ["interface"])
+        df = detect_entries(df)
+        df, _ = map_consecutive(df, ["entryid"])
+        df, _ = map_consecutive(df, ["rpcid"])
+        df, _ = map_consecutive(df, ["rpctype"])
+
+        rs_dir = os.path.join(data_root, "MSResource")
+        resource_raw = pd.concat(
+            pd.read_csv(os.path.join(rs_dir, f))
+            for f in sorted(os.listdir(rs_dir)) if f.endswith(".csv")
+        )
+        resource_df = build_resource_table(resource_raw)
+
+        df = filter_by_resource_coverage(df, resource_df)
+        df = filter_by_entry_occurrence(df, min_occurence=min_occurence)
+
+        # joint um/dm/msname factorization (preprocess.py:248-254)
+        unique_ms = list(set(df.um.values) | set(df.dm.values) | set(resource_df.msname.values))
+        ms2int = dict(zip(unique_ms, range(len(unique_ms))))
+        df["um"] = df.um.map(ms2int)
+        df["dm"] = df.dm.map(ms2int)
+        resource_df["msname"] = resource_df.msname.map(ms2int)
+
+        os.makedirs(processed_dir, exist_ok=True)
+        df.to_csv(p_df, index=False)
+        resource_df.to_csv(p_rs, index=False)
+
+    df["endTimestamp"] = df["timestamp"] + df["rt"].abs()
+    resource_df["msname"] = resource_df["msname"].astype(int)
+    return df, resource_df
+
+
+def run_ingest(data_root: str = "data", processed_dir: str = "processed", min_occurence: int = 100, verbose: bool = True):
+    """Full pipeline (reference preprocess.py main, :269-381)."""
+    os.makedirs(processed_dir, exist_ok=True)
+    df, resource_df = get_df(data_root, processed_dir, min_occurence=min_occurence)
+
+    # trace start-time bucket (preprocess.py:32-41)
+    tr2ts = (df.groupby("traceid")["timestamp"].min() // TS_BUCKET_MS * TS_BUCKET_MS)
+    joblib.dump(tr2ts, os.path.join(processed_dir, "tr2ts_map.joblib"))
+
+    # runtime-pattern id: the trace's um_dm_interface token string, factorized
+    # (preprocess.py:280-293)
+    tok = df["um"].astype(str) + "_" + df["dm"].astype(str) + "_" + df["interface"].astype(str)
+    corpus = tok.groupby(df["traceid"]).apply(" ".join)
+    tr2runtime = dict(zip(corpus.index, pd.factorize(corpus)[0]))
+    tr2delay = df["rt"].abs().groupby(df["traceid"]).max().to_dict()
+
+    tr2data = {}
+    entry2runtimes: dict = {}
+    runtime2span: dict = {}
+    runtime2pert: dict = {}
+
+    for entry, entry_group in df.groupby("entryid"):
+        for traceid, trace_df in entry_group.groupby("traceid"):
+            runtime_id = int(tr2runtime[traceid])
+            tr2data[traceid] = {
+                "entry_id": int(entry),
+                "runtime_id": runtime_id,
+                "timestamp": int(tr2ts[traceid]),
+                "y": torch.tensor(tr2delay[traceid]),
+            }
+            entry2runtimes.setdefault(int(entry), {})
+            entry2runtimes[int(entry)][runtime_id] = entry2runtimes[int(entry)].get(runtime_id, 0) + 1
+            if runtime_id not in runtime2span:
+                runtime2span[runtime_id] = build_span_graph(trace_df)
+            else:
+                runtime2span[runtime_id]["occurences"] += 1
+            if runtime_id not in runtime2pert:
+                runtime2pert[runtime_id] = build_pert_graph(trace_df)
+            else:
+                runtime2pert[runtime_id]["occurences"] += 1
+        if verbose:
+            print(f"entry {entry}: {entry_group['traceid'].nunique()} traces")
+
+    # normalize occurrence counts to probabilities (preprocess.py:372-375)
+    for entry, rt_counts in entry2runtimes.items():
+        total = sum(rt_counts.values())
+        for rt_id in rt_counts:
+            rt_counts[rt_id] = rt_counts[rt_id] / total
+
+    torch.save(runtime2span, os.path.join(processed_dir, "runtime2spangraph_map.pt"))
+    torch.save(runtime2pert, os.path.join(processed_dir, "runtime2pertgraph_map.pt"))
+    torch.save(tr2data, os.path.join(processed_dir, "tr2data.pt"))
+    joblib.dump(entry2runtimes, os.path.join(processed_dir, "entry2runtimes.joblib"))
+    return tr2data, entry2runtimes, runtime2span, runtime2pert, resource_df

@@ -1,0 +1,170 @@
+"""PERT-GNN graph-transformer model — MI355X-native re-implementation.
+
+State_dict-compatible with the reference ``model.py`` (SAGEDeterministic,
+/root/reference/model.py:10-114) WITHOUT any PyG dependency: the conv is our
+own TransformerConv equivalent whose submodule names match PyG 2.4.0
+(``lin_key``/``lin_query``/``lin_value``/``lin_edge``/``lin_skip``) so
+checkpoints keep the same keys, and whose math runs through
+``pertgnn.ops.functional`` (HIP kernels on GPU, eager oracle on CPU).
+
+Reference quirks intentionally reproduced:
+  * ``num_layers=1`` still builds 2 convs (model.py:24-52).
+  * dead ``edge_linear`` lazy module kept in the state_dict (model.py:68).
+  * ``local_pred`` returned but unused by the training loss (pert_gnn.py:245).
+  * pattern_probs consumed per-NODE (the rebuilt rt_probs of pert_gnn.py:224-230).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from ..ops import functional as ops
+
+
+class TransformerConv(nn.Module):
+    """Graph transformer conv, PyG-2.4.0 semantics with heads=1, concat=True,
+    root_weight=True, beta=False (reference model.py:25-52 configuration).
+
+    out_i = W_skip x_i + b + sum_{e:(j->i)} softmax_i(<W_q x_i, W_k x_j + W_e e_ij>/sqrt(H))
+            * (W_v x_j + W_e e_ij)
+    """
+
+    def __init__(self, in_channels: int, out_channels: int, heads: int = 1, edge_dim: int | None = None):
+        super().__init__()
+        assert heads == 1, "reference configuration uses heads=1 (model.py:29)"
+        self.in_channels = in_channels
+        self.out_channels = out_channels
+        self.heads = heads
+        self.edge_dim = edge_dim
+        self.lin_key = nn.Linear(in_channels, heads * out_channels)
+        self.lin_query = nn.Linear(in_channels, heads * out_channels)
+        self.lin_value = nn.Linear(in_channels, heads * out_channels)
+        self.lin_edge = nn.Linear(edge_dim, heads * out_channels, bias=False)
+        self.lin_skip = nn.Linear(in_channels, out_channels, bias=True)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        # PyG Linear default init is glorot for weight, zeros for bias.
+        for lin in (self.lin_key, self.lin_query, self.lin_value, self.lin_edge, self.lin_skip):
+            nn.init.xavier_uniform_(lin.weight)
+            if lin.bias is not None:
+                nn.init.zeros_(lin.bias)
+
+    def forward(self, x, edge_index, edge_embeds, csr=None, num_nodes=None):
+        n = x.shape[0] if num_nodes is None else num_nodes
+        q = ops.linear(x, self.lin_query.weight, self.lin_query.bias)
+        k = ops.linear(x, self.lin_key.weight, self.lin_key.bias)
+        v = ops.linear(x, self.lin_value.weight, self.lin_value.bias)
+        e = ops.linear(edge_embeds, self.lin_edge.weight, None)
+        skip = ops.linear(x, self.lin_skip.weight, self.lin_skip.bias)
+        return ops.edge_attention(q, k, v, e, skip, edge_index, n, csr=csr)
+
+
+class SAGEDeterministic(nn.Module):
+    """API- and checkpoint-compatible with reference model.py:10-114."""
+
+    def __init__(
+        self,
+        in_channels,
+        cat_dims,
+        entry_id_max,
+        interface_id_max,
+        rpctype_id_max,
+        hidden_channels,
+        num_layers,
+        dropout,
+    ):
+        super().__init__()
+        self.convs = nn.ModuleList()
+        self.convs.append(
+            TransformerConv(
+                in_channels=in_channels + hidden_channels,
+                out_channels=hidden_channels,
+                heads=1,
+                edge_dim=hidden_channels * 2,
+            )
+        )
+        self.bns = nn.ModuleList()
+        self.bns.append(nn.BatchNorm1d(hidden_channels))
+        for _ in range(num_layers - 2):
+            self.convs.append(
+                TransformerConv(
+                    in_channels=hidden_channels,
+                    out_channels=hidden_channels,
+                    heads=1,
+                    edge_dim=hidden_channels * 2,
+                )
+            )
+            self.bns.append(nn.BatchNorm1d(hidden_channels))
+        self.convs.append(
+            TransformerConv(
+                in_channels=hidden_channels,
+                out_channels=hidden_channels,
+                heads=1,
+                edge_dim=hidden_channels * 2,
+            )
+        )
+        self.local_linear = nn.Linear(hidden_channels, 1)
+        self.global_linear1 = nn.Linear(hidden_channels * 2, hidden_channels)
+        self.global_linear2 = nn.Linear(hidden_channels, 1)
+        self.cat_embedding = nn.ModuleList()
+        for num_categories in cat_dims:
+            self.cat_embedding.append(nn.Embedding(num_categories, hidden_channels))
+
+        self.dropout = dropout
+        self.entry_embeds = nn.Embedding(entry_id_max + 1, hidden_channels)
+        self.interface_embeds = nn.Embedding(interface_id_max + 1, hidden_channels)
+        self.rpctype_embeds = nn.Embedding(rpctype_id_max + 1, hidden_channels)
+        # Dead module, never called in forward — kept lazily-uninitialized so the
+        # state_dict matches the reference (model.py:68).
+        self.edge_linear = nn.LazyLinear(hidden_channels * 2)
+
+    def reset_parameters(self):
+        for conv in self.convs:
+            conv.reset_parameters()
+        for bn in self.bns:
+            bn.reset_parameters()
+
+    def forward(
+        self,
+        x,
+        cat_X,
+        edge_index,
+        edge_attr,
+        pattern_num_nodes,
+        pattern_probs,
+        entry_id,
+        batch,
+        csr=None,
+        num_graphs=None,
+    ):
+        if num_graphs is None:
+            num_graphs = int(batch.max().item()) + 1 if batch.numel() else 0
+        x = ops.embed_concat_node(x, cat_X, [t.weight for t in self.cat_embedding])
+        edge_embeds = ops.embed_concat_edge(
+            edge_attr, self.interface_embeds.weight, self.rpctype_embeds.weight
+        )
+        n = x.shape[0]
+
+        for i, conv in enumerate(self.convs[:-1]):
+            x = conv(x, edge_index, edge_embeds, csr=csr, num_nodes=n)
+            bn = self.bns[i]
+            x = ops.batchnorm_relu(
+                x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
+                bn.momentum, bn.eps, self.training, fuse_relu=True,
+            )
+            if self.training and bn.track_running_stats and bn.num_batches_tracked is not None:
+                bn.num_batches_tracked += 1
+            x = F.dropout(x, p=self.dropout, training=self.training)
+        x = self.convs[-1](x, edge_index, edge_embeds, csr=csr, num_nodes=n)
+        local_predict = ops.linear(x, self.local_linear.weight, self.local_linear.bias)
+        mean_x = ops.pattern_pool(x, pattern_probs, pattern_num_nodes, batch, num_graphs)
+        global_predict = torch.cat([mean_x, self.entry_embeds(entry_id)], dim=1)
+        h = ops.linear(global_predict, self.global_linear1.weight, self.global_linear1.bias)
+        global_predict = ops.linear(F.relu(h), self.global_linear2.weight, self.global_linear2.bias)
+        # reference comment says "ensure non-negative" but applies no clamp
+        # (model.py:113-114) — reproduced as-is.
+        return global_predict, local_predict

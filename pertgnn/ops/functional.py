@@ -1,0 +1,240 @@
+"""Dispatching functional ops: HIP kernels on GPU, eager oracle on CPU.
+
+Every op takes/returns plain tensors so the model code is path-agnostic.
+Edge order contract: edges are ALWAYS destination-sorted (CSR order) inside a
+collated batch — the collator (pertgnn/data/collate.py, reference K17) emits
+them that way, so ``row_ptr``/``csr_src`` describe ``edge_attr``/``e`` rows
+directly and no per-kernel permutation is needed.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+
+from . import reference as ref
+from .backend import ext, use_hip
+
+
+# ---------------------------------------------------------------------------
+# fused edge attention (K3-K6 fwd, K15 bwd)
+# ---------------------------------------------------------------------------
+
+class _EdgeAttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, e, skip, row_ptr, csr_src, col_ptr, csc_dst, csc_eid):
+        m = ext()
+        out, alpha = m.edge_attn_fwd(q, k, v, e, row_ptr, csr_src, skip)
+        ctx.save_for_backward(q, k, v, e, alpha, row_ptr, csr_src, col_ptr, csc_dst, csc_eid)
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        q, k, v, e, alpha, row_ptr, csr_src, col_ptr, csc_dst, csc_eid = ctx.saved_tensors
+        m = ext()
+        dq, dk, dv, de = m.edge_attn_bwd(
+            g.contiguous(), q, k, v, e, alpha, row_ptr, csr_src, col_ptr, csc_dst, csc_eid
+        )
+        # skip grad is g itself
+        return dq, dk, dv, de, g, None, None, None, None, None
+
+
+def edge_attention(q, k, v, e, skip, edge_index, num_nodes, csr=None):
+    """out_i = skip_i + sum_e softmax_i(<q_i, k_src+e>/sqrt(H)) (v_src+e).
+
+    ``csr``: (row_ptr, csr_src, col_ptr, csc_dst, csc_eid) from the collator;
+    required on the HIP path.
+    """
+    if use_hip(q):
+        row_ptr, csr_src, col_ptr, csc_dst, csc_eid = csr
+        return _EdgeAttentionFn.apply(q, k, v, e, skip, row_ptr, csr_src, col_ptr, csc_dst, csc_eid)
+    return ref.edge_attention(q, k, v, e, edge_index, num_nodes, skip)
+
+
+# ---------------------------------------------------------------------------
+# pattern pool (K9+K10)
+# ---------------------------------------------------------------------------
+
+class _PatternPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, probs, nnodes, batch, num_graphs):
+        m = ext()
+        out = m.seg_pool_fwd(x, probs, nnodes, batch, num_graphs)
+        ctx.save_for_backward(probs, nnodes, batch)
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        probs, nnodes, batch = ctx.saved_tensors
+        m = ext()
+        dx = m.seg_pool_bwd(g.contiguous(), probs, nnodes, batch)
+        return dx, None, None, None, None
+
+
+def pattern_pool(x, pattern_probs, pattern_num_nodes, batch, num_graphs):
+    if use_hip(x):
+        return _PatternPoolFn.apply(x, pattern_probs, pattern_num_nodes, batch, num_graphs)
+    return ref.pattern_pool(x, pattern_probs, pattern_num_nodes, batch, num_graphs)
+
+
+# ---------------------------------------------------------------------------
+# embedding gathers (K1 + K11)
+# ---------------------------------------------------------------------------
+
+class _EmbedNodeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x_raw, cat_idx, table):
+        m = ext()
+        out = m.embed_node_fwd(x_raw, cat_idx, table)
+        ctx.save_for_backward(cat_idx)
+        ctx.f = x_raw.shape[1]
+        ctx.rows = table.shape[0]
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        (cat_idx,) = ctx.saved_tensors
+        m = ext()
+        g = g.contiguous()
+        dx_raw = g[:, : ctx.f].contiguous()
+        dtable = m.embed_scatter_add(g, cat_idx, ctx.f, ctx.rows)
+        return dx_raw, None, dtable
+
+
+class _EmbedEdgeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, edge_attr, ifc_table, rpc_table):
+        m = ext()
+        out = m.embed_edge_fwd(edge_attr, ifc_table, rpc_table)
+        ctx.save_for_backward(edge_attr)
+        ctx.rows = (ifc_table.shape[0], rpc_table.shape[0])
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        (edge_attr,) = ctx.saved_tensors
+        m = ext()
+        g = g.contiguous()
+        d_ifc, d_rpc = m.embed_edge_bwd(g, edge_attr, ctx.rows[0], ctx.rows[1])
+        return None, d_ifc, d_rpc
+
+
+def embed_concat_node(x_raw, cat_X, tables):
+    if len(tables) == 1 and use_hip(x_raw):
+        return _EmbedNodeFn.apply(x_raw, cat_X[:, 0].contiguous(), tables[0])
+    return ref.embed_concat_node(x_raw, cat_X, tables)
+
+
+def embed_concat_edge(edge_attr, interface_table, rpctype_table):
+    if use_hip(interface_table):
+        return _EmbedEdgeFn.apply(edge_attr, interface_table, rpctype_table)
+    return ref.embed_concat_edge(edge_attr, interface_table, rpctype_table)
+
+
+# ---------------------------------------------------------------------------
+# batchnorm + relu (K7+K8)
+# ---------------------------------------------------------------------------
+
+class _BNReLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu):
+        m = ext()
+        y, save_mean, save_invstd = m.bn_relu_fwd(
+            x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu
+        )
+        ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+        ctx.fuse_relu = fuse_relu
+        return y
+
+    @staticmethod
+    def backward(ctx, g):
+        x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
+        m = ext()
+        dx, dgamma, dbeta = m.bn_relu_bwd(
+            g.contiguous(), x, gamma, save_mean, save_invstd, y, ctx.fuse_relu
+        )
+        return dx, dgamma, dbeta, None, None, None, None, None, None
+
+
+def batchnorm_relu(x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu=True):
+    """BatchNorm1d over N per channel, optional fused ReLU (model.py:101-102)."""
+    if use_hip(x):
+        return _BNReLUFn.apply(x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu)
+    y = torch.nn.functional.batch_norm(
+        x, running_mean, running_var, gamma, beta, training, momentum, eps
+    )
+    return torch.nn.functional.relu(y) if fuse_relu else y
+
+
+# ---------------------------------------------------------------------------
+# loss + metrics (K12/K13)
+# ---------------------------------------------------------------------------
+
+class _QuantileLossFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, y, y_hat, tau):
+        m = ext()
+        loss = m.quantile_loss_fwd(y, y_hat, tau)
+        ctx.save_for_backward(y, y_hat)
+        ctx.tau = tau
+        return loss
+
+    @staticmethod
+    def backward(ctx, g):
+        y, y_hat = ctx.saved_tensors
+        m = ext()
+        dy_hat = m.quantile_loss_bwd(g, y, y_hat, ctx.tau)
+        return None, dy_hat, None
+
+
+def quantile_loss(y, y_hat, tau):
+    if use_hip(y_hat):
+        return _QuantileLossFn.apply(y, y_hat, tau)
+    return ref.quantile_loss(y, y_hat, tau)
+
+
+def eval_metrics(y, y_hat, tau):
+    if use_hip(y_hat):
+        return ext().eval_metrics(y, y_hat, tau)
+    return ref.eval_metrics(y, y_hat, tau)
+
+
+# ---------------------------------------------------------------------------
+# linear (K2) — MFMA GEMM on the HIP path
+# ---------------------------------------------------------------------------
+
+class _LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        m = ext()
+        y = m.linear_fwd(x, w, b if b is not None else torch.empty(0, dtype=x.dtype, device=x.device))
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, g):
+        x, w = ctx.saved_tensors
+        m = ext()
+        g = g.contiguous()
+        dx, dw, db = m.linear_bwd(g, x, w, ctx.has_bias)
+        return dx, dw, (db if ctx.has_bias else None)
+
+
+def linear(x, w, b=None):
+    """y = x @ w^T + b with torch.nn.Linear weight layout [out,in]."""
+    if use_hip(x) and hasattr(ext(), "linear_fwd"):
+        return _LinearFn.apply(x, w, b)
+    return torch.nn.functional.linear(x, w, b)
+
+
+__all__ = [
+    "edge_attention",
+    "pattern_pool",
+    "embed_concat_node",
+    "embed_concat_edge",
+    "batchnorm_relu",
+    "quantile_loss",
+    "eval_metrics",
+    "linear",
+]
