@@ -1,0 +1,307 @@
+// pertgnn._C — Python bindings for the CDNA4 kernel library.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include <c10/cuda/CUDAStream.h>
+
+namespace {
+
+#define CHECK_IN(t)                                                      \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                      \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+inline hipStream_t cur_stream() {
+  return (hipStream_t)c10::cuda::getCurrentCUDAStream().stream();
+}
+
+}  // namespace
+
+// launchers (csrc/hip/*.hip)
+void launch_edge_attn_fwd(const float*, const float*, const float*,
+                          const float*, const int*, const int*, const float*,
+                          float*, float*, int, int, hipStream_t);
+void launch_edge_attn_bwd(const float*, const float*, const float*,
+                          const float*, const float*, const float*, const int*,
+                          const int*, const int*, const int*, float*, float*,
+                          float*, float*, float*, float*, int, int, long,
+                          hipStream_t);
+void launch_seg_pool_fwd(const float*, const float*, const float*, const int*,
+                         float*, int, int, hipStream_t);
+void launch_seg_pool_bwd(const float*, const float*, const float*, const long*,
+                         float*, long, int, hipStream_t);
+void launch_embed_node_fwd(const float*, const long*, const float*, float*,
+                           long, int, int, hipStream_t);
+void launch_embed_scatter_add(const float*, const long*, float*, long, int,
+                              int, hipStream_t);
+void launch_embed_edge_fwd(const long*, const float*, const float*, float*,
+                           long, int, int, hipStream_t);
+void launch_embed_edge_bwd(const float*, const long*, float*, float*, long,
+                           int, int, hipStream_t);
+void launch_gather_rows(const long*, const float*, float*, long, int,
+                        hipStream_t);
+void launch_scatter_add_rows(const float*, const long*, float*, long, int,
+                             hipStream_t);
+void launch_bn_fwd(const float*, const float*, const float*, float*, float*,
+                   float*, float*, float*, float*, long, int, float, float,
+                   bool, bool, hipStream_t);
+void launch_bn_bwd(const float*, const float*, const float*, const float*,
+                   const float*, const float*, float*, float*, float*, float*,
+                   long, int, bool, hipStream_t);
+void launch_quantile_loss_fwd(const float*, const float*, float*, long, float,
+                              hipStream_t);
+void launch_quantile_loss_bwd(const float*, const float*, const float*, float*,
+                              long, float, hipStream_t);
+void launch_eval_metrics(const float*, const float*, float*, long, float,
+                         hipStream_t);
+void launch_adam(float*, const float*, float*, float*, long, float, float,
+                 float, float, int, hipStream_t);
+
+// ---------------------------------------------------------------------------
+
+std::vector<torch::Tensor> edge_attn_fwd(torch::Tensor q, torch::Tensor k,
+                                         torch::Tensor v, torch::Tensor e,
+                                         torch::Tensor row_ptr,
+                                         torch::Tensor csr_src,
+                                         torch::Tensor skip) {
+  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v); CHECK_IN(e);
+  CHECK_IN(row_ptr); CHECK_IN(csr_src);
+  const int n = q.size(0);
+  const int h = q.size(1);
+  TORCH_CHECK(h <= 512, "H must be <= 512");
+  auto out = torch::empty_like(q);
+  auto alpha = torch::empty({e.size(0)}, q.options());
+  const float* skip_p = nullptr;
+  if (skip.defined() && skip.numel() > 0) {
+    CHECK_IN(skip);
+    skip_p = skip.data_ptr<float>();
+  }
+  launch_edge_attn_fwd(q.data_ptr<float>(), k.data_ptr<float>(),
+                       v.data_ptr<float>(), e.data_ptr<float>(),
+                       row_ptr.data_ptr<int>(), csr_src.data_ptr<int>(),
+                       skip_p, out.data_ptr<float>(), alpha.data_ptr<float>(),
+                       n, h, cur_stream());
+  return {out, alpha};
+}
+
+std::vector<torch::Tensor> edge_attn_bwd(
+    torch::Tensor g, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor e, torch::Tensor alpha, torch::Tensor row_ptr,
+    torch::Tensor csr_src, torch::Tensor col_ptr, torch::Tensor csc_dst,
+    torch::Tensor csc_eid) {
+  CHECK_IN(g); CHECK_IN(q); CHECK_IN(alpha);
+  const int n = q.size(0);
+  const int h = q.size(1);
+  const long ne = e.size(0);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto de = torch::empty_like(e);
+  auto dek = torch::empty_like(e);
+  auto dev = torch::empty_like(e);
+  launch_edge_attn_bwd(g.data_ptr<float>(), q.data_ptr<float>(),
+                       k.data_ptr<float>(), v.data_ptr<float>(),
+                       e.data_ptr<float>(), alpha.data_ptr<float>(),
+                       row_ptr.data_ptr<int>(), csr_src.data_ptr<int>(),
+                       col_ptr.data_ptr<int>(), csc_eid.data_ptr<int>(),
+                       dq.data_ptr<float>(), dk.data_ptr<float>(),
+                       dv.data_ptr<float>(), de.data_ptr<float>(),
+                       dek.data_ptr<float>(), dev.data_ptr<float>(), n, h, ne,
+                       cur_stream());
+  return {dq, dk, dv, de};
+}
+
+torch::Tensor seg_pool_fwd(torch::Tensor x, torch::Tensor probs,
+                           torch::Tensor nn, torch::Tensor batch_ptr,
+                           int64_t num_graphs) {
+  CHECK_IN(x); CHECK_IN(probs); CHECK_IN(nn); CHECK_IN(batch_ptr);
+  const int h = x.size(1);
+  auto out = torch::empty({num_graphs, h}, x.options());
+  launch_seg_pool_fwd(x.data_ptr<float>(), probs.data_ptr<float>(),
+                      nn.data_ptr<float>(), batch_ptr.data_ptr<int>(),
+                      out.data_ptr<float>(), (int)num_graphs, h, cur_stream());
+  return out;
+}
+
+torch::Tensor seg_pool_bwd(torch::Tensor gout, torch::Tensor probs,
+                           torch::Tensor nn, torch::Tensor batch) {
+  CHECK_IN(gout); CHECK_IN(batch);
+  const long n = batch.size(0);
+  const int h = gout.size(1);
+  auto dx = torch::empty({n, h}, gout.options());
+  launch_seg_pool_bwd(gout.data_ptr<float>(), probs.data_ptr<float>(),
+                      nn.data_ptr<float>(), batch.data_ptr<long>(),
+                      dx.data_ptr<float>(), n, h, cur_stream());
+  return dx;
+}
+
+torch::Tensor embed_node_fwd(torch::Tensor x_raw, torch::Tensor idx,
+                             torch::Tensor table) {
+  CHECK_IN(x_raw); CHECK_IN(idx); CHECK_IN(table);
+  const long n = x_raw.size(0);
+  const int f = x_raw.size(1);
+  const int h = table.size(1);
+  auto out = torch::empty({n, f + h}, x_raw.options());
+  launch_embed_node_fwd(x_raw.data_ptr<float>(), idx.data_ptr<long>(),
+                        table.data_ptr<float>(), out.data_ptr<float>(), n, f,
+                        h, cur_stream());
+  return out;
+}
+
+torch::Tensor embed_scatter_add(torch::Tensor g, torch::Tensor idx, int64_t f,
+                                int64_t rows) {
+  CHECK_IN(g); CHECK_IN(idx);
+  const long n = g.size(0);
+  const int h = g.size(1) - f;
+  auto dtable = torch::zeros({rows, h}, g.options());
+  launch_embed_scatter_add(g.data_ptr<float>(), idx.data_ptr<long>(),
+                           dtable.data_ptr<float>(), n, (int)f, h,
+                           cur_stream());
+  return dtable;
+}
+
+torch::Tensor embed_edge_fwd(torch::Tensor attr, torch::Tensor ifc,
+                             torch::Tensor rpc) {
+  CHECK_IN(attr); CHECK_IN(ifc); CHECK_IN(rpc);
+  const long e = attr.size(0);
+  const int h = ifc.size(1);
+  const int astride = attr.size(1);
+  auto out = torch::empty({e, 2 * h}, ifc.options());
+  launch_embed_edge_fwd(attr.data_ptr<long>(), ifc.data_ptr<float>(),
+                        rpc.data_ptr<float>(), out.data_ptr<float>(), e, h,
+                        astride, cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> embed_edge_bwd(torch::Tensor g, torch::Tensor attr,
+                                          int64_t rows_ifc, int64_t rows_rpc) {
+  CHECK_IN(g); CHECK_IN(attr);
+  const long e = attr.size(0);
+  const int h = g.size(1) / 2;
+  const int astride = attr.size(1);
+  auto difc = torch::zeros({rows_ifc, h}, g.options());
+  auto drpc = torch::zeros({rows_rpc, h}, g.options());
+  launch_embed_edge_bwd(g.data_ptr<float>(), attr.data_ptr<long>(),
+                        difc.data_ptr<float>(), drpc.data_ptr<float>(), e, h,
+                        astride, cur_stream());
+  return {difc, drpc};
+}
+
+torch::Tensor gather_rows(torch::Tensor idx, torch::Tensor table) {
+  CHECK_IN(idx); CHECK_IN(table);
+  const long n = idx.size(0);
+  const int h = table.size(1);
+  auto out = torch::empty({n, h}, table.options());
+  launch_gather_rows(idx.data_ptr<long>(), table.data_ptr<float>(),
+                     out.data_ptr<float>(), n, h, cur_stream());
+  return out;
+}
+
+torch::Tensor scatter_add_rows(torch::Tensor g, torch::Tensor idx,
+                               int64_t rows) {
+  CHECK_IN(g); CHECK_IN(idx);
+  const long n = g.size(0);
+  const int h = g.size(1);
+  auto dtable = torch::zeros({rows, h}, g.options());
+  launch_scatter_add_rows(g.data_ptr<float>(), idx.data_ptr<long>(),
+                          dtable.data_ptr<float>(), n, h, cur_stream());
+  return dtable;
+}
+
+std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
+                                       torch::Tensor beta,
+                                       torch::Tensor running_mean,
+                                       torch::Tensor running_var,
+                                       double momentum, double eps,
+                                       bool training, bool relu) {
+  CHECK_IN(x); CHECK_IN(gamma); CHECK_IN(beta);
+  const long n = x.size(0);
+  const int h = x.size(1);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({h}, x.options());
+  auto invstd = torch::empty({h}, x.options());
+  auto partials = torch::empty({2 * h}, x.options());
+  launch_bn_fwd(x.data_ptr<float>(), gamma.data_ptr<float>(),
+                beta.data_ptr<float>(), running_mean.data_ptr<float>(),
+                running_var.data_ptr<float>(), mean.data_ptr<float>(),
+                invstd.data_ptr<float>(), partials.data_ptr<float>(),
+                y.data_ptr<float>(), n, h, (float)momentum, (float)eps,
+                training, relu, cur_stream());
+  return {y, mean, invstd};
+}
+
+std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor g, torch::Tensor x,
+                                       torch::Tensor gamma, torch::Tensor mean,
+                                       torch::Tensor invstd, torch::Tensor y,
+                                       bool relu) {
+  CHECK_IN(g); CHECK_IN(x);
+  const long n = x.size(0);
+  const int h = x.size(1);
+  auto dx = torch::empty_like(x);
+  auto dgamma = torch::empty({h}, x.options());
+  auto dbeta = torch::empty({h}, x.options());
+  auto partials = torch::empty({2 * h}, x.options());
+  launch_bn_bwd(g.data_ptr<float>(), x.data_ptr<float>(), y.data_ptr<float>(),
+                mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                gamma.data_ptr<float>(), partials.data_ptr<float>(),
+                dx.data_ptr<float>(), dgamma.data_ptr<float>(),
+                dbeta.data_ptr<float>(), n, h, relu, cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
+torch::Tensor quantile_loss_fwd(torch::Tensor y, torch::Tensor y_hat,
+                                double tau) {
+  CHECK_IN(y); CHECK_IN(y_hat);
+  auto out = torch::empty({}, y_hat.options());
+  launch_quantile_loss_fwd(y.data_ptr<float>(), y_hat.data_ptr<float>(),
+                           out.data_ptr<float>(), y.size(0), (float)tau,
+                           cur_stream());
+  return out;
+}
+
+torch::Tensor quantile_loss_bwd(torch::Tensor g, torch::Tensor y,
+                                torch::Tensor y_hat, double tau) {
+  auto gc = g.contiguous();
+  auto dy_hat = torch::empty_like(y_hat);
+  launch_quantile_loss_bwd(gc.data_ptr<float>(), y.data_ptr<float>(),
+                           y_hat.data_ptr<float>(), dy_hat.data_ptr<float>(),
+                           y.size(0), (float)tau, cur_stream());
+  return dy_hat;
+}
+
+std::vector<torch::Tensor> eval_metrics(torch::Tensor y, torch::Tensor y_hat,
+                                        double tau) {
+  CHECK_IN(y); CHECK_IN(y_hat);
+  auto out3 = torch::empty({3}, y_hat.options());
+  launch_eval_metrics(y.data_ptr<float>(), y_hat.data_ptr<float>(),
+                      out3.data_ptr<float>(), y.size(0), (float)tau,
+                      cur_stream());
+  return {out3[0], out3[1], out3[2]};
+}
+
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, double lr, double b1, double b2, double eps,
+               int64_t step) {
+  CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v);
+  launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+              v.data_ptr<float>(), p.numel(), (float)lr, (float)b1, (float)b2,
+              (float)eps, (int)step, cur_stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("edge_attn_fwd", &edge_attn_fwd);
+  mod.def("edge_attn_bwd", &edge_attn_bwd);
+  mod.def("seg_pool_fwd", &seg_pool_fwd);
+  mod.def("seg_pool_bwd", &seg_pool_bwd);
+  mod.def("embed_node_fwd", &embed_node_fwd);
+  mod.def("embed_scatter_add", &embed_scatter_add);
+  mod.def("embed_edge_fwd", &embed_edge_fwd);
+  mod.def("embed_edge_bwd", &embed_edge_bwd);
+  mod.def("gather_rows", &gather_rows);
+  mod.def("scatter_add_rows", &scatter_add_rows);
+  mod.def("bn_relu_fwd", &bn_relu_fwd);
+  mod.def("bn_relu_bwd", &bn_relu_bwd);
+  mod.def("quantile_loss_fwd", &quantile_loss_fwd);
+  mod.def("quantile_loss_bwd", &quantile_loss_bwd);
+  mod.def("eval_metrics", &eval_metrics);
+  mod.def("adam_step", &adam_step);
+}

@@ -1,0 +1,32 @@
+// Shared helpers for the pertgnn CDNA4 (gfx950) kernel library.
+// Wave size is 64 on CDNA — every cross-lane idiom below is 64-wide.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#define PERTGNN_WAVE 64
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(_e), __FILE__,       \
+             __LINE__);                                                        \
+      abort();                                                                 \
+    }                                                                          \
+  } while (0)
+
+// full-wave (64-lane) sum/max reductions via xor shuffles
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+__device__ __forceinline__ float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }

@@ -1,0 +1,530 @@
+// Segment / elementwise / reduction kernels (gfx950):
+//   K9+K10  pattern-weighted global-add-pool (fwd: wave-per-graph segment sum
+//           over the batch-ptr — deterministic, no atomics; bwd: gather)
+//   K1+K11  embedding gathers fused with concat (+ scatter-add backward)
+//   K7+K8   BatchNorm1d (+fused ReLU) forward/backward
+//   K12/K13 quantile loss + eval-metric reductions
+//   K14     Adam step (flat master buffers — one kernel for the whole model)
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// pattern pool: out[b] = sum_{i in graph b} x[i] * p[i] / n[i]
+// ---------------------------------------------------------------------------
+
+#define WAVES_PER_BLOCK 4
+
+template <int VPT>
+__global__ void seg_pool_fwd_kernel(const float* __restrict__ x,
+                                    const float* __restrict__ probs,
+                                    const float* __restrict__ nn,
+                                    const int* __restrict__ batch_ptr,
+                                    float* __restrict__ out, int b, int h) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int g = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (g >= b) return;
+  float acc[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
+  for (int i = batch_ptr[g]; i < batch_ptr[g + 1]; ++i) {
+    const float w = probs[i] / nn[i];
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) acc[j] += x[(long)i * h + c] * w;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    if (c < h) out[(long)g * h + c] = acc[j];
+  }
+}
+
+__global__ void seg_pool_bwd_kernel(const float* __restrict__ gout,
+                                    const float* __restrict__ probs,
+                                    const float* __restrict__ nn,
+                                    const long* __restrict__ batch,
+                                    float* __restrict__ dx, long n, int h) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const long numel = n * h;
+  for (long t = i; t < numel; t += stride) {
+    const long row = t / h;
+    const int c = (int)(t - row * h);
+    dx[t] = gout[batch[row] * h + c] * probs[row] / nn[row];
+  }
+}
+
+void launch_seg_pool_fwd(const float* x, const float* probs, const float* nn,
+                         const int* batch_ptr, float* out, int b, int h,
+                         hipStream_t stream) {
+  if (b == 0) return;
+  const dim3 grid(ceil_div(b, WAVES_PER_BLOCK));
+  const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
+  const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  switch (vpt) {
+#define CASE(V)                                                                \
+  case V:                                                                      \
+    seg_pool_fwd_kernel<V><<<grid, block, 0, stream>>>(x, probs, nn,           \
+                                                       batch_ptr, out, b, h); \
+    break;
+    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+#undef CASE
+    default: abort();
+  }
+}
+
+void launch_seg_pool_bwd(const float* gout, const float* probs,
+                         const float* nn, const long* batch, float* dx, long n,
+                         int h, hipStream_t stream) {
+  if (n == 0) return;
+  const long numel = n * h;
+  const int tpb = 256;
+  const int blocks = (int)min((numel + tpb - 1) / tpb, (long)4096);
+  seg_pool_bwd_kernel<<<dim3(blocks), dim3(tpb), 0, stream>>>(gout, probs, nn,
+                                                              batch, dx, n, h);
+}
+
+// ---------------------------------------------------------------------------
+// embedding gathers fused with concat
+// ---------------------------------------------------------------------------
+
+__global__ void embed_node_fwd_kernel(const float* __restrict__ x_raw,
+                                      const long* __restrict__ idx,
+                                      const float* __restrict__ table,
+                                      float* __restrict__ out, long n, int f,
+                                      int h) {
+  const int w = f + h;
+  const long numel = n * w;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i0; t < numel; t += stride) {
+    const long row = t / w;
+    const int c = (int)(t - row * w);
+    out[t] = (c < f) ? x_raw[row * f + c] : table[idx[row] * h + (c - f)];
+  }
+}
+
+__global__ void embed_scatter_add_kernel(const float* __restrict__ g,
+                                         const long* __restrict__ idx,
+                                         float* __restrict__ dtable, long n,
+                                         int f, int h) {
+  const int w = f + h;
+  const long numel = n * h;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i0; t < numel; t += stride) {
+    const long row = t / h;
+    const int c = (int)(t - row * h);
+    atomicAdd(&dtable[idx[row] * h + c], g[row * w + f + c]);
+  }
+}
+
+__global__ void embed_edge_fwd_kernel(const long* __restrict__ attr,
+                                      const float* __restrict__ ifc,
+                                      const float* __restrict__ rpc,
+                                      float* __restrict__ out, long e, int h,
+                                      int astride) {
+  const int w = 2 * h;
+  const long numel = e * w;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i0; t < numel; t += stride) {
+    const long row = t / w;
+    const int c = (int)(t - row * w);
+    out[t] = (c < h) ? ifc[attr[row * astride] * h + c]
+                     : rpc[attr[row * astride + 1] * h + (c - h)];
+  }
+}
+
+__global__ void embed_edge_bwd_kernel(const float* __restrict__ g,
+                                      const long* __restrict__ attr,
+                                      float* __restrict__ difc,
+                                      float* __restrict__ drpc, long e, int h,
+                                      int astride) {
+  const int w = 2 * h;
+  const long numel = e * w;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i0; t < numel; t += stride) {
+    const long row = t / w;
+    const int c = (int)(t - row * w);
+    if (c < h)
+      atomicAdd(&difc[attr[row * astride] * h + c], g[t]);
+    else
+      atomicAdd(&drpc[attr[row * astride + 1] * h + (c - h)], g[t]);
+  }
+}
+
+// entry embedding gather: out[b] = table[idx[b]] — plain gather (fwd) +
+// scatter-add (bwd); reuses the edge kernels' grid-stride shape.
+__global__ void gather_rows_kernel(const long* __restrict__ idx,
+                                   const float* __restrict__ table,
+                                   float* __restrict__ out, long n, int h) {
+  const long numel = n * h;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i0; t < numel; t += stride) {
+    const long row = t / h;
+    const int c = (int)(t - row * h);
+    out[t] = table[idx[row] * h + c];
+  }
+}
+
+__global__ void scatter_add_rows_kernel(const float* __restrict__ g,
+                                        const long* __restrict__ idx,
+                                        float* __restrict__ dtable, long n,
+                                        int h) {
+  const long numel = n * h;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i0; t < numel; t += stride) {
+    const long row = t / h;
+    const int c = (int)(t - row * h);
+    atomicAdd(&dtable[idx[row] * h + c], g[t]);
+  }
+}
+
+static int grid_for(long numel, int tpb = 256, long cap = 4096) {
+  return (int)min((numel + tpb - 1) / tpb, cap);
+}
+
+void launch_embed_node_fwd(const float* x_raw, const long* idx,
+                           const float* table, float* out, long n, int f,
+                           int h, hipStream_t s) {
+  if (n == 0) return;
+  embed_node_fwd_kernel<<<grid_for(n * (f + h)), 256, 0, s>>>(x_raw, idx, table,
+                                                              out, n, f, h);
+}
+void launch_embed_scatter_add(const float* g, const long* idx, float* dtable,
+                              long n, int f, int h, hipStream_t s) {
+  if (n == 0) return;
+  embed_scatter_add_kernel<<<grid_for(n * h), 256, 0, s>>>(g, idx, dtable, n,
+                                                           f, h);
+}
+void launch_embed_edge_fwd(const long* attr, const float* ifc,
+                           const float* rpc, float* out, long e, int h,
+                           int astride, hipStream_t s) {
+  if (e == 0) return;
+  embed_edge_fwd_kernel<<<grid_for(e * 2 * h), 256, 0, s>>>(attr, ifc, rpc, out,
+                                                            e, h, astride);
+}
+void launch_embed_edge_bwd(const float* g, const long* attr, float* difc,
+                           float* drpc, long e, int h, int astride,
+                           hipStream_t s) {
+  if (e == 0) return;
+  embed_edge_bwd_kernel<<<grid_for(e * 2 * h), 256, 0, s>>>(g, attr, difc, drpc,
+                                                            e, h, astride);
+}
+void launch_gather_rows(const long* idx, const float* table, float* out,
+                        long n, int h, hipStream_t s) {
+  if (n == 0) return;
+  gather_rows_kernel<<<grid_for(n * h), 256, 0, s>>>(idx, table, out, n, h);
+}
+void launch_scatter_add_rows(const float* g, const long* idx, float* dtable,
+                             long n, int h, hipStream_t s) {
+  if (n == 0) return;
+  scatter_add_rows_kernel<<<grid_for(n * h), 256, 0, s>>>(g, idx, dtable, n, h);
+}
+
+// ---------------------------------------------------------------------------
+// BatchNorm1d (+ReLU) — two-stage column reduction, coalesced row tiles
+// ---------------------------------------------------------------------------
+
+// stage A: per-block partial sum/sumsq over a row range, accumulated into
+// global partials[2*h] with one atomicAdd per channel per block.
+__global__ void bn_stats_partial_kernel(const float* __restrict__ x, long n,
+                                        int h, float* __restrict__ partials) {
+  extern __shared__ float smem[];  // [2*h]
+  for (int c = threadIdx.x; c < 2 * h; c += blockDim.x) smem[c] = 0.f;
+  __syncthreads();
+  const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * rows_per_block;
+  const long r1 = min(n, r0 + rows_per_block);
+  for (long r = r0; r < r1; ++r) {
+    for (int c = threadIdx.x; c < h; c += blockDim.x) {
+      const float v = x[r * h + c];
+      atomicAdd(&smem[c], v);          // LDS atomics — cheap, per-CU
+      atomicAdd(&smem[h + c], v * v);
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < 2 * h; c += blockDim.x)
+    if (smem[c] != 0.f) atomicAdd(&partials[c], smem[c]);
+}
+
+// stage B: finalize mean/invstd (+ running-stat update, training only)
+__global__ void bn_finalize_kernel(const float* __restrict__ partials, long n,
+                                   int h, float eps, float momentum,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   int update_running) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= h) return;
+  const float m = partials[c] / n;
+  const float var = fmaxf(partials[h + c] / n - m * m, 0.f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+  if (update_running) {
+    const float unbiased = (n > 1) ? var * n / (n - 1) : var;
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
+// stage C: normalize + affine (+ReLU)
+__global__ void bn_apply_kernel(const float* __restrict__ x,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ invstd,
+                                const float* __restrict__ gamma,
+                                const float* __restrict__ beta,
+                                float* __restrict__ y, long n, int h,
+                                int relu) {
+  const long numel = n * h;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i0; t < numel; t += stride) {
+    const int c = (int)(t % h);
+    float v = (x[t] - mean[c]) * invstd[c] * gamma[c] + beta[c];
+    if (relu) v = fmaxf(v, 0.f);
+    y[t] = v;
+  }
+}
+
+// backward stage A: per-channel sums of gm and gm*xhat (gm = relu-masked g)
+__global__ void bn_bwd_partial_kernel(const float* __restrict__ g,
+                                      const float* __restrict__ x,
+                                      const float* __restrict__ y,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ invstd, long n,
+                                      int h, int relu,
+                                      float* __restrict__ partials) {
+  extern __shared__ float smem[];  // [2*h]
+  for (int c = threadIdx.x; c < 2 * h; c += blockDim.x) smem[c] = 0.f;
+  __syncthreads();
+  const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * rows_per_block;
+  const long r1 = min(n, r0 + rows_per_block);
+  for (long r = r0; r < r1; ++r) {
+    for (int c = threadIdx.x; c < h; c += blockDim.x) {
+      float gm = g[r * h + c];
+      if (relu && y[r * h + c] <= 0.f) gm = 0.f;
+      const float xhat = (x[r * h + c] - mean[c]) * invstd[c];
+      atomicAdd(&smem[c], gm);
+      atomicAdd(&smem[h + c], gm * xhat);
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < 2 * h; c += blockDim.x)
+    if (smem[c] != 0.f) atomicAdd(&partials[c], smem[c]);
+}
+
+// backward stage B: dx = gamma*invstd*(gm - sum_gm/n - xhat*sum_gmx/n)
+__global__ void bn_bwd_apply_kernel(
+    const float* __restrict__ g, const float* __restrict__ x,
+    const float* __restrict__ y, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ partials, float* __restrict__ dx, long n, int h,
+    int relu) {
+  const long numel = n * h;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const float invn = 1.f / n;
+  for (long t = i0; t < numel; t += stride) {
+    const int c = (int)(t % h);
+    float gm = g[t];
+    if (relu && y[t] <= 0.f) gm = 0.f;
+    const float xhat = (x[t] - mean[c]) * invstd[c];
+    dx[t] = gamma[c] * invstd[c] *
+            (gm - partials[c] * invn - xhat * partials[h + c] * invn);
+  }
+}
+
+// dgamma[c] = sum gm*xhat = partials[h+c]; dbeta[c] = sum gm = partials[c]
+__global__ void bn_grad_affine_kernel(const float* __restrict__ partials,
+                                      float* __restrict__ dgamma,
+                                      float* __restrict__ dbeta, int h) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= h) return;
+  dgamma[c] = partials[h + c];
+  dbeta[c] = partials[c];
+}
+
+__global__ void bn_eval_stats_kernel(const float* __restrict__ rm,
+                                     const float* __restrict__ rv,
+                                     float* __restrict__ mean,
+                                     float* __restrict__ invstd, int h,
+                                     float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= h) return;
+  mean[c] = rm[c];
+  invstd[c] = rsqrtf(rv[c] + eps);
+}
+
+void launch_bn_eval_stats(const float* running_mean, const float* running_var,
+                          float* mean, float* invstd, int h, float eps,
+                          hipStream_t s) {
+  bn_eval_stats_kernel<<<ceil_div(h, 256), 256, 0, s>>>(running_mean,
+                                                        running_var, mean,
+                                                        invstd, h, eps);
+}
+
+void launch_bn_fwd(const float* x, const float* gamma, const float* beta,
+                   float* running_mean, float* running_var, float* mean,
+                   float* invstd, float* partials, float* y, long n, int h,
+                   float momentum, float eps, bool training, bool relu,
+                   hipStream_t s) {
+  if (n == 0) return;
+  if (training) {
+    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+    const int nblocks = (int)min((long)512, (n + 63) / 64);
+    bn_stats_partial_kernel<<<nblocks, 256, 2 * h * sizeof(float), s>>>(
+        x, n, h, partials);
+    bn_finalize_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
+        partials, n, h, eps, momentum, mean, invstd, running_mean, running_var,
+        1);
+  } else {
+    launch_bn_eval_stats(running_mean, running_var, mean, invstd, h, eps, s);
+  }
+  bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma, beta,
+                                                  y, n, h, relu ? 1 : 0);
+}
+
+void launch_bn_bwd(const float* g, const float* x, const float* y,
+                   const float* mean, const float* invstd, const float* gamma,
+                   float* partials, float* dx, float* dgamma, float* dbeta,
+                   long n, int h, bool relu, hipStream_t s) {
+  if (n == 0) return;
+  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+  const int nblocks = (int)min((long)512, (n + 63) / 64);
+  bn_bwd_partial_kernel<<<nblocks, 256, 2 * h * sizeof(float), s>>>(
+      g, x, y, mean, invstd, n, h, relu ? 1 : 0, partials);
+  bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
+      g, x, y, mean, invstd, gamma, partials, dx, n, h, relu ? 1 : 0);
+  bn_grad_affine_kernel<<<ceil_div(h, 256), 256, 0, s>>>(partials, dgamma,
+                                                         dbeta, h);
+}
+
+// ---------------------------------------------------------------------------
+// quantile loss + eval metrics
+// ---------------------------------------------------------------------------
+
+__global__ void quantile_loss_fwd_kernel(const float* __restrict__ y,
+                                         const float* __restrict__ y_hat,
+                                         float* __restrict__ out, long b,
+                                         float tau) {
+  __shared__ float smem[256];
+  float acc = 0.f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < b;
+       i += (long)gridDim.x * blockDim.x) {
+    const float e = y[i] - y_hat[i];
+    acc += fmaxf(tau * e, (tau - 1.f) * e);
+  }
+  smem[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) smem[threadIdx.x] += smem[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(out, smem[0] / b);
+}
+
+__global__ void quantile_loss_bwd_kernel(const float* __restrict__ g,
+                                         const float* __restrict__ y,
+                                         const float* __restrict__ y_hat,
+                                         float* __restrict__ dy_hat, long b,
+                                         float tau) {
+  const float gs = g[0] / b;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < b;
+       i += (long)gridDim.x * blockDim.x) {
+    const float e = y[i] - y_hat[i];
+    float d;
+    if (e > 0.f) d = -tau;
+    else if (e < 0.f) d = 1.f - tau;
+    else d = 0.5f * (-tau) + 0.5f * (1.f - tau);  // torch.maximum tie split
+    dy_hat[i] = gs * d;
+  }
+}
+
+__global__ void eval_metrics_kernel(const float* __restrict__ y,
+                                    const float* __restrict__ y_hat,
+                                    float* __restrict__ out3, long b,
+                                    float tau) {
+  __shared__ float smem[3][256];
+  float mae = 0.f, mape = 0.f, q = 0.f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < b;
+       i += (long)gridDim.x * blockDim.x) {
+    const float err = y_hat[i] - y[i];
+    const float a = fabsf(err);
+    mae += a;
+    mape += a / y[i];
+    const float e = -err;
+    q += fmaxf(tau * e, (tau - 1.f) * e);
+  }
+  smem[0][threadIdx.x] = mae;
+  smem[1][threadIdx.x] = mape;
+  smem[2][threadIdx.x] = q;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off)
+      for (int k = 0; k < 3; ++k)
+        smem[k][threadIdx.x] += smem[k][threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0)
+    for (int k = 0; k < 3; ++k) atomicAdd(&out3[k], smem[k][0]);
+}
+
+void launch_quantile_loss_fwd(const float* y, const float* y_hat, float* out,
+                              long b, float tau, hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(out, 0, sizeof(float), s));
+  quantile_loss_fwd_kernel<<<grid_for(b, 256, 256), 256, 0, s>>>(y, y_hat, out,
+                                                                 b, tau);
+}
+void launch_quantile_loss_bwd(const float* g, const float* y,
+                              const float* y_hat, float* dy_hat, long b,
+                              float tau, hipStream_t s) {
+  quantile_loss_bwd_kernel<<<grid_for(b), 256, 0, s>>>(g, y, y_hat, dy_hat, b,
+                                                       tau);
+}
+void launch_eval_metrics(const float* y, const float* y_hat, float* out3,
+                         long b, float tau, hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(out3, 0, 3 * sizeof(float), s));
+  eval_metrics_kernel<<<grid_for(b, 256, 256), 256, 0, s>>>(y, y_hat, out3, b,
+                                                            tau);
+}
+
+// ---------------------------------------------------------------------------
+// Adam on flat master buffers (torch.optim.Adam math, K14)
+// ---------------------------------------------------------------------------
+
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            long numel, float lr, float b1, float b2,
+                            float eps, float bias1, float bias2) {
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const float step_size = lr / bias1;
+  for (long t = i0; t < numel; t += stride) {
+    const float gt = g[t];
+    const float mt = b1 * m[t] + (1.f - b1) * gt;
+    const float vt = b2 * v[t] + (1.f - b2) * gt * gt;
+    m[t] = mt;
+    v[t] = vt;
+    const float denom = sqrtf(vt) / sqrtf(bias2) + eps;
+    p[t] -= step_size * mt / denom;
+  }
+}
+
+void launch_adam(float* p, const float* g, float* m, float* v, long numel,
+                 float lr, float b1, float b2, float eps, int step,
+                 hipStream_t s) {
+  if (numel == 0) return;
+  const float bias1 = 1.f - powf(b1, (float)step);
+  const float bias2 = 1.f - powf(b2, (float)step);
+  adam_kernel<<<grid_for(numel), 256, 0, s>>>(p, g, m, v, numel, lr, b1, b2,
+                                              eps, bias1, bias2);
+}

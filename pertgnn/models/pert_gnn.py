@@ -162,7 +162,8 @@ class SAGEDeterministic(nn.Module):
         x = self.convs[-1](x, edge_index, edge_embeds, csr=csr, num_nodes=n)
         local_predict = ops.linear(x, self.local_linear.weight, self.local_linear.bias)
         mean_x = ops.pattern_pool(x, pattern_probs, pattern_num_nodes, batch, num_graphs)
-        global_predict = torch.cat([mean_x, self.entry_embeds(entry_id)], dim=1)
+        entry_vec = ops.embedding(entry_id, self.entry_embeds.weight)
+        global_predict = torch.cat([mean_x, entry_vec], dim=1)
         h = ops.linear(global_predict, self.global_linear1.weight, self.global_linear1.bias)
         global_predict = ops.linear(F.relu(h), self.global_linear2.weight, self.global_linear2.bias)
         # reference comment says "ensure non-negative" but applies no clamp

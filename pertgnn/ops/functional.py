@@ -59,7 +59,11 @@ class _PatternPoolFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, probs, nnodes, batch, num_graphs):
         m = ext()
-        out = m.seg_pool_fwd(x, probs, nnodes, batch, num_graphs)
+        # nodes are contiguous per graph (collation order) -> batch_ptr
+        counts = torch.bincount(batch, minlength=num_graphs)
+        batch_ptr = torch.zeros(num_graphs + 1, dtype=torch.int32, device=x.device)
+        batch_ptr[1:] = torch.cumsum(counts, 0).to(torch.int32)
+        out = m.seg_pool_fwd(x, probs.contiguous(), nnodes.contiguous(), batch_ptr, num_graphs)
         ctx.save_for_backward(probs, nnodes, batch)
         return out
 
@@ -129,6 +133,33 @@ def embed_concat_edge(edge_attr, interface_table, rpctype_table):
     if use_hip(interface_table):
         return _EmbedEdgeFn.apply(edge_attr, interface_table, rpctype_table)
     return ref.embed_concat_edge(edge_attr, interface_table, rpctype_table)
+
+
+# ---------------------------------------------------------------------------
+# row-gather embedding (entry_embeds, K1)
+# ---------------------------------------------------------------------------
+
+class _EmbeddingFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, idx, table):
+        m = ext()
+        out = m.gather_rows(idx, table)
+        ctx.save_for_backward(idx)
+        ctx.rows = table.shape[0]
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        (idx,) = ctx.saved_tensors
+        m = ext()
+        return None, m.scatter_add_rows(g.contiguous(), idx, ctx.rows)
+
+
+def embedding(idx, table):
+    """out[i] = table[idx[i]] — row gather with scatter-add backward."""
+    if use_hip(table):
+        return _EmbeddingFn.apply(idx.contiguous(), table)
+    return table.index_select(0, idx)
 
 
 # ---------------------------------------------------------------------------
@@ -230,6 +261,7 @@ def linear(x, w, b=None):
 
 __all__ = [
     "edge_attention",
+    "embedding",
     "pattern_pool",
     "embed_concat_node",
     "embed_concat_edge",

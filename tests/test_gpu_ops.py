@@ -1,0 +1,259 @@
+"""GPU parity tests: every HIP kernel against the fp32 eager oracle.
+
+Run on an MI355X box:  python -m pytest tests -m gpu -x -q
+"""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from pertgnn.data.collate import build_csr
+    from pertgnn.ops import functional as F
+    from pertgnn.ops import reference as ref
+    from pertgnn.ops.backend import require_ext
+else:
+    pytest.skip("no GPU available", allow_module_level=True)
+
+
+DEV = torch.device("cuda:0")
+
+
+def _graph(n, e, h, seed=0, skewed=False):
+    g = torch.Generator().manual_seed(seed)
+    src = torch.randint(0, n, (e,), generator=g)
+    if skewed:  # hub: half the edges into node 0 (PERT root fan-in pattern)
+        dst = torch.cat([torch.zeros(e // 2, dtype=torch.long),
+                         torch.randint(0, n, (e - e // 2,), generator=g)])
+    else:
+        dst = torch.randint(0, n, (e,), generator=g)
+    edge_index = torch.stack([src, dst])
+    perm, row_ptr, csr_src, col_ptr, csc_dst, csc_eid = build_csr(edge_index, n)
+    edge_index = edge_index.index_select(1, perm)
+    q = torch.randn(n, h, generator=g)
+    k = torch.randn(n, h, generator=g)
+    v = torch.randn(n, h, generator=g)
+    ee = torch.randn(e, h, generator=g)
+    skip = torch.randn(n, h, generator=g)
+    return edge_index, (row_ptr, csr_src, col_ptr, csc_dst, csc_eid), q, k, v, ee, skip
+
+
+@pytest.mark.parametrize("n,e,h,skewed", [
+    (50, 200, 32, False),
+    (200, 1000, 64, False),
+    (333, 2000, 256, True),
+    (100, 700, 512, False),
+    (10, 0, 64, False),      # empty edge set
+    (1, 1, 128, False),      # single node self-edge
+])
+def test_edge_attention_parity(n, e, h, skewed):
+    require_ext()
+    edge_index, csr, q, k, v, ee, skip = _graph(n, e, h, seed=1, skewed=skewed)
+
+    def run(device, use_csr):
+        args = [t.to(device).requires_grad_(True) for t in (q, k, v, ee, skip)]
+        out = F.edge_attention(args[0], args[1], args[2], args[3], args[4],
+                               edge_index.to(device), n,
+                               csr=tuple(t.to(device) for t in csr) if use_csr else None)
+        loss = (out * torch.arange(out.numel(), device=device, dtype=torch.float32
+                                   ).view_as(out).sin()).sum()
+        loss.backward()
+        return out.detach().cpu(), [a.grad.cpu() for a in args]
+
+    out_cpu, grads_cpu = run(torch.device("cpu"), False)
+    out_gpu, grads_gpu = run(DEV, True)
+    assert torch.allclose(out_gpu, out_cpu, atol=2e-4, rtol=1e-4), \
+        (out_gpu - out_cpu).abs().max()
+    for gg, gc, name in zip(grads_gpu, grads_cpu, "qkves"):
+        assert torch.allclose(gg, gc, atol=5e-4, rtol=1e-3), \
+            (name, (gg - gc).abs().max())
+
+
+def test_pattern_pool_parity():
+    require_ext()
+    g = torch.Generator().manual_seed(3)
+    n, h, b = 500, 256, 17
+    x = torch.randn(n, h, generator=g)
+    probs = torch.rand(n, 1, generator=g)
+    nn_ = torch.randint(1, 9, (n, 1), generator=g).float()
+    batch = torch.sort(torch.randint(0, b, (n,), generator=g)).values
+
+    def run(device):
+        xx = x.to(device).requires_grad_(True)
+        out = F.pattern_pool(xx, probs.to(device), nn_.to(device), batch.to(device), b)
+        out.sum().backward()
+        return out.detach().cpu(), xx.grad.cpu()
+
+    oc, gc = run(torch.device("cpu"))
+    og, gg = run(DEV)
+    assert torch.allclose(og, oc, atol=1e-4)
+    assert torch.allclose(gg, gc, atol=1e-5)
+
+
+def test_embed_ops_parity():
+    require_ext()
+    g = torch.Generator().manual_seed(5)
+    n, f, h, e = 300, 9, 128, 800
+    x_raw = torch.randn(n, f, generator=g)
+    cat = torch.randint(0, 40, (n, 1), generator=g)
+    table = torch.randn(40, h, generator=g)
+    attr = torch.stack([torch.randint(0, 30, (e,), generator=g),
+                        torch.randint(0, 7, (e,), generator=g)], dim=1)
+    ifc = torch.randn(30, h, generator=g)
+    rpc = torch.randn(7, h, generator=g)
+
+    def run(device):
+        t = table.to(device).requires_grad_(True)
+        i = ifc.to(device).requires_grad_(True)
+        r = rpc.to(device).requires_grad_(True)
+        o1 = F.embed_concat_node(x_raw.to(device), cat.to(device), [t])
+        o2 = F.embed_concat_edge(attr.to(device), i, r)
+        (o1.pow(2).sum() + o2.pow(2).sum()).backward()
+        return o1.detach().cpu(), o2.detach().cpu(), t.grad.cpu(), i.grad.cpu(), r.grad.cpu()
+
+    rc = run(torch.device("cpu"))
+    rg = run(DEV)
+    for a, b_, tol in zip(rg, rc, [1e-5, 1e-5, 1e-3, 1e-3, 1e-3]):
+        assert torch.allclose(a, b_, atol=tol, rtol=1e-4), (a - b_).abs().max()
+
+
+def test_embedding_gather_parity():
+    require_ext()
+    g = torch.Generator().manual_seed(6)
+    table = torch.randn(25, 64, generator=g)
+    idx = torch.randint(0, 25, (100,), generator=g)
+
+    t = table.to(DEV).requires_grad_(True)
+    out = F.embedding(idx.to(DEV), t)
+    out.sum().backward()
+    t2 = table.clone().requires_grad_(True)
+    out2 = t2.index_select(0, idx)
+    out2.sum().backward()
+    assert torch.allclose(out.cpu(), out2.detach(), atol=1e-6)
+    assert torch.allclose(t.grad.cpu(), t2.grad, atol=1e-4)
+
+
+@pytest.mark.parametrize("training,relu", [(True, True), (True, False), (False, True)])
+def test_batchnorm_relu_parity(training, relu):
+    require_ext()
+    g = torch.Generator().manual_seed(7)
+    n, h = 400, 256
+    x = torch.randn(n, h, generator=g) * 3 + 1
+    gamma = torch.rand(h, generator=g) + 0.5
+    beta = torch.randn(h, generator=g)
+    rm = torch.randn(h, generator=g)
+    rv = torch.rand(h, generator=g) + 0.5
+
+    def run(device, force_eager):
+        import os
+        if force_eager:
+            os.environ["PERTGNN_FORCE_EAGER"] = "1"
+        try:
+            xx = x.to(device).requires_grad_(True)
+            ga = gamma.to(device).requires_grad_(True)
+            be = beta.to(device).requires_grad_(True)
+            rmm = rm.to(device).clone()
+            rvv = rv.to(device).clone()
+            y = F.batchnorm_relu(xx, ga, be, rmm, rvv, 0.1, 1e-5, training, fuse_relu=relu)
+            if training:
+                y.pow(2).sum().backward()
+                return (y.detach().cpu(), rmm.cpu(), rvv.cpu(),
+                        xx.grad.cpu(), ga.grad.cpu(), be.grad.cpu())
+            return (y.detach().cpu(), rmm.cpu(), rvv.cpu(), None, None, None)
+        finally:
+            if force_eager:
+                os.environ.pop("PERTGNN_FORCE_EAGER", None)
+
+    rc = run(torch.device("cpu"), False)
+    rg = run(DEV, False)
+    for a, b_ in zip(rg, rc):
+        if a is None:
+            continue
+        assert torch.allclose(a, b_, atol=5e-3, rtol=1e-3), (a - b_).abs().max()
+
+
+def test_quantile_loss_parity():
+    require_ext()
+    g = torch.Generator().manual_seed(8)
+    b = 513
+    y = torch.rand(b, generator=g) * 100
+    y_hat = torch.rand(b, generator=g) * 100
+
+    yh_c = y_hat.clone().requires_grad_(True)
+    lc = ref.quantile_loss(y, yh_c, 0.7)
+    lc.backward()
+    yh_g = y_hat.to(DEV).requires_grad_(True)
+    lg = F.quantile_loss(y.to(DEV), yh_g, 0.7)
+    lg.backward()
+    assert torch.allclose(lg.cpu(), lc.detach(), atol=1e-4)
+    assert torch.allclose(yh_g.grad.cpu(), yh_c.grad, atol=1e-6)
+
+    m_g = F.eval_metrics(y.to(DEV), y_hat.to(DEV), 0.7)
+    m_c = ref.eval_metrics(y, y_hat, 0.7)
+    for a, b_ in zip(m_g, m_c):
+        assert torch.allclose(a.cpu(), b_, rtol=1e-4), (a, b_)
+
+
+def test_adam_parity():
+    require_ext()
+    import pertgnn._C as C
+    g = torch.Generator().manual_seed(9)
+    numel = 10000
+    p0 = torch.randn(numel, generator=g)
+    grad = torch.randn(numel, generator=g)
+
+    # torch reference
+    p_t = p0.clone().requires_grad_(True)
+    opt = torch.optim.Adam([p_t], lr=1e-3)
+    for step in range(3):
+        p_t.grad = grad.clone()
+        opt.step()
+
+    # HIP kernel
+    p_h = p0.clone().to(DEV)
+    m = torch.zeros(numel, device=DEV)
+    v = torch.zeros(numel, device=DEV)
+    for step in range(1, 4):
+        C.adam_step(p_h, grad.to(DEV), m, v, 1e-3, 0.9, 0.999, 1e-8, step)
+    torch.cuda.synchronize()
+    assert torch.allclose(p_h.cpu(), p_t.detach(), atol=1e-6), \
+        (p_h.cpu() - p_t.detach()).abs().max()
+
+
+def test_model_end_to_end_gpu_vs_cpu():
+    require_ext()
+    from pertgnn.models import SAGEDeterministic
+    from pertgnn.data.collate import collate
+    import bench as bench_mod
+
+    torch.manual_seed(0)
+    batches, stats = bench_mod.build_synthetic_batches(1, 16, seed=0, device=torch.device("cpu"))
+    b = batches[0]
+    model = SAGEDeterministic(9, [stats["cat_max"] + 1], stats["entry_max"],
+                              stats["ifc_max"], stats["rpc_max"], 64, 3, 0.0)
+    model.train()
+
+    def run(model, b, device):
+        m = model.to(device)
+        bb = b.to(device)
+        gp, lp = m(bb.x, bb.cat_X, bb.edge_index, bb.edge_attr,
+                   bb.pattern_num_nodes, bb.rt_probs, bb.entry_id, bb.batch,
+                   csr=bb.csr, num_graphs=bb.num_graphs)
+        loss = F.quantile_loss(bb.y, gp.flatten(), 0.5)
+        loss.backward()
+        grads = {n: p.grad.cpu().clone() for n, p in m.named_parameters() if p.grad is not None}
+        m.zero_grad()
+        return gp.detach().cpu(), float(loss), grads
+
+    import copy
+    model_cpu = copy.deepcopy(model)
+    gp_c, loss_c, grads_c = run(model_cpu, b, torch.device("cpu"))
+    gp_g, loss_g, grads_g = run(model, b, DEV)
+    assert torch.allclose(gp_g, gp_c, atol=2e-3, rtol=1e-3), (gp_g - gp_c).abs().max()
+    assert abs(loss_g - loss_c) < 2e-3
+    for n in grads_c:
+        assert n in grads_g, n
+        assert torch.allclose(grads_g[n], grads_c[n], atol=5e-3, rtol=5e-3), \
+            (n, (grads_g[n] - grads_c[n]).abs().max())
