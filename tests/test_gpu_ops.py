@@ -53,7 +53,7 @@ def test_edge_attention_parity(n, e, h, skewed):
     edge_index, csr, q, k, v, ee, skip = _graph(n, e, h, seed=1, skewed=skewed)
 
     def run(device, use_csr):
-        args = [t.to(device).requires_grad_(True) for t in (q, k, v, ee, skip)]
+        args = [t.detach().to(device).requires_grad_(True) for t in (q, k, v, ee, skip)]
         out = F.edge_attention(args[0], args[1], args[2], args[3], args[4],
                                edge_index.to(device), n,
                                csr=tuple(t.to(device) for t in csr) if use_csr else None)
@@ -81,7 +81,7 @@ def test_pattern_pool_parity():
     batch = torch.sort(torch.randint(0, b, (n,), generator=g)).values
 
     def run(device):
-        xx = x.to(device).requires_grad_(True)
+        xx = x.detach().to(device).requires_grad_(True)
         out = F.pattern_pool(xx, probs.to(device), nn_.to(device), batch.to(device), b)
         out.sum().backward()
         return out.detach().cpu(), xx.grad.cpu()
@@ -105,9 +105,9 @@ def test_embed_ops_parity():
     rpc = torch.randn(7, h, generator=g)
 
     def run(device):
-        t = table.to(device).requires_grad_(True)
-        i = ifc.to(device).requires_grad_(True)
-        r = rpc.to(device).requires_grad_(True)
+        t = table.detach().to(device).requires_grad_(True)
+        i = ifc.detach().to(device).requires_grad_(True)
+        r = rpc.detach().to(device).requires_grad_(True)
         o1 = F.embed_concat_node(x_raw.to(device), cat.to(device), [t])
         o2 = F.embed_concat_edge(attr.to(device), i, r)
         (o1.pow(2).sum() + o2.pow(2).sum()).backward()
@@ -151,9 +151,9 @@ def test_batchnorm_relu_parity(training, relu):
         if force_eager:
             os.environ["PERTGNN_FORCE_EAGER"] = "1"
         try:
-            xx = x.to(device).requires_grad_(True)
-            ga = gamma.to(device).requires_grad_(True)
-            be = beta.to(device).requires_grad_(True)
+            xx = x.detach().to(device).requires_grad_(True)
+            ga = gamma.detach().to(device).requires_grad_(True)
+            be = beta.detach().to(device).requires_grad_(True)
             rmm = rm.to(device).clone()
             rvv = rv.to(device).clone()
             y = F.batchnorm_relu(xx, ga, be, rmm, rvv, 0.1, 1e-5, training, fuse_relu=relu)
