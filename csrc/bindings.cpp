@@ -55,6 +55,13 @@ void launch_eval_metrics(const float*, const float*, float*, long, float,
                          hipStream_t);
 void launch_adam(float*, const float*, float*, float*, long, float, float,
                  float, float, int, hipStream_t);
+void launch_gemm_f32_nt(const float*, const float*, const float*, float*, int,
+                        int, int, bool, hipStream_t);
+void launch_gemm_f32_nn(const float*, const float*, const float*, float*, int,
+                        int, int, bool, hipStream_t);
+void launch_gemm_f32_tn(const float*, const float*, float*, int, int, int,
+                        hipStream_t);
+void launch_colsum(const float*, float*, long, int, hipStream_t);
 
 // ---------------------------------------------------------------------------
 
@@ -287,7 +294,77 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
               (float)eps, (int)step, cur_stream());
 }
 
+// y = x @ w^T + b (torch Linear layout: w [out,in])
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b) {
+  CHECK_IN(x); CHECK_IN(w);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = w.size(0);
+  TORCH_CHECK(w.size(1) == k, "weight shape mismatch");
+  auto y = torch::empty({m, n}, x.options());
+  const float* bias = nullptr;
+  if (b.defined() && b.numel() > 0) {
+    CHECK_IN(b);
+    bias = b.data_ptr<float>();
+  }
+  launch_gemm_f32_nt(x.data_ptr<float>(), w.data_ptr<float>(), bias,
+                     y.data_ptr<float>(), m, n, k, false, cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> linear_bwd(torch::Tensor g, torch::Tensor x,
+                                      torch::Tensor w, bool has_bias) {
+  CHECK_IN(g); CHECK_IN(x); CHECK_IN(w);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = w.size(0);
+  auto dx = torch::empty({m, k}, x.options());
+  auto dw = torch::empty({n, k}, w.options());
+  launch_gemm_f32_nn(g.data_ptr<float>(), w.data_ptr<float>(), nullptr,
+                     dx.data_ptr<float>(), m, n, k, false, cur_stream());
+  launch_gemm_f32_tn(g.data_ptr<float>(), x.data_ptr<float>(),
+                     dw.data_ptr<float>(), m, n, k, cur_stream());
+  torch::Tensor db = torch::empty({0}, g.options());
+  if (has_bias) {
+    db = torch::empty({n}, g.options());
+    launch_colsum(g.data_ptr<float>(), db.data_ptr<float>(), m, n,
+                  cur_stream());
+  }
+  return {dx, dw, db};
+}
+
+// raw GEMM entry points (tests / future fused paths)
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
+  CHECK_IN(a); CHECK_IN(b);
+  auto c = torch::empty({a.size(0), b.size(0)}, a.options());
+  launch_gemm_f32_nt(a.data_ptr<float>(), b.data_ptr<float>(), nullptr,
+                     c.data_ptr<float>(), a.size(0), b.size(0), a.size(1),
+                     false, cur_stream());
+  return c;
+}
+torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b) {
+  CHECK_IN(a); CHECK_IN(b);
+  auto c = torch::empty({a.size(0), b.size(1)}, a.options());
+  launch_gemm_f32_nn(a.data_ptr<float>(), b.data_ptr<float>(), nullptr,
+                     c.data_ptr<float>(), a.size(0), a.size(1), b.size(1),
+                     false, cur_stream());
+  return c;
+}
+torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b) {
+  CHECK_IN(a); CHECK_IN(b);
+  auto c = torch::empty({a.size(1), b.size(1)}, a.options());
+  launch_gemm_f32_tn(a.data_ptr<float>(), b.data_ptr<float>(),
+                     c.data_ptr<float>(), a.size(0), a.size(1), b.size(1),
+                     cur_stream());
+  return c;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("linear_fwd", &linear_fwd);
+  mod.def("linear_bwd", &linear_bwd);
+  mod.def("gemm_nt", &gemm_nt);
+  mod.def("gemm_nn", &gemm_nn);
+  mod.def("gemm_tn", &gemm_tn);
   mod.def("edge_attn_fwd", &edge_attn_fwd);
   mod.def("edge_attn_bwd", &edge_attn_bwd);
   mod.def("seg_pool_fwd", &seg_pool_fwd);

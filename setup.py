@@ -24,6 +24,7 @@ setup(
                 "csrc/bindings.cpp",
                 "csrc/hip/edge_attn.hip",
                 "csrc/hip/segops.hip",
+                "csrc/hip/gemm.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

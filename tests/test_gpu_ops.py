@@ -257,3 +257,64 @@ def test_model_end_to_end_gpu_vs_cpu():
         assert n in grads_g, n
         assert torch.allclose(grads_g[n], grads_c[n], atol=5e-3, rtol=5e-3), \
             (n, (grads_g[n] - grads_c[n]).abs().max())
+
+
+@pytest.mark.parametrize("m,n,k", [
+    (1000, 256, 265),   # layer-1 shape (9+H), K not multiple of 32
+    (777, 512, 512),
+    (65, 1, 256),       # head linear
+    (4096, 256, 256),
+    (17, 64, 128),
+])
+def test_gemm_parity(m, n, k):
+    require_ext()
+    import pertgnn._C as C
+    g = torch.Generator().manual_seed(11)
+    a = torch.randn(m, k, generator=g).to(DEV)
+    w = torch.randn(n, k, generator=g).to(DEV)
+    # NT: a @ w^T
+    ref_nt = a @ w.t()
+    out_nt = C.gemm_nt(a, w)
+    torch.cuda.synchronize()
+    assert torch.allclose(out_nt, ref_nt, atol=1e-3, rtol=1e-4), \
+        (out_nt - ref_nt).abs().max()
+    # NN: gout @ w  (gout [m,n], w [n,k])
+    gout = torch.randn(m, n, generator=torch.Generator().manual_seed(2)).to(DEV)
+    ref_nn = gout @ w
+    out_nn = C.gemm_nn(gout, w)
+    torch.cuda.synchronize()
+    assert torch.allclose(out_nn, ref_nn, atol=1e-3, rtol=1e-4), \
+        (out_nn - ref_nn).abs().max()
+    # TN: gout^T @ a
+    ref_tn = gout.t() @ a
+    out_tn = C.gemm_tn(gout, a)
+    torch.cuda.synchronize()
+    assert torch.allclose(out_tn, ref_tn, atol=5e-2, rtol=1e-3), \
+        (out_tn - ref_tn).abs().max()
+
+
+def test_linear_parity():
+    require_ext()
+    g = torch.Generator().manual_seed(13)
+    m, k, n = 2000, 265, 256
+    x = torch.randn(m, k, generator=g)
+    w = torch.randn(n, k, generator=g) * 0.05
+    b = torch.randn(n, generator=g)
+
+    xx = x.detach().to(DEV).requires_grad_(True)
+    ww = w.detach().to(DEV).requires_grad_(True)
+    bb = b.detach().to(DEV).requires_grad_(True)
+    y = F.linear(xx, ww, bb)
+    y.sin().sum().backward()
+
+    x2 = x.clone().requires_grad_(True)
+    w2 = w.clone().requires_grad_(True)
+    b2 = b.clone().requires_grad_(True)
+    y2 = torch.nn.functional.linear(x2, w2, b2)
+    y2.sin().sum().backward()
+
+    assert torch.allclose(y.detach().cpu(), y2.detach(), atol=1e-3, rtol=1e-4)
+    assert torch.allclose(xx.grad.cpu(), x2.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(ww.grad.cpu(), w2.grad, atol=5e-2, rtol=1e-3), \
+        (ww.grad.cpu() - w2.grad).abs().max()
+    assert torch.allclose(bb.grad.cpu(), b2.grad, atol=1e-2, rtol=1e-3)
