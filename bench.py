@@ -108,7 +108,8 @@ def main():
 
     from pertgnn.models import SAGEDeterministic
     from pertgnn.ops import functional as F
-    from pertgnn.parallel import Comm, GradBucketAllReduce
+    from pertgnn.parallel import Comm
+    from pertgnn.train.optim import FlatGradAllReduce, FusedAdam
 
     comm = Comm()
     n_gpus = max(args.gpus, comm.world_size)
@@ -125,8 +126,8 @@ def main():
         stats["rpc_max"], args.hidden, args.layers, 0.0,
     ).to(device)
     comm.broadcast_module_(model)
-    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr)
-    engine = GradBucketAllReduce(model, comm) if comm.distributed else None
+    optimizer = FusedAdam(model.parameters(), lr=args.lr)
+    engine = FlatGradAllReduce(optimizer, comm) if comm.distributed else None
 
     model.train()
 

@@ -59,9 +59,11 @@ void launch_gemm_f32_nt(const float*, const float*, const float*, float*, int,
                         int, int, bool, hipStream_t);
 void launch_gemm_f32_nn(const float*, const float*, const float*, float*, int,
                         int, int, bool, hipStream_t);
-void launch_gemm_f32_tn(const float*, const float*, float*, int, int, int,
-                        hipStream_t);
+void launch_gemm_f32_tn(const float*, const float*, float*, float*, int, int,
+                        int, hipStream_t);
 void launch_colsum(const float*, float*, long, int, hipStream_t);
+void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
+                                  int, int, int, int, hipStream_t);
 
 // ---------------------------------------------------------------------------
 
@@ -320,16 +322,16 @@ std::vector<torch::Tensor> linear_bwd(torch::Tensor g, torch::Tensor x,
   const int n = w.size(0);
   auto dx = torch::empty({m, k}, x.options());
   auto dw = torch::empty({n, k}, w.options());
+  torch::Tensor db = torch::empty({0}, g.options());
+  float* db_ptr = nullptr;
+  if (has_bias) {
+    db = torch::empty({n}, g.options());
+    db_ptr = db.data_ptr<float>();
+  }
   launch_gemm_f32_nn(g.data_ptr<float>(), w.data_ptr<float>(), nullptr,
                      dx.data_ptr<float>(), m, n, k, false, cur_stream());
   launch_gemm_f32_tn(g.data_ptr<float>(), x.data_ptr<float>(),
-                     dw.data_ptr<float>(), m, n, k, cur_stream());
-  torch::Tensor db = torch::empty({0}, g.options());
-  if (has_bias) {
-    db = torch::empty({n}, g.options());
-    launch_colsum(g.data_ptr<float>(), db.data_ptr<float>(), m, n,
-                  cur_stream());
-  }
+                     dw.data_ptr<float>(), db_ptr, m, n, k, cur_stream());
   return {dx, dw, db};
 }
 
@@ -354,12 +356,27 @@ torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b) {
   CHECK_IN(a); CHECK_IN(b);
   auto c = torch::empty({a.size(1), b.size(1)}, a.options());
   launch_gemm_f32_tn(a.data_ptr<float>(), b.data_ptr<float>(),
-                     c.data_ptr<float>(), a.size(0), a.size(1), b.size(1),
-                     cur_stream());
+                     c.data_ptr<float>(), nullptr, a.size(0), a.size(1),
+                     b.size(1), cur_stream());
   return c;
 }
 
+// dtable[r] = sum over g rows whose table index is r; order/ptr group the
+// g-row indices by table row (deterministic, no atomics).
+torch::Tensor embed_grouped_scatter(torch::Tensor g, torch::Tensor order,
+                                    torch::Tensor ptr, int64_t rows,
+                                    int64_t h, int64_t col_off) {
+  CHECK_IN(g); CHECK_IN(order); CHECK_IN(ptr);
+  auto dtable = torch::empty({rows, h}, g.options());
+  launch_embed_grouped_scatter(g.data_ptr<float>(), order.data_ptr<int>(),
+                               ptr.data_ptr<int>(), dtable.data_ptr<float>(),
+                               (int)rows, (int)h, (int)g.size(1), (int)col_off,
+                               cur_stream());
+  return dtable;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("embed_grouped_scatter", &embed_grouped_scatter);
   mod.def("linear_fwd", &linear_fwd);
   mod.def("linear_bwd", &linear_bwd);
   mod.def("gemm_nt", &gemm_nt);

@@ -32,13 +32,27 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 // ---------------------------------------------------------------------------
 
 // operand stored [free][contract] in global (row stride = ld): transpose-stage.
-// thread t loads a float4 along contract; zero-fill outside bounds.
+// thread t loads a float4 along contract; zero-fill outside bounds.  Interior
+// tiles with 16B-alignable rows (ld % 4 == 0) take the vectorized fast path.
 __device__ __forceinline__ void stage_transpose(
     const float* __restrict__ g, long ld, int free0, int contract0,
     int free_max, int contract_max, float* lds /* [GEMM_BK][GEMM_BM] */) {
   const int t = threadIdx.x;
-  const int fr = t / 8;           // 0..31 ? 256/8 = 32 rows per pass; need 64
+  const int fr = t / 8;           // 32 rows per pass; two passes cover 64
   const int cq = (t % 8) * 4;     // contract quad
+  const bool fast = (free0 + GEMM_BM <= free_max) &&
+                    (contract0 + GEMM_BK <= contract_max) && ((ld & 3) == 0);
+  if (fast) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int f = fr + half * 32;
+      const f32x4 v = *reinterpret_cast<const f32x4*>(
+          &g[(long)(free0 + f) * ld + contract0 + cq]);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) lds[(cq + u) * GEMM_BM + f] = v[u];
+    }
+    return;
+  }
 #pragma unroll
   for (int half = 0; half < 2; ++half) {
     const int f = fr + half * 32;
@@ -59,8 +73,21 @@ __device__ __forceinline__ void stage_direct(
     const float* __restrict__ g, long ld, int contract0, int free0,
     int contract_max, int free_max, float* lds /* [GEMM_BK][GEMM_BM] */) {
   const int t = threadIdx.x;
-  const int c = t / 16;            // 0..15; two passes cover 32
+  const int c = t / 16;            // two passes cover 32 contract rows
   const int fq = (t % 16) * 4;     // free quad
+  const bool fast = (contract0 + GEMM_BK <= contract_max) &&
+                    (free0 + GEMM_BM <= free_max) && ((ld & 3) == 0) &&
+                    ((free0 & 3) == 0);
+  if (fast) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int cc = c + half * 16;
+      const f32x4 v = *reinterpret_cast<const f32x4*>(
+          &g[(long)(contract0 + cc) * ld + free0 + fq]);
+      *reinterpret_cast<f32x4*>(&lds[cc * GEMM_BM + fq]) = v;
+    }
+    return;
+  }
 #pragma unroll
   for (int half = 0; half < 2; ++half) {
     const int cc = c + half * 16;
@@ -221,8 +248,9 @@ __global__ void gemm_f32_nn_kernel(const float* __restrict__ a,
 __launch_bounds__(GEMM_THREADS)
 __global__ void gemm_f32_tn_kernel(const float* __restrict__ a,
                                    const float* __restrict__ b,
-                                   float* __restrict__ c, int m, int n, int k2,
-                                   int slices) {
+                                   float* __restrict__ c,
+                                   float* __restrict__ dbias, int m, int n,
+                                   int k2, int slices) {
   __shared__ float lds_a[2][GEMM_BK * GEMM_BM];
   __shared__ float lds_b[2][GEMM_BK * GEMM_BM];
   const int tiles_k = (k2 + GEMM_BN - 1) / GEMM_BN;
@@ -246,6 +274,11 @@ __global__ void gemm_f32_tn_kernel(const float* __restrict__ a,
 
   WaveTile wt;
   wt.zero();
+  // bias grad rides along: tile_k==0 blocks column-sum their staged A (=g)
+  // tiles straight from LDS (no extra global pass over g).
+  const bool do_bias = (dbias != nullptr) && (tile_k == 0);
+  float dbsum = 0.f;
+  const int bcol = threadIdx.x;  // threads 0..63 own A-tile columns
   int buf = 0;
   stage_direct(a, n, c_beg, n0, c_end, n, lds_a[0]);
   stage_direct(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
@@ -254,10 +287,18 @@ __global__ void gemm_f32_tn_kernel(const float* __restrict__ a,
     stage_direct(a, n, cc, n0, c_end, n, lds_a[buf ^ 1]);
     stage_direct(b, k2, cc, k0, c_end, k2, lds_b[buf ^ 1]);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+    if (do_bias && bcol < GEMM_BM)
+#pragma unroll
+      for (int r = 0; r < GEMM_BK; ++r) dbsum += lds_a[buf][r * GEMM_BM + bcol];
     __syncthreads();
     buf ^= 1;
   }
   wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+  if (do_bias && bcol < GEMM_BM) {
+#pragma unroll
+    for (int r = 0; r < GEMM_BK; ++r) dbsum += lds_a[buf][r * GEMM_BM + bcol];
+    if (n0 + bcol < n) atomicAdd(&dbias[n0 + bcol], dbsum);
+  }
 
   // accumulate into C with atomics (one slice may be the only writer)
   const int fcol = lane & 15;
@@ -318,8 +359,8 @@ void launch_gemm_f32_nn(const float* a, const float* b, const float* bias,
       a, b, bias, c, m, n, k2, relu ? 1 : 0);
 }
 
-void launch_gemm_f32_tn(const float* a, const float* b, float* c, int m, int n,
-                        int k2, hipStream_t s) {
+void launch_gemm_f32_tn(const float* a, const float* b, float* c, float* dbias,
+                        int m, int n, int k2, hipStream_t s) {
   const int tiles_n = (n + GEMM_BM - 1) / GEMM_BM;
   const int tiles_k = (k2 + GEMM_BN - 1) / GEMM_BN;
   const int tiles = tiles_n * tiles_k;
@@ -328,8 +369,9 @@ void launch_gemm_f32_tn(const float* a, const float* b, float* c, int m, int n,
   while (tiles * slices < 512 && slices < 64 &&
          (long)slices * GEMM_BK * 4 < m) slices *= 2;
   if (slices > 1) HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
+  if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
  hipLaunchKernelGGL(( gemm_f32_tn_kernel), dim3(dim3(tiles * slices)), dim3(dim3(GEMM_THREADS)), 0, s, 
-      a, b, c, m, n, k2, slices);
+      a, b, c, dbias, m, n, k2, slices);
 }
 
 void launch_colsum(const float* g, float* out, long m, int n, hipStream_t s) {

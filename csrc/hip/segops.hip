@@ -157,6 +157,56 @@ __global__ void embed_edge_bwd_kernel(const float* __restrict__ g,
   }
 }
 
+// Deterministic grouped embedding backward: rows of g pre-grouped by table
+// index (order/ptr built once per backward from a sort) — one wave per table
+// row segment-sums its gradient rows; no atomics, coalesced columns.
+template <int VPT>
+__global__ void embed_grouped_scatter_kernel(
+    const float* __restrict__ g, const int* __restrict__ order,
+    const int* __restrict__ ptr, float* __restrict__ dtable, int rows, int h,
+    int gstride, int col_off) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (row >= rows) return;
+  float acc[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
+  for (int p = ptr[row]; p < ptr[row + 1]; ++p) {
+    const long r = order[p];
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) acc[j] += g[r * gstride + col_off + c];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    if (c < h) dtable[(long)row * h + c] = acc[j];
+  }
+}
+
+void launch_embed_grouped_scatter(const float* g, const int* order,
+                                  const int* ptr, float* dtable, int rows,
+                                  int h, int gstride, int col_off,
+                                  hipStream_t s) {
+  if (rows == 0) return;
+  const dim3 grid(ceil_div(rows, WAVES_PER_BLOCK));
+  const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
+  const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  switch (vpt) {
+#define CASE(V)                                                                \
+  case V:                                                                      \
+    embed_grouped_scatter_kernel<V><<<grid, block, 0, s>>>(                    \
+        g, order, ptr, dtable, rows, h, gstride, col_off);                     \
+    break;
+    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+#undef CASE
+    default: abort();
+  }
+}
+
 // entry embedding gather: out[b] = table[idx[b]] — plain gather (fwd) +
 // scatter-add (bwd); reuses the edge kernels' grid-stride shape.
 __global__ void gather_rows_kernel(const long* __restrict__ idx,
@@ -232,26 +282,35 @@ void launch_scatter_add_rows(const float* g, const long* idx, float* dtable,
 // BatchNorm1d (+ReLU) — two-stage column reduction, coalesced row tiles
 // ---------------------------------------------------------------------------
 
-// stage A: per-block partial sum/sumsq over a row range, accumulated into
-// global partials[2*h] with one atomicAdd per channel per block.
+// stage A: per-block partial sum/sumsq over a row range; thread t owns
+// channels t and t+256 in REGISTERS (H <= 512), one atomicAdd per channel
+// per block at the end — coalesced loads, no LDS traffic.
 __global__ void bn_stats_partial_kernel(const float* __restrict__ x, long n,
                                         int h, float* __restrict__ partials) {
-  extern __shared__ float smem[];  // [2*h]
-  for (int c = threadIdx.x; c < 2 * h; c += blockDim.x) smem[c] = 0.f;
-  __syncthreads();
+  const int c0 = threadIdx.x;
+  const int c1 = threadIdx.x + 256;
+  float s0 = 0.f, q0 = 0.f, s1 = 0.f, q1 = 0.f;
   const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
   const long r0 = (long)blockIdx.x * rows_per_block;
   const long r1 = min(n, r0 + rows_per_block);
   for (long r = r0; r < r1; ++r) {
-    for (int c = threadIdx.x; c < h; c += blockDim.x) {
-      const float v = x[r * h + c];
-      atomicAdd(&smem[c], v);          // LDS atomics — cheap, per-CU
-      atomicAdd(&smem[h + c], v * v);
+    if (c0 < h) {
+      const float v = x[r * h + c0];
+      s0 += v; q0 += v * v;
+    }
+    if (c1 < h) {
+      const float v = x[r * h + c1];
+      s1 += v; q1 += v * v;
     }
   }
-  __syncthreads();
-  for (int c = threadIdx.x; c < 2 * h; c += blockDim.x)
-    if (smem[c] != 0.f) atomicAdd(&partials[c], smem[c]);
+  if (c0 < h && r0 < r1) {
+    atomicAdd(&partials[c0], s0);
+    atomicAdd(&partials[h + c0], q0);
+  }
+  if (c1 < h && r0 < r1) {
+    atomicAdd(&partials[c1], s1);
+    atomicAdd(&partials[h + c1], q1);
+  }
 }
 
 // stage B: finalize mean/invstd (+ running-stat update, training only)
@@ -302,24 +361,34 @@ __global__ void bn_bwd_partial_kernel(const float* __restrict__ g,
                                       const float* __restrict__ invstd, long n,
                                       int h, int relu,
                                       float* __restrict__ partials) {
-  extern __shared__ float smem[];  // [2*h]
-  for (int c = threadIdx.x; c < 2 * h; c += blockDim.x) smem[c] = 0.f;
-  __syncthreads();
+  const int c0 = threadIdx.x;
+  const int c1 = threadIdx.x + 256;
+  float s0 = 0.f, q0 = 0.f, s1 = 0.f, q1 = 0.f;
   const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
   const long r0 = (long)blockIdx.x * rows_per_block;
   const long r1 = min(n, r0 + rows_per_block);
   for (long r = r0; r < r1; ++r) {
-    for (int c = threadIdx.x; c < h; c += blockDim.x) {
-      float gm = g[r * h + c];
-      if (relu && y[r * h + c] <= 0.f) gm = 0.f;
-      const float xhat = (x[r * h + c] - mean[c]) * invstd[c];
-      atomicAdd(&smem[c], gm);
-      atomicAdd(&smem[h + c], gm * xhat);
+    if (c0 < h) {
+      float gm = g[r * h + c0];
+      if (relu && y[r * h + c0] <= 0.f) gm = 0.f;
+      s0 += gm;
+      q0 += gm * (x[r * h + c0] - mean[c0]) * invstd[c0];
+    }
+    if (c1 < h) {
+      float gm = g[r * h + c1];
+      if (relu && y[r * h + c1] <= 0.f) gm = 0.f;
+      s1 += gm;
+      q1 += gm * (x[r * h + c1] - mean[c1]) * invstd[c1];
     }
   }
-  __syncthreads();
-  for (int c = threadIdx.x; c < 2 * h; c += blockDim.x)
-    if (smem[c] != 0.f) atomicAdd(&partials[c], smem[c]);
+  if (c0 < h && r0 < r1) {
+    atomicAdd(&partials[c0], s0);
+    atomicAdd(&partials[h + c0], q0);
+  }
+  if (c1 < h && r0 < r1) {
+    atomicAdd(&partials[c1], s1);
+    atomicAdd(&partials[h + c1], q1);
+  }
 }
 
 // backward stage B: dx = gamma*invstd*(gm - sum_gm/n - xhat*sum_gmx/n)
@@ -381,7 +450,7 @@ void launch_bn_fwd(const float* x, const float* gamma, const float* beta,
   if (training) {
     HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
     const int nblocks = (int)min((long)512, (n + 63) / 64);
-    bn_stats_partial_kernel<<<nblocks, 256, 2 * h * sizeof(float), s>>>(
+    bn_stats_partial_kernel<<<nblocks, 256, 0, s>>>(
         x, n, h, partials);
     bn_finalize_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
         partials, n, h, eps, momentum, mean, invstd, running_mean, running_var,
@@ -400,7 +469,7 @@ void launch_bn_bwd(const float* g, const float* x, const float* y,
   if (n == 0) return;
   HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
   const int nblocks = (int)min((long)512, (n + 63) / 64);
-  bn_bwd_partial_kernel<<<nblocks, 256, 2 * h * sizeof(float), s>>>(
+  bn_bwd_partial_kernel<<<nblocks, 256, 0, s>>>(
       g, x, y, mean, invstd, n, h, relu ? 1 : 0, partials);
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
       g, x, y, mean, invstd, gamma, partials, dx, n, h, relu ? 1 : 0);

@@ -18,7 +18,8 @@ import torch
 from pertgnn.data.collate import BatchLoader
 from pertgnn.data.dataset import build_data_list, split_60_20_20
 from pertgnn.models import SAGEDeterministic
-from pertgnn.parallel import Comm, GradBucketAllReduce
+from pertgnn.parallel import Comm
+from pertgnn.train.optim import FlatGradAllReduce, FusedAdam
 from pertgnn.train import evaluate, load_checkpoint, save_checkpoint, train_epoch
 from pertgnn.utils import JsonlLogger
 
@@ -127,8 +128,8 @@ def main(argv=None):
         rpctype_id_max, args.hidden_channels, args.num_layers, args.dropout,
     ).to(device)
     comm.broadcast_module_(model)
-    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr)
-    engine = GradBucketAllReduce(model, comm) if comm.distributed else None
+    optimizer = FusedAdam(model.parameters(), lr=args.lr)
+    engine = FlatGradAllReduce(optimizer, comm) if comm.distributed else None
     log = JsonlLogger(args.metrics_jsonl, rank=comm.rank)
 
     start_epoch = 1

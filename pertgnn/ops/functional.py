@@ -16,6 +16,16 @@ from . import reference as ref
 from .backend import ext, use_hip
 
 
+def _group_by(idx: torch.Tensor, rows: int):
+    """Group positions by index value: returns (order int32, ptr int32[rows+1])
+    for the deterministic grouped scatter kernels."""
+    order = torch.argsort(idx)
+    counts = torch.bincount(idx, minlength=rows)
+    ptr = torch.zeros(rows + 1, dtype=torch.int32, device=idx.device)
+    ptr[1:] = counts.cumsum(0).to(torch.int32)
+    return order.to(torch.int32), ptr
+
+
 # ---------------------------------------------------------------------------
 # fused edge attention (K3-K6 fwd, K15 bwd)
 # ---------------------------------------------------------------------------
@@ -101,7 +111,9 @@ class _EmbedNodeFn(torch.autograd.Function):
         m = ext()
         g = g.contiguous()
         dx_raw = g[:, : ctx.f].contiguous()
-        dtable = m.embed_scatter_add(g, cat_idx, ctx.f, ctx.rows)
+        h = g.shape[1] - ctx.f
+        order, ptr = _group_by(cat_idx, ctx.rows)
+        dtable = m.embed_grouped_scatter(g, order, ptr, ctx.rows, h, ctx.f)
         return dx_raw, None, dtable
 
 
@@ -119,7 +131,11 @@ class _EmbedEdgeFn(torch.autograd.Function):
         (edge_attr,) = ctx.saved_tensors
         m = ext()
         g = g.contiguous()
-        d_ifc, d_rpc = m.embed_edge_bwd(g, edge_attr, ctx.rows[0], ctx.rows[1])
+        h = g.shape[1] // 2
+        o0, p0 = _group_by(edge_attr[:, 0], ctx.rows[0])
+        o1, p1 = _group_by(edge_attr[:, 1], ctx.rows[1])
+        d_ifc = m.embed_grouped_scatter(g, o0, p0, ctx.rows[0], h, 0)
+        d_rpc = m.embed_grouped_scatter(g, o1, p1, ctx.rows[1], h, h)
         return None, d_ifc, d_rpc
 
 
@@ -152,7 +168,9 @@ class _EmbeddingFn(torch.autograd.Function):
     def backward(ctx, g):
         (idx,) = ctx.saved_tensors
         m = ext()
-        return None, m.scatter_add_rows(g.contiguous(), idx, ctx.rows)
+        g = g.contiguous()
+        order, ptr = _group_by(idx, ctx.rows)
+        return None, m.embed_grouped_scatter(g, order, ptr, ctx.rows, g.shape[1], 0)
 
 
 def embedding(idx, table):
