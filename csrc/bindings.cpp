@@ -63,7 +63,8 @@ void launch_gemm_f32_tn(const float*, const float*, float*, float*, int, int,
                         int, hipStream_t);
 void launch_colsum(const float*, float*, long, int, hipStream_t);
 void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
-                                  int, int, int, int, hipStream_t);
+                                  float*, long, int, int, int, int, int,
+                                  hipStream_t);
 
 // ---------------------------------------------------------------------------
 
@@ -367,10 +368,16 @@ torch::Tensor embed_grouped_scatter(torch::Tensor g, torch::Tensor order,
                                     torch::Tensor ptr, int64_t rows,
                                     int64_t h, int64_t col_off) {
   CHECK_IN(g); CHECK_IN(order); CHECK_IN(ptr);
+  const long num_src = order.size(0);
+  // sub-waves per row: enough that each wave sums ~<=512 gradient rows
+  long p = (num_src / std::max<int64_t>(rows, 1) + 511) / 512;
+  p = std::min<long>(std::max<long>(p, 1), 64);
   auto dtable = torch::empty({rows, h}, g.options());
+  auto partial = torch::empty({rows * p, h}, g.options());
   launch_embed_grouped_scatter(g.data_ptr<float>(), order.data_ptr<int>(),
-                               ptr.data_ptr<int>(), dtable.data_ptr<float>(),
-                               (int)rows, (int)h, (int)g.size(1), (int)col_off,
+                               ptr.data_ptr<int>(), partial.data_ptr<float>(),
+                               dtable.data_ptr<float>(), num_src, (int)rows,
+                               (int)p, (int)h, (int)g.size(1), (int)col_off,
                                cur_stream());
   return dtable;
 }

@@ -159,21 +159,26 @@ __global__ void embed_edge_bwd_kernel(const float* __restrict__ g,
 }
 
 // Deterministic grouped embedding backward: rows of g pre-grouped by table
-// index (order/ptr built once per backward from a sort) — one wave per table
-// row segment-sums its gradient rows; no atomics, coalesced columns.
+// index (order/ptr built once per backward from a sort).  Low-cardinality
+// tables (rpctype has ~5 rows over 5e4 edges) would serialize a whole segment
+// on one wave, so the reduction is TWO-PHASE: P sub-waves per table row each
+// sum a strided interleave of the segment (phase 1, deterministic), then one
+// wave per row folds the P partials in fixed order (phase 2).  No atomics.
 template <int VPT>
-__global__ void embed_grouped_scatter_kernel(
+__global__ void embed_grouped_p1_kernel(
     const float* __restrict__ g, const int* __restrict__ order,
-    const int* __restrict__ ptr, float* __restrict__ dtable, int rows, int h,
-    int gstride, int col_off) {
+    const int* __restrict__ ptr, float* __restrict__ partial, int rows, int P,
+    int h, int gstride, int col_off) {
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
-  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
-  if (row >= rows) return;
+  const int w = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (w >= rows * P) return;
+  const int row = w / P;
+  const int sub = w % P;
   float acc[VPT];
 #pragma unroll
   for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
-  for (int p = ptr[row]; p < ptr[row + 1]; ++p) {
+  for (int p = ptr[row] + sub; p < ptr[row + 1]; p += P) {
     const long r = order[p];
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
@@ -184,23 +189,52 @@ __global__ void embed_grouped_scatter_kernel(
 #pragma unroll
   for (int j = 0; j < VPT; ++j) {
     const int c = lane + j * PERTGNN_WAVE;
+    if (c < h) partial[(long)w * h + c] = acc[j];
+  }
+}
+
+template <int VPT>
+__global__ void embed_grouped_p2_kernel(const float* __restrict__ partial,
+                                        float* __restrict__ dtable, int rows,
+                                        int P, int h) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (row >= rows) return;
+  float acc[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
+  for (int sub = 0; sub < P; ++sub) {
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) acc[j] += partial[((long)row * P + sub) * h + c];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
     if (c < h) dtable[(long)row * h + c] = acc[j];
   }
 }
 
 void launch_embed_grouped_scatter(const float* g, const int* order,
-                                  const int* ptr, float* dtable, int rows,
+                                  const int* ptr, float* partial,
+                                  float* dtable, long num_src, int rows, int P,
                                   int h, int gstride, int col_off,
                                   hipStream_t s) {
   if (rows == 0) return;
-  const dim3 grid(ceil_div(rows, WAVES_PER_BLOCK));
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
   const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
   switch (vpt) {
 #define CASE(V)                                                                \
   case V:                                                                      \
-   hipLaunchKernelGGL(( embed_grouped_scatter_kernel<V>), dim3(grid), dim3(block), 0, s,                     \
-        g, order, ptr, dtable, rows, h, gstride, col_off);                     \
+   hipLaunchKernelGGL(( embed_grouped_p1_kernel<V>)                                                 \
+        , dim3(dim3(ceil_div((long)rows * P, WAVES_PER_BLOCK))), dim3(block), 0, s,     \
+            g, order, ptr, partial, rows, P, h, gstride, col_off);             \
+   hipLaunchKernelGGL(( embed_grouped_p2_kernel<V>)                                                 \
+        , dim3(dim3(ceil_div(rows, WAVES_PER_BLOCK))), dim3(block), 0, s,               \
+            partial, dtable, rows, P, h);                                      \
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
