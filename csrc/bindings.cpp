@@ -369,9 +369,13 @@ torch::Tensor embed_grouped_scatter(torch::Tensor g, torch::Tensor order,
                                     int64_t h, int64_t col_off) {
   CHECK_IN(g); CHECK_IN(order); CHECK_IN(ptr);
   const long num_src = order.size(0);
-  // sub-waves per row: enough that each wave sums ~<=512 gradient rows
-  long p = (num_src / std::max<int64_t>(rows, 1) + 511) / 512;
-  p = std::min<long>(std::max<long>(p, 1), 64);
+  // sub-waves per row: target ~16k waves to hide gather latency across the
+  // whole chip, but never more sub-waves than a row could populate
+  long p = 16384 / std::max<int64_t>(rows, 1);
+  const long per_row = (num_src + std::max<int64_t>(rows, 1) - 1) /
+                       std::max<int64_t>(rows, 1);
+  p = std::min<long>(p, per_row);
+  p = std::min<long>(std::max<long>(p, 1), 256);
   auto dtable = torch::empty({rows, h}, g.options());
   auto partial = torch::empty({rows * p, h}, g.options());
   launch_embed_grouped_scatter(g.data_ptr<float>(), order.data_ptr<int>(),
