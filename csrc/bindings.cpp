@@ -65,6 +65,14 @@ void launch_colsum(const float*, float*, long, int, hipStream_t);
 void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   float*, long, int, int, int, int, int,
                                   hipStream_t);
+void launch_edge_attn_fused_fwd(const float*, const float*, const float*,
+                                const long*, int, const int*, const int*,
+                                float*, float*, int, int, hipStream_t);
+void launch_edge_attn_fused_bwd(const float*, const float*, const float*,
+                                const float*, const long*, int, const float*,
+                                const int*, const int*, const int*,
+                                const int*, float*, float*, float*, float*,
+                                int, int, long, hipStream_t);
 
 // ---------------------------------------------------------------------------
 
@@ -386,7 +394,49 @@ torch::Tensor embed_grouped_scatter(torch::Tensor g, torch::Tensor order,
   return dtable;
 }
 
+std::vector<torch::Tensor> edge_attn_fused_fwd(
+    torch::Tensor qkvs, torch::Tensor pifc, torch::Tensor prpc,
+    torch::Tensor ea, torch::Tensor row_ptr, torch::Tensor csr_src) {
+  CHECK_IN(qkvs); CHECK_IN(pifc); CHECK_IN(prpc); CHECK_IN(ea);
+  const int n = qkvs.size(0);
+  const int h = qkvs.size(1) / 4;
+  TORCH_CHECK(h <= 512, "H must be <= 512");
+  auto out = torch::empty({n, h}, qkvs.options());
+  auto alpha = torch::empty({ea.size(0)}, qkvs.options());
+  launch_edge_attn_fused_fwd(
+      qkvs.data_ptr<float>(), pifc.data_ptr<float>(), prpc.data_ptr<float>(),
+      ea.data_ptr<long>(), (int)ea.size(1), row_ptr.data_ptr<int>(),
+      csr_src.data_ptr<int>(), out.data_ptr<float>(), alpha.data_ptr<float>(),
+      n, h, cur_stream());
+  return {out, alpha};
+}
+
+std::vector<torch::Tensor> edge_attn_fused_bwd(
+    torch::Tensor g, torch::Tensor qkvs, torch::Tensor pifc,
+    torch::Tensor prpc, torch::Tensor ea, torch::Tensor alpha,
+    torch::Tensor row_ptr, torch::Tensor csr_src, torch::Tensor col_ptr,
+    torch::Tensor csc_eid) {
+  CHECK_IN(g); CHECK_IN(qkvs); CHECK_IN(alpha);
+  const int n = qkvs.size(0);
+  const int h = qkvs.size(1) / 4;
+  const long ne = ea.size(0);
+  auto dqkvs = torch::empty_like(qkvs);
+  auto de = torch::empty({ne, h}, qkvs.options());
+  auto dek = torch::empty({ne, h}, qkvs.options());
+  auto dev = torch::empty({ne, h}, qkvs.options());
+  launch_edge_attn_fused_bwd(
+      g.data_ptr<float>(), qkvs.data_ptr<float>(), pifc.data_ptr<float>(),
+      prpc.data_ptr<float>(), ea.data_ptr<long>(), (int)ea.size(1),
+      alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
+      csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
+      csc_eid.data_ptr<int>(), dqkvs.data_ptr<float>(), de.data_ptr<float>(),
+      dek.data_ptr<float>(), dev.data_ptr<float>(), n, h, ne, cur_stream());
+  return {dqkvs, de};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("edge_attn_fused_fwd", &edge_attn_fused_fwd);
+  mod.def("edge_attn_fused_bwd", &edge_attn_fused_bwd);
   mod.def("embed_grouped_scatter", &embed_grouped_scatter);
   mod.def("linear_fwd", &linear_fwd);
   mod.def("linear_bwd", &linear_bwd);

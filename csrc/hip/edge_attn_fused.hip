@@ -1,0 +1,260 @@
+// Fused-layout CSR edge attention (gfx950) — production path.
+//
+// Differences from edge_attn.hip (which remains as the reference-layout
+// kernel used by the op-level parity tests):
+//   * q/k/v/skip live in ONE [N,4H] tensor (output of the fused QKVS GEMM)
+//     — one GEMM launch and one gradient buffer instead of four;
+//   * the per-edge embedding e_ij = W_e [ifc(a0) ‖ rpc(a1)] is NOT
+//     materialized: by linearity it equals P_ifc[a0] + P_rpc[a1] where
+//     P_ifc = ifc_table @ W_e[:, :H]^T (a tiny [V,H] table computed once per
+//     layer) — the kernel gathers from these L2-resident tables instead of
+//     streaming an [E,H] operand from HBM (halves forward edge traffic).
+//
+// Backward: row kernel writes dq + dskip segments of dqkvs and per-edge
+// dek/dev scratch; col kernel (CSC) segment-sums dk/dv into dqkvs; de =
+// dek + dev feeds the grouped per-vocab scatter (python side) to produce
+// dP_ifc/dP_rpc.  Deterministic — no atomics anywhere.
+
+#include "common.h"
+#include <cmath>
+
+#define WAVES_PER_BLOCK 4
+
+template <int VPT>
+__global__ void edge_attn_fused_fwd_kernel(
+    const float* __restrict__ qkvs,  // [N, 4h]
+    const float* __restrict__ pifc,  // [Vi, h]
+    const float* __restrict__ prpc,  // [Vr, h]
+    const long* __restrict__ ea, int astride,
+    const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
+    float* __restrict__ out, float* __restrict__ alpha, int n, int h,
+    float scale) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (row >= n) return;
+  const long ld = 4L * h;
+
+  float qr[VPT], acc[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    qr[j] = (c < h) ? qkvs[row * ld + c] : 0.f;
+    acc[j] = 0.f;
+  }
+
+  const int beg = row_ptr[row], end = row_ptr[row + 1];
+  float m = -INFINITY, s = 0.f;
+  for (int p = beg; p < end; ++p) {
+    const long src = csr_src[p];
+    const long a0 = ea[(long)p * astride];
+    const long a1 = ea[(long)p * astride + 1];
+    float part = 0.f;
+    float ec[VPT];
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) {
+        ec[j] = pifc[a0 * h + c] + prpc[a1 * h + c];
+        part += qr[j] * (qkvs[src * ld + h + c] + ec[j]);
+      }
+    }
+    const float logit = wave_reduce_sum(part) * scale;
+    if (lane == (p - beg) % PERTGNN_WAVE) alpha[p] = logit;
+    const float m_new = fmaxf(m, logit);
+    const float corr = __expf(m - m_new);
+    const float pexp = __expf(logit - m_new);
+    s = s * corr + pexp;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) {
+        const float ve = qkvs[src * ld + 2 * h + c] + ec[j];
+        acc[j] = acc[j] * corr + pexp * ve;
+      }
+    }
+    m = m_new;
+  }
+
+  const float inv_s = (s > 0.f) ? 1.f / s : 0.f;
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    if (c < h)
+      out[(long)row * h + c] = acc[j] * inv_s + qkvs[row * ld + 3 * h + c];
+  }
+  for (int p = beg + lane; p < end; p += PERTGNN_WAVE)
+    alpha[p] = __expf(alpha[p] - m) * inv_s;
+}
+
+template <int VPT>
+__global__ void edge_attn_fused_bwd_row_kernel(
+    const float* __restrict__ g, const float* __restrict__ qkvs,
+    const float* __restrict__ pifc, const float* __restrict__ prpc,
+    const long* __restrict__ ea, int astride, const float* __restrict__ alpha,
+    const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
+    float* __restrict__ dqkvs, float* __restrict__ dek,
+    float* __restrict__ dev, int n, int h, float scale) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (row >= n) return;
+  const long ld = 4L * h;
+
+  float gr[VPT], qr[VPT], dqacc[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    gr[j] = (c < h) ? g[(long)row * h + c] : 0.f;
+    qr[j] = (c < h) ? qkvs[row * ld + c] : 0.f;
+    dqacc[j] = 0.f;
+  }
+
+  const int beg = row_ptr[row], end = row_ptr[row + 1];
+  float sdot = 0.f;
+  for (int p = beg; p < end; ++p) {
+    const long src = csr_src[p];
+    const long a0 = ea[(long)p * astride];
+    const long a1 = ea[(long)p * astride + 1];
+    float part = 0.f;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h)
+        part += gr[j] * (qkvs[src * ld + 2 * h + c] + pifc[a0 * h + c] +
+                         prpc[a1 * h + c]);
+    }
+    const float dalpha = wave_reduce_sum(part);
+    sdot += alpha[p] * dalpha;
+    if (lane == 0) dev[(long)p * h] = dalpha;
+  }
+  for (int p = beg; p < end; ++p) {
+    const long src = csr_src[p];
+    const long a0 = ea[(long)p * astride];
+    const long a1 = ea[(long)p * astride + 1];
+    const float dalpha_l0 = (lane == 0) ? dev[(long)p * h] : 0.f;
+    const float dalpha = __shfl(dalpha_l0, 0, PERTGNN_WAVE);
+    const float a = alpha[p];
+    const float dl = a * (dalpha - sdot) * scale;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) {
+        const float kec =
+            qkvs[src * ld + h + c] + pifc[a0 * h + c] + prpc[a1 * h + c];
+        dqacc[j] += dl * kec;
+        dek[(long)p * h + c] = dl * qr[j];
+        dev[(long)p * h + c] = a * gr[j];
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    if (c < h) {
+      dqkvs[row * ld + c] = dqacc[j];         // dq
+      dqkvs[row * ld + 3 * h + c] = gr[j];    // dskip = g
+    }
+  }
+}
+
+template <int VPT>
+__global__ void edge_attn_fused_bwd_col_kernel(
+    const float* __restrict__ dek, const float* __restrict__ dev,
+    const int* __restrict__ col_ptr, const int* __restrict__ csc_eid,
+    float* __restrict__ dqkvs, int n, int h) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (row >= n) return;
+  const long ld = 4L * h;
+  float ka[VPT], va[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) { ka[j] = 0.f; va[j] = 0.f; }
+  for (int p = col_ptr[row]; p < col_ptr[row + 1]; ++p) {
+    const long eid = csc_eid[p];
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) {
+        ka[j] += dek[eid * h + c];
+        va[j] += dev[eid * h + c];
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    if (c < h) {
+      dqkvs[(long)row * ld + h + c] = ka[j];
+      dqkvs[(long)row * ld + 2 * h + c] = va[j];
+    }
+  }
+}
+
+__global__ void add2_kernel(const float* __restrict__ a,
+                            const float* __restrict__ b,
+                            float* __restrict__ out, long numel) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < numel; t += stride) out[t] = a[t] + b[t];
+}
+
+// ---------------------------------------------------------------------------
+
+void launch_edge_attn_fused_fwd(const float* qkvs, const float* pifc,
+                                const float* prpc, const long* ea, int astride,
+                                const int* row_ptr, const int* csr_src,
+                                float* out, float* alpha, int n, int h,
+                                hipStream_t stream) {
+  if (n == 0) return;
+  const float scale = 1.f / std::sqrt((float)h);
+  const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
+  const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
+  const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  switch (vpt) {
+#define CASE(V)                                                                \
+  case V:                                                                      \
+    edge_attn_fused_fwd_kernel<V><<<grid, block, 0, stream>>>(                 \
+        qkvs, pifc, prpc, ea, astride, row_ptr, csr_src, out, alpha, n, h,     \
+        scale);                                                                \
+    break;
+    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+#undef CASE
+    default: abort();
+  }
+}
+
+void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
+                                const float* pifc, const float* prpc,
+                                const long* ea, int astride,
+                                const float* alpha, const int* row_ptr,
+                                const int* csr_src, const int* col_ptr,
+                                const int* csc_eid, float* dqkvs, float* de,
+                                float* dek, float* dev, int n, int h,
+                                long num_edges, hipStream_t stream) {
+  if (n == 0) return;
+  const float scale = 1.f / std::sqrt((float)h);
+  const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
+  const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
+  const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  switch (vpt) {
+#define CASE(V)                                                                \
+  case V:                                                                      \
+    edge_attn_fused_bwd_row_kernel<V><<<grid, block, 0, stream>>>(             \
+        g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,      \
+        dek, dev, n, h, scale);                                                \
+    edge_attn_fused_bwd_col_kernel<V><<<grid, block, 0, stream>>>(             \
+        dek, dev, col_ptr, csc_eid, dqkvs, n, h);                              \
+    break;
+    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+#undef CASE
+    default: abort();
+  }
+  const long numel = num_edges * h;
+  if (numel > 0) {
+    const int tpb = 256;
+    const int blocks = (int)min((numel + tpb - 1) / tpb, (long)4096);
+    add2_kernel<<<dim3(blocks), dim3(tpb), 0, stream>>>(dek, dev, de, numel);
+  }
+}

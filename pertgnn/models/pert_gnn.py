@@ -62,6 +62,21 @@ class TransformerConv(nn.Module):
         skip = ops.linear(x, self.lin_skip.weight, self.lin_skip.bias)
         return ops.edge_attention(q, k, v, e, skip, edge_index, n, csr=csr)
 
+    def forward_fused(self, x, edge_attr, ifc_weight, rpc_weight, csr):
+        """HIP fast path: one [N,K]x[4H,K]^T GEMM for q/k/v/skip and
+        L2-resident per-vocab P tables instead of the [E,2H] edge-embed
+        stream (exact refactoring by linearity of lin_edge)."""
+        h = self.out_channels
+        w4 = torch.cat([self.lin_query.weight, self.lin_key.weight,
+                        self.lin_value.weight, self.lin_skip.weight], dim=0)
+        b4 = torch.cat([self.lin_query.bias, self.lin_key.bias,
+                        self.lin_value.bias, self.lin_skip.bias], dim=0)
+        qkvs = ops.linear(x, w4, b4)
+        we = self.lin_edge.weight  # [H, 2H]
+        pifc = ops.linear(ifc_weight, we[:, :h].contiguous(), None)
+        prpc = ops.linear(rpc_weight, we[:, h:].contiguous(), None)
+        return ops.edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr)
+
 
 class SAGEDeterministic(nn.Module):
     """API- and checkpoint-compatible with reference model.py:10-114."""
@@ -143,14 +158,27 @@ class SAGEDeterministic(nn.Module):
     ):
         if num_graphs is None:
             num_graphs = int(batch.max().item()) + 1 if batch.numel() else 0
+        from ..ops.backend import use_hip
+
+        fused = csr is not None and use_hip(x)
         x = ops.embed_concat_node(x, cat_X, [t.weight for t in self.cat_embedding])
-        edge_embeds = ops.embed_concat_edge(
-            edge_attr, self.interface_embeds.weight, self.rpctype_embeds.weight
-        )
+        edge_embeds = None
+        if not fused:
+            edge_embeds = ops.embed_concat_edge(
+                edge_attr, self.interface_embeds.weight, self.rpctype_embeds.weight
+            )
         n = x.shape[0]
 
+        def run_conv(conv, x):
+            if fused:
+                return conv.forward_fused(
+                    x, edge_attr, self.interface_embeds.weight,
+                    self.rpctype_embeds.weight, csr,
+                )
+            return conv(x, edge_index, edge_embeds, csr=csr, num_nodes=n)
+
         for i, conv in enumerate(self.convs[:-1]):
-            x = conv(x, edge_index, edge_embeds, csr=csr, num_nodes=n)
+            x = run_conv(conv, x)
             bn = self.bns[i]
             x = ops.batchnorm_relu(
                 x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
@@ -159,7 +187,7 @@ class SAGEDeterministic(nn.Module):
             if self.training and bn.track_running_stats and bn.num_batches_tracked is not None:
                 bn.num_batches_tracked += 1
             x = F.dropout(x, p=self.dropout, training=self.training)
-        x = self.convs[-1](x, edge_index, edge_embeds, csr=csr, num_nodes=n)
+        x = run_conv(self.convs[-1], x)
         local_predict = ops.linear(x, self.local_linear.weight, self.local_linear.bias)
         mean_x = ops.pattern_pool(x, pattern_probs, pattern_num_nodes, batch, num_graphs)
         entry_vec = ops.embedding(entry_id, self.entry_embeds.weight)

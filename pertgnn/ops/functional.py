@@ -62,6 +62,45 @@ def edge_attention(q, k, v, e, skip, edge_index, num_nodes, csr=None):
 
 
 # ---------------------------------------------------------------------------
+# fused-layout edge attention: qkvs [N,4H] + per-vocab P tables (fast path)
+# ---------------------------------------------------------------------------
+
+class _EdgeAttentionFusedFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_eid):
+        m = ext()
+        out, alpha = m.edge_attn_fused_fwd(qkvs, pifc, prpc, edge_attr, row_ptr, csr_src)
+        ctx.save_for_backward(qkvs, pifc, prpc, edge_attr, alpha,
+                              row_ptr, csr_src, col_ptr, csc_eid)
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        qkvs, pifc, prpc, edge_attr, alpha, row_ptr, csr_src, col_ptr, csc_eid = ctx.saved_tensors
+        m = ext()
+        dqkvs, de = m.edge_attn_fused_bwd(
+            g.contiguous(), qkvs, pifc, prpc, edge_attr, alpha,
+            row_ptr, csr_src, col_ptr, csc_eid,
+        )
+        # dP tables: grouped (deterministic) per-vocab segment sums of de
+        o0, p0 = _group_by(edge_attr[:, 0], pifc.shape[0])
+        o1, p1 = _group_by(edge_attr[:, 1], prpc.shape[0])
+        dpifc = m.embed_grouped_scatter(de, o0, p0, pifc.shape[0], de.shape[1], 0)
+        dprpc = m.embed_grouped_scatter(de, o1, p1, prpc.shape[0], de.shape[1], 0)
+        return dqkvs, dpifc, dprpc, None, None, None, None, None
+
+
+def edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr):
+    """HIP-only fast path: out_i = skip_i + softmax-weighted aggregate where
+    q/k/v/skip are the four H-segments of ``qkvs`` and the edge embedding is
+    P_ifc[a0] + P_rpc[a1] (exact refactoring of lin_edge(concat(ifc, rpc)))."""
+    row_ptr, csr_src, col_ptr, _csc_dst, csc_eid = csr
+    return _EdgeAttentionFusedFn.apply(
+        qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_eid
+    )
+
+
+# ---------------------------------------------------------------------------
 # pattern pool (K9+K10)
 # ---------------------------------------------------------------------------
 
@@ -279,6 +318,7 @@ def linear(x, w, b=None):
 
 __all__ = [
     "edge_attention",
+    "edge_attention_fused",
     "embedding",
     "pattern_pool",
     "embed_concat_node",

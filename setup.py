@@ -23,6 +23,7 @@ setup(
             sources=[
                 "csrc/bindings.cpp",
                 "csrc/hip/edge_attn.hip",
+                "csrc/hip/edge_attn_fused.hip",
                 "csrc/hip/segops.hip",
                 "csrc/hip/gemm.hip",
             ],
