@@ -65,6 +65,8 @@ void launch_colsum(const float*, float*, long, int, hipStream_t);
 void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   float*, long, int, int, int, int, int,
                                   hipStream_t);
+void launch_vocab_scatter(const float*, const long*, long, float*, long, int,
+                          int, int, int, hipStream_t);
 void launch_edge_attn_fused_fwd(const float*, const float*, const float*,
                                 const long*, int, const int*, const int*,
                                 float*, float*, int, int, hipStream_t);
@@ -434,7 +436,24 @@ std::vector<torch::Tensor> edge_attn_fused_bwd(
   return {dqkvs, de};
 }
 
+// dtable[v] += sum of g[r, col_off:col_off+h] over rows r with idx[r]==v.
+// idx may be a strided column view of an [N,A] attr tensor.
+torch::Tensor vocab_scatter(torch::Tensor g, torch::Tensor idx, int64_t rows,
+                            int64_t h, int64_t col_off) {
+  CHECK_IN(g);
+  TORCH_CHECK(idx.is_cuda() && idx.dim() == 1, "idx must be 1-D CUDA");
+  const size_t lds = (size_t)rows * h * sizeof(float);
+  TORCH_CHECK(lds <= 160 * 1024, "vocab too large for LDS accumulator");
+  auto dtable = torch::empty({rows, h}, g.options());
+  launch_vocab_scatter(g.data_ptr<float>(), idx.data_ptr<long>(),
+                       idx.stride(0), dtable.data_ptr<float>(), g.size(0),
+                       (int)rows, (int)h, (int)g.size(1), (int)col_off,
+                       cur_stream());
+  return dtable;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("vocab_scatter", &vocab_scatter);
   mod.def("edge_attn_fused_fwd", &edge_attn_fused_fwd);
   mod.def("edge_attn_fused_bwd", &edge_attn_fused_bwd);
   mod.def("embed_grouped_scatter", &embed_grouped_scatter);

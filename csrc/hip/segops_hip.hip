@@ -242,6 +242,52 @@ void launch_embed_grouped_scatter(const float* g, const int* order,
   }
 }
 
+// Single-pass vocab accumulator for small vocabularies: blocks stream
+// CONTIGUOUS gradient rows (coalesced, no sort/gather), accumulate into an
+// LDS-resident [V,h] table (LDS atomics — contention only within the CU),
+// then one global atomicAdd per (v,c) per block.  Weight-gradient class:
+// fp32 reduction order varies run to run (same class as split-K wgrad).
+__global__ void vocab_scatter_kernel(const float* __restrict__ g,
+                                     const long* __restrict__ idx,
+                                     long idx_stride, float* __restrict__ dtable,
+                                     long n, int rows, int h, int gstride,
+                                     int col_off) {
+  extern __shared__ float acc[];  // [rows*h]
+  const long vh = (long)rows * h;
+  for (long t = threadIdx.x; t < vh; t += blockDim.x) acc[t] = 0.f;
+  __syncthreads();
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * rows_per_block;
+  const long r1 = min(n, r0 + rows_per_block);
+  for (long r = r0 + wid; r < r1; r += WAVES_PER_BLOCK) {
+    const long v = idx[r * idx_stride];
+    for (int c = lane; c < h; c += PERTGNN_WAVE)
+      atomicAdd(&acc[v * h + c], g[r * gstride + col_off + c]);
+  }
+  __syncthreads();
+  for (long t = threadIdx.x; t < vh; t += blockDim.x)
+    if (acc[t] != 0.f) atomicAdd(&dtable[t], acc[t]);
+}
+
+void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
+                          float* dtable, long n, int rows, int h, int gstride,
+                          int col_off, hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(dtable, 0, (long)rows * h * sizeof(float), s));
+  if (n == 0) return;
+  const size_t lds = (size_t)rows * h * sizeof(float);
+  if (lds > 64 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        (const void*)vocab_scatter_kernel,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
+  }
+  const int blocks = (int)min((long)1024, (n + 63) / 64);
+ hipLaunchKernelGGL(( vocab_scatter_kernel), dim3(dim3(blocks)), dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)),
+                         lds, s, g, idx, idx_stride, dtable, n, rows, h,
+                                   gstride, col_off);
+}
+
 // entry embedding gather: out[b] = table[idx[b]] — plain gather (fwd) +
 // scatter-add (bwd); reuses the edge kernels' grid-stride shape.
 __global__ void gather_rows_kernel(const long* __restrict__ idx,

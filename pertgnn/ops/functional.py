@@ -16,6 +16,16 @@ from . import reference as ref
 from .backend import ext, use_hip
 
 
+def _table_grad(m, g, idx, rows, h, col_off):
+    """dtable[v] = segment-sum of g[:, col_off:col_off+h] by idx — LDS vocab
+    accumulator for small tables, deterministic two-phase grouped scatter
+    otherwise."""
+    if rows * h * 4 <= 160 * 1024:
+        return m.vocab_scatter(g, idx, rows, h, col_off)
+    order, ptr = _group_by(idx.contiguous(), rows)
+    return m.embed_grouped_scatter(g, order, ptr, rows, h, col_off)
+
+
 def _group_by(idx: torch.Tensor, rows: int):
     """Group positions by index value: returns (order int32, ptr int32[rows+1])
     for the deterministic grouped scatter kernels."""
@@ -82,11 +92,9 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
             g.contiguous(), qkvs, pifc, prpc, edge_attr, alpha,
             row_ptr, csr_src, col_ptr, csc_eid,
         )
-        # dP tables: grouped (deterministic) per-vocab segment sums of de
-        o0, p0 = _group_by(edge_attr[:, 0], pifc.shape[0])
-        o1, p1 = _group_by(edge_attr[:, 1], prpc.shape[0])
-        dpifc = m.embed_grouped_scatter(de, o0, p0, pifc.shape[0], de.shape[1], 0)
-        dprpc = m.embed_grouped_scatter(de, o1, p1, prpc.shape[0], de.shape[1], 0)
+        # dP tables: per-vocab segment sums of de
+        dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
+        dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
         return dqkvs, dpifc, dprpc, None, None, None, None, None
 
 
@@ -151,8 +159,7 @@ class _EmbedNodeFn(torch.autograd.Function):
         g = g.contiguous()
         dx_raw = g[:, : ctx.f].contiguous()
         h = g.shape[1] - ctx.f
-        order, ptr = _group_by(cat_idx, ctx.rows)
-        dtable = m.embed_grouped_scatter(g, order, ptr, ctx.rows, h, ctx.f)
+        dtable = _table_grad(m, g, cat_idx, ctx.rows, h, ctx.f)
         return dx_raw, None, dtable
 
 
@@ -171,10 +178,8 @@ class _EmbedEdgeFn(torch.autograd.Function):
         m = ext()
         g = g.contiguous()
         h = g.shape[1] // 2
-        o0, p0 = _group_by(edge_attr[:, 0], ctx.rows[0])
-        o1, p1 = _group_by(edge_attr[:, 1], ctx.rows[1])
-        d_ifc = m.embed_grouped_scatter(g, o0, p0, ctx.rows[0], h, 0)
-        d_rpc = m.embed_grouped_scatter(g, o1, p1, ctx.rows[1], h, h)
+        d_ifc = _table_grad(m, g, edge_attr[:, 0], ctx.rows[0], h, 0)
+        d_rpc = _table_grad(m, g, edge_attr[:, 1], ctx.rows[1], h, h)
         return None, d_ifc, d_rpc
 
 
@@ -208,8 +213,7 @@ class _EmbeddingFn(torch.autograd.Function):
         (idx,) = ctx.saved_tensors
         m = ext()
         g = g.contiguous()
-        order, ptr = _group_by(idx, ctx.rows)
-        return None, m.embed_grouped_scatter(g, order, ptr, ctx.rows, g.shape[1], 0)
+        return None, _table_grad(m, g, idx, ctx.rows, g.shape[1], 0)
 
 
 def embedding(idx, table):
