@@ -62,6 +62,12 @@ void launch_gemm_f32_nn(const float*, const float*, const float*, float*, int,
 void launch_gemm_f32_tn(const float*, const float*, float*, float*, int, int,
                         int, hipStream_t);
 void launch_colsum(const float*, float*, long, int, hipStream_t);
+void launch_gemm_bf16_nt(const float*, const float*, const float*, float*,
+                         int, int, int, bool, hipStream_t);
+void launch_gemm_bf16_nn(const float*, const float*, const float*, float*,
+                         int, int, int, bool, hipStream_t);
+void launch_gemm_bf16_tn(const float*, const float*, float*, float*, int, int,
+                         int, hipStream_t);
 void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   float*, long, int, int, int, int, int,
                                   hipStream_t);
@@ -307,7 +313,68 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
               (float)eps, (int)step, cur_stream());
 }
 
-// y = x @ w^T + b (torch Linear layout: w [out,in])
+// y = x @ w^T + b (torch Linear layout: w [out,in]); bf16 variant rounds
+// operands to bf16 in the matrix cores (fp32 accumulate + output).
+torch::Tensor linear_fwd_bf16(torch::Tensor x, torch::Tensor w,
+                              torch::Tensor b) {
+  CHECK_IN(x); CHECK_IN(w);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = w.size(0);
+  auto y = torch::empty({m, n}, x.options());
+  const float* bias = nullptr;
+  if (b.defined() && b.numel() > 0) bias = b.data_ptr<float>();
+  launch_gemm_bf16_nt(x.data_ptr<float>(), w.data_ptr<float>(), bias,
+                      y.data_ptr<float>(), m, n, k, false, cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> linear_bwd_bf16(torch::Tensor g, torch::Tensor x,
+                                           torch::Tensor w, bool has_bias) {
+  CHECK_IN(g); CHECK_IN(x); CHECK_IN(w);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = w.size(0);
+  auto dx = torch::empty({m, k}, x.options());
+  auto dw = torch::empty({n, k}, w.options());
+  torch::Tensor db = torch::empty({0}, g.options());
+  float* db_ptr = nullptr;
+  if (has_bias) {
+    db = torch::empty({n}, g.options());
+    db_ptr = db.data_ptr<float>();
+  }
+  launch_gemm_bf16_nn(g.data_ptr<float>(), w.data_ptr<float>(), nullptr,
+                      dx.data_ptr<float>(), m, n, k, false, cur_stream());
+  launch_gemm_bf16_tn(g.data_ptr<float>(), x.data_ptr<float>(),
+                      dw.data_ptr<float>(), db_ptr, m, n, k, cur_stream());
+  return {dx, dw, db};
+}
+
+torch::Tensor gemm_nt_bf16(torch::Tensor a, torch::Tensor b) {
+  CHECK_IN(a); CHECK_IN(b);
+  auto c = torch::empty({a.size(0), b.size(0)}, a.options());
+  launch_gemm_bf16_nt(a.data_ptr<float>(), b.data_ptr<float>(), nullptr,
+                      c.data_ptr<float>(), a.size(0), b.size(0), a.size(1),
+                      false, cur_stream());
+  return c;
+}
+torch::Tensor gemm_nn_bf16(torch::Tensor a, torch::Tensor b) {
+  CHECK_IN(a); CHECK_IN(b);
+  auto c = torch::empty({a.size(0), b.size(1)}, a.options());
+  launch_gemm_bf16_nn(a.data_ptr<float>(), b.data_ptr<float>(), nullptr,
+                      c.data_ptr<float>(), a.size(0), a.size(1), b.size(1),
+                      false, cur_stream());
+  return c;
+}
+torch::Tensor gemm_tn_bf16(torch::Tensor a, torch::Tensor b) {
+  CHECK_IN(a); CHECK_IN(b);
+  auto c = torch::empty({a.size(1), b.size(1)}, a.options());
+  launch_gemm_bf16_tn(a.data_ptr<float>(), b.data_ptr<float>(),
+                      c.data_ptr<float>(), nullptr, a.size(0), a.size(1),
+                      b.size(1), cur_stream());
+  return c;
+}
+
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b) {
   CHECK_IN(x); CHECK_IN(w);
   const int m = x.size(0);
@@ -458,6 +525,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("edge_attn_fused_bwd", &edge_attn_fused_bwd);
   mod.def("embed_grouped_scatter", &embed_grouped_scatter);
   mod.def("linear_fwd", &linear_fwd);
+  mod.def("linear_fwd_bf16", &linear_fwd_bf16);
+  mod.def("linear_bwd_bf16", &linear_bwd_bf16);
+  mod.def("gemm_nt_bf16", &gemm_nt_bf16);
+  mod.def("gemm_nn_bf16", &gemm_nn_bf16);
+  mod.def("gemm_tn_bf16", &gemm_tn_bf16);
   mod.def("linear_bwd", &linear_bwd);
   mod.def("gemm_nt", &gemm_nt);
   mod.def("gemm_nn", &gemm_nn);

@@ -295,13 +295,33 @@ def eval_metrics(y, y_hat, tau):
 # linear (K2) — MFMA GEMM on the HIP path
 # ---------------------------------------------------------------------------
 
+_GEMM_PRECISION = "fp32"
+
+
+def set_gemm_precision(prec: str):
+    """Select the matmul compute precision on the HIP path: "fp32" (exact,
+    v_mfma_f32_16x16x4_f32) or "bf16" (operands rounded to bf16 in the matrix
+    cores, fp32 accumulate — BASELINE configs 2/5 mixed-precision mode).
+    Everything outside the matmuls stays fp32 either way."""
+    global _GEMM_PRECISION
+    assert prec in ("fp32", "bf16"), prec
+    _GEMM_PRECISION = prec
+
+
+def gemm_precision() -> str:
+    return _GEMM_PRECISION
+
+
 class _LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b):
         m = ext()
-        y = m.linear_fwd(x, w, b if b is not None else torch.empty(0, dtype=x.dtype, device=x.device))
+        bf16 = _GEMM_PRECISION == "bf16"
+        bb = b if b is not None else torch.empty(0, dtype=x.dtype, device=x.device)
+        y = m.linear_fwd_bf16(x, w, bb) if bf16 else m.linear_fwd(x, w, bb)
         ctx.save_for_backward(x, w)
         ctx.has_bias = b is not None
+        ctx.bf16 = bf16
         return y
 
     @staticmethod
@@ -309,7 +329,10 @@ class _LinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         m = ext()
         g = g.contiguous()
-        dx, dw, db = m.linear_bwd(g, x, w, ctx.has_bias)
+        if ctx.bf16:
+            dx, dw, db = m.linear_bwd_bf16(g, x, w, ctx.has_bias)
+        else:
+            dx, dw, db = m.linear_bwd(g, x, w, ctx.has_bias)
         return dx, dw, (db if ctx.has_bias else None)
 
 
@@ -321,6 +344,8 @@ def linear(x, w, b=None):
 
 
 __all__ = [
+    "set_gemm_precision",
+    "gemm_precision",
     "edge_attention",
     "edge_attention_fused",
     "embedding",

@@ -318,3 +318,65 @@ def test_linear_parity():
     assert torch.allclose(ww.grad.cpu(), w2.grad, atol=5e-2, rtol=1e-3), \
         (ww.grad.cpu() - w2.grad).abs().max()
     assert torch.allclose(bb.grad.cpu(), b2.grad, atol=1e-2, rtol=1e-3)
+
+
+@pytest.mark.parametrize("m,n,k", [(1000, 256, 265), (2048, 1024, 256), (300, 128, 512)])
+def test_gemm_bf16_parity(m, n, k):
+    """bf16-compute GEMM vs torch matmul on bf16-rounded operands."""
+    require_ext()
+    import pertgnn._C as C
+    g = torch.Generator().manual_seed(21)
+    a = torch.randn(m, k, generator=g).to(DEV)
+    w = torch.randn(n, k, generator=g).to(DEV)
+    ar = a.to(torch.bfloat16).float()
+    wr = w.to(torch.bfloat16).float()
+
+    out = C.gemm_nt_bf16(a, w)
+    ref_ = ar @ wr.t()
+    torch.cuda.synchronize()
+    assert torch.allclose(out, ref_, atol=1e-2, rtol=1e-2), (out - ref_).abs().max()
+
+    gout = torch.randn(m, n, generator=torch.Generator().manual_seed(3)).to(DEV)
+    gr = gout.to(torch.bfloat16).float()
+    out_nn = C.gemm_nn_bf16(gout, w)
+    ref_nn = gr @ wr
+    torch.cuda.synchronize()
+    assert torch.allclose(out_nn, ref_nn, atol=1e-2, rtol=1e-2), (out_nn - ref_nn).abs().max()
+
+    out_tn = C.gemm_tn_bf16(gout, a)
+    ref_tn = gr.t() @ ar
+    torch.cuda.synchronize()
+    # split-K accumulation over large M: scale tolerance with sqrt(m)
+    assert torch.allclose(out_tn, ref_tn, atol=0.5, rtol=1e-2), (out_tn - ref_tn).abs().max()
+
+
+def test_model_bf16_close_to_fp32():
+    """End-to-end forward in bf16-GEMM mode stays close to the fp32 path."""
+    require_ext()
+    from pertgnn.models import SAGEDeterministic
+    import bench as bench_mod
+    from pertgnn.ops.functional import set_gemm_precision
+
+    torch.manual_seed(0)
+    batches, stats = bench_mod.build_synthetic_batches(1, 16, seed=0, device=DEV)
+    b = batches[0]
+    model = SAGEDeterministic(9, [stats["cat_max"] + 1], stats["entry_max"],
+                              stats["ifc_max"], stats["rpc_max"], 64, 3, 0.0).to(DEV)
+    model.eval()
+
+    def run():
+        with torch.no_grad():
+            gp, _ = model(b.x, b.cat_X, b.edge_index, b.edge_attr,
+                          b.pattern_num_nodes, b.rt_probs, b.entry_id, b.batch,
+                          csr=b.csr, num_graphs=b.num_graphs)
+        return gp
+
+    try:
+        set_gemm_precision("fp32")
+        out32 = run()
+        set_gemm_precision("bf16")
+        out16 = run()
+    finally:
+        set_gemm_precision("fp32")
+    rel = (out16 - out32).abs().max() / out32.abs().max().clamp_min(1e-6)
+    assert rel < 0.05, rel
