@@ -1,0 +1,367 @@
+#include "hip/hip_runtime.h"
+// bf16-compute GEMM path (gfx950): fp32 operands in HBM, rounded to bf16
+// during LDS staging, v_mfma_f32_16x16x32_bf16 matrix cores (≈2 PF ceiling,
+// 16x the f32-MFMA rate), fp32 accumulate, fp32 output.  This is the
+// mixed-precision mode of BASELINE configs 2/5: memory traffic identical to
+// the f32 path (activations stay fp32 end-to-end), only the matmul operands
+// are bf16-rounded.
+//
+// LDS layout: [free][BK+PAD] bf16 rows (k-contiguous), so each lane's MFMA
+// fragment (8 consecutive k) is ONE ds_read_b128; the +8-element row pad
+// makes the 16-lane fragment reads conflict-free (guide §6 G4).
+
+#include "common.h"
+
+#define BGEMM_BK 64
+#define BGEMM_PAD 8
+#define BGEMM_THREADS 256
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+__device__ __forceinline__ __bf16 to_bf16(float x) { return (__bf16)x; }
+
+// global [free][contract] (contract-minor) -> LDS [free][BK+PAD]
+template <int BF>
+__device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
+                                            long ld, int free0, int contract0,
+                                            int free_max, int contract_max,
+                                            __bf16* lds) {
+  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  const int t = threadIdx.x;
+  constexpr int QUADS = BGEMM_BK / 4;  // 16 float4 per row
+  const int f = t / QUADS;             // 16 free rows per pass
+  const int cq = (t % QUADS) * 4;
+  constexpr int FSTEP = BGEMM_THREADS / QUADS;  // 16
+  const bool fast = (free0 + BF <= free_max) &&
+                    (contract0 + BGEMM_BK <= contract_max) && ((ld & 3) == 0);
+#pragma unroll
+  for (int half = 0; half < BF / FSTEP; ++half) {
+    const int ff = f + half * FSTEP;
+    if (fast) {
+      const f32x4 v = *reinterpret_cast<const f32x4*>(
+          &g[(long)(free0 + ff) * ld + contract0 + cq]);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) lds[ff * LDW + cq + u] = to_bf16(v[u]);
+    } else {
+      const int gf = free0 + ff;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int gc = contract0 + cq + u;
+        lds[ff * LDW + cq + u] =
+            (gf < free_max && gc < contract_max) ? to_bf16(g[(long)gf * ld + gc])
+                                                 : (__bf16)0.f;
+      }
+    }
+  }
+}
+
+// global [contract][free] (contract-major) -> LDS [free][BK+PAD] (transpose)
+template <int BF>
+__device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
+                                            long ld, int contract0, int free0,
+                                            int contract_max, int free_max,
+                                            __bf16* lds) {
+  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  const int t = threadIdx.x;
+  constexpr int QUADS = BF / 4;
+  const int c = t / QUADS;
+  const int fq = (t % QUADS) * 4;
+  constexpr int CSTEP = BGEMM_THREADS / QUADS;
+  const bool fast = (contract0 + BGEMM_BK <= contract_max) &&
+                    (free0 + BF <= free_max) && ((ld & 3) == 0) &&
+                    ((free0 & 3) == 0);
+#pragma unroll
+  for (int half = 0; half < BGEMM_BK / CSTEP; ++half) {
+    const int cc = c + half * CSTEP;
+    if (fast) {
+      const f32x4 v = *reinterpret_cast<const f32x4*>(
+          &g[(long)(contract0 + cc) * ld + free0 + fq]);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) lds[(fq + u) * LDW + cc] = to_bf16(v[u]);
+    } else {
+      const int gc = contract0 + cc;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int gf = free0 + fq + u;
+        lds[(fq + u) * LDW + cc] =
+            (gc < contract_max && gf < free_max) ? to_bf16(g[(long)gc * ld + gf])
+                                                 : (__bf16)0.f;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wave tile: FM x FN fragments of 16x16; K-step 32 per MFMA, BK=64 = 2 steps
+// fragment layout (mfma_f32_16x16x32_bf16): lane l holds 8 k-consecutive
+// elements at k = (l>>4)*8 of row/col (l&15); C/D: col=l&15, row=(l>>4)*4+r.
+// ---------------------------------------------------------------------------
+
+template <int FM, int FN>
+struct BWaveTile {
+  f32x4 acc[FM][FN];
+  __device__ __forceinline__ void zero() {
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+      for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+  }
+  __device__ __forceinline__ void mma(const __bf16* lds_a, const __bf16* lds_b,
+                                      int wm, int wn, int lane) {
+    constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+    const int fi = lane & 15;
+    const int fk = (lane >> 4) * 8;
+#pragma unroll
+    for (int s = 0; s < BGEMM_BK / 32; ++s) {
+      const int k = s * 32 + fk;
+      bf16x8 a[FM], b[FN];
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+        a[mi] = *reinterpret_cast<const bf16x8*>(
+            &lds_a[(wm + mi * 16 + fi) * LDW + k]);
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni)
+        b[ni] = *reinterpret_cast<const bf16x8*>(
+            &lds_b[(wn + ni * 16 + fi) * LDW + k]);
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+    }
+  }
+  __device__ __forceinline__ void store(float* __restrict__ c, long ldc,
+                                        int row0, int col0, int m_max,
+                                        int n_max, const float* bias, int relu,
+                                        int lane) {
+    const int fcol = lane & 15;
+    const int frow = (lane >> 4) * 4;
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = row0 + mi * 16 + frow + r;
+          const int col = col0 + ni * 16 + fcol;
+          if (row < m_max && col < n_max) {
+            float v = acc[mi][ni][r];
+            if (bias) v += bias[col];
+            if (relu) v = fmaxf(v, 0.f);
+            c[(long)row * ldc + col] = v;
+          }
+        }
+  }
+};
+
+// ---------------------------------------------------------------------------
+// kernels — same three layouts as the f32 suite
+// ---------------------------------------------------------------------------
+
+template <int BM, int BN>
+__launch_bounds__(BGEMM_THREADS)
+__global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
+                                    const float* __restrict__ b,
+                                    const float* __restrict__ bias,
+                                    float* __restrict__ c, int m, int n, int k,
+                                    int relu) {
+  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  constexpr int FM = BM / 32, FN = BN / 32;
+  __shared__ __bf16 lds_a[2][BM * LDW];
+  __shared__ __bf16 lds_b[2][BN * LDW];
+  const int tiles_n = (n + BN - 1) / BN;
+  const int m0 = (blockIdx.x / tiles_n) * BM;
+  const int n0 = (blockIdx.x % tiles_n) * BN;
+  const int wave = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int wm = (wave >> 1) * (BM / 2);
+  const int wn = (wave & 1) * (BN / 2);
+
+  BWaveTile<FM, FN> wt;
+  wt.zero();
+  int buf = 0;
+  bstage_cmin<BM>(a, k, m0, 0, m, k, lds_a[0]);
+  bstage_cmin<BN>(b, k, n0, 0, n, k, lds_b[0]);
+  __syncthreads();
+  for (int k0 = BGEMM_BK; k0 < k; k0 += BGEMM_BK) {
+    bstage_cmin<BM>(a, k, m0, k0, m, k, lds_a[buf ^ 1]);
+    bstage_cmin<BN>(b, k, n0, k0, n, k, lds_b[buf ^ 1]);
+    wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+    __syncthreads();
+    buf ^= 1;
+  }
+  wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+  wt.store(c, n, m0 + wm, n0 + wn, m, n, bias, relu, lane);
+}
+
+template <int BM, int BN>
+__launch_bounds__(BGEMM_THREADS)
+__global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
+                                    const float* __restrict__ b,
+                                    const float* __restrict__ bias,
+                                    float* __restrict__ c, int m, int n,
+                                    int k2, int relu) {
+  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  constexpr int FM = BM / 32, FN = BN / 32;
+  __shared__ __bf16 lds_a[2][BM * LDW];
+  __shared__ __bf16 lds_b[2][BN * LDW];
+  const int tiles_n = (k2 + BN - 1) / BN;
+  const int m0 = (blockIdx.x / tiles_n) * BM;
+  const int n0 = (blockIdx.x % tiles_n) * BN;
+  const int wave = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int wm = (wave >> 1) * (BM / 2);
+  const int wn = (wave & 1) * (BN / 2);
+
+  BWaveTile<FM, FN> wt;
+  wt.zero();
+  int buf = 0;
+  bstage_cmin<BM>(a, n, m0, 0, m, n, lds_a[0]);
+  bstage_cmaj<BN>(b, k2, 0, n0, n, k2, lds_b[0]);
+  __syncthreads();
+  for (int c0 = BGEMM_BK; c0 < n; c0 += BGEMM_BK) {
+    bstage_cmin<BM>(a, n, m0, c0, m, n, lds_a[buf ^ 1]);
+    bstage_cmaj<BN>(b, k2, c0, n0, n, k2, lds_b[buf ^ 1]);
+    wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+    __syncthreads();
+    buf ^= 1;
+  }
+  wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+  wt.store(c, k2, m0 + wm, n0 + wn, m, k2, bias, relu, lane);
+}
+
+template <int BM, int BN>
+__launch_bounds__(BGEMM_THREADS)
+__global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
+                                    const float* __restrict__ b,
+                                    float* __restrict__ c,
+                                    float* __restrict__ dbias, int m, int n,
+                                    int k2, int slices) {
+  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  constexpr int FM = BM / 32, FN = BN / 32;
+  __shared__ __bf16 lds_a[2][BM * LDW];
+  __shared__ __bf16 lds_b[2][BN * LDW];
+  const int tiles_k = (k2 + BN - 1) / BN;
+  const int tile_id = blockIdx.x / slices;
+  const int slice = blockIdx.x % slices;
+  const int n0 = (tile_id / tiles_k) * BM;
+  const int k0 = (tile_id % tiles_k) * BN;
+  if (n0 >= n) return;
+  const int wave = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int wm = (wave >> 1) * (BM / 2);
+  const int wn = (wave & 1) * (BN / 2);
+
+  const int per_slice =
+      ((m + slices - 1) / slices + BGEMM_BK - 1) / BGEMM_BK * BGEMM_BK;
+  const int c_beg = slice * per_slice;
+  const int c_end = min(m, c_beg + per_slice);
+  if (c_beg >= c_end) return;
+
+  BWaveTile<FM, FN> wt;
+  wt.zero();
+  // fused bias grad: sum the STAGED bf16 A (=g) tile columns.  NOTE: this
+  // sums bf16-rounded g — for exact-f32 db the caller uses the f32 path.
+  const bool do_bias = (dbias != nullptr) && (k0 == 0);
+  float dbsum = 0.f;
+  const int bcol = threadIdx.x;
+  int buf = 0;
+  bstage_cmaj<BM>(a, n, c_beg, n0, c_end, n, lds_a[0]);
+  bstage_cmaj<BN>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
+  __syncthreads();
+  for (int cc = c_beg + BGEMM_BK; cc < c_end; cc += BGEMM_BK) {
+    bstage_cmaj<BM>(a, n, cc, n0, c_end, n, lds_a[buf ^ 1]);
+    bstage_cmaj<BN>(b, k2, cc, k0, c_end, k2, lds_b[buf ^ 1]);
+    wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+    if (do_bias && bcol < BM)
+#pragma unroll
+      for (int r = 0; r < BGEMM_BK; ++r)
+        dbsum += (float)lds_a[buf][bcol * LDW + r];
+    __syncthreads();
+    buf ^= 1;
+  }
+  wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+  if (do_bias && bcol < BM) {
+#pragma unroll
+    for (int r = 0; r < BGEMM_BK; ++r)
+      dbsum += (float)lds_a[buf][bcol * LDW + r];
+    if (n0 + bcol < n) atomicAdd(&dbias[n0 + bcol], dbsum);
+  }
+
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < FN; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = n0 + wm + mi * 16 + frow + r;
+        const int col = k0 + wn + ni * 16 + fcol;
+        if (row < n && col < k2) {
+          if (slices == 1)
+            c[(long)row * k2 + col] = wt.acc[mi][ni][r];
+          else
+            atomicAdd(&c[(long)row * k2 + col], wt.acc[mi][ni][r]);
+        }
+      }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+void launch_gemm_bf16_nt(const float* a, const float* b, const float* bias,
+                         float* c, int m, int n, int k, bool relu,
+                         hipStream_t s) {
+  if (m >= 512 && n >= 128) {
+    const int grid = ((m + 127) / 128) * ((n + 127) / 128);
+   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
+        a, b, bias, c, m, n, k, relu ? 1 : 0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((n + 63) / 64);
+   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<64, 64>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
+        a, b, bias, c, m, n, k, relu ? 1 : 0);
+  }
+}
+
+void launch_gemm_bf16_nn(const float* a, const float* b, const float* bias,
+                         float* c, int m, int n, int k2, bool relu,
+                         hipStream_t s) {
+  if (m >= 512 && k2 >= 128) {
+    const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
+   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
+        a, b, bias, c, m, n, k2, relu ? 1 : 0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((k2 + 63) / 64);
+   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<64, 64>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
+        a, b, bias, c, m, n, k2, relu ? 1 : 0);
+  }
+}
+
+void launch_gemm_bf16_tn(const float* a, const float* b, float* c,
+                         float* dbias, int m, int n, int k2, hipStream_t s) {
+  const bool big = (n >= 128 && k2 >= 128);
+  const int bm = big ? 128 : 64;
+  const int bn = big ? 128 : 64;
+  const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
+  int slices = 1;
+  while (tiles * slices < 512 && slices < 64 &&
+         (long)slices * BGEMM_BK * 4 < m)
+    slices *= 2;
+  if (slices > 1)
+    HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
+  if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
+  if (big)
+   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128>)
+        , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+  else
+   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64>)
+        , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+}

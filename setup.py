@@ -26,6 +26,7 @@ setup(
                 "csrc/hip/edge_attn_fused.hip",
                 "csrc/hip/segops.hip",
                 "csrc/hip/gemm.hip",
+                "csrc/hip/gemm_bf16.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
