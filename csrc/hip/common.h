@@ -30,3 +30,14 @@ __device__ __forceinline__ float wave_reduce_max(float v) {
 }
 
 static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
+
+// XCD-aware blockIdx remap (guide §5.5 T1, bijective form): the dispatcher
+// places block b on XCD b%8; this remap gives each XCD a CONTIGUOUS chunk of
+// the logical grid so neighboring tiles (which share operand panels) hit the
+// same XCD-private L2.
+__device__ __forceinline__ int xcd_swizzle(int bid, int nwg) {
+  const int xcd = bid % 8;
+  const int q = nwg / 8;
+  const int r = nwg % 8;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + bid / 8;
+}

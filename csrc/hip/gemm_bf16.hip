@@ -17,10 +17,19 @@
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
 __device__ __forceinline__ __bf16 to_bf16(float x) { return (__bf16)x; }
 
+__device__ __forceinline__ bf16x4 pack4(float a, float b, float c, float d) {
+  bf16x4 r;
+  r[0] = (__bf16)a; r[1] = (__bf16)b; r[2] = (__bf16)c; r[3] = (__bf16)d;
+  return r;
+}
+
 // global [free][contract] (contract-minor) -> LDS [free][BK+PAD]
+// 4 bf16 packed into one 8-byte LDS write (scalar u16 LDS writes are ~2x
+// slower — guide G13 applies to LDS too).
 template <int BF>
 __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
                                             long ld, int free0, int contract0,
@@ -40,22 +49,26 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
     if (fast) {
       const f32x4 v = *reinterpret_cast<const f32x4*>(
           &g[(long)(free0 + ff) * ld + contract0 + cq]);
-#pragma unroll
-      for (int u = 0; u < 4; ++u) lds[ff * LDW + cq + u] = to_bf16(v[u]);
+      *reinterpret_cast<bf16x4*>(&lds[ff * LDW + cq]) =
+          pack4(v[0], v[1], v[2], v[3]);
     } else {
       const int gf = free0 + ff;
+      float v[4];
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         const int gc = contract0 + cq + u;
-        lds[ff * LDW + cq + u] =
-            (gf < free_max && gc < contract_max) ? to_bf16(g[(long)gf * ld + gc])
-                                                 : (__bf16)0.f;
+        v[u] = (gf < free_max && gc < contract_max) ? g[(long)gf * ld + gc]
+                                                    : 0.f;
       }
+      *reinterpret_cast<bf16x4*>(&lds[ff * LDW + cq]) =
+          pack4(v[0], v[1], v[2], v[3]);
     }
   }
 }
 
-// global [contract][free] (contract-major) -> LDS [free][BK+PAD] (transpose)
+// global [contract][free] (contract-major) -> LDS [free][BK+PAD]: each thread
+// transposes a 4x4 block in registers (4 coalesced f32x4 loads from 4
+// contract rows), then writes 4 packed 8-byte LDS rows.
 template <int BF>
 __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
                                             long ld, int contract0, int free0,
@@ -63,31 +76,37 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
                                             __bf16* lds) {
   constexpr int LDW = BGEMM_BK + BGEMM_PAD;
   const int t = threadIdx.x;
-  constexpr int QUADS = BF / 4;
-  const int c = t / QUADS;
-  const int fq = (t % QUADS) * 4;
-  constexpr int CSTEP = BGEMM_THREADS / QUADS;
+  constexpr int FQUADS = BF / 4;
+  const int cb = (t / FQUADS) * 4;     // contract block of 4
+  const int fq = (t % FQUADS) * 4;     // free quad
+  constexpr int CSTEP = (BGEMM_THREADS / FQUADS) * 4;
   const bool fast = (contract0 + BGEMM_BK <= contract_max) &&
                     (free0 + BF <= free_max) && ((ld & 3) == 0) &&
                     ((free0 & 3) == 0);
 #pragma unroll
   for (int half = 0; half < BGEMM_BK / CSTEP; ++half) {
-    const int cc = c + half * CSTEP;
-    if (fast) {
-      const f32x4 v = *reinterpret_cast<const f32x4*>(
-          &g[(long)(contract0 + cc) * ld + free0 + fq]);
+    const int cc = cb + half * CSTEP;
+    float v[4][4];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) lds[(fq + u) * LDW + cc] = to_bf16(v[u]);
-    } else {
-      const int gc = contract0 + cc;
+    for (int u = 0; u < 4; ++u) {
+      if (fast) {
+        *reinterpret_cast<f32x4*>(v[u]) = *reinterpret_cast<const f32x4*>(
+            &g[(long)(contract0 + cc + u) * ld + free0 + fq]);
+      } else {
+        const int gc = contract0 + cc + u;
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int gf = free0 + fq + u;
-        lds[(fq + u) * LDW + cc] =
-            (gc < contract_max && gf < free_max) ? to_bf16(g[(long)gc * ld + gf])
-                                                 : (__bf16)0.f;
+        for (int w = 0; w < 4; ++w) {
+          const int gf = free0 + fq + w;
+          v[u][w] = (gc < contract_max && gf < free_max)
+                        ? g[(long)gc * ld + gf]
+                        : 0.f;
+        }
       }
     }
+#pragma unroll
+    for (int w = 0; w < 4; ++w)
+      *reinterpret_cast<bf16x4*>(&lds[(fq + w) * LDW + cc]) =
+          pack4(v[0][w], v[1][w], v[2][w], v[3][w]);
   }
 }
 
@@ -170,9 +189,10 @@ __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
   constexpr int FM = BM / 32, FN = BN / 32;
   __shared__ __bf16 lds_a[2][BM * LDW];
   __shared__ __bf16 lds_b[2][BN * LDW];
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_n = (n + BN - 1) / BN;
-  const int m0 = (blockIdx.x / tiles_n) * BM;
-  const int n0 = (blockIdx.x % tiles_n) * BN;
+  const int m0 = (bid / tiles_n) * BM;
+  const int n0 = (bid % tiles_n) * BN;
   const int wave = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
   const int wm = (wave >> 1) * (BM / 2);
@@ -206,9 +226,10 @@ __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
   constexpr int FM = BM / 32, FN = BN / 32;
   __shared__ __bf16 lds_a[2][BM * LDW];
   __shared__ __bf16 lds_b[2][BN * LDW];
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_n = (k2 + BN - 1) / BN;
-  const int m0 = (blockIdx.x / tiles_n) * BM;
-  const int n0 = (blockIdx.x % tiles_n) * BN;
+  const int m0 = (bid / tiles_n) * BM;
+  const int n0 = (bid % tiles_n) * BN;
   const int wave = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
   const int wm = (wave >> 1) * (BM / 2);
@@ -242,9 +263,10 @@ __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
   constexpr int FM = BM / 32, FN = BN / 32;
   __shared__ __bf16 lds_a[2][BM * LDW];
   __shared__ __bf16 lds_b[2][BN * LDW];
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_k = (k2 + BN - 1) / BN;
-  const int tile_id = blockIdx.x / slices;
-  const int slice = blockIdx.x % slices;
+  const int tile_id = bid / slices;
+  const int slice = bid % slices;
   const int n0 = (tile_id / tiles_k) * BM;
   const int k0 = (tile_id % tiles_k) * BN;
   if (n0 >= n) return;

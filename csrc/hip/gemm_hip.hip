@@ -175,9 +175,10 @@ __global__ void gemm_f32_nt_kernel(const float* __restrict__ a,
   constexpr int FM = BM / 32, FN = BN / 32;  // 2x2 wave grid
   __shared__ float lds_a[2][GEMM_BK * BM];
   __shared__ float lds_b[2][GEMM_BK * BN];
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_n = (n + BN - 1) / BN;
-  const int m0 = (blockIdx.x / tiles_n) * BM;
-  const int n0 = (blockIdx.x % tiles_n) * BN;
+  const int m0 = (bid / tiles_n) * BM;
+  const int n0 = (bid % tiles_n) * BN;
   const int wave = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
   const int wm = (wave >> 1) * (BM / 2);
@@ -211,9 +212,10 @@ __global__ void gemm_f32_nn_kernel(const float* __restrict__ a,
   constexpr int FM = BM / 32, FN = BN / 32;
   __shared__ float lds_a[2][GEMM_BK * BM];
   __shared__ float lds_b[2][GEMM_BK * BN];
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_n = (k2 + BN - 1) / BN;
-  const int m0 = (blockIdx.x / tiles_n) * BM;
-  const int n0 = (blockIdx.x % tiles_n) * BN;
+  const int m0 = (bid / tiles_n) * BM;
+  const int n0 = (bid % tiles_n) * BN;
   const int wave = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
   const int wm = (wave >> 1) * (BM / 2);
@@ -247,9 +249,10 @@ __global__ void gemm_f32_tn_kernel(const float* __restrict__ a,
   constexpr int FM = BM / 32, FN = BN / 32;
   __shared__ float lds_a[2][GEMM_BK * BM];
   __shared__ float lds_b[2][GEMM_BK * BN];
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_k = (k2 + BN - 1) / BN;
-  const int tile_id = blockIdx.x / slices;
-  const int slice = blockIdx.x % slices;
+  const int tile_id = bid / slices;
+  const int slice = bid % slices;
   const int n0 = (tile_id / tiles_k) * BM;
   const int k0 = (tile_id % tiles_k) * BN;
   if (n0 >= n) return;
