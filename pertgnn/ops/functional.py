@@ -116,8 +116,11 @@ class _PatternPoolFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, probs, nnodes, batch, num_graphs):
         m = ext()
-        # nodes are contiguous per graph (collation order) -> batch_ptr
-        counts = torch.bincount(batch, minlength=num_graphs)
+        # nodes are contiguous per graph (collation order) -> batch_ptr.
+        # scatter_add instead of bincount: bincount syncs on the device max,
+        # which would break hipGraph capture of the training step.
+        counts = torch.zeros(num_graphs, dtype=torch.long, device=x.device)
+        counts.scatter_add_(0, batch, torch.ones_like(batch))
         batch_ptr = torch.zeros(num_graphs + 1, dtype=torch.int32, device=x.device)
         batch_ptr[1:] = torch.cumsum(counts, 0).to(torch.int32)
         out = m.seg_pool_fwd(x, probs.contiguous(), nnodes.contiguous(), batch_ptr, num_graphs)
