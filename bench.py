@@ -128,6 +128,9 @@ def main():
     batches, stats = build_synthetic_batches(
         args.n_batches, args.batch_size, seed=100 + comm.rank, device=device
     )
+    # vocab sizes must agree across ranks (per-rank synthetic seeds differ)
+    for key in ("cat_max", "entry_max", "ifc_max", "rpc_max"):
+        stats[key] = int(comm.all_reduce_scalar(float(stats[key]), op="max"))
 
     model = SAGEDeterministic(
         9, [stats["cat_max"] + 1], stats["entry_max"], stats["ifc_max"],

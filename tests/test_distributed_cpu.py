@@ -99,3 +99,52 @@ def test_ddp_grads_match_average_of_ranks():
     for n in expected:
         assert torch.allclose(torch.from_numpy(g0[n]), expected[n], atol=1e-6), n
         assert torch.allclose(torch.from_numpy(g0[n]), torch.from_numpy(g1[n]), atol=1e-6), n
+
+
+def _worker_flat(rank, world_size, port, result_queue):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world_size), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    from pertgnn.parallel import Comm
+    from pertgnn.train.optim import FlatGradAllReduce, FusedAdam
+
+    comm = Comm(backend="gloo")
+    model = _build_model()
+    comm.broadcast_module_(model)
+    opt = FusedAdam(model.parameters(), lr=1e-2)
+    engine = FlatGradAllReduce(opt, comm, bucket_cap_mb=0.0005)  # many buckets
+    model.train()
+    for step in range(2):
+        opt.zero_grad()
+        engine.reset()
+        loss = _loss_on(model, seed=200 + rank)
+        loss.backward()
+        engine.finalize()
+        opt.step()
+    grads = opt.flat_grad.numpy().copy()
+    params = opt.flat_param.numpy().copy()
+    result_queue.put((rank, grads, params))
+    comm.barrier()
+    comm.finalize()
+
+
+@pytest.mark.timeout(120)
+def test_flat_engine_ranks_stay_in_sync():
+    """FusedAdam + FlatGradAllReduce: after identical averaged grads, both
+    ranks' flat parameters must be bitwise identical."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_flat, args=(r, world, 29513, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=100) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    results.sort(key=lambda t: t[0])
+    (_, g0, p0), (_, g1, p1) = results
+    import numpy as np
+    assert np.array_equal(g0, g1), "averaged grads must match bitwise"
+    assert np.array_equal(p0, p1), "params must stay in sync"
+    assert np.abs(g0).sum() > 0
