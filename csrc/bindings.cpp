@@ -47,6 +47,19 @@ void launch_bn_fwd(const float*, const float*, const float*, float*, float*,
 void launch_bn_bwd(const float*, const float*, const float*, const float*,
                    const float*, const float*, float*, float*, float*, float*,
                    long, int, bool, hipStream_t);
+void launch_bn_stats_only(const float*, long, int, float*, hipStream_t);
+void launch_bn_finalize_apply(const float*, const float*, long, const float*,
+                              const float*, float*, float*, float*, float*,
+                              float*, long, int, float, float, bool, bool,
+                              hipStream_t);
+void launch_bn_bwd_partials_only(const float*, const float*, const float*,
+                                 const float*, const float*, long, int, bool,
+                                 float*, hipStream_t);
+void launch_bn_bwd_apply_only(const float*, const float*, const float*,
+                              const float*, const float*, const float*,
+                              const float*, long, float*, long, int, bool,
+                              hipStream_t);
+void launch_bn_grad_affine(const float*, float*, float*, int, hipStream_t);
 void launch_quantile_loss_fwd(const float*, const float*, float*, long, float,
                               hipStream_t);
 void launch_quantile_loss_bwd(const float*, const float*, const float*, float*,
@@ -81,6 +94,12 @@ void launch_edge_attn_fused_bwd(const float*, const float*, const float*,
                                 const int*, const int*, const int*,
                                 const int*, float*, float*, float*, float*,
                                 int, int, long, hipStream_t);
+std::vector<torch::Tensor> collate_native(
+    std::vector<torch::Tensor>, std::vector<torch::Tensor>,
+    std::vector<torch::Tensor>, std::vector<torch::Tensor>,
+    std::vector<torch::Tensor>, std::vector<torch::Tensor>,
+    std::vector<torch::Tensor>, std::vector<torch::Tensor>,
+    std::vector<torch::Tensor>, bool);
 
 // ---------------------------------------------------------------------------
 
@@ -519,8 +538,77 @@ torch::Tensor vocab_scatter(torch::Tensor g, torch::Tensor idx, int64_t rows,
   return dtable;
 }
 
+torch::Tensor bn_stats(torch::Tensor x) {
+  CHECK_IN(x);
+  auto partials = torch::empty({2 * x.size(1)}, x.options());
+  launch_bn_stats_only(x.data_ptr<float>(), x.size(0), x.size(1),
+                       partials.data_ptr<float>(), cur_stream());
+  return partials;
+}
+
+std::vector<torch::Tensor> bn_finalize_apply(
+    torch::Tensor x, torch::Tensor partials, int64_t count,
+    torch::Tensor gamma, torch::Tensor beta, torch::Tensor running_mean,
+    torch::Tensor running_var, double momentum, double eps, bool training,
+    bool relu) {
+  CHECK_IN(x); CHECK_IN(partials);
+  const int h = x.size(1);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({h}, x.options());
+  auto invstd = torch::empty({h}, x.options());
+  launch_bn_finalize_apply(
+      x.data_ptr<float>(), partials.data_ptr<float>(), count,
+      gamma.data_ptr<float>(), beta.data_ptr<float>(),
+      running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
+      mean.data_ptr<float>(), invstd.data_ptr<float>(), y.data_ptr<float>(),
+      x.size(0), h, (float)momentum, (float)eps, training, relu,
+      cur_stream());
+  return {y, mean, invstd};
+}
+
+torch::Tensor bn_bwd_partials(torch::Tensor g, torch::Tensor x,
+                              torch::Tensor y, torch::Tensor mean,
+                              torch::Tensor invstd, bool relu) {
+  CHECK_IN(g); CHECK_IN(x);
+  auto partials = torch::empty({2 * x.size(1)}, x.options());
+  launch_bn_bwd_partials_only(g.data_ptr<float>(), x.data_ptr<float>(),
+                              y.data_ptr<float>(), mean.data_ptr<float>(),
+                              invstd.data_ptr<float>(), x.size(0), x.size(1),
+                              relu, partials.data_ptr<float>(), cur_stream());
+  return partials;
+}
+
+std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor g, torch::Tensor x,
+                                        torch::Tensor y, torch::Tensor mean,
+                                        torch::Tensor invstd,
+                                        torch::Tensor gamma,
+                                        torch::Tensor partials_global,
+                                        torch::Tensor partials_local,
+                                        int64_t count, bool relu) {
+  CHECK_IN(g); CHECK_IN(x);
+  const int h = x.size(1);
+  auto dx = torch::empty_like(x);
+  auto dgamma = torch::empty({h}, x.options());
+  auto dbeta = torch::empty({h}, x.options());
+  launch_bn_bwd_apply_only(g.data_ptr<float>(), x.data_ptr<float>(),
+                           y.data_ptr<float>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                           partials_global.data_ptr<float>(), count,
+                           dx.data_ptr<float>(), x.size(0), h, relu,
+                           cur_stream());
+  launch_bn_grad_affine(partials_local.data_ptr<float>(),
+                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), h,
+                        cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("bn_stats", &bn_stats);
+  mod.def("bn_finalize_apply", &bn_finalize_apply);
+  mod.def("bn_bwd_partials", &bn_bwd_partials);
+  mod.def("bn_bwd_apply", &bn_bwd_apply);
   mod.def("vocab_scatter", &vocab_scatter);
+  mod.def("collate_native", &collate_native);
   mod.def("edge_attn_fused_fwd", &edge_attn_fused_fwd);
   mod.def("edge_attn_fused_bwd", &edge_attn_fused_bwd);
   mod.def("embed_grouped_scatter", &embed_grouped_scatter);

@@ -476,12 +476,12 @@ __global__ void bn_bwd_apply_kernel(
     const float* __restrict__ g, const float* __restrict__ x,
     const float* __restrict__ y, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ partials, float* __restrict__ dx, long n, int h,
-    int relu) {
+    const float* __restrict__ partials, float* __restrict__ dx, long n,
+    long count, int h, int relu) {
   const long numel = n * h;
   const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  const float invn = 1.f / n;
+  const float invn = 1.f / count;
   for (long t = i0; t < numel; t += stride) {
     const int c = (int)(t % h);
     float gm = g[t];
@@ -521,6 +521,65 @@ void launch_bn_eval_stats(const float* running_mean, const float* running_var,
                                                         invstd, h, eps);
 }
 
+// --- granular launchers for sync-BN (partials are all-reduced across ranks
+// between the stats and apply phases; `count` = GLOBAL row count) ---
+
+void launch_bn_stats_only(const float* x, long n, int h, float* partials,
+                          hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+  if (n == 0) return;
+  const int nblocks = (int)min((long)512, (n + 63) / 64);
+  bn_stats_partial_kernel<<<nblocks, 256, 0, s>>>(x, n, h, partials);
+}
+
+void launch_bn_finalize_apply(const float* x, const float* partials,
+                              long count, const float* gamma,
+                              const float* beta, float* running_mean,
+                              float* running_var, float* mean, float* invstd,
+                              float* y, long n, int h, float momentum,
+                              float eps, bool training, bool relu,
+                              hipStream_t s) {
+  if (training) {
+    bn_finalize_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
+        partials, count, h, eps, momentum, mean, invstd, running_mean,
+        running_var, 1);
+  } else {
+    bn_eval_stats_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
+        running_mean, running_var, mean, invstd, h, eps);
+  }
+  if (n > 0)
+    bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma,
+                                                    beta, y, n, h,
+                                                    relu ? 1 : 0);
+}
+
+void launch_bn_bwd_partials_only(const float* g, const float* x,
+                                 const float* y, const float* mean,
+                                 const float* invstd, long n, int h, bool relu,
+                                 float* partials, hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+  if (n == 0) return;
+  const int nblocks = (int)min((long)512, (n + 63) / 64);
+  bn_bwd_partial_kernel<<<nblocks, 256, 0, s>>>(g, x, y, mean, invstd, n, h,
+                                                relu ? 1 : 0, partials);
+}
+
+void launch_bn_bwd_apply_only(const float* g, const float* x, const float* y,
+                              const float* mean, const float* invstd,
+                              const float* gamma, const float* partials,
+                              long count, float* dx, long n, int h, bool relu,
+                              hipStream_t s) {
+  if (n == 0) return;
+  bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
+      g, x, y, mean, invstd, gamma, partials, dx, n, count, h, relu ? 1 : 0);
+}
+
+void launch_bn_grad_affine(const float* partials, float* dgamma, float* dbeta,
+                           int h, hipStream_t s) {
+  bn_grad_affine_kernel<<<ceil_div(h, 256), 256, 0, s>>>(partials, dgamma,
+                                                         dbeta, h);
+}
+
 void launch_bn_fwd(const float* x, const float* gamma, const float* beta,
                    float* running_mean, float* running_var, float* mean,
                    float* invstd, float* partials, float* y, long n, int h,
@@ -552,7 +611,7 @@ void launch_bn_bwd(const float* g, const float* x, const float* y,
   bn_bwd_partial_kernel<<<nblocks, 256, 0, s>>>(
       g, x, y, mean, invstd, n, h, relu ? 1 : 0, partials);
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
-      g, x, y, mean, invstd, gamma, partials, dx, n, h, relu ? 1 : 0);
+      g, x, y, mean, invstd, gamma, partials, dx, n, n, h, relu ? 1 : 0);
   bn_grad_affine_kernel<<<ceil_div(h, 256), 256, 0, s>>>(partials, dgamma,
                                                          dbeta, h);
 }

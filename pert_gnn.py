@@ -50,6 +50,10 @@ def build_parser():
     parser.add_argument("--seed", type=int, default=0)
     parser.add_argument("--synthetic", action="store_true",
                         help="generate a synthetic dataset in-place if processed/ is missing")
+    parser.add_argument("--sync_bn", action="store_true",
+                        help="exact-parity BatchNorm under DDP (statistics all-reduced)")
+    parser.add_argument("--precision", choices=["fp32", "bf16"], default="fp32",
+                        help="matmul compute precision on GPU")
     return parser
 
 
@@ -112,9 +116,19 @@ def main(argv=None):
     valid_shard = comm.shard(valid_list)
     test_shard = comm.shard(test_list)
 
-    train_loader = BatchLoader(train_shard, args.batch_size, shuffle=True, seed=args.seed + comm.rank)
-    valid_loader = BatchLoader(valid_shard, args.batch_size, shuffle=False)
-    test_loader = BatchLoader(test_shard, args.batch_size, shuffle=False)
+    on_gpu = torch.cuda.is_available()
+    from pertgnn.data.collate import collate_native
+    from pertgnn.data.prefetch import PrefetchLoader
+
+    collate_fn = (lambda samples: collate_native(samples, pin=True)) if on_gpu else collate_native
+    train_loader = BatchLoader(train_shard, args.batch_size, shuffle=True,
+                               seed=args.seed + comm.rank, collate_fn=collate_fn)
+    valid_loader = BatchLoader(valid_shard, args.batch_size, shuffle=False, collate_fn=collate_fn)
+    test_loader = BatchLoader(test_shard, args.batch_size, shuffle=False, collate_fn=collate_fn)
+    if on_gpu:
+        train_loader = PrefetchLoader(train_loader, comm.device)
+        valid_loader = PrefetchLoader(valid_loader, comm.device)
+        test_loader = PrefetchLoader(test_loader, comm.device)
 
     # vocab scans (pert_gnn.py:306-328)
     unique_ms_max = max(int(g["ms_id"].max()) for g in runtime2graph.values())
@@ -128,6 +142,11 @@ def main(argv=None):
         rpctype_id_max, args.hidden_channels, args.num_layers, args.dropout,
     ).to(device)
     comm.broadcast_module_(model)
+    if args.sync_bn and comm.distributed:
+        model.enable_sync_bn(comm)
+    if torch.cuda.is_available():
+        from pertgnn.ops.functional import set_gemm_precision
+        set_gemm_precision(args.precision)
     optimizer = FusedAdam(model.parameters(), lr=args.lr)
     engine = FlatGradAllReduce(optimizer, comm) if comm.distributed else None
     log = JsonlLogger(args.metrics_jsonl, rank=comm.rank)

@@ -37,6 +37,7 @@ class GraphBatch:
     csc_dst: torch.Tensor            # [E]
     csc_eid: torch.Tensor            # [E] position into CSR edge order
     num_graphs: int
+    batch_ptr: torch.Tensor | None = None  # [B+1] int32 (native collator)
 
     @property
     def csr(self):
@@ -56,6 +57,7 @@ class GraphBatch:
             col_ptr=self.col_ptr.to(**kw), csc_dst=self.csc_dst.to(**kw),
             csc_eid=self.csc_eid.to(**kw),
             num_graphs=self.num_graphs,
+            batch_ptr=self.batch_ptr.to(**kw) if self.batch_ptr is not None else None,
         )
 
     def pin_memory(self):
@@ -65,6 +67,38 @@ class GraphBatch:
                 for f in self.__dataclass_fields__
             }
         )
+
+
+def collate_native(samples, pin: bool = False):
+    """C++ single-pass collator (csrc/collate.cpp, reference K17): concat +
+    edge offset + counting-sort CSR/CSC + batch vector in one native call,
+    optionally into pinned host memory for async H2D."""
+    from ..ops.backend import ext
+
+    m = ext()
+    if m is None or not hasattr(m, "collate_native"):
+        return collate(samples)
+    out = m.collate_native(
+        [s.x for s in samples],
+        [s.edge_index for s in samples],
+        [s.edge_attr for s in samples],
+        [s.cat_X for s in samples],
+        [s.rt_probs for s in samples],
+        [s.pattern_num_nodes for s in samples],
+        [s.node_depth for s in samples],
+        [s.entry_id for s in samples],
+        [s.y for s in samples],
+        pin,
+    )
+    (x, cat_X, edge_index, edge_attr, probs, pnn, nd, entry, batch,
+     batch_ptr, y, row_ptr, csr_src, col_ptr, csc_dst, csc_eid) = out
+    return GraphBatch(
+        x=x, cat_X=cat_X, edge_index=edge_index, edge_attr=edge_attr,
+        rt_probs=probs, pattern_num_nodes=pnn, node_depth=nd,
+        entry_id=entry, batch=batch, y=y, row_ptr=row_ptr, csr_src=csr_src,
+        col_ptr=col_ptr, csc_dst=csc_dst, csc_eid=csc_eid,
+        num_graphs=len(samples), batch_ptr=batch_ptr,
+    )
 
 
 def build_csr(edge_index: torch.Tensor, num_nodes: int):

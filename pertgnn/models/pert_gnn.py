@@ -136,6 +136,13 @@ class SAGEDeterministic(nn.Module):
         # Dead module, never called in forward — kept lazily-uninitialized so the
         # state_dict matches the reference (model.py:68).
         self.edge_linear = nn.LazyLinear(hidden_channels * 2)
+        # sync-BN comm (None = per-replica BN, the fast default under DDP)
+        self._bn_comm = None
+
+    def enable_sync_bn(self, comm):
+        """Exact-parity BN under DDP: batch statistics all-reduced across
+        ranks (SURVEY.md §7 hard part 4). Pass None to disable."""
+        self._bn_comm = comm
 
     def reset_parameters(self):
         for conv in self.convs:
@@ -183,6 +190,7 @@ class SAGEDeterministic(nn.Module):
             x = ops.batchnorm_relu(
                 x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
                 bn.momentum, bn.eps, self.training, fuse_relu=True,
+                comm=self._bn_comm,
             )
             if self.training and bn.track_running_stats and bn.num_batches_tracked is not None:
                 bn.num_batches_tracked += 1

@@ -232,29 +232,114 @@ def embedding(idx, table):
 
 class _BNReLUFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu):
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu, comm):
         m = ext()
-        y, save_mean, save_invstd = m.bn_relu_fwd(
-            x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu
-        )
+        sync = comm is not None and getattr(comm, "distributed", False) and training
+        if sync:
+            # sync-BN (SURVEY.md §7 hard part 4, exact-parity mode): partial
+            # sums + global count all-reduced before normalization.
+            partials = m.bn_stats(x)
+            comm.all_reduce_(partials)
+            count = int(comm.all_reduce_scalar(float(x.shape[0])))
+            y, save_mean, save_invstd = m.bn_finalize_apply(
+                x, partials, count, gamma, beta, running_mean, running_var,
+                momentum, eps, training, fuse_relu)
+        else:
+            count = x.shape[0]
+            y, save_mean, save_invstd = m.bn_relu_fwd(
+                x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu
+            )
         ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
         ctx.fuse_relu = fuse_relu
+        ctx.comm = comm if sync else None
+        ctx.count = count
         return y
 
     @staticmethod
     def backward(ctx, g):
         x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
         m = ext()
-        dx, dgamma, dbeta = m.bn_relu_bwd(
-            g.contiguous(), x, gamma, save_mean, save_invstd, y, ctx.fuse_relu
-        )
-        return dx, dgamma, dbeta, None, None, None, None, None, None
+        g = g.contiguous()
+        if ctx.comm is not None:
+            partials_local = m.bn_bwd_partials(g, x, y, save_mean, save_invstd, ctx.fuse_relu)
+            partials_global = partials_local.clone()
+            ctx.comm.all_reduce_(partials_global)
+            dx, dgamma, dbeta = m.bn_bwd_apply(
+                g, x, y, save_mean, save_invstd, gamma,
+                partials_global, partials_local, ctx.count, ctx.fuse_relu)
+        else:
+            dx, dgamma, dbeta = m.bn_relu_bwd(
+                g, x, gamma, save_mean, save_invstd, y, ctx.fuse_relu
+            )
+        return dx, dgamma, dbeta, None, None, None, None, None, None, None
 
 
-def batchnorm_relu(x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu=True):
-    """BatchNorm1d over N per channel, optional fused ReLU (model.py:101-102)."""
+class _EagerSyncBNFn(torch.autograd.Function):
+    """Eager sync-BN with the explicit cross-rank backward: the gradient of
+    x_i includes every rank's loss through the SHARED mean/var, so the
+    backward all-reduces the per-channel sums of gm and gm*xhat and divides
+    by the GLOBAL count (torch.nn.SyncBatchNorm semantics)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps,
+                fuse_relu, comm):
+        with torch.no_grad():
+            n_local = x.shape[0]
+            h = x.shape[1]
+            pack = torch.cat([x.sum(0), (x * x).sum(0),
+                              torch.tensor([float(n_local)], dtype=x.dtype, device=x.device)])
+            comm.all_reduce_(pack)
+            count = float(pack[-1])
+            mean = pack[:h] / count
+            var = (pack[h:2 * h] / count - mean * mean).clamp_min(0)
+            invstd = (var + eps).rsqrt()
+            unbiased = var * (count / max(count - 1.0, 1.0))
+            running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+            running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+            y = (x - mean) * invstd * gamma + beta
+            if fuse_relu:
+                y = torch.nn.functional.relu(y)
+        ctx.save_for_backward(x, gamma, mean, invstd, y)
+        ctx.fuse_relu = fuse_relu
+        ctx.comm = comm
+        ctx.count = count
+        return y
+
+    @staticmethod
+    def backward(ctx, g):
+        x, gamma, mean, invstd, y = ctx.saved_tensors
+        gm = g.clone()
+        if ctx.fuse_relu:
+            gm[y <= 0] = 0
+        xhat = (x - mean) * invstd
+        s1_local = gm.sum(0)
+        s2_local = (gm * xhat).sum(0)
+        pack = torch.cat([s1_local, s2_local])
+        ctx.comm.all_reduce_(pack)
+        h = x.shape[1]
+        s1_g = pack[:h] / ctx.count
+        s2_g = pack[h:] / ctx.count
+        dx = gamma * invstd * (gm - s1_g - xhat * s2_g)
+        return dx, s2_local, s1_local, None, None, None, None, None, None
+
+
+def _eager_sync_batchnorm(x, gamma, beta, running_mean, running_var, momentum,
+                          eps, fuse_relu, comm):
+    return _EagerSyncBNFn.apply(x, gamma, beta, running_mean, running_var,
+                                momentum, eps, fuse_relu, comm)
+
+
+def batchnorm_relu(x, gamma, beta, running_mean, running_var, momentum, eps,
+                   training, fuse_relu=True, comm=None):
+    """BatchNorm1d over N per channel, optional fused ReLU (model.py:101-102).
+    With ``comm`` (distributed) and training=True, runs sync-BN: statistics
+    over the GLOBAL batch (exact single-process parity)."""
     if use_hip(x):
-        return _BNReLUFn.apply(x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu)
+        return _BNReLUFn.apply(x, gamma, beta, running_mean, running_var,
+                               momentum, eps, training, fuse_relu, comm)
+    if comm is not None and getattr(comm, "distributed", False) and training:
+        return _eager_sync_batchnorm(x, gamma, beta, running_mean, running_var,
+                                     momentum, eps, fuse_relu, comm)
     y = torch.nn.functional.batch_norm(
         x, running_mean, running_var, gamma, beta, training, momentum, eps
     )

@@ -22,6 +22,7 @@ setup(
             name="pertgnn._C",
             sources=[
                 "csrc/bindings.cpp",
+                "csrc/collate.cpp",
                 "csrc/hip/edge_attn.hip",
                 "csrc/hip/edge_attn_fused.hip",
                 "csrc/hip/segops.hip",

@@ -148,3 +148,74 @@ def test_flat_engine_ranks_stay_in_sync():
     assert np.array_equal(g0, g1), "averaged grads must match bitwise"
     assert np.array_equal(p0, p1), "params must stay in sync"
     assert np.abs(g0).sum() > 0
+
+
+def _worker_syncbn(rank, world_size, port, result_queue):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world_size), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    from pertgnn.ops.functional import batchnorm_relu
+    from pertgnn.parallel import Comm
+
+    comm = Comm(backend="gloo")
+    torch.manual_seed(7)
+    full = torch.randn(8, 6) * 2 + 1   # global batch of 8 rows
+    x = full[rank * 4:(rank + 1) * 4].clone().requires_grad_(True)
+    gamma = torch.rand(6).add_(0.5).requires_grad_(True)
+    beta = torch.randn(6).requires_grad_(True)
+    rm = torch.zeros(6)
+    rv = torch.ones(6)
+    y = batchnorm_relu(x, gamma, beta, rm, rv, 0.1, 1e-5, True, fuse_relu=True, comm=comm)
+    # local-mean loss (reference loop semantics per rank)
+    loss = y.pow(2).mean()
+    loss.backward()
+    result_queue.put((rank, x.grad.numpy().copy(), gamma.grad.numpy().copy(),
+                      beta.grad.numpy().copy(), rm.numpy().copy(), rv.numpy().copy(),
+                      y.detach().numpy().copy()))
+    comm.barrier()
+    comm.finalize()
+
+
+@pytest.mark.timeout(120)
+def test_sync_bn_matches_single_process():
+    """2-rank sync-BN == single-process BN over the concatenated batch:
+    forward outputs, running stats, and (after the DDP 1/W averaging) every
+    gradient."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_syncbn, args=(r, world, 29514, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=100) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    results.sort(key=lambda t: t[0])
+
+    # single-process reference over the concatenated batch
+    torch.manual_seed(7)
+    full = torch.randn(8, 6) * 2 + 1
+    x = full.clone().requires_grad_(True)
+    gamma = torch.rand(6).add_(0.5).requires_grad_(True)
+    beta = torch.randn(6).requires_grad_(True)
+    rm = torch.zeros(6)
+    rv = torch.ones(6)
+    y = torch.nn.functional.relu(torch.nn.functional.batch_norm(
+        x, rm, rv, gamma, beta, True, 0.1, 1e-5))
+    # average of per-rank local-mean losses == mean over global batch
+    loss = (y[:4].pow(2).mean() + y[4:].pow(2).mean()) / 2
+    loss.backward()
+
+    (_, dx0, dg0, db0, rm0, rv0, y0), (_, dx1, dg1, db1, rm1, rv1, y1) = results
+    import numpy as np
+    assert np.allclose(np.concatenate([y0, y1]), y.detach().numpy(), atol=1e-5)
+    assert np.allclose(rm0, rm.numpy(), atol=1e-6) and np.allclose(rm1, rm.numpy(), atol=1e-6)
+    assert np.allclose(rv0, rv.numpy(), atol=1e-5)
+    # DDP engine averages grads across ranks: (g0+g1)/2 == single-process
+    assert np.allclose((dg0 + dg1) / 2, gamma.grad.numpy(), atol=1e-5)
+    assert np.allclose((db0 + db1) / 2, beta.grad.numpy(), atol=1e-5)
+    # dx: per-rank dx must equal W * single-process dx rows (engine divides
+    # weight grads, and dx feeds W-averaged weight grads upstream)
+    assert np.allclose(dx0, 2 * x.grad.numpy()[:4], atol=1e-5), np.abs(dx0 - 2*x.grad.numpy()[:4]).max()
+    assert np.allclose(dx1, 2 * x.grad.numpy()[4:], atol=1e-5)

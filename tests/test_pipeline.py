@@ -143,3 +143,30 @@ def test_end_to_end_training_epoch(synthetic_workspace):
                 first = float(loss)
             last = float(loss)
     assert last < first
+
+
+def test_native_collator_matches_python(synthetic_workspace):
+    """csrc/collate.cpp vs the torch collator — identical batch content."""
+    from pertgnn.ops.backend import ext
+    import pytest as _pytest
+    if ext() is None:
+        _pytest.skip("extension not built")
+    from pertgnn.data.collate import collate_native
+    root, (tr2data, entry2runtimes, _, runtime2pert, resource_df) = synthetic_workspace
+    data_list = build_data_list(tr2data, entry2runtimes, runtime2pert, resource_df, limit=40)
+    py = collate(data_list[:12])
+    nat = collate_native(data_list[:12])
+    assert torch.equal(py.x, nat.x)
+    assert torch.equal(py.cat_X, nat.cat_X)
+    assert torch.equal(py.edge_index, nat.edge_index)
+    assert torch.equal(py.edge_attr, nat.edge_attr)
+    assert torch.equal(py.rt_probs, nat.rt_probs)
+    assert torch.equal(py.batch, nat.batch)
+    assert torch.equal(py.y, nat.y)
+    assert torch.equal(py.row_ptr, nat.row_ptr)
+    assert torch.equal(py.csr_src, nat.csr_src)
+    assert torch.equal(py.col_ptr, nat.col_ptr)
+    assert torch.equal(py.csc_dst, nat.csc_dst)
+    assert torch.equal(py.csc_eid, nat.csc_eid)
+    assert torch.equal(nat.batch_ptr[1:].long(), torch.cumsum(
+        torch.bincount(py.batch, minlength=py.num_graphs), 0).int().long())
