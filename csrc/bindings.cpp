@@ -66,8 +66,8 @@ void launch_quantile_loss_bwd(const float*, const float*, const float*, float*,
                               long, float, hipStream_t);
 void launch_eval_metrics(const float*, const float*, float*, long, float,
                          hipStream_t);
-void launch_adam(float*, const float*, float*, float*, long, float, float,
-                 float, float, int, hipStream_t);
+void launch_adam(float*, const float*, float*, float*, float*, long, float,
+                 float, float, float, hipStream_t);
 void launch_gemm_f32_nt(const float*, const float*, const float*, float*, int,
                         int, int, bool, hipStream_t);
 void launch_gemm_f32_nn(const float*, const float*, const float*, float*, int,
@@ -324,12 +324,12 @@ std::vector<torch::Tensor> eval_metrics(torch::Tensor y, torch::Tensor y_hat,
 }
 
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
-               torch::Tensor v, double lr, double b1, double b2, double eps,
-               int64_t step) {
-  CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v);
+               torch::Tensor v, torch::Tensor state, double lr, double b1,
+               double b2, double eps) {
+  CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v); CHECK_IN(state);
   launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
-              v.data_ptr<float>(), p.numel(), (float)lr, (float)b1, (float)b2,
-              (float)eps, (int)step, cur_stream());
+              v.data_ptr<float>(), state.data_ptr<float>(), p.numel(),
+              (float)lr, (float)b1, (float)b2, (float)eps, cur_stream());
 }
 
 // y = x @ w^T + b (torch Linear layout: w [out,in]); bf16 variant rounds

@@ -710,30 +710,41 @@ void launch_eval_metrics(const float* y, const float* y_hat, float* out3,
 // Adam on flat master buffers (torch.optim.Adam math, K14)
 // ---------------------------------------------------------------------------
 
+// step/bias state lives ON DEVICE (state[0]=step, state[1]=1-b1^t,
+// state[2]=1-b2^t) so the whole Adam update is hipGraph-replayable: each
+// replay increments the step and recomputes the bias corrections.
+__global__ void adam_tick_kernel(float* __restrict__ state, float b1,
+                                 float b2) {
+  const float t = state[0] + 1.f;
+  state[0] = t;
+  state[1] = 1.f - powf(b1, t);
+  state[2] = 1.f - powf(b2, t);
+}
+
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
-                            long numel, float lr, float b1, float b2,
-                            float eps, float bias1, float bias2) {
+                            const float* __restrict__ state, long numel,
+                            float lr, float b1, float b2, float eps) {
   const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  const float step_size = lr / bias1;
+  const float step_size = lr / state[1];
+  const float sqrt_bias2 = sqrtf(state[2]);
   for (long t = i0; t < numel; t += stride) {
     const float gt = g[t];
     const float mt = b1 * m[t] + (1.f - b1) * gt;
     const float vt = b2 * v[t] + (1.f - b2) * gt * gt;
     m[t] = mt;
     v[t] = vt;
-    const float denom = sqrtf(vt) / sqrtf(bias2) + eps;
+    const float denom = sqrtf(vt) / sqrt_bias2 + eps;
     p[t] -= step_size * mt / denom;
   }
 }
 
-void launch_adam(float* p, const float* g, float* m, float* v, long numel,
-                 float lr, float b1, float b2, float eps, int step,
+void launch_adam(float* p, const float* g, float* m, float* v, float* state,
+                 long numel, float lr, float b1, float b2, float eps,
                  hipStream_t s) {
   if (numel == 0) return;
-  const float bias1 = 1.f - powf(b1, (float)step);
-  const float bias2 = 1.f - powf(b2, (float)step);
- hipLaunchKernelGGL(( adam_kernel), dim3(grid_for(numel)), dim3(256), 0, s, p, g, m, v, numel, lr, b1, b2,
-                                              eps, bias1, bias2);
+ hipLaunchKernelGGL(( adam_tick_kernel), dim3(1), dim3(1), 0, s, state, b1, b2);
+ hipLaunchKernelGGL(( adam_kernel), dim3(grid_for(numel)), dim3(256), 0, s, p, g, m, v, state, numel, lr,
+                                              b1, b2, eps);
 }

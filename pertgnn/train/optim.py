@@ -40,6 +40,8 @@ class FusedAdam:
         self.flat_grad = torch.zeros(total, dtype=torch.float32, device=device)
         self.exp_avg = torch.zeros(total, dtype=torch.float32, device=device)
         self.exp_avg_sq = torch.zeros(total, dtype=torch.float32, device=device)
+        # device-side [step, 1-b1^t, 1-b2^t] so the step is hipGraph-replayable
+        self.dev_state = torch.zeros(3, dtype=torch.float32, device=device)
         self.offsets = []
         off = 0
         for p in self.params:
@@ -63,8 +65,8 @@ class FusedAdam:
         b1, b2 = self.betas
         if self.flat_param.is_cuda and has_hip():
             ext().adam_step(self.flat_param, self.flat_grad, self.exp_avg,
-                            self.exp_avg_sq, self.lr, b1, b2, self.eps,
-                            self.step_count)
+                            self.exp_avg_sq, self.dev_state, self.lr, b1, b2,
+                            self.eps)
             return
         # eager fallback — identical formula (torch.optim.Adam)
         g = self.flat_grad
@@ -79,6 +81,7 @@ class FusedAdam:
     def state_dict(self):
         return {
             "step": self.step_count,
+            "dev_state": self.dev_state,
             "exp_avg": self.exp_avg,
             "exp_avg_sq": self.exp_avg_sq,
             "lr": self.lr,
@@ -88,6 +91,8 @@ class FusedAdam:
 
     def load_state_dict(self, sd):
         self.step_count = sd["step"]
+        if "dev_state" in sd:
+            self.dev_state.copy_(sd["dev_state"].to(self.dev_state.device))
         self.exp_avg.copy_(sd["exp_avg"].to(self.exp_avg.device))
         self.exp_avg_sq.copy_(sd["exp_avg_sq"].to(self.exp_avg_sq.device))
         self.lr = sd.get("lr", self.lr)

@@ -215,8 +215,9 @@ def test_adam_parity():
     p_h = p0.clone().to(DEV)
     m = torch.zeros(numel, device=DEV)
     v = torch.zeros(numel, device=DEV)
+    state = torch.zeros(3, device=DEV)
     for step in range(1, 4):
-        C.adam_step(p_h, grad.to(DEV), m, v, 1e-3, 0.9, 0.999, 1e-8, step)
+        C.adam_step(p_h, grad.to(DEV), m, v, state, 1e-3, 0.9, 0.999, 1e-8)
     torch.cuda.synchronize()
     assert torch.allclose(p_h.cpu(), p_t.detach(), atol=1e-6), \
         (p_h.cpu() - p_t.detach()).abs().max()
