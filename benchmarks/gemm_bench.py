@@ -3,9 +3,13 @@
     python benchmarks/gemm_bench.py [--iters 50]
 """
 import argparse
+import sys
 import time
+from pathlib import Path
 
 import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
 
 
 def bench(fn, iters, *args):
