@@ -41,16 +41,21 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
   const int f = t / QUADS;             // 16 free rows per pass
   const int cq = (t % QUADS) * 4;
   constexpr int FSTEP = BGEMM_THREADS / QUADS;  // 16
-  const bool fast = (free0 + BF <= free_max) &&
-                    (contract0 + BGEMM_BK <= contract_max) && ((ld & 3) == 0);
+  const bool interior = (free0 + BF <= free_max) &&
+                        (contract0 + BGEMM_BK <= contract_max);
+  const bool aligned = (ld & 3) == 0;
 #pragma unroll
   for (int half = 0; half < BF / FSTEP; ++half) {
     const int ff = f + half * FSTEP;
-    if (fast) {
+    if (interior && aligned) {
       const f32x4 v = *reinterpret_cast<const f32x4*>(
           &g[(long)(free0 + ff) * ld + contract0 + cq]);
       *reinterpret_cast<bf16x4*>(&lds[ff * LDW + cq]) =
           pack4(v[0], v[1], v[2], v[3]);
+    } else if (interior) {  // odd leading dim (layer-1 K=9+H): unchecked scalars
+      const float* row = &g[(long)(free0 + ff) * ld + contract0 + cq];
+      *reinterpret_cast<bf16x4*>(&lds[ff * LDW + cq]) =
+          pack4(row[0], row[1], row[2], row[3]);
     } else {
       const int gf = free0 + ff;
       float v[4];
@@ -80,18 +85,22 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
   const int cb = (t / FQUADS) * 4;     // contract block of 4
   const int fq = (t % FQUADS) * 4;     // free quad
   constexpr int CSTEP = (BGEMM_THREADS / FQUADS) * 4;
-  const bool fast = (contract0 + BGEMM_BK <= contract_max) &&
-                    (free0 + BF <= free_max) && ((ld & 3) == 0) &&
-                    ((free0 & 3) == 0);
+  const bool interior = (contract0 + BGEMM_BK <= contract_max) &&
+                        (free0 + BF <= free_max);
+  const bool aligned = ((ld & 3) == 0) && ((free0 & 3) == 0);
 #pragma unroll
   for (int half = 0; half < BGEMM_BK / CSTEP; ++half) {
     const int cc = cb + half * CSTEP;
     float v[4][4];
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
-      if (fast) {
+      if (interior && aligned) {
         *reinterpret_cast<f32x4*>(v[u]) = *reinterpret_cast<const f32x4*>(
             &g[(long)(contract0 + cc + u) * ld + free0 + fq]);
+      } else if (interior) {
+        const float* row = &g[(long)(contract0 + cc + u) * ld + free0 + fq];
+#pragma unroll
+        for (int w = 0; w < 4; ++w) v[u][w] = row[w];
       } else {
         const int gc = contract0 + cc + u;
 #pragma unroll

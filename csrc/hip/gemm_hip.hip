@@ -37,9 +37,9 @@ __device__ __forceinline__ void stage_transpose(
   const int t = threadIdx.x;
   const int fr = t / 8;           // 32 free rows per pass
   const int cq = (t % 8) * 4;     // contract quad
-  const bool fast = (free0 + BF <= free_max) &&
-                    (contract0 + GEMM_BK <= contract_max) && ((ld & 3) == 0);
-  if (fast) {
+  const bool interior = (free0 + BF <= free_max) &&
+                        (contract0 + GEMM_BK <= contract_max);
+  if (interior && (ld & 3) == 0) {
 #pragma unroll
     for (int half = 0; half < BF / 32; ++half) {
       const int f = fr + half * 32;
@@ -47,6 +47,16 @@ __device__ __forceinline__ void stage_transpose(
           &g[(long)(free0 + f) * ld + contract0 + cq]);
 #pragma unroll
       for (int u = 0; u < 4; ++u) lds[(cq + u) * BF + f] = v[u];
+    }
+    return;
+  }
+  if (interior) {  // odd leading dim: unchecked scalar loads
+#pragma unroll
+    for (int half = 0; half < BF / 32; ++half) {
+      const int f = fr + half * 32;
+      const float* row = &g[(long)(free0 + f) * ld + contract0 + cq];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) lds[(cq + u) * BF + f] = row[u];
     }
     return;
   }
@@ -75,16 +85,25 @@ __device__ __forceinline__ void stage_direct(
   const int c = t / QUADS;                // contract rows per pass
   const int fq = (t % QUADS) * 4;
   constexpr int CSTEP = GEMM_THREADS / QUADS;
-  const bool fast = (contract0 + GEMM_BK <= contract_max) &&
-                    (free0 + BF <= free_max) && ((ld & 3) == 0) &&
-                    ((free0 & 3) == 0);
-  if (fast) {
+  const bool interior = (contract0 + GEMM_BK <= contract_max) &&
+                        (free0 + BF <= free_max);
+  if (interior && ((ld & 3) == 0) && ((free0 & 3) == 0)) {
 #pragma unroll
     for (int half = 0; half < GEMM_BK / CSTEP; ++half) {
       const int cc = c + half * CSTEP;
       const f32x4 v = *reinterpret_cast<const f32x4*>(
           &g[(long)(contract0 + cc) * ld + free0 + fq]);
       *reinterpret_cast<f32x4*>(&lds[cc * BF + fq]) = v;
+    }
+    return;
+  }
+  if (interior) {
+#pragma unroll
+    for (int half = 0; half < GEMM_BK / CSTEP; ++half) {
+      const int cc = c + half * CSTEP;
+      const float* row = &g[(long)(contract0 + cc) * ld + free0 + fq];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) lds[cc * BF + fq + u] = row[u];
     }
     return;
   }
