@@ -26,7 +26,7 @@ void launch_edge_attn_bwd(const float*, const float*, const float*,
                           float*, float*, float*, float*, int, int, long,
                           hipStream_t);
 void launch_seg_pool_fwd(const float*, const float*, const float*, const int*,
-                         float*, int, int, hipStream_t);
+                         float*, float*, int, int, int, hipStream_t);
 void launch_seg_pool_bwd(const float*, const float*, const float*, const long*,
                          float*, long, int, hipStream_t);
 void launch_embed_node_fwd(const float*, const long*, const float*, float*,
@@ -160,10 +160,17 @@ torch::Tensor seg_pool_fwd(torch::Tensor x, torch::Tensor probs,
                            int64_t num_graphs) {
   CHECK_IN(x); CHECK_IN(probs); CHECK_IN(nn); CHECK_IN(batch_ptr);
   const int h = x.size(1);
+  const long n = x.size(0);
+  long p = 8192 / std::max<int64_t>(num_graphs, 1);
+  const long per_g = (n + std::max<int64_t>(num_graphs, 1) - 1) /
+                     std::max<int64_t>(num_graphs, 1);
+  p = std::min<long>(std::max<long>(std::min(p, per_g), 1), 64);
+  auto partial = torch::empty({num_graphs * p, h}, x.options());
   auto out = torch::empty({num_graphs, h}, x.options());
   launch_seg_pool_fwd(x.data_ptr<float>(), probs.data_ptr<float>(),
                       nn.data_ptr<float>(), batch_ptr.data_ptr<int>(),
-                      out.data_ptr<float>(), (int)num_graphs, h, cur_stream());
+                      partial.data_ptr<float>(), out.data_ptr<float>(),
+                      (int)num_graphs, (int)p, h, cur_stream());
   return out;
 }
 

@@ -14,12 +14,45 @@
 
 #define WAVES_PER_BLOCK 4
 
+// two-level: P sub-waves per graph each sum a strided node interleave into
+// partial[g*P+sub], then one wave per graph folds the P partials in order
+// (deterministic, no atomics; single waves per graph would leave the chip
+// ~64/256 CUs busy at trace-scale batches).
 template <int VPT>
-__global__ void seg_pool_fwd_kernel(const float* __restrict__ x,
-                                    const float* __restrict__ probs,
-                                    const float* __restrict__ nn,
-                                    const int* __restrict__ batch_ptr,
-                                    float* __restrict__ out, int b, int h) {
+__global__ void seg_pool_p1_kernel(const float* __restrict__ x,
+                                   const float* __restrict__ probs,
+                                   const float* __restrict__ nn,
+                                   const int* __restrict__ batch_ptr,
+                                   float* __restrict__ partial, int b, int P,
+                                   int h) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int w = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (w >= b * P) return;
+  const int g = w / P;
+  const int sub = w % P;
+  float acc[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
+  for (int i = batch_ptr[g] + sub; i < batch_ptr[g + 1]; i += P) {
+    const float wt = probs[i] / nn[i];
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) acc[j] += x[(long)i * h + c] * wt;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    if (c < h) partial[(long)w * h + c] = acc[j];
+  }
+}
+
+template <int VPT>
+__global__ void seg_pool_p2_kernel(const float* __restrict__ partial,
+                                   float* __restrict__ out, int b, int P,
+                                   int h) {
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
   const int g = blockIdx.x * WAVES_PER_BLOCK + wid;
@@ -27,12 +60,11 @@ __global__ void seg_pool_fwd_kernel(const float* __restrict__ x,
   float acc[VPT];
 #pragma unroll
   for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
-  for (int i = batch_ptr[g]; i < batch_ptr[g + 1]; ++i) {
-    const float w = probs[i] / nn[i];
+  for (int sub = 0; sub < P; ++sub) {
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
       const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) acc[j] += x[(long)i * h + c] * w;
+      if (c < h) acc[j] += partial[((long)g * P + sub) * h + c];
     }
   }
 #pragma unroll
@@ -58,17 +90,20 @@ __global__ void seg_pool_bwd_kernel(const float* __restrict__ gout,
 }
 
 void launch_seg_pool_fwd(const float* x, const float* probs, const float* nn,
-                         const int* batch_ptr, float* out, int b, int h,
-                         hipStream_t stream) {
+                         const int* batch_ptr, float* partial, float* out,
+                         int b, int P, int h, hipStream_t stream) {
   if (b == 0) return;
-  const dim3 grid(ceil_div(b, WAVES_PER_BLOCK));
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
   const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
   switch (vpt) {
 #define CASE(V)                                                                \
   case V:                                                                      \
-   hipLaunchKernelGGL(( seg_pool_fwd_kernel<V>), dim3(grid), dim3(block), 0, stream, x, probs, nn,           \
-                                                       batch_ptr, out, b, h); \
+   hipLaunchKernelGGL(( seg_pool_p1_kernel<V>)                                                      \
+        , dim3(dim3(ceil_div((long)b * P, WAVES_PER_BLOCK))), dim3(block), 0, stream,   \
+            x, probs, nn, batch_ptr, partial, b, P, h);                        \
+   hipLaunchKernelGGL(( seg_pool_p2_kernel<V>)                                                      \
+        , dim3(dim3(ceil_div(b, WAVES_PER_BLOCK))), dim3(block), 0, stream,             \
+            partial, out, b, P, h);                                            \
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
@@ -282,7 +317,7 @@ void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
         (const void*)vocab_scatter_kernel,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
-  const int blocks = (int)min((long)256, (n + 63) / 64);
+  const int blocks = (int)min((long)128, (n + 63) / 64);
  hipLaunchKernelGGL(( vocab_scatter_kernel), dim3(dim3(blocks)), dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)),
                          lds, s, g, idx, idx_stride, dtable, n, rows, h,
                                    gstride, col_off);
