@@ -317,7 +317,7 @@ void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
         (const void*)vocab_scatter_kernel,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
-  const int blocks = (int)min((long)128, (n + 63) / 64);
+  const int blocks = (int)min((long)256, (n + 63) / 64);
  hipLaunchKernelGGL(( vocab_scatter_kernel), dim3(dim3(blocks)), dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)),
                          lds, s, g, idx, idx_stride, dtable, n, rows, h,
                                    gstride, col_off);
