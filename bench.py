@@ -104,7 +104,7 @@ def main():
     ap.add_argument("--n-batches", type=int, default=8, help="distinct resident batches to cycle")
     ap.add_argument("--lr", type=float, default=3e-4)
     ap.add_argument("--tau", type=float, default=0.5)
-    ap.add_argument("--precision", choices=["fp32", "bf16"], default="bf16",
+    ap.add_argument("--precision", choices=["fp32", "bf16", "fp16"], default="bf16",
                     help="matmul compute precision (weights/activations stay fp32)")
     ap.add_argument("--no-hipgraph", action="store_true",
                     help="disable hipGraph capture of the training step")

@@ -81,6 +81,12 @@ void launch_gemm_bf16_nn(const float*, const float*, const float*, float*,
                          int, int, int, bool, hipStream_t);
 void launch_gemm_bf16_tn(const float*, const float*, float*, float*, int, int,
                          int, hipStream_t);
+void launch_gemm_fp16_nt(const float*, const float*, const float*, float*,
+                         int, int, int, bool, hipStream_t);
+void launch_gemm_fp16_nn(const float*, const float*, const float*, float*,
+                         int, int, int, bool, hipStream_t);
+void launch_gemm_fp16_tn(const float*, const float*, float*, float*, int, int,
+                         int, hipStream_t);
 void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   float*, long, int, int, int, int, int,
                                   hipStream_t);
@@ -401,6 +407,41 @@ torch::Tensor gemm_tn_bf16(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor linear_fwd_fp16(torch::Tensor x, torch::Tensor w,
+                              torch::Tensor b) {
+  CHECK_IN(x); CHECK_IN(w);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = w.size(0);
+  auto y = torch::empty({m, n}, x.options());
+  const float* bias = nullptr;
+  if (b.defined() && b.numel() > 0) bias = b.data_ptr<float>();
+  launch_gemm_fp16_nt(x.data_ptr<float>(), w.data_ptr<float>(), bias,
+                      y.data_ptr<float>(), m, n, k, false, cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> linear_bwd_fp16(torch::Tensor g, torch::Tensor x,
+                                           torch::Tensor w, bool has_bias) {
+  CHECK_IN(g); CHECK_IN(x); CHECK_IN(w);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = w.size(0);
+  auto dx = torch::empty({m, k}, x.options());
+  auto dw = torch::empty({n, k}, w.options());
+  torch::Tensor db = torch::empty({0}, g.options());
+  float* db_ptr = nullptr;
+  if (has_bias) {
+    db = torch::empty({n}, g.options());
+    db_ptr = db.data_ptr<float>();
+  }
+  launch_gemm_fp16_nn(g.data_ptr<float>(), w.data_ptr<float>(), nullptr,
+                      dx.data_ptr<float>(), m, n, k, false, cur_stream());
+  launch_gemm_fp16_tn(g.data_ptr<float>(), x.data_ptr<float>(),
+                      dw.data_ptr<float>(), db_ptr, m, n, k, cur_stream());
+  return {dx, dw, db};
+}
+
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b) {
   CHECK_IN(x); CHECK_IN(w);
   const int m = x.size(0);
@@ -621,6 +662,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("embed_grouped_scatter", &embed_grouped_scatter);
   mod.def("linear_fwd", &linear_fwd);
   mod.def("linear_fwd_bf16", &linear_fwd_bf16);
+  mod.def("linear_fwd_fp16", &linear_fwd_fp16);
+  mod.def("linear_bwd_fp16", &linear_bwd_fp16);
   mod.def("linear_bwd_bf16", &linear_bwd_bf16);
   mod.def("gemm_nt_bf16", &gemm_nt_bf16);
   mod.def("gemm_nn_bf16", &gemm_nn_bf16);

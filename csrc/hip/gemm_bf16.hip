@@ -18,23 +18,43 @@
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((ext_vector_type(8))) _Float16 f16x8;
+typedef __attribute__((ext_vector_type(4))) _Float16 f16x4;
 
-__device__ __forceinline__ __bf16 to_bf16(float x) { return (__bf16)x; }
+// element-type traits: bf16 and fp16 share the whole kernel structure; only
+// the packed vector types and the MFMA intrinsic differ.
+template <typename T16> struct Vec16;
+template <> struct Vec16<__bf16> {
+  using v4 = bf16x4;
+  using v8 = bf16x8;
+  static __device__ __forceinline__ f32x4 mfma(v8 a, v8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <> struct Vec16<_Float16> {
+  using v4 = f16x4;
+  using v8 = f16x8;
+  static __device__ __forceinline__ f32x4 mfma(v8 a, v8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+};
 
-__device__ __forceinline__ bf16x4 pack4(float a, float b, float c, float d) {
-  bf16x4 r;
-  r[0] = (__bf16)a; r[1] = (__bf16)b; r[2] = (__bf16)c; r[3] = (__bf16)d;
+template <typename T16>
+__device__ __forceinline__ typename Vec16<T16>::v4 pack4(float a, float b,
+                                                         float c, float d) {
+  typename Vec16<T16>::v4 r;
+  r[0] = (T16)a; r[1] = (T16)b; r[2] = (T16)c; r[3] = (T16)d;
   return r;
 }
 
 // global [free][contract] (contract-minor) -> LDS [free][BK+PAD]
 // 4 bf16 packed into one 8-byte LDS write (scalar u16 LDS writes are ~2x
 // slower — guide G13 applies to LDS too).
-template <int BF>
+template <int BF, typename T16>
 __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
                                             long ld, int free0, int contract0,
                                             int free_max, int contract_max,
-                                            __bf16* lds) {
+                                            T16* lds) {
   constexpr int LDW = BGEMM_BK + BGEMM_PAD;
   const int t = threadIdx.x;
   constexpr int QUADS = BGEMM_BK / 4;  // 16 float4 per row
@@ -50,12 +70,12 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
     if (interior && aligned) {
       const f32x4 v = *reinterpret_cast<const f32x4*>(
           &g[(long)(free0 + ff) * ld + contract0 + cq]);
-      *reinterpret_cast<bf16x4*>(&lds[ff * LDW + cq]) =
-          pack4(v[0], v[1], v[2], v[3]);
+      *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[ff * LDW + cq]) =
+          pack4<T16>(v[0], v[1], v[2], v[3]);
     } else if (interior) {  // odd leading dim (layer-1 K=9+H): unchecked scalars
       const float* row = &g[(long)(free0 + ff) * ld + contract0 + cq];
-      *reinterpret_cast<bf16x4*>(&lds[ff * LDW + cq]) =
-          pack4(row[0], row[1], row[2], row[3]);
+      *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[ff * LDW + cq]) =
+          pack4<T16>(row[0], row[1], row[2], row[3]);
     } else {
       const int gf = free0 + ff;
       float v[4];
@@ -65,8 +85,8 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
         v[u] = (gf < free_max && gc < contract_max) ? g[(long)gf * ld + gc]
                                                     : 0.f;
       }
-      *reinterpret_cast<bf16x4*>(&lds[ff * LDW + cq]) =
-          pack4(v[0], v[1], v[2], v[3]);
+      *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[ff * LDW + cq]) =
+          pack4<T16>(v[0], v[1], v[2], v[3]);
     }
   }
 }
@@ -74,11 +94,11 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
 // global [contract][free] (contract-major) -> LDS [free][BK+PAD]: each thread
 // transposes a 4x4 block in registers (4 coalesced f32x4 loads from 4
 // contract rows), then writes 4 packed 8-byte LDS rows.
-template <int BF>
+template <int BF, typename T16>
 __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
                                             long ld, int contract0, int free0,
                                             int contract_max, int free_max,
-                                            __bf16* lds) {
+                                            T16* lds) {
   constexpr int LDW = BGEMM_BK + BGEMM_PAD;
   const int t = threadIdx.x;
   constexpr int FQUADS = BF / 4;
@@ -114,8 +134,8 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
     }
 #pragma unroll
     for (int w = 0; w < 4; ++w)
-      *reinterpret_cast<bf16x4*>(&lds[(fq + w) * LDW + cc]) =
-          pack4(v[0][w], v[1][w], v[2][w], v[3][w]);
+      *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[(fq + w) * LDW + cc]) =
+          pack4<T16>(v[0][w], v[1][w], v[2][w], v[3][w]);
   }
 }
 
@@ -125,7 +145,7 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
 // elements at k = (l>>4)*8 of row/col (l&15); C/D: col=l&15, row=(l>>4)*4+r.
 // ---------------------------------------------------------------------------
 
-template <int FM, int FN>
+template <int FM, int FN, typename T16>
 struct BWaveTile {
   f32x4 acc[FM][FN];
   __device__ __forceinline__ void zero() {
@@ -134,7 +154,7 @@ struct BWaveTile {
 #pragma unroll
       for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
   }
-  __device__ __forceinline__ void mma(const __bf16* lds_a, const __bf16* lds_b,
+  __device__ __forceinline__ void mma(const T16* lds_a, const T16* lds_b,
                                       int wm, int wn, int lane) {
     constexpr int LDW = BGEMM_BK + BGEMM_PAD;
     const int fi = lane & 15;
@@ -142,21 +162,21 @@ struct BWaveTile {
 #pragma unroll
     for (int s = 0; s < BGEMM_BK / 32; ++s) {
       const int k = s * 32 + fk;
-      bf16x8 a[FM], b[FN];
+      using v8 = typename Vec16<T16>::v8;
+      v8 a[FM], b[FN];
 #pragma unroll
       for (int mi = 0; mi < FM; ++mi)
-        a[mi] = *reinterpret_cast<const bf16x8*>(
+        a[mi] = *reinterpret_cast<const v8*>(
             &lds_a[(wm + mi * 16 + fi) * LDW + k]);
 #pragma unroll
       for (int ni = 0; ni < FN; ++ni)
-        b[ni] = *reinterpret_cast<const bf16x8*>(
+        b[ni] = *reinterpret_cast<const v8*>(
             &lds_b[(wn + ni * 16 + fi) * LDW + k]);
 #pragma unroll
       for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
         for (int ni = 0; ni < FN; ++ni)
-          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+          acc[mi][ni] = Vec16<T16>::mfma(a[mi], b[ni], acc[mi][ni]);
     }
   }
   __device__ __forceinline__ void store(float* __restrict__ c, long ldc,
@@ -187,7 +207,7 @@ struct BWaveTile {
 // kernels — same three layouts as the f32 suite
 // ---------------------------------------------------------------------------
 
-template <int BM, int BN>
+template <int BM, int BN, typename T16 = __bf16>
 __launch_bounds__(BGEMM_THREADS)
 __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
                                     const float* __restrict__ b,
@@ -196,8 +216,8 @@ __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
                                     int relu) {
   constexpr int LDW = BGEMM_BK + BGEMM_PAD;
   constexpr int FM = BM / 32, FN = BN / 32;
-  __shared__ __bf16 lds_a[2][BM * LDW];
-  __shared__ __bf16 lds_b[2][BN * LDW];
+  __shared__ T16 lds_a[2][BM * LDW];
+  __shared__ T16 lds_b[2][BN * LDW];
   const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_n = (n + BN - 1) / BN;
   const int m0 = (bid / tiles_n) * BM;
@@ -207,15 +227,15 @@ __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
   const int wm = (wave >> 1) * (BM / 2);
   const int wn = (wave & 1) * (BN / 2);
 
-  BWaveTile<FM, FN> wt;
+  BWaveTile<FM, FN, T16> wt;
   wt.zero();
   int buf = 0;
-  bstage_cmin<BM>(a, k, m0, 0, m, k, lds_a[0]);
-  bstage_cmin<BN>(b, k, n0, 0, n, k, lds_b[0]);
+  bstage_cmin<BM, T16>(a, k, m0, 0, m, k, lds_a[0]);
+  bstage_cmin<BN, T16>(b, k, n0, 0, n, k, lds_b[0]);
   __syncthreads();
   for (int k0 = BGEMM_BK; k0 < k; k0 += BGEMM_BK) {
-    bstage_cmin<BM>(a, k, m0, k0, m, k, lds_a[buf ^ 1]);
-    bstage_cmin<BN>(b, k, n0, k0, n, k, lds_b[buf ^ 1]);
+    bstage_cmin<BM, T16>(a, k, m0, k0, m, k, lds_a[buf ^ 1]);
+    bstage_cmin<BN, T16>(b, k, n0, k0, n, k, lds_b[buf ^ 1]);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
     __syncthreads();
     buf ^= 1;
@@ -224,7 +244,7 @@ __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
   wt.store(c, n, m0 + wm, n0 + wn, m, n, bias, relu, lane);
 }
 
-template <int BM, int BN>
+template <int BM, int BN, typename T16 = __bf16>
 __launch_bounds__(BGEMM_THREADS)
 __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
                                     const float* __restrict__ b,
@@ -233,8 +253,8 @@ __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
                                     int k2, int relu) {
   constexpr int LDW = BGEMM_BK + BGEMM_PAD;
   constexpr int FM = BM / 32, FN = BN / 32;
-  __shared__ __bf16 lds_a[2][BM * LDW];
-  __shared__ __bf16 lds_b[2][BN * LDW];
+  __shared__ T16 lds_a[2][BM * LDW];
+  __shared__ T16 lds_b[2][BN * LDW];
   const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_n = (k2 + BN - 1) / BN;
   const int m0 = (bid / tiles_n) * BM;
@@ -244,15 +264,15 @@ __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
   const int wm = (wave >> 1) * (BM / 2);
   const int wn = (wave & 1) * (BN / 2);
 
-  BWaveTile<FM, FN> wt;
+  BWaveTile<FM, FN, T16> wt;
   wt.zero();
   int buf = 0;
-  bstage_cmin<BM>(a, n, m0, 0, m, n, lds_a[0]);
-  bstage_cmaj<BN>(b, k2, 0, n0, n, k2, lds_b[0]);
+  bstage_cmin<BM, T16>(a, n, m0, 0, m, n, lds_a[0]);
+  bstage_cmaj<BN, T16>(b, k2, 0, n0, n, k2, lds_b[0]);
   __syncthreads();
   for (int c0 = BGEMM_BK; c0 < n; c0 += BGEMM_BK) {
-    bstage_cmin<BM>(a, n, m0, c0, m, n, lds_a[buf ^ 1]);
-    bstage_cmaj<BN>(b, k2, c0, n0, n, k2, lds_b[buf ^ 1]);
+    bstage_cmin<BM, T16>(a, n, m0, c0, m, n, lds_a[buf ^ 1]);
+    bstage_cmaj<BN, T16>(b, k2, c0, n0, n, k2, lds_b[buf ^ 1]);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
     __syncthreads();
     buf ^= 1;
@@ -261,7 +281,7 @@ __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
   wt.store(c, k2, m0 + wm, n0 + wn, m, k2, bias, relu, lane);
 }
 
-template <int BM, int BN>
+template <int BM, int BN, typename T16 = __bf16>
 __launch_bounds__(BGEMM_THREADS)
 __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
                                     const float* __restrict__ b,
@@ -270,8 +290,8 @@ __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
                                     int k2, int slices) {
   constexpr int LDW = BGEMM_BK + BGEMM_PAD;
   constexpr int FM = BM / 32, FN = BN / 32;
-  __shared__ __bf16 lds_a[2][BM * LDW];
-  __shared__ __bf16 lds_b[2][BN * LDW];
+  __shared__ T16 lds_a[2][BM * LDW];
+  __shared__ T16 lds_b[2][BN * LDW];
   const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_k = (k2 + BN - 1) / BN;
   const int tile_id = bid / slices;
@@ -290,7 +310,7 @@ __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
   const int c_end = min(m, c_beg + per_slice);
   if (c_beg >= c_end) return;
 
-  BWaveTile<FM, FN> wt;
+  BWaveTile<FM, FN, T16> wt;
   wt.zero();
   // fused bias grad: sum the STAGED bf16 A (=g) tile columns.  NOTE: this
   // sums bf16-rounded g — for exact-f32 db the caller uses the f32 path.
@@ -298,12 +318,12 @@ __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
   float dbsum = 0.f;
   const int bcol = threadIdx.x;
   int buf = 0;
-  bstage_cmaj<BM>(a, n, c_beg, n0, c_end, n, lds_a[0]);
-  bstage_cmaj<BN>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
+  bstage_cmaj<BM, T16>(a, n, c_beg, n0, c_end, n, lds_a[0]);
+  bstage_cmaj<BN, T16>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
   __syncthreads();
   for (int cc = c_beg + BGEMM_BK; cc < c_end; cc += BGEMM_BK) {
-    bstage_cmaj<BM>(a, n, cc, n0, c_end, n, lds_a[buf ^ 1]);
-    bstage_cmaj<BN>(b, k2, cc, k0, c_end, k2, lds_b[buf ^ 1]);
+    bstage_cmaj<BM, T16>(a, n, cc, n0, c_end, n, lds_a[buf ^ 1]);
+    bstage_cmaj<BN, T16>(b, k2, cc, k0, c_end, k2, lds_b[buf ^ 1]);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
     if (do_bias && bcol < BM)
 #pragma unroll
@@ -391,6 +411,68 @@ void launch_gemm_bf16_tn(const float* a, const float* b, float* c,
                                                               slices);
   else
     gemm_bf16_tn_kernel<64, 64>
+        <<<dim3(tiles * slices), dim3(BGEMM_THREADS), 0, s>>>(a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+}
+
+
+// ---------------------------------------------------------------------------
+// fp16 launchers (BASELINE config 5 — mfma_f32_16x16x32_f16, fp32 accumulate)
+// ---------------------------------------------------------------------------
+
+void launch_gemm_fp16_nt(const float* a, const float* b, const float* bias,
+                         float* c, int m, int n, int k, bool relu,
+                         hipStream_t s) {
+  if (m >= 512 && n >= 128) {
+    const int grid = ((m + 127) / 128) * ((n + 127) / 128);
+    gemm_bf16_nt_kernel<128, 128, _Float16>
+        <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(a, b, bias, c, m, n, k,
+                                                    relu ? 1 : 0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((n + 63) / 64);
+    gemm_bf16_nt_kernel<64, 64, _Float16>
+        <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(a, b, bias, c, m, n, k,
+                                                    relu ? 1 : 0);
+  }
+}
+
+void launch_gemm_fp16_nn(const float* a, const float* b, const float* bias,
+                         float* c, int m, int n, int k2, bool relu,
+                         hipStream_t s) {
+  if (m >= 512 && k2 >= 128) {
+    const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
+    gemm_bf16_nn_kernel<128, 128, _Float16>
+        <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(a, b, bias, c, m, n, k2,
+                                                    relu ? 1 : 0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((k2 + 63) / 64);
+    gemm_bf16_nn_kernel<64, 64, _Float16>
+        <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(a, b, bias, c, m, n, k2,
+                                                    relu ? 1 : 0);
+  }
+}
+
+void launch_gemm_fp16_tn(const float* a, const float* b, float* c,
+                         float* dbias, int m, int n, int k2, hipStream_t s) {
+  const bool big = (n >= 128 && k2 >= 128);
+  const int bm = big ? 128 : 64;
+  const int bn = big ? 128 : 64;
+  const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
+  int slices = 1;
+  while (tiles * slices < 512 && slices < 64 &&
+         (long)slices * BGEMM_BK * 4 < m)
+    slices *= 2;
+  if (slices > 1)
+    HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
+  if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
+  if (big)
+    gemm_bf16_tn_kernel<128, 128, _Float16>
+        <<<dim3(tiles * slices), dim3(BGEMM_THREADS), 0, s>>>(a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+  else
+    gemm_bf16_tn_kernel<64, 64, _Float16>
         <<<dim3(tiles * slices), dim3(BGEMM_THREADS), 0, s>>>(a, b, c, dbias,
                                                               m, n, k2,
                                                               slices);

@@ -52,7 +52,7 @@ def build_parser():
                         help="generate a synthetic dataset in-place if processed/ is missing")
     parser.add_argument("--sync_bn", action="store_true",
                         help="exact-parity BatchNorm under DDP (statistics all-reduced)")
-    parser.add_argument("--precision", choices=["fp32", "bf16"], default="fp32",
+    parser.add_argument("--precision", choices=["fp32", "bf16", "fp16"], default="fp32",
                         help="matmul compute precision on GPU")
     return parser
 

@@ -388,11 +388,11 @@ _GEMM_PRECISION = "fp32"
 
 def set_gemm_precision(prec: str):
     """Select the matmul compute precision on the HIP path: "fp32" (exact,
-    v_mfma_f32_16x16x4_f32) or "bf16" (operands rounded to bf16 in the matrix
-    cores, fp32 accumulate — BASELINE configs 2/5 mixed-precision mode).
-    Everything outside the matmuls stays fp32 either way."""
+    v_mfma_f32_16x16x4_f32), "bf16" or "fp16" (operands rounded to 16-bit in
+    the matrix cores, fp32 accumulate — BASELINE configs 2/5 mixed-precision
+    modes).  Everything outside the matmuls stays fp32 either way."""
     global _GEMM_PRECISION
-    assert prec in ("fp32", "bf16"), prec
+    assert prec in ("fp32", "bf16", "fp16"), prec
     _GEMM_PRECISION = prec
 
 
@@ -404,12 +404,17 @@ class _LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b):
         m = ext()
-        bf16 = _GEMM_PRECISION == "bf16"
+        prec = _GEMM_PRECISION
         bb = b if b is not None else torch.empty(0, dtype=x.dtype, device=x.device)
-        y = m.linear_fwd_bf16(x, w, bb) if bf16 else m.linear_fwd(x, w, bb)
+        if prec == "bf16":
+            y = m.linear_fwd_bf16(x, w, bb)
+        elif prec == "fp16":
+            y = m.linear_fwd_fp16(x, w, bb)
+        else:
+            y = m.linear_fwd(x, w, bb)
         ctx.save_for_backward(x, w)
         ctx.has_bias = b is not None
-        ctx.bf16 = bf16
+        ctx.prec = prec
         return y
 
     @staticmethod
@@ -417,8 +422,10 @@ class _LinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         m = ext()
         g = g.contiguous()
-        if ctx.bf16:
+        if ctx.prec == "bf16":
             dx, dw, db = m.linear_bwd_bf16(g, x, w, ctx.has_bias)
+        elif ctx.prec == "fp16":
+            dx, dw, db = m.linear_bwd_fp16(g, x, w, ctx.has_bias)
         else:
             dx, dw, db = m.linear_bwd(g, x, w, ctx.has_bias)
         return dx, dw, (db if ctx.has_bias else None)

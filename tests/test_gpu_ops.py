@@ -381,3 +381,26 @@ def test_model_bf16_close_to_fp32():
         set_gemm_precision("fp32")
     rel = (out16 - out32).abs().max() / out32.abs().max().clamp_min(1e-6)
     assert rel < 0.05, rel
+
+
+def test_gemm_fp16_parity():
+    """fp16-compute GEMM vs torch matmul on fp16-rounded operands."""
+    require_ext()
+    import pertgnn._C as C
+    g = torch.Generator().manual_seed(31)
+    m, n, k = 1500, 512, 256
+    x = torch.randn(m, k, generator=g).to(DEV)
+    w = torch.randn(n, k, generator=g).to(DEV) * 0.05
+    b = torch.randn(n, generator=g).to(DEV)
+    y = C.linear_fwd_fp16(x, w, b)
+    ref_ = x.half().float() @ w.half().float().t() + b
+    torch.cuda.synchronize()
+    assert torch.allclose(y, ref_, atol=1e-2, rtol=1e-2), (y - ref_).abs().max()
+
+    gout = torch.randn(m, n, generator=g).to(DEV)
+    dx, dw, db = C.linear_bwd_fp16(gout, x, w, True)
+    torch.cuda.synchronize()
+    gr = gout.half().float()
+    assert torch.allclose(dx, gr @ w.half().float(), atol=1e-2, rtol=1e-2)
+    assert torch.allclose(dw, gr.t() @ x.half().float(), atol=0.5, rtol=1e-2)
+    assert torch.allclose(db, gr.sum(0), atol=0.5, rtol=1e-2)
