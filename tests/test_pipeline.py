@@ -170,3 +170,37 @@ def test_native_collator_matches_python(synthetic_workspace):
     assert torch.equal(py.csc_eid, nat.csc_eid)
     assert torch.equal(nat.batch_ptr[1:].long(), torch.cumsum(
         torch.bincount(py.batch, minlength=py.num_graphs), 0).int().long())
+
+
+def test_checkpoint_resume_cli(synthetic_workspace, tmp_path):
+    """save_checkpoint/load_checkpoint round-trip with FusedAdam state."""
+    from pertgnn.train import load_checkpoint, save_checkpoint
+    from pertgnn.train.optim import FusedAdam
+
+    root, (tr2data, entry2runtimes, _, runtime2pert, resource_df) = synthetic_workspace
+    data_list = build_data_list(tr2data, entry2runtimes, runtime2pert, resource_df, limit=30)
+    model = SAGEDeterministic(9, [64], 8, 50, 8, hidden_channels=16, num_layers=1, dropout=0.0)
+    opt = FusedAdam(model.parameters(), lr=1e-3)
+    loader = BatchLoader(data_list, batch_size=8, shuffle=False)
+    model.train()
+    for b in loader:
+        opt.zero_grad()
+        gp, _ = model(b.x, b.cat_X, b.edge_index, b.edge_attr,
+                      b.pattern_num_nodes, b.rt_probs, b.entry_id, b.batch,
+                      csr=b.csr, num_graphs=b.num_graphs)
+        F.quantile_loss(b.y, gp.flatten(), 0.5).backward()
+        opt.step()
+        break
+    p = tmp_path / "ck.pt"
+    save_checkpoint(str(p), model, opt, epoch=3, extra={"note": "x"})
+
+    model2 = SAGEDeterministic(9, [64], 8, 50, 8, hidden_channels=16, num_layers=1, dropout=0.0)
+    opt2 = FusedAdam(model2.parameters(), lr=1e-3)
+    epoch, extra = load_checkpoint(str(p), model2, opt2)
+    assert epoch == 3 and extra["note"] == "x"
+    assert opt2.step_count == opt.step_count
+    for (n1, p1), (n2, p2) in zip(model.named_parameters(), model2.named_parameters()):
+        if isinstance(p1, torch.nn.parameter.UninitializedParameter):
+            continue
+        assert torch.equal(p1.detach(), p2.detach()), n1
+    assert torch.equal(opt.exp_avg, opt2.exp_avg)
