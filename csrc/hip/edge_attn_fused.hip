@@ -10,9 +10,15 @@
 //     layer) — the kernel gathers from these L2-resident tables instead of
 //     streaming an [E,H] operand from HBM (halves forward edge traffic).
 //
+// Column mapping: when H is a multiple of 256 each lane owns VPT CONTIGUOUS
+// columns (lane*VPT..) so every per-edge row access is f32x4 loads (guide
+// G13); otherwise the strided lane+j*64 mapping with per-column guards.
+// The mapping is an internal detail — reductions are wave-wide and every
+// tensor is read/written with the same mapping.
+//
 // Backward: row kernel writes dq + dskip segments of dqkvs and per-edge
 // dek/dev scratch; col kernel (CSC) segment-sums dk/dv into dqkvs; de =
-// dek + dev feeds the grouped per-vocab scatter (python side) to produce
+// dek + dev feeds the per-vocab scatter (python side) to produce
 // dP_ifc/dP_rpc.  Deterministic — no atomics anywhere.
 
 #include "common.h"
@@ -20,7 +26,63 @@
 
 #define WAVES_PER_BLOCK 4
 
-template <int VPT>
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+// row-slice accessors for the two column mappings
+template <int VPT, bool VEC>
+struct Slice {
+  static __device__ __forceinline__ void load(const float* base, int lane,
+                                              int h, float (&dst)[VPT]) {
+    if constexpr (VEC) {
+#pragma unroll
+      for (int q = 0; q < VPT; q += 4)
+        *reinterpret_cast<f32x4_t*>(&dst[q]) =
+            *reinterpret_cast<const f32x4_t*>(&base[lane * VPT + q]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPT; ++j) {
+        const int c = lane + j * PERTGNN_WAVE;
+        dst[j] = (c < h) ? base[c] : 0.f;
+      }
+    }
+  }
+  static __device__ __forceinline__ void load_add(const float* a,
+                                                  const float* b, int lane,
+                                                  int h, float (&dst)[VPT]) {
+    if constexpr (VEC) {
+#pragma unroll
+      for (int q = 0; q < VPT; q += 4) {
+        const f32x4_t va = *reinterpret_cast<const f32x4_t*>(&a[lane * VPT + q]);
+        const f32x4_t vb = *reinterpret_cast<const f32x4_t*>(&b[lane * VPT + q]);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) dst[q + u] = va[u] + vb[u];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPT; ++j) {
+        const int c = lane + j * PERTGNN_WAVE;
+        dst[j] = (c < h) ? a[c] + b[c] : 0.f;
+      }
+    }
+  }
+  static __device__ __forceinline__ void store(float* base, int lane, int h,
+                                               const float (&src)[VPT]) {
+    if constexpr (VEC) {
+#pragma unroll
+      for (int q = 0; q < VPT; q += 4)
+        *reinterpret_cast<f32x4_t*>(&base[lane * VPT + q]) =
+            *reinterpret_cast<const f32x4_t*>(&src[q]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPT; ++j) {
+        const int c = lane + j * PERTGNN_WAVE;
+        if (c < h) base[c] = src[j];
+      }
+    }
+  }
+};
+
+template <int VPT, bool VEC>
 __global__ void edge_attn_fused_fwd_kernel(
     const float* __restrict__ qkvs,  // [N, 4h]
     const float* __restrict__ pifc,  // [Vi, h]
@@ -29,6 +91,7 @@ __global__ void edge_attn_fused_fwd_kernel(
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
     float* __restrict__ out, float* __restrict__ alpha, int n, int h,
     float scale) {
+  using S = Slice<VPT, VEC>;
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
   const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
@@ -36,12 +99,9 @@ __global__ void edge_attn_fused_fwd_kernel(
   const long ld = 4L * h;
 
   float qr[VPT], acc[VPT];
+  S::load(&qkvs[row * ld], lane, h, qr);
 #pragma unroll
-  for (int j = 0; j < VPT; ++j) {
-    const int c = lane + j * PERTGNN_WAVE;
-    qr[j] = (c < h) ? qkvs[row * ld + c] : 0.f;
-    acc[j] = 0.f;
-  }
+  for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
 
   const int beg = row_ptr[row], end = row_ptr[row + 1];
   float m = -INFINITY, s = 0.f;
@@ -49,45 +109,37 @@ __global__ void edge_attn_fused_fwd_kernel(
     const long src = csr_src[p];
     const long a0 = ea[(long)p * astride];
     const long a1 = ea[(long)p * astride + 1];
+    float ec[VPT], ke[VPT];
+    S::load_add(&pifc[a0 * h], &prpc[a1 * h], lane, h, ec);
+    S::load(&qkvs[src * ld + h], lane, h, ke);
     float part = 0.f;
-    float ec[VPT];
 #pragma unroll
-    for (int j = 0; j < VPT; ++j) {
-      const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) {
-        ec[j] = pifc[a0 * h + c] + prpc[a1 * h + c];
-        part += qr[j] * (qkvs[src * ld + h + c] + ec[j]);
-      }
-    }
+    for (int j = 0; j < VPT; ++j) part += qr[j] * (ke[j] + ec[j]);
     const float logit = wave_reduce_sum(part) * scale;
     if (lane == (p - beg) % PERTGNN_WAVE) alpha[p] = logit;
     const float m_new = fmaxf(m, logit);
     const float corr = __expf(m - m_new);
     const float pexp = __expf(logit - m_new);
     s = s * corr + pexp;
+    float ve[VPT];
+    S::load(&qkvs[src * ld + 2 * h], lane, h, ve);
 #pragma unroll
-    for (int j = 0; j < VPT; ++j) {
-      const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) {
-        const float ve = qkvs[src * ld + 2 * h + c] + ec[j];
-        acc[j] = acc[j] * corr + pexp * ve;
-      }
-    }
+    for (int j = 0; j < VPT; ++j)
+      acc[j] = acc[j] * corr + pexp * (ve[j] + ec[j]);
     m = m_new;
   }
 
   const float inv_s = (s > 0.f) ? 1.f / s : 0.f;
+  float sk[VPT], res[VPT];
+  S::load(&qkvs[row * ld + 3 * h], lane, h, sk);
 #pragma unroll
-  for (int j = 0; j < VPT; ++j) {
-    const int c = lane + j * PERTGNN_WAVE;
-    if (c < h)
-      out[(long)row * h + c] = acc[j] * inv_s + qkvs[row * ld + 3 * h + c];
-  }
+  for (int j = 0; j < VPT; ++j) res[j] = acc[j] * inv_s + sk[j];
+  S::store(&out[(long)row * h], lane, h, res);
   for (int p = beg + lane; p < end; p += PERTGNN_WAVE)
     alpha[p] = __expf(alpha[p] - m) * inv_s;
 }
 
-template <int VPT>
+template <int VPT, bool VEC>
 __global__ void edge_attn_fused_bwd_row_kernel(
     const float* __restrict__ g, const float* __restrict__ qkvs,
     const float* __restrict__ pifc, const float* __restrict__ prpc,
@@ -95,6 +147,7 @@ __global__ void edge_attn_fused_bwd_row_kernel(
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
     float* __restrict__ dqkvs, float* __restrict__ dek,
     float* __restrict__ dev, int n, int h, float scale) {
+  using S = Slice<VPT, VEC>;
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
   const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
@@ -102,13 +155,10 @@ __global__ void edge_attn_fused_bwd_row_kernel(
   const long ld = 4L * h;
 
   float gr[VPT], qr[VPT], dqacc[VPT];
+  S::load(&g[(long)row * h], lane, h, gr);
+  S::load(&qkvs[row * ld], lane, h, qr);
 #pragma unroll
-  for (int j = 0; j < VPT; ++j) {
-    const int c = lane + j * PERTGNN_WAVE;
-    gr[j] = (c < h) ? g[(long)row * h + c] : 0.f;
-    qr[j] = (c < h) ? qkvs[row * ld + c] : 0.f;
-    dqacc[j] = 0.f;
-  }
+  for (int j = 0; j < VPT; ++j) dqacc[j] = 0.f;
 
   const int beg = row_ptr[row], end = row_ptr[row + 1];
   float sdot = 0.f;
@@ -116,14 +166,12 @@ __global__ void edge_attn_fused_bwd_row_kernel(
     const long src = csr_src[p];
     const long a0 = ea[(long)p * astride];
     const long a1 = ea[(long)p * astride + 1];
+    float ec[VPT], ve[VPT];
+    S::load_add(&pifc[a0 * h], &prpc[a1 * h], lane, h, ec);
+    S::load(&qkvs[src * ld + 2 * h], lane, h, ve);
     float part = 0.f;
 #pragma unroll
-    for (int j = 0; j < VPT; ++j) {
-      const int c = lane + j * PERTGNN_WAVE;
-      if (c < h)
-        part += gr[j] * (qkvs[src * ld + 2 * h + c] + pifc[a0 * h + c] +
-                         prpc[a1 * h + c]);
-    }
+    for (int j = 0; j < VPT; ++j) part += gr[j] * (ve[j] + ec[j]);
     const float dalpha = wave_reduce_sum(part);
     sdot += alpha[p] * dalpha;
     if (lane == 0) dev[(long)p * h] = dalpha;
@@ -136,33 +184,28 @@ __global__ void edge_attn_fused_bwd_row_kernel(
     const float dalpha = __shfl(dalpha_l0, 0, PERTGNN_WAVE);
     const float a = alpha[p];
     const float dl = a * (dalpha - sdot) * scale;
+    float ec[VPT], ke[VPT], dekv[VPT], devv[VPT];
+    S::load_add(&pifc[a0 * h], &prpc[a1 * h], lane, h, ec);
+    S::load(&qkvs[src * ld + h], lane, h, ke);
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
-      const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) {
-        const float kec =
-            qkvs[src * ld + h + c] + pifc[a0 * h + c] + prpc[a1 * h + c];
-        dqacc[j] += dl * kec;
-        dek[(long)p * h + c] = dl * qr[j];
-        dev[(long)p * h + c] = a * gr[j];
-      }
+      dqacc[j] += dl * (ke[j] + ec[j]);
+      dekv[j] = dl * qr[j];
+      devv[j] = a * gr[j];
     }
+    S::store(&dek[(long)p * h], lane, h, dekv);
+    S::store(&dev[(long)p * h], lane, h, devv);
   }
-#pragma unroll
-  for (int j = 0; j < VPT; ++j) {
-    const int c = lane + j * PERTGNN_WAVE;
-    if (c < h) {
-      dqkvs[row * ld + c] = dqacc[j];         // dq
-      dqkvs[row * ld + 3 * h + c] = gr[j];    // dskip = g
-    }
-  }
+  S::store(&dqkvs[row * ld], lane, h, dqacc);          // dq
+  S::store(&dqkvs[row * ld + 3 * h], lane, h, gr);     // dskip = g
 }
 
-template <int VPT>
+template <int VPT, bool VEC>
 __global__ void edge_attn_fused_bwd_col_kernel(
     const float* __restrict__ dek, const float* __restrict__ dev,
     const int* __restrict__ col_ptr, const int* __restrict__ csc_eid,
     float* __restrict__ dqkvs, int n, int h) {
+  using S = Slice<VPT, VEC>;
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
   const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
@@ -173,23 +216,17 @@ __global__ void edge_attn_fused_bwd_col_kernel(
   for (int j = 0; j < VPT; ++j) { ka[j] = 0.f; va[j] = 0.f; }
   for (int p = col_ptr[row]; p < col_ptr[row + 1]; ++p) {
     const long eid = csc_eid[p];
+    float dk1[VPT], dv1[VPT];
+    S::load(&dek[eid * h], lane, h, dk1);
+    S::load(&dev[eid * h], lane, h, dv1);
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
-      const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) {
-        ka[j] += dek[eid * h + c];
-        va[j] += dev[eid * h + c];
-      }
+      ka[j] += dk1[j];
+      va[j] += dv1[j];
     }
   }
-#pragma unroll
-  for (int j = 0; j < VPT; ++j) {
-    const int c = lane + j * PERTGNN_WAVE;
-    if (c < h) {
-      dqkvs[(long)row * ld + h + c] = ka[j];
-      dqkvs[(long)row * ld + 2 * h + c] = va[j];
-    }
-  }
+  S::store(&dqkvs[(long)row * ld + h], lane, h, ka);
+  S::store(&dqkvs[(long)row * ld + 2 * h], lane, h, va);
 }
 
 __global__ void add2_kernel(const float* __restrict__ a,
@@ -212,12 +249,18 @@ void launch_edge_attn_fused_fwd(const float* qkvs, const float* pifc,
   const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
   const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  const bool vec = (h % 256 == 0);
   switch (vpt) {
 #define CASE(V)                                                                \
   case V:                                                                      \
-    edge_attn_fused_fwd_kernel<V><<<grid, block, 0, stream>>>(                 \
-        qkvs, pifc, prpc, ea, astride, row_ptr, csr_src, out, alpha, n, h,     \
-        scale);                                                                \
+    if (vec && (V % 4 == 0))                                                   \
+      edge_attn_fused_fwd_kernel<V, true><<<grid, block, 0, stream>>>(         \
+          qkvs, pifc, prpc, ea, astride, row_ptr, csr_src, out, alpha, n, h,   \
+          scale);                                                              \
+    else                                                                       \
+      edge_attn_fused_fwd_kernel<V, false><<<grid, block, 0, stream>>>(        \
+          qkvs, pifc, prpc, ea, astride, row_ptr, csr_src, out, alpha, n, h,   \
+          scale);                                                              \
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
@@ -238,14 +281,23 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
   const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
   const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  const bool vec = (h % 256 == 0);
   switch (vpt) {
 #define CASE(V)                                                                \
   case V:                                                                      \
-    edge_attn_fused_bwd_row_kernel<V><<<grid, block, 0, stream>>>(             \
-        g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,      \
-        dek, dev, n, h, scale);                                                \
-    edge_attn_fused_bwd_col_kernel<V><<<grid, block, 0, stream>>>(             \
-        dek, dev, col_ptr, csc_eid, dqkvs, n, h);                              \
+    if (vec && (V % 4 == 0)) {                                                 \
+      edge_attn_fused_bwd_row_kernel<V, true><<<grid, block, 0, stream>>>(     \
+          g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,    \
+          dek, dev, n, h, scale);                                              \
+      edge_attn_fused_bwd_col_kernel<V, true><<<grid, block, 0, stream>>>(     \
+          dek, dev, col_ptr, csc_eid, dqkvs, n, h);                            \
+    } else {                                                                   \
+      edge_attn_fused_bwd_row_kernel<V, false><<<grid, block, 0, stream>>>(    \
+          g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,    \
+          dek, dev, n, h, scale);                                              \
+      edge_attn_fused_bwd_col_kernel<V, false><<<grid, block, 0, stream>>>(    \
+          dek, dev, col_ptr, csc_eid, dqkvs, n, h);                            \
+    }                                                                          \
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE

@@ -14,6 +14,23 @@ import torch
 from ..ops import functional as F
 
 
+class _nvtx:
+    """roctx phase ranges (torch.cuda.nvtx maps to roctx on ROCm) — makes
+    collate/H2D/fwd/bwd/allreduce/step phases visible in rocprofv3 traces."""
+
+    def __init__(self, name):
+        self.name = name
+        self.on = torch.cuda.is_available()
+
+    def __enter__(self):
+        if self.on:
+            torch.cuda.nvtx.range_push(self.name)
+
+    def __exit__(self, *a):
+        if self.on:
+            torch.cuda.nvtx.range_pop()
+
+
 def _forward(model, b):
     return model(
         b.x, b.cat_X, b.edge_index, b.edge_attr,
@@ -29,17 +46,22 @@ def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
     mape_sum = 0.0
     n_graphs = 0
     for batch in loader:
-        b = batch.to(device, non_blocking=non_blocking) if device is not None else batch
+        with _nvtx("h2d"):
+            b = batch.to(device, non_blocking=non_blocking) if device is not None else batch
         optimizer.zero_grad(set_to_none=False)
         if engine is not None:
             engine.reset()
-        global_pred, _local_pred = _forward(model, b)
-        pred = global_pred.flatten()
-        loss = F.quantile_loss(b.y, pred, tau)
-        loss.backward()
-        if engine is not None:
-            engine.finalize()
-        optimizer.step()
+        with _nvtx("forward"):
+            global_pred, _local_pred = _forward(model, b)
+            pred = global_pred.flatten()
+            loss = F.quantile_loss(b.y, pred, tau)
+        with _nvtx("backward"):
+            loss.backward()
+        with _nvtx("allreduce"):
+            if engine is not None:
+                engine.finalize()
+        with _nvtx("optimizer"):
+            optimizer.step()
         with torch.no_grad():
             total_loss += float(loss.detach()) * b.num_graphs
             mape_sum += float(((pred.detach() - b.y).abs() / b.y).sum())
