@@ -650,7 +650,54 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor g, torch::Tensor x,
   return {dx, dgamma, dbeta};
 }
 
+// split backward entry points so python can overlap dgrad (current stream)
+// with wgrad+bias-grad (side stream): prec 0=f32, 1=bf16, 2=fp16
+torch::Tensor linear_dgrad(torch::Tensor g, torch::Tensor w, int64_t prec) {
+  CHECK_IN(g); CHECK_IN(w);
+  const int m = g.size(0);
+  const int n = w.size(0);
+  const int k = w.size(1);
+  auto dx = torch::empty({m, k}, g.options());
+  if (prec == 1)
+    launch_gemm_bf16_nn(g.data_ptr<float>(), w.data_ptr<float>(), nullptr,
+                        dx.data_ptr<float>(), m, n, k, false, cur_stream());
+  else if (prec == 2)
+    launch_gemm_fp16_nn(g.data_ptr<float>(), w.data_ptr<float>(), nullptr,
+                        dx.data_ptr<float>(), m, n, k, false, cur_stream());
+  else
+    launch_gemm_f32_nn(g.data_ptr<float>(), w.data_ptr<float>(), nullptr,
+                       dx.data_ptr<float>(), m, n, k, false, cur_stream());
+  return dx;
+}
+
+std::vector<torch::Tensor> linear_wgrad(torch::Tensor g, torch::Tensor x,
+                                        bool has_bias, int64_t prec) {
+  CHECK_IN(g); CHECK_IN(x);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = g.size(1);
+  auto dw = torch::empty({n, k}, x.options());
+  torch::Tensor db = torch::empty({0}, g.options());
+  float* db_ptr = nullptr;
+  if (has_bias) {
+    db = torch::empty({n}, g.options());
+    db_ptr = db.data_ptr<float>();
+  }
+  if (prec == 1)
+    launch_gemm_bf16_tn(g.data_ptr<float>(), x.data_ptr<float>(),
+                        dw.data_ptr<float>(), db_ptr, m, n, k, cur_stream());
+  else if (prec == 2)
+    launch_gemm_fp16_tn(g.data_ptr<float>(), x.data_ptr<float>(),
+                        dw.data_ptr<float>(), db_ptr, m, n, k, cur_stream());
+  else
+    launch_gemm_f32_tn(g.data_ptr<float>(), x.data_ptr<float>(),
+                       dw.data_ptr<float>(), db_ptr, m, n, k, cur_stream());
+  return {dw, db};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("linear_dgrad", &linear_dgrad);
+  mod.def("linear_wgrad", &linear_wgrad);
   mod.def("bn_stats", &bn_stats);
   mod.def("bn_finalize_apply", &bn_finalize_apply);
   mod.def("bn_bwd_partials", &bn_bwd_partials);

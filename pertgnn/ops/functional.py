@@ -92,9 +92,21 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
             g.contiguous(), qkvs, pifc, prpc, edge_attr, alpha,
             row_ptr, csr_src, col_ptr, csc_eid,
         )
-        # dP tables: per-vocab segment sums of de
-        dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
-        dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
+        # dP tables (per-vocab segment sums of de) are independent of the
+        # CSC dk/dv pass queued above — overlap them on the side stream.
+        cur = torch.cuda.current_stream()
+        side = _side_stream()
+        ev = torch.cuda.Event()
+        ev.record(cur)
+        side.wait_event(ev)
+        with torch.cuda.stream(side):
+            dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
+            dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
+        ev2 = torch.cuda.Event()
+        ev2.record(side)
+        cur.wait_event(ev2)
+        _mark_cross_stream(dpifc, cur)
+        _mark_cross_stream(dprpc, cur)
         return dqkvs, dpifc, dprpc, None, None, None, None, None
 
 
@@ -384,6 +396,24 @@ def eval_metrics(y, y_hat, tau):
 # ---------------------------------------------------------------------------
 
 _GEMM_PRECISION = "fp32"
+_SIDE_STREAM = None
+
+
+def _side_stream():
+    """Side stream for independent backward branches (wgrad overlaps dgrad);
+    cross-stream edges are captured into hipGraphs as graph dependencies."""
+    global _SIDE_STREAM
+    if _SIDE_STREAM is None:
+        _SIDE_STREAM = torch.cuda.Stream()
+    return _SIDE_STREAM
+
+
+def _mark_cross_stream(t, consumer_stream):
+    """Tensors allocated on the side stream are consumed on the main stream:
+    tell the caching allocator (no-op during graph capture, where the private
+    pool owns the memory for the whole graph)."""
+    if not torch.cuda.is_current_stream_capturing():
+        t.record_stream(consumer_stream)
 
 
 def set_gemm_precision(prec: str):
@@ -422,12 +452,21 @@ class _LinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         m = ext()
         g = g.contiguous()
-        if ctx.prec == "bf16":
-            dx, dw, db = m.linear_bwd_bf16(g, x, w, ctx.has_bias)
-        elif ctx.prec == "fp16":
-            dx, dw, db = m.linear_bwd_fp16(g, x, w, ctx.has_bias)
-        else:
-            dx, dw, db = m.linear_bwd(g, x, w, ctx.has_bias)
+        prec = {"fp32": 0, "bf16": 1, "fp16": 2}[ctx.prec]
+        cur = torch.cuda.current_stream()
+        side = _side_stream()
+        ev = torch.cuda.Event()
+        ev.record(cur)
+        side.wait_event(ev)
+        with torch.cuda.stream(side):
+            dw, db = m.linear_wgrad(g, x, ctx.has_bias, prec)
+        dx = m.linear_dgrad(g, w, prec)  # current stream, overlapped with wgrad
+        ev2 = torch.cuda.Event()
+        ev2.record(side)
+        cur.wait_event(ev2)
+        _mark_cross_stream(dw, cur)
+        if ctx.has_bias:
+            _mark_cross_stream(db, cur)
         return dx, dw, (db if ctx.has_bias else None)
 
 
