@@ -94,6 +94,10 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
         )
         # dP tables (per-vocab segment sums of de) are independent of the
         # CSC dk/dv pass queued above — overlap them on the side stream.
+        if not _overlap_enabled():
+            dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
+            dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
+            return dqkvs, dpifc, dprpc, None, None, None, None, None
         cur = torch.cuda.current_stream()
         side = _side_stream()
         ev = torch.cuda.Event()
@@ -397,6 +401,15 @@ def eval_metrics(y, y_hat, tau):
 
 _GEMM_PRECISION = "fp32"
 _SIDE_STREAM = None
+_OVERLAP = None
+
+
+def _overlap_enabled() -> bool:
+    global _OVERLAP
+    if _OVERLAP is None:
+        import os
+        _OVERLAP = os.environ.get("PERTGNN_NO_SIDE_STREAM", "0") != "1"
+    return _OVERLAP
 
 
 def _side_stream():
@@ -453,6 +466,10 @@ class _LinearFn(torch.autograd.Function):
         m = ext()
         g = g.contiguous()
         prec = {"fp32": 0, "bf16": 1, "fp16": 2}[ctx.prec]
+        if not _overlap_enabled():
+            dw, db = m.linear_wgrad(g, x, ctx.has_bias, prec)
+            dx = m.linear_dgrad(g, w, prec)
+            return dx, dw, (db if ctx.has_bias else None)
         cur = torch.cuda.current_stream()
         side = _side_stream()
         ev = torch.cuda.Event()
