@@ -703,7 +703,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_bwd_partials", &bn_bwd_partials);
   mod.def("bn_bwd_apply", &bn_bwd_apply);
   mod.def("vocab_scatter", &vocab_scatter);
-  mod.def("collate_native", &collate_native);
+  mod.def("collate_native", &collate_native,
+          py::call_guard<py::gil_scoped_release>());
   mod.def("edge_attn_fused_fwd", &edge_attn_fused_fwd);
   mod.def("edge_attn_fused_bwd", &edge_attn_fused_bwd);
   mod.def("embed_grouped_scatter", &embed_grouped_scatter);

@@ -126,9 +126,11 @@ def main(argv=None):
     valid_loader = BatchLoader(valid_shard, args.batch_size, shuffle=False, collate_fn=collate_fn)
     test_loader = BatchLoader(test_shard, args.batch_size, shuffle=False, collate_fn=collate_fn)
     if on_gpu:
-        train_loader = PrefetchLoader(train_loader, comm.device)
-        valid_loader = PrefetchLoader(valid_loader, comm.device)
-        test_loader = PrefetchLoader(test_loader, comm.device)
+        from pertgnn.data.prefetch import ThreadedLoader
+
+        train_loader = PrefetchLoader(ThreadedLoader(train_loader), comm.device)
+        valid_loader = PrefetchLoader(ThreadedLoader(valid_loader), comm.device)
+        test_loader = PrefetchLoader(ThreadedLoader(test_loader), comm.device)
 
     # vocab scans (pert_gnn.py:306-328)
     unique_ms_max = max(int(g["ms_id"].max()) for g in runtime2graph.values())

@@ -49,3 +49,48 @@ class PrefetchLoader:
                     t.record_stream(torch.cuda.current_stream())
             pending = self._start_copy(nxt_host) if nxt_host is not None else None
             yield cur
+
+
+class ThreadedLoader:
+    """Runs the underlying loader (collation) in a worker thread — the native
+    collator releases the GIL, so batch i+1 is collated while batch i trains.
+    Bounded queue (depth 2) keeps at most two host batches in flight."""
+
+    def __init__(self, loader, depth: int = 2):
+        self.loader = loader
+        self.depth = depth
+
+    def __len__(self):
+        return len(self.loader)
+
+    @property
+    def dataset(self):
+        return self.loader.dataset
+
+    def __iter__(self):
+        import queue
+        import threading
+
+        q: "queue.Queue" = queue.Queue(maxsize=self.depth)
+        SENTINEL = object()
+        err = []
+
+        def worker():
+            try:
+                for b in self.loader:
+                    q.put(b)
+            except BaseException as e:  # propagate into the consumer
+                err.append(e)
+            finally:
+                q.put(SENTINEL)
+
+        t = threading.Thread(target=worker, daemon=True)
+        t.start()
+        while True:
+            item = q.get()
+            if item is SENTINEL:
+                break
+            yield item
+        t.join()
+        if err:
+            raise err[0]
