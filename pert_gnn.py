@@ -160,8 +160,10 @@ def main(argv=None):
         log.print0(f"resumed from {args.checkpoint} at epoch {ep}")
 
     for epoch in range(start_epoch, args.epochs + 1):
+        epoch_stats: dict = {}
         train_mae, train_mape = train_epoch(
-            model, train_loader, optimizer, args.tau, device, engine=engine, comm=comm
+            model, train_loader, optimizer, args.tau, device, engine=engine,
+            comm=comm, stats_out=epoch_stats,
         )
         valid_mae, valid_mape, valid_q = evaluate(model, valid_loader, args.tau, device, comm=comm)
         test_mae, test_mape, test_q = evaluate(model, test_loader, args.tau, device, comm=comm)
@@ -172,6 +174,7 @@ def main(argv=None):
         )
         log.log({
             "epoch": epoch, "train_loss": train_mae, "train_mape": train_mape,
+            **epoch_stats,
             "valid_mae": valid_mae, "valid_mape": valid_mape, "valid_q": valid_q,
             "test_mae": test_mae, "test_mape": test_mape, "test_q": test_q,
         })

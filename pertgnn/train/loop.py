@@ -40,11 +40,18 @@ def _forward(model, b):
 
 
 def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
-                non_blocking=True):
+                non_blocking=True, stats_out=None):
+    """Returns (avg_loss, avg_mape); with stats_out (dict) also records
+    graphs/sec, edges/sec and nodes/sec for the epoch (whole job)."""
+    import time as _time
+
     model.train()
     total_loss = 0.0
     mape_sum = 0.0
     n_graphs = 0
+    n_edges = 0
+    n_nodes = 0
+    t0 = _time.perf_counter()
     for batch in loader:
         with _nvtx("h2d"):
             b = batch.to(device, non_blocking=non_blocking) if device is not None else batch
@@ -66,10 +73,25 @@ def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
             total_loss += float(loss.detach()) * b.num_graphs
             mape_sum += float(((pred.detach() - b.y).abs() / b.y).sum())
             n_graphs += b.num_graphs
+            n_edges += b.edge_index.shape[1]
+            n_nodes += b.x.shape[0]
+    if device is not None and torch.cuda.is_available():
+        torch.cuda.synchronize()
+    elapsed = _time.perf_counter() - t0
     if comm is not None and comm.distributed:
         total_loss = comm.all_reduce_scalar(total_loss)
         mape_sum = comm.all_reduce_scalar(mape_sum)
         n_graphs = int(comm.all_reduce_scalar(float(n_graphs)))
+        n_edges = int(comm.all_reduce_scalar(float(n_edges)))
+        n_nodes = int(comm.all_reduce_scalar(float(n_nodes)))
+        elapsed = comm.all_reduce_scalar(elapsed, op="max")
+    if stats_out is not None:
+        stats_out.update({
+            "epoch_s": elapsed,
+            "graphs_per_s": n_graphs / max(elapsed, 1e-9),
+            "edges_per_s": n_edges / max(elapsed, 1e-9),
+            "nodes_per_s": n_nodes / max(elapsed, 1e-9),
+        })
     n = max(n_graphs, 1)
     return total_loss / n, mape_sum / n
 
