@@ -11,8 +11,8 @@
 // makes the 16-lane fragment reads conflict-free (guide §6 G4).
 
 #include "common.h"
+#include <cstdlib>
 
-#define BGEMM_BK 64
 #define BGEMM_PAD 8
 #define BGEMM_THREADS 256
 
@@ -51,19 +51,19 @@ __device__ __forceinline__ typename Vec16<T16>::v4 pack4(float a, float b,
 // global [free][contract] (contract-minor) -> LDS [free][BK+PAD]
 // 4 bf16 packed into one 8-byte LDS write (scalar u16 LDS writes are ~2x
 // slower — guide G13 applies to LDS too).
-template <int BF, typename T16>
+template <int BF, int BK, typename T16>
 __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
                                             long ld, int free0, int contract0,
                                             int free_max, int contract_max,
                                             T16* lds) {
-  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  constexpr int LDW = BK + BGEMM_PAD;
   const int t = threadIdx.x;
-  constexpr int QUADS = BGEMM_BK / 4;  // 16 float4 per row
+  constexpr int QUADS = BK / 4;
   const int f = t / QUADS;             // 16 free rows per pass
   const int cq = (t % QUADS) * 4;
   constexpr int FSTEP = BGEMM_THREADS / QUADS;  // 16
   const bool interior = (free0 + BF <= free_max) &&
-                        (contract0 + BGEMM_BK <= contract_max);
+                        (contract0 + BK <= contract_max);
   const bool aligned = (ld & 3) == 0;
 #pragma unroll
   for (int half = 0; half < BF / FSTEP; ++half) {
@@ -95,22 +95,22 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
 // global [contract][free] (contract-major) -> LDS [free][BK+PAD]: each thread
 // transposes a 4x4 block in registers (4 coalesced f32x4 loads from 4
 // contract rows), then writes 4 packed 8-byte LDS rows.
-template <int BF, typename T16>
+template <int BF, int BK, typename T16>
 __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
                                             long ld, int contract0, int free0,
                                             int contract_max, int free_max,
                                             T16* lds) {
-  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  constexpr int LDW = BK + BGEMM_PAD;
   const int t = threadIdx.x;
   constexpr int FQUADS = BF / 4;
   const int cb = (t / FQUADS) * 4;     // contract block of 4
   const int fq = (t % FQUADS) * 4;     // free quad
   constexpr int CSTEP = (BGEMM_THREADS / FQUADS) * 4;
-  const bool interior = (contract0 + BGEMM_BK <= contract_max) &&
+  const bool interior = (contract0 + BK <= contract_max) &&
                         (free0 + BF <= free_max);
   const bool aligned = ((ld & 3) == 0) && ((free0 & 3) == 0);
 #pragma unroll
-  for (int half = 0; half < BGEMM_BK / CSTEP; ++half) {
+  for (int half = 0; half < (BK > CSTEP ? BK / CSTEP : 1); ++half) {
     const int cc = cb + half * CSTEP;
     float v[4][4];
 #pragma unroll
@@ -146,7 +146,7 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
 // elements at k = (l>>4)*8 of row/col (l&15); C/D: col=l&15, row=(l>>4)*4+r.
 // ---------------------------------------------------------------------------
 
-template <int FM, int FN, typename T16>
+template <int FM, int FN, int BK, typename T16>
 struct BWaveTile {
   f32x4 acc[FM][FN];
   __device__ __forceinline__ void zero() {
@@ -157,11 +157,11 @@ struct BWaveTile {
   }
   __device__ __forceinline__ void mma(const T16* lds_a, const T16* lds_b,
                                       int wm, int wn, int lane) {
-    constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+    constexpr int LDW = BK + BGEMM_PAD;
     const int fi = lane & 15;
     const int fk = (lane >> 4) * 8;
 #pragma unroll
-    for (int s = 0; s < BGEMM_BK / 32; ++s) {
+    for (int s = 0; s < BK / 32; ++s) {
       const int k = s * 32 + fk;
       using v8 = typename Vec16<T16>::v8;
       v8 a[FM], b[FN];
@@ -208,14 +208,14 @@ struct BWaveTile {
 // kernels — same three layouts as the f32 suite
 // ---------------------------------------------------------------------------
 
-template <int BM, int BN, typename T16 = __bf16>
+template <int BM, int BN, int BK = 64, typename T16 = __bf16>
 __launch_bounds__(BGEMM_THREADS)
 __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
                                     const float* __restrict__ b,
                                     const float* __restrict__ bias,
                                     float* __restrict__ c, int m, int n, int k,
                                     int relu) {
-  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  constexpr int LDW = BK + BGEMM_PAD;
   constexpr int FM = BM / 32, FN = BN / 32;
   __shared__ T16 lds_a[2][BM * LDW];
   __shared__ T16 lds_b[2][BN * LDW];
@@ -228,15 +228,15 @@ __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
   const int wm = (wave >> 1) * (BM / 2);
   const int wn = (wave & 1) * (BN / 2);
 
-  BWaveTile<FM, FN, T16> wt;
+  BWaveTile<FM, FN, BK, T16> wt;
   wt.zero();
   int buf = 0;
-  bstage_cmin<BM, T16>(a, k, m0, 0, m, k, lds_a[0]);
-  bstage_cmin<BN, T16>(b, k, n0, 0, n, k, lds_b[0]);
+  bstage_cmin<BM, BK, T16>(a, k, m0, 0, m, k, lds_a[0]);
+  bstage_cmin<BN, BK, T16>(b, k, n0, 0, n, k, lds_b[0]);
   __syncthreads();
-  for (int k0 = BGEMM_BK; k0 < k; k0 += BGEMM_BK) {
-    bstage_cmin<BM, T16>(a, k, m0, k0, m, k, lds_a[buf ^ 1]);
-    bstage_cmin<BN, T16>(b, k, n0, k0, n, k, lds_b[buf ^ 1]);
+  for (int k0 = BK; k0 < k; k0 += BK) {
+    bstage_cmin<BM, BK, T16>(a, k, m0, k0, m, k, lds_a[buf ^ 1]);
+    bstage_cmin<BN, BK, T16>(b, k, n0, k0, n, k, lds_b[buf ^ 1]);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
     __syncthreads();
     buf ^= 1;
@@ -245,14 +245,14 @@ __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
   wt.store(c, n, m0 + wm, n0 + wn, m, n, bias, relu, lane);
 }
 
-template <int BM, int BN, typename T16 = __bf16>
+template <int BM, int BN, int BK = 64, typename T16 = __bf16>
 __launch_bounds__(BGEMM_THREADS)
 __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
                                     const float* __restrict__ b,
                                     const float* __restrict__ bias,
                                     float* __restrict__ c, int m, int n,
                                     int k2, int relu) {
-  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  constexpr int LDW = BK + BGEMM_PAD;
   constexpr int FM = BM / 32, FN = BN / 32;
   __shared__ T16 lds_a[2][BM * LDW];
   __shared__ T16 lds_b[2][BN * LDW];
@@ -265,15 +265,15 @@ __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
   const int wm = (wave >> 1) * (BM / 2);
   const int wn = (wave & 1) * (BN / 2);
 
-  BWaveTile<FM, FN, T16> wt;
+  BWaveTile<FM, FN, BK, T16> wt;
   wt.zero();
   int buf = 0;
-  bstage_cmin<BM, T16>(a, n, m0, 0, m, n, lds_a[0]);
-  bstage_cmaj<BN, T16>(b, k2, 0, n0, n, k2, lds_b[0]);
+  bstage_cmin<BM, BK, T16>(a, n, m0, 0, m, n, lds_a[0]);
+  bstage_cmaj<BN, BK, T16>(b, k2, 0, n0, n, k2, lds_b[0]);
   __syncthreads();
-  for (int c0 = BGEMM_BK; c0 < n; c0 += BGEMM_BK) {
-    bstage_cmin<BM, T16>(a, n, m0, c0, m, n, lds_a[buf ^ 1]);
-    bstage_cmaj<BN, T16>(b, k2, c0, n0, n, k2, lds_b[buf ^ 1]);
+  for (int c0 = BK; c0 < n; c0 += BK) {
+    bstage_cmin<BM, BK, T16>(a, n, m0, c0, m, n, lds_a[buf ^ 1]);
+    bstage_cmaj<BN, BK, T16>(b, k2, c0, n0, n, k2, lds_b[buf ^ 1]);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
     __syncthreads();
     buf ^= 1;
@@ -282,14 +282,14 @@ __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
   wt.store(c, k2, m0 + wm, n0 + wn, m, k2, bias, relu, lane);
 }
 
-template <int BM, int BN, typename T16 = __bf16>
+template <int BM, int BN, int BK = 64, typename T16 = __bf16>
 __launch_bounds__(BGEMM_THREADS)
 __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
                                     const float* __restrict__ b,
                                     float* __restrict__ c,
                                     float* __restrict__ dbias, int m, int n,
                                     int k2, int slices) {
-  constexpr int LDW = BGEMM_BK + BGEMM_PAD;
+  constexpr int LDW = BK + BGEMM_PAD;
   constexpr int FM = BM / 32, FN = BN / 32;
   __shared__ T16 lds_a[2][BM * LDW];
   __shared__ T16 lds_b[2][BN * LDW];
@@ -306,12 +306,12 @@ __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
   const int wn = (wave & 1) * (BN / 2);
 
   const int per_slice =
-      ((m + slices - 1) / slices + BGEMM_BK - 1) / BGEMM_BK * BGEMM_BK;
+      ((m + slices - 1) / slices + BK - 1) / BK * BK;
   const int c_beg = slice * per_slice;
   const int c_end = min(m, c_beg + per_slice);
   if (c_beg >= c_end) return;
 
-  BWaveTile<FM, FN, T16> wt;
+  BWaveTile<FM, FN, BK, T16> wt;
   wt.zero();
   // fused bias grad: sum the STAGED bf16 A (=g) tile columns.  NOTE: this
   // sums bf16-rounded g — for exact-f32 db the caller uses the f32 path.
@@ -319,16 +319,16 @@ __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
   float dbsum = 0.f;
   const int bcol = threadIdx.x;
   int buf = 0;
-  bstage_cmaj<BM, T16>(a, n, c_beg, n0, c_end, n, lds_a[0]);
-  bstage_cmaj<BN, T16>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
+  bstage_cmaj<BM, BK, T16>(a, n, c_beg, n0, c_end, n, lds_a[0]);
+  bstage_cmaj<BN, BK, T16>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
   __syncthreads();
-  for (int cc = c_beg + BGEMM_BK; cc < c_end; cc += BGEMM_BK) {
-    bstage_cmaj<BM, T16>(a, n, cc, n0, c_end, n, lds_a[buf ^ 1]);
-    bstage_cmaj<BN, T16>(b, k2, cc, k0, c_end, k2, lds_b[buf ^ 1]);
+  for (int cc = c_beg + BK; cc < c_end; cc += BK) {
+    bstage_cmaj<BM, BK, T16>(a, n, cc, n0, c_end, n, lds_a[buf ^ 1]);
+    bstage_cmaj<BN, BK, T16>(b, k2, cc, k0, c_end, k2, lds_b[buf ^ 1]);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
     if (do_bias && bcol < BM)
 #pragma unroll
-      for (int r = 0; r < BGEMM_BK; ++r)
+      for (int r = 0; r < BK; ++r)
         dbsum += (float)lds_a[buf][bcol * LDW + r];
     __syncthreads();
     buf ^= 1;
@@ -336,7 +336,7 @@ __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
   wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
   if (do_bias && bcol < BM) {
 #pragma unroll
-    for (int r = 0; r < BGEMM_BK; ++r)
+    for (int r = 0; r < BK; ++r)
       dbsum += (float)lds_a[buf][bcol * LDW + r];
     if (n0 + bcol < n) atomicAdd(&dbias[n0 + bcol], dbsum);
   }
@@ -364,13 +364,25 @@ __global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
 // launchers
 // ---------------------------------------------------------------------------
 
+static int gemm_bk() {
+  static int bk = [] {
+    const char* e = getenv("PERTGNN_GEMM_BK");
+    return (e && atoi(e) == 64) ? 64 : 32;  // default 32 (4 blocks/CU)
+  }();
+  return bk;
+}
+
 void launch_gemm_bf16_nt(const float* a, const float* b, const float* bias,
                          float* c, int m, int n, int k, bool relu,
                          hipStream_t s) {
   if (m >= 512 && n >= 128) {
     const int grid = ((m + 127) / 128) * ((n + 127) / 128);
-   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
-        a, b, bias, c, m, n, k, relu ? 1 : 0);
+    if (gemm_bk() == 32)
+     hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128, 32>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
+          a, b, bias, c, m, n, k, relu ? 1 : 0);
+    else
+     hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128, 64>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
+          a, b, bias, c, m, n, k, relu ? 1 : 0);
   } else {
     const int grid = ((m + 63) / 64) * ((n + 63) / 64);
    hipLaunchKernelGGL(( gemm_bf16_nt_kernel<64, 64>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
@@ -383,8 +395,12 @@ void launch_gemm_bf16_nn(const float* a, const float* b, const float* bias,
                          hipStream_t s) {
   if (m >= 512 && k2 >= 128) {
     const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
-   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
-        a, b, bias, c, m, n, k2, relu ? 1 : 0);
+    if (gemm_bk() == 32)
+     hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128, 32>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
+          a, b, bias, c, m, n, k2, relu ? 1 : 0);
+    else
+     hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128, 64>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
+          a, b, bias, c, m, n, k2, relu ? 1 : 0);
   } else {
     const int grid = ((m + 63) / 64) * ((k2 + 63) / 64);
    hipLaunchKernelGGL(( gemm_bf16_nn_kernel<64, 64>), dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, 
@@ -400,17 +416,23 @@ void launch_gemm_bf16_tn(const float* a, const float* b, float* c,
   const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
   int slices = 1;
   while (tiles * slices < 512 && slices < 64 &&
-         (long)slices * BGEMM_BK * 4 < m)
+         (long)slices * 64 * 4 < m)
     slices *= 2;
   if (slices > 1)
     HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
   if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
-  if (big)
-   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128>)
-        , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
-                                                              m, n, k2,
-                                                              slices);
-  else
+  if (big) {
+    if (gemm_bk() == 32)
+     hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128, 32>)
+          , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
+                                                                m, n, k2,
+                                                                slices);
+    else
+     hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128, 64>)
+          , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
+                                                                m, n, k2,
+                                                                slices);
+  } else
    hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64>)
         , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
                                                               m, n, k2,
@@ -427,12 +449,12 @@ void launch_gemm_fp16_nt(const float* a, const float* b, const float* bias,
                          hipStream_t s) {
   if (m >= 512 && n >= 128) {
     const int grid = ((m + 127) / 128) * ((n + 127) / 128);
-   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128, _Float16>)
+   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128, 32, _Float16>)
         , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k,
                                                     relu ? 1 : 0);
   } else {
     const int grid = ((m + 63) / 64) * ((n + 63) / 64);
-   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<64, 64, _Float16>)
+   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<64, 64, 64, _Float16>)
         , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k,
                                                     relu ? 1 : 0);
   }
@@ -443,12 +465,12 @@ void launch_gemm_fp16_nn(const float* a, const float* b, const float* bias,
                          hipStream_t s) {
   if (m >= 512 && k2 >= 128) {
     const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
-   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128, _Float16>)
+   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128, 32, _Float16>)
         , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k2,
                                                     relu ? 1 : 0);
   } else {
     const int grid = ((m + 63) / 64) * ((k2 + 63) / 64);
-   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<64, 64, _Float16>)
+   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<64, 64, 64, _Float16>)
         , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k2,
                                                     relu ? 1 : 0);
   }
@@ -462,18 +484,18 @@ void launch_gemm_fp16_tn(const float* a, const float* b, float* c,
   const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
   int slices = 1;
   while (tiles * slices < 512 && slices < 64 &&
-         (long)slices * BGEMM_BK * 4 < m)
+         (long)slices * 64 * 4 < m)
     slices *= 2;
   if (slices > 1)
     HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
   if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
   if (big)
-   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128, _Float16>)
+   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128, 32, _Float16>)
         , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
                                                               m, n, k2,
                                                               slices);
   else
-   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64, _Float16>)
+   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64, 64, _Float16>)
         , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
                                                               m, n, k2,
                                                               slices);
