@@ -64,19 +64,37 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
   const bool interior = (free0 + BF <= free_max) &&
                         (contract0 + BK <= contract_max);
   const bool aligned = (ld & 3) == 0;
+  // the path choice is HOISTED outside the loops: a branch inside the
+  // staging loop makes hipcc emit per-element load+vmcnt(0) chains
+  // (guide §5 ".s-level traps" (c))
+  if (interior && aligned) {
+    f32x4 v[BF / FSTEP];
 #pragma unroll
-  for (int half = 0; half < BF / FSTEP; ++half) {
-    const int ff = f + half * FSTEP;
-    if (interior && aligned) {
-      const f32x4 v = *reinterpret_cast<const f32x4*>(
-          &g[(long)(free0 + ff) * ld + contract0 + cq]);
-      *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[ff * LDW + cq]) =
-          pack4<T16>(v[0], v[1], v[2], v[3]);
-    } else if (interior) {  // odd leading dim (layer-1 K=9+H): unchecked scalars
-      const float* row = &g[(long)(free0 + ff) * ld + contract0 + cq];
-      *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[ff * LDW + cq]) =
-          pack4<T16>(row[0], row[1], row[2], row[3]);
-    } else {
+    for (int half = 0; half < BF / FSTEP; ++half)
+      v[half] = *reinterpret_cast<const f32x4*>(
+          &g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq]);
+#pragma unroll
+    for (int half = 0; half < BF / FSTEP; ++half)
+      *reinterpret_cast<typename Vec16<T16>::v4*>(
+          &lds[(f + half * FSTEP) * LDW + cq]) =
+          pack4<T16>(v[half][0], v[half][1], v[half][2], v[half][3]);
+  } else if (interior) {  // odd leading dim (layer-1 K=9+H): unchecked scalars
+    float v[BF / FSTEP][4];
+#pragma unroll
+    for (int half = 0; half < BF / FSTEP; ++half) {
+      const float* row = &g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) v[half][u] = row[u];
+    }
+#pragma unroll
+    for (int half = 0; half < BF / FSTEP; ++half)
+      *reinterpret_cast<typename Vec16<T16>::v4*>(
+          &lds[(f + half * FSTEP) * LDW + cq]) =
+          pack4<T16>(v[half][0], v[half][1], v[half][2], v[half][3]);
+  } else {
+#pragma unroll
+    for (int half = 0; half < BF / FSTEP; ++half) {
+      const int ff = f + half * FSTEP;
       const int gf = free0 + ff;
       float v[4];
 #pragma unroll
@@ -108,20 +126,50 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
   const bool interior = (contract0 + BK <= contract_max) &&
                         (free0 + BF <= free_max);
   const bool aligned = ((ld & 3) == 0) && ((free0 & 3) == 0);
+  constexpr int HALVES = (BK > CSTEP ? BK / CSTEP : 1);
+  // path choice hoisted outside the loops (guide §5 ".s-level traps" (c))
+  if (interior && aligned) {
+    f32x4 v[HALVES][4];
 #pragma unroll
-  for (int half = 0; half < (BK > CSTEP ? BK / CSTEP : 1); ++half) {
-    const int cc = cb + half * CSTEP;
-    float v[4][4];
+    for (int half = 0; half < HALVES; ++half)
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      if (interior && aligned) {
-        *reinterpret_cast<f32x4*>(v[u]) = *reinterpret_cast<const f32x4*>(
-            &g[(long)(contract0 + cc + u) * ld + free0 + fq]);
-      } else if (interior) {
-        const float* row = &g[(long)(contract0 + cc + u) * ld + free0 + fq];
+      for (int u = 0; u < 4; ++u)
+        v[half][u] = *reinterpret_cast<const f32x4*>(
+            &g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq]);
 #pragma unroll
-        for (int w = 0; w < 4; ++w) v[u][w] = row[w];
-      } else {
+    for (int half = 0; half < HALVES; ++half)
+#pragma unroll
+      for (int w = 0; w < 4; ++w)
+        *reinterpret_cast<typename Vec16<T16>::v4*>(
+            &lds[(fq + w) * LDW + cb + half * CSTEP]) =
+            pack4<T16>(v[half][0][w], v[half][1][w], v[half][2][w],
+                       v[half][3][w]);
+  } else if (interior) {
+    float v[HALVES][4][4];
+#pragma unroll
+    for (int half = 0; half < HALVES; ++half)
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const float* row =
+            &g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq];
+#pragma unroll
+        for (int w = 0; w < 4; ++w) v[half][u][w] = row[w];
+      }
+#pragma unroll
+    for (int half = 0; half < HALVES; ++half)
+#pragma unroll
+      for (int w = 0; w < 4; ++w)
+        *reinterpret_cast<typename Vec16<T16>::v4*>(
+            &lds[(fq + w) * LDW + cb + half * CSTEP]) =
+            pack4<T16>(v[half][0][w], v[half][1][w], v[half][2][w],
+                       v[half][3][w]);
+  } else {
+#pragma unroll
+    for (int half = 0; half < HALVES; ++half) {
+      const int cc = cb + half * CSTEP;
+      float v[4][4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
         const int gc = contract0 + cc + u;
 #pragma unroll
         for (int w = 0; w < 4; ++w) {
@@ -131,11 +179,11 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
                         : 0.f;
         }
       }
-    }
 #pragma unroll
-    for (int w = 0; w < 4; ++w)
-      *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[(fq + w) * LDW + cc]) =
-          pack4<T16>(v[0][w], v[1][w], v[2][w], v[3][w]);
+      for (int w = 0; w < 4; ++w)
+        *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[(fq + w) * LDW + cc]) =
+            pack4<T16>(v[0][w], v[1][w], v[2][w], v[3][w]);
+    }
   }
 }
 
