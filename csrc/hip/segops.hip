@@ -305,21 +305,60 @@ __global__ void vocab_scatter_kernel(const float* __restrict__ g,
     if (acc[t] != 0.f) atomicAdd(&dtable[t], acc[t]);
 }
 
+// column-sliced variant: each of the 4 waves owns an exclusive h/4 column
+// slice of the LDS table and iterates ALL of the block's rows — no LDS
+// atomics at all (the cross-wave contention dominated the atomic version;
+// PMC: 87% wave stall).  Requires h % 256 == 0.
+__global__ void vocab_scatter_sliced_kernel(const float* __restrict__ g,
+                                            const long* __restrict__ idx,
+                                            long idx_stride,
+                                            float* __restrict__ dtable, long n,
+                                            int rows, int h, int gstride,
+                                            int col_off) {
+  extern __shared__ float acc[];  // [rows*h]
+  const long vh = (long)rows * h;
+  for (long t = threadIdx.x; t < vh; t += blockDim.x) acc[t] = 0.f;
+  __syncthreads();
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int slice = h / WAVES_PER_BLOCK;       // 64 at h=256
+  const int c0 = wid * slice;
+  const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * rows_per_block;
+  const long r1 = min(n, r0 + rows_per_block);
+  for (long r = r0; r < r1; ++r) {
+    const long v = idx[r * idx_stride];
+    for (int c = c0 + lane; c < c0 + slice; c += PERTGNN_WAVE)
+      acc[v * h + c] += g[r * gstride + col_off + c];
+  }
+  __syncthreads();
+  for (long t = threadIdx.x; t < vh; t += blockDim.x)
+    if (acc[t] != 0.f) atomicAdd(&dtable[t], acc[t]);
+}
+
 void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
                           float* dtable, long n, int rows, int h, int gstride,
                           int col_off, hipStream_t s) {
   HIP_CHECK(hipMemsetAsync(dtable, 0, (long)rows * h * sizeof(float), s));
   if (n == 0) return;
   const size_t lds = (size_t)rows * h * sizeof(float);
+  const bool sliced = (h % (4 * PERTGNN_WAVE) == 0);
+  const void* fn = sliced ? (const void*)vocab_scatter_sliced_kernel
+                          : (const void*)vocab_scatter_kernel;
   if (lds > 64 * 1024) {
     HIP_CHECK(hipFuncSetAttribute(
-        (const void*)vocab_scatter_kernel,
-        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
+        fn, hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
   const int blocks = (int)min((long)256, (n + 63) / 64);
-  vocab_scatter_kernel<<<dim3(blocks), dim3(WAVES_PER_BLOCK * PERTGNN_WAVE),
-                         lds, s>>>(g, idx, idx_stride, dtable, n, rows, h,
-                                   gstride, col_off);
+  if (sliced)
+    vocab_scatter_sliced_kernel<<<dim3(blocks),
+                                  dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds,
+                                  s>>>(g, idx, idx_stride, dtable, n, rows, h,
+                                       gstride, col_off);
+  else
+    vocab_scatter_kernel<<<dim3(blocks), dim3(WAVES_PER_BLOCK * PERTGNN_WAVE),
+                           lds, s>>>(g, idx, idx_stride, dtable, n, rows, h,
+                                     gstride, col_off);
 }
 
 // entry embedding gather: out[b] = table[idx[b]] — plain gather (fwd) +
@@ -408,7 +447,25 @@ __global__ void bn_stats_partial_kernel(const float* __restrict__ x, long n,
   const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
   const long r0 = (long)blockIdx.x * rows_per_block;
   const long r1 = min(n, r0 + rows_per_block);
-  for (long r = r0; r < r1; ++r) {
+  long r = r0;
+  // 4-row unroll: independent loads overlap (single-row loop is latency-bound)
+  for (; r + 4 <= r1; r += 4) {
+    if (c0 < h) {
+      float v[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) v[u] = x[(r + u) * h + c0];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) { s0 += v[u]; q0 += v[u] * v[u]; }
+    }
+    if (c1 < h) {
+      float v[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) v[u] = x[(r + u) * h + c1];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) { s1 += v[u]; q1 += v[u] * v[u]; }
+    }
+  }
+  for (; r < r1; ++r) {
     if (c0 < h) {
       const float v = x[r * h + c0];
       s0 += v; q0 += v * v;
@@ -482,18 +539,55 @@ __global__ void bn_bwd_partial_kernel(const float* __restrict__ g,
   const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
   const long r0 = (long)blockIdx.x * rows_per_block;
   const long r1 = min(n, r0 + rows_per_block);
-  for (long r = r0; r < r1; ++r) {
+  const float m0 = (c0 < h) ? mean[c0] : 0.f;
+  const float i0 = (c0 < h) ? invstd[c0] : 0.f;
+  const float m1 = (c1 < h) ? mean[c1] : 0.f;
+  const float i1 = (c1 < h) ? invstd[c1] : 0.f;
+  long r = r0;
+  for (; r + 4 <= r1; r += 4) {
+    if (c0 < h) {
+      float gv[4], yv[4], xv[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        gv[u] = g[(r + u) * h + c0];
+        yv[u] = y[(r + u) * h + c0];
+        xv[u] = x[(r + u) * h + c0];
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        float gm = (relu && yv[u] <= 0.f) ? 0.f : gv[u];
+        s0 += gm;
+        q0 += gm * (xv[u] - m0) * i0;
+      }
+    }
+    if (c1 < h) {
+      float gv[4], yv[4], xv[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        gv[u] = g[(r + u) * h + c1];
+        yv[u] = y[(r + u) * h + c1];
+        xv[u] = x[(r + u) * h + c1];
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        float gm = (relu && yv[u] <= 0.f) ? 0.f : gv[u];
+        s1 += gm;
+        q1 += gm * (xv[u] - m1) * i1;
+      }
+    }
+  }
+  for (; r < r1; ++r) {
     if (c0 < h) {
       float gm = g[r * h + c0];
       if (relu && y[r * h + c0] <= 0.f) gm = 0.f;
       s0 += gm;
-      q0 += gm * (x[r * h + c0] - mean[c0]) * invstd[c0];
+      q0 += gm * (x[r * h + c0] - m0) * i0;
     }
     if (c1 < h) {
       float gm = g[r * h + c1];
       if (relu && y[r * h + c1] <= 0.f) gm = 0.f;
       s1 += gm;
-      q1 += gm * (x[r * h + c1] - mean[c1]) * invstd[c1];
+      q1 += gm * (x[r * h + c1] - m1) * i1;
     }
   }
   if (c0 < h && r0 < r1) {
