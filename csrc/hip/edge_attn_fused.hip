@@ -204,7 +204,7 @@ template <int VPT, bool VEC>
 __global__ void edge_attn_fused_bwd_col_kernel(
     const float* __restrict__ dek, const float* __restrict__ dev,
     const int* __restrict__ col_ptr, const int* __restrict__ csc_eid,
-    float* __restrict__ dqkvs, int n, int h) {
+    float* __restrict__ dqkvs, float* __restrict__ de, int n, int h) {
   using S = Slice<VPT, VEC>;
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
@@ -216,14 +216,18 @@ __global__ void edge_attn_fused_bwd_col_kernel(
   for (int j = 0; j < VPT; ++j) { ka[j] = 0.f; va[j] = 0.f; }
   for (int p = col_ptr[row]; p < col_ptr[row + 1]; ++p) {
     const long eid = csc_eid[p];
-    float dk1[VPT], dv1[VPT];
+    float dk1[VPT], dv1[VPT], des[VPT];
     S::load(&dek[eid * h], lane, h, dk1);
     S::load(&dev[eid * h], lane, h, dv1);
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
       ka[j] += dk1[j];
       va[j] += dv1[j];
+      des[j] = dk1[j] + dv1[j];
     }
+    // every edge appears exactly once in the CSC sweep: de rides along,
+    // replacing a separate elementwise pass over dek/dev
+    S::store(&de[eid * h], lane, h, des);
   }
   S::store(&dqkvs[(long)row * ld + h], lane, h, ka);
   S::store(&dqkvs[(long)row * ld + 2 * h], lane, h, va);
@@ -290,23 +294,17 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
           g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,    \
           dek, dev, n, h, scale);                                              \
       edge_attn_fused_bwd_col_kernel<V, true><<<grid, block, 0, stream>>>(     \
-          dek, dev, col_ptr, csc_eid, dqkvs, n, h);                            \
+          dek, dev, col_ptr, csc_eid, dqkvs, de, n, h);                        \
     } else {                                                                   \
       edge_attn_fused_bwd_row_kernel<V, false><<<grid, block, 0, stream>>>(    \
           g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,    \
           dek, dev, n, h, scale);                                              \
       edge_attn_fused_bwd_col_kernel<V, false><<<grid, block, 0, stream>>>(    \
-          dek, dev, col_ptr, csc_eid, dqkvs, n, h);                            \
+          dek, dev, col_ptr, csc_eid, dqkvs, de, n, h);                        \
     }                                                                          \
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
     default: abort();
-  }
-  const long numel = num_edges * h;
-  if (numel > 0) {
-    const int tpb = 256;
-    const int blocks = (int)min((numel + tpb - 1) / tpb, (long)4096);
-    add2_kernel<<<dim3(blocks), dim3(tpb), 0, stream>>>(dek, dev, de, numel);
   }
 }
