@@ -476,3 +476,20 @@ def test_span_attr_layout_gpu():
                        csr=csr, num_graphs=2)
     assert torch.allclose(gp_gpu.cpu(), gp_cpu, atol=2e-3, rtol=1e-3), \
         (gp_gpu.cpu() - gp_cpu).abs().max()
+
+
+def test_fused_attention_empty_and_isolated_rows():
+    """Fused path with zero edges and isolated nodes: out == skip segment."""
+    require_ext()
+    import pertgnn._C as C
+    n, h = 40, 256
+    qkvs = torch.randn(n, 4 * h, device=DEV)
+    pifc = torch.randn(5, h, device=DEV)
+    prpc = torch.randn(3, h, device=DEV)
+    ea = torch.zeros(0, 2, dtype=torch.long, device=DEV)
+    row_ptr = torch.zeros(n + 1, dtype=torch.int32, device=DEV)
+    csr_src = torch.zeros(0, dtype=torch.int32, device=DEV)
+    out, alpha = C.edge_attn_fused_fwd(qkvs, pifc, prpc, ea, row_ptr, csr_src)
+    torch.cuda.synchronize()
+    assert alpha.numel() == 0
+    assert torch.allclose(out, qkvs[:, 3 * h:], atol=1e-6)
