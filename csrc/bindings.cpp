@@ -92,6 +92,8 @@ void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   hipStream_t);
 void launch_vocab_scatter(const float*, const long*, long, float*, long, int,
                           int, int, int, hipStream_t);
+void launch_vocab_scatter_dual(const float*, const long*, int, float*, float*,
+                               long, int, int, int, hipStream_t);
 void launch_edge_attn_fused_fwd(const float*, const float*, const float*,
                                 const long*, int, const int*, const int*,
                                 float*, float*, int, int, hipStream_t);
@@ -695,7 +697,25 @@ std::vector<torch::Tensor> linear_wgrad(torch::Tensor g, torch::Tensor x,
   return {dw, db};
 }
 
+// both dP tables in one pass over de (h % 256 == 0 required)
+std::vector<torch::Tensor> vocab_scatter_dual(torch::Tensor g,
+                                              torch::Tensor ea, int64_t rows0,
+                                              int64_t rows1) {
+  CHECK_IN(g); CHECK_IN(ea);
+  const int h = g.size(1);
+  TORCH_CHECK(h % 256 == 0, "dual scatter needs h % 256 == 0");
+  TORCH_CHECK((size_t)(rows0 + rows1) * h * 4 <= 160 * 1024, "tables too large");
+  auto dt0 = torch::empty({rows0, h}, g.options());
+  auto dt1 = torch::empty({rows1, h}, g.options());
+  launch_vocab_scatter_dual(g.data_ptr<float>(), ea.data_ptr<long>(),
+                            (int)ea.size(1), dt0.data_ptr<float>(),
+                            dt1.data_ptr<float>(), g.size(0), (int)rows0,
+                            (int)rows1, h, cur_stream());
+  return {dt0, dt1};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("vocab_scatter_dual", &vocab_scatter_dual);
   mod.def("linear_dgrad", &linear_dgrad);
   mod.def("linear_wgrad", &linear_wgrad);
   mod.def("bn_stats", &bn_stats);

@@ -95,8 +95,12 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
         # dP tables (per-vocab segment sums of de) are independent of the
         # CSC dk/dv pass queued above — overlap them on the side stream.
         if not _overlap_enabled():
-            dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
-            dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
+            h = de.shape[1]
+            if h % 256 == 0 and (pifc.shape[0] + prpc.shape[0]) * h * 4 <= 160 * 1024:
+                dpifc, dprpc = m.vocab_scatter_dual(de, edge_attr, pifc.shape[0], prpc.shape[0])
+            else:
+                dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], h, 0)
+                dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], h, 0)
             return dqkvs, dpifc, dprpc, None, None, None, None, None
         cur = torch.cuda.current_stream()
         side = _side_stream()
