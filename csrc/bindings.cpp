@@ -102,6 +102,20 @@ void launch_edge_attn_fused_bwd(const float*, const float*, const float*,
                                 const int*, const int*, const int*,
                                 const int*, float*, float*, float*, float*,
                                 int, int, long, hipStream_t);
+void launch_edge_attn_fused_fwd16(const void*, const float*, const float*,
+                                  const long*, int, const int*, const int*,
+                                  float*, float*, int, int, hipStream_t);
+void launch_edge_attn_fused_bwd16(const float*, const void*, const float*,
+                                  const float*, const long*, int,
+                                  const float*, const int*, const int*,
+                                  const int*, const int*, void*, float*,
+                                  float*, float*, int, int, long, hipStream_t);
+void launch_gemm_bf16_nt_o16(const float*, const float*, const float*, void*,
+                             int, int, int, hipStream_t);
+void launch_gemm_bf16_nn_a16(const void*, const float*, float*, int, int, int,
+                             hipStream_t);
+void launch_gemm_bf16_tn_a16(const void*, const float*, float*, float*, int,
+                             int, int, hipStream_t);
 std::vector<torch::Tensor> collate_native(
     std::vector<torch::Tensor>, std::vector<torch::Tensor>,
     std::vector<torch::Tensor>, std::vector<torch::Tensor>,
@@ -539,13 +553,22 @@ std::vector<torch::Tensor> edge_attn_fused_fwd(
   const int n = qkvs.size(0);
   const int h = qkvs.size(1) / 4;
   TORCH_CHECK(h <= 512, "H must be <= 512");
-  auto out = torch::empty({n, h}, qkvs.options());
-  auto alpha = torch::empty({ea.size(0)}, qkvs.options());
-  launch_edge_attn_fused_fwd(
-      qkvs.data_ptr<float>(), pifc.data_ptr<float>(), prpc.data_ptr<float>(),
-      ea.data_ptr<long>(), (int)ea.size(1), row_ptr.data_ptr<int>(),
-      csr_src.data_ptr<int>(), out.data_ptr<float>(), alpha.data_ptr<float>(),
-      n, h, cur_stream());
+  auto fopt = qkvs.options().dtype(torch::kFloat32);
+  auto out = torch::empty({n, h}, fopt);
+  auto alpha = torch::empty({ea.size(0)}, fopt);
+  if (qkvs.scalar_type() == torch::kBFloat16) {
+    launch_edge_attn_fused_fwd16(
+        qkvs.data_ptr(), pifc.data_ptr<float>(),
+        prpc.data_ptr<float>(), ea.data_ptr<long>(), (int)ea.size(1),
+        row_ptr.data_ptr<int>(), csr_src.data_ptr<int>(),
+        out.data_ptr<float>(), alpha.data_ptr<float>(), n, h, cur_stream());
+  } else {
+    launch_edge_attn_fused_fwd(
+        qkvs.data_ptr<float>(), pifc.data_ptr<float>(), prpc.data_ptr<float>(),
+        ea.data_ptr<long>(), (int)ea.size(1), row_ptr.data_ptr<int>(),
+        csr_src.data_ptr<int>(), out.data_ptr<float>(), alpha.data_ptr<float>(),
+        n, h, cur_stream());
+  }
   return {out, alpha};
 }
 
@@ -558,17 +581,29 @@ std::vector<torch::Tensor> edge_attn_fused_bwd(
   const int n = qkvs.size(0);
   const int h = qkvs.size(1) / 4;
   const long ne = ea.size(0);
+  auto fopt = qkvs.options().dtype(torch::kFloat32);
   auto dqkvs = torch::empty_like(qkvs);
-  auto de = torch::empty({ne, h}, qkvs.options());
-  auto dek = torch::empty({ne, h}, qkvs.options());
-  auto dev = torch::empty({ne, h}, qkvs.options());
-  launch_edge_attn_fused_bwd(
-      g.data_ptr<float>(), qkvs.data_ptr<float>(), pifc.data_ptr<float>(),
-      prpc.data_ptr<float>(), ea.data_ptr<long>(), (int)ea.size(1),
-      alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
-      csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
-      csc_eid.data_ptr<int>(), dqkvs.data_ptr<float>(), de.data_ptr<float>(),
-      dek.data_ptr<float>(), dev.data_ptr<float>(), n, h, ne, cur_stream());
+  auto de = torch::empty({ne, h}, fopt);
+  auto dek = torch::empty({ne, h}, fopt);
+  auto dev = torch::empty({ne, h}, fopt);
+  if (qkvs.scalar_type() == torch::kBFloat16) {
+    launch_edge_attn_fused_bwd16(
+        g.data_ptr<float>(), qkvs.data_ptr(),
+        pifc.data_ptr<float>(), prpc.data_ptr<float>(), ea.data_ptr<long>(),
+        (int)ea.size(1), alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
+        csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
+        csc_eid.data_ptr<int>(), dqkvs.data_ptr(),
+        de.data_ptr<float>(), dek.data_ptr<float>(), dev.data_ptr<float>(), n,
+        h, ne, cur_stream());
+  } else {
+    launch_edge_attn_fused_bwd(
+        g.data_ptr<float>(), qkvs.data_ptr<float>(), pifc.data_ptr<float>(),
+        prpc.data_ptr<float>(), ea.data_ptr<long>(), (int)ea.size(1),
+        alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
+        csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
+        csc_eid.data_ptr<int>(), dqkvs.data_ptr<float>(), de.data_ptr<float>(),
+        dek.data_ptr<float>(), dev.data_ptr<float>(), n, h, ne, cur_stream());
+  }
   return {dqkvs, de};
 }
 
@@ -714,7 +749,57 @@ std::vector<torch::Tensor> vocab_scatter_dual(torch::Tensor g,
   return {dt0, dt1};
 }
 
+// bf16-activation-mode linear: fp32 x/w in, bf16 C out; backward from bf16 g
+torch::Tensor linear_fwd_bf16_o16(torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor b) {
+  CHECK_IN(x); CHECK_IN(w);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = w.size(0);
+  auto y = torch::empty({m, n}, x.options().dtype(torch::kBFloat16));
+  const float* bias = nullptr;
+  if (b.defined() && b.numel() > 0) bias = b.data_ptr<float>();
+  launch_gemm_bf16_nt_o16(x.data_ptr<float>(), w.data_ptr<float>(), bias,
+                          y.data_ptr(), m, n, k,
+                          cur_stream());
+  return y;
+}
+
+torch::Tensor linear_dgrad16(torch::Tensor g, torch::Tensor w) {
+  CHECK_IN(g); CHECK_IN(w);
+  const int m = g.size(0);
+  const int n = w.size(0);
+  const int k = w.size(1);
+  auto dx = torch::empty({m, k}, w.options());
+  launch_gemm_bf16_nn_a16(g.data_ptr(),
+                          w.data_ptr<float>(), dx.data_ptr<float>(), m, n, k,
+                          cur_stream());
+  return dx;
+}
+
+std::vector<torch::Tensor> linear_wgrad16(torch::Tensor g, torch::Tensor x,
+                                          bool has_bias) {
+  CHECK_IN(g); CHECK_IN(x);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = g.size(1);
+  auto dw = torch::empty({n, k}, x.options());
+  torch::Tensor db = torch::empty({0}, x.options());
+  float* db_ptr = nullptr;
+  if (has_bias) {
+    db = torch::empty({n}, x.options());
+    db_ptr = db.data_ptr<float>();
+  }
+  launch_gemm_bf16_tn_a16(g.data_ptr(),
+                          x.data_ptr<float>(), dw.data_ptr<float>(), db_ptr,
+                          m, n, k, cur_stream());
+  return {dw, db};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("linear_fwd_bf16_o16", &linear_fwd_bf16_o16);
+  mod.def("linear_dgrad16", &linear_dgrad16);
+  mod.def("linear_wgrad16", &linear_wgrad16);
   mod.def("vocab_scatter_dual", &vocab_scatter_dual);
   mod.def("linear_dgrad", &linear_dgrad);
   mod.def("linear_wgrad", &linear_wgrad);

@@ -27,8 +27,10 @@
 #define WAVES_PER_BLOCK 4
 
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4_t;
 
-// row-slice accessors for the two column mappings
+// row-slice accessors for the two column mappings; fp32 and bf16 overloads
+// (bf16 rows are the qkvs/dqkvs tensors in bf16-resident-activation mode).
 template <int VPT, bool VEC>
 struct Slice {
   static __device__ __forceinline__ void load(const float* base, int lane,
@@ -43,6 +45,24 @@ struct Slice {
       for (int j = 0; j < VPT; ++j) {
         const int c = lane + j * PERTGNN_WAVE;
         dst[j] = (c < h) ? base[c] : 0.f;
+      }
+    }
+  }
+  static __device__ __forceinline__ void load(const __bf16* base, int lane,
+                                              int h, float (&dst)[VPT]) {
+    if constexpr (VEC) {
+#pragma unroll
+      for (int q = 0; q < VPT; q += 4) {
+        const bf16x4_t v =
+            *reinterpret_cast<const bf16x4_t*>(&base[lane * VPT + q]);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) dst[q + u] = (float)v[u];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPT; ++j) {
+        const int c = lane + j * PERTGNN_WAVE;
+        dst[j] = (c < h) ? (float)base[c] : 0.f;
       }
     }
   }
@@ -80,11 +100,29 @@ struct Slice {
       }
     }
   }
+  static __device__ __forceinline__ void store(__bf16* base, int lane, int h,
+                                               const float (&src)[VPT]) {
+    if constexpr (VEC) {
+#pragma unroll
+      for (int q = 0; q < VPT; q += 4) {
+        bf16x4_t v;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) v[u] = (__bf16)src[q + u];
+        *reinterpret_cast<bf16x4_t*>(&base[lane * VPT + q]) = v;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPT; ++j) {
+        const int c = lane + j * PERTGNN_WAVE;
+        if (c < h) base[c] = (__bf16)src[j];
+      }
+    }
+  }
 };
 
-template <int VPT, bool VEC>
+template <int VPT, bool VEC, typename QT = float>
 __global__ void edge_attn_fused_fwd_kernel(
-    const float* __restrict__ qkvs,  // [N, 4h]
+    const QT* __restrict__ qkvs,  // [N, 4h]
     const float* __restrict__ pifc,  // [Vi, h]
     const float* __restrict__ prpc,  // [Vr, h]
     const long* __restrict__ ea, int astride,
@@ -139,13 +177,13 @@ __global__ void edge_attn_fused_fwd_kernel(
     alpha[p] = __expf(alpha[p] - m) * inv_s;
 }
 
-template <int VPT, bool VEC>
+template <int VPT, bool VEC, typename QT = float>
 __global__ void edge_attn_fused_bwd_row_kernel(
-    const float* __restrict__ g, const float* __restrict__ qkvs,
+    const float* __restrict__ g, const QT* __restrict__ qkvs,
     const float* __restrict__ pifc, const float* __restrict__ prpc,
     const long* __restrict__ ea, int astride, const float* __restrict__ alpha,
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
-    float* __restrict__ dqkvs, float* __restrict__ dek,
+    QT* __restrict__ dqkvs, float* __restrict__ dek,
     float* __restrict__ dev, int n, int h, float scale) {
   using S = Slice<VPT, VEC>;
   const int wid = threadIdx.x / PERTGNN_WAVE;
@@ -200,11 +238,11 @@ __global__ void edge_attn_fused_bwd_row_kernel(
   S::store(&dqkvs[row * ld + 3 * h], lane, h, gr);     // dskip = g
 }
 
-template <int VPT, bool VEC>
+template <int VPT, bool VEC, typename QT = float>
 __global__ void edge_attn_fused_bwd_col_kernel(
     const float* __restrict__ dek, const float* __restrict__ dev,
     const int* __restrict__ col_ptr, const int* __restrict__ csc_eid,
-    float* __restrict__ dqkvs, float* __restrict__ de, int n, int h) {
+    QT* __restrict__ dqkvs, float* __restrict__ de, int n, int h) {
   using S = Slice<VPT, VEC>;
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
@@ -306,5 +344,69 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
     default: abort();
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// bf16-qkvs launchers (activation-resident bf16 mode; requires h % 256 == 0)
+// ---------------------------------------------------------------------------
+
+void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
+                                  const float* prpc, const long* ea,
+                                  int astride, const int* row_ptr,
+                                  const int* csr_src, float* out, float* alpha,
+                                  int n, int h, hipStream_t stream) {
+  const __bf16* qkvs = (const __bf16*)qkvs_v;
+  if (n == 0) return;
+  const float scale = 1.f / std::sqrt((float)h);
+  const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
+  const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
+  const int vpt = h / PERTGNN_WAVE;
+  if (vpt == 4)
+    edge_attn_fused_fwd_kernel<4, true, __bf16><<<grid, block, 0, stream>>>(
+        qkvs, pifc, prpc, ea, astride, row_ptr, csr_src, out, alpha, n, h,
+        scale);
+  else if (vpt == 8)
+    edge_attn_fused_fwd_kernel<8, true, __bf16><<<grid, block, 0, stream>>>(
+        qkvs, pifc, prpc, ea, astride, row_ptr, csr_src, out, alpha, n, h,
+        scale);
+  else
+    abort();
+}
+
+void launch_edge_attn_fused_bwd16(const float* g, const void* qkvs_v,
+                                  const float* pifc, const float* prpc,
+                                  const long* ea, int astride,
+                                  const float* alpha, const int* row_ptr,
+                                  const int* csr_src, const int* col_ptr,
+                                  const int* csc_eid, void* dqkvs_v, float* de,
+                                  float* dek, float* dev, int n, int h,
+                                  long num_edges, hipStream_t stream) {
+  const __bf16* qkvs = (const __bf16*)qkvs_v;
+  __bf16* dqkvs = (__bf16*)dqkvs_v;
+  if (n == 0) return;
+  const float scale = 1.f / std::sqrt((float)h);
+  const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
+  const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
+  const int vpt = h / PERTGNN_WAVE;
+  if (vpt == 4) {
+    edge_attn_fused_bwd_row_kernel<4, true, __bf16>
+        <<<grid, block, 0, stream>>>(g, qkvs, pifc, prpc, ea, astride, alpha,
+                                     row_ptr, csr_src, dqkvs, dek, dev, n, h,
+                                     scale);
+    edge_attn_fused_bwd_col_kernel<4, true, __bf16>
+        <<<grid, block, 0, stream>>>(dek, dev, col_ptr, csc_eid, dqkvs, de, n,
+                                     h);
+  } else if (vpt == 8) {
+    edge_attn_fused_bwd_row_kernel<8, true, __bf16>
+        <<<grid, block, 0, stream>>>(g, qkvs, pifc, prpc, ea, astride, alpha,
+                                     row_ptr, csr_src, dqkvs, dek, dev, n, h,
+                                     scale);
+    edge_attn_fused_bwd_col_kernel<8, true, __bf16>
+        <<<grid, block, 0, stream>>>(dek, dev, col_ptr, csc_eid, dqkvs, de, n,
+                                     h);
+  } else {
+    abort();
   }
 }

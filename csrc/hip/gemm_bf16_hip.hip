@@ -48,11 +48,23 @@ __device__ __forceinline__ typename Vec16<T16>::v4 pack4(float a, float b,
   return r;
 }
 
+// 4 consecutive source elements -> float[4] (vector load for both dtypes)
+__device__ __forceinline__ void ld4(const float* p, float (&v)[4]) {
+  *reinterpret_cast<f32x4*>(v) = *reinterpret_cast<const f32x4*>(p);
+}
+__device__ __forceinline__ void ld4(const __bf16* p, float (&v)[4]) {
+  const bf16x4 b = *reinterpret_cast<const bf16x4*>(p);
+#pragma unroll
+  for (int u = 0; u < 4; ++u) v[u] = (float)b[u];
+}
+__device__ __forceinline__ float ld1(const float* p) { return *p; }
+__device__ __forceinline__ float ld1(const __bf16* p) { return (float)*p; }
+
 // global [free][contract] (contract-minor) -> LDS [free][BK+PAD]
 // 4 bf16 packed into one 8-byte LDS write (scalar u16 LDS writes are ~2x
 // slower — guide G13 applies to LDS too).
-template <int BF, int BK, typename T16>
-__device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
+template <int BF, int BK, typename T16, typename TA>
+__device__ __forceinline__ void bstage_cmin(const TA* __restrict__ g,
                                             long ld, int free0, int contract0,
                                             int free_max, int contract_max,
                                             T16* lds) {
@@ -69,11 +81,10 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
   // staging loop makes hipcc emit per-element load+vmcnt(0) chains
   // (guide §5 ".s-level traps" (c))
   if (interior && aligned) {
-    f32x4 v[BF / FSTEP];
+    float v[BF / FSTEP][4];
 #pragma unroll
     for (int half = 0; half < BF / FSTEP; ++half)
-      v[half] = *reinterpret_cast<const f32x4*>(
-          &g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq]);
+      ld4(&g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq], v[half]);
 #pragma unroll
     for (int half = 0; half < BF / FSTEP; ++half)
       *reinterpret_cast<typename Vec16<T16>::v4*>(
@@ -83,9 +94,9 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
     float v[BF / FSTEP][4];
 #pragma unroll
     for (int half = 0; half < BF / FSTEP; ++half) {
-      const float* row = &g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq];
+      const TA* row = &g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) v[half][u] = row[u];
+      for (int u = 0; u < 4; ++u) v[half][u] = ld1(row + u);
     }
 #pragma unroll
     for (int half = 0; half < BF / FSTEP; ++half)
@@ -101,8 +112,9 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         const int gc = contract0 + cq + u;
-        v[u] = (gf < free_max && gc < contract_max) ? g[(long)gf * ld + gc]
-                                                    : 0.f;
+        v[u] = (gf < free_max && gc < contract_max)
+                   ? ld1(&g[(long)gf * ld + gc])
+                   : 0.f;
       }
       *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[ff * LDW + cq]) =
           pack4<T16>(v[0], v[1], v[2], v[3]);
@@ -113,8 +125,8 @@ __device__ __forceinline__ void bstage_cmin(const float* __restrict__ g,
 // global [contract][free] (contract-major) -> LDS [free][BK+PAD]: each thread
 // transposes a 4x4 block in registers (4 coalesced f32x4 loads from 4
 // contract rows), then writes 4 packed 8-byte LDS rows.
-template <int BF, int BK, typename T16>
-__device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
+template <int BF, int BK, typename T16, typename TA>
+__device__ __forceinline__ void bstage_cmaj(const TA* __restrict__ g,
                                             long ld, int contract0, int free0,
                                             int contract_max, int free_max,
                                             T16* lds) {
@@ -130,13 +142,13 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
   constexpr int HALVES = (BK > CSTEP ? BK / CSTEP : 1);
   // path choice hoisted outside the loops (guide §5 ".s-level traps" (c))
   if (interior && aligned) {
-    f32x4 v[HALVES][4];
+    float v[HALVES][4][4];
 #pragma unroll
     for (int half = 0; half < HALVES; ++half)
 #pragma unroll
       for (int u = 0; u < 4; ++u)
-        v[half][u] = *reinterpret_cast<const f32x4*>(
-            &g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq]);
+        ld4(&g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq],
+            v[half][u]);
 #pragma unroll
     for (int half = 0; half < HALVES; ++half)
 #pragma unroll
@@ -151,10 +163,10 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
     for (int half = 0; half < HALVES; ++half)
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        const float* row =
+        const TA* row =
             &g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq];
 #pragma unroll
-        for (int w = 0; w < 4; ++w) v[half][u][w] = row[w];
+        for (int w = 0; w < 4; ++w) v[half][u][w] = ld1(row + w);
       }
 #pragma unroll
     for (int half = 0; half < HALVES; ++half)
@@ -176,7 +188,7 @@ __device__ __forceinline__ void bstage_cmaj(const float* __restrict__ g,
         for (int w = 0; w < 4; ++w) {
           const int gf = free0 + fq + w;
           v[u][w] = (gc < contract_max && gf < free_max)
-                        ? g[(long)gc * ld + gf]
+                        ? ld1(&g[(long)gc * ld + gf])
                         : 0.f;
         }
       }
@@ -228,7 +240,8 @@ struct BWaveTile {
           acc[mi][ni] = Vec16<T16>::mfma(a[mi], b[ni], acc[mi][ni]);
     }
   }
-  __device__ __forceinline__ void store(float* __restrict__ c, long ldc,
+  template <typename TO>
+  __device__ __forceinline__ void store(TO* __restrict__ c, long ldc,
                                         int row0, int col0, int m_max,
                                         int n_max, const float* bias, int relu,
                                         int lane) {
@@ -246,7 +259,7 @@ struct BWaveTile {
             float v = acc[mi][ni][r];
             if (bias) v += bias[col];
             if (relu) v = fmaxf(v, 0.f);
-            c[(long)row * ldc + col] = v;
+            c[(long)row * ldc + col] = (TO)v;
           }
         }
   }
@@ -256,12 +269,13 @@ struct BWaveTile {
 // kernels — same three layouts as the f32 suite
 // ---------------------------------------------------------------------------
 
-template <int BM, int BN, int BK = 64, typename T16 = __bf16>
+template <int BM, int BN, int BK = 64, typename T16 = __bf16,
+          typename TA = float, typename TO = float>
 __launch_bounds__(BGEMM_THREADS)
-__global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
+__global__ void gemm_bf16_nt_kernel(const TA* __restrict__ a,
                                     const float* __restrict__ b,
                                     const float* __restrict__ bias,
-                                    float* __restrict__ c, int m, int n, int k,
+                                    TO* __restrict__ c, int m, int n, int k,
                                     int relu) {
   constexpr int LDW = BK + BGEMM_PAD;
   constexpr int FM = BM / 32, FN = BN / 32;
@@ -293,12 +307,13 @@ __global__ void gemm_bf16_nt_kernel(const float* __restrict__ a,
   wt.store(c, n, m0 + wm, n0 + wn, m, n, bias, relu, lane);
 }
 
-template <int BM, int BN, int BK = 64, typename T16 = __bf16>
+template <int BM, int BN, int BK = 64, typename T16 = __bf16,
+          typename TA = float, typename TO = float>
 __launch_bounds__(BGEMM_THREADS)
-__global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
+__global__ void gemm_bf16_nn_kernel(const TA* __restrict__ a,
                                     const float* __restrict__ b,
                                     const float* __restrict__ bias,
-                                    float* __restrict__ c, int m, int n,
+                                    TO* __restrict__ c, int m, int n,
                                     int k2, int relu) {
   constexpr int LDW = BK + BGEMM_PAD;
   constexpr int FM = BM / 32, FN = BN / 32;
@@ -330,9 +345,10 @@ __global__ void gemm_bf16_nn_kernel(const float* __restrict__ a,
   wt.store(c, k2, m0 + wm, n0 + wn, m, k2, bias, relu, lane);
 }
 
-template <int BM, int BN, int BK = 64, typename T16 = __bf16>
+template <int BM, int BN, int BK = 64, typename T16 = __bf16,
+          typename TA = float>
 __launch_bounds__(BGEMM_THREADS)
-__global__ void gemm_bf16_tn_kernel(const float* __restrict__ a,
+__global__ void gemm_bf16_tn_kernel(const TA* __restrict__ a,
                                     const float* __restrict__ b,
                                     float* __restrict__ c,
                                     float* __restrict__ dbias, int m, int n,
@@ -544,6 +560,69 @@ void launch_gemm_fp16_tn(const float* a, const float* b, float* c,
                                                               slices);
   else
    hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64, 64, _Float16>)
+        , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+}
+
+
+// ---------------------------------------------------------------------------
+// mixed-dtype launchers for the bf16-resident-activation mode: the QKVS
+// forward writes its C as bf16; the backward GEMMs read the bf16 gradient
+// directly (no conversion pass — bf16 LDS copy).
+// ---------------------------------------------------------------------------
+
+void launch_gemm_bf16_nt_o16(const float* a, const float* b, const float* bias,
+                             void* c_v, int m, int n, int k, hipStream_t s) {
+  __bf16* c = (__bf16*)c_v;
+  if (m >= 512 && n >= 128) {
+    const int grid = ((m + 127) / 128) * ((n + 127) / 128);
+   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128, 32, __bf16, float, __bf16>)
+        , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k, 0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((n + 63) / 64);
+   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<64, 64, 64, __bf16, float, __bf16>)
+        , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k, 0);
+  }
+}
+
+void launch_gemm_bf16_nn_a16(const void* a_v, const float* b, float* c, int m,
+                             int n, int k2, hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  if (m >= 512 && k2 >= 128) {
+    const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
+   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128, 32, __bf16, __bf16, float>)
+        , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, nullptr, c, m, n, k2,
+                                                    0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((k2 + 63) / 64);
+   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<64, 64, 64, __bf16, __bf16, float>)
+        , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, nullptr, c, m, n, k2,
+                                                    0);
+  }
+}
+
+void launch_gemm_bf16_tn_a16(const void* a_v, const float* b, float* c,
+                             float* dbias, int m, int n, int k2,
+                             hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  const bool big = (n >= 128 && k2 >= 128);
+  const int bm = big ? 128 : 64;
+  const int bn = big ? 128 : 64;
+  const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
+  int slices = 1;
+  while (tiles * slices < 512 && slices < 64 && (long)slices * 64 * 4 < m)
+    slices *= 2;
+  if (slices > 1)
+    HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
+  if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
+  if (big)
+   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128, 32, __bf16, __bf16>)
+        , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+  else
+   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64, 64, __bf16, __bf16>)
         , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
                                                               m, n, k2,
                                                               slices);

@@ -65,13 +65,19 @@ class TransformerConv(nn.Module):
     def forward_fused(self, x, edge_attr, ifc_weight, rpc_weight, csr):
         """HIP fast path: one [N,K]x[4H,K]^T GEMM for q/k/v/skip and
         L2-resident per-vocab P tables instead of the [E,2H] edge-embed
-        stream (exact refactoring by linearity of lin_edge)."""
+        stream (exact refactoring by linearity of lin_edge).  In bf16/fp16
+        precision with H%256==0 the qkvs tensor is kept bf16-resident
+        through the attention kernels."""
         h = self.out_channels
         w4 = torch.cat([self.lin_query.weight, self.lin_key.weight,
                         self.lin_value.weight, self.lin_skip.weight], dim=0)
         b4 = torch.cat([self.lin_query.bias, self.lin_key.bias,
                         self.lin_value.bias, self.lin_skip.bias], dim=0)
-        qkvs = ops.linear(x, w4, b4)
+        if (ops.gemm_precision() != "fp32" and h % 256 == 0
+                and ops.act16_enabled()):
+            qkvs = ops.linear16(x, w4, b4)
+        else:
+            qkvs = ops.linear(x, w4, b4)
         we = self.lin_edge.weight  # [H, 2H]
         pifc = ops.linear(ifc_weight, we[:, :h].contiguous(), None)
         prpc = ops.linear(rpc_weight, we[:, h:].contiguous(), None)

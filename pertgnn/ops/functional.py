@@ -71,6 +71,48 @@ def edge_attention(q, k, v, e, skip, edge_index, num_nodes, csr=None):
     return ref.edge_attention(q, k, v, e, edge_index, num_nodes, skip)
 
 
+class _Linear16Fn(torch.autograd.Function):
+    """bf16-activation-mode linear: fp32 x/w in, bf16 out (the QKVS tensor
+    stays bf16 through the attention kernels — halves the edge-gather traffic
+    and the largest C-writes; operands were already bf16-rounded inside the
+    matrix cores, so numerics match the plain bf16-GEMM mode)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        m = ext()
+        bb = b if b is not None else torch.empty(0, dtype=x.dtype, device=x.device)
+        y = m.linear_fwd_bf16_o16(x, w, bb)
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, g):
+        x, w = ctx.saved_tensors
+        m = ext()
+        g = g.contiguous()
+        dx = m.linear_dgrad16(g, w)
+        dw, db = m.linear_wgrad16(g, x, ctx.has_bias)
+        return dx, dw, (db if ctx.has_bias else None)
+
+
+def linear16(x, w, b=None):
+    return _Linear16Fn.apply(x, w, b)
+
+
+_ACT16 = None
+
+
+def act16_enabled() -> bool:
+    """bf16-resident qkvs activations (bf16/fp16 precision modes, H%256==0).
+    Disable with PERTGNN_NO_ACT16=1."""
+    global _ACT16
+    if _ACT16 is None:
+        import os
+        _ACT16 = os.environ.get("PERTGNN_NO_ACT16", "0") != "1"
+    return _ACT16
+
+
 # ---------------------------------------------------------------------------
 # fused-layout edge attention: qkvs [N,4H] + per-vocab P tables (fast path)
 # ---------------------------------------------------------------------------
@@ -505,6 +547,8 @@ def linear(x, w, b=None):
 __all__ = [
     "set_gemm_precision",
     "gemm_precision",
+    "linear16",
+    "act16_enabled",
     "edge_attention",
     "edge_attention_fused",
     "embedding",
