@@ -73,7 +73,7 @@ class TransformerConv(nn.Module):
                         self.lin_value.weight, self.lin_skip.weight], dim=0)
         b4 = torch.cat([self.lin_query.bias, self.lin_key.bias,
                         self.lin_value.bias, self.lin_skip.bias], dim=0)
-        if (ops.gemm_precision() != "fp32" and h % 256 == 0
+        if (ops.gemm_precision() == "bf16" and h % 256 == 0
                 and ops.act16_enabled()):
             qkvs = ops.linear16(x, w4, b4)
         else:

@@ -493,3 +493,48 @@ def test_fused_attention_empty_and_isolated_rows():
     torch.cuda.synchronize()
     assert alpha.numel() == 0
     assert torch.allclose(out, qkvs[:, 3 * h:], atol=1e-6)
+
+
+def test_model_act16_close_to_fp32():
+    """bf16-resident-activation mode (H=256, the flagship shape) stays close
+    to the fp32 path end-to-end, forward AND gradients."""
+    require_ext()
+    import copy
+    from pertgnn.models import SAGEDeterministic
+    import bench as bench_mod
+    from pertgnn.ops.functional import set_gemm_precision
+
+    torch.manual_seed(1)
+    batches, stats = bench_mod.build_synthetic_batches(1, 16, seed=2, device=DEV)
+    b = batches[0]
+    model = SAGEDeterministic(9, [stats["cat_max"] + 1], stats["entry_max"],
+                              stats["ifc_max"], stats["rpc_max"], 256, 3, 0.0).to(DEV)
+    model.train()
+
+    def run(m):
+        gp, _ = m(b.x, b.cat_X, b.edge_index, b.edge_attr,
+                  b.pattern_num_nodes, b.rt_probs, b.entry_id, b.batch,
+                  csr=b.csr, num_graphs=b.num_graphs)
+        loss = F.quantile_loss(b.y, gp.flatten(), 0.5)
+        loss.backward()
+        grads = {n: p.grad.clone() for n, p in m.named_parameters() if p.grad is not None}
+        m.zero_grad()
+        return gp.detach(), float(loss.detach()), grads
+
+    m32 = copy.deepcopy(model)
+    try:
+        set_gemm_precision("fp32")
+        gp32, l32, g32 = run(m32)
+        set_gemm_precision("bf16")
+        gp16, l16, g16 = run(model)
+    finally:
+        set_gemm_precision("fp32")
+    rel = (gp16 - gp32).abs().max() / gp32.abs().max().clamp_min(1e-6)
+    assert rel < 0.05, rel
+    assert abs(l16 - l32) / max(abs(l32), 1e-6) < 0.05
+    # gradient direction must agree (cosine) for the big weights
+    for n in ("convs.1.lin_query.weight", "convs.0.lin_edge.weight",
+              "cat_embedding.0.weight"):
+        a, c = g16[n].flatten(), g32[n].flatten()
+        cos = torch.dot(a, c) / (a.norm() * c.norm()).clamp_min(1e-12)
+        assert cos > 0.99, (n, float(cos))
