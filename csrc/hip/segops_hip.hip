@@ -396,6 +396,55 @@ __global__ void vocab_scatter_dual_kernel(const float* __restrict__ g,
     if (acc1[t] != 0.f) atomicAdd(&dt1[t], acc1[t]);
 }
 
+// wave-private variant: each wave accumulates into its OWN (rows0+rows1,h)
+// LDS table (no atomics at all — lanes own distinct columns), then the four
+// tables are combined in LDS and flushed once.  Gated on 4x table fitting in
+// 160KB LDS.
+__global__ void vocab_scatter_dual_priv_kernel(const float* __restrict__ g,
+                                               const long* __restrict__ ea,
+                                               int astride,
+                                               float* __restrict__ dt0,
+                                               float* __restrict__ dt1, long n,
+                                               int rows0, int rows1, int h) {
+  typedef __attribute__((ext_vector_type(4))) float f4;
+  extern __shared__ float acc[];  // [4][(rows0+rows1)*h]
+  const long vh = (long)(rows0 + rows1) * h;
+  for (long t = threadIdx.x; t < 4 * vh; t += blockDim.x) acc[t] = 0.f;
+  __syncthreads();
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  float* my = acc + (long)wid * vh;
+  float* my1 = my + (long)rows0 * h;
+  const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * rows_per_block;
+  const long r1 = min(n, r0 + rows_per_block);
+  const int nq = h / (4 * PERTGNN_WAVE);
+  for (long r = r0 + wid; r < r1; r += WAVES_PER_BLOCK) {
+    const long v0 = ea[r * astride];
+    const long v1 = ea[r * astride + 1];
+    for (int q = 0; q < nq; ++q) {
+      const int c = (q * PERTGNN_WAVE + lane) * 4;
+      const f4 xv = *reinterpret_cast<const f4*>(&g[r * h + c]);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        my[v0 * h + c + u] += xv[u];
+        my1[v1 * h + c + u] += xv[u];
+      }
+    }
+  }
+  __syncthreads();
+  // combine tables 1..3 into 0 (threads partition elements), then flush
+  for (long t = threadIdx.x; t < vh; t += blockDim.x)
+    acc[t] += acc[vh + t] + acc[2 * vh + t] + acc[3 * vh + t];
+  __syncthreads();
+  for (long t = threadIdx.x; t < (long)rows0 * h; t += blockDim.x)
+    if (acc[t] != 0.f) atomicAdd(&dt0[t], acc[t]);
+  for (long t = threadIdx.x; t < (long)rows1 * h; t += blockDim.x) {
+    const float v = acc[(long)rows0 * h + t];
+    if (v != 0.f) atomicAdd(&dt1[t], v);
+  }
+}
+
 void launch_vocab_scatter_dual(const float* g, const long* ea, int astride,
                                float* dt0, float* dt1, long n, int rows0,
                                int rows1, int h, hipStream_t s) {
@@ -403,15 +452,26 @@ void launch_vocab_scatter_dual(const float* g, const long* ea, int astride,
   HIP_CHECK(hipMemsetAsync(dt1, 0, (long)rows1 * h * sizeof(float), s));
   if (n == 0) return;
   const size_t lds = (size_t)(rows0 + rows1) * h * sizeof(float);
-  if (lds > 64 * 1024) {
-    HIP_CHECK(hipFuncSetAttribute((const void*)vocab_scatter_dual_kernel,
+  const bool priv = 4 * lds <= 160 * 1024;
+  const void* fn = priv ? (const void*)vocab_scatter_dual_priv_kernel
+                        : (const void*)vocab_scatter_dual_kernel;
+  const size_t use_lds = priv ? 4 * lds : lds;
+  if (use_lds > 64 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(fn,
                                   hipFuncAttributeMaxDynamicSharedMemorySize,
-                                  (int)lds));
+                                  (int)use_lds));
   }
   const int blocks = (int)min((long)256, (n + 63) / 64);
- hipLaunchKernelGGL(( vocab_scatter_dual_kernel), dim3(dim3(blocks)),
-                              dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)), lds, s, 
-      g, ea, astride, dt0, dt1, n, rows0, rows1, h);
+  if (priv)
+   hipLaunchKernelGGL(( vocab_scatter_dual_priv_kernel), dim3(dim3(blocks)),
+                                     dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)),
+                                     use_lds, s, g, ea, astride, dt0, dt1, n,
+                                                   rows0, rows1, h);
+  else
+   hipLaunchKernelGGL(( vocab_scatter_dual_kernel), dim3(dim3(blocks)),
+                                dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)), use_lds,
+                                s, g, ea, astride, dt0, dt1, n, rows0,
+                                     rows1, h);
 }
 
 void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
