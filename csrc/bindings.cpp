@@ -101,15 +101,17 @@ void launch_edge_attn_fused_bwd(const float*, const float*, const float*,
                                 const float*, const long*, int, const float*,
                                 const int*, const int*, const int*,
                                 const int*, float*, float*, float*, float*,
-                                int, int, long, hipStream_t);
+                                float*, int, int, long, hipStream_t);
 void launch_edge_attn_fused_fwd16(const void*, const float*, const float*,
                                   const long*, int, const int*, const int*,
                                   float*, float*, int, int, hipStream_t);
 void launch_edge_attn_fused_bwd16(const float*, const void*, const float*,
                                   const float*, const long*, int,
                                   const float*, const int*, const int*,
-                                  const int*, const int*, void*, float*,
-                                  float*, float*, int, int, long, hipStream_t);
+                                  const int*, const int*, void*, void*, void*,
+                                  void*, float*, int, int, long, hipStream_t);
+void launch_vocab_scatter_dual16(const void*, const long*, int, float*,
+                                 float*, long, int, int, int, hipStream_t);
 void launch_gemm_bf16_nt_o16(const float*, const float*, const float*, void*,
                              int, int, int, hipStream_t);
 void launch_gemm_bf16_nn_a16(const void*, const float*, float*, int, int, int,
@@ -583,26 +585,31 @@ std::vector<torch::Tensor> edge_attn_fused_bwd(
   const long ne = ea.size(0);
   auto fopt = qkvs.options().dtype(torch::kFloat32);
   auto dqkvs = torch::empty_like(qkvs);
-  auto de = torch::empty({ne, h}, fopt);
-  auto dek = torch::empty({ne, h}, fopt);
-  auto dev = torch::empty({ne, h}, fopt);
-  if (qkvs.scalar_type() == torch::kBFloat16) {
+  const bool b16 = qkvs.scalar_type() == torch::kBFloat16;
+  auto eopt = b16 ? qkvs.options() : fopt;  // edge scratch matches qkvs dtype
+  auto de = torch::empty({ne, h}, eopt);
+  auto dek = torch::empty({ne, h}, eopt);
+  auto dev = torch::empty({ne, h}, eopt);
+  if (b16) {
+    auto dal = torch::empty({ne}, fopt);
     launch_edge_attn_fused_bwd16(
         g.data_ptr<float>(), qkvs.data_ptr(),
         pifc.data_ptr<float>(), prpc.data_ptr<float>(), ea.data_ptr<long>(),
         (int)ea.size(1), alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
         csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
-        csc_eid.data_ptr<int>(), dqkvs.data_ptr(),
-        de.data_ptr<float>(), dek.data_ptr<float>(), dev.data_ptr<float>(), n,
-        h, ne, cur_stream());
+        csc_eid.data_ptr<int>(), dqkvs.data_ptr(), de.data_ptr(),
+        dek.data_ptr(), dev.data_ptr(), dal.data_ptr<float>(), n, h, ne,
+        cur_stream());
   } else {
+    auto dal = torch::empty({ne}, fopt);
     launch_edge_attn_fused_bwd(
         g.data_ptr<float>(), qkvs.data_ptr<float>(), pifc.data_ptr<float>(),
         prpc.data_ptr<float>(), ea.data_ptr<long>(), (int)ea.size(1),
         alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
         csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
         csc_eid.data_ptr<int>(), dqkvs.data_ptr<float>(), de.data_ptr<float>(),
-        dek.data_ptr<float>(), dev.data_ptr<float>(), n, h, ne, cur_stream());
+        dek.data_ptr<float>(), dev.data_ptr<float>(), dal.data_ptr<float>(),
+        n, h, ne, cur_stream());
   }
   return {dqkvs, de};
 }
@@ -740,12 +747,19 @@ std::vector<torch::Tensor> vocab_scatter_dual(torch::Tensor g,
   const int h = g.size(1);
   TORCH_CHECK(h % 256 == 0, "dual scatter needs h % 256 == 0");
   TORCH_CHECK((size_t)(rows0 + rows1) * h * 4 <= 160 * 1024, "tables too large");
-  auto dt0 = torch::empty({rows0, h}, g.options());
-  auto dt1 = torch::empty({rows1, h}, g.options());
-  launch_vocab_scatter_dual(g.data_ptr<float>(), ea.data_ptr<long>(),
-                            (int)ea.size(1), dt0.data_ptr<float>(),
-                            dt1.data_ptr<float>(), g.size(0), (int)rows0,
-                            (int)rows1, h, cur_stream());
+  auto fopt = g.options().dtype(torch::kFloat32);
+  auto dt0 = torch::empty({rows0, h}, fopt);
+  auto dt1 = torch::empty({rows1, h}, fopt);
+  if (g.scalar_type() == torch::kBFloat16)
+    launch_vocab_scatter_dual16(g.data_ptr(), ea.data_ptr<long>(),
+                                (int)ea.size(1), dt0.data_ptr<float>(),
+                                dt1.data_ptr<float>(), g.size(0), (int)rows0,
+                                (int)rows1, h, cur_stream());
+  else
+    launch_vocab_scatter_dual(g.data_ptr<float>(), ea.data_ptr<long>(),
+                              (int)ea.size(1), dt0.data_ptr<float>(),
+                              dt1.data_ptr<float>(), g.size(0), (int)rows0,
+                              (int)rows1, h, cur_stream());
   return {dt0, dt1};
 }
 

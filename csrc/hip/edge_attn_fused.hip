@@ -177,14 +177,15 @@ __global__ void edge_attn_fused_fwd_kernel(
     alpha[p] = __expf(alpha[p] - m) * inv_s;
 }
 
-template <int VPT, bool VEC, typename QT = float>
+template <int VPT, bool VEC, typename QT = float, typename ET = float>
 __global__ void edge_attn_fused_bwd_row_kernel(
     const float* __restrict__ g, const QT* __restrict__ qkvs,
     const float* __restrict__ pifc, const float* __restrict__ prpc,
     const long* __restrict__ ea, int astride, const float* __restrict__ alpha,
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
-    QT* __restrict__ dqkvs, float* __restrict__ dek,
-    float* __restrict__ dev, int n, int h, float scale) {
+    QT* __restrict__ dqkvs, ET* __restrict__ dek,
+    ET* __restrict__ dev, float* __restrict__ dal, int n, int h,
+    float scale) {
   using S = Slice<VPT, VEC>;
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
@@ -212,13 +213,13 @@ __global__ void edge_attn_fused_bwd_row_kernel(
     for (int j = 0; j < VPT; ++j) part += gr[j] * (ve[j] + ec[j]);
     const float dalpha = wave_reduce_sum(part);
     sdot += alpha[p] * dalpha;
-    if (lane == 0) dev[(long)p * h] = dalpha;
+    if (lane == 0) dal[p] = dalpha;
   }
   for (int p = beg; p < end; ++p) {
     const long src = csr_src[p];
     const long a0 = ea[(long)p * astride];
     const long a1 = ea[(long)p * astride + 1];
-    const float dalpha_l0 = (lane == 0) ? dev[(long)p * h] : 0.f;
+    const float dalpha_l0 = (lane == 0) ? dal[p] : 0.f;
     const float dalpha = __shfl(dalpha_l0, 0, PERTGNN_WAVE);
     const float a = alpha[p];
     const float dl = a * (dalpha - sdot) * scale;
@@ -238,11 +239,11 @@ __global__ void edge_attn_fused_bwd_row_kernel(
   S::store(&dqkvs[row * ld + 3 * h], lane, h, gr);     // dskip = g
 }
 
-template <int VPT, bool VEC, typename QT = float>
+template <int VPT, bool VEC, typename QT = float, typename ET = float>
 __global__ void edge_attn_fused_bwd_col_kernel(
-    const float* __restrict__ dek, const float* __restrict__ dev,
+    const ET* __restrict__ dek, const ET* __restrict__ dev,
     const int* __restrict__ col_ptr, const int* __restrict__ csc_eid,
-    QT* __restrict__ dqkvs, float* __restrict__ de, int n, int h) {
+    QT* __restrict__ dqkvs, ET* __restrict__ de, int n, int h) {
   using S = Slice<VPT, VEC>;
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
@@ -316,8 +317,8 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
                                 const float* alpha, const int* row_ptr,
                                 const int* csr_src, const int* col_ptr,
                                 const int* csc_eid, float* dqkvs, float* de,
-                                float* dek, float* dev, int n, int h,
-                                long num_edges, hipStream_t stream) {
+                                float* dek, float* dev, float* dal, int n,
+                                int h, long num_edges, hipStream_t stream) {
   if (n == 0) return;
   const float scale = 1.f / std::sqrt((float)h);
   const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
@@ -330,13 +331,13 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
     if (vec && (V % 4 == 0)) {                                                 \
       edge_attn_fused_bwd_row_kernel<V, true><<<grid, block, 0, stream>>>(     \
           g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,    \
-          dek, dev, n, h, scale);                                              \
+          dek, dev, dal, n, h, scale);                                         \
       edge_attn_fused_bwd_col_kernel<V, true><<<grid, block, 0, stream>>>(     \
           dek, dev, col_ptr, csc_eid, dqkvs, de, n, h);                        \
     } else {                                                                   \
       edge_attn_fused_bwd_row_kernel<V, false><<<grid, block, 0, stream>>>(    \
           g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,    \
-          dek, dev, n, h, scale);                                              \
+          dek, dev, dal, n, h, scale);                                         \
       edge_attn_fused_bwd_col_kernel<V, false><<<grid, block, 0, stream>>>(    \
           dek, dev, col_ptr, csc_eid, dqkvs, de, n, h);                        \
     }                                                                          \
@@ -380,30 +381,34 @@ void launch_edge_attn_fused_bwd16(const float* g, const void* qkvs_v,
                                   const long* ea, int astride,
                                   const float* alpha, const int* row_ptr,
                                   const int* csr_src, const int* col_ptr,
-                                  const int* csc_eid, void* dqkvs_v, float* de,
-                                  float* dek, float* dev, int n, int h,
-                                  long num_edges, hipStream_t stream) {
+                                  const int* csc_eid, void* dqkvs_v,
+                                  void* de_v, void* dek_v, void* dev_v,
+                                  float* dal, int n, int h, long num_edges,
+                                  hipStream_t stream) {
   const __bf16* qkvs = (const __bf16*)qkvs_v;
   __bf16* dqkvs = (__bf16*)dqkvs_v;
+  __bf16* de = (__bf16*)de_v;
+  __bf16* dek = (__bf16*)dek_v;
+  __bf16* dev = (__bf16*)dev_v;
   if (n == 0) return;
   const float scale = 1.f / std::sqrt((float)h);
   const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
   const int vpt = h / PERTGNN_WAVE;
   if (vpt == 4) {
-    edge_attn_fused_bwd_row_kernel<4, true, __bf16>
+    edge_attn_fused_bwd_row_kernel<4, true, __bf16, __bf16>
         <<<grid, block, 0, stream>>>(g, qkvs, pifc, prpc, ea, astride, alpha,
-                                     row_ptr, csr_src, dqkvs, dek, dev, n, h,
-                                     scale);
-    edge_attn_fused_bwd_col_kernel<4, true, __bf16>
+                                     row_ptr, csr_src, dqkvs, dek, dev, dal,
+                                     n, h, scale);
+    edge_attn_fused_bwd_col_kernel<4, true, __bf16, __bf16>
         <<<grid, block, 0, stream>>>(dek, dev, col_ptr, csc_eid, dqkvs, de, n,
                                      h);
   } else if (vpt == 8) {
-    edge_attn_fused_bwd_row_kernel<8, true, __bf16>
+    edge_attn_fused_bwd_row_kernel<8, true, __bf16, __bf16>
         <<<grid, block, 0, stream>>>(g, qkvs, pifc, prpc, ea, astride, alpha,
-                                     row_ptr, csr_src, dqkvs, dek, dev, n, h,
-                                     scale);
-    edge_attn_fused_bwd_col_kernel<8, true, __bf16>
+                                     row_ptr, csr_src, dqkvs, dek, dev, dal,
+                                     n, h, scale);
+    edge_attn_fused_bwd_col_kernel<8, true, __bf16, __bf16>
         <<<grid, block, 0, stream>>>(dek, dev, col_ptr, csc_eid, dqkvs, de, n,
                                      h);
   } else {

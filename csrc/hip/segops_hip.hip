@@ -358,13 +358,24 @@ __global__ void vocab_scatter_vec_kernel(const float* __restrict__ g,
 // dual-table variant: accumulates BOTH vocab tables (interface + rpctype)
 // in one pass over g — the two dP reductions of the attention backward share
 // the same 56MB de stream.
-__global__ void vocab_scatter_dual_kernel(const float* __restrict__ g,
+typedef __attribute__((ext_vector_type(4))) float vs_f4;
+typedef __attribute__((ext_vector_type(4))) __bf16 vs_b4;
+__device__ __forceinline__ void vs_ld4(const float* p, float (&v)[4]) {
+  *reinterpret_cast<vs_f4*>(v) = *reinterpret_cast<const vs_f4*>(p);
+}
+__device__ __forceinline__ void vs_ld4(const __bf16* p, float (&v)[4]) {
+  const vs_b4 b = *reinterpret_cast<const vs_b4*>(p);
+#pragma unroll
+  for (int u = 0; u < 4; ++u) v[u] = (float)b[u];
+}
+
+template <typename GT>
+__global__ void vocab_scatter_dual_kernel(const GT* __restrict__ g,
                                           const long* __restrict__ ea,
                                           int astride,
                                           float* __restrict__ dt0,
                                           float* __restrict__ dt1, long n,
                                           int rows0, int rows1, int h) {
-  typedef __attribute__((ext_vector_type(4))) float f4;
   extern __shared__ float acc[];  // [(rows0+rows1)*h]
   float* acc1 = acc + (long)rows0 * h;
   const long vh = (long)(rows0 + rows1) * h;
@@ -381,7 +392,8 @@ __global__ void vocab_scatter_dual_kernel(const float* __restrict__ g,
     const long v1 = ea[r * astride + 1];
     for (int q = 0; q < nq; ++q) {
       const int c = (q * PERTGNN_WAVE + lane) * 4;
-      const f4 xv = *reinterpret_cast<const f4*>(&g[r * h + c]);
+      float xv[4];
+      vs_ld4(&g[r * h + c], xv);
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         atomicAdd(&acc[v0 * h + c + u], xv[u]);
@@ -400,13 +412,13 @@ __global__ void vocab_scatter_dual_kernel(const float* __restrict__ g,
 // LDS table (no atomics at all — lanes own distinct columns), then the four
 // tables are combined in LDS and flushed once.  Gated on 4x table fitting in
 // 160KB LDS.
-__global__ void vocab_scatter_dual_priv_kernel(const float* __restrict__ g,
+template <typename GT>
+__global__ void vocab_scatter_dual_priv_kernel(const GT* __restrict__ g,
                                                const long* __restrict__ ea,
                                                int astride,
                                                float* __restrict__ dt0,
                                                float* __restrict__ dt1, long n,
                                                int rows0, int rows1, int h) {
-  typedef __attribute__((ext_vector_type(4))) float f4;
   extern __shared__ float acc[];  // [4][(rows0+rows1)*h]
   const long vh = (long)(rows0 + rows1) * h;
   for (long t = threadIdx.x; t < 4 * vh; t += blockDim.x) acc[t] = 0.f;
@@ -424,7 +436,8 @@ __global__ void vocab_scatter_dual_priv_kernel(const float* __restrict__ g,
     const long v1 = ea[r * astride + 1];
     for (int q = 0; q < nq; ++q) {
       const int c = (q * PERTGNN_WAVE + lane) * 4;
-      const f4 xv = *reinterpret_cast<const f4*>(&g[r * h + c]);
+      float xv[4];
+      vs_ld4(&g[r * h + c], xv);
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         my[v0 * h + c + u] += xv[u];
@@ -445,33 +458,37 @@ __global__ void vocab_scatter_dual_priv_kernel(const float* __restrict__ g,
   }
 }
 
-void launch_vocab_scatter_dual(const float* g, const long* ea, int astride,
-                               float* dt0, float* dt1, long n, int rows0,
-                               int rows1, int h, hipStream_t s) {
+template <typename GT>
+static void vocab_dual_impl(const GT* g, const long* ea, int astride,
+                            float* dt0, float* dt1, long n, int rows0,
+                            int rows1, int h, hipStream_t s) {
   HIP_CHECK(hipMemsetAsync(dt0, 0, (long)rows0 * h * sizeof(float), s));
   HIP_CHECK(hipMemsetAsync(dt1, 0, (long)rows1 * h * sizeof(float), s));
   if (n == 0) return;
   const size_t lds = (size_t)(rows0 + rows1) * h * sizeof(float);
-  const bool priv = 4 * lds <= 160 * 1024;
-  const void* fn = priv ? (const void*)vocab_scatter_dual_priv_kernel
-                        : (const void*)vocab_scatter_dual_kernel;
-  const size_t use_lds = priv ? 4 * lds : lds;
-  if (use_lds > 64 * 1024) {
-    HIP_CHECK(hipFuncSetAttribute(fn,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize,
-                                  (int)use_lds));
+  if (lds > 64 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        (const void*)vocab_scatter_dual_kernel<GT>,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
   const int blocks = (int)min((long)256, (n + 63) / 64);
-  if (priv)
-   hipLaunchKernelGGL(( vocab_scatter_dual_priv_kernel), dim3(dim3(blocks)),
-                                     dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)),
-                                     use_lds, s, g, ea, astride, dt0, dt1, n,
-                                                   rows0, rows1, h);
-  else
-   hipLaunchKernelGGL(( vocab_scatter_dual_kernel), dim3(dim3(blocks)),
-                                dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)), use_lds,
-                                s, g, ea, astride, dt0, dt1, n, rows0,
-                                     rows1, h);
+ hipLaunchKernelGGL(( vocab_scatter_dual_kernel<GT>), dim3(dim3(blocks)),
+                                  dim3(dim3(WAVES_PER_BLOCK * PERTGNN_WAVE)), lds,
+                                  s, g, ea, astride, dt0, dt1, n, rows0,
+                                       rows1, h);
+}
+
+void launch_vocab_scatter_dual(const float* g, const long* ea, int astride,
+                               float* dt0, float* dt1, long n, int rows0,
+                               int rows1, int h, hipStream_t s) {
+  vocab_dual_impl<float>(g, ea, astride, dt0, dt1, n, rows0, rows1, h, s);
+}
+
+void launch_vocab_scatter_dual16(const void* g, const long* ea, int astride,
+                                 float* dt0, float* dt1, long n, int rows0,
+                                 int rows1, int h, hipStream_t s) {
+  vocab_dual_impl<__bf16>((const __bf16*)g, ea, astride, dt0, dt1, n, rows0,
+                          rows1, h, s);
 }
 
 void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,

@@ -141,8 +141,9 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
             if h % 256 == 0 and (pifc.shape[0] + prpc.shape[0]) * h * 4 <= 160 * 1024:
                 dpifc, dprpc = m.vocab_scatter_dual(de, edge_attr, pifc.shape[0], prpc.shape[0])
             else:
-                dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], h, 0)
-                dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], h, 0)
+                de32 = de.float() if de.dtype != torch.float32 else de
+                dpifc = _table_grad(m, de32, edge_attr[:, 0], pifc.shape[0], h, 0)
+                dprpc = _table_grad(m, de32, edge_attr[:, 1], prpc.shape[0], h, 0)
             return dqkvs, dpifc, dprpc, None, None, None, None, None
         cur = torch.cuda.current_stream()
         side = _side_stream()
@@ -150,8 +151,9 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
         ev.record(cur)
         side.wait_event(ev)
         with torch.cuda.stream(side):
-            dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
-            dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
+            de32 = de.float() if de.dtype != torch.float32 else de
+            dpifc = _table_grad(m, de32, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
+            dprpc = _table_grad(m, de32, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
         ev2 = torch.cuda.Event()
         ev2.record(side)
         cur.wait_event(ev2)
