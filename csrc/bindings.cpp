@@ -31,16 +31,10 @@ void launch_seg_pool_bwd(const float*, const float*, const float*, const long*,
                          float*, long, int, hipStream_t);
 void launch_embed_node_fwd(const float*, const long*, const float*, float*,
                            long, int, int, hipStream_t);
-void launch_embed_scatter_add(const float*, const long*, float*, long, int,
-                              int, hipStream_t);
 void launch_embed_edge_fwd(const long*, const float*, const float*, float*,
                            long, int, int, hipStream_t);
-void launch_embed_edge_bwd(const float*, const long*, float*, float*, long,
-                           int, int, hipStream_t);
 void launch_gather_rows(const long*, const float*, float*, long, int,
                         hipStream_t);
-void launch_scatter_add_rows(const float*, const long*, float*, long, int,
-                             hipStream_t);
 void launch_bn_fwd(const float*, const float*, const float*, float*, float*,
                    float*, float*, float*, float*, long, int, float, float,
                    bool, bool, hipStream_t);
@@ -223,18 +217,6 @@ torch::Tensor embed_node_fwd(torch::Tensor x_raw, torch::Tensor idx,
   return out;
 }
 
-torch::Tensor embed_scatter_add(torch::Tensor g, torch::Tensor idx, int64_t f,
-                                int64_t rows) {
-  CHECK_IN(g); CHECK_IN(idx);
-  const long n = g.size(0);
-  const int h = g.size(1) - f;
-  auto dtable = torch::zeros({rows, h}, g.options());
-  launch_embed_scatter_add(g.data_ptr<float>(), idx.data_ptr<long>(),
-                           dtable.data_ptr<float>(), n, (int)f, h,
-                           cur_stream());
-  return dtable;
-}
-
 torch::Tensor embed_edge_fwd(torch::Tensor attr, torch::Tensor ifc,
                              torch::Tensor rpc) {
   CHECK_IN(attr); CHECK_IN(ifc); CHECK_IN(rpc);
@@ -248,20 +230,6 @@ torch::Tensor embed_edge_fwd(torch::Tensor attr, torch::Tensor ifc,
   return out;
 }
 
-std::vector<torch::Tensor> embed_edge_bwd(torch::Tensor g, torch::Tensor attr,
-                                          int64_t rows_ifc, int64_t rows_rpc) {
-  CHECK_IN(g); CHECK_IN(attr);
-  const long e = attr.size(0);
-  const int h = g.size(1) / 2;
-  const int astride = attr.size(1);
-  auto difc = torch::zeros({rows_ifc, h}, g.options());
-  auto drpc = torch::zeros({rows_rpc, h}, g.options());
-  launch_embed_edge_bwd(g.data_ptr<float>(), attr.data_ptr<long>(),
-                        difc.data_ptr<float>(), drpc.data_ptr<float>(), e, h,
-                        astride, cur_stream());
-  return {difc, drpc};
-}
-
 torch::Tensor gather_rows(torch::Tensor idx, torch::Tensor table) {
   CHECK_IN(idx); CHECK_IN(table);
   const long n = idx.size(0);
@@ -270,17 +238,6 @@ torch::Tensor gather_rows(torch::Tensor idx, torch::Tensor table) {
   launch_gather_rows(idx.data_ptr<long>(), table.data_ptr<float>(),
                      out.data_ptr<float>(), n, h, cur_stream());
   return out;
-}
-
-torch::Tensor scatter_add_rows(torch::Tensor g, torch::Tensor idx,
-                               int64_t rows) {
-  CHECK_IN(g); CHECK_IN(idx);
-  const long n = g.size(0);
-  const int h = g.size(1);
-  auto dtable = torch::zeros({rows, h}, g.options());
-  launch_scatter_add_rows(g.data_ptr<float>(), idx.data_ptr<long>(),
-                          dtable.data_ptr<float>(), n, h, cur_stream());
-  return dtable;
 }
 
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
@@ -844,11 +801,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("seg_pool_fwd", &seg_pool_fwd);
   mod.def("seg_pool_bwd", &seg_pool_bwd);
   mod.def("embed_node_fwd", &embed_node_fwd);
-  mod.def("embed_scatter_add", &embed_scatter_add);
   mod.def("embed_edge_fwd", &embed_edge_fwd);
-  mod.def("embed_edge_bwd", &embed_edge_bwd);
   mod.def("gather_rows", &gather_rows);
-  mod.def("scatter_add_rows", &scatter_add_rows);
   mod.def("bn_relu_fwd", &bn_relu_fwd);
   mod.def("bn_relu_bwd", &bn_relu_bwd);
   mod.def("quantile_loss_fwd", &quantile_loss_fwd);

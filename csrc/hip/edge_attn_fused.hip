@@ -272,13 +272,6 @@ __global__ void edge_attn_fused_bwd_col_kernel(
   S::store(&dqkvs[(long)row * ld + 2 * h], lane, h, va);
 }
 
-__global__ void add2_kernel(const float* __restrict__ a,
-                            const float* __restrict__ b,
-                            float* __restrict__ out, long numel) {
-  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long t = i; t < numel; t += stride) out[t] = a[t] + b[t];
-}
 
 // ---------------------------------------------------------------------------
 

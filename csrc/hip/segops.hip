@@ -141,21 +141,6 @@ __global__ void embed_node_fwd_kernel(const float* __restrict__ x_raw,
   }
 }
 
-__global__ void embed_scatter_add_kernel(const float* __restrict__ g,
-                                         const long* __restrict__ idx,
-                                         float* __restrict__ dtable, long n,
-                                         int f, int h) {
-  const int w = f + h;
-  const long numel = n * h;
-  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long t = i0; t < numel; t += stride) {
-    const long row = t / h;
-    const int c = (int)(t - row * h);
-    atomicAdd(&dtable[idx[row] * h + c], g[row * w + f + c]);
-  }
-}
-
 __global__ void embed_edge_fwd_kernel(const long* __restrict__ attr,
                                       const float* __restrict__ ifc,
                                       const float* __restrict__ rpc,
@@ -170,25 +155,6 @@ __global__ void embed_edge_fwd_kernel(const long* __restrict__ attr,
     const int c = (int)(t - row * w);
     out[t] = (c < h) ? ifc[attr[row * astride] * h + c]
                      : rpc[attr[row * astride + 1] * h + (c - h)];
-  }
-}
-
-__global__ void embed_edge_bwd_kernel(const float* __restrict__ g,
-                                      const long* __restrict__ attr,
-                                      float* __restrict__ difc,
-                                      float* __restrict__ drpc, long e, int h,
-                                      int astride) {
-  const int w = 2 * h;
-  const long numel = e * w;
-  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long t = i0; t < numel; t += stride) {
-    const long row = t / w;
-    const int c = (int)(t - row * w);
-    if (c < h)
-      atomicAdd(&difc[attr[row * astride] * h + c], g[t]);
-    else
-      atomicAdd(&drpc[attr[row * astride + 1] * h + (c - h)], g[t]);
   }
 }
 
@@ -555,12 +521,6 @@ void launch_embed_node_fwd(const float* x_raw, const long* idx,
   embed_node_fwd_kernel<<<grid_for(n * (f + h)), 256, 0, s>>>(x_raw, idx, table,
                                                               out, n, f, h);
 }
-void launch_embed_scatter_add(const float* g, const long* idx, float* dtable,
-                              long n, int f, int h, hipStream_t s) {
-  if (n == 0) return;
-  embed_scatter_add_kernel<<<grid_for(n * h), 256, 0, s>>>(g, idx, dtable, n,
-                                                           f, h);
-}
 void launch_embed_edge_fwd(const long* attr, const float* ifc,
                            const float* rpc, float* out, long e, int h,
                            int astride, hipStream_t s) {
@@ -568,22 +528,10 @@ void launch_embed_edge_fwd(const long* attr, const float* ifc,
   embed_edge_fwd_kernel<<<grid_for(e * 2 * h), 256, 0, s>>>(attr, ifc, rpc, out,
                                                             e, h, astride);
 }
-void launch_embed_edge_bwd(const float* g, const long* attr, float* difc,
-                           float* drpc, long e, int h, int astride,
-                           hipStream_t s) {
-  if (e == 0) return;
-  embed_edge_bwd_kernel<<<grid_for(e * 2 * h), 256, 0, s>>>(g, attr, difc, drpc,
-                                                            e, h, astride);
-}
 void launch_gather_rows(const long* idx, const float* table, float* out,
                         long n, int h, hipStream_t s) {
   if (n == 0) return;
   gather_rows_kernel<<<grid_for(n * h), 256, 0, s>>>(idx, table, out, n, h);
-}
-void launch_scatter_add_rows(const float* g, const long* idx, float* dtable,
-                             long n, int h, hipStream_t s) {
-  if (n == 0) return;
-  scatter_add_rows_kernel<<<grid_for(n * h), 256, 0, s>>>(g, idx, dtable, n, h);
 }
 
 // ---------------------------------------------------------------------------
