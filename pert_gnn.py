@@ -93,7 +93,13 @@ def main(argv=None):
     torch.manual_seed(args.seed + comm.rank)
 
     if torch.cuda.is_available():
-        device = comm.device
+        # reference semantics: single process honors --device (pert_gnn.py:36);
+        # under torchrun each rank uses its LOCAL_RANK device
+        if comm.distributed:
+            device = comm.device
+        else:
+            device = torch.device(f"cuda:{args.device % torch.cuda.device_count()}")
+            torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
 
