@@ -84,8 +84,8 @@ void launch_gemm_fp16_tn(const float*, const float*, float*, float*, int, int,
 void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   float*, long, int, int, int, int, int,
                                   hipStream_t);
-void launch_vocab_scatter(const float*, const long*, long, float*, float*,
-                          long, int, int, int, int, hipStream_t);
+void launch_vocab_scatter(const float*, const long*, long, float*, long, int,
+                          int, int, int, hipStream_t);
 void launch_vocab_scatter_dual(const float*, const long*, int, float*, float*,
                                long, int, int, int, hipStream_t);
 void launch_edge_attn_fused_fwd(const float*, const float*, const float*,
@@ -106,7 +106,6 @@ void launch_edge_attn_fused_bwd16(const float*, const void*, const float*,
                                   void*, float*, int, int, long, hipStream_t);
 void launch_vocab_scatter_dual16(const void*, const long*, int, float*,
                                  float*, long, int, int, int, hipStream_t);
-// (same signature for fp32 variant; partial is the [nblocks, vh] workspace)
 void launch_gemm_bf16_nt_o16(const float*, const float*, const float*, void*,
                              int, int, int, hipStream_t);
 void launch_gemm_bf16_nn_a16(const void*, const float*, float*, int, int, int,
@@ -574,23 +573,17 @@ std::vector<torch::Tensor> edge_attn_fused_bwd(
 
 // dtable[v] += sum of g[r, col_off:col_off+h] over rows r with idx[r]==v.
 // idx may be a strided column view of an [N,A] attr tensor.
-static int64_t vocab_blocks(int64_t n) {
-  return std::min<int64_t>(128, (n + 63) / 64);
-}
-
 torch::Tensor vocab_scatter(torch::Tensor g, torch::Tensor idx, int64_t rows,
                             int64_t h, int64_t col_off) {
   CHECK_IN(g);
   TORCH_CHECK(idx.is_cuda() && idx.dim() == 1, "idx must be 1-D CUDA");
   const size_t lds = (size_t)rows * h * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024, "vocab too large for LDS accumulator");
-  auto dtable = torch::empty({rows, h}, g.options().dtype(torch::kFloat32));
-  auto partial = torch::empty({vocab_blocks(g.size(0)) * rows * h},
-                              g.options().dtype(torch::kFloat32));
+  auto dtable = torch::empty({rows, h}, g.options());
   launch_vocab_scatter(g.data_ptr<float>(), idx.data_ptr<long>(),
-                       idx.stride(0), partial.data_ptr<float>(),
-                       dtable.data_ptr<float>(), g.size(0), (int)rows, (int)h,
-                       (int)g.size(1), (int)col_off, cur_stream());
+                       idx.stride(0), dtable.data_ptr<float>(), g.size(0),
+                       (int)rows, (int)h, (int)g.size(1), (int)col_off,
+                       cur_stream());
   return dtable;
 }
 
@@ -712,21 +705,19 @@ std::vector<torch::Tensor> vocab_scatter_dual(torch::Tensor g,
   TORCH_CHECK(h % 256 == 0, "dual scatter needs h % 256 == 0");
   TORCH_CHECK((size_t)(rows0 + rows1) * h * 4 <= 160 * 1024, "tables too large");
   auto fopt = g.options().dtype(torch::kFloat32);
-  const int64_t vh = (rows0 + rows1) * h;
-  auto dt01 = torch::empty({vh}, fopt);
-  auto partial = torch::empty({vocab_blocks(g.size(0)) * vh}, fopt);
+  auto dt0 = torch::empty({rows0, h}, fopt);
+  auto dt1 = torch::empty({rows1, h}, fopt);
   if (g.scalar_type() == torch::kBFloat16)
     launch_vocab_scatter_dual16(g.data_ptr(), ea.data_ptr<long>(),
-                                (int)ea.size(1), partial.data_ptr<float>(),
-                                dt01.data_ptr<float>(), g.size(0), (int)rows0,
+                                (int)ea.size(1), dt0.data_ptr<float>(),
+                                dt1.data_ptr<float>(), g.size(0), (int)rows0,
                                 (int)rows1, h, cur_stream());
   else
     launch_vocab_scatter_dual(g.data_ptr<float>(), ea.data_ptr<long>(),
-                              (int)ea.size(1), partial.data_ptr<float>(),
-                              dt01.data_ptr<float>(), g.size(0), (int)rows0,
+                              (int)ea.size(1), dt0.data_ptr<float>(),
+                              dt1.data_ptr<float>(), g.size(0), (int)rows0,
                               (int)rows1, h, cur_stream());
-  return {dt01.narrow(0, 0, rows0 * h).view({rows0, h}),
-          dt01.narrow(0, rows0 * h, rows1 * h).view({rows1, h})};
+  return {dt0, dt1};
 }
 
 // bf16-activation-mode linear: fp32 x/w in, bf16 C out; backward from bf16 g

@@ -244,13 +244,12 @@ void launch_embed_grouped_scatter(const float* g, const int* order,
 
 // Single-pass vocab accumulator for small vocabularies: blocks stream
 // CONTIGUOUS gradient rows (coalesced, no sort/gather), accumulate into an
-// LDS-resident [V,h] table (LDS ds_add_f32 — contention only within the CU),
-// then write the block's PARTIAL slab with plain stores; a second kernel
-// folds the slabs in block order (deterministic — the global-atomic flush
-// this replaces was the dominant cost AND non-deterministic).
+// LDS-resident [V,h] table (LDS atomics — contention only within the CU),
+// then one global atomicAdd per (v,c) per block.  Weight-gradient class:
+// fp32 reduction order varies run to run (same class as split-K wgrad).
 __global__ void vocab_scatter_kernel(const float* __restrict__ g,
                                      const long* __restrict__ idx,
-                                     long idx_stride, float* __restrict__ partial,
+                                     long idx_stride, float* __restrict__ dtable,
                                      long n, int rows, int h, int gstride,
                                      int col_off) {
   extern __shared__ float acc[];  // [rows*h]
@@ -269,7 +268,7 @@ __global__ void vocab_scatter_kernel(const float* __restrict__ g,
   }
   __syncthreads();
   for (long t = threadIdx.x; t < vh; t += blockDim.x)
-    partial[(long)blockIdx.x * vh + t] = acc[t];
+    if (acc[t] != 0.f) atomicAdd(&dtable[t], acc[t]);
 }
 
 // vectorized variant (h % 256 == 0, aligned): lane owns 4 contiguous columns
@@ -278,7 +277,7 @@ __global__ void vocab_scatter_kernel(const float* __restrict__ g,
 __global__ void vocab_scatter_vec_kernel(const float* __restrict__ g,
                                          const long* __restrict__ idx,
                                          long idx_stride,
-                                         float* __restrict__ partial, long n,
+                                         float* __restrict__ dtable, long n,
                                          int rows, int h, int gstride,
                                          int col_off) {
   typedef __attribute__((ext_vector_type(4))) float f4;
@@ -318,25 +317,12 @@ __global__ void vocab_scatter_vec_kernel(const float* __restrict__ g,
   }
   __syncthreads();
   for (long t = threadIdx.x; t < vh; t += blockDim.x)
-    partial[(long)blockIdx.x * vh + t] = acc[t];
+    if (acc[t] != 0.f) atomicAdd(&dtable[t], acc[t]);
 }
 
 // dual-table variant: accumulates BOTH vocab tables (interface + rpctype)
 // in one pass over g — the two dP reductions of the attention backward share
 // the same 56MB de stream.
-static int vocab_nblocks(long n);
-
-// partial-table reduce: dtable[c] = sum_b partial[b][c] (ordered, no atomics)
-__global__ void vocab_reduce_kernel(const float* __restrict__ partial,
-                                    float* __restrict__ dtable, long vh,
-                                    int nblocks) {
-  const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= vh) return;
-  float acc = 0.f;
-  for (int b = 0; b < nblocks; ++b) acc += partial[(long)b * vh + c];
-  dtable[c] = acc;
-}
-
 typedef __attribute__((ext_vector_type(4))) float vs_f4;
 typedef __attribute__((ext_vector_type(4))) __bf16 vs_b4;
 __device__ __forceinline__ void vs_ld4(const float* p, float (&v)[4]) {
@@ -352,7 +338,8 @@ template <typename GT>
 __global__ void vocab_scatter_dual_kernel(const GT* __restrict__ g,
                                           const long* __restrict__ ea,
                                           int astride,
-                                          float* __restrict__ partial, long n,
+                                          float* __restrict__ dt0,
+                                          float* __restrict__ dt1, long n,
                                           int rows0, int rows1, int h) {
   extern __shared__ float acc[];  // [(rows0+rows1)*h]
   float* acc1 = acc + (long)rows0 * h;
@@ -380,58 +367,101 @@ __global__ void vocab_scatter_dual_kernel(const GT* __restrict__ g,
     }
   }
   __syncthreads();
-  for (long t = threadIdx.x; t < vh; t += blockDim.x)
-    partial[(long)blockIdx.x * vh + t] = acc[t];
+  for (long t = threadIdx.x; t < (long)rows0 * h; t += blockDim.x)
+    if (acc[t] != 0.f) atomicAdd(&dt0[t], acc[t]);
+  for (long t = threadIdx.x; t < (long)rows1 * h; t += blockDim.x)
+    if (acc1[t] != 0.f) atomicAdd(&dt1[t], acc1[t]);
 }
 
-static int vocab_nblocks(long n) { return (int)min((long)128, (n + 63) / 64); }
+// wave-private variant: each wave accumulates into its OWN (rows0+rows1,h)
+// LDS table (no atomics at all — lanes own distinct columns), then the four
+// tables are combined in LDS and flushed once.  Gated on 4x table fitting in
+// 160KB LDS.
+template <typename GT>
+__global__ void vocab_scatter_dual_priv_kernel(const GT* __restrict__ g,
+                                               const long* __restrict__ ea,
+                                               int astride,
+                                               float* __restrict__ dt0,
+                                               float* __restrict__ dt1, long n,
+                                               int rows0, int rows1, int h) {
+  extern __shared__ float acc[];  // [4][(rows0+rows1)*h]
+  const long vh = (long)(rows0 + rows1) * h;
+  for (long t = threadIdx.x; t < 4 * vh; t += blockDim.x) acc[t] = 0.f;
+  __syncthreads();
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  float* my = acc + (long)wid * vh;
+  float* my1 = my + (long)rows0 * h;
+  const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * rows_per_block;
+  const long r1 = min(n, r0 + rows_per_block);
+  const int nq = h / (4 * PERTGNN_WAVE);
+  for (long r = r0 + wid; r < r1; r += WAVES_PER_BLOCK) {
+    const long v0 = ea[r * astride];
+    const long v1 = ea[r * astride + 1];
+    for (int q = 0; q < nq; ++q) {
+      const int c = (q * PERTGNN_WAVE + lane) * 4;
+      float xv[4];
+      vs_ld4(&g[r * h + c], xv);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        my[v0 * h + c + u] += xv[u];
+        my1[v1 * h + c + u] += xv[u];
+      }
+    }
+  }
+  __syncthreads();
+  // combine tables 1..3 into 0 (threads partition elements), then flush
+  for (long t = threadIdx.x; t < vh; t += blockDim.x)
+    acc[t] += acc[vh + t] + acc[2 * vh + t] + acc[3 * vh + t];
+  __syncthreads();
+  for (long t = threadIdx.x; t < (long)rows0 * h; t += blockDim.x)
+    if (acc[t] != 0.f) atomicAdd(&dt0[t], acc[t]);
+  for (long t = threadIdx.x; t < (long)rows1 * h; t += blockDim.x) {
+    const float v = acc[(long)rows0 * h + t];
+    if (v != 0.f) atomicAdd(&dt1[t], v);
+  }
+}
 
 template <typename GT>
 static void vocab_dual_impl(const GT* g, const long* ea, int astride,
-                            float* partial, float* dt01, long n, int rows0,
+                            float* dt0, float* dt1, long n, int rows0,
                             int rows1, int h, hipStream_t s) {
-  const long vh = (long)(rows0 + rows1) * h;
-  if (n == 0) {
-    HIP_CHECK(hipMemsetAsync(dt01, 0, vh * sizeof(float), s));
-    return;
-  }
-  const size_t lds = (size_t)vh * sizeof(float);
+  HIP_CHECK(hipMemsetAsync(dt0, 0, (long)rows0 * h * sizeof(float), s));
+  HIP_CHECK(hipMemsetAsync(dt1, 0, (long)rows1 * h * sizeof(float), s));
+  if (n == 0) return;
+  const size_t lds = (size_t)(rows0 + rows1) * h * sizeof(float);
   if (lds > 64 * 1024) {
     HIP_CHECK(hipFuncSetAttribute(
         (const void*)vocab_scatter_dual_kernel<GT>,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
-  const int blocks = vocab_nblocks(n);
+  const int blocks = (int)min((long)256, (n + 63) / 64);
   vocab_scatter_dual_kernel<GT><<<dim3(blocks),
                                   dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds,
-                                  s>>>(g, ea, astride, partial, n, rows0,
+                                  s>>>(g, ea, astride, dt0, dt1, n, rows0,
                                        rows1, h);
-  vocab_reduce_kernel<<<dim3(ceil_div(vh, 256)), dim3(256), 0, s>>>(
-      partial, dt01, vh, blocks);
 }
 
 void launch_vocab_scatter_dual(const float* g, const long* ea, int astride,
-                               float* partial, float* dt01, long n, int rows0,
+                               float* dt0, float* dt1, long n, int rows0,
                                int rows1, int h, hipStream_t s) {
-  vocab_dual_impl<float>(g, ea, astride, partial, dt01, n, rows0, rows1, h, s);
+  vocab_dual_impl<float>(g, ea, astride, dt0, dt1, n, rows0, rows1, h, s);
 }
 
 void launch_vocab_scatter_dual16(const void* g, const long* ea, int astride,
-                                 float* partial, float* dt01, long n,
-                                 int rows0, int rows1, int h, hipStream_t s) {
-  vocab_dual_impl<__bf16>((const __bf16*)g, ea, astride, partial, dt01, n,
-                          rows0, rows1, h, s);
+                                 float* dt0, float* dt1, long n, int rows0,
+                                 int rows1, int h, hipStream_t s) {
+  vocab_dual_impl<__bf16>((const __bf16*)g, ea, astride, dt0, dt1, n, rows0,
+                          rows1, h, s);
 }
 
 void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
-                          float* partial, float* dtable, long n, int rows,
-                          int h, int gstride, int col_off, hipStream_t s) {
-  const long vh = (long)rows * h;
-  if (n == 0) {
-    HIP_CHECK(hipMemsetAsync(dtable, 0, vh * sizeof(float), s));
-    return;
-  }
-  const size_t lds = (size_t)vh * sizeof(float);
+                          float* dtable, long n, int rows, int h, int gstride,
+                          int col_off, hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(dtable, 0, (long)rows * h * sizeof(float), s));
+  if (n == 0) return;
+  const size_t lds = (size_t)rows * h * sizeof(float);
   const bool vec = (h % (4 * PERTGNN_WAVE) == 0) && ((gstride & 3) == 0) &&
                    ((col_off & 3) == 0);
   const void* fn = vec ? (const void*)vocab_scatter_vec_kernel
@@ -440,17 +470,15 @@ void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
     HIP_CHECK(hipFuncSetAttribute(
         fn, hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
-  const int blocks = vocab_nblocks(n);
+  const int blocks = (int)min((long)256, (n + 63) / 64);
   if (vec)
     vocab_scatter_vec_kernel<<<dim3(blocks),
                                dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds, s>>>(
-        g, idx, idx_stride, partial, n, rows, h, gstride, col_off);
+        g, idx, idx_stride, dtable, n, rows, h, gstride, col_off);
   else
     vocab_scatter_kernel<<<dim3(blocks), dim3(WAVES_PER_BLOCK * PERTGNN_WAVE),
-                           lds, s>>>(g, idx, idx_stride, partial, n, rows, h,
+                           lds, s>>>(g, idx, idx_stride, dtable, n, rows, h,
                                      gstride, col_off);
-  vocab_reduce_kernel<<<dim3(ceil_div(vh, 256)), dim3(256), 0, s>>>(
-      partial, dtable, vh, blocks);
 }
 
 // entry embedding gather: out[b] = table[idx[b]] — plain gather (fwd) +
