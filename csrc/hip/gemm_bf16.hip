@@ -435,10 +435,24 @@ static int gemm_bk() {
   return bk;
 }
 
+static bool gemm_n64() {  // narrow-BN experiment: 128x64 tiles, 4 waves/SIMD
+  static bool v = [] {
+    const char* e = getenv("PERTGNN_GEMM_N64");
+    return e && atoi(e) == 1;
+  }();
+  return v;
+}
+
 void launch_gemm_bf16_nt(const float* a, const float* b, const float* bias,
                          float* c, int m, int n, int k, bool relu,
                          hipStream_t s) {
   if (m >= 512 && n >= 128) {
+    if (gemm_n64()) {
+      const int grid = ((m + 127) / 128) * ((n + 63) / 64);
+      gemm_bf16_nt_kernel<128, 64, 32><<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(
+          a, b, bias, c, m, n, k, relu ? 1 : 0);
+      return;
+    }
     const int grid = ((m + 127) / 128) * ((n + 127) / 128);
     if (gemm_bk() == 32)
       gemm_bf16_nt_kernel<128, 128, 32><<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(
@@ -457,6 +471,12 @@ void launch_gemm_bf16_nn(const float* a, const float* b, const float* bias,
                          float* c, int m, int n, int k2, bool relu,
                          hipStream_t s) {
   if (m >= 512 && k2 >= 128) {
+    if (gemm_n64()) {
+      const int grid = ((m + 127) / 128) * ((k2 + 63) / 64);
+      gemm_bf16_nn_kernel<128, 64, 32><<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(
+          a, b, bias, c, m, n, k2, relu ? 1 : 0);
+      return;
+    }
     const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
     if (gemm_bk() == 32)
       gemm_bf16_nn_kernel<128, 128, 32><<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(
