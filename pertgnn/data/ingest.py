@@ -145,7 +145,12 @@ def get_df(data_root: str, processed_dir: str, min_occurence: int = 100):
     return df, resource_df
 
 
-def run_ingest(data_root: str = "data", processed_dir: str = "processed", min_occurence: int = 100, verbose: bool = True):
+def _build_runtime_graphs(trace_df):
+    return build_span_graph(trace_df), build_pert_graph(trace_df)
+
+
+def run_ingest(data_root: str = "data", processed_dir: str = "processed",
+               min_occurence: int = 100, verbose: bool = True, n_jobs: int = -1):
     """Full pipeline (reference preprocess.py main, :269-381)."""
     os.makedirs(processed_dir, exist_ok=True)
     df, resource_df = get_df(data_root, processed_dir, min_occurence=min_occurence)
@@ -161,32 +166,55 @@ def run_ingest(data_root: str = "data", processed_dir: str = "processed", min_oc
     tr2runtime = dict(zip(corpus.index, pd.factorize(corpus)[0]))
     tr2delay = df["rt"].abs().groupby(df["traceid"]).max().to_dict()
 
+    # The reference walks every trace of every entry in Python
+    # (preprocess.py:295-369, the "10+ hours" hot loop, README.md:12).  The
+    # same observable result is computed vectorized: per-trace records and
+    # runtime occurrence counts from grouped frames, and each runtime
+    # pattern's graphs built exactly ONCE from its first trace.
+    per_trace = (
+        df.groupby("traceid").agg(entryid=("entryid", "first")).reset_index()
+    )
+    per_trace["runtime_id"] = per_trace["traceid"].map(tr2runtime)
+
     tr2data = {}
+    for traceid, entry, runtime_id in per_trace.itertuples(index=False):
+        tr2data[traceid] = {
+            "entry_id": int(entry),
+            "runtime_id": int(runtime_id),
+            "timestamp": int(tr2ts[traceid]),
+            "y": torch.tensor(tr2delay[traceid]),
+        }
+
     entry2runtimes: dict = {}
+    counts = per_trace.groupby(["entryid", "runtime_id"]).size()
+    for (entry, runtime_id), cnt in counts.items():
+        entry2runtimes.setdefault(int(entry), {})[int(runtime_id)] = int(cnt)
+
+    # one representative trace per runtime pattern
+    rep_trace = per_trace.groupby("runtime_id")["traceid"].first()
+    occurrences = per_trace.groupby("runtime_id").size().to_dict()
+    by_trace = dict(tuple(df[df["traceid"].isin(set(rep_trace.values))].groupby("traceid")))
+
+    # per-pattern graph building is embarrassingly parallel
+    items = list(rep_trace.items())
+    if n_jobs != 1 and len(items) > 64:
+        from joblib import Parallel, delayed
+
+        built = Parallel(n_jobs=n_jobs, batch_size=64)(
+            delayed(_build_runtime_graphs)(by_trace[tid]) for _, tid in items
+        )
+    else:
+        built = [_build_runtime_graphs(by_trace[tid]) for _, tid in items]
     runtime2span: dict = {}
     runtime2pert: dict = {}
-
-    for entry, entry_group in df.groupby("entryid"):
-        for traceid, trace_df in entry_group.groupby("traceid"):
-            runtime_id = int(tr2runtime[traceid])
-            tr2data[traceid] = {
-                "entry_id": int(entry),
-                "runtime_id": runtime_id,
-                "timestamp": int(tr2ts[traceid]),
-                "y": torch.tensor(tr2delay[traceid]),
-            }
-            entry2runtimes.setdefault(int(entry), {})
-            entry2runtimes[int(entry)][runtime_id] = entry2runtimes[int(entry)].get(runtime_id, 0) + 1
-            if runtime_id not in runtime2span:
-                runtime2span[runtime_id] = build_span_graph(trace_df)
-            else:
-                runtime2span[runtime_id]["occurences"] += 1
-            if runtime_id not in runtime2pert:
-                runtime2pert[runtime_id] = build_pert_graph(trace_df)
-            else:
-                runtime2pert[runtime_id]["occurences"] += 1
-        if verbose:
-            print(f"entry {entry}: {entry_group['traceid'].nunique()} traces")
+    for (runtime_id, _tid), (g_span, g_pert) in zip(items, built):
+        g_span["occurences"] = int(occurrences[runtime_id])
+        g_pert["occurences"] = int(occurrences[runtime_id])
+        runtime2span[int(runtime_id)] = g_span
+        runtime2pert[int(runtime_id)] = g_pert
+    if verbose:
+        print(f"{len(tr2data)} traces, {len(runtime2pert)} runtime patterns, "
+              f"{len(entry2runtimes)} entries")
 
     # normalize occurrence counts to probabilities (preprocess.py:372-375)
     for entry, rt_counts in entry2runtimes.items():

@@ -112,7 +112,10 @@ def generate_traces(cfg: SyntheticConfig):
             # child calls: strictly later timestamps, strictly smaller rt
             slot_ms = [pat.entry_ms]
             for ci, (parent_slot, child, iface, rpct) in enumerate(pat.calls):
-                ts = t0 + 1 + ci + int(rng.integers(0, 5))
+                # deterministic per-pattern call order: traces of one runtime
+                # pattern produce identical um_dm_interface sequences (the
+                # Alibaba data repeats patterns; only rt varies per trace)
+                ts = t0 + 1 + 3 * ci
                 rt = max(1, int(total_rt * float(rng.uniform(0.05, 0.5))))
                 rt = min(rt, total_rt - 1)
                 call_rows.append(

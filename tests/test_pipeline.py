@@ -129,8 +129,9 @@ def test_end_to_end_training_epoch(synthetic_workspace):
     opt = torch.optim.Adam(model.parameters(), lr=3e-3)
     loader = BatchLoader(train, batch_size=16, shuffle=True, seed=0)
     model.train()
-    first = last = None
+    epoch_means = []
     for epoch in range(3):
+        losses = []
         for b in loader:
             opt.zero_grad()
             gp, lp = model(b.x, b.cat_X, b.edge_index, b.edge_attr,
@@ -139,10 +140,9 @@ def test_end_to_end_training_epoch(synthetic_workspace):
             loss = F.quantile_loss(b.y, gp.flatten(), 0.5)
             loss.backward()
             opt.step()
-            if first is None:
-                first = float(loss)
-            last = float(loss)
-    assert last < first
+            losses.append(float(loss.detach()))
+        epoch_means.append(sum(losses) / len(losses))
+    assert epoch_means[-1] < epoch_means[0]
 
 
 def test_native_collator_matches_python(synthetic_workspace):
