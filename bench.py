@@ -25,7 +25,7 @@ import torch
 def build_synthetic_batches(n_batches: int, graphs_per_batch: int, seed: int,
                             device) -> tuple[list, dict]:
     """Trace-scale synthetic PERT graphs -> pre-collated GPU-resident batches."""
-    from pertgnn.data.collate import collate
+    from pertgnn.data.collate import collate_native as collate
     from pertgnn.data.graphs import build_pert_graph
     from pertgnn.data.dataset import TraceSample
     from pertgnn.data.synthetic import SyntheticConfig, generate_traces
