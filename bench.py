@@ -100,8 +100,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--layers", type=int, default=8)
     ap.add_argument("--hidden", type=int, default=256)
-    ap.add_argument("--batch-size", type=int, default=256, help="graphs per GPU per step")
-    ap.add_argument("--n-batches", type=int, default=8, help="distinct resident batches to cycle")
+    ap.add_argument("--batch-size", type=int, default=1024, help="graphs per GPU per step")
+    ap.add_argument("--n-batches", type=int, default=4, help="distinct resident batches to cycle")
     ap.add_argument("--lr", type=float, default=3e-4)
     ap.add_argument("--tau", type=float, default=0.5)
     ap.add_argument("--precision", choices=["fp32", "bf16", "fp16"], default="bf16",
