@@ -334,7 +334,11 @@ __device__ __forceinline__ void vs_ld4(const __bf16* p, float (&v)[4]) {
   for (int u = 0; u < 4; ++u) v[u] = (float)b[u];
 }
 
-template <typename GT>
+// U rows are prefetched per loop iteration (independent index + gradient
+// loads issue together before the LDS-atomic block) — the plain one-row loop
+// exposes two dependent memory latencies per row and, at 1 block/CU, ran at
+// ~200 GB/s; prefetch + 3 blocks/CU recovers the latency-bound gap.
+template <typename GT, int U>
 __global__ void vocab_scatter_dual_kernel(const GT* __restrict__ g,
                                           const long* __restrict__ ea,
                                           int astride,
@@ -352,7 +356,30 @@ __global__ void vocab_scatter_dual_kernel(const GT* __restrict__ g,
   const long r0 = (long)blockIdx.x * rows_per_block;
   const long r1 = min(n, r0 + rows_per_block);
   const int nq = h / (4 * PERTGNN_WAVE);
-  for (long r = r0 + wid; r < r1; r += WAVES_PER_BLOCK) {
+  const long step = WAVES_PER_BLOCK;
+  long r = r0 + wid;
+  for (; r + (U - 1) * step < r1; r += U * step) {
+    long v0[U], v1[U];
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+      v0[u] = ea[(r + u * step) * astride];
+      v1[u] = ea[(r + u * step) * astride + 1];
+    }
+    for (int q = 0; q < nq; ++q) {
+      const int c = (q * PERTGNN_WAVE + lane) * 4;
+      float xv[U][4];
+#pragma unroll
+      for (int u = 0; u < U; ++u) vs_ld4(&g[(r + u * step) * h + c], xv[u]);
+#pragma unroll
+      for (int u = 0; u < U; ++u)
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          atomicAdd(&acc[v0[u] * h + c + k], xv[u][k]);
+          atomicAdd(&acc1[v1[u] * h + c + k], xv[u][k]);
+        }
+    }
+  }
+  for (; r < r1; r += step) {
     const long v0 = ea[r * astride];
     const long v1 = ea[r * astride + 1];
     for (int q = 0; q < nq; ++q) {
@@ -373,53 +400,78 @@ __global__ void vocab_scatter_dual_kernel(const GT* __restrict__ g,
     if (acc1[t] != 0.f) atomicAdd(&dt1[t], acc1[t]);
 }
 
-// wave-private variant: each wave accumulates into its OWN (rows0+rows1,h)
-// LDS table (no atomics at all — lanes own distinct columns), then the four
-// tables are combined in LDS and flushed once.  Gated on 4x table fitting in
-// 160KB LDS.
-template <typename GT>
+// Wave-private variant — the production path.  Measured decomposition
+// (benchmarks/vocab_micro.hip, E=216k/h=256/V=47): the LDS-atomic kernel is
+// 100% ds_add_f32-bound at ~80 cycles/instruction (538 us; loads alone are
+// 23 us; bank layout, unrolling, occupancy and index skew all change
+// nothing).  Plain read+add+write into per-WAVE private tables removes the
+// atomics entirely: 75 us for the same reduction (7.2x).  Each wave owns a
+// private [rows0+rows1, HH] LDS table for one HH-wide column slice
+// (grid.y = h/HH selects the slice); tables merge in LDS, one global-atomic
+// flush per block.
+template <typename GT, int U, int HH>
 __global__ void vocab_scatter_dual_priv_kernel(const GT* __restrict__ g,
                                                const long* __restrict__ ea,
                                                int astride,
                                                float* __restrict__ dt0,
                                                float* __restrict__ dt1, long n,
                                                int rows0, int rows1, int h) {
-  extern __shared__ float acc[];  // [4][(rows0+rows1)*h]
-  const long vh = (long)(rows0 + rows1) * h;
-  for (long t = threadIdx.x; t < 4 * vh; t += blockDim.x) acc[t] = 0.f;
+  constexpr int CPL = HH / PERTGNN_WAVE;  // columns per lane
+  extern __shared__ float acc[];  // [WAVES_PER_BLOCK][(rows0+rows1)*HH]
+  const long vwh = (long)(rows0 + rows1) * HH;
+  for (long t = threadIdx.x; t < WAVES_PER_BLOCK * vwh; t += blockDim.x)
+    acc[t] = 0.f;
   __syncthreads();
   const int wid = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
-  float* my = acc + (long)wid * vh;
-  float* my1 = my + (long)rows0 * h;
+  float* my0 = acc + (long)wid * vwh;
+  float* my1 = my0 + (long)rows0 * HH;
+  const int c0 = blockIdx.y * HH;
   const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
   const long r0 = (long)blockIdx.x * rows_per_block;
   const long r1 = min(n, r0 + rows_per_block);
-  const int nq = h / (4 * PERTGNN_WAVE);
-  for (long r = r0 + wid; r < r1; r += WAVES_PER_BLOCK) {
+  const long step = WAVES_PER_BLOCK;
+  long r = r0 + wid;
+  for (; r + (U - 1) * step < r1; r += U * step) {
+    long v0[U], v1[U];
+    float xv[U][CPL];
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+      v0[u] = ea[(r + u * step) * astride];
+      v1[u] = ea[(r + u * step) * astride + 1];
+#pragma unroll
+      for (int k = 0; k < CPL; ++k)
+        xv[u][k] = (float)g[(r + u * step) * h + c0 + lane * CPL + k];
+    }
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+#pragma unroll
+      for (int k = 0; k < CPL; ++k) my0[v0[u] * HH + lane * CPL + k] += xv[u][k];
+#pragma unroll
+      for (int k = 0; k < CPL; ++k) my1[v1[u] * HH + lane * CPL + k] += xv[u][k];
+    }
+  }
+  for (; r < r1; r += step) {
     const long v0 = ea[r * astride];
     const long v1 = ea[r * astride + 1];
-    for (int q = 0; q < nq; ++q) {
-      const int c = (q * PERTGNN_WAVE + lane) * 4;
-      float xv[4];
-      vs_ld4(&g[r * h + c], xv);
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        my[v0 * h + c + u] += xv[u];
-        my1[v1 * h + c + u] += xv[u];
-      }
+    for (int k = 0; k < CPL; ++k) {
+      const float x = (float)g[r * h + c0 + lane * CPL + k];
+      my0[v0 * HH + lane * CPL + k] += x;
+      my1[v1 * HH + lane * CPL + k] += x;
     }
   }
   __syncthreads();
-  // combine tables 1..3 into 0 (threads partition elements), then flush
-  for (long t = threadIdx.x; t < vh; t += blockDim.x)
-    acc[t] += acc[vh + t] + acc[2 * vh + t] + acc[3 * vh + t];
+  for (long t = threadIdx.x; t < vwh; t += blockDim.x)
+    acc[t] += acc[vwh + t] + acc[2 * vwh + t] + acc[3 * vwh + t];
   __syncthreads();
-  for (long t = threadIdx.x; t < (long)rows0 * h; t += blockDim.x)
-    if (acc[t] != 0.f) atomicAdd(&dt0[t], acc[t]);
-  for (long t = threadIdx.x; t < (long)rows1 * h; t += blockDim.x) {
-    const float v = acc[(long)rows0 * h + t];
-    if (v != 0.f) atomicAdd(&dt1[t], v);
+  for (long t = threadIdx.x; t < (long)rows0 * HH; t += blockDim.x) {
+    const float v = acc[t];
+    if (v != 0.f) atomicAdd(&dt0[(t / HH) * h + c0 + t % HH], v);
+  }
+  for (long t = threadIdx.x; t < (long)rows1 * HH; t += blockDim.x) {
+    const float v = acc[(long)rows0 * HH + t];
+    if (v != 0.f) atomicAdd(&dt1[(t / HH) * h + c0 + t % HH], v);
   }
 }
 
@@ -430,17 +482,42 @@ static void vocab_dual_impl(const GT* g, const long* ea, int astride,
   HIP_CHECK(hipMemsetAsync(dt0, 0, (long)rows0 * h * sizeof(float), s));
   HIP_CHECK(hipMemsetAsync(dt1, 0, (long)rows1 * h * sizeof(float), s));
   if (n == 0) return;
-  const size_t lds = (size_t)(rows0 + rows1) * h * sizeof(float);
+  // wave-private path (7.2x the LDS-atomic kernel, see kernel comment)
+  const long vrows = rows0 + rows1;
+  const int blocks = (int)min((long)128, (n + 255) / 256);
+  const size_t lds128 = (size_t)WAVES_PER_BLOCK * vrows * 128 * sizeof(float);
+  const size_t lds64 = (size_t)WAVES_PER_BLOCK * vrows * 64 * sizeof(float);
+  if (h % 128 == 0 && lds128 <= 160 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        (const void*)vocab_scatter_dual_priv_kernel<GT, 4, 128>,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds128));
+    vocab_scatter_dual_priv_kernel<GT, 4, 128>
+        <<<dim3(max(blocks, 1), h / 128),
+           dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds128, s>>>(
+            g, ea, astride, dt0, dt1, n, rows0, rows1, h);
+    return;
+  }
+  if (h % 64 == 0 && lds64 <= 160 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        (const void*)vocab_scatter_dual_priv_kernel<GT, 4, 64>,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds64));
+    vocab_scatter_dual_priv_kernel<GT, 4, 64>
+        <<<dim3(max(blocks, 1), h / 64),
+           dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds64, s>>>(
+            g, ea, astride, dt0, dt1, n, rows0, rows1, h);
+    return;
+  }
+  const size_t lds = (size_t)vrows * h * sizeof(float);
   if (lds > 64 * 1024) {
     HIP_CHECK(hipFuncSetAttribute(
-        (const void*)vocab_scatter_dual_kernel<GT>,
+        (const void*)vocab_scatter_dual_kernel<GT, 4>,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
-  const int blocks = (int)min((long)256, (n + 63) / 64);
-  vocab_scatter_dual_kernel<GT><<<dim3(blocks),
-                                  dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds,
-                                  s>>>(g, ea, astride, dt0, dt1, n, rows0,
-                                       rows1, h);
+  const int ablocks = (int)min((long)768, (n + 63) / 64);
+  vocab_scatter_dual_kernel<GT, 4><<<dim3(max(ablocks, 1)),
+                                     dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds,
+                                     s>>>(g, ea, astride, dt0, dt1, n, rows0,
+                                          rows1, h);
 }
 
 void launch_vocab_scatter_dual(const float* g, const long* ea, int astride,
@@ -456,11 +533,91 @@ void launch_vocab_scatter_dual16(const void* g, const long* ea, int astride,
                           rows1, h, s);
 }
 
+// single-table wave-private variant (same rationale/measurements as the dual
+// kernel above: plain LDS read+add+write beats ds_add_f32 ~7x)
+template <int U, int HH>
+__global__ void vocab_scatter_priv_kernel(const float* __restrict__ g,
+                                          const long* __restrict__ idx,
+                                          long idx_stride,
+                                          float* __restrict__ dtable, long n,
+                                          int rows, int h, int gstride,
+                                          int col_off) {
+  constexpr int CPL = HH / PERTGNN_WAVE;
+  extern __shared__ float acc[];  // [WAVES_PER_BLOCK][rows*HH]
+  const long vwh = (long)rows * HH;
+  for (long t = threadIdx.x; t < WAVES_PER_BLOCK * vwh; t += blockDim.x)
+    acc[t] = 0.f;
+  __syncthreads();
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  float* my = acc + (long)wid * vwh;
+  const int c0 = blockIdx.y * HH;
+  const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
+  const long r0 = (long)blockIdx.x * rows_per_block;
+  const long r1 = min(n, r0 + rows_per_block);
+  const long step = WAVES_PER_BLOCK;
+  long r = r0 + wid;
+  for (; r + (U - 1) * step < r1; r += U * step) {
+    long v[U];
+    float xv[U][CPL];
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+      v[u] = idx[(r + u * step) * idx_stride];
+#pragma unroll
+      for (int k = 0; k < CPL; ++k)
+        xv[u][k] = g[(r + u * step) * gstride + col_off + c0 + lane * CPL + k];
+    }
+#pragma unroll
+    for (int u = 0; u < U; ++u)
+#pragma unroll
+      for (int k = 0; k < CPL; ++k) my[v[u] * HH + lane * CPL + k] += xv[u][k];
+  }
+  for (; r < r1; r += step) {
+    const long v = idx[r * idx_stride];
+#pragma unroll
+    for (int k = 0; k < CPL; ++k)
+      my[v * HH + lane * CPL + k] += g[r * gstride + col_off + c0 + lane * CPL + k];
+  }
+  __syncthreads();
+  for (long t = threadIdx.x; t < vwh; t += blockDim.x)
+    acc[t] += acc[vwh + t] + acc[2 * vwh + t] + acc[3 * vwh + t];
+  __syncthreads();
+  for (long t = threadIdx.x; t < vwh; t += blockDim.x) {
+    const float val = acc[t];
+    if (val != 0.f) atomicAdd(&dtable[(t / HH) * h + c0 + t % HH], val);
+  }
+}
+
 void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
                           float* dtable, long n, int rows, int h, int gstride,
                           int col_off, hipStream_t s) {
   HIP_CHECK(hipMemsetAsync(dtable, 0, (long)rows * h * sizeof(float), s));
   if (n == 0) return;
+  {
+    const int blocks = (int)min((long)128, (n + 255) / 256);
+    const size_t lds128 = (size_t)WAVES_PER_BLOCK * rows * 128 * sizeof(float);
+    const size_t lds64 = (size_t)WAVES_PER_BLOCK * rows * 64 * sizeof(float);
+    if (h % 128 == 0 && lds128 <= 160 * 1024) {
+      HIP_CHECK(hipFuncSetAttribute(
+          (const void*)vocab_scatter_priv_kernel<4, 128>,
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds128));
+      vocab_scatter_priv_kernel<4, 128>
+          <<<dim3(max(blocks, 1), h / 128),
+             dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds128, s>>>(
+              g, idx, idx_stride, dtable, n, rows, h, gstride, col_off);
+      return;
+    }
+    if (h % 64 == 0 && lds64 <= 160 * 1024) {
+      HIP_CHECK(hipFuncSetAttribute(
+          (const void*)vocab_scatter_priv_kernel<4, 64>,
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds64));
+      vocab_scatter_priv_kernel<4, 64>
+          <<<dim3(max(blocks, 1), h / 64),
+             dim3(WAVES_PER_BLOCK * PERTGNN_WAVE), lds64, s>>>(
+              g, idx, idx_stride, dtable, n, rows, h, gstride, col_off);
+      return;
+    }
+  }
   const size_t lds = (size_t)rows * h * sizeof(float);
   const bool vec = (h % (4 * PERTGNN_WAVE) == 0) && ((gstride & 3) == 0) &&
                    ((col_off & 3) == 0);
