@@ -62,141 +62,163 @@ __device__ __forceinline__ float ld1(const __bf16* p) { return (float)*p; }
 // global [free][contract] (contract-minor) -> LDS [free][BK+PAD]
 // 4 bf16 packed into one 8-byte LDS write (scalar u16 LDS writes are ~2x
 // slower — guide G13 applies to LDS too).
+//
+// Split into load() / commit() so the mainloop can ISSUE the next tile's
+// global loads, run the MFMA work on the current LDS buffer while they are
+// in flight, and only then wait + write LDS: the fused form stalled every
+// wave on s_waitcnt(vmcnt) BEFORE any MFMA issued (PMC: 72% SQ_WAIT_ANY on
+// the NT kernel).
+template <int BF, int BK, typename T16, typename TA>
+struct StageCmin {
+  static constexpr int LDW = BK + BGEMM_PAD;
+  static constexpr int QUADS = BK / 4;
+  static constexpr int FSTEP = BGEMM_THREADS / QUADS;
+  static constexpr int HALVES = BF / FSTEP;
+  float v[HALVES][4];
+
+  __device__ __forceinline__ void load(const TA* __restrict__ g, long ld,
+                                       int free0, int contract0, int free_max,
+                                       int contract_max) {
+    const int t = threadIdx.x;
+    const int f = t / QUADS;
+    const int cq = (t % QUADS) * 4;
+    const bool interior = (free0 + BF <= free_max) &&
+                          (contract0 + BK <= contract_max);
+    const bool aligned = (ld & 3) == 0;
+    // path choice HOISTED outside the loops: a branch inside the staging
+    // loop makes hipcc emit per-element load+vmcnt(0) chains
+    // (guide §5 ".s-level traps" (c))
+    if (interior && aligned) {
+#pragma unroll
+      for (int half = 0; half < HALVES; ++half)
+        ld4(&g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq],
+            v[half]);
+    } else if (interior) {  // odd leading dim: unchecked scalars
+#pragma unroll
+      for (int half = 0; half < HALVES; ++half) {
+        const TA* row =
+            &g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) v[half][u] = ld1(row + u);
+      }
+    } else {
+#pragma unroll
+      for (int half = 0; half < HALVES; ++half) {
+        const int gf = free0 + f + half * FSTEP;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int gc = contract0 + cq + u;
+          v[half][u] = (gf < free_max && gc < contract_max)
+                           ? ld1(&g[(long)gf * ld + gc])
+                           : 0.f;
+        }
+      }
+    }
+  }
+
+  __device__ __forceinline__ void commit(T16* lds) const {
+    const int t = threadIdx.x;
+    const int f = t / QUADS;
+    const int cq = (t % QUADS) * 4;
+#pragma unroll
+    for (int half = 0; half < HALVES; ++half)
+      *reinterpret_cast<typename Vec16<T16>::v4*>(
+          &lds[(f + half * FSTEP) * LDW + cq]) =
+          pack4<T16>(v[half][0], v[half][1], v[half][2], v[half][3]);
+  }
+};
+
 template <int BF, int BK, typename T16, typename TA>
 __device__ __forceinline__ void bstage_cmin(const TA* __restrict__ g,
                                             long ld, int free0, int contract0,
                                             int free_max, int contract_max,
                                             T16* lds) {
-  constexpr int LDW = BK + BGEMM_PAD;
-  const int t = threadIdx.x;
-  constexpr int QUADS = BK / 4;
-  const int f = t / QUADS;             // 16 free rows per pass
-  const int cq = (t % QUADS) * 4;
-  constexpr int FSTEP = BGEMM_THREADS / QUADS;  // 16
-  const bool interior = (free0 + BF <= free_max) &&
-                        (contract0 + BK <= contract_max);
-  const bool aligned = (ld & 3) == 0;
-  // the path choice is HOISTED outside the loops: a branch inside the
-  // staging loop makes hipcc emit per-element load+vmcnt(0) chains
-  // (guide §5 ".s-level traps" (c))
-  if (interior && aligned) {
-    float v[BF / FSTEP][4];
-#pragma unroll
-    for (int half = 0; half < BF / FSTEP; ++half)
-      ld4(&g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq], v[half]);
-#pragma unroll
-    for (int half = 0; half < BF / FSTEP; ++half)
-      *reinterpret_cast<typename Vec16<T16>::v4*>(
-          &lds[(f + half * FSTEP) * LDW + cq]) =
-          pack4<T16>(v[half][0], v[half][1], v[half][2], v[half][3]);
-  } else if (interior) {  // odd leading dim (layer-1 K=9+H): unchecked scalars
-    float v[BF / FSTEP][4];
-#pragma unroll
-    for (int half = 0; half < BF / FSTEP; ++half) {
-      const TA* row = &g[(long)(free0 + f + half * FSTEP) * ld + contract0 + cq];
-#pragma unroll
-      for (int u = 0; u < 4; ++u) v[half][u] = ld1(row + u);
-    }
-#pragma unroll
-    for (int half = 0; half < BF / FSTEP; ++half)
-      *reinterpret_cast<typename Vec16<T16>::v4*>(
-          &lds[(f + half * FSTEP) * LDW + cq]) =
-          pack4<T16>(v[half][0], v[half][1], v[half][2], v[half][3]);
-  } else {
-#pragma unroll
-    for (int half = 0; half < BF / FSTEP; ++half) {
-      const int ff = f + half * FSTEP;
-      const int gf = free0 + ff;
-      float v[4];
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int gc = contract0 + cq + u;
-        v[u] = (gf < free_max && gc < contract_max)
-                   ? ld1(&g[(long)gf * ld + gc])
-                   : 0.f;
-      }
-      *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[ff * LDW + cq]) =
-          pack4<T16>(v[0], v[1], v[2], v[3]);
-    }
-  }
+  StageCmin<BF, BK, T16, TA> s;
+  s.load(g, ld, free0, contract0, free_max, contract_max);
+  s.commit(lds);
 }
 
 // global [contract][free] (contract-major) -> LDS [free][BK+PAD]: each thread
 // transposes a 4x4 block in registers (4 coalesced f32x4 loads from 4
-// contract rows), then writes 4 packed 8-byte LDS rows.
+// contract rows), then writes 4 packed 8-byte LDS rows.  Split load/commit
+// for the same software-pipelining reason as StageCmin.
+template <int BF, int BK, typename T16, typename TA>
+struct StageCmaj {
+  static constexpr int LDW = BK + BGEMM_PAD;
+  static constexpr int FQUADS = BF / 4;
+  static constexpr int CSTEP = (BGEMM_THREADS / FQUADS) * 4;
+  static constexpr int HALVES = (BK > CSTEP ? BK / CSTEP : 1);
+  float v[HALVES][4][4];
+
+  __device__ __forceinline__ void load(const TA* __restrict__ g, long ld,
+                                       int contract0, int free0,
+                                       int contract_max, int free_max) {
+    const int t = threadIdx.x;
+    const int cb = (t / FQUADS) * 4;
+    const int fq = (t % FQUADS) * 4;
+    const bool interior = (contract0 + BK <= contract_max) &&
+                          (free0 + BF <= free_max);
+    const bool aligned = ((ld & 3) == 0) && ((free0 & 3) == 0);
+    // path choice hoisted outside the loops (guide §5 ".s-level traps" (c))
+    if (interior && aligned) {
+#pragma unroll
+      for (int half = 0; half < HALVES; ++half)
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+          ld4(&g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq],
+              v[half][u]);
+    } else if (interior) {
+#pragma unroll
+      for (int half = 0; half < HALVES; ++half)
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const TA* row =
+              &g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq];
+#pragma unroll
+          for (int w = 0; w < 4; ++w) v[half][u][w] = ld1(row + w);
+        }
+    } else {
+#pragma unroll
+      for (int half = 0; half < HALVES; ++half) {
+        const int cc = cb + half * CSTEP;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int gc = contract0 + cc + u;
+#pragma unroll
+          for (int w = 0; w < 4; ++w) {
+            const int gf = free0 + fq + w;
+            v[half][u][w] = (gc < contract_max && gf < free_max)
+                                ? ld1(&g[(long)gc * ld + gf])
+                                : 0.f;
+          }
+        }
+      }
+    }
+  }
+
+  __device__ __forceinline__ void commit(T16* lds) const {
+    const int t = threadIdx.x;
+    const int cb = (t / FQUADS) * 4;
+    const int fq = (t % FQUADS) * 4;
+#pragma unroll
+    for (int half = 0; half < HALVES; ++half)
+#pragma unroll
+      for (int w = 0; w < 4; ++w)
+        *reinterpret_cast<typename Vec16<T16>::v4*>(
+            &lds[(fq + w) * LDW + cb + half * CSTEP]) =
+            pack4<T16>(v[half][0][w], v[half][1][w], v[half][2][w],
+                       v[half][3][w]);
+  }
+};
+
 template <int BF, int BK, typename T16, typename TA>
 __device__ __forceinline__ void bstage_cmaj(const TA* __restrict__ g,
                                             long ld, int contract0, int free0,
                                             int contract_max, int free_max,
                                             T16* lds) {
-  constexpr int LDW = BK + BGEMM_PAD;
-  const int t = threadIdx.x;
-  constexpr int FQUADS = BF / 4;
-  const int cb = (t / FQUADS) * 4;     // contract block of 4
-  const int fq = (t % FQUADS) * 4;     // free quad
-  constexpr int CSTEP = (BGEMM_THREADS / FQUADS) * 4;
-  const bool interior = (contract0 + BK <= contract_max) &&
-                        (free0 + BF <= free_max);
-  const bool aligned = ((ld & 3) == 0) && ((free0 & 3) == 0);
-  constexpr int HALVES = (BK > CSTEP ? BK / CSTEP : 1);
-  // path choice hoisted outside the loops (guide §5 ".s-level traps" (c))
-  if (interior && aligned) {
-    float v[HALVES][4][4];
-#pragma unroll
-    for (int half = 0; half < HALVES; ++half)
-#pragma unroll
-      for (int u = 0; u < 4; ++u)
-        ld4(&g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq],
-            v[half][u]);
-#pragma unroll
-    for (int half = 0; half < HALVES; ++half)
-#pragma unroll
-      for (int w = 0; w < 4; ++w)
-        *reinterpret_cast<typename Vec16<T16>::v4*>(
-            &lds[(fq + w) * LDW + cb + half * CSTEP]) =
-            pack4<T16>(v[half][0][w], v[half][1][w], v[half][2][w],
-                       v[half][3][w]);
-  } else if (interior) {
-    float v[HALVES][4][4];
-#pragma unroll
-    for (int half = 0; half < HALVES; ++half)
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const TA* row =
-            &g[(long)(contract0 + cb + half * CSTEP + u) * ld + free0 + fq];
-#pragma unroll
-        for (int w = 0; w < 4; ++w) v[half][u][w] = ld1(row + w);
-      }
-#pragma unroll
-    for (int half = 0; half < HALVES; ++half)
-#pragma unroll
-      for (int w = 0; w < 4; ++w)
-        *reinterpret_cast<typename Vec16<T16>::v4*>(
-            &lds[(fq + w) * LDW + cb + half * CSTEP]) =
-            pack4<T16>(v[half][0][w], v[half][1][w], v[half][2][w],
-                       v[half][3][w]);
-  } else {
-#pragma unroll
-    for (int half = 0; half < HALVES; ++half) {
-      const int cc = cb + half * CSTEP;
-      float v[4][4];
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int gc = contract0 + cc + u;
-#pragma unroll
-        for (int w = 0; w < 4; ++w) {
-          const int gf = free0 + fq + w;
-          v[u][w] = (gc < contract_max && gf < free_max)
-                        ? ld1(&g[(long)gc * ld + gf])
-                        : 0.f;
-        }
-      }
-#pragma unroll
-      for (int w = 0; w < 4; ++w)
-        *reinterpret_cast<typename Vec16<T16>::v4*>(&lds[(fq + w) * LDW + cc]) =
-            pack4<T16>(v[0][w], v[1][w], v[2][w], v[3][w]);
-    }
-  }
+  StageCmaj<BF, BK, T16, TA> s;
+  s.load(g, ld, contract0, free0, contract_max, free_max);
+  s.commit(lds);
 }
 
 // ---------------------------------------------------------------------------
@@ -295,10 +317,14 @@ __global__ void gemm_bf16_nt_kernel(const TA* __restrict__ a,
   bstage_cmin<BM, BK, T16>(a, k, m0, 0, m, k, lds_a[0]);
   bstage_cmin<BN, BK, T16>(b, k, n0, 0, n, k, lds_b[0]);
   __syncthreads();
+  StageCmin<BM, BK, T16, TA> sa;
+  StageCmin<BN, BK, T16, float> sb;
   for (int k0 = BK; k0 < k; k0 += BK) {
-    bstage_cmin<BM, BK, T16>(a, k, m0, k0, m, k, lds_a[buf ^ 1]);
-    bstage_cmin<BN, BK, T16>(b, k, n0, k0, n, k, lds_b[buf ^ 1]);
-    wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+    sa.load(a, k, m0, k0, m, k);        // issue loads…
+    sb.load(b, k, n0, k0, n, k);
+    wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);  // …MFMA while in flight
+    sa.commit(lds_a[buf ^ 1]);
+    sb.commit(lds_b[buf ^ 1]);
     __syncthreads();
     buf ^= 1;
   }
@@ -308,7 +334,7 @@ __global__ void gemm_bf16_nt_kernel(const TA* __restrict__ a,
 
 template <int BM, int BN, int BK = 64, typename T16 = __bf16,
           typename TA = float, typename TO = float>
-__launch_bounds__(BGEMM_THREADS)
+__launch_bounds__(BGEMM_THREADS, 3)
 __global__ void gemm_bf16_nn_kernel(const TA* __restrict__ a,
                                     const float* __restrict__ b,
                                     const float* __restrict__ bias,
@@ -333,10 +359,14 @@ __global__ void gemm_bf16_nn_kernel(const TA* __restrict__ a,
   bstage_cmin<BM, BK, T16>(a, n, m0, 0, m, n, lds_a[0]);
   bstage_cmaj<BN, BK, T16>(b, k2, 0, n0, n, k2, lds_b[0]);
   __syncthreads();
+  StageCmin<BM, BK, T16, TA> sa;
+  StageCmaj<BN, BK, T16, float> sb;
   for (int c0 = BK; c0 < n; c0 += BK) {
-    bstage_cmin<BM, BK, T16>(a, n, m0, c0, m, n, lds_a[buf ^ 1]);
-    bstage_cmaj<BN, BK, T16>(b, k2, c0, n0, n, k2, lds_b[buf ^ 1]);
+    sa.load(a, n, m0, c0, m, n);
+    sb.load(b, k2, c0, n0, n, k2);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
+    sa.commit(lds_a[buf ^ 1]);
+    sb.commit(lds_b[buf ^ 1]);
     __syncthreads();
     buf ^= 1;
   }
@@ -385,14 +415,18 @@ __global__ void gemm_bf16_tn_kernel(const TA* __restrict__ a,
   bstage_cmaj<BM, BK, T16>(a, n, c_beg, n0, c_end, n, lds_a[0]);
   bstage_cmaj<BN, BK, T16>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
   __syncthreads();
+  StageCmaj<BM, BK, T16, TA> sa;
+  StageCmaj<BN, BK, T16, float> sb;
   for (int cc = c_beg + BK; cc < c_end; cc += BK) {
-    bstage_cmaj<BM, BK, T16>(a, n, cc, n0, c_end, n, lds_a[buf ^ 1]);
-    bstage_cmaj<BN, BK, T16>(b, k2, cc, k0, c_end, k2, lds_b[buf ^ 1]);
+    sa.load(a, n, cc, n0, c_end, n);
+    sb.load(b, k2, cc, k0, c_end, k2);
     wt.mma(lds_a[buf], lds_b[buf], wm, wn, lane);
     if (do_bias && bcol < BM)
 #pragma unroll
       for (int r = 0; r < BK; ++r)
         dbsum += (float)lds_a[buf][bcol * LDW + r];
+    sa.commit(lds_a[buf ^ 1]);
+    sb.commit(lds_b[buf ^ 1]);
     __syncthreads();
     buf ^= 1;
   }
