@@ -54,6 +54,35 @@ void launch_bn_bwd_apply_only(const float*, const float*, const float*,
                               const float*, long, float*, long, int, bool,
                               hipStream_t);
 void launch_bn_grad_affine(const float*, float*, float*, int, hipStream_t);
+// act16 variants (bf16 activation / gradient streams, fp32 statistics)
+void launch_bn_fwd16(const float*, const float*, const float*, float*, float*,
+                     float*, float*, float*, void*, long, int, float, float,
+                     bool, bool, hipStream_t);
+void launch_bn_finalize_apply16(const float*, const float*, long, const float*,
+                                const float*, float*, float*, float*, float*,
+                                void*, long, int, float, float, bool, bool,
+                                hipStream_t);
+void launch_bn_bwd_partials_only16(const void*, const float*, const void*,
+                                   const float*, const float*, long, int, bool,
+                                   float*, hipStream_t);
+void launch_bn_bwd_apply_only16(const void*, const float*, const void*,
+                                const float*, const float*, const float*,
+                                const float*, long, float*, long, int, bool,
+                                hipStream_t);
+void launch_bn_bwd16(const void*, const float*, const void*, const float*,
+                     const float*, const float*, float*, float*, float*,
+                     float*, long, int, bool, hipStream_t);
+void launch_seg_pool_fwd16(const void*, const float*, const float*,
+                           const int*, float*, float*, int, int, int,
+                           hipStream_t);
+void launch_seg_pool_bwd16(const float*, const float*, const float*,
+                           const long*, void*, long, int, hipStream_t);
+void launch_gemm_bf16_nt_a16o16(const void*, const float*, const float*,
+                                void*, int, int, int, hipStream_t);
+void launch_gemm_bf16_nn_a16o16(const void*, const float*, void*, int, int,
+                                int, hipStream_t);
+void launch_gemm_bf16_tn_a16b16(const void*, const void*, float*, float*, int,
+                                int, int, hipStream_t);
 void launch_quantile_loss_fwd(const float*, const float*, float*, long, float,
                               hipStream_t);
 void launch_quantile_loss_bwd(const float*, const float*, const float*, float*,
@@ -767,6 +796,184 @@ std::vector<torch::Tensor> linear_wgrad16(torch::Tensor g, torch::Tensor x,
   return {dw, db};
 }
 
+
+// ---------------------------------------------------------------------------
+// act16-v2: bf16 activation streams between BN <-> GEMMs <-> pool.
+// Statistics, affine params, weights and weight-grads stay fp32.
+// ---------------------------------------------------------------------------
+
+std::vector<torch::Tensor> bn_relu_fwd16(torch::Tensor x, torch::Tensor gamma,
+                                         torch::Tensor beta,
+                                         torch::Tensor running_mean,
+                                         torch::Tensor running_var,
+                                         double momentum, double eps,
+                                         bool training, bool relu) {
+  CHECK_IN(x); CHECK_IN(gamma); CHECK_IN(beta);
+  const long n = x.size(0);
+  const int h = x.size(1);
+  auto y = torch::empty({n, h}, x.options().dtype(torch::kBFloat16));
+  auto mean = torch::empty({h}, x.options());
+  auto invstd = torch::empty({h}, x.options());
+  auto partials = torch::empty({2 * h}, x.options());
+  launch_bn_fwd16(x.data_ptr<float>(), gamma.data_ptr<float>(),
+                  beta.data_ptr<float>(), running_mean.data_ptr<float>(),
+                  running_var.data_ptr<float>(), mean.data_ptr<float>(),
+                  invstd.data_ptr<float>(), partials.data_ptr<float>(),
+                  y.data_ptr(), n, h, (float)momentum, (float)eps, training,
+                  relu, cur_stream());
+  return {y, mean, invstd};
+}
+
+std::vector<torch::Tensor> bn_relu_bwd16(torch::Tensor g, torch::Tensor x,
+                                         torch::Tensor gamma,
+                                         torch::Tensor mean,
+                                         torch::Tensor invstd, torch::Tensor y,
+                                         bool relu) {
+  CHECK_IN(g); CHECK_IN(x);
+  const long n = x.size(0);
+  const int h = x.size(1);
+  auto dx = torch::empty_like(x);
+  auto dgamma = torch::empty({h}, x.options());
+  auto dbeta = torch::empty({h}, x.options());
+  auto partials = torch::empty({2 * h}, x.options());
+  launch_bn_bwd16(g.data_ptr(), x.data_ptr<float>(), y.data_ptr(),
+                  mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                  gamma.data_ptr<float>(), partials.data_ptr<float>(),
+                  dx.data_ptr<float>(), dgamma.data_ptr<float>(),
+                  dbeta.data_ptr<float>(), n, h, relu, cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
+std::vector<torch::Tensor> bn_finalize_apply16(
+    torch::Tensor x, torch::Tensor partials, int64_t count,
+    torch::Tensor gamma, torch::Tensor beta, torch::Tensor running_mean,
+    torch::Tensor running_var, double momentum, double eps, bool training,
+    bool relu) {
+  CHECK_IN(x); CHECK_IN(partials);
+  const int h = x.size(1);
+  auto y = torch::empty({x.size(0), h}, x.options().dtype(torch::kBFloat16));
+  auto mean = torch::empty({h}, x.options());
+  auto invstd = torch::empty({h}, x.options());
+  launch_bn_finalize_apply16(
+      x.data_ptr<float>(), partials.data_ptr<float>(), count,
+      gamma.data_ptr<float>(), beta.data_ptr<float>(),
+      running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
+      mean.data_ptr<float>(), invstd.data_ptr<float>(), y.data_ptr(),
+      x.size(0), h, (float)momentum, (float)eps, training, relu,
+      cur_stream());
+  return {y, mean, invstd};
+}
+
+torch::Tensor bn_bwd_partials16(torch::Tensor g, torch::Tensor x,
+                                torch::Tensor y, torch::Tensor mean,
+                                torch::Tensor invstd, bool relu) {
+  CHECK_IN(g); CHECK_IN(x);
+  auto partials = torch::empty({2 * x.size(1)}, x.options());
+  launch_bn_bwd_partials_only16(g.data_ptr(), x.data_ptr<float>(),
+                                y.data_ptr(), mean.data_ptr<float>(),
+                                invstd.data_ptr<float>(), x.size(0), x.size(1),
+                                relu, partials.data_ptr<float>(),
+                                cur_stream());
+  return partials;
+}
+
+std::vector<torch::Tensor> bn_bwd_apply16(
+    torch::Tensor g, torch::Tensor x, torch::Tensor y, torch::Tensor mean,
+    torch::Tensor invstd, torch::Tensor gamma, torch::Tensor partials_global,
+    torch::Tensor partials_local, int64_t count, bool relu) {
+  CHECK_IN(g); CHECK_IN(x);
+  const int h = x.size(1);
+  auto dx = torch::empty_like(x);
+  auto dgamma = torch::empty({h}, x.options());
+  auto dbeta = torch::empty({h}, x.options());
+  launch_bn_bwd_apply_only16(g.data_ptr(), x.data_ptr<float>(), y.data_ptr(),
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             gamma.data_ptr<float>(),
+                             partials_global.data_ptr<float>(), count,
+                             dx.data_ptr<float>(), x.size(0), h, relu,
+                             cur_stream());
+  launch_bn_grad_affine(partials_local.data_ptr<float>(),
+                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), h,
+                        cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
+torch::Tensor seg_pool_fwd16(torch::Tensor x, torch::Tensor probs,
+                             torch::Tensor nn, torch::Tensor batch_ptr,
+                             int64_t num_graphs) {
+  CHECK_IN(x); CHECK_IN(probs); CHECK_IN(nn); CHECK_IN(batch_ptr);
+  const int h = x.size(1);
+  const long n = x.size(0);
+  long p = 8192 / std::max<int64_t>(num_graphs, 1);
+  const long per_g = (n + std::max<int64_t>(num_graphs, 1) - 1) /
+                     std::max<int64_t>(num_graphs, 1);
+  p = std::min<long>(std::max<long>(std::min(p, per_g), 1), 64);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto partial = torch::empty({num_graphs * p, h}, fopt);
+  auto out = torch::empty({num_graphs, h}, fopt);
+  launch_seg_pool_fwd16(x.data_ptr(), probs.data_ptr<float>(),
+                        nn.data_ptr<float>(), batch_ptr.data_ptr<int>(),
+                        partial.data_ptr<float>(), out.data_ptr<float>(),
+                        (int)num_graphs, (int)p, h, cur_stream());
+  return out;
+}
+
+torch::Tensor seg_pool_bwd16(torch::Tensor gout, torch::Tensor probs,
+                             torch::Tensor nn, torch::Tensor batch) {
+  CHECK_IN(gout); CHECK_IN(batch);
+  const long n = batch.size(0);
+  const int h = gout.size(1);
+  auto dx = torch::empty({n, h}, gout.options().dtype(torch::kBFloat16));
+  launch_seg_pool_bwd16(gout.data_ptr<float>(), probs.data_ptr<float>(),
+                        nn.data_ptr<float>(), batch.data_ptr<long>(),
+                        dx.data_ptr(), n, h, cur_stream());
+  return dx;
+}
+
+torch::Tensor linear_fwd_a16o16(torch::Tensor x, torch::Tensor w,
+                                torch::Tensor b) {
+  CHECK_IN(x); CHECK_IN(w);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = w.size(0);
+  auto y = torch::empty({m, n}, x.options());  // bf16
+  const float* bias = nullptr;
+  if (b.defined() && b.numel() > 0) bias = b.data_ptr<float>();
+  launch_gemm_bf16_nt_a16o16(x.data_ptr(), w.data_ptr<float>(), bias,
+                             y.data_ptr(), m, n, k, cur_stream());
+  return y;
+}
+
+torch::Tensor linear_dgrad16_o16(torch::Tensor g, torch::Tensor w) {
+  CHECK_IN(g); CHECK_IN(w);
+  const int m = g.size(0);
+  const int n = w.size(0);
+  const int k = w.size(1);
+  auto dx = torch::empty({m, k}, g.options());  // bf16
+  launch_gemm_bf16_nn_a16o16(g.data_ptr(), w.data_ptr<float>(), dx.data_ptr(),
+                             m, n, k, cur_stream());
+  return dx;
+}
+
+std::vector<torch::Tensor> linear_wgrad16_b16(torch::Tensor g, torch::Tensor x,
+                                              bool has_bias) {
+  CHECK_IN(g); CHECK_IN(x);
+  const int m = x.size(0);
+  const int k = x.size(1);
+  const int n = g.size(1);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto dw = torch::empty({n, k}, fopt);
+  torch::Tensor db = torch::empty({0}, fopt);
+  float* db_ptr = nullptr;
+  if (has_bias) {
+    db = torch::empty({n}, fopt);
+    db_ptr = db.data_ptr<float>();
+  }
+  launch_gemm_bf16_tn_a16b16(g.data_ptr(), x.data_ptr(), dw.data_ptr<float>(),
+                             db_ptr, m, n, k, cur_stream());
+  return {dw, db};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("linear_fwd_bf16_o16", &linear_fwd_bf16_o16);
   mod.def("linear_dgrad16", &linear_dgrad16);
@@ -803,6 +1010,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("embed_node_fwd", &embed_node_fwd);
   mod.def("embed_edge_fwd", &embed_edge_fwd);
   mod.def("gather_rows", &gather_rows);
+  mod.def("bn_relu_fwd16", &bn_relu_fwd16);
+  mod.def("bn_relu_bwd16", &bn_relu_bwd16);
+  mod.def("bn_finalize_apply16", &bn_finalize_apply16);
+  mod.def("bn_bwd_partials16", &bn_bwd_partials16);
+  mod.def("bn_bwd_apply16", &bn_bwd_apply16);
+  mod.def("seg_pool_fwd16", &seg_pool_fwd16);
+  mod.def("seg_pool_bwd16", &seg_pool_bwd16);
+  mod.def("linear_fwd_a16o16", &linear_fwd_a16o16);
+  mod.def("linear_dgrad16_o16", &linear_dgrad16_o16);
+  mod.def("linear_wgrad16_b16", &linear_wgrad16_b16);
   mod.def("bn_relu_fwd", &bn_relu_fwd);
   mod.def("bn_relu_bwd", &bn_relu_bwd);
   mod.def("quantile_loss_fwd", &quantile_loss_fwd);

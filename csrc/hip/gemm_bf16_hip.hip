@@ -376,10 +376,10 @@ __global__ void gemm_bf16_nn_kernel(const TA* __restrict__ a,
 }
 
 template <int BM, int BN, int BK = 64, typename T16 = __bf16,
-          typename TA = float>
+          typename TA = float, typename TB = float>
 __launch_bounds__(BGEMM_THREADS)
 __global__ void gemm_bf16_tn_kernel(const TA* __restrict__ a,
-                                    const float* __restrict__ b,
+                                    const TB* __restrict__ b,
                                     float* __restrict__ c,
                                     float* __restrict__ dbias, int m, int n,
                                     int k2, int slices) {
@@ -417,7 +417,7 @@ __global__ void gemm_bf16_tn_kernel(const TA* __restrict__ a,
   bstage_cmaj<BN, BK, T16>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
   __syncthreads();
   StageCmaj<BM, BK, T16, TA> sa;
-  StageCmaj<BN, BK, T16, float> sb;
+  StageCmaj<BN, BK, T16, TB> sb;
   for (int cc = c_beg + BK; cc < c_end; cc += BK) {
     sa.load(a, n, cc, n0, c_end, n);
     sb.load(b, k2, cc, k0, c_end, k2);
@@ -677,6 +677,74 @@ void launch_gemm_bf16_tn_a16(const void* a_v, const float* b, float* c,
                                                               slices);
   else
    hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64, 64, __bf16, __bf16>)
+        , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+}
+
+// ---------------------------------------------------------------------------
+// act16-v2 launchers: fully bf16 activation IO (x16 in / qkvs16 out on the
+// forward, bf16 gradients through dgrad, bf16 x through wgrad).  Weights and
+// accumulation stay fp32; only stream dtypes change — at the flagship shapes
+// the three model GEMMs are stream-bound, so halving A/C bytes is the lever
+// (isolated: hipBLASLt bf16 reaches 1.9-3.5 TB/s effective on these shapes).
+// ---------------------------------------------------------------------------
+
+void launch_gemm_bf16_nt_a16o16(const void* a_v, const float* b,
+                                const float* bias, void* c_v, int m, int n,
+                                int k, hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  __bf16* c = (__bf16*)c_v;
+  if (m >= 512 && n >= 128) {
+    const int grid = ((m + 127) / 128) * ((n + 127) / 128);
+   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128, 32, __bf16, __bf16, __bf16>)
+        , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k, 0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((n + 63) / 64);
+   hipLaunchKernelGGL(( gemm_bf16_nt_kernel<64, 64, 64, __bf16, __bf16, __bf16>)
+        , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k, 0);
+  }
+}
+
+void launch_gemm_bf16_nn_a16o16(const void* a_v, const float* b, void* c_v,
+                                int m, int n, int k2, hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  __bf16* c = (__bf16*)c_v;
+  if (m >= 512 && k2 >= 128) {
+    const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
+   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128, 32, __bf16, __bf16, __bf16>)
+        , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, nullptr, c, m, n, k2,
+                                                    0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((k2 + 63) / 64);
+   hipLaunchKernelGGL(( gemm_bf16_nn_kernel<64, 64, 64, __bf16, __bf16, __bf16>)
+        , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, nullptr, c, m, n, k2,
+                                                    0);
+  }
+}
+
+void launch_gemm_bf16_tn_a16b16(const void* a_v, const void* b_v, float* c,
+                                float* dbias, int m, int n, int k2,
+                                hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  const __bf16* b = (const __bf16*)b_v;
+  const bool big = (n >= 128 && k2 >= 128);
+  const int bm = big ? 128 : 64;
+  const int bn = big ? 128 : 64;
+  const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
+  int slices = 1;
+  while (tiles * slices < 512 && slices < 64 && (long)slices * 64 * 4 < m)
+    slices *= 2;
+  if (slices > 1)
+    HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
+  if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
+  if (big)
+   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128, 32, __bf16, __bf16, __bf16>)
+        , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+  else
+   hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64, 64, __bf16, __bf16, __bf16>)
         , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
                                                               m, n, k2,
                                                               slices);

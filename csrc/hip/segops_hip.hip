@@ -18,8 +18,8 @@
 // partial[g*P+sub], then one wave per graph folds the P partials in order
 // (deterministic, no atomics; single waves per graph would leave the chip
 // ~64/256 CUs busy at trace-scale batches).
-template <int VPT>
-__global__ void seg_pool_p1_kernel(const float* __restrict__ x,
+template <int VPT, typename TX = float>
+__global__ void seg_pool_p1_kernel(const TX* __restrict__ x,
                                    const float* __restrict__ probs,
                                    const float* __restrict__ nn,
                                    const int* __restrict__ batch_ptr,
@@ -39,7 +39,7 @@ __global__ void seg_pool_p1_kernel(const float* __restrict__ x,
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
       const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) acc[j] += x[(long)i * h + c] * wt;
+      if (c < h) acc[j] += (float)x[(long)i * h + c] * wt;
     }
   }
 #pragma unroll
@@ -74,18 +74,19 @@ __global__ void seg_pool_p2_kernel(const float* __restrict__ partial,
   }
 }
 
+template <typename TD = float>
 __global__ void seg_pool_bwd_kernel(const float* __restrict__ gout,
                                     const float* __restrict__ probs,
                                     const float* __restrict__ nn,
                                     const long* __restrict__ batch,
-                                    float* __restrict__ dx, long n, int h) {
+                                    TD* __restrict__ dx, long n, int h) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   const long numel = n * h;
   for (long t = i; t < numel; t += stride) {
     const long row = t / h;
     const int c = (int)(t - row * h);
-    dx[t] = gout[batch[row] * h + c] * probs[row] / nn[row];
+    dx[t] = (TD)(gout[batch[row] * h + c] * probs[row] / nn[row]);
   }
 }
 
@@ -120,6 +121,40 @@ void launch_seg_pool_bwd(const float* gout, const float* probs,
   const int blocks = (int)min((numel + tpb - 1) / tpb, (long)4096);
  hipLaunchKernelGGL(( seg_pool_bwd_kernel), dim3(dim3(blocks)), dim3(dim3(tpb)), 0, stream, gout, probs, nn,
                                                               batch, dx, n, h);
+}
+
+// act16 variants: x16 in (bf16 last-layer activations), bf16 dx out
+void launch_seg_pool_fwd16(const void* x, const float* probs, const float* nn,
+                           const int* batch_ptr, float* partial, float* out,
+                           int b, int P, int h, hipStream_t stream) {
+  if (b == 0) return;
+  const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
+  const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  switch (vpt) {
+#define CASE(V)                                                                \
+  case V:                                                                      \
+   hipLaunchKernelGGL(( seg_pool_p1_kernel<V, __bf16>)                                              \
+        , dim3(dim3(ceil_div((long)b * P, WAVES_PER_BLOCK))), dim3(block), 0, stream,   \
+            (const __bf16*)x, probs, nn, batch_ptr, partial, b, P, h);         \
+   hipLaunchKernelGGL(( seg_pool_p2_kernel<V>)                                                      \
+        , dim3(dim3(ceil_div(b, WAVES_PER_BLOCK))), dim3(block), 0, stream,             \
+            partial, out, b, P, h);                                            \
+    break;
+    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+#undef CASE
+    default: abort();
+  }
+}
+
+void launch_seg_pool_bwd16(const float* gout, const float* probs,
+                           const float* nn, const long* batch, void* dx,
+                           long n, int h, hipStream_t stream) {
+  if (n == 0) return;
+  const long numel = n * h;
+  const int tpb = 256;
+  const int blocks = (int)min((numel + tpb - 1) / tpb, (long)4096);
+ hipLaunchKernelGGL(( seg_pool_bwd_kernel), dim3(dim3(blocks)), dim3(dim3(tpb)), 0, stream, 
+      gout, probs, nn, batch, (__bf16*)dx, n, h);
 }
 
 // ---------------------------------------------------------------------------
@@ -767,12 +802,16 @@ __global__ void bn_finalize_kernel(const float* __restrict__ partials, long n,
 }
 
 // stage C: normalize + affine (+ReLU)
+// TY = output/activation dtype: fp32 in exact mode, bf16 in the act16 mode
+// (standard mixed-precision BN: statistics and normalization math stay fp32,
+// only the activation stream is 16-bit).
+template <typename TY>
 __global__ void bn_apply_kernel(const float* __restrict__ x,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
-                                float* __restrict__ y, long n, int h,
+                                TY* __restrict__ y, long n, int h,
                                 int relu) {
   const long numel = n * h;
   const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -781,14 +820,15 @@ __global__ void bn_apply_kernel(const float* __restrict__ x,
     const int c = (int)(t % h);
     float v = (x[t] - mean[c]) * invstd[c] * gamma[c] + beta[c];
     if (relu) v = fmaxf(v, 0.f);
-    y[t] = v;
+    y[t] = (TY)v;
   }
 }
 
 // backward stage A: per-channel sums of gm and gm*xhat (gm = relu-masked g)
-__global__ void bn_bwd_partial_kernel(const float* __restrict__ g,
+template <typename TG, typename TY>
+__global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
                                       const float* __restrict__ x,
-                                      const float* __restrict__ y,
+                                      const TY* __restrict__ y,
                                       const float* __restrict__ mean,
                                       const float* __restrict__ invstd, long n,
                                       int h, int relu,
@@ -809,8 +849,8 @@ __global__ void bn_bwd_partial_kernel(const float* __restrict__ g,
       float gv[4], yv[4], xv[4];
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        gv[u] = g[(r + u) * h + c0];
-        yv[u] = y[(r + u) * h + c0];
+        gv[u] = (float)g[(r + u) * h + c0];
+        yv[u] = (float)y[(r + u) * h + c0];
         xv[u] = x[(r + u) * h + c0];
       }
 #pragma unroll
@@ -824,8 +864,8 @@ __global__ void bn_bwd_partial_kernel(const float* __restrict__ g,
       float gv[4], yv[4], xv[4];
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        gv[u] = g[(r + u) * h + c1];
-        yv[u] = y[(r + u) * h + c1];
+        gv[u] = (float)g[(r + u) * h + c1];
+        yv[u] = (float)y[(r + u) * h + c1];
         xv[u] = x[(r + u) * h + c1];
       }
 #pragma unroll
@@ -838,14 +878,14 @@ __global__ void bn_bwd_partial_kernel(const float* __restrict__ g,
   }
   for (; r < r1; ++r) {
     if (c0 < h) {
-      float gm = g[r * h + c0];
-      if (relu && y[r * h + c0] <= 0.f) gm = 0.f;
+      float gm = (float)g[r * h + c0];
+      if (relu && (float)y[r * h + c0] <= 0.f) gm = 0.f;
       s0 += gm;
       q0 += gm * (x[r * h + c0] - m0) * i0;
     }
     if (c1 < h) {
-      float gm = g[r * h + c1];
-      if (relu && y[r * h + c1] <= 0.f) gm = 0.f;
+      float gm = (float)g[r * h + c1];
+      if (relu && (float)y[r * h + c1] <= 0.f) gm = 0.f;
       s1 += gm;
       q1 += gm * (x[r * h + c1] - m1) * i1;
     }
@@ -861,9 +901,10 @@ __global__ void bn_bwd_partial_kernel(const float* __restrict__ g,
 }
 
 // backward stage B: dx = gamma*invstd*(gm - sum_gm/n - xhat*sum_gmx/n)
+template <typename TG, typename TY>
 __global__ void bn_bwd_apply_kernel(
-    const float* __restrict__ g, const float* __restrict__ x,
-    const float* __restrict__ y, const float* __restrict__ mean,
+    const TG* __restrict__ g, const float* __restrict__ x,
+    const TY* __restrict__ y, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
     const float* __restrict__ partials, float* __restrict__ dx, long n,
     long count, int h, int relu) {
@@ -873,8 +914,8 @@ __global__ void bn_bwd_apply_kernel(
   const float invn = 1.f / count;
   for (long t = i0; t < numel; t += stride) {
     const int c = (int)(t % h);
-    float gm = g[t];
-    if (relu && y[t] <= 0.f) gm = 0.f;
+    float gm = (float)g[t];
+    if (relu && (float)y[t] <= 0.f) gm = 0.f;
     const float xhat = (x[t] - mean[c]) * invstd[c];
     dx[t] = gamma[c] * invstd[c] *
             (gm - partials[c] * invn - xhat * partials[h + c] * invn);
@@ -1001,6 +1042,93 @@ void launch_bn_bwd(const float* g, const float* x, const float* y,
       g, x, y, mean, invstd, n, h, relu ? 1 : 0, partials);
  hipLaunchKernelGGL(( bn_bwd_apply_kernel), dim3(grid_for(n * h)), dim3(256), 0, s, 
       g, x, y, mean, invstd, gamma, partials, dx, n, n, h, relu ? 1 : 0);
+ hipLaunchKernelGGL(( bn_grad_affine_kernel), dim3(ceil_div(h, 256)), dim3(256), 0, s, partials, dgamma,
+                                                         dbeta, h);
+}
+
+// --- act16 launchers: bf16 activation stream (y) and bf16 incoming grad
+// (g), fp32 statistics/affine math and fp32 dx out (feeds the attention
+// backward).  Same kernels, TY/TG = __bf16. ---
+
+void launch_bn_fwd16(const float* x, const float* gamma, const float* beta,
+                     float* running_mean, float* running_var, float* mean,
+                     float* invstd, float* partials, void* y, long n, int h,
+                     float momentum, float eps, bool training, bool relu,
+                     hipStream_t s) {
+  if (n == 0) return;
+  if (training) {
+    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+    const int nblocks = (int)min((long)512, (n + 63) / 64);
+   hipLaunchKernelGGL(( bn_stats_partial_kernel), dim3(nblocks), dim3(256), 0, s, x, n, h, partials);
+   hipLaunchKernelGGL(( bn_finalize_kernel), dim3(ceil_div(h, 256)), dim3(256), 0, s, 
+        partials, n, h, eps, momentum, mean, invstd, running_mean, running_var,
+        1);
+  } else {
+    launch_bn_eval_stats(running_mean, running_var, mean, invstd, h, eps, s);
+  }
+ hipLaunchKernelGGL(( bn_apply_kernel), dim3(grid_for(n * h)), dim3(256), 0, s, x, mean, invstd, gamma, beta,
+                                                  (__bf16*)y, n, h,
+                                                  relu ? 1 : 0);
+}
+
+void launch_bn_finalize_apply16(const float* x, const float* partials,
+                                long count, const float* gamma,
+                                const float* beta, float* running_mean,
+                                float* running_var, float* mean, float* invstd,
+                                void* y, long n, int h, float momentum,
+                                float eps, bool training, bool relu,
+                                hipStream_t s) {
+  if (training) {
+   hipLaunchKernelGGL(( bn_finalize_kernel), dim3(ceil_div(h, 256)), dim3(256), 0, s, 
+        partials, count, h, eps, momentum, mean, invstd, running_mean,
+        running_var, 1);
+  } else {
+   hipLaunchKernelGGL(( bn_eval_stats_kernel), dim3(ceil_div(h, 256)), dim3(256), 0, s, 
+        running_mean, running_var, mean, invstd, h, eps);
+  }
+  if (n > 0)
+   hipLaunchKernelGGL(( bn_apply_kernel), dim3(grid_for(n * h)), dim3(256), 0, s, x, mean, invstd, gamma,
+                                                    beta, (__bf16*)y, n, h,
+                                                    relu ? 1 : 0);
+}
+
+void launch_bn_bwd_partials_only16(const void* g, const float* x,
+                                   const void* y, const float* mean,
+                                   const float* invstd, long n, int h,
+                                   bool relu, float* partials, hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+  if (n == 0) return;
+  const int nblocks = (int)min((long)512, (n + 63) / 64);
+ hipLaunchKernelGGL(( bn_bwd_partial_kernel), dim3(nblocks), dim3(256), 0, s, (const __bf16*)g, x,
+                                                (const __bf16*)y, mean, invstd,
+                                                n, h, relu ? 1 : 0, partials);
+}
+
+void launch_bn_bwd_apply_only16(const void* g, const float* x, const void* y,
+                                const float* mean, const float* invstd,
+                                const float* gamma, const float* partials,
+                                long count, float* dx, long n, int h,
+                                bool relu, hipStream_t s) {
+  if (n == 0) return;
+ hipLaunchKernelGGL(( bn_bwd_apply_kernel), dim3(grid_for(n * h)), dim3(256), 0, s, 
+      (const __bf16*)g, x, (const __bf16*)y, mean, invstd, gamma, partials, dx,
+      n, count, h, relu ? 1 : 0);
+}
+
+void launch_bn_bwd16(const void* g, const float* x, const void* y,
+                     const float* mean, const float* invstd,
+                     const float* gamma, float* partials, float* dx,
+                     float* dgamma, float* dbeta, long n, int h, bool relu,
+                     hipStream_t s) {
+  if (n == 0) return;
+  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+  const int nblocks = (int)min((long)512, (n + 63) / 64);
+ hipLaunchKernelGGL(( bn_bwd_partial_kernel), dim3(nblocks), dim3(256), 0, s, 
+      (const __bf16*)g, x, (const __bf16*)y, mean, invstd, n, h, relu ? 1 : 0,
+      partials);
+ hipLaunchKernelGGL(( bn_bwd_apply_kernel), dim3(grid_for(n * h)), dim3(256), 0, s, 
+      (const __bf16*)g, x, (const __bf16*)y, mean, invstd, gamma, partials, dx,
+      n, n, h, relu ? 1 : 0);
  hipLaunchKernelGGL(( bn_grad_affine_kernel), dim3(ceil_div(h, 256)), dim3(256), 0, s, partials, dgamma,
                                                          dbeta, h);
 }

@@ -190,13 +190,19 @@ class SAGEDeterministic(nn.Module):
                 )
             return conv(x, edge_index, edge_embeds, csr=csr, num_nodes=n)
 
+        # act16: BN emits bf16 activations so the next conv's fused QKVS GEMM
+        # (and its backward) reads/writes 16-bit streams; gate matches the
+        # linear16 gate in TransformerConv.forward_fused so dtypes line up.
+        hidden = self.bns[0].weight.shape[0] if len(self.bns) else 0
+        out16 = (fused and ops.gemm_precision() == "bf16"
+                 and hidden % 256 == 0 and ops.act16_enabled())
         for i, conv in enumerate(self.convs[:-1]):
             x = run_conv(conv, x)
             bn = self.bns[i]
             x = ops.batchnorm_relu(
                 x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
                 bn.momentum, bn.eps, self.training, fuse_relu=True,
-                comm=self._bn_comm,
+                comm=self._bn_comm, out16=out16,
             )
             if self.training and bn.track_running_stats and bn.num_batches_tracked is not None:
                 bn.num_batches_tracked += 1

@@ -80,8 +80,11 @@ class _Linear16Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b):
         m = ext()
-        bb = b if b is not None else torch.empty(0, dtype=x.dtype, device=x.device)
-        y = m.linear_fwd_bf16_o16(x, w, bb)
+        bb = b if b is not None else torch.empty(0, dtype=torch.float32, device=x.device)
+        if x.dtype == torch.bfloat16:
+            y = m.linear_fwd_a16o16(x, w, bb)   # x16 in, bf16 out
+        else:
+            y = m.linear_fwd_bf16_o16(x, w, bb)
         ctx.save_for_backward(x, w)
         ctx.has_bias = b is not None
         return y
@@ -91,8 +94,12 @@ class _Linear16Fn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         m = ext()
         g = g.contiguous()
-        dx = m.linear_dgrad16(g, w)
-        dw, db = m.linear_wgrad16(g, x, ctx.has_bias)
+        if x.dtype == torch.bfloat16:
+            dx = m.linear_dgrad16_o16(g, w)       # bf16 dx for the x16 producer
+            dw, db = m.linear_wgrad16_b16(g, x, ctx.has_bias)
+        else:
+            dx = m.linear_dgrad16(g, w)
+            dw, db = m.linear_wgrad16(g, x, ctx.has_bias)
         return dx, dw, (db if ctx.has_bias else None)
 
 
@@ -296,7 +303,7 @@ def embedding(idx, table):
 
 class _BNReLUFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu, comm):
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu, comm, out16=False):
         m = ext()
         sync = comm is not None and getattr(comm, "distributed", False) and training
         if sync:
@@ -305,12 +312,14 @@ class _BNReLUFn(torch.autograd.Function):
             partials = m.bn_stats(x)
             comm.all_reduce_(partials)
             count = int(comm.all_reduce_scalar(float(x.shape[0])))
-            y, save_mean, save_invstd = m.bn_finalize_apply(
+            fin = m.bn_finalize_apply16 if out16 else m.bn_finalize_apply
+            y, save_mean, save_invstd = fin(
                 x, partials, count, gamma, beta, running_mean, running_var,
                 momentum, eps, training, fuse_relu)
         else:
             count = x.shape[0]
-            y, save_mean, save_invstd = m.bn_relu_fwd(
+            fwd = m.bn_relu_fwd16 if out16 else m.bn_relu_fwd
+            y, save_mean, save_invstd = fwd(
                 x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu
             )
         ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
@@ -324,18 +333,22 @@ class _BNReLUFn(torch.autograd.Function):
         x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
         m = ext()
         g = g.contiguous()
+        y16 = y.dtype == torch.bfloat16
         if ctx.comm is not None:
-            partials_local = m.bn_bwd_partials(g, x, y, save_mean, save_invstd, ctx.fuse_relu)
+            bwd_partials = m.bn_bwd_partials16 if y16 else m.bn_bwd_partials
+            bwd_apply = m.bn_bwd_apply16 if y16 else m.bn_bwd_apply
+            partials_local = bwd_partials(g, x, y, save_mean, save_invstd, ctx.fuse_relu)
             partials_global = partials_local.clone()
             ctx.comm.all_reduce_(partials_global)
-            dx, dgamma, dbeta = m.bn_bwd_apply(
+            dx, dgamma, dbeta = bwd_apply(
                 g, x, y, save_mean, save_invstd, gamma,
                 partials_global, partials_local, ctx.count, ctx.fuse_relu)
         else:
-            dx, dgamma, dbeta = m.bn_relu_bwd(
+            bwd = m.bn_relu_bwd16 if y16 else m.bn_relu_bwd
+            dx, dgamma, dbeta = bwd(
                 g, x, gamma, save_mean, save_invstd, y, ctx.fuse_relu
             )
-        return dx, dgamma, dbeta, None, None, None, None, None, None, None
+        return dx, dgamma, dbeta, None, None, None, None, None, None, None, None
 
 
 class _EagerSyncBNFn(torch.autograd.Function):
@@ -394,13 +407,16 @@ def _eager_sync_batchnorm(x, gamma, beta, running_mean, running_var, momentum,
 
 
 def batchnorm_relu(x, gamma, beta, running_mean, running_var, momentum, eps,
-                   training, fuse_relu=True, comm=None):
+                   training, fuse_relu=True, comm=None, out16=False):
     """BatchNorm1d over N per channel, optional fused ReLU (model.py:101-102).
     With ``comm`` (distributed) and training=True, runs sync-BN: statistics
-    over the GLOBAL batch (exact single-process parity)."""
+    over the GLOBAL batch (exact single-process parity).  ``out16`` emits the
+    normalized activations as bf16 (act16 mode): statistics/affine math stays
+    fp32, only the activation stream narrows — the downstream QKVS GEMM and
+    its wgrad then read half the bytes."""
     if use_hip(x):
         return _BNReLUFn.apply(x, gamma, beta, running_mean, running_var,
-                               momentum, eps, training, fuse_relu, comm)
+                               momentum, eps, training, fuse_relu, comm, out16)
     if comm is not None and getattr(comm, "distributed", False) and training:
         return _eager_sync_batchnorm(x, gamma, beta, running_mean, running_var,
                                      momentum, eps, fuse_relu, comm)
