@@ -62,6 +62,14 @@ def main():
     t = bench(C.linear_wgrad16, args.iters, g16, x, True)
     row("wgrad TN g16/x32->f32", t, (m*n*2 + m*h*4 + n*h*4))
 
+    # act16-v2 production entries (bf16 A / bf16 C)
+    t = bench(C.linear_fwd_a16o16, args.iters, x16, w4, b4)
+    row("fwd NT  a16->bf16", t, (m*h*2 + n*h*4 + m*n*2))
+    t = bench(C.linear_dgrad16_o16, args.iters, g16, w4)
+    row("dgrad NN bf16->bf16", t, (m*n*2 + n*h*4 + m*h*2))
+    t = bench(C.linear_wgrad16_b16, args.iters, g16, x16, True)
+    row("wgrad TN a16b16->f32", t, (m*n*2 + m*h*2 + n*h*4))
+
     # rocBLAS reference on bf16
     wt16 = w4.bfloat16()
     t = bench(lambda: x16 @ wt16.t(), args.iters)

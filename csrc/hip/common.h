@@ -23,6 +23,14 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
   return v;
 }
 
+// reduction over an aligned W-lane sub-group of the wave (W power of two)
+template <int W>
+__device__ __forceinline__ float subwave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = W / 2; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
 __device__ __forceinline__ float wave_reduce_max(float v) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));

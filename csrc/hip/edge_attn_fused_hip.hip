@@ -24,6 +24,7 @@
 
 #include "common.h"
 #include <cmath>
+#include <cstdlib>
 
 #define WAVES_PER_BLOCK 4
 
@@ -121,7 +122,11 @@ struct Slice {
   }
 };
 
-template <int VPT, bool VEC, typename QT = float>
+// LPR = lanes per row: CSR rows average degree ~1.3 on PERT graphs, so a
+// full 64-lane wave per row is latency-bound with most lanes idle between
+// the few row loads.  LPR=16 packs 4 rows into a wave (each sub-group owns
+// h/LPR contiguous columns); reductions become sub-wave shfl_xor trees.
+template <int VPT, bool VEC, typename QT = float, int LPR = PERTGNN_WAVE>
 __global__ void edge_attn_fused_fwd_kernel(
     const QT* __restrict__ qkvs,  // [N, 4h]
     const float* __restrict__ pifc,  // [Vi, h]
@@ -131,9 +136,13 @@ __global__ void edge_attn_fused_fwd_kernel(
     float* __restrict__ out, float* __restrict__ alpha, int n, int h,
     float scale) {
   using S = Slice<VPT, VEC>;
+  static_assert(VEC || LPR == PERTGNN_WAVE, "sub-wave rows need VEC layout");
+  constexpr int RPW = PERTGNN_WAVE / LPR;
   const int wid = threadIdx.x / PERTGNN_WAVE;
-  const int lane = threadIdx.x % PERTGNN_WAVE;
-  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  const int lane0 = threadIdx.x % PERTGNN_WAVE;
+  const int sub = lane0 / LPR;
+  const int lane = lane0 % LPR;
+  const int row = (blockIdx.x * WAVES_PER_BLOCK + wid) * RPW + sub;
   if (row >= n) return;
   const long ld = 4L * h;
 
@@ -154,8 +163,8 @@ __global__ void edge_attn_fused_fwd_kernel(
     float part = 0.f;
 #pragma unroll
     for (int j = 0; j < VPT; ++j) part += qr[j] * (ke[j] + ec[j]);
-    const float logit = wave_reduce_sum(part) * scale;
-    if (lane == (p - beg) % PERTGNN_WAVE) alpha[p] = logit;
+    const float logit = subwave_reduce_sum<LPR>(part) * scale;
+    if (lane == (p - beg) % LPR) alpha[p] = logit;
     const float m_new = fmaxf(m, logit);
     const float corr = __expf(m - m_new);
     const float pexp = __expf(logit - m_new);
@@ -174,11 +183,12 @@ __global__ void edge_attn_fused_fwd_kernel(
 #pragma unroll
   for (int j = 0; j < VPT; ++j) res[j] = acc[j] * inv_s + sk[j];
   S::store(&out[(long)row * h], lane, h, res);
-  for (int p = beg + lane; p < end; p += PERTGNN_WAVE)
+  for (int p = beg + lane; p < end; p += LPR)
     alpha[p] = __expf(alpha[p] - m) * inv_s;
 }
 
-template <int VPT, bool VEC, typename QT = float, typename ET = float>
+template <int VPT, bool VEC, typename QT = float, typename ET = float,
+          int LPR = PERTGNN_WAVE>
 __global__ void edge_attn_fused_bwd_row_kernel(
     const float* __restrict__ g, const QT* __restrict__ qkvs,
     const float* __restrict__ pifc, const float* __restrict__ prpc,
@@ -188,9 +198,12 @@ __global__ void edge_attn_fused_bwd_row_kernel(
     ET* __restrict__ dev, float* __restrict__ dal, int n, int h,
     float scale) {
   using S = Slice<VPT, VEC>;
+  static_assert(VEC || LPR == PERTGNN_WAVE, "sub-wave rows need VEC layout");
+  constexpr int RPW = PERTGNN_WAVE / LPR;
   const int wid = threadIdx.x / PERTGNN_WAVE;
-  const int lane = threadIdx.x % PERTGNN_WAVE;
-  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  const int lane0 = threadIdx.x % PERTGNN_WAVE;
+  const int lane = lane0 % LPR;
+  const int row = (blockIdx.x * WAVES_PER_BLOCK + wid) * RPW + lane0 / LPR;
   if (row >= n) return;
   const long ld = 4L * h;
 
@@ -212,7 +225,7 @@ __global__ void edge_attn_fused_bwd_row_kernel(
     float part = 0.f;
 #pragma unroll
     for (int j = 0; j < VPT; ++j) part += gr[j] * (ve[j] + ec[j]);
-    const float dalpha = wave_reduce_sum(part);
+    const float dalpha = subwave_reduce_sum<LPR>(part);
     sdot += alpha[p] * dalpha;
     if (lane == 0) dal[p] = dalpha;
   }
@@ -220,8 +233,7 @@ __global__ void edge_attn_fused_bwd_row_kernel(
     const long src = csr_src[p];
     const long a0 = ea[(long)p * astride];
     const long a1 = ea[(long)p * astride + 1];
-    const float dalpha_l0 = (lane == 0) ? dal[p] : 0.f;
-    const float dalpha = __shfl(dalpha_l0, 0, PERTGNN_WAVE);
+    const float dalpha = dal[p];  // broadcast load (sub-group shares p)
     const float a = alpha[p];
     const float dl = a * (dalpha - sdot) * scale;
     float ec[VPT], ke[VPT], dekv[VPT], devv[VPT];
@@ -240,15 +252,19 @@ __global__ void edge_attn_fused_bwd_row_kernel(
   S::store(&dqkvs[row * ld + 3 * h], lane, h, gr);     // dskip = g
 }
 
-template <int VPT, bool VEC, typename QT = float, typename ET = float>
+template <int VPT, bool VEC, typename QT = float, typename ET = float,
+          int LPR = PERTGNN_WAVE>
 __global__ void edge_attn_fused_bwd_col_kernel(
     const ET* __restrict__ dek, const ET* __restrict__ dev,
     const int* __restrict__ col_ptr, const int* __restrict__ csc_eid,
     QT* __restrict__ dqkvs, ET* __restrict__ de, int n, int h) {
   using S = Slice<VPT, VEC>;
+  static_assert(VEC || LPR == PERTGNN_WAVE, "sub-wave rows need VEC layout");
+  constexpr int RPW = PERTGNN_WAVE / LPR;
   const int wid = threadIdx.x / PERTGNN_WAVE;
-  const int lane = threadIdx.x % PERTGNN_WAVE;
-  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  const int lane0 = threadIdx.x % PERTGNN_WAVE;
+  const int lane = lane0 % LPR;
+  const int row = (blockIdx.x * WAVES_PER_BLOCK + wid) * RPW + lane0 / LPR;
   if (row >= n) return;
   const long ld = 4L * h;
   float ka[VPT], va[VPT];
@@ -347,6 +363,19 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
 // bf16-qkvs launchers (activation-resident bf16 mode; requires h % 256 == 0)
 // ---------------------------------------------------------------------------
 
+// sub-wave row packing: PERT CSR rows average degree ~1.3, so a full wave
+// per row is latency-bound with idle lanes.  LPR=32 (two rows per wave,
+// 8 cols/lane at H=256) measured best: 14.87 vs 15.33 ms/step at LPR=64;
+// LPR=16 loses it back to register pressure.  PERTGNN_ATTN_LPR overrides.
+static int attn_lpr() {
+  static int v = [] {
+    const char* e = getenv("PERTGNN_ATTN_LPR");
+    const int x = e ? atoi(e) : 32;
+    return (x == 16 || x == 32 || x == 64) ? x : 32;
+  }();
+  return v;
+}
+
 void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
                                   const float* prpc, const long* ea,
                                   int astride, const int* row_ptr,
@@ -355,19 +384,31 @@ void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
   const __bf16* qkvs = (const __bf16*)qkvs_v;
   if (n == 0) return;
   const float scale = 1.f / std::sqrt((float)h);
-  const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
-  const int vpt = h / PERTGNN_WAVE;
-  if (vpt == 4)
-   hipLaunchKernelGGL(( edge_attn_fused_fwd_kernel<4, true, __bf16>), dim3(grid), dim3(block), 0, stream, 
-        qkvs, pifc, prpc, ea, astride, row_ptr, csr_src, out, alpha, n, h,
-        scale);
-  else if (vpt == 8)
-   hipLaunchKernelGGL(( edge_attn_fused_fwd_kernel<8, true, __bf16>), dim3(grid), dim3(block), 0, stream, 
-        qkvs, pifc, prpc, ea, astride, row_ptr, csr_src, out, alpha, n, h,
-        scale);
-  else
+  const int lpr = attn_lpr();
+  const int rpb = WAVES_PER_BLOCK * (PERTGNN_WAVE / lpr);
+  const dim3 grid(ceil_div(n, rpb));
+#define FWD16(VPT, LPR)                                                        \
+ hipLaunchKernelGGL(( edge_attn_fused_fwd_kernel<VPT, true, __bf16, LPR>)                           \
+      , dim3(grid), dim3(block), 0, stream, qkvs, pifc, prpc, ea, astride, row_ptr,     \
+                                   csr_src, out, alpha, n, h, scale)
+  if (h == 256) {
+    if (lpr == 16) FWD16(16, 16);
+    else if (lpr == 32) FWD16(8, 32);
+    else FWD16(4, 64);
+  } else if (h == 512) {
+    if (lpr == 16 || lpr == 32) FWD16(16, 32);
+    else FWD16(8, 64);
+  } else if (h % 256 == 0 && h / PERTGNN_WAVE <= 32) {
+    const int vpt = h / PERTGNN_WAVE;
+    if (vpt == 4) FWD16(4, 64);
+    else if (vpt == 8) FWD16(8, 64);
+    else if (vpt == 16) FWD16(16, 64);
+    else FWD16(32, 64);
+  } else {
     abort();
+  }
+#undef FWD16
 }
 
 void launch_edge_attn_fused_bwd16(const float* g, const void* qkvs_v,
@@ -386,26 +427,35 @@ void launch_edge_attn_fused_bwd16(const float* g, const void* qkvs_v,
   __bf16* dev = (__bf16*)dev_v;
   if (n == 0) return;
   const float scale = 1.f / std::sqrt((float)h);
-  const dim3 grid(ceil_div(n, WAVES_PER_BLOCK));
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
-  const int vpt = h / PERTGNN_WAVE;
-  if (vpt == 4) {
-   hipLaunchKernelGGL(( edge_attn_fused_bwd_row_kernel<4, true, __bf16, __bf16>)
-        , dim3(grid), dim3(block), 0, stream, g, qkvs, pifc, prpc, ea, astride, alpha,
-                                     row_ptr, csr_src, dqkvs, dek, dev, dal,
-                                     n, h, scale);
-   hipLaunchKernelGGL(( edge_attn_fused_bwd_col_kernel<4, true, __bf16, __bf16>)
-        , dim3(grid), dim3(block), 0, stream, dek, dev, col_ptr, csc_eid, dqkvs, de, n,
-                                     h);
-  } else if (vpt == 8) {
-   hipLaunchKernelGGL(( edge_attn_fused_bwd_row_kernel<8, true, __bf16, __bf16>)
-        , dim3(grid), dim3(block), 0, stream, g, qkvs, pifc, prpc, ea, astride, alpha,
-                                     row_ptr, csr_src, dqkvs, dek, dev, dal,
-                                     n, h, scale);
-   hipLaunchKernelGGL(( edge_attn_fused_bwd_col_kernel<8, true, __bf16, __bf16>)
-        , dim3(grid), dim3(block), 0, stream, dek, dev, col_ptr, csc_eid, dqkvs, de, n,
-                                     h);
+  const int lpr = attn_lpr();
+  const int rpb = WAVES_PER_BLOCK * (PERTGNN_WAVE / lpr);
+  const dim3 grid(ceil_div(n, rpb));
+#define BWD16(VPT, LPR)                                                        \
+  do {                                                                         \
+   hipLaunchKernelGGL(( edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, __bf16, LPR>)             \
+        , dim3(grid), dim3(block), 0, stream, g, qkvs, pifc, prpc, ea, astride, alpha,  \
+                                     row_ptr, csr_src, dqkvs, dek, dev, dal,   \
+                                     n, h, scale);                             \
+   hipLaunchKernelGGL(( edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR>)             \
+        , dim3(grid), dim3(block), 0, stream, dek, dev, col_ptr, csc_eid, dqkvs, de, n, \
+                                     h);                                       \
+  } while (0)
+  if (h == 256) {
+    if (lpr == 16) BWD16(16, 16);
+    else if (lpr == 32) BWD16(8, 32);
+    else BWD16(4, 64);
+  } else if (h == 512) {
+    if (lpr == 16 || lpr == 32) BWD16(16, 32);
+    else BWD16(8, 64);
+  } else if (h % 256 == 0 && h / PERTGNN_WAVE <= 32) {
+    const int vpt = h / PERTGNN_WAVE;
+    if (vpt == 4) BWD16(4, 64);
+    else if (vpt == 8) BWD16(8, 64);
+    else if (vpt == 16) BWD16(16, 64);
+    else BWD16(32, 64);
   } else {
     abort();
   }
+#undef BWD16
 }
