@@ -73,6 +73,14 @@ class TransformerConv(nn.Module):
                         self.lin_value.weight, self.lin_skip.weight], dim=0)
         b4 = torch.cat([self.lin_query.bias, self.lin_key.bias,
                         self.lin_value.bias, self.lin_skip.bias], dim=0)
+        # pad an odd contraction dim (layer 1: K = 9 + H + embeds) to a
+        # multiple of 8 so the GEMM staging takes the vectorized interior
+        # path (zero columns are exact; autograd slices the grads back)
+        kdim = x.shape[1]
+        if kdim % 8:
+            pad = 8 - kdim % 8
+            x = torch.nn.functional.pad(x, (0, pad))
+            w4 = torch.nn.functional.pad(w4, (0, pad))
         if (ops.gemm_precision() == "bf16" and h % 256 == 0
                 and ops.act16_enabled()):
             qkvs = ops.linear16(x, w4, b4)
