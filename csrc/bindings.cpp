@@ -54,23 +54,24 @@ void launch_bn_bwd_apply_only(const float*, const float*, const float*,
                               const float*, long, float*, long, int, bool,
                               hipStream_t);
 void launch_bn_grad_affine(const float*, float*, float*, int, hipStream_t);
-// act16 variants (bf16 activation / gradient streams, fp32 statistics)
-void launch_bn_fwd16(const float*, const float*, const float*, float*, float*,
+// act16 variants (bf16 x/y/g/dx streams, fp32 statistics)
+void launch_bn_stats_only16(const void*, long, int, float*, hipStream_t);
+void launch_bn_fwd16(const void*, const float*, const float*, float*, float*,
                      float*, float*, float*, void*, long, int, float, float,
                      bool, bool, hipStream_t);
-void launch_bn_finalize_apply16(const float*, const float*, long, const float*,
+void launch_bn_finalize_apply16(const void*, const float*, long, const float*,
                                 const float*, float*, float*, float*, float*,
                                 void*, long, int, float, float, bool, bool,
                                 hipStream_t);
-void launch_bn_bwd_partials_only16(const void*, const float*, const void*,
+void launch_bn_bwd_partials_only16(const void*, const void*, const void*,
                                    const float*, const float*, long, int, bool,
                                    float*, hipStream_t);
-void launch_bn_bwd_apply_only16(const void*, const float*, const void*,
+void launch_bn_bwd_apply_only16(const void*, const void*, const void*,
                                 const float*, const float*, const float*,
-                                const float*, long, float*, long, int, bool,
+                                const float*, long, void*, long, int, bool,
                                 hipStream_t);
-void launch_bn_bwd16(const void*, const float*, const void*, const float*,
-                     const float*, const float*, float*, float*, float*,
+void launch_bn_bwd16(const void*, const void*, const void*, const float*,
+                     const float*, const float*, float*, void*, float*,
                      float*, long, int, bool, hipStream_t);
 void launch_seg_pool_fwd16(const void*, const float*, const float*,
                            const int*, float*, float*, int, int, int,
@@ -127,10 +128,10 @@ void launch_edge_attn_fused_bwd(const float*, const float*, const float*,
                                 float*, int, int, long, hipStream_t);
 void launch_edge_attn_fused_fwd16(const void*, const float*, const float*,
                                   const long*, int, const int*, const int*,
-                                  float*, float*, int, int, hipStream_t);
-void launch_edge_attn_fused_bwd16(const float*, const void*, const float*,
-                                  const float*, const long*, int,
-                                  const float*, const int*, const int*,
+                                  void*, int, float*, int, int, hipStream_t);
+void launch_edge_attn_fused_bwd16(const void*, int, const void*,
+                                  const float*, const float*, const long*,
+                                  int, const float*, const int*, const int*,
                                   const int*, const int*, void*, void*, void*,
                                   void*, float*, int, int, long, hipStream_t);
 void launch_vocab_scatter_dual16(const void*, const long*, int, float*,
@@ -536,20 +537,24 @@ torch::Tensor embed_grouped_scatter(torch::Tensor g, torch::Tensor order,
 
 std::vector<torch::Tensor> edge_attn_fused_fwd(
     torch::Tensor qkvs, torch::Tensor pifc, torch::Tensor prpc,
-    torch::Tensor ea, torch::Tensor row_ptr, torch::Tensor csr_src) {
+    torch::Tensor ea, torch::Tensor row_ptr, torch::Tensor csr_src,
+    bool out16) {
   CHECK_IN(qkvs); CHECK_IN(pifc); CHECK_IN(prpc); CHECK_IN(ea);
   const int n = qkvs.size(0);
   const int h = qkvs.size(1) / 4;
   TORCH_CHECK(h <= 512, "H must be <= 512");
   auto fopt = qkvs.options().dtype(torch::kFloat32);
-  auto out = torch::empty({n, h}, fopt);
+  const bool b16 = qkvs.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(!out16 || b16, "out16 requires bf16 qkvs");
+  auto out = torch::empty({n, h}, out16 ? qkvs.options() : fopt);
   auto alpha = torch::empty({ea.size(0)}, fopt);
-  if (qkvs.scalar_type() == torch::kBFloat16) {
+  if (b16) {
     launch_edge_attn_fused_fwd16(
         qkvs.data_ptr(), pifc.data_ptr<float>(),
         prpc.data_ptr<float>(), ea.data_ptr<long>(), (int)ea.size(1),
         row_ptr.data_ptr<int>(), csr_src.data_ptr<int>(),
-        out.data_ptr<float>(), alpha.data_ptr<float>(), n, h, cur_stream());
+        out.data_ptr(), out16 ? 1 : 0, alpha.data_ptr<float>(), n, h,
+        cur_stream());
   } else {
     launch_edge_attn_fused_fwd(
         qkvs.data_ptr<float>(), pifc.data_ptr<float>(), prpc.data_ptr<float>(),
@@ -578,8 +583,9 @@ std::vector<torch::Tensor> edge_attn_fused_bwd(
   auto dev = torch::empty({ne, h}, eopt);
   if (b16) {
     auto dal = torch::empty({ne}, fopt);
+    const int g16 = g.scalar_type() == torch::kBFloat16 ? 1 : 0;
     launch_edge_attn_fused_bwd16(
-        g.data_ptr<float>(), qkvs.data_ptr(),
+        g.data_ptr(), g16, qkvs.data_ptr(),
         pifc.data_ptr<float>(), prpc.data_ptr<float>(), ea.data_ptr<long>(),
         (int)ea.size(1), alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
         csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
@@ -618,9 +624,14 @@ torch::Tensor vocab_scatter(torch::Tensor g, torch::Tensor idx, int64_t rows,
 
 torch::Tensor bn_stats(torch::Tensor x) {
   CHECK_IN(x);
-  auto partials = torch::empty({2 * x.size(1)}, x.options());
-  launch_bn_stats_only(x.data_ptr<float>(), x.size(0), x.size(1),
-                       partials.data_ptr<float>(), cur_stream());
+  auto partials = torch::empty({2 * x.size(1)},
+                               x.options().dtype(torch::kFloat32));
+  if (x.scalar_type() == torch::kBFloat16)
+    launch_bn_stats_only16(x.data_ptr(), x.size(0), x.size(1),
+                           partials.data_ptr<float>(), cur_stream());
+  else
+    launch_bn_stats_only(x.data_ptr<float>(), x.size(0), x.size(1),
+                         partials.data_ptr<float>(), cur_stream());
   return partials;
 }
 
@@ -811,11 +822,12 @@ std::vector<torch::Tensor> bn_relu_fwd16(torch::Tensor x, torch::Tensor gamma,
   CHECK_IN(x); CHECK_IN(gamma); CHECK_IN(beta);
   const long n = x.size(0);
   const int h = x.size(1);
+  auto fopt = x.options().dtype(torch::kFloat32);
   auto y = torch::empty({n, h}, x.options().dtype(torch::kBFloat16));
-  auto mean = torch::empty({h}, x.options());
-  auto invstd = torch::empty({h}, x.options());
-  auto partials = torch::empty({2 * h}, x.options());
-  launch_bn_fwd16(x.data_ptr<float>(), gamma.data_ptr<float>(),
+  auto mean = torch::empty({h}, fopt);
+  auto invstd = torch::empty({h}, fopt);
+  auto partials = torch::empty({2 * h}, fopt);
+  launch_bn_fwd16(x.data_ptr(), gamma.data_ptr<float>(),
                   beta.data_ptr<float>(), running_mean.data_ptr<float>(),
                   running_var.data_ptr<float>(), mean.data_ptr<float>(),
                   invstd.data_ptr<float>(), partials.data_ptr<float>(),
@@ -832,14 +844,15 @@ std::vector<torch::Tensor> bn_relu_bwd16(torch::Tensor g, torch::Tensor x,
   CHECK_IN(g); CHECK_IN(x);
   const long n = x.size(0);
   const int h = x.size(1);
+  auto fopt = x.options().dtype(torch::kFloat32);
   auto dx = torch::empty_like(x);
-  auto dgamma = torch::empty({h}, x.options());
-  auto dbeta = torch::empty({h}, x.options());
-  auto partials = torch::empty({2 * h}, x.options());
-  launch_bn_bwd16(g.data_ptr(), x.data_ptr<float>(), y.data_ptr(),
+  auto dgamma = torch::empty({h}, fopt);
+  auto dbeta = torch::empty({h}, fopt);
+  auto partials = torch::empty({2 * h}, fopt);
+  launch_bn_bwd16(g.data_ptr(), x.data_ptr(), y.data_ptr(),
                   mean.data_ptr<float>(), invstd.data_ptr<float>(),
                   gamma.data_ptr<float>(), partials.data_ptr<float>(),
-                  dx.data_ptr<float>(), dgamma.data_ptr<float>(),
+                  dx.data_ptr(), dgamma.data_ptr<float>(),
                   dbeta.data_ptr<float>(), n, h, relu, cur_stream());
   return {dx, dgamma, dbeta};
 }
@@ -851,11 +864,12 @@ std::vector<torch::Tensor> bn_finalize_apply16(
     bool relu) {
   CHECK_IN(x); CHECK_IN(partials);
   const int h = x.size(1);
+  auto fopt = x.options().dtype(torch::kFloat32);
   auto y = torch::empty({x.size(0), h}, x.options().dtype(torch::kBFloat16));
-  auto mean = torch::empty({h}, x.options());
-  auto invstd = torch::empty({h}, x.options());
+  auto mean = torch::empty({h}, fopt);
+  auto invstd = torch::empty({h}, fopt);
   launch_bn_finalize_apply16(
-      x.data_ptr<float>(), partials.data_ptr<float>(), count,
+      x.data_ptr(), partials.data_ptr<float>(), count,
       gamma.data_ptr<float>(), beta.data_ptr<float>(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       mean.data_ptr<float>(), invstd.data_ptr<float>(), y.data_ptr(),
@@ -868,8 +882,9 @@ torch::Tensor bn_bwd_partials16(torch::Tensor g, torch::Tensor x,
                                 torch::Tensor y, torch::Tensor mean,
                                 torch::Tensor invstd, bool relu) {
   CHECK_IN(g); CHECK_IN(x);
-  auto partials = torch::empty({2 * x.size(1)}, x.options());
-  launch_bn_bwd_partials_only16(g.data_ptr(), x.data_ptr<float>(),
+  auto partials = torch::empty({2 * x.size(1)},
+                               x.options().dtype(torch::kFloat32));
+  launch_bn_bwd_partials_only16(g.data_ptr(), x.data_ptr(),
                                 y.data_ptr(), mean.data_ptr<float>(),
                                 invstd.data_ptr<float>(), x.size(0), x.size(1),
                                 relu, partials.data_ptr<float>(),
@@ -883,14 +898,15 @@ std::vector<torch::Tensor> bn_bwd_apply16(
     torch::Tensor partials_local, int64_t count, bool relu) {
   CHECK_IN(g); CHECK_IN(x);
   const int h = x.size(1);
+  auto fopt = x.options().dtype(torch::kFloat32);
   auto dx = torch::empty_like(x);
-  auto dgamma = torch::empty({h}, x.options());
-  auto dbeta = torch::empty({h}, x.options());
-  launch_bn_bwd_apply_only16(g.data_ptr(), x.data_ptr<float>(), y.data_ptr(),
+  auto dgamma = torch::empty({h}, fopt);
+  auto dbeta = torch::empty({h}, fopt);
+  launch_bn_bwd_apply_only16(g.data_ptr(), x.data_ptr(), y.data_ptr(),
                              mean.data_ptr<float>(), invstd.data_ptr<float>(),
                              gamma.data_ptr<float>(),
                              partials_global.data_ptr<float>(), count,
-                             dx.data_ptr<float>(), x.size(0), h, relu,
+                             dx.data_ptr(), x.size(0), h, relu,
                              cur_stream());
   launch_bn_grad_affine(partials_local.data_ptr<float>(),
                         dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), h,

@@ -125,14 +125,15 @@ struct Slice {
 // full 64-lane wave per row is latency-bound with most lanes idle between
 // the few row loads.  LPR=16 packs 4 rows into a wave (each sub-group owns
 // h/LPR contiguous columns); reductions become sub-wave shfl_xor trees.
-template <int VPT, bool VEC, typename QT = float, int LPR = PERTGNN_WAVE>
+template <int VPT, bool VEC, typename QT = float, int LPR = PERTGNN_WAVE,
+          typename TO = float>
 __global__ void edge_attn_fused_fwd_kernel(
     const QT* __restrict__ qkvs,  // [N, 4h]
     const float* __restrict__ pifc,  // [Vi, h]
     const float* __restrict__ prpc,  // [Vr, h]
     const long* __restrict__ ea, int astride,
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
-    float* __restrict__ out, float* __restrict__ alpha, int n, int h,
+    TO* __restrict__ out, float* __restrict__ alpha, int n, int h,
     float scale) {
   using S = Slice<VPT, VEC>;
   static_assert(VEC || LPR == PERTGNN_WAVE, "sub-wave rows need VEC layout");
@@ -187,9 +188,9 @@ __global__ void edge_attn_fused_fwd_kernel(
 }
 
 template <int VPT, bool VEC, typename QT = float, typename ET = float,
-          int LPR = PERTGNN_WAVE>
+          int LPR = PERTGNN_WAVE, typename TG = float>
 __global__ void edge_attn_fused_bwd_row_kernel(
-    const float* __restrict__ g, const QT* __restrict__ qkvs,
+    const TG* __restrict__ g, const QT* __restrict__ qkvs,
     const float* __restrict__ pifc, const float* __restrict__ prpc,
     const long* __restrict__ ea, int astride, const float* __restrict__ alpha,
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
@@ -378,8 +379,9 @@ static int attn_lpr() {
 void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
                                   const float* prpc, const long* ea,
                                   int astride, const int* row_ptr,
-                                  const int* csr_src, float* out, float* alpha,
-                                  int n, int h, hipStream_t stream) {
+                                  const int* csr_src, void* out_v, int out16,
+                                  float* alpha, int n, int h,
+                                  hipStream_t stream) {
   const __bf16* qkvs = (const __bf16*)qkvs_v;
   if (n == 0) return;
   const float scale = 1.f / std::sqrt((float)h);
@@ -388,9 +390,18 @@ void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
   const int rpb = WAVES_PER_BLOCK * (PERTGNN_WAVE / lpr);
   const dim3 grid(ceil_div(n, rpb));
 #define FWD16(VPT, LPR)                                                        \
-  edge_attn_fused_fwd_kernel<VPT, true, __bf16, LPR>                           \
-      <<<grid, block, 0, stream>>>(qkvs, pifc, prpc, ea, astride, row_ptr,     \
-                                   csr_src, out, alpha, n, h, scale)
+  do {                                                                         \
+    if (out16)                                                                 \
+      edge_attn_fused_fwd_kernel<VPT, true, __bf16, LPR, __bf16>               \
+          <<<grid, block, 0, stream>>>(qkvs, pifc, prpc, ea, astride, row_ptr, \
+                                       csr_src, (__bf16*)out_v, alpha, n, h,   \
+                                       scale);                                 \
+    else                                                                       \
+      edge_attn_fused_fwd_kernel<VPT, true, __bf16, LPR, float>                \
+          <<<grid, block, 0, stream>>>(qkvs, pifc, prpc, ea, astride, row_ptr, \
+                                       csr_src, (float*)out_v, alpha, n, h,    \
+                                       scale);                                 \
+  } while (0)
   if (h == 256) {
     if (lpr == 16) FWD16(16, 16);
     else if (lpr == 32) FWD16(8, 32);
@@ -410,7 +421,8 @@ void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
 #undef FWD16
 }
 
-void launch_edge_attn_fused_bwd16(const float* g, const void* qkvs_v,
+void launch_edge_attn_fused_bwd16(const void* g_v, int g16,
+                                  const void* qkvs_v,
                                   const float* pifc, const float* prpc,
                                   const long* ea, int astride,
                                   const float* alpha, const int* row_ptr,
@@ -432,10 +444,16 @@ void launch_edge_attn_fused_bwd16(const float* g, const void* qkvs_v,
   const dim3 grid(ceil_div(n, rpb));
 #define BWD16(VPT, LPR)                                                        \
   do {                                                                         \
-    edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, __bf16, LPR>             \
-        <<<grid, block, 0, stream>>>(g, qkvs, pifc, prpc, ea, astride, alpha,  \
-                                     row_ptr, csr_src, dqkvs, dek, dev, dal,   \
-                                     n, h, scale);                             \
+    if (g16)                                                                   \
+      edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, __bf16, LPR, __bf16>   \
+          <<<grid, block, 0, stream>>>((const __bf16*)g_v, qkvs, pifc, prpc,   \
+                                       ea, astride, alpha, row_ptr, csr_src,   \
+                                       dqkvs, dek, dev, dal, n, h, scale);     \
+    else                                                                       \
+      edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, __bf16, LPR, float>    \
+          <<<grid, block, 0, stream>>>((const float*)g_v, qkvs, pifc, prpc,    \
+                                       ea, astride, alpha, row_ptr, csr_src,   \
+                                       dqkvs, dek, dev, dal, n, h, scale);     \
     edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR>             \
         <<<grid, block, 0, stream>>>(dek, dev, col_ptr, csc_eid, dqkvs, de, n, \
                                      h);                                       \

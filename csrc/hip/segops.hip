@@ -733,7 +733,8 @@ void launch_gather_rows(const long* idx, const float* table, float* out,
 // stage A: per-block partial sum/sumsq over a row range; thread t owns
 // channels t and t+256 in REGISTERS (H <= 512), one atomicAdd per channel
 // per block at the end — coalesced loads, no LDS traffic.
-__global__ void bn_stats_partial_kernel(const float* __restrict__ x, long n,
+template <typename TX = float>
+__global__ void bn_stats_partial_kernel(const TX* __restrict__ x, long n,
                                         int h, float* __restrict__ partials) {
   const int c0 = threadIdx.x;
   const int c1 = threadIdx.x + 256;
@@ -804,8 +805,8 @@ __global__ void bn_finalize_kernel(const float* __restrict__ partials, long n,
 // TY = output/activation dtype: fp32 in exact mode, bf16 in the act16 mode
 // (standard mixed-precision BN: statistics and normalization math stay fp32,
 // only the activation stream is 16-bit).
-template <typename TY>
-__global__ void bn_apply_kernel(const float* __restrict__ x,
+template <typename TY, typename TX = float>
+__global__ void bn_apply_kernel(const TX* __restrict__ x,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
                                 const float* __restrict__ gamma,
@@ -821,7 +822,7 @@ __global__ void bn_apply_kernel(const float* __restrict__ x,
       const long t = q * 4;
       const int c = (int)(t % h);
       float xv[4];
-      *reinterpret_cast<bnf4*>(xv) = *reinterpret_cast<const bnf4*>(&x[t]);
+      vs_ld4(&x[t], xv);
       struct alignas(4 * sizeof(TY)) TY4 { TY v[4]; };
       TY4 o;
 #pragma unroll
@@ -840,16 +841,16 @@ __global__ void bn_apply_kernel(const float* __restrict__ x,
   const long stride = (long)gridDim.x * blockDim.x;
   for (long t = i0; t < numel; t += stride) {
     const int c = (int)(t % h);
-    float v = (x[t] - mean[c]) * invstd[c] * gamma[c] + beta[c];
+    float v = ((float)x[t] - mean[c]) * invstd[c] * gamma[c] + beta[c];
     if (relu) v = fmaxf(v, 0.f);
     y[t] = (TY)v;
   }
 }
 
 // backward stage A: per-channel sums of gm and gm*xhat (gm = relu-masked g)
-template <typename TG, typename TY>
+template <typename TG, typename TY, typename TX = float>
 __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
-                                      const float* __restrict__ x,
+                                      const TX* __restrict__ x,
                                       const TY* __restrict__ y,
                                       const float* __restrict__ mean,
                                       const float* __restrict__ invstd, long n,
@@ -873,7 +874,7 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
       for (int u = 0; u < 4; ++u) {
         gv[u] = (float)g[(r + u) * h + c0];
         yv[u] = (float)y[(r + u) * h + c0];
-        xv[u] = x[(r + u) * h + c0];
+        xv[u] = (float)x[(r + u) * h + c0];
       }
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
@@ -888,7 +889,7 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
       for (int u = 0; u < 4; ++u) {
         gv[u] = (float)g[(r + u) * h + c1];
         yv[u] = (float)y[(r + u) * h + c1];
-        xv[u] = x[(r + u) * h + c1];
+        xv[u] = (float)x[(r + u) * h + c1];
       }
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
@@ -903,13 +904,13 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
       float gm = (float)g[r * h + c0];
       if (relu && (float)y[r * h + c0] <= 0.f) gm = 0.f;
       s0 += gm;
-      q0 += gm * (x[r * h + c0] - m0) * i0;
+      q0 += gm * ((float)x[r * h + c0] - m0) * i0;
     }
     if (c1 < h) {
       float gm = (float)g[r * h + c1];
       if (relu && (float)y[r * h + c1] <= 0.f) gm = 0.f;
       s1 += gm;
-      q1 += gm * (x[r * h + c1] - m1) * i1;
+      q1 += gm * ((float)x[r * h + c1] - m1) * i1;
     }
   }
   if (c0 < h && r0 < r1) {
@@ -925,12 +926,12 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
 // backward stage B: dx = gamma*invstd*(gm - sum_gm/n - xhat*sum_gmx/n)
 // 4-wide vectorized (16-bit scalar loads left the kernel latency-bound:
 // 163 us vs 133 us fp32 at 165k x 256 — 4x-unrolled loads recover it)
-template <typename TG, typename TY>
+template <typename TG, typename TY, typename TX = float>
 __global__ void bn_bwd_apply_kernel(
-    const TG* __restrict__ g, const float* __restrict__ x,
+    const TG* __restrict__ g, const TX* __restrict__ x,
     const TY* __restrict__ y, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ partials, float* __restrict__ dx, long n,
+    const float* __restrict__ partials, TX* __restrict__ dx, long n,
     long count, int h, int relu) {
   typedef __attribute__((ext_vector_type(4))) float bnf4;
   const float invn = 1.f / count;
@@ -948,17 +949,19 @@ __global__ void bn_bwd_apply_kernel(
       if (relu)
 #pragma unroll
         for (int u = 0; u < 4; ++u) yv[u] = (float)y[t + u];
-      *reinterpret_cast<bnf4*>(xv) = *reinterpret_cast<const bnf4*>(&x[t]);
-      bnf4 o;
+      vs_ld4(&x[t], xv);
+      struct alignas(4 * sizeof(TX)) TX4 { TX v[4]; };
+      TX4 o;
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         float gm = gv[u];
         if (relu && yv[u] <= 0.f) gm = 0.f;
         const float xhat = (xv[u] - mean[c + u]) * invstd[c + u];
-        o[u] = gamma[c + u] * invstd[c + u] *
-               (gm - partials[c + u] * invn - xhat * partials[h + c + u] * invn);
+        o.v[u] = (TX)(gamma[c + u] * invstd[c + u] *
+                      (gm - partials[c + u] * invn -
+                       xhat * partials[h + c + u] * invn));
       }
-      *reinterpret_cast<bnf4*>(&dx[t]) = o;
+      *reinterpret_cast<TX4*>(&dx[t]) = o;
     }
     (void)hq;
     return;
@@ -970,9 +973,9 @@ __global__ void bn_bwd_apply_kernel(
     const int c = (int)(t % h);
     float gm = (float)g[t];
     if (relu && (float)y[t] <= 0.f) gm = 0.f;
-    const float xhat = (x[t] - mean[c]) * invstd[c];
-    dx[t] = gamma[c] * invstd[c] *
-            (gm - partials[c] * invn - xhat * partials[h + c] * invn);
+    const float xhat = ((float)x[t] - mean[c]) * invstd[c];
+    dx[t] = (TX)(gamma[c] * invstd[c] *
+                 (gm - partials[c] * invn - xhat * partials[h + c] * invn));
   }
 }
 
@@ -1100,11 +1103,20 @@ void launch_bn_bwd(const float* g, const float* x, const float* y,
                                                          dbeta, h);
 }
 
-// --- act16 launchers: bf16 activation stream (y) and bf16 incoming grad
-// (g), fp32 statistics/affine math and fp32 dx out (feeds the attention
-// backward).  Same kernels, TY/TG = __bf16. ---
+// --- act16 launchers: fully bf16 activation streams — x (the conv output /
+// BN input), y, incoming grad g, and dx are all bf16; statistics, affine
+// params and their grads stay fp32 (standard mixed-precision BN). ---
 
-void launch_bn_fwd16(const float* x, const float* gamma, const float* beta,
+void launch_bn_stats_only16(const void* x, long n, int h, float* partials,
+                            hipStream_t s) {
+  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+  if (n == 0) return;
+  const int nblocks = (int)min((long)512, (n + 63) / 64);
+  bn_stats_partial_kernel<<<nblocks, 256, 0, s>>>((const __bf16*)x, n, h,
+                                                  partials);
+}
+
+void launch_bn_fwd16(const void* x, const float* gamma, const float* beta,
                      float* running_mean, float* running_var, float* mean,
                      float* invstd, float* partials, void* y, long n, int h,
                      float momentum, float eps, bool training, bool relu,
@@ -1113,19 +1125,21 @@ void launch_bn_fwd16(const float* x, const float* gamma, const float* beta,
   if (training) {
     HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
     const int nblocks = (int)min((long)512, (n + 63) / 64);
-    bn_stats_partial_kernel<<<nblocks, 256, 0, s>>>(x, n, h, partials);
+    bn_stats_partial_kernel<<<nblocks, 256, 0, s>>>((const __bf16*)x, n, h,
+                                                    partials);
     bn_finalize_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
         partials, n, h, eps, momentum, mean, invstd, running_mean, running_var,
         1);
   } else {
     launch_bn_eval_stats(running_mean, running_var, mean, invstd, h, eps, s);
   }
-  bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma, beta,
+  bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>((const __bf16*)x, mean,
+                                                  invstd, gamma, beta,
                                                   (__bf16*)y, n, h,
                                                   relu ? 1 : 0);
 }
 
-void launch_bn_finalize_apply16(const float* x, const float* partials,
+void launch_bn_finalize_apply16(const void* x, const float* partials,
                                 long count, const float* gamma,
                                 const float* beta, float* running_mean,
                                 float* running_var, float* mean, float* invstd,
@@ -1141,48 +1155,49 @@ void launch_bn_finalize_apply16(const float* x, const float* partials,
         running_mean, running_var, mean, invstd, h, eps);
   }
   if (n > 0)
-    bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma,
-                                                    beta, (__bf16*)y, n, h,
+    bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>((const __bf16*)x, mean,
+                                                    invstd, gamma, beta,
+                                                    (__bf16*)y, n, h,
                                                     relu ? 1 : 0);
 }
 
-void launch_bn_bwd_partials_only16(const void* g, const float* x,
+void launch_bn_bwd_partials_only16(const void* g, const void* x,
                                    const void* y, const float* mean,
                                    const float* invstd, long n, int h,
                                    bool relu, float* partials, hipStream_t s) {
   HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
   if (n == 0) return;
   const int nblocks = (int)min((long)512, (n + 63) / 64);
-  bn_bwd_partial_kernel<<<nblocks, 256, 0, s>>>((const __bf16*)g, x,
-                                                (const __bf16*)y, mean, invstd,
-                                                n, h, relu ? 1 : 0, partials);
+  bn_bwd_partial_kernel<<<nblocks, 256, 0, s>>>(
+      (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd, n,
+      h, relu ? 1 : 0, partials);
 }
 
-void launch_bn_bwd_apply_only16(const void* g, const float* x, const void* y,
+void launch_bn_bwd_apply_only16(const void* g, const void* x, const void* y,
                                 const float* mean, const float* invstd,
                                 const float* gamma, const float* partials,
-                                long count, float* dx, long n, int h,
+                                long count, void* dx, long n, int h,
                                 bool relu, hipStream_t s) {
   if (n == 0) return;
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
-      (const __bf16*)g, x, (const __bf16*)y, mean, invstd, gamma, partials, dx,
-      n, count, h, relu ? 1 : 0);
+      (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd,
+      gamma, partials, (__bf16*)dx, n, count, h, relu ? 1 : 0);
 }
 
-void launch_bn_bwd16(const void* g, const float* x, const void* y,
+void launch_bn_bwd16(const void* g, const void* x, const void* y,
                      const float* mean, const float* invstd,
-                     const float* gamma, float* partials, float* dx,
+                     const float* gamma, float* partials, void* dx,
                      float* dgamma, float* dbeta, long n, int h, bool relu,
                      hipStream_t s) {
   if (n == 0) return;
   HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
   const int nblocks = (int)min((long)512, (n + 63) / 64);
   bn_bwd_partial_kernel<<<nblocks, 256, 0, s>>>(
-      (const __bf16*)g, x, (const __bf16*)y, mean, invstd, n, h, relu ? 1 : 0,
-      partials);
+      (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd, n,
+      h, relu ? 1 : 0, partials);
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
-      (const __bf16*)g, x, (const __bf16*)y, mean, invstd, gamma, partials, dx,
-      n, n, h, relu ? 1 : 0);
+      (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd,
+      gamma, partials, (__bf16*)dx, n, n, h, relu ? 1 : 0);
   bn_grad_affine_kernel<<<ceil_div(h, 256), 256, 0, s>>>(partials, dgamma,
                                                          dbeta, h);
 }

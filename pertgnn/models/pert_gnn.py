@@ -62,7 +62,8 @@ class TransformerConv(nn.Module):
         skip = ops.linear(x, self.lin_skip.weight, self.lin_skip.bias)
         return ops.edge_attention(q, k, v, e, skip, edge_index, n, csr=csr)
 
-    def forward_fused(self, x, edge_attr, ifc_weight, rpc_weight, csr):
+    def forward_fused(self, x, edge_attr, ifc_weight, rpc_weight, csr,
+                      out16=False):
         """HIP fast path: one [N,K]x[4H,K]^T GEMM for q/k/v/skip and
         L2-resident per-vocab P tables instead of the [E,2H] edge-embed
         stream (exact refactoring by linearity of lin_edge).  In bf16/fp16
@@ -89,7 +90,8 @@ class TransformerConv(nn.Module):
         we = self.lin_edge.weight  # [H, 2H]
         pifc = ops.linear(ifc_weight, we[:, :h].contiguous(), None)
         prpc = ops.linear(rpc_weight, we[:, h:].contiguous(), None)
-        return ops.edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr)
+        return ops.edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr,
+                                        out16=out16)
 
 
 class SAGEDeterministic(nn.Module):
@@ -190,11 +192,11 @@ class SAGEDeterministic(nn.Module):
             )
         n = x.shape[0]
 
-        def run_conv(conv, x):
+        def run_conv(conv, x, out16=False):
             if fused:
                 return conv.forward_fused(
                     x, edge_attr, self.interface_embeds.weight,
-                    self.rpctype_embeds.weight, csr,
+                    self.rpctype_embeds.weight, csr, out16=out16,
                 )
             return conv(x, edge_index, edge_embeds, csr=csr, num_nodes=n)
 
@@ -205,7 +207,7 @@ class SAGEDeterministic(nn.Module):
         out16 = (fused and ops.gemm_precision() == "bf16"
                  and hidden % 256 == 0 and ops.act16_enabled())
         for i, conv in enumerate(self.convs[:-1]):
-            x = run_conv(conv, x)
+            x = run_conv(conv, x, out16=out16)
             bn = self.bns[i]
             x = ops.batchnorm_relu(
                 x, bn.weight, bn.bias, bn.running_mean, bn.running_var,

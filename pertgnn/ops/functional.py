@@ -126,9 +126,9 @@ def act16_enabled() -> bool:
 
 class _EdgeAttentionFusedFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_eid):
+    def forward(ctx, qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_eid, out16=False):
         m = ext()
-        out, alpha = m.edge_attn_fused_fwd(qkvs, pifc, prpc, edge_attr, row_ptr, csr_src)
+        out, alpha = m.edge_attn_fused_fwd(qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, out16)
         ctx.save_for_backward(qkvs, pifc, prpc, edge_attr, alpha,
                               row_ptr, csr_src, col_ptr, csc_eid)
         return out
@@ -151,7 +151,7 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
                 de32 = de.float() if de.dtype != torch.float32 else de
                 dpifc = _table_grad(m, de32, edge_attr[:, 0], pifc.shape[0], h, 0)
                 dprpc = _table_grad(m, de32, edge_attr[:, 1], prpc.shape[0], h, 0)
-            return dqkvs, dpifc, dprpc, None, None, None, None, None
+            return dqkvs, dpifc, dprpc, None, None, None, None, None, None
         cur = torch.cuda.current_stream()
         side = _side_stream()
         ev = torch.cuda.Event()
@@ -166,16 +166,18 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
         cur.wait_event(ev2)
         _mark_cross_stream(dpifc, cur)
         _mark_cross_stream(dprpc, cur)
-        return dqkvs, dpifc, dprpc, None, None, None, None, None
+        return dqkvs, dpifc, dprpc, None, None, None, None, None, None
 
 
-def edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr):
+def edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr, out16=False):
     """HIP-only fast path: out_i = skip_i + softmax-weighted aggregate where
     q/k/v/skip are the four H-segments of ``qkvs`` and the edge embedding is
-    P_ifc[a0] + P_rpc[a1] (exact refactoring of lin_edge(concat(ifc, rpc)))."""
+    P_ifc[a0] + P_rpc[a1] (exact refactoring of lin_edge(concat(ifc, rpc))).
+    ``out16`` writes the aggregate bf16 (non-final layers feed BN, whose
+    act16 path consumes/produces bf16 streams)."""
     row_ptr, csr_src, col_ptr, _csc_dst, csc_eid = csr
     return _EdgeAttentionFusedFn.apply(
-        qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_eid
+        qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_eid, out16
     )
 
 
@@ -312,13 +314,15 @@ class _BNReLUFn(torch.autograd.Function):
             partials = m.bn_stats(x)
             comm.all_reduce_(partials)
             count = int(comm.all_reduce_scalar(float(x.shape[0])))
-            fin = m.bn_finalize_apply16 if out16 else m.bn_finalize_apply
+            fin = (m.bn_finalize_apply16 if x.dtype == torch.bfloat16
+                   else m.bn_finalize_apply)
             y, save_mean, save_invstd = fin(
                 x, partials, count, gamma, beta, running_mean, running_var,
                 momentum, eps, training, fuse_relu)
         else:
             count = x.shape[0]
-            fwd = m.bn_relu_fwd16 if out16 else m.bn_relu_fwd
+            fwd = (m.bn_relu_fwd16 if x.dtype == torch.bfloat16
+                   else m.bn_relu_fwd)
             y, save_mean, save_invstd = fwd(
                 x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu
             )

@@ -489,7 +489,7 @@ def test_fused_attention_empty_and_isolated_rows():
     ea = torch.zeros(0, 2, dtype=torch.long, device=DEV)
     row_ptr = torch.zeros(n + 1, dtype=torch.int32, device=DEV)
     csr_src = torch.zeros(0, dtype=torch.int32, device=DEV)
-    out, alpha = C.edge_attn_fused_fwd(qkvs, pifc, prpc, ea, row_ptr, csr_src)
+    out, alpha = C.edge_attn_fused_fwd(qkvs, pifc, prpc, ea, row_ptr, csr_src, False)
     torch.cuda.synchronize()
     assert alpha.numel() == 0
     assert torch.allclose(out, qkvs[:, 3 * h:], atol=1e-6)
