@@ -69,11 +69,12 @@ __device__ __forceinline__ float ld1(const __bf16* p) { return (float)*p; }
 // in flight, and only then wait + write LDS: the fused form stalled every
 // wave on s_waitcnt(vmcnt) BEFORE any MFMA issued (PMC: 72% SQ_WAIT_ANY on
 // the NT kernel).
-template <int BF, int BK, typename T16, typename TA>
+template <int BF, int BK, typename T16, typename TA,
+          int THREADS = BGEMM_THREADS>
 struct StageCmin {
   static constexpr int LDW = BK + BGEMM_PAD;
   static constexpr int QUADS = BK / 4;
-  static constexpr int FSTEP = BGEMM_THREADS / QUADS;
+  static constexpr int FSTEP = THREADS / QUADS;
   static constexpr int HALVES = BF / FSTEP;
   float v[HALVES][4];
 
@@ -129,12 +130,13 @@ struct StageCmin {
   }
 };
 
-template <int BF, int BK, typename T16, typename TA>
+template <int BF, int BK, typename T16, int THREADS = BGEMM_THREADS,
+          typename TA>
 __device__ __forceinline__ void bstage_cmin(const TA* __restrict__ g,
                                             long ld, int free0, int contract0,
                                             int free_max, int contract_max,
                                             T16* lds) {
-  StageCmin<BF, BK, T16, TA> s;
+  StageCmin<BF, BK, T16, TA, THREADS> s;
   s.load(g, ld, free0, contract0, free_max, contract_max);
   s.commit(lds);
 }
@@ -143,12 +145,16 @@ __device__ __forceinline__ void bstage_cmin(const TA* __restrict__ g,
 // transposes a 4x4 block in registers (4 coalesced f32x4 loads from 4
 // contract rows), then writes 4 packed 8-byte LDS rows.  Split load/commit
 // for the same software-pipelining reason as StageCmin.
-template <int BF, int BK, typename T16, typename TA>
+template <int BF, int BK, typename T16, typename TA,
+          int THREADS = BGEMM_THREADS>
 struct StageCmaj {
   static constexpr int LDW = BK + BGEMM_PAD;
   static constexpr int FQUADS = BF / 4;
-  static constexpr int CSTEP = (BGEMM_THREADS / FQUADS) * 4;
+  static constexpr int CSTEP = (THREADS / FQUADS) * 4;
   static constexpr int HALVES = (BK > CSTEP ? BK / CSTEP : 1);
+  // with more threads than BK/4 * FQUADS lanes of work, the surplus threads
+  // idle through staging (ACTIVE guarded below) but still hit the barrier
+  static constexpr bool SURPLUS = CSTEP > BK;
   float v[HALVES][4][4];
 
   __device__ __forceinline__ void load(const TA* __restrict__ g, long ld,
@@ -157,6 +163,7 @@ struct StageCmaj {
     const int t = threadIdx.x;
     const int cb = (t / FQUADS) * 4;
     const int fq = (t % FQUADS) * 4;
+    if (SURPLUS && cb >= BK) return;
     const bool interior = (contract0 + BK <= contract_max) &&
                           (free0 + BF <= free_max);
     const bool aligned = ((ld & 3) == 0) && ((free0 & 3) == 0);
@@ -201,6 +208,7 @@ struct StageCmaj {
     const int t = threadIdx.x;
     const int cb = (t / FQUADS) * 4;
     const int fq = (t % FQUADS) * 4;
+    if (SURPLUS && cb >= BK) return;
 #pragma unroll
     for (int half = 0; half < HALVES; ++half)
 #pragma unroll
@@ -212,12 +220,13 @@ struct StageCmaj {
   }
 };
 
-template <int BF, int BK, typename T16, typename TA>
+template <int BF, int BK, typename T16, int THREADS = BGEMM_THREADS,
+          typename TA>
 __device__ __forceinline__ void bstage_cmaj(const TA* __restrict__ g,
                                             long ld, int contract0, int free0,
                                             int contract_max, int free_max,
                                             T16* lds) {
-  StageCmaj<BF, BK, T16, TA> s;
+  StageCmaj<BF, BK, T16, TA, THREADS> s;
   s.load(g, ld, contract0, free0, contract_max, free_max);
   s.commit(lds);
 }
@@ -292,15 +301,17 @@ struct BWaveTile {
 // ---------------------------------------------------------------------------
 
 template <int BM, int BN, int BK = 64, typename T16 = __bf16,
-          typename TA = float, typename TO = float>
-__launch_bounds__(BGEMM_THREADS)
+          typename TA = float, typename TO = float,
+          int THREADS = BGEMM_THREADS>
+__launch_bounds__(THREADS)
 __global__ void gemm_bf16_nt_kernel(const TA* __restrict__ a,
                                     const float* __restrict__ b,
                                     const float* __restrict__ bias,
                                     TO* __restrict__ c, int m, int n, int k,
                                     int relu) {
   constexpr int LDW = BK + BGEMM_PAD;
-  constexpr int FM = BM / 32, FN = BN / 32;
+  constexpr int WCOL = THREADS / PERTGNN_WAVE / 2;  // wave columns
+  constexpr int FM = (BM / 2) / 16, FN = (BN / WCOL) / 16;
   __shared__ T16 lds_a[2][BM * LDW];
   __shared__ T16 lds_b[2][BN * LDW];
   const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
@@ -309,17 +320,17 @@ __global__ void gemm_bf16_nt_kernel(const TA* __restrict__ a,
   const int n0 = (bid % tiles_n) * BN;
   const int wave = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
-  const int wm = (wave >> 1) * (BM / 2);
-  const int wn = (wave & 1) * (BN / 2);
+  const int wm = (wave / WCOL) * (BM / 2);
+  const int wn = (wave % WCOL) * (BN / WCOL);
 
   BWaveTile<FM, FN, BK, T16> wt;
   wt.zero();
   int buf = 0;
-  bstage_cmin<BM, BK, T16>(a, k, m0, 0, m, k, lds_a[0]);
-  bstage_cmin<BN, BK, T16>(b, k, n0, 0, n, k, lds_b[0]);
+  bstage_cmin<BM, BK, T16, THREADS>(a, k, m0, 0, m, k, lds_a[0]);
+  bstage_cmin<BN, BK, T16, THREADS>(b, k, n0, 0, n, k, lds_b[0]);
   __syncthreads();
-  StageCmin<BM, BK, T16, TA> sa;
-  StageCmin<BN, BK, T16, float> sb;
+  StageCmin<BM, BK, T16, TA, THREADS> sa;
+  StageCmin<BN, BK, T16, float, THREADS> sb;
   for (int k0 = BK; k0 < k; k0 += BK) {
     sa.load(a, k, m0, k0, m, k);        // issue loads…
     sb.load(b, k, n0, k0, n, k);
@@ -334,15 +345,17 @@ __global__ void gemm_bf16_nt_kernel(const TA* __restrict__ a,
 }
 
 template <int BM, int BN, int BK = 64, typename T16 = __bf16,
-          typename TA = float, typename TO = float>
-__launch_bounds__(BGEMM_THREADS, 3)
+          typename TA = float, typename TO = float,
+          int THREADS = BGEMM_THREADS>
+__launch_bounds__(THREADS, 3)
 __global__ void gemm_bf16_nn_kernel(const TA* __restrict__ a,
                                     const float* __restrict__ b,
                                     const float* __restrict__ bias,
                                     TO* __restrict__ c, int m, int n,
                                     int k2, int relu) {
   constexpr int LDW = BK + BGEMM_PAD;
-  constexpr int FM = BM / 32, FN = BN / 32;
+  constexpr int WCOL = THREADS / PERTGNN_WAVE / 2;
+  constexpr int FM = (BM / 2) / 16, FN = (BN / WCOL) / 16;
   __shared__ T16 lds_a[2][BM * LDW];
   __shared__ T16 lds_b[2][BN * LDW];
   const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
@@ -351,17 +364,17 @@ __global__ void gemm_bf16_nn_kernel(const TA* __restrict__ a,
   const int n0 = (bid % tiles_n) * BN;
   const int wave = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
-  const int wm = (wave >> 1) * (BM / 2);
-  const int wn = (wave & 1) * (BN / 2);
+  const int wm = (wave / WCOL) * (BM / 2);
+  const int wn = (wave % WCOL) * (BN / WCOL);
 
   BWaveTile<FM, FN, BK, T16> wt;
   wt.zero();
   int buf = 0;
-  bstage_cmin<BM, BK, T16>(a, n, m0, 0, m, n, lds_a[0]);
-  bstage_cmaj<BN, BK, T16>(b, k2, 0, n0, n, k2, lds_b[0]);
+  bstage_cmin<BM, BK, T16, THREADS>(a, n, m0, 0, m, n, lds_a[0]);
+  bstage_cmaj<BN, BK, T16, THREADS>(b, k2, 0, n0, n, k2, lds_b[0]);
   __syncthreads();
-  StageCmin<BM, BK, T16, TA> sa;
-  StageCmaj<BN, BK, T16, float> sb;
+  StageCmin<BM, BK, T16, TA, THREADS> sa;
+  StageCmaj<BN, BK, T16, float, THREADS> sb;
   for (int c0 = BK; c0 < n; c0 += BK) {
     sa.load(a, n, m0, c0, m, n);
     sb.load(b, k2, c0, n0, n, k2);
@@ -376,15 +389,17 @@ __global__ void gemm_bf16_nn_kernel(const TA* __restrict__ a,
 }
 
 template <int BM, int BN, int BK = 64, typename T16 = __bf16,
-          typename TA = float, typename TB = float>
-__launch_bounds__(BGEMM_THREADS)
+          typename TA = float, typename TB = float,
+          int THREADS = BGEMM_THREADS>
+__launch_bounds__(THREADS)
 __global__ void gemm_bf16_tn_kernel(const TA* __restrict__ a,
                                     const TB* __restrict__ b,
                                     float* __restrict__ c,
                                     float* __restrict__ dbias, int m, int n,
                                     int k2, int slices) {
   constexpr int LDW = BK + BGEMM_PAD;
-  constexpr int FM = BM / 32, FN = BN / 32;
+  constexpr int WCOL = THREADS / PERTGNN_WAVE / 2;
+  constexpr int FM = (BM / 2) / 16, FN = (BN / WCOL) / 16;
   __shared__ T16 lds_a[2][BM * LDW];
   __shared__ T16 lds_b[2][BN * LDW];
   const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
@@ -396,8 +411,8 @@ __global__ void gemm_bf16_tn_kernel(const TA* __restrict__ a,
   if (n0 >= n) return;
   const int wave = threadIdx.x / PERTGNN_WAVE;
   const int lane = threadIdx.x % PERTGNN_WAVE;
-  const int wm = (wave >> 1) * (BM / 2);
-  const int wn = (wave & 1) * (BN / 2);
+  const int wm = (wave / WCOL) * (BM / 2);
+  const int wn = (wave % WCOL) * (BN / WCOL);
 
   const int per_slice =
       ((m + slices - 1) / slices + BK - 1) / BK * BK;
@@ -413,11 +428,11 @@ __global__ void gemm_bf16_tn_kernel(const TA* __restrict__ a,
   float dbsum = 0.f;
   const int bcol = threadIdx.x;
   int buf = 0;
-  bstage_cmaj<BM, BK, T16>(a, n, c_beg, n0, c_end, n, lds_a[0]);
-  bstage_cmaj<BN, BK, T16>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
+  bstage_cmaj<BM, BK, T16, THREADS>(a, n, c_beg, n0, c_end, n, lds_a[0]);
+  bstage_cmaj<BN, BK, T16, THREADS>(b, k2, c_beg, k0, c_end, k2, lds_b[0]);
   __syncthreads();
-  StageCmaj<BM, BK, T16, TA> sa;
-  StageCmaj<BN, BK, T16, TB> sb;
+  StageCmaj<BM, BK, T16, TA, THREADS> sa;
+  StageCmaj<BN, BK, T16, TB, THREADS> sb;
   for (int cc = c_beg + BK; cc < c_end; cc += BK) {
     sa.load(a, n, cc, n0, c_end, n);
     sb.load(b, k2, cc, k0, c_end, k2);
@@ -468,6 +483,14 @@ static int gemm_bk() {
     return (e && atoi(e) == 64) ? 64 : 32;  // default 32 (4 blocks/CU)
   }();
   return bk;
+}
+
+static bool gemm_t512() {  // 8-wave (512-thread) blocks for the a16 GEMMs
+  static bool v = [] {
+    const char* e = getenv("PERTGNN_GEMM_T512");
+    return e && atoi(e) == 1;
+  }();
+  return v;
 }
 
 static bool gemm_n64() {  // narrow-BN experiment: 128x64 tiles, 4 waves/SIMD
@@ -697,6 +720,11 @@ void launch_gemm_bf16_nt_a16o16(const void* a_v, const float* b,
   __bf16* c = (__bf16*)c_v;
   if (m >= 512 && n >= 128) {
     const int grid = ((m + 127) / 128) * ((n + 127) / 128);
+    if (gemm_t512()) {
+     hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128, 32, __bf16, __bf16, __bf16, 512>)
+          , dim3(dim3(grid)), dim3(dim3(512)), 0, s, a, b, bias, c, m, n, k, 0);
+      return;
+    }
    hipLaunchKernelGGL(( gemm_bf16_nt_kernel<128, 128, 32, __bf16, __bf16, __bf16>)
         , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, bias, c, m, n, k, 0);
   } else {
@@ -712,6 +740,11 @@ void launch_gemm_bf16_nn_a16o16(const void* a_v, const float* b, void* c_v,
   __bf16* c = (__bf16*)c_v;
   if (m >= 512 && k2 >= 128) {
     const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
+    if (gemm_t512()) {
+     hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128, 32, __bf16, __bf16, __bf16, 512>)
+          , dim3(dim3(grid)), dim3(dim3(512)), 0, s, a, b, nullptr, c, m, n, k2, 0);
+      return;
+    }
    hipLaunchKernelGGL(( gemm_bf16_nn_kernel<128, 128, 32, __bf16, __bf16, __bf16>)
         , dim3(dim3(grid)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, nullptr, c, m, n, k2,
                                                     0);
@@ -738,12 +771,18 @@ void launch_gemm_bf16_tn_a16b16(const void* a_v, const void* b_v, float* c,
   if (slices > 1)
     HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
   if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
-  if (big)
+  if (big) {
+    if (gemm_t512()) {
+     hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128, 32, __bf16, __bf16, __bf16, 512>)
+          , dim3(dim3(tiles * slices)), dim3(dim3(512)), 0, s, a, b, c, dbias, m, n,
+                                                      k2, slices);
+      return;
+    }
    hipLaunchKernelGGL(( gemm_bf16_tn_kernel<128, 128, 32, __bf16, __bf16, __bf16>)
         , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
                                                               m, n, k2,
                                                               slices);
-  else
+  } else
    hipLaunchKernelGGL(( gemm_bf16_tn_kernel<64, 64, 64, __bf16, __bf16, __bf16>)
         , dim3(dim3(tiles * slices)), dim3(dim3(BGEMM_THREADS)), 0, s, a, b, c, dbias,
                                                               m, n, k2,
