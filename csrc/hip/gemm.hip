@@ -17,6 +17,14 @@
 // (16 lanes read 16 consecutive floats of one LDS row).
 
 #include "common.h"
+#include <cstdlib>
+
+// PERTGNN_DETERMINISTIC=1: collapse split-K wgrad to one slice so the dw/db
+// reduction order is fixed (read per call — tests flip it at runtime)
+static inline bool pertgnn_deterministic() {
+  const char* e = getenv("PERTGNN_DETERMINISTIC");
+  return e && e[0] == '1';
+}
 
 #define GEMM_BK 32
 #define GEMM_THREADS 256
@@ -385,7 +393,8 @@ void launch_gemm_f32_tn(const float* a, const float* b, float* c, float* dbias,
   const int bn = big ? 128 : 64;
   const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
   int slices = 1;
-  while (tiles * slices < 512 && slices < 64 && (long)slices * GEMM_BK * 4 < m)
+  while (!pertgnn_deterministic() && tiles * slices < 512 && slices < 64 &&
+         (long)slices * GEMM_BK * 4 < m)
     slices *= 2;
   if (slices > 1)
     HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));

@@ -10,6 +10,13 @@
 // makes the 16-lane fragment reads conflict-free (guide §6 G4).
 
 #include "common.h"
+
+// PERTGNN_DETERMINISTIC=1: collapse split-K wgrad to one slice so the dw/db
+// reduction order is fixed (read per call — tests flip it at runtime)
+static inline bool pertgnn_deterministic() {
+  const char* e = getenv("PERTGNN_DETERMINISTIC");
+  return e && e[0] == '1';
+}
 #include <cstdlib>
 
 #define BGEMM_PAD 8
@@ -555,7 +562,7 @@ void launch_gemm_bf16_tn(const float* a, const float* b, float* c,
   const int bn = big ? 128 : 64;
   const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
   int slices = 1;
-  while (tiles * slices < 512 && slices < 64 &&
+  while (!pertgnn_deterministic() && tiles * slices < 512 && slices < 64 &&
          (long)slices * 64 * 4 < m)
     slices *= 2;
   if (slices > 1)
@@ -623,7 +630,7 @@ void launch_gemm_fp16_tn(const float* a, const float* b, float* c,
   const int bn = big ? 128 : 64;
   const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
   int slices = 1;
-  while (tiles * slices < 512 && slices < 64 &&
+  while (!pertgnn_deterministic() && tiles * slices < 512 && slices < 64 &&
          (long)slices * 64 * 4 < m)
     slices *= 2;
   if (slices > 1)
@@ -687,7 +694,8 @@ void launch_gemm_bf16_tn_a16(const void* a_v, const float* b, float* c,
   const int bn = big ? 128 : 64;
   const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
   int slices = 1;
-  while (tiles * slices < 512 && slices < 64 && (long)slices * 64 * 4 < m)
+  while (!pertgnn_deterministic() && tiles * slices < 512 && slices < 64 &&
+         (long)slices * 64 * 4 < m)
     slices *= 2;
   if (slices > 1)
     HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
@@ -765,7 +773,8 @@ void launch_gemm_bf16_tn_a16b16(const void* a_v, const void* b_v, float* c,
   const int bn = big ? 128 : 64;
   const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
   int slices = 1;
-  while (tiles * slices < 512 && slices < 64 && (long)slices * 64 * 4 < m)
+  while (!pertgnn_deterministic() && tiles * slices < 512 && slices < 64 &&
+         (long)slices * 64 * 4 < m)
     slices *= 2;
   if (slices > 1)
     HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));

@@ -7,6 +7,7 @@
 //   K12/K13 quantile loss + eval-metric reductions
 //   K14     Adam step (flat master buffers — one kernel for the whole model)
 #include "common.h"
+#include <cstdlib>
 
 // ---------------------------------------------------------------------------
 // pattern pool: out[b] = sum_{i in graph b} x[i] * p[i] / n[i]
@@ -734,7 +735,7 @@ void launch_gather_rows(const long* idx, const float* table, float* out,
 // stage A: per-block partial sum/sumsq over a row range; thread t owns
 // channels t and t+256 in REGISTERS (H <= 512), one atomicAdd per channel
 // per block at the end — coalesced loads, no LDS traffic.
-template <typename TX = float>
+template <typename TX = float, bool SLAB = false>
 __global__ void bn_stats_partial_kernel(const TX* __restrict__ x, long n,
                                         int h, float* __restrict__ partials) {
   const int c0 = threadIdx.x;
@@ -771,6 +772,13 @@ __global__ void bn_stats_partial_kernel(const TX* __restrict__ x, long n,
       s1 += v; q1 += v * v;
     }
   }
+  if (SLAB) {
+    // deterministic mode: per-block slab row, reduced in fixed block order
+    float* row = partials + (long)blockIdx.x * 2 * h;
+    if (c0 < h) { row[c0] = s0; row[h + c0] = q0; }
+    if (c1 < h) { row[c1] = s1; row[h + c1] = q1; }
+    return;
+  }
   if (c0 < h && r0 < r1) {
     atomicAdd(&partials[c0], s0);
     atomicAdd(&partials[h + c0], q0);
@@ -780,6 +788,50 @@ __global__ void bn_stats_partial_kernel(const TX* __restrict__ x, long n,
     atomicAdd(&partials[h + c1], q1);
   }
 }
+
+// fixed-order fold of the deterministic-mode slab rows
+__global__ void bn_slab_reduce_kernel(const float* __restrict__ slab, int nb,
+                                      int w, float* __restrict__ out) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= w) return;
+  float s = 0.f;
+  for (int b = 0; b < nb; ++b) s += slab[(long)b * w + c];
+  out[c] = s;
+}
+
+// lazily-allocated device slab for the deterministic BN reductions
+// (512 blocks x 2 x 1024 channels); deterministic mode is documented as
+// incompatible with hipGraph capture, so the lazy hipMalloc is safe.
+static float* bn_det_slab() {
+  static float* p = [] {
+    float* q = nullptr;
+    HIP_CHECK(hipMalloc(&q, (size_t)512 * 2 * 1024 * sizeof(float)));
+    return q;
+  }();
+  return p;
+}
+
+static inline bool pertgnn_deterministic_seg() {
+  const char* e = getenv("PERTGNN_DETERMINISTIC");
+  return e && e[0] == '1';
+}
+
+template <typename TX>
+static void bn_stats_dispatch(const TX* x, long n, int h, float* partials,
+                              hipStream_t s) {
+  const int nblocks = (int)min((long)512, (n + 63) / 64);
+  if (pertgnn_deterministic_seg() && h <= 1024) {
+    float* slab = bn_det_slab();
+   hipLaunchKernelGGL(( bn_stats_partial_kernel<TX, true>), dim3(nblocks), dim3(256), 0, s, x, n, h, slab);
+   hipLaunchKernelGGL(( bn_slab_reduce_kernel), dim3(ceil_div(2 * h, 256)), dim3(256), 0, s, 
+        slab, nblocks, 2 * h, partials);
+  } else {
+    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+   hipLaunchKernelGGL(( bn_stats_partial_kernel<TX, false>), dim3(nblocks), dim3(256), 0, s, x, n, h,
+                                                               partials);
+  }
+}
+
 
 // stage B: finalize mean/invstd (+ running-stat update, training only)
 __global__ void bn_finalize_kernel(const float* __restrict__ partials, long n,
@@ -849,7 +901,7 @@ __global__ void bn_apply_kernel(const TX* __restrict__ x,
 }
 
 // backward stage A: per-channel sums of gm and gm*xhat (gm = relu-masked g)
-template <typename TG, typename TY, typename TX = float>
+template <typename TG, typename TY, typename TX = float, bool SLAB = false>
 __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
                                       const TX* __restrict__ x,
                                       const TY* __restrict__ y,
@@ -914,6 +966,12 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
       q1 += gm * ((float)x[r * h + c1] - m1) * i1;
     }
   }
+  if (SLAB) {
+    float* row = partials + (long)blockIdx.x * 2 * h;
+    if (c0 < h) { row[c0] = s0; row[h + c0] = q0; }
+    if (c1 < h) { row[c1] = s1; row[h + c1] = q1; }
+    return;
+  }
   if (c0 < h && r0 < r1) {
     atomicAdd(&partials[c0], s0);
     atomicAdd(&partials[h + c0], q0);
@@ -921,6 +979,26 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
   if (c1 < h && r0 < r1) {
     atomicAdd(&partials[c1], s1);
     atomicAdd(&partials[h + c1], q1);
+  }
+}
+
+
+template <typename TG, typename TY, typename TX>
+static void bn_bwd_partials_dispatch(const TG* g, const TX* x, const TY* y,
+                                     const float* mean, const float* invstd,
+                                     long n, int h, bool relu, float* partials,
+                                     hipStream_t s) {
+  const int nblocks = (int)min((long)512, (n + 63) / 64);
+  if (pertgnn_deterministic_seg() && h <= 1024) {
+    float* slab = bn_det_slab();
+   hipLaunchKernelGGL(( bn_bwd_partial_kernel<TG, TY, TX, true>), dim3(nblocks), dim3(256), 0, s, 
+        g, x, y, mean, invstd, n, h, relu ? 1 : 0, slab);
+   hipLaunchKernelGGL(( bn_slab_reduce_kernel), dim3(ceil_div(2 * h, 256)), dim3(256), 0, s, 
+        slab, nblocks, 2 * h, partials);
+  } else {
+    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+   hipLaunchKernelGGL(( bn_bwd_partial_kernel<TG, TY, TX, false>), dim3(nblocks), dim3(256), 0, s, 
+        g, x, y, mean, invstd, n, h, relu ? 1 : 0, partials);
   }
 }
 
@@ -1014,10 +1092,11 @@ void launch_bn_eval_stats(const float* running_mean, const float* running_var,
 
 void launch_bn_stats_only(const float* x, long n, int h, float* partials,
                           hipStream_t s) {
-  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
-  if (n == 0) return;
-  const int nblocks = (int)min((long)512, (n + 63) / 64);
- hipLaunchKernelGGL(( bn_stats_partial_kernel), dim3(nblocks), dim3(256), 0, s, x, n, h, partials);
+  if (n == 0) {
+    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+    return;
+  }
+  bn_stats_dispatch(x, n, h, partials, s);
 }
 
 void launch_bn_finalize_apply(const float* x, const float* partials,
@@ -1045,11 +1124,11 @@ void launch_bn_bwd_partials_only(const float* g, const float* x,
                                  const float* y, const float* mean,
                                  const float* invstd, long n, int h, bool relu,
                                  float* partials, hipStream_t s) {
-  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
-  if (n == 0) return;
-  const int nblocks = (int)min((long)512, (n + 63) / 64);
- hipLaunchKernelGGL(( bn_bwd_partial_kernel), dim3(nblocks), dim3(256), 0, s, g, x, y, mean, invstd, n, h,
-                                                relu ? 1 : 0, partials);
+  if (n == 0) {
+    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+    return;
+  }
+  bn_bwd_partials_dispatch(g, x, y, mean, invstd, n, h, relu, partials, s);
 }
 
 void launch_bn_bwd_apply_only(const float* g, const float* x, const float* y,
@@ -1075,10 +1154,7 @@ void launch_bn_fwd(const float* x, const float* gamma, const float* beta,
                    hipStream_t s) {
   if (n == 0) return;
   if (training) {
-    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
-    const int nblocks = (int)min((long)512, (n + 63) / 64);
-   hipLaunchKernelGGL(( bn_stats_partial_kernel), dim3(nblocks), dim3(256), 0, s, 
-        x, n, h, partials);
+    bn_stats_dispatch(x, n, h, partials, s);
    hipLaunchKernelGGL(( bn_finalize_kernel), dim3(ceil_div(h, 256)), dim3(256), 0, s, 
         partials, n, h, eps, momentum, mean, invstd, running_mean, running_var,
         1);
@@ -1094,10 +1170,7 @@ void launch_bn_bwd(const float* g, const float* x, const float* y,
                    float* partials, float* dx, float* dgamma, float* dbeta,
                    long n, int h, bool relu, hipStream_t s) {
   if (n == 0) return;
-  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
-  const int nblocks = (int)min((long)512, (n + 63) / 64);
- hipLaunchKernelGGL(( bn_bwd_partial_kernel), dim3(nblocks), dim3(256), 0, s, 
-      g, x, y, mean, invstd, n, h, relu ? 1 : 0, partials);
+  bn_bwd_partials_dispatch(g, x, y, mean, invstd, n, h, relu, partials, s);
  hipLaunchKernelGGL(( bn_bwd_apply_kernel), dim3(grid_for(n * h)), dim3(256), 0, s, 
       g, x, y, mean, invstd, gamma, partials, dx, n, n, h, relu ? 1 : 0);
  hipLaunchKernelGGL(( bn_grad_affine_kernel), dim3(ceil_div(h, 256)), dim3(256), 0, s, partials, dgamma,
@@ -1110,11 +1183,11 @@ void launch_bn_bwd(const float* g, const float* x, const float* y,
 
 void launch_bn_stats_only16(const void* x, long n, int h, float* partials,
                             hipStream_t s) {
-  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
-  if (n == 0) return;
-  const int nblocks = (int)min((long)512, (n + 63) / 64);
- hipLaunchKernelGGL(( bn_stats_partial_kernel), dim3(nblocks), dim3(256), 0, s, (const __bf16*)x, n, h,
-                                                  partials);
+  if (n == 0) {
+    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+    return;
+  }
+  bn_stats_dispatch((const __bf16*)x, n, h, partials, s);
 }
 
 void launch_bn_fwd16(const void* x, const float* gamma, const float* beta,
@@ -1124,10 +1197,7 @@ void launch_bn_fwd16(const void* x, const float* gamma, const float* beta,
                      hipStream_t s) {
   if (n == 0) return;
   if (training) {
-    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
-    const int nblocks = (int)min((long)512, (n + 63) / 64);
-   hipLaunchKernelGGL(( bn_stats_partial_kernel), dim3(nblocks), dim3(256), 0, s, (const __bf16*)x, n, h,
-                                                    partials);
+    bn_stats_dispatch((const __bf16*)x, n, h, partials, s);
    hipLaunchKernelGGL(( bn_finalize_kernel), dim3(ceil_div(h, 256)), dim3(256), 0, s, 
         partials, n, h, eps, momentum, mean, invstd, running_mean, running_var,
         1);
@@ -1166,12 +1236,13 @@ void launch_bn_bwd_partials_only16(const void* g, const void* x,
                                    const void* y, const float* mean,
                                    const float* invstd, long n, int h,
                                    bool relu, float* partials, hipStream_t s) {
-  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
-  if (n == 0) return;
-  const int nblocks = (int)min((long)512, (n + 63) / 64);
- hipLaunchKernelGGL(( bn_bwd_partial_kernel), dim3(nblocks), dim3(256), 0, s, 
-      (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd, n,
-      h, relu ? 1 : 0, partials);
+  if (n == 0) {
+    HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
+    return;
+  }
+  bn_bwd_partials_dispatch((const __bf16*)g, (const __bf16*)x,
+                           (const __bf16*)y, mean, invstd, n, h, relu,
+                           partials, s);
 }
 
 void launch_bn_bwd_apply_only16(const void* g, const void* x, const void* y,
@@ -1191,11 +1262,9 @@ void launch_bn_bwd16(const void* g, const void* x, const void* y,
                      float* dgamma, float* dbeta, long n, int h, bool relu,
                      hipStream_t s) {
   if (n == 0) return;
-  HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
-  const int nblocks = (int)min((long)512, (n + 63) / 64);
- hipLaunchKernelGGL(( bn_bwd_partial_kernel), dim3(nblocks), dim3(256), 0, s, 
-      (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd, n,
-      h, relu ? 1 : 0, partials);
+  bn_bwd_partials_dispatch((const __bf16*)g, (const __bf16*)x,
+                           (const __bf16*)y, mean, invstd, n, h, relu,
+                           partials, s);
  hipLaunchKernelGGL(( bn_bwd_apply_kernel), dim3(grid_for(n * h)), dim3(256), 0, s, 
       (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd,
       gamma, partials, (__bf16*)dx, n, n, h, relu ? 1 : 0);

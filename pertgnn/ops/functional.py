@@ -16,24 +16,50 @@ from . import reference as ref
 from .backend import ext, use_hip
 
 
+def deterministic() -> bool:
+    """PERTGNN_DETERMINISTIC=1 makes the weight-gradient reductions
+    run-to-run bitwise reproducible: vocab/embedding table grads take the
+    two-phase grouped scatter (fixed reduction order, no atomics) and the
+    split-K wgrad GEMMs collapse to one K-slice (the C++ launchers read the
+    same env).  The activation-gradient path is deterministic either way.
+    Costs ~10-15% step time; incompatible with hipGraph capture (the grouping
+    uses bincount, which syncs) — use --no-hipgraph in bench.py."""
+    import os
+    return os.environ.get("PERTGNN_DETERMINISTIC", "0") == "1"
+
+
 def _table_grad(m, g, idx, rows, h, col_off):
     """dtable[v] = segment-sum of g[:, col_off:col_off+h] by idx — LDS vocab
     accumulator for small tables, deterministic two-phase grouped scatter
-    otherwise."""
-    if rows * h * 4 <= 160 * 1024:
+    otherwise (always, under PERTGNN_DETERMINISTIC=1)."""
+    if rows * h * 4 <= 160 * 1024 and not deterministic():
         return m.vocab_scatter(g, idx, rows, h, col_off)
     order, ptr = _group_by(idx.contiguous(), rows)
     return m.embed_grouped_scatter(g, order, ptr, rows, h, col_off)
 
 
+_GROUP_CACHE: dict = {}
+
+
 def _group_by(idx: torch.Tensor, rows: int):
     """Group positions by index value: returns (order int32, ptr int32[rows+1])
-    for the deterministic grouped scatter kernels."""
-    order = torch.argsort(idx)
+    for the deterministic grouped scatter kernels.  Cached per
+    (data_ptr, numel, rows): batch index tensors are static across steps, so
+    the sort + bincount (which syncs) runs once per resident batch."""
+    key = (idx.data_ptr(), idx.numel(), rows)
+    hit = _GROUP_CACHE.get(key)
+    if hit is not None:
+        return hit
+    order = torch.argsort(idx, stable=True)  # ties in input order: the
+    # grouped kernels' reduction order is then fully determined
     counts = torch.bincount(idx, minlength=rows)
     ptr = torch.zeros(rows + 1, dtype=torch.int32, device=idx.device)
     ptr[1:] = counts.cumsum(0).to(torch.int32)
-    return order.to(torch.int32), ptr
+    out = (order.to(torch.int32), ptr)
+    if len(_GROUP_CACHE) > 256:  # bounded: resident batches reuse, streams churn
+        _GROUP_CACHE.clear()
+    _GROUP_CACHE[key] = out
+    return out
 
 
 # ---------------------------------------------------------------------------
@@ -145,7 +171,8 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
         # CSC dk/dv pass queued above — overlap them on the side stream.
         if not _overlap_enabled():
             h = de.shape[1]
-            if h % 256 == 0 and (pifc.shape[0] + prpc.shape[0]) * h * 4 <= 160 * 1024:
+            if (h % 256 == 0 and not deterministic()
+                    and (pifc.shape[0] + prpc.shape[0]) * h * 4 <= 160 * 1024):
                 dpifc, dprpc = m.vocab_scatter_dual(de, edge_attr, pifc.shape[0], prpc.shape[0])
             else:
                 de32 = de.float() if de.dtype != torch.float32 else de

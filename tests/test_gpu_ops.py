@@ -538,3 +538,48 @@ def test_model_act16_close_to_fp32():
         a, c = g16[n].flatten(), g32[n].flatten()
         cos = torch.dot(a, c) / (a.norm() * c.norm()).clamp_min(1e-12)
         assert cos > 0.99, (n, float(cos))
+
+
+def test_deterministic_mode_bitwise(monkeypatch):
+    """PERTGNN_DETERMINISTIC=1: weight grads are bitwise identical across
+    runs (grouped table scatter + single-slice wgrad GEMMs), and still agree
+    with the default (atomic) reductions numerically."""
+    require_ext()
+    import os
+    import bench as bench_mod
+    from pertgnn.models import SAGEDeterministic
+    from pertgnn.ops.functional import set_gemm_precision
+
+    torch.manual_seed(2)
+    batches, stats = bench_mod.build_synthetic_batches(1, 32, seed=7, device=DEV)
+    b = batches[0]
+    model = SAGEDeterministic(9, [stats["cat_max"] + 1], stats["entry_max"],
+                              stats["ifc_max"], stats["rpc_max"], 256, 3, 0.0).to(DEV)
+    model.train()
+
+    def grads():
+        gp, _ = model(b.x, b.cat_X, b.edge_index, b.edge_attr,
+                      b.pattern_num_nodes, b.rt_probs, b.entry_id, b.batch,
+                      csr=b.csr, num_graphs=b.num_graphs)
+        loss = F.quantile_loss(b.y, gp.flatten(), 0.5)
+        loss.backward()
+        out = {n: p.grad.clone() for n, p in model.named_parameters()
+               if p.grad is not None}
+        model.zero_grad()
+        return out
+
+    try:
+        set_gemm_precision("bf16")
+        monkeypatch.setenv("PERTGNN_DETERMINISTIC", "1")
+        g1 = grads()
+        g2 = grads()
+        monkeypatch.delenv("PERTGNN_DETERMINISTIC")
+        g_atomic = grads()
+    finally:
+        set_gemm_precision("fp32")
+    for n in g1:
+        assert torch.equal(g1[n], g2[n]), f"{n} not bitwise reproducible"
+    # the deterministic reductions compute the same math (fp32 sums, different
+    # order) — loose agreement with the atomic path
+    for n in ("convs.1.lin_query.weight", "interface_embeds.weight"):
+        assert torch.allclose(g1[n], g_atomic[n], atol=1e-2, rtol=1e-2), n
