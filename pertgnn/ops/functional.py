@@ -49,17 +49,19 @@ def _group_by(idx: torch.Tensor, rows: int):
     key = (idx.data_ptr(), idx.numel(), rows)
     hit = _GROUP_CACHE.get(key)
     if hit is not None:
-        return hit
+        return hit[1], hit[2]
     order = torch.argsort(idx, stable=True)  # ties in input order: the
     # grouped kernels' reduction order is then fully determined
     counts = torch.bincount(idx, minlength=rows)
     ptr = torch.zeros(rows + 1, dtype=torch.int32, device=idx.device)
     ptr[1:] = counts.cumsum(0).to(torch.int32)
-    out = (order.to(torch.int32), ptr)
+    order32 = order.to(torch.int32)
     if len(_GROUP_CACHE) > 256:  # bounded: resident batches reuse, streams churn
         _GROUP_CACHE.clear()
-    _GROUP_CACHE[key] = out
-    return out
+    # the cached idx reference pins its storage so the allocator cannot hand
+    # the same data_ptr to a different tensor while the entry is live
+    _GROUP_CACHE[key] = (idx, order32, ptr)
+    return order32, ptr
 
 
 # ---------------------------------------------------------------------------
