@@ -24,13 +24,47 @@ from torch import nn
 from ..ops import functional as ops
 
 
+class _SegLinearView:
+    """Read-through view of one H-row segment of the conv's fused ``w4``/``b4``
+    parameters — keeps the PyG-style ``conv.lin_query.weight`` access (and the
+    eager CPU forward) working while the storage is the single fused tensor
+    the QKVS GEMM consumes."""
+
+    def __init__(self, conv, seg, has_bias=True):
+        object.__setattr__(self, "_conv", conv)
+        object.__setattr__(self, "_seg", seg)
+        object.__setattr__(self, "_has_bias", has_bias)
+
+    @property
+    def weight(self):
+        c = self._conv
+        h = c.out_channels
+        return c.w4[self._seg * h:(self._seg + 1) * h, :c.in_channels]
+
+    @property
+    def bias(self):
+        if not self._has_bias:
+            return None
+        c = self._conv
+        h = c.out_channels
+        return c.b4[self._seg * h:(self._seg + 1) * h]
+
+
 class TransformerConv(nn.Module):
     """Graph transformer conv, PyG-2.4.0 semantics with heads=1, concat=True,
     root_weight=True, beta=False (reference model.py:25-52 configuration).
 
     out_i = W_skip x_i + b + sum_{e:(j->i)} softmax_i(<W_q x_i, W_k x_j + W_e e_ij>/sqrt(H))
             * (W_v x_j + W_e e_ij)
-    """
+
+    The four projection matrices live in ONE fused parameter ``w4`` [4H, Kp]
+    (rows: query, key, value, skip; Kp = K padded to a multiple of 8 with
+    zero columns so the GEMM staging vectorizes) + fused bias ``b4`` [4H] —
+    the hot QKVS GEMM and its weight-grad then touch a single tensor with no
+    per-step cat/slice glue.  ``lin_query``/``lin_key``/``lin_value``/
+    ``lin_skip`` stay accessible as views, and the state_dict still uses the
+    reference's per-projection keys (checkpoint compatibility), remapped in
+    ``_save_to_state_dict``/``_load_from_state_dict``."""
 
     def __init__(self, in_channels: int, out_channels: int, heads: int = 1, edge_dim: int | None = None):
         super().__init__()
@@ -39,27 +73,74 @@ class TransformerConv(nn.Module):
         self.out_channels = out_channels
         self.heads = heads
         self.edge_dim = edge_dim
-        self.lin_key = nn.Linear(in_channels, heads * out_channels)
-        self.lin_query = nn.Linear(in_channels, heads * out_channels)
-        self.lin_value = nn.Linear(in_channels, heads * out_channels)
+        self.k_padded = (in_channels + 7) // 8 * 8
+        self.w4 = nn.Parameter(torch.zeros(4 * out_channels, self.k_padded))
+        self.b4 = nn.Parameter(torch.zeros(4 * out_channels))
+        self.lin_query = _SegLinearView(self, 0)
+        self.lin_key = _SegLinearView(self, 1)
+        self.lin_value = _SegLinearView(self, 2)
+        self.lin_skip = _SegLinearView(self, 3)
         self.lin_edge = nn.Linear(edge_dim, heads * out_channels, bias=False)
-        self.lin_skip = nn.Linear(in_channels, out_channels, bias=True)
         self.reset_parameters()
 
     def reset_parameters(self):
         # PyG Linear default init is glorot for weight, zeros for bias.
+        with torch.no_grad():
+            self.w4.zero_()
+            self.b4.zero_()
         for lin in (self.lin_key, self.lin_query, self.lin_value, self.lin_edge, self.lin_skip):
             nn.init.xavier_uniform_(lin.weight)
             if lin.bias is not None:
                 nn.init.zeros_(lin.bias)
 
+    _SEG_NAMES = ("lin_query", "lin_key", "lin_value", "lin_skip")
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        # emit the reference's per-projection keys instead of w4/b4
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+        del destination[prefix + "w4"]
+        del destination[prefix + "b4"]
+        for i, name in enumerate(self._SEG_NAMES):
+            view = getattr(self, name)
+            w = view.weight
+            b = view.bias
+            destination[prefix + name + ".weight"] = w if keep_vars else w.detach().clone()
+            destination[prefix + name + ".bias"] = b if keep_vars else b.detach().clone()
+
+    def _load_from_state_dict(self, state_dict, prefix, local_metadata, strict,
+                              missing_keys, unexpected_keys, error_msgs):
+        h = self.out_channels
+        with torch.no_grad():
+            for i, name in enumerate(self._SEG_NAMES):
+                wk, bk = prefix + name + ".weight", prefix + name + ".bias"
+                if wk in state_dict:
+                    self.w4[i * h:(i + 1) * h, :self.in_channels].copy_(state_dict[wk])
+                    self.w4[i * h:(i + 1) * h, self.in_channels:].zero_()
+                    state_dict = {k: v for k, v in state_dict.items() if k != wk}
+                elif strict:
+                    missing_keys.append(wk)
+                if bk in state_dict:
+                    self.b4[i * h:(i + 1) * h].copy_(state_dict[bk])
+                    state_dict = {k: v for k, v in state_dict.items() if k != bk}
+                elif strict:
+                    missing_keys.append(bk)
+        # let the default machinery handle lin_edge (and flag stray w4/b4)
+        filtered = {k: v for k, v in state_dict.items()
+                    if not any(k == prefix + n + s for n in self._SEG_NAMES
+                               for s in (".weight", ".bias"))}
+        super()._load_from_state_dict(filtered, prefix, local_metadata, strict,
+                                      missing_keys, unexpected_keys, error_msgs)
+        for k in (prefix + "w4", prefix + "b4"):
+            if k in missing_keys:
+                missing_keys.remove(k)
+
     def forward(self, x, edge_index, edge_embeds, csr=None, num_nodes=None):
         n = x.shape[0] if num_nodes is None else num_nodes
-        q = ops.linear(x, self.lin_query.weight, self.lin_query.bias)
-        k = ops.linear(x, self.lin_key.weight, self.lin_key.bias)
-        v = ops.linear(x, self.lin_value.weight, self.lin_value.bias)
+        q = ops.linear(x, self.lin_query.weight.contiguous(), self.lin_query.bias)
+        k = ops.linear(x, self.lin_key.weight.contiguous(), self.lin_key.bias)
+        v = ops.linear(x, self.lin_value.weight.contiguous(), self.lin_value.bias)
         e = ops.linear(edge_embeds, self.lin_edge.weight, None)
-        skip = ops.linear(x, self.lin_skip.weight, self.lin_skip.bias)
+        skip = ops.linear(x, self.lin_skip.weight.contiguous(), self.lin_skip.bias)
         return ops.edge_attention(q, k, v, e, skip, edge_index, n, csr=csr)
 
     def forward_fused(self, x, edge_attr, ifc_weight, rpc_weight, csr,
@@ -70,23 +151,17 @@ class TransformerConv(nn.Module):
         precision with H%256==0 the qkvs tensor is kept bf16-resident
         through the attention kernels."""
         h = self.out_channels
-        w4 = torch.cat([self.lin_query.weight, self.lin_key.weight,
-                        self.lin_value.weight, self.lin_skip.weight], dim=0)
-        b4 = torch.cat([self.lin_query.bias, self.lin_key.bias,
-                        self.lin_value.bias, self.lin_skip.bias], dim=0)
-        # pad an odd contraction dim (layer 1: K = 9 + H + embeds) to a
-        # multiple of 8 so the GEMM staging takes the vectorized interior
-        # path (zero columns are exact; autograd slices the grads back)
-        kdim = x.shape[1]
-        if kdim % 8:
-            pad = 8 - kdim % 8
-            x = torch.nn.functional.pad(x, (0, pad))
-            w4 = torch.nn.functional.pad(w4, (0, pad))
+        # w4/b4 ARE the parameters (no per-step cat); the storage is
+        # pre-padded to k_padded with zero columns (zero-grad columns stay
+        # zero under Adam), so only x needs padding for the vectorized
+        # staging path
+        if x.shape[1] != self.k_padded:
+            x = torch.nn.functional.pad(x, (0, self.k_padded - x.shape[1]))
         if (ops.gemm_precision() == "bf16" and h % 256 == 0
                 and ops.act16_enabled()):
-            qkvs = ops.linear16(x, w4, b4)
+            qkvs = ops.linear16(x, self.w4, self.b4)
         else:
-            qkvs = ops.linear(x, w4, b4)
+            qkvs = ops.linear(x, self.w4, self.b4)
         we = self.lin_edge.weight  # [H, 2H]
         pifc = ops.linear(ifc_weight, we[:, :h].contiguous(), None)
         prpc = ops.linear(rpc_weight, we[:, h:].contiguous(), None)

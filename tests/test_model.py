@@ -70,7 +70,7 @@ def test_forward_shapes_and_grad():
     loss = gp.sum() + 0 * lp.sum()
     loss.backward()
     # gradients reach every trainable leaf that participates
-    assert m.convs[0].lin_query.weight.grad is not None
+    assert m.convs[0].w4.grad is not None  # fused QKVS parameter
     assert m.entry_embeds.weight.grad is not None
     assert m.cat_embedding[0].weight.grad is not None
 
