@@ -31,6 +31,10 @@ void launch_seg_pool_bwd(const float*, const float*, const float*, const long*,
                          float*, long, int, hipStream_t);
 void launch_embed_node_fwd(const float*, const long*, const float*, float*,
                            long, int, int, hipStream_t);
+void launch_embed_node_fwd16(const float*, const long*, const float*, void*,
+                             long, int, int, hipStream_t);
+void launch_vocab_scatter16(const void*, const long*, long, float*, long, int,
+                            int, int, int, hipStream_t);
 void launch_embed_edge_fwd(const long*, const float*, const float*, float*,
                            long, int, int, hipStream_t);
 void launch_gather_rows(const long*, const float*, float*, long, int,
@@ -235,11 +239,19 @@ torch::Tensor seg_pool_bwd(torch::Tensor gout, torch::Tensor probs,
 }
 
 torch::Tensor embed_node_fwd(torch::Tensor x_raw, torch::Tensor idx,
-                             torch::Tensor table) {
+                             torch::Tensor table, bool out16 = false) {
   CHECK_IN(x_raw); CHECK_IN(idx); CHECK_IN(table);
   const long n = x_raw.size(0);
   const int f = x_raw.size(1);
   const int h = table.size(1);
+  if (out16) {
+    auto out = torch::empty({n, f + h},
+                            x_raw.options().dtype(torch::kBFloat16));
+    launch_embed_node_fwd16(x_raw.data_ptr<float>(), idx.data_ptr<long>(),
+                            table.data_ptr<float>(), out.data_ptr(), n, f, h,
+                            cur_stream());
+    return out;
+  }
   auto out = torch::empty({n, f + h}, x_raw.options());
   launch_embed_node_fwd(x_raw.data_ptr<float>(), idx.data_ptr<long>(),
                         table.data_ptr<float>(), out.data_ptr<float>(), n, f,
@@ -614,7 +626,19 @@ torch::Tensor vocab_scatter(torch::Tensor g, torch::Tensor idx, int64_t rows,
   TORCH_CHECK(idx.is_cuda() && idx.dim() == 1, "idx must be 1-D CUDA");
   const size_t lds = (size_t)rows * h * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024, "vocab too large for LDS accumulator");
-  auto dtable = torch::empty({rows, h}, g.options());
+  auto dtable = torch::empty({rows, h}, g.options().dtype(torch::kFloat32));
+  if (g.scalar_type() == torch::kBFloat16) {
+    // wave-private kernel handles bf16 g directly when its tables fit
+    const size_t priv64 = (size_t)4 * rows * 64 * sizeof(float);
+    TORCH_CHECK(h % 64 == 0 && priv64 <= 160 * 1024,
+                "bf16 vocab_scatter needs the wave-private path "
+                "(upcast g to f32 for this shape)");
+    launch_vocab_scatter16(g.data_ptr(), idx.data_ptr<long>(), idx.stride(0),
+                           dtable.data_ptr<float>(), g.size(0), (int)rows,
+                           (int)h, (int)g.size(1), (int)col_off,
+                           cur_stream());
+    return dtable;
+  }
   launch_vocab_scatter(g.data_ptr<float>(), idx.data_ptr<long>(),
                        idx.stride(0), dtable.data_ptr<float>(), g.size(0),
                        (int)rows, (int)h, (int)g.size(1), (int)col_off,
@@ -1023,7 +1047,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("edge_attn_bwd", &edge_attn_bwd);
   mod.def("seg_pool_fwd", &seg_pool_fwd);
   mod.def("seg_pool_bwd", &seg_pool_bwd);
-  mod.def("embed_node_fwd", &embed_node_fwd);
+  mod.def("embed_node_fwd", &embed_node_fwd, py::arg("x_raw"), py::arg("idx"), py::arg("table"), py::arg("out16") = false);
   mod.def("embed_edge_fwd", &embed_edge_fwd);
   mod.def("gather_rows", &gather_rows);
   mod.def("bn_relu_fwd16", &bn_relu_fwd16);

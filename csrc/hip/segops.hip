@@ -161,10 +161,11 @@ void launch_seg_pool_bwd16(const float* gout, const float* probs,
 // embedding gathers fused with concat
 // ---------------------------------------------------------------------------
 
+template <typename TY = float>
 __global__ void embed_node_fwd_kernel(const float* __restrict__ x_raw,
                                       const long* __restrict__ idx,
                                       const float* __restrict__ table,
-                                      float* __restrict__ out, long n, int f,
+                                      TY* __restrict__ out, long n, int f,
                                       int h) {
   const int w = f + h;
   const long numel = n * w;
@@ -173,7 +174,8 @@ __global__ void embed_node_fwd_kernel(const float* __restrict__ x_raw,
   for (long t = i0; t < numel; t += stride) {
     const long row = t / w;
     const int c = (int)(t - row * w);
-    out[t] = (c < f) ? x_raw[row * f + c] : table[idx[row] * h + (c - f)];
+    out[t] = (TY)((c < f) ? x_raw[row * f + c]
+                          : table[idx[row] * h + (c - f)]);
   }
 }
 
@@ -571,8 +573,8 @@ void launch_vocab_scatter_dual16(const void* g, const long* ea, int astride,
 
 // single-table wave-private variant (same rationale/measurements as the dual
 // kernel above: plain LDS read+add+write beats ds_add_f32 ~7x)
-template <int U, int HH>
-__global__ void vocab_scatter_priv_kernel(const float* __restrict__ g,
+template <int U, int HH, typename GT = float>
+__global__ void vocab_scatter_priv_kernel(const GT* __restrict__ g,
                                           const long* __restrict__ idx,
                                           long idx_stride,
                                           float* __restrict__ dtable, long n,
@@ -601,7 +603,7 @@ __global__ void vocab_scatter_priv_kernel(const float* __restrict__ g,
       v[u] = idx[(r + u * step) * idx_stride];
 #pragma unroll
       for (int k = 0; k < CPL; ++k)
-        xv[u][k] = g[(r + u * step) * gstride + col_off + c0 + lane * CPL + k];
+        xv[u][k] = (float)g[(r + u * step) * gstride + col_off + c0 + lane * CPL + k];
     }
 #pragma unroll
     for (int u = 0; u < U; ++u)
@@ -612,7 +614,8 @@ __global__ void vocab_scatter_priv_kernel(const float* __restrict__ g,
     const long v = idx[r * idx_stride];
 #pragma unroll
     for (int k = 0; k < CPL; ++k)
-      my[v * HH + lane * CPL + k] += g[r * gstride + col_off + c0 + lane * CPL + k];
+      my[v * HH + lane * CPL + k] +=
+          (float)g[r * gstride + col_off + c0 + lane * CPL + k];
   }
   __syncthreads();
   for (long t = threadIdx.x; t < vwh; t += blockDim.x)
@@ -622,6 +625,40 @@ __global__ void vocab_scatter_priv_kernel(const float* __restrict__ g,
     const float val = acc[t];
     if (val != 0.f) atomicAdd(&dtable[(t / HH) * h + c0 + t % HH], val);
   }
+}
+
+// bf16-gradient variant: wave-private path only (the caller upcasts to f32
+// when the private tables do not fit)
+void launch_vocab_scatter16(const void* g_v, const long* idx, long idx_stride,
+                            float* dtable, long n, int rows, int h,
+                            int gstride, int col_off, hipStream_t s) {
+  const __bf16* g = (const __bf16*)g_v;
+  HIP_CHECK(hipMemsetAsync(dtable, 0, (long)rows * h * sizeof(float), s));
+  if (n == 0) return;
+  const int blocks = (int)min((long)128, (n + 255) / 256);
+  const size_t lds128 = (size_t)WAVES_PER_BLOCK * rows * 128 * sizeof(float);
+  const size_t lds64 = (size_t)WAVES_PER_BLOCK * rows * 64 * sizeof(float);
+  if (h % 128 == 0 && lds128 <= 160 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        (const void*)vocab_scatter_priv_kernel<4, 128, __bf16>,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds128));
+    vocab_scatter_priv_kernel<4, 128, __bf16>
+        <<<dim3(max(blocks, 1), h / 128), dim3(WAVES_PER_BLOCK * PERTGNN_WAVE),
+           lds128, s>>>(g, idx, idx_stride, dtable, n, rows, h, gstride,
+                        col_off);
+    return;
+  }
+  if (h % 64 == 0 && lds64 <= 160 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        (const void*)vocab_scatter_priv_kernel<4, 64, __bf16>,
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds64));
+    vocab_scatter_priv_kernel<4, 64, __bf16>
+        <<<dim3(max(blocks, 1), h / 64), dim3(WAVES_PER_BLOCK * PERTGNN_WAVE),
+           lds64, s>>>(g, idx, idx_stride, dtable, n, rows, h, gstride,
+                       col_off);
+    return;
+  }
+  abort();  // caller guarantees fit (python upcasts otherwise)
 }
 
 void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
@@ -713,6 +750,16 @@ void launch_embed_node_fwd(const float* x_raw, const long* idx,
   if (n == 0) return;
   embed_node_fwd_kernel<<<grid_for(n * (f + h)), 256, 0, s>>>(x_raw, idx, table,
                                                               out, n, f, h);
+}
+
+void launch_embed_node_fwd16(const float* x_raw, const long* idx,
+                             const float* table, void* out, long n, int f,
+                             int h, hipStream_t stream) {
+  const long numel = n * (long)(f + h);
+  const int tpb = 256;
+  const int blocks = (int)min((numel + tpb - 1) / tpb, (long)4096);
+  embed_node_fwd_kernel<<<dim3(blocks), dim3(tpb), 0, stream>>>(
+      x_raw, idx, table, (__bf16*)out, n, f, h);
 }
 void launch_embed_edge_fwd(const long* attr, const float* ifc,
                            const float* rpc, float* out, long e, int h,

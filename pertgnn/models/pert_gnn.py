@@ -259,7 +259,12 @@ class SAGEDeterministic(nn.Module):
         from ..ops.backend import use_hip
 
         fused = csr is not None and use_hip(x)
-        x = ops.embed_concat_node(x, cat_X, [t.weight for t in self.cat_embedding])
+        hidden = self.bns[0].weight.shape[0] if len(self.bns) else 0
+        out16 = (fused and ops.gemm_precision() == "bf16"
+                 and hidden % 256 == 0 and ops.act16_enabled())
+        x = ops.embed_concat_node(x, cat_X,
+                                  [t.weight for t in self.cat_embedding],
+                                  out16=out16)
         edge_embeds = None
         if not fused:
             edge_embeds = ops.embed_concat_edge(
@@ -275,12 +280,9 @@ class SAGEDeterministic(nn.Module):
                 )
             return conv(x, edge_index, edge_embeds, csr=csr, num_nodes=n)
 
-        # act16: BN emits bf16 activations so the next conv's fused QKVS GEMM
-        # (and its backward) reads/writes 16-bit streams; gate matches the
-        # linear16 gate in TransformerConv.forward_fused so dtypes line up.
-        hidden = self.bns[0].weight.shape[0] if len(self.bns) else 0
-        out16 = (fused and ops.gemm_precision() == "bf16"
-                 and hidden % 256 == 0 and ops.act16_enabled())
+        # act16: BN (and the embed concat above) emit bf16 activations so
+        # every conv's fused QKVS GEMM and its backward read/write 16-bit
+        # streams; gate matches the linear16 gate in forward_fused.
         for i, conv in enumerate(self.convs[:-1]):
             x = run_conv(conv, x, out16=out16)
             bn = self.bns[i]

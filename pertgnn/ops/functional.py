@@ -31,10 +31,17 @@ def deterministic() -> bool:
 def _table_grad(m, g, idx, rows, h, col_off):
     """dtable[v] = segment-sum of g[:, col_off:col_off+h] by idx — LDS vocab
     accumulator for small tables, deterministic two-phase grouped scatter
-    otherwise (always, under PERTGNN_DETERMINISTIC=1)."""
+    otherwise (always, under PERTGNN_DETERMINISTIC=1).  bf16 g is consumed
+    directly by the wave-private kernel when its tables fit; other paths
+    upcast."""
+    if g.dtype == torch.bfloat16 and (
+            deterministic() or h % 64 != 0 or 4 * rows * 64 * 4 > 160 * 1024):
+        g = g.float()
     if rows * h * 4 <= 160 * 1024 and not deterministic():
         return m.vocab_scatter(g, idx, rows, h, col_off)
     order, ptr = _group_by(idx.contiguous(), rows)
+    if g.dtype != torch.float32:
+        g = g.float()
     return m.embed_grouped_scatter(g, order, ptr, rows, h, col_off)
 
 
@@ -249,9 +256,9 @@ def pattern_pool(x, pattern_probs, pattern_num_nodes, batch, num_graphs):
 
 class _EmbedNodeFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x_raw, cat_idx, table):
+    def forward(ctx, x_raw, cat_idx, table, out16=False):
         m = ext()
-        out = m.embed_node_fwd(x_raw, cat_idx, table)
+        out = m.embed_node_fwd(x_raw, cat_idx, table, out16)
         ctx.save_for_backward(cat_idx)
         ctx.f = x_raw.shape[1]
         ctx.rows = table.shape[0]
@@ -263,9 +270,11 @@ class _EmbedNodeFn(torch.autograd.Function):
         m = ext()
         g = g.contiguous()
         dx_raw = g[:, : ctx.f].contiguous()
+        if dx_raw.dtype != torch.float32:
+            dx_raw = dx_raw.float()
         h = g.shape[1] - ctx.f
         dtable = _table_grad(m, g, cat_idx, ctx.rows, h, ctx.f)
-        return dx_raw, None, dtable
+        return dx_raw, None, dtable, None
 
 
 class _EmbedEdgeFn(torch.autograd.Function):
@@ -288,9 +297,12 @@ class _EmbedEdgeFn(torch.autograd.Function):
         return None, d_ifc, d_rpc
 
 
-def embed_concat_node(x_raw, cat_X, tables):
+def embed_concat_node(x_raw, cat_X, tables, out16=False):
+    """``out16`` emits the concat bf16 so layer 1's QKVS GEMM takes the
+    bf16-A path — numerically identical to the fp32 tensor (the GEMM staging
+    rounds operands to bf16 either way)."""
     if len(tables) == 1 and use_hip(x_raw):
-        return _EmbedNodeFn.apply(x_raw, cat_X[:, 0].contiguous(), tables[0])
+        return _EmbedNodeFn.apply(x_raw, cat_X[:, 0].contiguous(), tables[0], out16)
     return ref.embed_concat_node(x_raw, cat_X, tables)
 
 
