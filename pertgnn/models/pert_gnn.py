@@ -24,6 +24,22 @@ from torch import nn
 from ..ops import functional as ops
 
 
+class _EdgeLinearView:
+    """Read-through view of the conv's split edge-projection parameters
+    (``we_ifc``/``we_rpc``) as the reference's single ``lin_edge.weight``
+    [H, 2H] (bias=False)."""
+
+    def __init__(self, conv):
+        object.__setattr__(self, "_conv", conv)
+
+    @property
+    def weight(self):
+        c = self._conv
+        return torch.cat([c.we_ifc, c.we_rpc], dim=1)
+
+    bias = None
+
+
 class _SegLinearView:
     """Read-through view of one H-row segment of the conv's fused ``w4``/``b4``
     parameters — keeps the PyG-style ``conv.lin_query.weight`` access (and the
@@ -80,7 +96,13 @@ class TransformerConv(nn.Module):
         self.lin_key = _SegLinearView(self, 1)
         self.lin_value = _SegLinearView(self, 2)
         self.lin_skip = _SegLinearView(self, 3)
-        self.lin_edge = nn.Linear(edge_dim, heads * out_channels, bias=False)
+        # lin_edge [H, 2H] stored as the two halves the fused path consumes
+        # (P_ifc = ifc_table @ we_ifc^T, P_rpc = rpc_table @ we_rpc^T) — no
+        # per-step slicing or slice-grad padding
+        assert edge_dim == 2 * out_channels
+        self.we_ifc = nn.Parameter(torch.empty(out_channels, out_channels))
+        self.we_rpc = nn.Parameter(torch.empty(out_channels, out_channels))
+        self.lin_edge = _EdgeLinearView(self)
         self.reset_parameters()
 
     def reset_parameters(self):
@@ -88,24 +110,36 @@ class TransformerConv(nn.Module):
         with torch.no_grad():
             self.w4.zero_()
             self.b4.zero_()
-        for lin in (self.lin_key, self.lin_query, self.lin_value, self.lin_edge, self.lin_skip):
+        for lin in (self.lin_key, self.lin_query, self.lin_value, self.lin_skip):
             nn.init.xavier_uniform_(lin.weight)
             if lin.bias is not None:
                 nn.init.zeros_(lin.bias)
+        # glorot over the FULL [H, 2H] edge matrix (fan_in = 2H, matching the
+        # reference's single lin_edge), then split into the stored halves
+        h = self.out_channels
+        w = torch.empty(h, 2 * h)
+        nn.init.xavier_uniform_(w)
+        with torch.no_grad():
+            self.we_ifc.copy_(w[:, :h])
+            self.we_rpc.copy_(w[:, h:])
 
     _SEG_NAMES = ("lin_query", "lin_key", "lin_value", "lin_skip")
 
     def _save_to_state_dict(self, destination, prefix, keep_vars):
-        # emit the reference's per-projection keys instead of w4/b4
+        # emit the reference's per-projection keys instead of w4/b4/we_*
         super()._save_to_state_dict(destination, prefix, keep_vars)
         del destination[prefix + "w4"]
         del destination[prefix + "b4"]
+        del destination[prefix + "we_ifc"]
+        del destination[prefix + "we_rpc"]
         for i, name in enumerate(self._SEG_NAMES):
             view = getattr(self, name)
             w = view.weight
             b = view.bias
             destination[prefix + name + ".weight"] = w if keep_vars else w.detach().clone()
             destination[prefix + name + ".bias"] = b if keep_vars else b.detach().clone()
+        we = self.lin_edge.weight
+        destination[prefix + "lin_edge.weight"] = we if keep_vars else we.detach()
 
     def _load_from_state_dict(self, state_dict, prefix, local_metadata, strict,
                               missing_keys, unexpected_keys, error_msgs):
@@ -124,13 +158,22 @@ class TransformerConv(nn.Module):
                     state_dict = {k: v for k, v in state_dict.items() if k != bk}
                 elif strict:
                     missing_keys.append(bk)
-        # let the default machinery handle lin_edge (and flag stray w4/b4)
+        we_key = prefix + "lin_edge.weight"
+        if we_key in state_dict:
+            with torch.no_grad():
+                w = state_dict[we_key]
+                self.we_ifc.copy_(w[:, :h])
+                self.we_rpc.copy_(w[:, h:])
+        elif strict:
+            missing_keys.append(we_key)
         filtered = {k: v for k, v in state_dict.items()
-                    if not any(k == prefix + n + s for n in self._SEG_NAMES
-                               for s in (".weight", ".bias"))}
+                    if k != we_key
+                    and not any(k == prefix + n + s for n in self._SEG_NAMES
+                                for s in (".weight", ".bias"))}
         super()._load_from_state_dict(filtered, prefix, local_metadata, strict,
                                       missing_keys, unexpected_keys, error_msgs)
-        for k in (prefix + "w4", prefix + "b4"):
+        for k in (prefix + "w4", prefix + "b4", prefix + "we_ifc",
+                  prefix + "we_rpc"):
             if k in missing_keys:
                 missing_keys.remove(k)
 
@@ -162,9 +205,8 @@ class TransformerConv(nn.Module):
             qkvs = ops.linear16(x, self.w4, self.b4)
         else:
             qkvs = ops.linear(x, self.w4, self.b4)
-        we = self.lin_edge.weight  # [H, 2H]
-        pifc = ops.linear(ifc_weight, we[:, :h].contiguous(), None)
-        prpc = ops.linear(rpc_weight, we[:, h:].contiguous(), None)
+        pifc = ops.linear(ifc_weight, self.we_ifc, None)
+        prpc = ops.linear(rpc_weight, self.we_rpc, None)
         return ops.edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr,
                                         out16=out16)
 

@@ -533,7 +533,7 @@ def test_model_act16_close_to_fp32():
     assert rel < 0.05, rel
     assert abs(l16 - l32) / max(abs(l32), 1e-6) < 0.05
     # gradient direction must agree (cosine) for the big weights
-    for n in ("convs.1.w4", "convs.0.lin_edge.weight",
+    for n in ("convs.1.w4", "convs.0.we_ifc",
               "cat_embedding.0.weight"):
         a, c = g16[n].flatten(), g32[n].flatten()
         cos = torch.dot(a, c) / (a.norm() * c.norm()).clamp_min(1e-12)
