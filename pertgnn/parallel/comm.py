@@ -64,10 +64,16 @@ class Comm:
         return float(t.item())
 
     def broadcast_module_(self, module: torch.nn.Module, src: int = 0):
-        """Broadcast initial weights + buffers from rank src."""
+        """Broadcast initial weights + buffers from rank src.
+
+        Iterates parameters()/buffers() directly — NOT state_dict(), whose
+        values can be detached re-mapped clones (the conv's fused w4/b4 are
+        exported under the reference's per-projection keys) that an in-place
+        broadcast would silently not write back."""
         if not self.distributed:
             return
-        for p in module.state_dict().values():
+        import itertools
+        for p in itertools.chain(module.parameters(), module.buffers()):
             # skip lazily-uninitialized params (the model's dead edge_linear)
             if isinstance(p, torch.nn.parameter.UninitializedParameter) or \
                isinstance(p, torch.nn.parameter.UninitializedBuffer):

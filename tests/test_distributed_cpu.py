@@ -326,3 +326,52 @@ def test_ddp_training_equals_single_process_big_batch():
     # the per-step losses above (1e-4); params stay in the same neighborhood.
     ref_flat = opt.flat_param.numpy()
     assert np.allclose(params0, ref_flat, atol=2e-2), np.abs(params0 - ref_flat).max()
+
+
+def _worker_bcast(rank, world_size, port, result_queue):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world_size), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    from pertgnn.parallel import Comm
+
+    comm = Comm(backend="gloo")
+    torch.manual_seed(1000 + rank)  # DIFFERENT init per rank on purpose
+    model = _build_model()
+    comm.broadcast_module_(model)
+    params = {n: p.detach().numpy().copy() for n, p in model.named_parameters()
+              if not isinstance(p, torch.nn.parameter.UninitializedParameter)
+              and p.numel() > 0}
+    bufs = {n: b.detach().numpy().copy() for n, b in model.named_buffers()
+            if torch.is_tensor(b) and b.numel() > 0}
+    result_queue.put((rank, params, bufs))
+    comm.barrier()
+    comm.finalize()
+
+
+@pytest.mark.timeout(120)
+def test_broadcast_module_syncs_divergent_init():
+    """Regression: broadcast_module_ must write the REAL parameter storage
+    (the conv's fused w4/b4 are exported in state_dict as remapped clones —
+    broadcasting those would silently leave ranks desynchronized)."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_bcast, args=(r, world, 29516, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, params, bufs = q.get(timeout=110)
+        results[rank] = (params, bufs)
+    for p in procs:
+        p.join(timeout=110)
+        assert p.exitcode == 0
+    p0, b0 = results[0]
+    p1, b1 = results[1]
+    assert set(p0) == set(p1)
+    import numpy as np
+    for n in p0:
+        assert np.array_equal(p0[n], p1[n]), f"param {n} not synced"
+    for n in b0:
+        assert np.array_equal(b0[n], b1[n]), f"buffer {n} not synced"
