@@ -162,6 +162,7 @@ def main():
     # per-kernel launch gaps (guide: capture launch-bound inner loops).
     timed_step = step
     if on_gpu and not args.no_hipgraph:
+        captured = False
         try:
             for i in range(len(batches)):
                 step(i)  # allocation warmup per batch shape
@@ -176,11 +177,15 @@ def main():
                     pool = g.pool()
                 gobjs.append(g)
             torch.cuda.synchronize()
-
-            def timed_step(i):
-                gobjs[i % len(gobjs)].replay()
+            captured = True
         except Exception as exc:  # pragma: no cover
             print(f"# hipGraph capture unavailable ({exc}); stepping eagerly")
+        # all ranks must agree on the stepping mode (a rank that failed
+        # capture must not leave the others replaying graphs alone)
+        if comm.all_reduce_scalar(1.0 if captured else 0.0, op="min") >= 1.0:
+            def timed_step(i):
+                gobjs[i % len(gobjs)].replay()
+        else:
             timed_step = step
 
     for i in range(args.warmup):

@@ -45,7 +45,8 @@ class Comm:
     def all_reduce_(self, t: torch.Tensor, op: str = "sum", async_op: bool = False):
         if not self.distributed:
             return None
-        red = dist.ReduceOp.SUM if op == "sum" else dist.ReduceOp.MAX
+        red = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
+               "min": dist.ReduceOp.MIN}[op]
         return dist.all_reduce(t, op=red, async_op=async_op)
 
     def broadcast_(self, t: torch.Tensor, src: int = 0):
