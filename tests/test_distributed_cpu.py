@@ -375,3 +375,56 @@ def test_broadcast_module_syncs_divergent_init():
         assert np.array_equal(p0[n], p1[n]), f"param {n} not synced"
     for n in b0:
         assert np.array_equal(b0[n], b1[n]), f"buffer {n} not synced"
+
+
+def _worker_bench(rank, world_size, port, result_queue):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world_size), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    import io
+    import sys as _sys
+    import contextlib
+
+    argv = ["bench.py", "--steps", "2", "--warmup", "1", "--batch-size", "6",
+            "--n-batches", "1", "--layers", "2", "--hidden", "32"]
+    out = io.StringIO()
+    old = _sys.argv
+    _sys.argv = argv
+    try:
+        import bench as bench_mod
+        with contextlib.redirect_stdout(out):
+            bench_mod.main()
+    finally:
+        _sys.argv = old
+    result_queue.put((rank, out.getvalue()))
+
+
+@pytest.mark.timeout(180)
+def test_bench_main_world2_gloo():
+    """The driver's scaling run executes bench.py with WORLD_SIZE>1 — run the
+    EXACT same main() at world_size=2 over gloo on CPU: vocab-stat sync,
+    weight broadcast, flat-grad engine, barriers, max-over-ranks timing and
+    the rank-0 JSON contract."""
+    import json
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_bench, args=(r, world, 29517, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    outs = {}
+    for _ in range(world):
+        rank, text = q.get(timeout=170)
+        outs[rank] = text
+    for p in procs:
+        p.join(timeout=170)
+        assert p.exitcode == 0
+    # rank 0 prints exactly one JSON line; rank 1 prints nothing
+    line = outs[0].strip().splitlines()[-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 12
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert outs[1].strip() == ""
