@@ -88,6 +88,12 @@ void launch_gemm_bf16_nn_a16o16(const void*, const float*, void*, int, int,
                                 int, hipStream_t);
 void launch_gemm_bf16_tn_a16b16(const void*, const void*, float*, float*, int,
                                 int, int, hipStream_t);
+void launch_gemm_fp16_nt_a16o16(const void*, const float*, const float*,
+                                void*, int, int, int, hipStream_t);
+void launch_gemm_fp16_nn_a16o16(const void*, const float*, void*, int, int,
+                                int, hipStream_t);
+void launch_gemm_fp16_tn_a16b16(const void*, const void*, float*, float*, int,
+                                int, int, hipStream_t);
 void launch_quantile_loss_fwd(const float*, const float*, float*, long, float,
                               hipStream_t);
 void launch_quantile_loss_bwd(const float*, const float*, const float*, float*,
@@ -971,7 +977,7 @@ torch::Tensor seg_pool_bwd16(torch::Tensor gout, torch::Tensor probs,
 }
 
 torch::Tensor linear_fwd_a16o16(torch::Tensor x, torch::Tensor w,
-                                torch::Tensor b) {
+                                torch::Tensor b, bool fp16c = false) {
   CHECK_IN(x); CHECK_IN(w);
   const int m = x.size(0);
   const int k = x.size(1);
@@ -979,24 +985,34 @@ torch::Tensor linear_fwd_a16o16(torch::Tensor x, torch::Tensor w,
   auto y = torch::empty({m, n}, x.options());  // bf16
   const float* bias = nullptr;
   if (b.defined() && b.numel() > 0) bias = b.data_ptr<float>();
-  launch_gemm_bf16_nt_a16o16(x.data_ptr(), w.data_ptr<float>(), bias,
-                             y.data_ptr(), m, n, k, cur_stream());
+  if (fp16c)
+    launch_gemm_fp16_nt_a16o16(x.data_ptr(), w.data_ptr<float>(), bias,
+                               y.data_ptr(), m, n, k, cur_stream());
+  else
+    launch_gemm_bf16_nt_a16o16(x.data_ptr(), w.data_ptr<float>(), bias,
+                               y.data_ptr(), m, n, k, cur_stream());
   return y;
 }
 
-torch::Tensor linear_dgrad16_o16(torch::Tensor g, torch::Tensor w) {
+torch::Tensor linear_dgrad16_o16(torch::Tensor g, torch::Tensor w,
+                                 bool fp16c = false) {
   CHECK_IN(g); CHECK_IN(w);
   const int m = g.size(0);
   const int n = w.size(0);
   const int k = w.size(1);
   auto dx = torch::empty({m, k}, g.options());  // bf16
-  launch_gemm_bf16_nn_a16o16(g.data_ptr(), w.data_ptr<float>(), dx.data_ptr(),
-                             m, n, k, cur_stream());
+  if (fp16c)
+    launch_gemm_fp16_nn_a16o16(g.data_ptr(), w.data_ptr<float>(),
+                               dx.data_ptr(), m, n, k, cur_stream());
+  else
+    launch_gemm_bf16_nn_a16o16(g.data_ptr(), w.data_ptr<float>(),
+                               dx.data_ptr(), m, n, k, cur_stream());
   return dx;
 }
 
 std::vector<torch::Tensor> linear_wgrad16_b16(torch::Tensor g, torch::Tensor x,
-                                              bool has_bias) {
+                                              bool has_bias,
+                                              bool fp16c = false) {
   CHECK_IN(g); CHECK_IN(x);
   const int m = x.size(0);
   const int k = x.size(1);
@@ -1009,8 +1025,14 @@ std::vector<torch::Tensor> linear_wgrad16_b16(torch::Tensor g, torch::Tensor x,
     db = torch::empty({n}, fopt);
     db_ptr = db.data_ptr<float>();
   }
-  launch_gemm_bf16_tn_a16b16(g.data_ptr(), x.data_ptr(), dw.data_ptr<float>(),
-                             db_ptr, m, n, k, cur_stream());
+  if (fp16c)
+    launch_gemm_fp16_tn_a16b16(g.data_ptr(), x.data_ptr(),
+                               dw.data_ptr<float>(), db_ptr, m, n, k,
+                               cur_stream());
+  else
+    launch_gemm_bf16_tn_a16b16(g.data_ptr(), x.data_ptr(),
+                               dw.data_ptr<float>(), db_ptr, m, n, k,
+                               cur_stream());
   return {dw, db};
 }
 
@@ -1057,9 +1079,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_bwd_apply16", &bn_bwd_apply16);
   mod.def("seg_pool_fwd16", &seg_pool_fwd16);
   mod.def("seg_pool_bwd16", &seg_pool_bwd16);
-  mod.def("linear_fwd_a16o16", &linear_fwd_a16o16);
-  mod.def("linear_dgrad16_o16", &linear_dgrad16_o16);
-  mod.def("linear_wgrad16_b16", &linear_wgrad16_b16);
+  mod.def("linear_fwd_a16o16", &linear_fwd_a16o16, py::arg("x"), py::arg("w"),
+          py::arg("b"), py::arg("fp16c") = false);
+  mod.def("linear_dgrad16_o16", &linear_dgrad16_o16, py::arg("g"),
+          py::arg("w"), py::arg("fp16c") = false);
+  mod.def("linear_wgrad16_b16", &linear_wgrad16_b16, py::arg("g"),
+          py::arg("x"), py::arg("has_bias"), py::arg("fp16c") = false);
   mod.def("bn_relu_fwd", &bn_relu_fwd);
   mod.def("bn_relu_bwd", &bn_relu_bwd);
   mod.def("quantile_loss_fwd", &quantile_loss_fwd);

@@ -796,3 +796,67 @@ void launch_gemm_bf16_tn_a16b16(const void* a_v, const void* b_v, float* c,
                                                               m, n, k2,
                                                               slices);
 }
+
+// fp16-COMPUTE variants of the act16 launchers: activation streams stay bf16
+// (storage dtype is independent of the MFMA operand dtype — staging converts
+// while packing), matrix cores run v_mfma_f32_16x16x32_f16.
+void launch_gemm_fp16_nt_a16o16(const void* a_v, const float* b,
+                                const float* bias, void* c_v, int m, int n,
+                                int k, hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  __bf16* c = (__bf16*)c_v;
+  if (m >= 512 && n >= 128) {
+    const int grid = ((m + 127) / 128) * ((n + 127) / 128);
+    gemm_bf16_nt_kernel<128, 128, 32, _Float16, __bf16, __bf16>
+        <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(a, b, bias, c, m, n, k, 0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((n + 63) / 64);
+    gemm_bf16_nt_kernel<64, 64, 64, _Float16, __bf16, __bf16>
+        <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(a, b, bias, c, m, n, k, 0);
+  }
+}
+
+void launch_gemm_fp16_nn_a16o16(const void* a_v, const float* b, void* c_v,
+                                int m, int n, int k2, hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  __bf16* c = (__bf16*)c_v;
+  if (m >= 512 && k2 >= 128) {
+    const int grid = ((m + 127) / 128) * ((k2 + 127) / 128);
+    gemm_bf16_nn_kernel<128, 128, 32, _Float16, __bf16, __bf16>
+        <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(a, b, nullptr, c, m, n, k2,
+                                                    0);
+  } else {
+    const int grid = ((m + 63) / 64) * ((k2 + 63) / 64);
+    gemm_bf16_nn_kernel<64, 64, 64, _Float16, __bf16, __bf16>
+        <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(a, b, nullptr, c, m, n, k2,
+                                                    0);
+  }
+}
+
+void launch_gemm_fp16_tn_a16b16(const void* a_v, const void* b_v, float* c,
+                                float* dbias, int m, int n, int k2,
+                                hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  const __bf16* b = (const __bf16*)b_v;
+  const bool big = (n >= 128 && k2 >= 128);
+  const int bm = big ? 128 : 64;
+  const int bn = big ? 128 : 64;
+  const int tiles = ((n + bm - 1) / bm) * ((k2 + bn - 1) / bn);
+  int slices = 1;
+  while (!pertgnn_deterministic() && tiles * slices < 512 && slices < 64 &&
+         (long)slices * 64 * 4 < m)
+    slices *= 2;
+  if (slices > 1)
+    HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
+  if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
+  if (big)
+    gemm_bf16_tn_kernel<128, 128, 32, _Float16, __bf16, __bf16>
+        <<<dim3(tiles * slices), dim3(BGEMM_THREADS), 0, s>>>(a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+  else
+    gemm_bf16_tn_kernel<64, 64, 64, _Float16, __bf16, __bf16>
+        <<<dim3(tiles * slices), dim3(BGEMM_THREADS), 0, s>>>(a, b, c, dbias,
+                                                              m, n, k2,
+                                                              slices);
+}

@@ -200,7 +200,7 @@ class TransformerConv(nn.Module):
         # staging path
         if x.shape[1] != self.k_padded:
             x = torch.nn.functional.pad(x, (0, self.k_padded - x.shape[1]))
-        if (ops.gemm_precision() == "bf16" and h % 256 == 0
+        if (ops.gemm_precision() in ("bf16", "fp16") and h % 256 == 0
                 and ops.act16_enabled()):
             qkvs = ops.linear16(x, self.w4, self.b4)
         else:
@@ -302,7 +302,7 @@ class SAGEDeterministic(nn.Module):
 
         fused = csr is not None and use_hip(x)
         hidden = self.bns[0].weight.shape[0] if len(self.bns) else 0
-        out16 = (fused and ops.gemm_precision() == "bf16"
+        out16 = (fused and ops.gemm_precision() in ("bf16", "fp16")
                  and hidden % 256 == 0 and ops.act16_enabled())
         x = ops.embed_concat_node(x, cat_X,
                                   [t.weight for t in self.cat_embedding],

@@ -116,12 +116,14 @@ class _Linear16Fn(torch.autograd.Function):
     def forward(ctx, x, w, b):
         m = ext()
         bb = b if b is not None else torch.empty(0, dtype=torch.float32, device=x.device)
+        fp16c = gemm_precision() == "fp16"  # fp16 matrix cores, bf16 IO
         if x.dtype == torch.bfloat16:
-            y = m.linear_fwd_a16o16(x, w, bb)   # x16 in, bf16 out
+            y = m.linear_fwd_a16o16(x, w, bb, fp16c)   # x16 in, bf16 out
         else:
             y = m.linear_fwd_bf16_o16(x, w, bb)
         ctx.save_for_backward(x, w)
         ctx.has_bias = b is not None
+        ctx.fp16c = fp16c
         return y
 
     @staticmethod
@@ -130,8 +132,8 @@ class _Linear16Fn(torch.autograd.Function):
         m = ext()
         g = g.contiguous()
         if x.dtype == torch.bfloat16:
-            dx = m.linear_dgrad16_o16(g, w)       # bf16 dx for the x16 producer
-            dw, db = m.linear_wgrad16_b16(g, x, ctx.has_bias)
+            dx = m.linear_dgrad16_o16(g, w, ctx.fp16c)
+            dw, db = m.linear_wgrad16_b16(g, x, ctx.has_bias, ctx.fp16c)
         else:
             dx = m.linear_dgrad16(g, w)
             dw, db = m.linear_wgrad16(g, x, ctx.has_bias)

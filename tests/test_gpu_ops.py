@@ -495,9 +495,11 @@ def test_fused_attention_empty_and_isolated_rows():
     assert torch.allclose(out, qkvs[:, 3 * h:], atol=1e-6)
 
 
-def test_model_act16_close_to_fp32():
-    """bf16-resident-activation mode (H=256, the flagship shape) stays close
-    to the fp32 path end-to-end, forward AND gradients."""
+@pytest.mark.parametrize("prec", ["bf16", "fp16"])
+def test_model_act16_close_to_fp32(prec):
+    """16-bit-activation mode (H=256, the flagship shape; bf16 or fp16 matrix
+    cores over bf16 streams) stays close to the fp32 path end-to-end,
+    forward AND gradients."""
     require_ext()
     import copy
     from pertgnn.models import SAGEDeterministic
@@ -525,7 +527,7 @@ def test_model_act16_close_to_fp32():
     try:
         set_gemm_precision("fp32")
         gp32, l32, g32 = run(m32)
-        set_gemm_precision("bf16")
+        set_gemm_precision(prec)
         gp16, l16, g16 = run(model)
     finally:
         set_gemm_precision("fp32")
