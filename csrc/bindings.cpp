@@ -101,7 +101,7 @@ void launch_quantile_loss_bwd(const float*, const float*, const float*, float*,
 void launch_eval_metrics(const float*, const float*, float*, long, float,
                          hipStream_t);
 void launch_adam(float*, const float*, float*, float*, float*, long, float,
-                 float, float, float, hipStream_t);
+                 float, float, float, float, hipStream_t);
 void launch_gemm_f32_nt(const float*, const float*, const float*, float*, int,
                         int, int, bool, hipStream_t);
 void launch_gemm_f32_nn(const float*, const float*, const float*, float*, int,
@@ -361,11 +361,11 @@ std::vector<torch::Tensor> eval_metrics(torch::Tensor y, torch::Tensor y_hat,
 
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor state, double lr, double b1,
-               double b2, double eps) {
+               double b2, double eps, double gscale = 1.0) {
   CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v); CHECK_IN(state);
   launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
               v.data_ptr<float>(), state.data_ptr<float>(), p.numel(),
-              (float)lr, (float)b1, (float)b2, (float)eps, cur_stream());
+              (float)lr, (float)b1, (float)b2, (float)eps, (float)gscale, cur_stream());
 }
 
 // y = x @ w^T + b (torch Linear layout: w [out,in]); bf16 variant rounds
@@ -1090,5 +1090,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("quantile_loss_fwd", &quantile_loss_fwd);
   mod.def("quantile_loss_bwd", &quantile_loss_bwd);
   mod.def("eval_metrics", &eval_metrics);
-  mod.def("adam_step", &adam_step);
+  mod.def("adam_step", &adam_step, py::arg("p"), py::arg("g"), py::arg("m"),
+          py::arg("v"), py::arg("state"), py::arg("lr"), py::arg("b1"),
+          py::arg("b2"), py::arg("eps"), py::arg("gscale") = 1.0);
 }

@@ -1425,13 +1425,14 @@ __global__ void adam_tick_kernel(float* __restrict__ state, float b1,
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             const float* __restrict__ state, long numel,
-                            float lr, float b1, float b2, float eps) {
+                            float lr, float b1, float b2, float eps,
+                            float gscale) {
   const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   const float step_size = lr / state[1];
   const float sqrt_bias2 = sqrtf(state[2]);
   for (long t = i0; t < numel; t += stride) {
-    const float gt = g[t];
+    const float gt = g[t] * gscale;  // static loss-scaling unscale
     const float mt = b1 * m[t] + (1.f - b1) * gt;
     const float vt = b2 * v[t] + (1.f - b2) * gt * gt;
     m[t] = mt;
@@ -1442,10 +1443,10 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
 }
 
 void launch_adam(float* p, const float* g, float* m, float* v, float* state,
-                 long numel, float lr, float b1, float b2, float eps,
+                 long numel, float lr, float b1, float b2, float eps, float gscale,
                  hipStream_t s) {
   if (numel == 0) return;
   adam_tick_kernel<<<1, 1, 0, s>>>(state, b1, b2);
   adam_kernel<<<grid_for(numel), 256, 0, s>>>(p, g, m, v, state, numel, lr,
-                                              b1, b2, eps);
+                                              b1, b2, eps, gscale);
 }

@@ -52,6 +52,8 @@ def build_parser():
                         help="generate a synthetic dataset in-place if processed/ is missing")
     parser.add_argument("--sync_bn", action="store_true",
                         help="exact-parity BatchNorm under DDP (statistics all-reduced)")
+    parser.add_argument("--loss_scale", type=float, default=1.0,
+                        help="static loss scaling (fp16 mode): loss*S backward, optimizer unscales")
     parser.add_argument("--precision", choices=["fp32", "bf16", "fp16"], default="fp32",
                         help="matmul compute precision on GPU")
     return parser
@@ -155,7 +157,7 @@ def main(argv=None):
     if torch.cuda.is_available():
         from pertgnn.ops.functional import set_gemm_precision
         set_gemm_precision(args.precision)
-    optimizer = FusedAdam(model.parameters(), lr=args.lr)
+    optimizer = FusedAdam(model.parameters(), lr=args.lr, grad_scale=args.loss_scale)
     engine = FlatGradAllReduce(optimizer, comm) if comm.distributed else None
     log = JsonlLogger(args.metrics_jsonl, rank=comm.rank)
 
@@ -169,7 +171,7 @@ def main(argv=None):
         epoch_stats: dict = {}
         train_mae, train_mape = train_epoch(
             model, train_loader, optimizer, args.tau, device, engine=engine,
-            comm=comm, stats_out=epoch_stats,
+            comm=comm, stats_out=epoch_stats, loss_scale=args.loss_scale,
         )
         valid_mae, valid_mape, valid_q = evaluate(model, valid_loader, args.tau, device, comm=comm)
         test_mae, test_mape, test_q = evaluate(model, test_loader, args.tau, device, comm=comm)

@@ -40,9 +40,12 @@ def _forward(model, b):
 
 
 def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
-                non_blocking=True, stats_out=None):
+                non_blocking=True, stats_out=None, loss_scale=1.0):
     """Returns (avg_loss, avg_mape); with stats_out (dict) also records
-    graphs/sec, edges/sec and nodes/sec for the epoch (whole job)."""
+    graphs/sec, edges/sec and nodes/sec for the epoch (whole job).
+    ``loss_scale`` > 1 enables static loss scaling (fp16 mode): the scaled
+    loss drives backward, the optimizer unscales (FusedAdam ``grad_scale``
+    must match), and the REPORTED loss stays unscaled."""
     import time as _time
 
     model.train()
@@ -63,7 +66,7 @@ def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
             pred = global_pred.flatten()
             loss = F.quantile_loss(b.y, pred, tau)
         with _nvtx("backward"):
-            loss.backward()
+            (loss * loss_scale).backward() if loss_scale != 1.0 else loss.backward()
         with _nvtx("allreduce"):
             if engine is not None:
                 engine.finalize()

@@ -25,13 +25,17 @@ def _trainable(params):
 
 
 class FusedAdam:
-    def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8):
+    def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                 grad_scale=1.0):
         self.params = _trainable(list(params))
         assert self.params, "no trainable parameters"
         assert all(p.dtype == torch.float32 for p in self.params), "fp32 master params only"
         self.lr = lr
         self.betas = betas
         self.eps = eps
+        # static loss scaling: the loop multiplies the loss by grad_scale,
+        # the optimizer divides the gradients back before the moment update
+        self.grad_scale = float(grad_scale)
         self.step_count = 0
         device = self.params[0].device
 
@@ -63,15 +67,16 @@ class FusedAdam:
     def step(self):
         self.step_count += 1
         b1, b2 = self.betas
+        inv = 1.0 / self.grad_scale
         if self.flat_param.is_cuda and has_hip():
             ext().adam_step(self.flat_param, self.flat_grad, self.exp_avg,
                             self.exp_avg_sq, self.dev_state, self.lr, b1, b2,
-                            self.eps)
+                            self.eps, inv)
             return
         # eager fallback — identical formula (torch.optim.Adam)
         g = self.flat_grad
-        self.exp_avg.mul_(b1).add_(g, alpha=1 - b1)
-        self.exp_avg_sq.mul_(b2).addcmul_(g, g, value=1 - b2)
+        self.exp_avg.mul_(b1).add_(g, alpha=(1 - b1) * inv)
+        self.exp_avg_sq.mul_(b2).addcmul_(g, g, value=(1 - b2) * inv * inv)
         bias1 = 1 - b1 ** self.step_count
         bias2 = 1 - b2 ** self.step_count
         denom = (self.exp_avg_sq.sqrt() / (bias2 ** 0.5)).add_(self.eps)

@@ -88,3 +88,26 @@ def test_fused_adam_state_roundtrip(tmp_path):
     opt2.load_state_dict(sd)
     assert opt2.step_count == opt.step_count
     assert torch.allclose(opt2.exp_avg, opt.exp_avg)
+
+
+def test_fused_adam_loss_scaling_equivalence():
+    """grad_scale=S with S-scaled gradients reproduces the unscaled step."""
+    from pertgnn.train.optim import FusedAdam
+
+    torch.manual_seed(5)
+    w0 = torch.randn(64, 8)
+    g = torch.randn(64, 8) * 1e-3
+
+    p_a = torch.nn.Parameter(w0.clone())
+    opt_a = FusedAdam([p_a], lr=1e-3)
+    p_a.grad.copy_(g)
+    opt_a.step()
+
+    S = 1024.0
+    p_b = torch.nn.Parameter(w0.clone())
+    opt_b = FusedAdam([p_b], lr=1e-3, grad_scale=S)
+    p_b.grad.copy_(g * S)
+    opt_b.step()
+
+    assert torch.allclose(p_a.data, p_b.data, atol=1e-7), \
+        (p_a.data - p_b.data).abs().max()
