@@ -58,6 +58,8 @@ def sanitize_edges(trace_df: pd.DataFrame, root_ms) -> pd.DataFrame:
 
 def min_node_depth(num_nodes: int, edge_index: np.ndarray, root: int) -> np.ndarray:
     """Iterative version of misc.py:52-63 (min depth, monotone guard)."""
+    if num_nodes == 0:
+        return np.zeros(0)
     depth = np.full(num_nodes, np.inf)
     adj: list[list[int]] = [[] for _ in range(num_nodes)]
     for s, d in edge_index.T:
@@ -73,6 +75,8 @@ def min_node_depth(num_nodes: int, edge_index: np.ndarray, root: int) -> np.ndar
 
 
 def _normalized_depth_int64(depth: np.ndarray) -> torch.Tensor:
+    if depth.size == 0:
+        return torch.zeros(0, 1, dtype=torch.long)
     depth = depth.copy()
     depth[np.isinf(depth)] = 0
     norm = depth.max() if depth.max() > 0 else 1.0
@@ -92,7 +96,7 @@ def build_span_graph(trace_df: pd.DataFrame):
     edge_index = torch.tensor(inv.reshape(2, -1), dtype=torch.long)
     num_nodes = int(edge_index.max().item()) + 1 if edge_index.numel() else 0
     ms2nid = {int(m): i for i, m in enumerate(uniq)}
-    root_nid = ms2nid.get(int(root_ms), 0)
+    root_nid = ms2nid.get(int(root_ms), 0) if root_ms is not None else 0
     depth = min_node_depth(len(uniq), edge_index.numpy(), root_nid)
     edge_attr = torch.tensor(
         df[["interface", "rpctype"]].to_numpy(dtype=np.int64), dtype=torch.long
@@ -189,7 +193,8 @@ def build_pert_graph(trace_df: pd.DataFrame):
     edge_index = torch.tensor(edges, dtype=torch.long).t().contiguous() if edges else torch.zeros(2, 0, dtype=torch.long)
     edge_attr = torch.tensor(attrs, dtype=torch.long).contiguous() if attrs else torch.zeros(0, 4, dtype=torch.long)
     nn_from_edges = int(edge_index.max().item()) + 1 if edge_index.numel() else 0
-    root_nid = int(stages[int(root_ms)][0]) if int(root_ms) in stages else 0
+    root_nid = (int(stages[int(root_ms)][0])
+                if root_ms is not None and int(root_ms) in stages else 0)
     depth = min_node_depth(num_nodes, edge_index.numpy(), root_nid)
     ms_id = torch.tensor(np.array(sorted_ms_id)[:, None], dtype=torch.long)
     return {

@@ -1,0 +1,124 @@
+"""Property-based tests for the span/PERT graph builders (hypothesis).
+
+Invariants derived from the reference semantics (misc.py:87-370) hold for
+arbitrary call frames, not just the synthetic generator's shapes.
+"""
+import numpy as np
+import pandas as pd
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from pertgnn.data.graphs import (build_pert_graph, build_span_graph,
+                                 find_root_ms, sanitize_edges)
+
+
+@st.composite
+def call_frames(draw):
+    n = draw(st.integers(min_value=1, max_value=14))
+    ms = st.integers(min_value=0, max_value=6)
+    rows = []
+    for i in range(n):
+        um = draw(ms)
+        dm = draw(ms)
+        rows.append({
+            "traceid": 0,
+            "timestamp": draw(st.integers(min_value=0, max_value=50)),
+            "rpcid": draw(st.integers(min_value=0, max_value=8)),
+            "um": um,
+            "rpctype": draw(st.integers(min_value=0, max_value=3)),
+            "dm": dm,
+            "interface": draw(st.integers(min_value=0, max_value=5)),
+            "rt": draw(st.integers(min_value=1, max_value=100)),
+        })
+    return pd.DataFrame(rows)
+
+
+def _brute_sanitize(df, root):
+    """Independent reimplementation of the reference order (misc.py:87-105)."""
+    out = []
+    seen_rpcid = set()
+    for _, r in df.iterrows():
+        if r["um"] == r["dm"]:
+            continue
+        if r["rpcid"] in seen_rpcid:
+            continue
+        seen_rpcid.add(r["rpcid"])
+        if r["dm"] == root:
+            continue
+        out.append(r)
+    # dup (um,dm): keep LAST
+    kept = []
+    seen_pair = set()
+    for r in reversed(out):
+        if (r["um"], r["dm"]) in seen_pair:
+            continue
+        seen_pair.add((r["um"], r["dm"]))
+        kept.append(r)
+    kept.reverse()
+    # one edge per unordered pair: keep FIRST
+    final = []
+    seen_unord = set()
+    for r in kept:
+        key = (min(r["um"], r["dm"]), max(r["um"], r["dm"]))
+        if key in seen_unord:
+            continue
+        seen_unord.add(key)
+        final.append(r)
+    return final
+
+
+@settings(max_examples=60, deadline=None)
+@given(call_frames())
+def test_sanitize_matches_bruteforce(df):
+    root = find_root_ms(df)
+    got = sanitize_edges(df, root)
+    want = _brute_sanitize(df, root)
+    assert len(got) == len(want)
+    for (_, g), w in zip(got.iterrows(), want):
+        assert (g["um"], g["dm"], g["rpcid"]) == (w["um"], w["dm"], w["rpcid"])
+
+
+@settings(max_examples=60, deadline=None)
+@given(call_frames())
+def test_pert_graph_invariants(df):
+    root = find_root_ms(df)
+    clean = sanitize_edges(df, root)
+    g = build_pert_graph(df)
+    ei, ea = g["edge_index"], g["edge_attr"]
+    rows = len(clean)
+    callers = clean["um"].value_counts()
+    # stage-node accounting: 2n+1 per caller, 1 per pure callee
+    pure_callees = set(clean["dm"]) - set(clean["um"])
+    expected_nodes = int(sum(2 * c + 1 for c in callers.values) + len(pure_callees))
+    assert g["ms_id"].shape[0] == expected_nodes
+    # edges: intra-chain (2n per caller) + one call + one return per row
+    expected_edges = int(sum(2 * c for c in callers.values)) + 2 * rows
+    assert ei.shape[1] == expected_edges
+    assert ea.shape == (expected_edges, 4)
+    if rows:
+        # intra edges carry [0,0,1,1]; call edges [.,.,1,0]; return [.,.,0,0]
+        intra = (ea[:, 2] == 1) & (ea[:, 3] == 1)
+        call = (ea[:, 2] == 1) & (ea[:, 3] == 0)
+        ret = (ea[:, 2] == 0) & (ea[:, 3] == 0)
+        assert int(call.sum()) == rows
+        assert int(ret.sum()) == rows
+        assert int(intra.sum()) == expected_edges - 2 * rows
+        assert (ea[intra][:, :2] == 0).all()
+    # no out-of-range endpoints
+    if ei.numel():
+        assert int(ei.max()) < expected_nodes
+        assert int(ei.min()) >= 0
+
+
+@settings(max_examples=60, deadline=None)
+@given(call_frames())
+def test_span_graph_invariants(df):
+    root = find_root_ms(df)
+    clean = sanitize_edges(df, root)
+    g = build_span_graph(df)
+    assert g["edge_index"].shape[1] == len(clean)
+    assert g["edge_attr"].shape == (len(clean), 2)
+    n_ms = len(set(clean["um"]) | set(clean["dm"]))
+    assert g["ms_id"].shape[0] == n_ms
+    if len(clean):
+        assert int(g["edge_index"].max()) < n_ms
