@@ -122,3 +122,51 @@ def test_span_graph_invariants(df):
     assert g["ms_id"].shape[0] == n_ms
     if len(clean):
         assert int(g["edge_index"].max()) < n_ms
+
+
+@st.composite
+def multi_trace_frames(draw):
+    n_traces = draw(st.integers(min_value=1, max_value=4))
+    rows = []
+    for t in range(n_traces):
+        n = draw(st.integers(min_value=1, max_value=6))
+        for i in range(n):
+            rows.append({
+                "traceid": t,
+                "timestamp": draw(st.integers(min_value=0, max_value=9)),
+                "rpcid": f"{t}.{i}",
+                "um": draw(st.sampled_from(["(?)", "A", "B"])),
+                "rpctype": draw(st.sampled_from(["http", "rpc", "db"])),
+                "dm": draw(st.sampled_from(["A", "B", "C"])),
+                "interface": draw(st.sampled_from(["i0", "i1"])),
+                "rt": draw(st.integers(min_value=-20, max_value=20)),
+            })
+    return pd.DataFrame(rows)
+
+
+@settings(max_examples=60, deadline=None)
+@given(multi_trace_frames())
+def test_detect_entries_matches_bruteforce(df):
+    """Vectorized entry detection == the reference's per-trace semantics
+    (preprocess.py:99-149): entry row is http & ts==min & |rt|==max; several
+    candidates tie-break on um=='(?)'; none/ambiguous -> drop the trace."""
+    from pertgnn.data.ingest import detect_entries
+
+    got = detect_entries(df.copy())
+
+    want = {}
+    for tid, tdf in df.groupby("traceid"):
+        ts_min = tdf["timestamp"].min()
+        rt_max = tdf["rt"].abs().max()
+        cand = tdf[(tdf["rpctype"] == "http")
+                   & (tdf["timestamp"] == ts_min)
+                   & (tdf["rt"].abs() == rt_max)]
+        if len(cand) > 1:
+            cand = cand[cand["um"] == "(?)"]
+        if len(cand) == 1:
+            row = cand.iloc[0]
+            want[tid] = f"{row['dm']}_{row['interface']}"
+
+    assert set(got["traceid"].unique()) == set(want)
+    for tid, entry in want.items():
+        assert (got[got["traceid"] == tid]["entryid"] == entry).all()
