@@ -157,9 +157,12 @@ __global__ void edge_attn_fused_fwd_kernel(
     const long src = csr_src[p];
     const long a0 = ea[(long)p * astride];
     const long a1 = ea[(long)p * astride + 1];
-    float ec[VPT], ke[VPT];
+    float ec[VPT], ke[VPT], ve[VPT];
+    // all three gathers issue together: the ve load must not sit behind the
+    // sub-wave reduce + expf chain (one exposed memory latency per edge)
     S::load_add(&pifc[a0 * h], &prpc[a1 * h], lane, h, ec);
     S::load(&qkvs[src * ld + h], lane, h, ke);
+    S::load(&qkvs[src * ld + 2 * h], lane, h, ve);
     float part = 0.f;
 #pragma unroll
     for (int j = 0; j < VPT; ++j) part += qr[j] * (ke[j] + ec[j]);
@@ -169,8 +172,6 @@ __global__ void edge_attn_fused_fwd_kernel(
     const float corr = __expf(m - m_new);
     const float pexp = __expf(logit - m_new);
     s = s * corr + pexp;
-    float ve[VPT];
-    S::load(&qkvs[src * ld + 2 * h], lane, h, ve);
 #pragma unroll
     for (int j = 0; j < VPT; ++j)
       acc[j] = acc[j] * corr + pexp * (ve[j] + ec[j]);
