@@ -1,5 +1,6 @@
 """End-to-end pipeline: synthetic traces -> ingest -> dataset -> collate -> model."""
 import os
+from pathlib import Path
 
 import joblib
 import pytest
@@ -204,3 +205,30 @@ def test_checkpoint_resume_cli(synthetic_workspace, tmp_path):
             continue
         assert torch.equal(p1.detach(), p2.detach()), n1
     assert torch.equal(opt.exp_avg, opt2.exp_avg)
+
+
+def test_cli_end_to_end_subprocess(tmp_path):
+    """The real `python pert_gnn.py` entrypoint: synthetic data, 2 epochs on
+    CPU with precision/loss_scale flags — catches argument-wiring and
+    module-level regressions no unit test sees."""
+    import json
+    import re
+    import subprocess
+    import sys
+
+    metrics = tmp_path / "m.jsonl"
+    cmd = [sys.executable, "pert_gnn.py", "--synthetic", "--graph_type", "pert",
+           "--epochs", "2", "--num_layers", "1", "--hidden_channels", "16",
+           "--batch_size", "32", "--seed", "3", "--loss_scale", "8",
+           "--processed_dir", str(tmp_path / "processed"),
+           "--metrics_jsonl", str(metrics)]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                       cwd=str(Path(__file__).resolve().parents[1]))
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("Epoch:")]
+    assert len(lines) == 2, r.stdout[-2000:]
+    # reference epoch line format
+    assert re.match(r"Epoch: 2, Train: [\d.]+, Test mae: [\d.]+, ", lines[1])
+    recs = [json.loads(l) for l in metrics.read_text().splitlines()]
+    assert len(recs) == 2 and recs[1]["epoch"] == 2
+    assert all("test_mae" in r or "test" in str(r) for r in recs)
