@@ -1064,7 +1064,6 @@ __global__ void bn_bwd_apply_kernel(
     const long numq = n * (h / 4);
     const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
     const long stride = (long)gridDim.x * blockDim.x;
-    const int hq = h / 4;
     for (long q = i0; q < numq; q += stride) {
       const long t = q * 4;
       const int c = (int)(t % h);
@@ -1088,7 +1087,6 @@ __global__ void bn_bwd_apply_kernel(
       }
       *reinterpret_cast<TX4*>(&dx[t]) = o;
     }
-    (void)hq;
     return;
   }
   const long numel = n * h;

@@ -15,8 +15,6 @@ Reference quirks intentionally reproduced:
 """
 from __future__ import annotations
 
-import math
-
 import torch
 import torch.nn.functional as F
 from torch import nn
