@@ -8,7 +8,6 @@ directly and no per-kernel permutation is needed.
 """
 from __future__ import annotations
 
-import math
 
 import torch
 
@@ -17,13 +16,14 @@ from .backend import ext, use_hip
 
 
 def deterministic() -> bool:
-    """PERTGNN_DETERMINISTIC=1 makes the weight-gradient reductions
-    run-to-run bitwise reproducible: vocab/embedding table grads take the
-    two-phase grouped scatter (fixed reduction order, no atomics) and the
-    split-K wgrad GEMMs collapse to one K-slice (the C++ launchers read the
-    same env).  The activation-gradient path is deterministic either way.
-    Costs ~10-15% step time; incompatible with hipGraph capture (the grouping
-    uses bincount, which syncs) — use --no-hipgraph in bench.py."""
+    """PERTGNN_DETERMINISTIC=1 makes training bitwise run-to-run
+    reproducible: BN statistics reduce through per-block slabs in fixed
+    order, vocab/embedding table grads take the two-phase grouped scatter
+    (stable sort, no atomics), and the split-K wgrad GEMMs collapse to one
+    K-slice (the C++ launchers read the same env).  The activation-gradient
+    path is deterministic either way.  Verification/debugging mode: combine
+    with --no-hipgraph (the first grouping per batch uses bincount, which
+    syncs); eager stepping dominates the cost at small batches."""
     import os
     return os.environ.get("PERTGNN_DETERMINISTIC", "0") == "1"
 
