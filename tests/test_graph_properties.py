@@ -170,3 +170,35 @@ def test_detect_entries_matches_bruteforce(df):
     assert set(got["traceid"].unique()) == set(want)
     for tid, entry in want.items():
         assert (got[got["traceid"] == tid]["entryid"] == entry).all()
+
+
+@settings(max_examples=50, deadline=None)
+@given(call_frames())
+def test_min_node_depth_matches_bfs(df):
+    """Iterative min-depth (monotone-guard DFS, misc.py:52-63 semantics)
+    equals plain BFS levels on the same graph."""
+    from collections import deque
+
+    from pertgnn.data.graphs import build_pert_graph, min_node_depth
+
+    g = build_pert_graph(df)
+    ei = g["edge_index"].numpy()
+    n = int(g["ms_id"].shape[0])
+    if n == 0:
+        return
+    got = min_node_depth(n, ei, 0)
+
+    adj = [[] for _ in range(n)]
+    for s_, d_ in ei.T:
+        adj[int(s_)].append(int(d_))
+    dist = [float("inf")] * n
+    dist[0] = 0
+    dq = deque([0])
+    while dq:
+        v = dq.popleft()
+        for nb in adj[v]:
+            if dist[nb] > dist[v] + 1:
+                dist[nb] = dist[v] + 1
+                dq.append(nb)
+    for v in range(n):
+        assert got[v] == dist[v], (v, got[v], dist[v])

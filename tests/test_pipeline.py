@@ -232,3 +232,37 @@ def test_cli_end_to_end_subprocess(tmp_path):
     recs = [json.loads(l) for l in metrics.read_text().splitlines()]
     assert len(recs) == 2 and recs[1]["epoch"] == 2
     assert all("test_mae" in r or "test" in str(r) for r in recs)
+
+
+def test_threaded_loader_equivalence(synthetic_workspace):
+    """ThreadedLoader yields the same batches as iterating the base loader."""
+    from pertgnn.data.prefetch import ThreadedLoader
+
+    root, (tr2data, entry2runtimes, _, runtime2pert, resource_df) = synthetic_workspace
+    data_list = build_data_list(tr2data, entry2runtimes, runtime2pert, resource_df, limit=24)
+    base = BatchLoader(data_list, batch_size=8, shuffle=False)
+    ref_batches = list(base)
+    thr_batches = list(ThreadedLoader(BatchLoader(data_list, batch_size=8, shuffle=False)))
+    assert len(ref_batches) == len(thr_batches)
+    for a, b in zip(ref_batches, thr_batches):
+        assert torch.equal(a.x, b.x)
+        assert torch.equal(a.edge_index, b.edge_index)
+        assert torch.equal(a.y, b.y)
+
+
+def test_ingest_cached_csv_reload(synthetic_workspace, tmp_path):
+    """get_df's second call reads the cached processed CSVs and reproduces
+    the same frame (preprocess.py:191-266 cache contract)."""
+    import pandas as pd
+
+    from pertgnn.data.ingest import get_df
+
+    root, _ = synthetic_workspace
+    processed = os.path.join(root, "processed")
+    df1, rs1 = get_df(os.path.join(root, "data"), processed)
+    assert os.path.isfile(os.path.join(processed, "processed_df.csv"))
+    df2, rs2 = get_df(os.path.join(root, "data"), processed)
+    pd.testing.assert_frame_equal(
+        df1.reset_index(drop=True), df2.reset_index(drop=True), check_dtype=False)
+    pd.testing.assert_frame_equal(
+        rs1.reset_index(drop=True), rs2.reset_index(drop=True), check_dtype=False)
