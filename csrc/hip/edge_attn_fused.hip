@@ -10,11 +10,15 @@
 //     layer) — the kernel gathers from these L2-resident tables instead of
 //     streaming an [E,H] operand from HBM (halves forward edge traffic).
 //
-// Column mapping: when H is a multiple of 256 each lane owns VPT CONTIGUOUS
-// columns (lane*VPT..) so every per-edge row access is f32x4 loads (guide
-// G13); otherwise the strided lane+j*64 mapping with per-column guards.
-// The mapping is an internal detail — reductions are wave-wide and every
-// tensor is read/written with the same mapping.
+// Row mapping: LPR lanes per CSR destination row (default 32 — PERT rows
+// average degree ~1.3, so packing 2 rows per wave doubles row throughput of
+// the latency-bound per-edge loop); reductions are sub-wave shfl_xor trees.
+//
+// Column mapping: in VEC mode each sub-wave lane owns VPT CONTIGUOUS columns
+// (lane*VPT..) so every per-edge row access is vectorized 16B loads (guide
+// G13); otherwise the strided lane+j*64 mapping with per-column guards
+// (LPR=64 only).  The mapping is an internal detail — every tensor is
+// read/written with the same mapping.
 //
 // Backward: row kernel writes dq + dskip segments of dqkvs and per-edge
 // dek/dev scratch; col kernel (CSC) segment-sums dk/dv into dqkvs; de =
