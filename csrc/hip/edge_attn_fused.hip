@@ -127,8 +127,9 @@ struct Slice {
 
 // LPR = lanes per row: CSR rows average degree ~1.3 on PERT graphs, so a
 // full 64-lane wave per row is latency-bound with most lanes idle between
-// the few row loads.  LPR=16 packs 4 rows into a wave (each sub-group owns
-// h/LPR contiguous columns); reductions become sub-wave shfl_xor trees.
+// the few row loads.  LPR<64 packs 64/LPR rows into a wave (each sub-group
+// owns h/LPR contiguous columns); reductions become sub-wave shfl_xor trees.
+// LPR=32 measured best at H=256 (see attn_lpr below).
 template <int VPT, bool VEC, typename QT = float, int LPR = PERTGNN_WAVE,
           typename TO = float>
 __global__ void edge_attn_fused_fwd_kernel(
