@@ -4,17 +4,30 @@
 
 #include <hip/hip_runtime.h>
 
+#include <stdexcept>
+#include <string>
+
 #define PERTGNN_WAVE 64
 
 #define HIP_CHECK(expr)                                                        \
   do {                                                                         \
     hipError_t _e = (expr);                                                    \
     if (_e != hipSuccess) {                                                    \
-      printf("HIP error %s at %s:%d\n", hipGetErrorString(_e), __FILE__,       \
-             __LINE__);                                                        \
-      abort();                                                                 \
+      throw std::runtime_error(std::string("HIP error ") +                     \
+                               hipGetErrorString(_e) + " at " + __FILE__ +     \
+                               ":" + std::to_string(__LINE__));                \
     }                                                                          \
   } while (0)
+
+// Launcher shape guard: throws (surfaces as a Python RuntimeError through
+// the binding layer) instead of abort()ing the process on a shape the
+// kernel template set does not cover.
+[[noreturn]] inline void pertgnn_shape_fail(const char* launcher,
+                                            const char* what, long v) {
+  throw std::runtime_error(std::string("pertgnn ") + launcher +
+                           ": unsupported shape (" + what + "=" +
+                           std::to_string(v) + ")");
+}
 
 // full-wave (64-lane) sum/max reductions via xor shuffles
 __device__ __forceinline__ float wave_reduce_sum(float v) {

@@ -220,7 +220,7 @@ void launch_edge_attn_fwd(const float* q, const float* k, const float* v,
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
-    default: abort();
+    default: pertgnn_shape_fail("edge_attn launcher", "vpt", vpt);
   }
 }
 
@@ -246,7 +246,7 @@ void launch_edge_attn_bwd(const float* g, const float* q, const float* k,
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
-    default: abort();
+    default: pertgnn_shape_fail("edge_attn launcher", "vpt", vpt);
   }
   const long numel = num_edges * h;
   if (numel > 0) {

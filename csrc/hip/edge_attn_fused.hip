@@ -323,7 +323,7 @@ void launch_edge_attn_fused_fwd(const float* qkvs, const float* pifc,
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
-    default: abort();
+    default: pertgnn_shape_fail("edge_attn_fused launcher", "vpt", vpt);
   }
 }
 
@@ -360,7 +360,7 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
-    default: abort();
+    default: pertgnn_shape_fail("edge_attn_fused launcher", "vpt", vpt);
   }
 }
 
@@ -422,7 +422,7 @@ void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
     else if (vpt == 16) FWD16(16, 64);
     else FWD16(32, 64);
   } else {
-    abort();
+    pertgnn_shape_fail("edge_attn_fused act16 launcher", "h", h);
   }
 #undef FWD16
 }
@@ -478,7 +478,7 @@ void launch_edge_attn_fused_bwd16(const void* g_v, int g16,
     else if (vpt == 16) BWD16(16, 64);
     else BWD16(32, 64);
   } else {
-    abort();
+    pertgnn_shape_fail("edge_attn_fused act16 launcher", "h", h);
   }
 #undef BWD16
 }

@@ -108,7 +108,7 @@ void launch_seg_pool_fwd(const float* x, const float* probs, const float* nn,
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
-    default: abort();
+    default: pertgnn_shape_fail("segops launcher", "vpt", vpt);
   }
 }
 
@@ -142,7 +142,7 @@ void launch_seg_pool_fwd16(const void* x, const float* probs, const float* nn,
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
-    default: abort();
+    default: pertgnn_shape_fail("segops launcher", "vpt", vpt);
   }
 }
 
@@ -276,7 +276,7 @@ void launch_embed_grouped_scatter(const float* g, const int* order,
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
-    default: abort();
+    default: pertgnn_shape_fail("segops launcher", "vpt", vpt);
   }
 }
 
@@ -658,7 +658,7 @@ void launch_vocab_scatter16(const void* g_v, const long* idx, long idx_stride,
                        col_off);
     return;
   }
-  abort();  // caller guarantees fit (python upcasts otherwise)
+  pertgnn_shape_fail("vocab_scatter", "table_bytes_over_lds_h", h);  // caller guarantees fit (python upcasts otherwise)
 }
 
 void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,

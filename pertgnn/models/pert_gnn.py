@@ -149,12 +149,14 @@ class TransformerConv(nn.Module):
                     self.w4[i * h:(i + 1) * h, :self.in_channels].copy_(state_dict[wk])
                     self.w4[i * h:(i + 1) * h, self.in_channels:].zero_()
                     state_dict = {k: v for k, v in state_dict.items() if k != wk}
-                elif strict:
+                else:
+                    # nn.Module semantics: missing keys are COLLECTED
+                    # regardless of strict (strict only controls raising)
                     missing_keys.append(wk)
                 if bk in state_dict:
                     self.b4[i * h:(i + 1) * h].copy_(state_dict[bk])
                     state_dict = {k: v for k, v in state_dict.items() if k != bk}
-                elif strict:
+                else:
                     missing_keys.append(bk)
         we_key = prefix + "lin_edge.weight"
         if we_key in state_dict:
@@ -162,7 +164,7 @@ class TransformerConv(nn.Module):
                 w = state_dict[we_key]
                 self.we_ifc.copy_(w[:, :h])
                 self.we_rpc.copy_(w[:, h:])
-        elif strict:
+        else:
             missing_keys.append(we_key)
         filtered = {k: v for k, v in state_dict.items()
                     if k != we_key

@@ -39,35 +39,47 @@ def _table_grad(m, g, idx, rows, h, col_off):
         g = g.float()
     if rows * h * 4 <= 160 * 1024 and not deterministic():
         return m.vocab_scatter(g, idx, rows, h, col_off)
-    order, ptr = _group_by(idx.contiguous(), rows)
+    order, ptr = _group_by(idx, rows)
     if g.dtype != torch.float32:
         g = g.float()
     return m.embed_grouped_scatter(g, order, ptr, rows, h, col_off)
 
 
-_GROUP_CACHE: dict = {}
+_GROUP_CACHE: "dict" = __import__("collections").OrderedDict()
+_GROUP_CACHE_BYTES = [0]
+_GROUP_CACHE_CAP = 64 * 1024 * 1024  # bytes of cached order/ptr tensors
 
 
 def _group_by(idx: torch.Tensor, rows: int):
     """Group positions by index value: returns (order int32, ptr int32[rows+1])
-    for the deterministic grouped scatter kernels.  Cached per
-    (data_ptr, numel, rows): batch index tensors are static across steps, so
-    the sort + bincount (which syncs) runs once per resident batch."""
-    key = (idx.data_ptr(), idx.numel(), rows)
+    for the deterministic grouped scatter kernels.  Cached so the sort +
+    bincount (which syncs) runs once per resident batch.
+
+    The key is taken from the VIEW the caller holds (data_ptr + stride +
+    numel), before any ``.contiguous()`` copy — callers like the fused
+    attention backward pass ``edge_attr[:, 0]`` column views, which are
+    stable across steps while a fresh ``.contiguous()`` tensor never is.
+    LRU-evicted at a byte budget (the cached view reference pins its base
+    storage so the allocator cannot recycle the keyed data_ptr while the
+    entry is live)."""
+    key = (idx.data_ptr(), idx.numel(), tuple(idx.stride()), rows)
     hit = _GROUP_CACHE.get(key)
     if hit is not None:
+        _GROUP_CACHE.move_to_end(key)
         return hit[1], hit[2]
-    order = torch.argsort(idx, stable=True)  # ties in input order: the
+    idx_c = idx.contiguous()
+    order = torch.argsort(idx_c, stable=True)  # ties in input order: the
     # grouped kernels' reduction order is then fully determined
-    counts = torch.bincount(idx, minlength=rows)
+    counts = torch.bincount(idx_c, minlength=rows)
     ptr = torch.zeros(rows + 1, dtype=torch.int32, device=idx.device)
     ptr[1:] = counts.cumsum(0).to(torch.int32)
     order32 = order.to(torch.int32)
-    if len(_GROUP_CACHE) > 256:  # bounded: resident batches reuse, streams churn
-        _GROUP_CACHE.clear()
-    # the cached idx reference pins its storage so the allocator cannot hand
-    # the same data_ptr to a different tensor while the entry is live
-    _GROUP_CACHE[key] = (idx, order32, ptr)
+    nbytes = order32.numel() * 4 + ptr.numel() * 4
+    while _GROUP_CACHE and _GROUP_CACHE_BYTES[0] + nbytes > _GROUP_CACHE_CAP:
+        _, old = _GROUP_CACHE.popitem(last=False)
+        _GROUP_CACHE_BYTES[0] -= old[3]
+    _GROUP_CACHE[key] = (idx, order32, ptr, nbytes)
+    _GROUP_CACHE_BYTES[0] += nbytes
     return order32, ptr
 
 
@@ -94,13 +106,40 @@ class _EdgeAttentionFn(torch.autograd.Function):
         return dq, dk, dv, de, g, None, None, None, None, None
 
 
+def _device_csr(edge_index, num_nodes):
+    """CSR/CSC arrays for an arbitrary-order edge list, on its own device.
+    Returns (perm, csr_tuple): ``perm`` maps original edge order -> CSR
+    order.  The collator normally provides these (pre-sorted, no perm);
+    this path serves direct module-API callers (reference call signature,
+    model.py:76-86) who pass only ``edge_index``."""
+    dev = edge_index.device
+    src, dst = edge_index[0], edge_index[1]
+    perm = torch.argsort(dst, stable=True)
+    src_s = src.index_select(0, perm)
+    dst_s = dst.index_select(0, perm)
+    row_ptr = torch.zeros(num_nodes + 1, dtype=torch.int32, device=dev)
+    row_ptr[1:] = torch.bincount(dst_s, minlength=num_nodes).cumsum(0).to(torch.int32)
+    perm2 = torch.argsort(src_s, stable=True)
+    col_ptr = torch.zeros(num_nodes + 1, dtype=torch.int32, device=dev)
+    col_ptr[1:] = torch.bincount(src_s, minlength=num_nodes).cumsum(0).to(torch.int32)
+    csr = (row_ptr, src_s.to(torch.int32), col_ptr,
+           dst_s.index_select(0, perm2).to(torch.int32), perm2.to(torch.int32))
+    return perm, csr
+
+
 def edge_attention(q, k, v, e, skip, edge_index, num_nodes, csr=None):
     """out_i = skip_i + sum_e softmax_i(<q_i, k_src+e>/sqrt(H)) (v_src+e).
 
-    ``csr``: (row_ptr, csr_src, col_ptr, csc_dst, csc_eid) from the collator;
-    required on the HIP path.
+    ``csr``: (row_ptr, csr_src, col_ptr, csc_dst, csc_eid) from the collator
+    (edges already CSR-ordered).  When absent on the HIP path — the reference
+    module-API call signature — CSR is built on the fly from ``edge_index``
+    and ``e`` is permuted to match (autograd routes ``de`` back through the
+    index_select).
     """
     if use_hip(q):
+        if csr is None:
+            perm, csr = _device_csr(edge_index, num_nodes)
+            e = e.index_select(0, perm)
         row_ptr, csr_src, col_ptr, csc_dst, csc_eid = csr
         return _EdgeAttentionFn.apply(q, k, v, e, skip, row_ptr, csr_src, col_ptr, csc_dst, csc_eid)
     return ref.edge_attention(q, k, v, e, edge_index, num_nodes, skip)

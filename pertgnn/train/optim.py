@@ -82,10 +82,19 @@ class FusedAdam:
         denom = (self.exp_avg_sq.sqrt() / (bias2 ** 0.5)).add_(self.eps)
         self.flat_param.addcdiv_(self.exp_avg, denom, value=-self.lr / bias1)
 
+    def _synced_step_count(self) -> int:
+        """True step count even after hipGraph replays of the captured step
+        (replay advances only the device-side counter, not Python's)."""
+        if self.flat_param.is_cuda:
+            dev = int(self.dev_state[0].item())
+            if dev > self.step_count:
+                self.step_count = dev
+        return self.step_count
+
     # -- torch-optimizer-compatible surface ---------------------------------
     def state_dict(self):
         return {
-            "step": self.step_count,
+            "step": self._synced_step_count(),
             "dev_state": self.dev_state,
             "exp_avg": self.exp_avg,
             "exp_avg_sq": self.exp_avg_sq,
