@@ -46,34 +46,34 @@ void launch_bn_bwd(const float*, const float*, const float*, const float*,
                    const float*, const float*, float*, float*, float*, float*,
                    long, int, bool, hipStream_t);
 void launch_bn_stats_only(const float*, long, int, float*, hipStream_t);
-void launch_bn_finalize_apply(const float*, const float*, long, const float*,
-                              const float*, float*, float*, float*, float*,
-                              float*, long, int, float, float, bool, bool,
-                              hipStream_t);
+void launch_bn_finalize_apply(const float*, const float*, const float*,
+                              const float*, const float*, float*, float*,
+                              float*, float*, float*, long, int, float, float,
+                              bool, bool, hipStream_t);
 void launch_bn_bwd_partials_only(const float*, const float*, const float*,
                                  const float*, const float*, long, int, bool,
                                  float*, hipStream_t);
 void launch_bn_bwd_apply_only(const float*, const float*, const float*,
                               const float*, const float*, const float*,
-                              const float*, long, float*, long, int, bool,
-                              hipStream_t);
+                              const float*, const float*, float*, long, int,
+                              bool, hipStream_t);
 void launch_bn_grad_affine(const float*, float*, float*, int, hipStream_t);
 // act16 variants (bf16 x/y/g/dx streams, fp32 statistics)
 void launch_bn_stats_only16(const void*, long, int, float*, hipStream_t);
 void launch_bn_fwd16(const void*, const float*, const float*, float*, float*,
                      float*, float*, float*, void*, long, int, float, float,
                      bool, bool, hipStream_t);
-void launch_bn_finalize_apply16(const void*, const float*, long, const float*,
-                                const float*, float*, float*, float*, float*,
-                                void*, long, int, float, float, bool, bool,
-                                hipStream_t);
+void launch_bn_finalize_apply16(const void*, const float*, const float*,
+                                const float*, const float*, float*, float*,
+                                float*, float*, void*, long, int, float, float,
+                                bool, bool, hipStream_t);
 void launch_bn_bwd_partials_only16(const void*, const void*, const void*,
                                    const float*, const float*, long, int, bool,
                                    float*, hipStream_t);
 void launch_bn_bwd_apply_only16(const void*, const void*, const void*,
                                 const float*, const float*, const float*,
-                                const float*, long, void*, long, int, bool,
-                                hipStream_t);
+                                const float*, const float*, void*, long, int,
+                                bool, hipStream_t);
 void launch_bn_bwd16(const void*, const void*, const void*, const float*,
                      const float*, const float*, float*, void*, float*,
                      float*, long, int, bool, hipStream_t);
@@ -652,31 +652,40 @@ torch::Tensor vocab_scatter(torch::Tensor g, torch::Tensor idx, int64_t rows,
   return dtable;
 }
 
+// sync-BN partials: [2h+1] — per-channel sum / sum-of-squares plus the
+// LOCAL row count in the tail slot, so a single all-reduce carries sums AND
+// the global count with no host round-trip (SURVEY.md §7 hard part 4).
 torch::Tensor bn_stats(torch::Tensor x) {
   CHECK_IN(x);
-  auto partials = torch::empty({2 * x.size(1)},
+  const long h = x.size(1);
+  auto partials = torch::empty({2 * h + 1},
                                x.options().dtype(torch::kFloat32));
   if (x.scalar_type() == torch::kBFloat16)
-    launch_bn_stats_only16(x.data_ptr(), x.size(0), x.size(1),
+    launch_bn_stats_only16(x.data_ptr(), x.size(0), h,
                            partials.data_ptr<float>(), cur_stream());
   else
-    launch_bn_stats_only(x.data_ptr<float>(), x.size(0), x.size(1),
+    launch_bn_stats_only(x.data_ptr<float>(), x.size(0), h,
                          partials.data_ptr<float>(), cur_stream());
+  partials.narrow(0, 2 * h, 1).fill_((float)x.size(0));
   return partials;
 }
 
 std::vector<torch::Tensor> bn_finalize_apply(
-    torch::Tensor x, torch::Tensor partials, int64_t count,
+    torch::Tensor x, torch::Tensor partials,
     torch::Tensor gamma, torch::Tensor beta, torch::Tensor running_mean,
     torch::Tensor running_var, double momentum, double eps, bool training,
     bool relu) {
   CHECK_IN(x); CHECK_IN(partials);
   const int h = x.size(1);
+  TORCH_CHECK(partials.numel() == 2 * h + 1,
+              "bn_finalize_apply expects [2h+1] partials with the global "
+              "count in the tail slot");
   auto y = torch::empty_like(x);
   auto mean = torch::empty({h}, x.options());
   auto invstd = torch::empty({h}, x.options());
   launch_bn_finalize_apply(
-      x.data_ptr<float>(), partials.data_ptr<float>(), count,
+      x.data_ptr<float>(), partials.data_ptr<float>(),
+      partials.data_ptr<float>() + 2 * h,
       gamma.data_ptr<float>(), beta.data_ptr<float>(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       mean.data_ptr<float>(), invstd.data_ptr<float>(), y.data_ptr<float>(),
@@ -689,11 +698,13 @@ torch::Tensor bn_bwd_partials(torch::Tensor g, torch::Tensor x,
                               torch::Tensor y, torch::Tensor mean,
                               torch::Tensor invstd, bool relu) {
   CHECK_IN(g); CHECK_IN(x);
-  auto partials = torch::empty({2 * x.size(1)}, x.options());
+  const long h = x.size(1);
+  auto partials = torch::empty({2 * h + 1}, x.options());
   launch_bn_bwd_partials_only(g.data_ptr<float>(), x.data_ptr<float>(),
                               y.data_ptr<float>(), mean.data_ptr<float>(),
-                              invstd.data_ptr<float>(), x.size(0), x.size(1),
+                              invstd.data_ptr<float>(), x.size(0), h,
                               relu, partials.data_ptr<float>(), cur_stream());
+  partials.narrow(0, 2 * h, 1).fill_((float)x.size(0));
   return partials;
 }
 
@@ -703,16 +714,20 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor g, torch::Tensor x,
                                         torch::Tensor gamma,
                                         torch::Tensor partials_global,
                                         torch::Tensor partials_local,
-                                        int64_t count, bool relu) {
+                                        bool relu) {
   CHECK_IN(g); CHECK_IN(x);
   const int h = x.size(1);
+  TORCH_CHECK(partials_global.numel() == 2 * h + 1,
+              "bn_bwd_apply expects [2h+1] partials with the global count "
+              "in the tail slot");
   auto dx = torch::empty_like(x);
   auto dgamma = torch::empty({h}, x.options());
   auto dbeta = torch::empty({h}, x.options());
   launch_bn_bwd_apply_only(g.data_ptr<float>(), x.data_ptr<float>(),
                            y.data_ptr<float>(), mean.data_ptr<float>(),
                            invstd.data_ptr<float>(), gamma.data_ptr<float>(),
-                           partials_global.data_ptr<float>(), count,
+                           partials_global.data_ptr<float>(),
+                           partials_global.data_ptr<float>() + 2 * h,
                            dx.data_ptr<float>(), x.size(0), h, relu,
                            cur_stream());
   launch_bn_grad_affine(partials_local.data_ptr<float>(),
@@ -888,18 +903,22 @@ std::vector<torch::Tensor> bn_relu_bwd16(torch::Tensor g, torch::Tensor x,
 }
 
 std::vector<torch::Tensor> bn_finalize_apply16(
-    torch::Tensor x, torch::Tensor partials, int64_t count,
+    torch::Tensor x, torch::Tensor partials,
     torch::Tensor gamma, torch::Tensor beta, torch::Tensor running_mean,
     torch::Tensor running_var, double momentum, double eps, bool training,
     bool relu) {
   CHECK_IN(x); CHECK_IN(partials);
   const int h = x.size(1);
+  TORCH_CHECK(partials.numel() == 2 * h + 1,
+              "bn_finalize_apply16 expects [2h+1] partials with the global "
+              "count in the tail slot");
   auto fopt = x.options().dtype(torch::kFloat32);
   auto y = torch::empty({x.size(0), h}, x.options().dtype(torch::kBFloat16));
   auto mean = torch::empty({h}, fopt);
   auto invstd = torch::empty({h}, fopt);
   launch_bn_finalize_apply16(
-      x.data_ptr(), partials.data_ptr<float>(), count,
+      x.data_ptr(), partials.data_ptr<float>(),
+      partials.data_ptr<float>() + 2 * h,
       gamma.data_ptr<float>(), beta.data_ptr<float>(),
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       mean.data_ptr<float>(), invstd.data_ptr<float>(), y.data_ptr(),
@@ -912,22 +931,27 @@ torch::Tensor bn_bwd_partials16(torch::Tensor g, torch::Tensor x,
                                 torch::Tensor y, torch::Tensor mean,
                                 torch::Tensor invstd, bool relu) {
   CHECK_IN(g); CHECK_IN(x);
-  auto partials = torch::empty({2 * x.size(1)},
+  const long h = x.size(1);
+  auto partials = torch::empty({2 * h + 1},
                                x.options().dtype(torch::kFloat32));
   launch_bn_bwd_partials_only16(g.data_ptr(), x.data_ptr(),
                                 y.data_ptr(), mean.data_ptr<float>(),
-                                invstd.data_ptr<float>(), x.size(0), x.size(1),
+                                invstd.data_ptr<float>(), x.size(0), h,
                                 relu, partials.data_ptr<float>(),
                                 cur_stream());
+  partials.narrow(0, 2 * h, 1).fill_((float)x.size(0));
   return partials;
 }
 
 std::vector<torch::Tensor> bn_bwd_apply16(
     torch::Tensor g, torch::Tensor x, torch::Tensor y, torch::Tensor mean,
     torch::Tensor invstd, torch::Tensor gamma, torch::Tensor partials_global,
-    torch::Tensor partials_local, int64_t count, bool relu) {
+    torch::Tensor partials_local, bool relu) {
   CHECK_IN(g); CHECK_IN(x);
   const int h = x.size(1);
+  TORCH_CHECK(partials_global.numel() == 2 * h + 1,
+              "bn_bwd_apply16 expects [2h+1] partials with the global count "
+              "in the tail slot");
   auto fopt = x.options().dtype(torch::kFloat32);
   auto dx = torch::empty_like(x);
   auto dgamma = torch::empty({h}, fopt);
@@ -935,7 +959,8 @@ std::vector<torch::Tensor> bn_bwd_apply16(
   launch_bn_bwd_apply_only16(g.data_ptr(), x.data_ptr(), y.data_ptr(),
                              mean.data_ptr<float>(), invstd.data_ptr<float>(),
                              gamma.data_ptr<float>(),
-                             partials_global.data_ptr<float>(), count,
+                             partials_global.data_ptr<float>(),
+                             partials_global.data_ptr<float>() + 2 * h,
                              dx.data_ptr(), x.size(0), h, relu,
                              cur_stream());
   launch_bn_grad_affine(partials_local.data_ptr<float>(),

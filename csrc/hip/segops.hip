@@ -880,21 +880,27 @@ static void bn_stats_dispatch(const TX* x, long n, int h, float* partials,
 
 
 // stage B: finalize mean/invstd (+ running-stat update, training only)
+// count_ptr (may be null): device-side GLOBAL row count — sync-BN appends
+// the local count to the stats partials so ONE all-reduce carries sums AND
+// count, with no host round-trip per BN layer (the count slot lives at
+// partials[2h], untouched by the 2h-wide reducers).
 __global__ void bn_finalize_kernel(const float* __restrict__ partials, long n,
                                    int h, float eps, float momentum,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
-                                   int update_running) {
+                                   int update_running,
+                                   const float* __restrict__ count_ptr = nullptr) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= h) return;
-  const float m = partials[c] / n;
-  const float var = fmaxf(partials[h + c] / n - m * m, 0.f);
+  const float nf = count_ptr ? *count_ptr : (float)n;
+  const float m = partials[c] / nf;
+  const float var = fmaxf(partials[h + c] / nf - m * m, 0.f);
   mean[c] = m;
   invstd[c] = rsqrtf(var + eps);
   if (update_running) {
-    const float unbiased = (n > 1) ? var * n / (n - 1) : var;
+    const float unbiased = (nf > 1.f) ? var * nf / (nf - 1.f) : var;
     running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
     running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
   }
@@ -1057,9 +1063,10 @@ __global__ void bn_bwd_apply_kernel(
     const TY* __restrict__ y, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
     const float* __restrict__ partials, TX* __restrict__ dx, long n,
-    long count, int h, int relu) {
+    long count, int h, int relu,
+    const float* __restrict__ count_ptr = nullptr) {
   typedef __attribute__((ext_vector_type(4))) float bnf4;
-  const float invn = 1.f / count;
+  const float invn = 1.f / (count_ptr ? *count_ptr : (float)count);
   if ((h & 3) == 0) {
     const long numq = n * (h / 4);
     const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1144,7 +1151,7 @@ void launch_bn_stats_only(const float* x, long n, int h, float* partials,
 }
 
 void launch_bn_finalize_apply(const float* x, const float* partials,
-                              long count, const float* gamma,
+                              const float* count_ptr, const float* gamma,
                               const float* beta, float* running_mean,
                               float* running_var, float* mean, float* invstd,
                               float* y, long n, int h, float momentum,
@@ -1152,8 +1159,8 @@ void launch_bn_finalize_apply(const float* x, const float* partials,
                               hipStream_t s) {
   if (training) {
     bn_finalize_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
-        partials, count, h, eps, momentum, mean, invstd, running_mean,
-        running_var, 1);
+        partials, n, h, eps, momentum, mean, invstd, running_mean,
+        running_var, 1, count_ptr);
   } else {
     bn_eval_stats_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
         running_mean, running_var, mean, invstd, h, eps);
@@ -1178,11 +1185,12 @@ void launch_bn_bwd_partials_only(const float* g, const float* x,
 void launch_bn_bwd_apply_only(const float* g, const float* x, const float* y,
                               const float* mean, const float* invstd,
                               const float* gamma, const float* partials,
-                              long count, float* dx, long n, int h, bool relu,
-                              hipStream_t s) {
+                              const float* count_ptr, float* dx, long n, int h,
+                              bool relu, hipStream_t s) {
   if (n == 0) return;
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
-      g, x, y, mean, invstd, gamma, partials, dx, n, count, h, relu ? 1 : 0);
+      g, x, y, mean, invstd, gamma, partials, dx, n, n, h, relu ? 1 : 0,
+      count_ptr);
 }
 
 void launch_bn_grad_affine(const float* partials, float* dgamma, float* dbeta,
@@ -1255,7 +1263,7 @@ void launch_bn_fwd16(const void* x, const float* gamma, const float* beta,
 }
 
 void launch_bn_finalize_apply16(const void* x, const float* partials,
-                                long count, const float* gamma,
+                                const float* count_ptr, const float* gamma,
                                 const float* beta, float* running_mean,
                                 float* running_var, float* mean, float* invstd,
                                 void* y, long n, int h, float momentum,
@@ -1263,8 +1271,8 @@ void launch_bn_finalize_apply16(const void* x, const float* partials,
                                 hipStream_t s) {
   if (training) {
     bn_finalize_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
-        partials, count, h, eps, momentum, mean, invstd, running_mean,
-        running_var, 1);
+        partials, n, h, eps, momentum, mean, invstd, running_mean,
+        running_var, 1, count_ptr);
   } else {
     bn_eval_stats_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
         running_mean, running_var, mean, invstd, h, eps);
@@ -1292,12 +1300,12 @@ void launch_bn_bwd_partials_only16(const void* g, const void* x,
 void launch_bn_bwd_apply_only16(const void* g, const void* x, const void* y,
                                 const float* mean, const float* invstd,
                                 const float* gamma, const float* partials,
-                                long count, void* dx, long n, int h,
-                                bool relu, hipStream_t s) {
+                                const float* count_ptr, void* dx, long n,
+                                int h, bool relu, hipStream_t s) {
   if (n == 0) return;
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
       (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd,
-      gamma, partials, (__bf16*)dx, n, count, h, relu ? 1 : 0);
+      gamma, partials, (__bf16*)dx, n, n, h, relu ? 1 : 0, count_ptr);
 }
 
 void launch_bn_bwd16(const void* g, const void* x, const void* y,

@@ -391,18 +391,19 @@ class _BNReLUFn(torch.autograd.Function):
         m = ext()
         sync = comm is not None and getattr(comm, "distributed", False) and training
         if sync:
-            # sync-BN (SURVEY.md §7 hard part 4, exact-parity mode): partial
-            # sums + global count all-reduced before normalization.
+            # sync-BN (SURVEY.md §7 hard part 4, exact-parity mode): ONE
+            # all-reduce carries the partial sums AND the row count (tail
+            # slot of the [2h+1] partials) — no host round-trip per layer,
+            # so the whole step stays launch-async (and hipGraph-capturable
+            # where the backend supports captured collectives).
             partials = m.bn_stats(x)
             comm.all_reduce_(partials)
-            count = int(comm.all_reduce_scalar(float(x.shape[0])))
             fin = (m.bn_finalize_apply16 if x.dtype == torch.bfloat16
                    else m.bn_finalize_apply)
             y, save_mean, save_invstd = fin(
-                x, partials, count, gamma, beta, running_mean, running_var,
+                x, partials, gamma, beta, running_mean, running_var,
                 momentum, eps, training, fuse_relu)
         else:
-            count = x.shape[0]
             fwd = (m.bn_relu_fwd16 if x.dtype == torch.bfloat16
                    else m.bn_relu_fwd)
             y, save_mean, save_invstd = fwd(
@@ -411,7 +412,6 @@ class _BNReLUFn(torch.autograd.Function):
         ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
         ctx.fuse_relu = fuse_relu
         ctx.comm = comm if sync else None
-        ctx.count = count
         return y
 
     @staticmethod
@@ -423,12 +423,14 @@ class _BNReLUFn(torch.autograd.Function):
         if ctx.comm is not None:
             bwd_partials = m.bn_bwd_partials16 if y16 else m.bn_bwd_partials
             bwd_apply = m.bn_bwd_apply16 if y16 else m.bn_bwd_apply
+            # [2h+1] partials: sums + local count; one all-reduce gives the
+            # global sums and count together (device-side, no .item()).
             partials_local = bwd_partials(g, x, y, save_mean, save_invstd, ctx.fuse_relu)
             partials_global = partials_local.clone()
             ctx.comm.all_reduce_(partials_global)
             dx, dgamma, dbeta = bwd_apply(
                 g, x, y, save_mean, save_invstd, gamma,
-                partials_global, partials_local, ctx.count, ctx.fuse_relu)
+                partials_global, partials_local, ctx.fuse_relu)
         else:
             bwd = m.bn_relu_bwd16 if y16 else m.bn_relu_bwd
             dx, dgamma, dbeta = bwd(
