@@ -2,38 +2,74 @@
 
 Measures training graphs/sec (whole job) for the 8-layer/256-dim PERT-GNN on
 synthetic Alibaba-shaped call graphs, one process
-per GPU over RCCL.  Weak scaling: per-GPU batch is fixed, value aggregates
-over all N ranks.
+per GPU over RCCL, and appends the latency-MAE anchor (the second half of the
+BASELINE metric) from a short training run on an ingested synthetic dataset.
 
     python bench.py --gpus 1 --steps 20 --warmup 5
     python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
         --master-addr 127.0.0.1 bench.py --gpus 8 --steps 20 --warmup 5
 
+(`python bench.py --gpus 8` without torchrun re-execs itself under
+torch.distributed.run with 8 local ranks.)
+
 Batches are pre-collated once and held resident on HBM (the north-star
 "graph-shard prefetch sized for 288 GB"); every timed step is a FULL training
 step: forward + quantile loss + backward + DDP all-reduce + Adam update.
+
+hipGraph stepping modes (--graph-mode):
+  * full  — the whole step, collectives included, is captured and replayed.
+  * split — fwd+loss+bwd captured; gradient all-reduce + Adam run eagerly
+    after each replay.  The capture then contains no RCCL ops, so it cannot
+    depend on collective-capture support; the eager comm tail costs ~3
+    launches + one all-reduce of the flat grad buffer (~0.2 ms at 8L/256H
+    over xGMI) — the safe default for world_size > 1.
+  * eager — no capture.
+  * auto  — full at world_size 1, split otherwise; every rank's chosen mode
+    is min-reduced so all ranks step the same way even if capture fails on
+    only some of them.
 """
 from __future__ import annotations
 
 import argparse
 import json
+import os
+import sys
 import time
 
 import torch
 
 
+def vocab_cfg(name: str, seed: int):
+    """Synthetic scale presets (SURVEY.md §6): 'small' is the round-1 config;
+    'realistic' approximates Alibaba-2021 vocabulary scale (reference scans
+    thousands of interfaces / hundreds of entries over 100k traces,
+    /root/reference/pert_gnn.py:298-299,325-328) — it exercises the grouped
+    table-gradient scatters (embedding tables far beyond LDS capacity) and
+    gives every batch a diverse pool of entry-union samples."""
+    from pertgnn.data.synthetic import SyntheticConfig
+
+    if name == "realistic":
+        return SyntheticConfig(
+            n_entries=320, patterns_per_entry=3, traces_per_entry=4,
+            min_calls=8, max_calls=28, n_microservices=1024,
+            n_interfaces=2048, seed=seed,
+        )
+    return SyntheticConfig(
+        n_entries=8, patterns_per_entry=3, traces_per_entry=4,
+        min_calls=8, max_calls=28, n_microservices=96, seed=seed,
+    )
+
+
 def build_synthetic_batches(n_batches: int, graphs_per_batch: int, seed: int,
-                            device) -> tuple[list, dict]:
+                            device, vocab: str = "small") -> tuple[list, dict]:
     """Trace-scale synthetic PERT graphs -> pre-collated GPU-resident batches."""
     from pertgnn.data.collate import collate_native as collate
     from pertgnn.data.graphs import build_pert_graph
     from pertgnn.data.dataset import TraceSample
-    from pertgnn.data.synthetic import SyntheticConfig, generate_traces
 
-    cfg = SyntheticConfig(
-        n_entries=8, patterns_per_entry=3, traces_per_entry=4,
-        min_calls=8, max_calls=28, n_microservices=96, seed=seed,
-    )
+    from pertgnn.data.synthetic import generate_traces
+
+    cfg = vocab_cfg(vocab, seed)
     call_df, resource_df = generate_traces(cfg)
     # factorize to int ids the way ingest does (vectorized, minimal)
     import pandas as pd
@@ -53,11 +89,11 @@ def build_synthetic_batches(n_batches: int, graphs_per_batch: int, seed: int,
     entries = {}
     for (tid), tdf in call_df.groupby("traceid"):
         g = build_pert_graph(tdf)
-        key = int(tdf["dm"].iloc[0]) % cfg.n_entries
+        key = int(tdf["dm"].iloc[0])  # entry microservice identity
         entries.setdefault(key, []).append(g)
 
     samples = []
-    for eid, graphs in entries.items():
+    for eid, (key, graphs) in enumerate(sorted(entries.items())):
         graphs = graphs[:3]
         nn = [g["num_nodes"] for g in graphs]
         offs = [0]
@@ -85,12 +121,67 @@ def build_synthetic_batches(n_batches: int, graphs_per_batch: int, seed: int,
     stats = {
         "avg_nodes": sum(b.x.shape[0] for b in batches) / len(batches),
         "avg_edges": sum(b.edge_index.shape[1] for b in batches) / len(batches),
+        "distinct_unions": len(samples),
         "cat_max": max(int(b.cat_X.max()) for b in batches),
         "entry_max": max(int(b.entry_id.max()) for b in batches),
         "ifc_max": max(int(b.edge_attr[:, 0].max()) for b in batches),
         "rpc_max": max(int(b.edge_attr[:, 1].max()) for b in batches),
     }
     return batches, stats
+
+
+def measure_latency_mae(epochs: int, device, seed: int = 7) -> dict:
+    """The accuracy half of the BASELINE metric: ingest a synthetic
+    Alibaba-shaped dataset through the full offline pipeline, train the
+    reference-default-shaped model (2 effective convs, H=32) for a few
+    epochs, and report the final test latency-MAE (reference epoch metric,
+    /root/reference/pert_gnn.py:290-294,348-350)."""
+    import tempfile
+
+    from pertgnn.data.collate import BatchLoader, collate_native
+    from pertgnn.data.dataset import build_data_list, split_60_20_20
+    from pertgnn.data.ingest import run_ingest
+    from pertgnn.data.synthetic import SyntheticConfig, write_dataset
+    from pertgnn.models import SAGEDeterministic
+    from pertgnn.train import evaluate, train_epoch
+    from pertgnn.train.optim import FusedAdam
+
+    torch.manual_seed(seed)
+    with tempfile.TemporaryDirectory() as root:
+        cfg = SyntheticConfig(n_entries=12, traces_per_entry=400, seed=seed)
+        write_dataset(root, cfg)
+        pdir = os.path.join(root, "processed")
+        run_ingest(data_root=os.path.join(root, "data"), processed_dir=pdir,
+                   verbose=False)
+        import joblib
+        import pandas as pd
+
+        tr2data = torch.load(os.path.join(pdir, "tr2data.pt"), weights_only=False)
+        runtime2graph = torch.load(os.path.join(pdir, "runtime2pertgraph_map.pt"),
+                                   weights_only=False)
+        entry2runtimes = joblib.load(os.path.join(pdir, "entry2runtimes.joblib"))
+        resource_df = pd.read_csv(os.path.join(pdir, "processed_resource_df.csv"))
+        resource_df["msname"] = resource_df["msname"].astype(int)
+        data_list = build_data_list(tr2data, entry2runtimes, runtime2graph,
+                                    resource_df)
+    train_list, _valid, test_list = split_60_20_20(data_list)
+    train_loader = BatchLoader(train_list, 170, shuffle=True, seed=seed,
+                               collate_fn=collate_native)
+    test_loader = BatchLoader(test_list, 170, shuffle=False,
+                              collate_fn=collate_native)
+    unique_ms_max = max(int(g["ms_id"].max()) for g in runtime2graph.values())
+    entry_id_max = max(int(s.entry_id) for s in data_list)
+    interface_id_max = max(int(s.edge_attr[:, 0].max()) for s in data_list)
+    rpctype_id_max = max(int(s.edge_attr[:, 1].max()) for s in data_list)
+    model = SAGEDeterministic(9, [unique_ms_max + 1], entry_id_max,
+                              interface_id_max, rpctype_id_max, 32, 1, 0.0)
+    model = model.to(device)
+    optimizer = FusedAdam(model.parameters(), lr=3e-3)
+    for _ in range(epochs):
+        train_epoch(model, train_loader, optimizer, 0.5, device)
+    mae, mape, _q = evaluate(model, test_loader, 0.5, device)
+    return {"latency_mae": mae, "mae_epochs": epochs,
+            "mae_dataset_traces": len(data_list), "mae_model": "2conv/32H"}
 
 
 def main():
@@ -106,9 +197,26 @@ def main():
     ap.add_argument("--tau", type=float, default=0.5)
     ap.add_argument("--precision", choices=["fp32", "bf16", "fp16"], default="bf16",
                     help="matmul compute precision (weights/activations stay fp32)")
+    ap.add_argument("--vocab", choices=["small", "realistic"], default="realistic",
+                    help="synthetic vocabulary scale (realistic = Alibaba-like: "
+                         "2k interfaces / 1k microservices / 256 entry-unions)")
+    ap.add_argument("--graph-mode", choices=["auto", "full", "split", "eager"],
+                    default="auto", help="hipGraph stepping mode (see module doc)")
     ap.add_argument("--no-hipgraph", action="store_true",
-                    help="disable hipGraph capture of the training step")
+                    help="disable hipGraph capture of the training step (= --graph-mode eager)")
+    ap.add_argument("--mae-epochs", type=int, default=25,
+                    help="epochs for the latency-MAE anchor probe (0 = skip)")
     args = ap.parse_args()
+
+    # `--gpus N` without a torchrun rendezvous: re-exec under
+    # torch.distributed.run with N local ranks (the flag is then authoritative)
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        os.execv(sys.executable, [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={args.gpus}",
+            "--master-addr", "127.0.0.1", "--master-port", "29531",
+            os.path.abspath(__file__), *sys.argv[1:],
+        ])
 
     from pertgnn.models import SAGEDeterministic
     from pertgnn.ops import functional as F
@@ -118,7 +226,9 @@ def main():
     from pertgnn.ops.functional import set_gemm_precision
 
     comm = Comm()
-    n_gpus = max(args.gpus, comm.world_size)
+    if args.gpus > 1 and comm.world_size != args.gpus and comm.rank == 0:
+        print(f"# note: --gpus {args.gpus} but WORLD_SIZE={comm.world_size}; "
+              "the rendezvous world size is authoritative", file=sys.stderr)
     on_gpu = torch.cuda.is_available()
     if on_gpu:
         set_gemm_precision(args.precision)
@@ -126,7 +236,8 @@ def main():
     torch.manual_seed(1234 + comm.rank)
 
     batches, stats = build_synthetic_batches(
-        args.n_batches, args.batch_size, seed=100 + comm.rank, device=device
+        args.n_batches, args.batch_size, seed=100 + comm.rank, device=device,
+        vocab=args.vocab,
     )
     # vocab sizes must agree across ranks (per-rank synthetic seeds differ)
     for key in ("cat_max", "entry_max", "ifc_max", "rpc_max"):
@@ -142,51 +253,98 @@ def main():
 
     model.train()
 
-    def step(i):
+    def step_compute(i):
+        """zero_grad + forward + loss + backward (no collectives unless the
+        engine hooks are enabled)."""
         b = batches[i % len(batches)]
         optimizer.zero_grad(set_to_none=False)
-        if engine is not None:
-            engine.reset()
         gp, _ = model(b.x, b.cat_X, b.edge_index, b.edge_attr,
                       b.pattern_num_nodes, b.rt_probs, b.entry_id, b.batch,
                       csr=b.csr, num_graphs=b.num_graphs)
         loss = F.quantile_loss(b.y, gp.flatten(), args.tau)
         loss.backward()
+        return loss
+
+    def step(i):
+        """Full eager training step (bucketed all-reduce overlapped with
+        backward via the engine hooks)."""
+        if engine is not None:
+            engine.reset()
+        loss = step_compute(i)
         if engine is not None:
             engine.finalize()
         optimizer.step()
         return loss
 
-    # hipGraph capture: the whole training step (fwd + loss + bwd + allreduce
-    # + Adam) is captured once per resident batch and replayed — removes all
-    # per-kernel launch gaps (guide: capture launch-bound inner loops).
+    mode = "eager" if args.no_hipgraph else args.graph_mode
+    if mode == "auto":
+        mode = "full" if comm.world_size == 1 else "split"
+    if not on_gpu:
+        mode = "eager"
+
+    # hipGraph capture (guide: capture launch-bound inner loops).  Each rank
+    # reports the capture level it achieved (2=full, 1=split, 0=eager) and
+    # the MINIMUM over ranks decides the common stepping mode — a rank that
+    # failed capture must not leave the others replaying graphs alone.
+    def try_capture(body):
+        for i in range(len(batches)):
+            body(i)  # allocation warmup per batch shape
+        torch.cuda.synchronize()
+        gobjs = []
+        pool = None
+        for i in range(len(batches)):
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, pool=pool):
+                body(i)
+            if pool is None:
+                pool = g.pool()
+            gobjs.append(g)
+        torch.cuda.synchronize()
+        return gobjs
+
     timed_step = step
-    if on_gpu and not args.no_hipgraph:
-        captured = False
+    level = 0
+    gobjs = None
+    if mode == "full":
         try:
-            for i in range(len(batches)):
-                step(i)  # allocation warmup per batch shape
-            torch.cuda.synchronize()
-            gobjs = []
-            pool = None
-            for i in range(len(batches)):
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g, pool=pool):
-                    step(i)
-                if pool is None:
-                    pool = g.pool()
-                gobjs.append(g)
-            torch.cuda.synchronize()
-            captured = True
-        except Exception as exc:  # pragma: no cover
-            print(f"# hipGraph capture unavailable ({exc}); stepping eagerly")
-        # all ranks must agree on the stepping mode (a rank that failed
-        # capture must not leave the others replaying graphs alone)
-        if comm.all_reduce_scalar(1.0 if captured else 0.0, op="min") >= 1.0:
-            def timed_step(i):
-                gobjs[i % len(gobjs)].replay()
-        else:
-            timed_step = step
+            gobjs = try_capture(step)
+            level = 2
+        except Exception as exc:  # pragma: no cover - hardware path
+            print(f"# full-step hipGraph capture unavailable ({exc}); "
+                  "trying compute-only capture", file=sys.stderr)
+            mode = "split"
+    if mode == "split" and level == 0:
+        try:
+            if engine is not None:
+                engine.enabled = False
+            try:
+                gobjs = try_capture(step_compute)
+            finally:
+                if engine is not None:
+                    engine.enabled = True
+            level = 1
+        except Exception as exc:  # pragma: no cover - hardware path
+            print(f"# compute hipGraph capture unavailable ({exc}); "
+                  "stepping eagerly", file=sys.stderr)
+    if mode == "eager":
+        level = 0
+
+    level = int(comm.all_reduce_scalar(float(level), op="min"))
+    if level == 2:
+        def timed_step(i):
+            gobjs[i % len(gobjs)].replay()
+    elif level == 1:
+        inv_ws = 1.0 / comm.world_size
+
+        def timed_step(i):
+            gobjs[i % len(gobjs)].replay()
+            if comm.distributed:
+                comm.all_reduce_(optimizer.flat_grad)
+                optimizer.flat_grad.mul_(inv_ws)
+            optimizer.step()
+    else:
+        timed_step = step
+    mode_used = {2: "full", 1: "split", 0: "eager"}[level]
 
     for i in range(args.warmup):
         timed_step(i)
@@ -204,6 +362,15 @@ def main():
     # max over ranks
     elapsed = comm.all_reduce_scalar(elapsed, op="max")
 
+    # latency-MAE anchor (rank 0 only; the other ranks idle at the final
+    # barrier) — skipped on CPU to keep the no-GPU default run fast
+    mae_info = {"latency_mae": None}
+    if args.mae_epochs > 0 and on_gpu and comm.rank == 0:
+        try:
+            mae_info = measure_latency_mae(args.mae_epochs, device)
+        except Exception as exc:  # pragma: no cover
+            print(f"# latency-MAE probe failed: {exc}", file=sys.stderr)
+
     graphs_per_sec = args.batch_size * comm.world_size * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1e3
     if comm.rank == 0:
@@ -220,11 +387,17 @@ def main():
             "vs_baseline": None,
             "dtype": args.precision if on_gpu else "fp32",
             "data": "synthetic",
+            "latency_mae": mae_info.get("latency_mae"),
+            "mae_detail": {k: v for k, v in mae_info.items() if k != "latency_mae"},
             "config": {
                 "model": f"PERT-GNN {args.layers}L/{args.hidden}H",
                 "global_batch": args.batch_size * comm.world_size,
                 "avg_nodes_per_batch": stats["avg_nodes"],
                 "avg_edges_per_batch": stats["avg_edges"],
+                "vocab": args.vocab,
+                "interfaces": stats["ifc_max"] + 1,
+                "entry_unions": stats["distinct_unions"],
+                "graph_mode": mode_used,
                 "parallelism": f"dp{comm.world_size}",
             },
         }))

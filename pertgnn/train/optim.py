@@ -123,6 +123,10 @@ class FlatGradAllReduce:
     def __init__(self, optimizer: FusedAdam, comm, bucket_cap_mb: float = 32.0):
         self.opt = optimizer
         self.comm = comm
+        # `enabled=False` silences the backward hooks (and finalize): used
+        # while capturing a compute-only hipGraph, where collectives must
+        # stay out of the captured stream (bench.py --graph-mode split)
+        self.enabled = True
         self.use_stream = comm.device.type == "cuda"
         self.comm_stream = torch.cuda.Stream() if self.use_stream else None
 
@@ -169,7 +173,7 @@ class FlatGradAllReduce:
             b["work"] = self.comm.all_reduce_(flat, async_op=True)
 
     def _hook(self, p):
-        if not self.comm.distributed:
+        if not self.comm.distributed or not self.enabled:
             return
         b = self.param2bucket[id(p)]
         b["pending"] -= 1
@@ -177,7 +181,7 @@ class FlatGradAllReduce:
             self._launch(b)
 
     def finalize(self):
-        if not self.comm.distributed:
+        if not self.comm.distributed or not self.enabled:
             return
         for b in self.buckets:
             if b["work"] is None:
