@@ -124,6 +124,9 @@ void launch_gemm_fp16_tn(const float*, const float*, float*, float*, int, int,
 void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   float*, long, int, int, int, int, int,
                                   hipStream_t);
+void launch_embed_grouped_scatter_bal(const float*, const int*, const int*,
+                                      const int*, const int*, float*, float*,
+                                      int, int, int, int, int, hipStream_t);
 void launch_vocab_scatter(const float*, const long*, long, float*, long, int,
                           int, int, int, hipStream_t);
 void launch_vocab_scatter_dual(const float*, const long*, int, float*, float*,
@@ -550,6 +553,30 @@ torch::Tensor embed_grouped_scatter(torch::Tensor g, torch::Tensor order,
                                dtable.data_ptr<float>(), num_src, (int)rows,
                                (int)p, (int)h, (int)g.size(1), (int)col_off,
                                cur_stream());
+  return dtable;
+}
+
+// Work-balanced deterministic grouped scatter: waves assigned per row in
+// proportion to its group size (row_map/wave_start built host-side at
+// _group_by time) — robust to the PERT interface-0 mega-group (quirk 6).
+torch::Tensor embed_grouped_scatter_bal(torch::Tensor g, torch::Tensor order,
+                                        torch::Tensor ptr,
+                                        torch::Tensor row_map,
+                                        torch::Tensor wave_start,
+                                        int64_t rows, int64_t h,
+                                        int64_t col_off) {
+  CHECK_IN(g); CHECK_IN(order); CHECK_IN(ptr);
+  CHECK_IN(row_map); CHECK_IN(wave_start);
+  TORCH_CHECK(wave_start.numel() == rows + 1,
+              "wave_start must have rows+1 entries");
+  const int n_waves = row_map.size(0);
+  auto dtable = torch::empty({rows, h}, g.options());
+  auto partial = torch::empty({(long)n_waves, h}, g.options());
+  launch_embed_grouped_scatter_bal(
+      g.data_ptr<float>(), order.data_ptr<int>(), ptr.data_ptr<int>(),
+      row_map.data_ptr<int>(), wave_start.data_ptr<int>(),
+      partial.data_ptr<float>(), dtable.data_ptr<float>(), n_waves,
+      (int)rows, (int)h, (int)g.size(1), (int)col_off, cur_stream());
   return dtable;
 }
 
@@ -1068,6 +1095,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("vocab_scatter_dual", &vocab_scatter_dual);
   mod.def("linear_dgrad", &linear_dgrad);
   mod.def("linear_wgrad", &linear_wgrad);
+  mod.def("embed_grouped_scatter_bal", &embed_grouped_scatter_bal);
   mod.def("bn_stats", &bn_stats);
   mod.def("bn_finalize_apply", &bn_finalize_apply);
   mod.def("bn_bwd_partials", &bn_bwd_partials);

@@ -256,6 +256,88 @@ __global__ void embed_grouped_p2_kernel(const float* __restrict__ partial,
   }
 }
 
+// Work-balanced variant: the uniform rows*P wave grid collapses under
+// skewed group sizes (quirk 6: PERT intra-ms edges all carry interface id 0,
+// so ONE table row owns ~half the edges — 8 sub-waves serialized ~12k
+// dependent gathers each, measured 39 ms vs 1 ms for an even distribution
+// of the same volume).  Here the host assigns waves PROPORTIONAL to group
+// size (row_map[w] = row of wave w, wave_start[row] = its first wave), so
+// every wave runs a bounded iteration count regardless of skew.  Reduction
+// order per row is (sub, stride P_row) — fixed by the data, deterministic.
+template <int VPT>
+__global__ void embed_grouped_bal_p1_kernel(
+    const float* __restrict__ g, const int* __restrict__ order,
+    const int* __restrict__ ptr, const int* __restrict__ row_map,
+    const int* __restrict__ wave_start, float* __restrict__ partial,
+    int n_waves, int h, int gstride, int col_off) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int w = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (w >= n_waves) return;
+  const int row = row_map[w];
+  const int w0 = wave_start[row];
+  const int P = wave_start[row + 1] - w0;
+  const int sub = w - w0;
+  float acc[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
+  for (int p = ptr[row] + sub; p < ptr[row + 1]; p += P) {
+    const long r = order[p];
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) acc[j] += g[r * gstride + col_off + c];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    if (c < h) partial[(long)w * h + c] = acc[j];
+  }
+}
+
+template <int VPT>
+__global__ void embed_grouped_bal_p2_kernel(
+    const float* __restrict__ partial, const int* __restrict__ wave_start,
+    float* __restrict__ dtable, int rows, int h) {
+  const int wid = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (row >= rows) return;
+  float acc[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
+  for (int s = wave_start[row]; s < wave_start[row + 1]; ++s) {
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * PERTGNN_WAVE;
+      if (c < h) acc[j] += partial[(long)s * h + c];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) {
+    const int c = lane + j * PERTGNN_WAVE;
+    if (c < h) dtable[(long)row * h + c] = acc[j];
+  }
+}
+
+void launch_embed_grouped_scatter_bal(const float* g, const int* order,
+                                      const int* ptr, const int* row_map,
+                                      const int* wave_start, float* partial,
+                                      float* dtable, int n_waves, int rows,
+                                      int h, int gstride, int col_off,
+                                      hipStream_t s) {
+  if (rows == 0) return;
+  const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
+  const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  switch (vpt) {
+#define CASE(V)                                                                  case V:                                                                          embed_grouped_bal_p1_kernel<V>                                                     <<<dim3(ceil_div(n_waves, WAVES_PER_BLOCK)), block, 0, s>>>(                       g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride,              col_off);                                                              embed_grouped_bal_p2_kernel<V>                                                     <<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(                          partial, wave_start, dtable, rows, h);                                 break;
+    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+#undef CASE
+    default: pertgnn_shape_fail("segops launcher", "vpt", vpt);
+  }
+}
+
 void launch_embed_grouped_scatter(const float* g, const int* order,
                                   const int* ptr, float* partial,
                                   float* dtable, long num_src, int rows, int P,

@@ -49,11 +49,12 @@ def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
     import time as _time
 
     model.train()
-    total_loss = 0.0
-    mape_sum = 0.0
     n_graphs = 0
     n_edges = 0
     n_nodes = 0
+    acc = None  # [loss*B sum, mape sum] accumulated ON DEVICE — the only
+    # host sync is the one .tolist() after the epoch (reference semantics
+    # float()'d per batch, pert_gnn.py:248-250, which serializes every step)
     t0 = _time.perf_counter()
     for batch in loader:
         with _nvtx("h2d"):
@@ -73,13 +74,16 @@ def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
         with _nvtx("optimizer"):
             optimizer.step()
         with torch.no_grad():
-            total_loss += float(loss.detach()) * b.num_graphs
-            mape_sum += float(((pred.detach() - b.y).abs() / b.y).sum())
+            if acc is None:
+                acc = torch.zeros(2, dtype=torch.float64, device=pred.device)
+            acc[0] += loss.detach().double() * b.num_graphs
+            acc[1] += ((pred.detach() - b.y).abs() / b.y).sum().double()
             n_graphs += b.num_graphs
             n_edges += b.edge_index.shape[1]
             n_nodes += b.x.shape[0]
     if device is not None and torch.cuda.is_available():
         torch.cuda.synchronize()
+    total_loss, mape_sum = (acc.tolist() if acc is not None else (0.0, 0.0))
     elapsed = _time.perf_counter() - t0
     if comm is not None and comm.distributed:
         total_loss = comm.all_reduce_scalar(total_loss)
@@ -102,17 +106,20 @@ def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
 @torch.no_grad()
 def evaluate(model, loader, tau, device, comm=None):
     model.eval()
-    mae = mape = qloss = 0.0
+    acc = None  # [mae, mape, qloss] sums on device; one sync per eval pass
     n_graphs = 0
     for batch in loader:
         b = batch.to(device) if device is not None else batch
         global_pred, _ = _forward(model, b)
         pred = global_pred.flatten()
         mae_s, mape_s, q_s = F.eval_metrics(b.y, pred, tau)
-        mae += float(mae_s)
-        mape += float(mape_s)
-        qloss += float(q_s)
+        if acc is None:
+            acc = torch.zeros(3, dtype=torch.float64, device=pred.device)
+        acc[0] += mae_s.double() if torch.is_tensor(mae_s) else mae_s
+        acc[1] += mape_s.double() if torch.is_tensor(mape_s) else mape_s
+        acc[2] += q_s.double() if torch.is_tensor(q_s) else q_s
         n_graphs += b.num_graphs
+    mae, mape, qloss = (acc.tolist() if acc is not None else (0.0, 0.0, 0.0))
     if comm is not None and comm.distributed:
         mae = comm.all_reduce_scalar(mae)
         mape = comm.all_reduce_scalar(mape)
