@@ -102,6 +102,9 @@ void launch_eval_metrics(const float*, const float*, float*, long, float,
                          hipStream_t);
 void launch_adam(float*, const float*, float*, float*, float*, long, float,
                  float, float, float, float, hipStream_t);
+void launch_adam_dynamic(float*, const float*, float*, float*, float*, float*,
+                         long, float, float, float, float, float, float,
+                         float, hipStream_t);
 void launch_gemm_f32_nt(const float*, const float*, const float*, float*, int,
                         int, int, bool, hipStream_t);
 void launch_gemm_f32_nn(const float*, const float*, const float*, float*, int,
@@ -369,6 +372,26 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
   launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
               v.data_ptr<float>(), state.data_ptr<float>(), p.numel(),
               (float)lr, (float)b1, (float)b2, (float)eps, (float)gscale, cur_stream());
+}
+
+// Adam with device-side dynamic loss scaling (sstate = [scale, clean-step
+// counter, found_inf]): scans the flat grad for non-finites, skips the
+// whole update on overflow, and adjusts the scale — all on device, so the
+// captured step replays with the scale evolving (GradScaler semantics).
+void adam_step_dynamic(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                       torch::Tensor v, torch::Tensor state,
+                       torch::Tensor sstate, double lr, double b1, double b2,
+                       double eps, double backoff, double growth,
+                       double growth_interval) {
+  CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v); CHECK_IN(state);
+  CHECK_IN(sstate);
+  TORCH_CHECK(sstate.numel() >= 3, "scaler state needs 3 elements");
+  launch_adam_dynamic(p.data_ptr<float>(), g.data_ptr<float>(),
+                      m.data_ptr<float>(), v.data_ptr<float>(),
+                      state.data_ptr<float>(), sstate.data_ptr<float>(),
+                      p.numel(), (float)lr, (float)b1, (float)b2, (float)eps,
+                      (float)backoff, (float)growth, (float)growth_interval,
+                      cur_stream());
 }
 
 // y = x @ w^T + b (torch Linear layout: w [out,in]); bf16 variant rounds
@@ -1143,6 +1166,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("quantile_loss_fwd", &quantile_loss_fwd);
   mod.def("quantile_loss_bwd", &quantile_loss_bwd);
   mod.def("eval_metrics", &eval_metrics);
+  mod.def("adam_step_dynamic", &adam_step_dynamic);
   mod.def("adam_step", &adam_step, py::arg("p"), py::arg("g"), py::arg("m"),
           py::arg("v"), py::arg("state"), py::arg("lr"), py::arg("b1"),
           py::arg("b2"), py::arg("eps"), py::arg("gscale") = 1.0);

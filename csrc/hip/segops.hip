@@ -1502,8 +1502,16 @@ void launch_eval_metrics(const float* y, const float* y_hat, float* out3,
 // step/bias state lives ON DEVICE (state[0]=step, state[1]=1-b1^t,
 // state[2]=1-b2^t) so the whole Adam update is hipGraph-replayable: each
 // replay increments the step and recomputes the bias corrections.
+// sstate (dynamic loss scaling, all device-resident so the whole step is
+// hipGraph-replayable with the scale evolving across replays):
+//   sstate[0] = current scale, sstate[1] = clean-step counter,
+//   sstate[2] = found_inf flag for THIS step.
+// A skipped step (non-finite grads) advances neither the Adam moments nor
+// the bias-correction counter — torch.cuda.amp.GradScaler semantics.
 __global__ void adam_tick_kernel(float* __restrict__ state, float b1,
-                                 float b2) {
+                                 float b2,
+                                 const float* __restrict__ sstate = nullptr) {
+  if (sstate && sstate[2] != 0.f) return;  // skipped step
   const float t = state[0] + 1.f;
   state[0] = t;
   state[1] = 1.f - powf(b1, t);
@@ -1514,13 +1522,16 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             const float* __restrict__ state, long numel,
                             float lr, float b1, float b2, float eps,
-                            float gscale) {
+                            float gscale,
+                            const float* __restrict__ sstate = nullptr) {
+  if (sstate && sstate[2] != 0.f) return;  // skipped step
   const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   const float step_size = lr / state[1];
   const float sqrt_bias2 = sqrtf(state[2]);
+  const float gs = sstate ? 1.f / sstate[0] : gscale;
   for (long t = i0; t < numel; t += stride) {
-    const float gt = g[t] * gscale;  // static loss-scaling unscale
+    const float gt = g[t] * gs;  // loss-scaling unscale
     const float mt = b1 * m[t] + (1.f - b1) * gt;
     const float vt = b2 * v[t] + (1.f - b2) * gt * gt;
     m[t] = mt;
@@ -1530,11 +1541,54 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
   }
 }
 
+__global__ void scaler_begin_kernel(float* __restrict__ sstate) {
+  sstate[2] = 0.f;
+}
+
+__global__ void grad_scan_kernel(const float* __restrict__ g, long numel,
+                                 float* __restrict__ sstate) {
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  float bad = 0.f;
+  for (long t = i0; t < numel; t += stride) {
+    const float x = g[t];
+    if (!isfinite(x)) { bad = 1.f; break; }
+  }
+  if (bad != 0.f) sstate[2] = 1.f;  // benign race: every writer stores 1
+}
+
+__global__ void scaler_update_kernel(float* __restrict__ sstate, float backoff,
+                                     float growth, float interval) {
+  if (sstate[2] != 0.f) {
+    sstate[0] = fmaxf(sstate[0] * backoff, 1.f);
+    sstate[1] = 0.f;
+  } else {
+    sstate[1] += 1.f;
+    if (sstate[1] >= interval) {
+      sstate[0] = fminf(sstate[0] * growth, 4294967296.f);
+      sstate[1] = 0.f;
+    }
+  }
+}
+
 void launch_adam(float* p, const float* g, float* m, float* v, float* state,
                  long numel, float lr, float b1, float b2, float eps, float gscale,
                  hipStream_t s) {
   if (numel == 0) return;
-  adam_tick_kernel<<<1, 1, 0, s>>>(state, b1, b2);
+  adam_tick_kernel<<<1, 1, 0, s>>>(state, b1, b2, nullptr);
   adam_kernel<<<grid_for(numel), 256, 0, s>>>(p, g, m, v, state, numel, lr,
-                                              b1, b2, eps, gscale);
+                                              b1, b2, eps, gscale, nullptr);
+}
+
+void launch_adam_dynamic(float* p, const float* g, float* m, float* v,
+                         float* state, float* sstate, long numel, float lr,
+                         float b1, float b2, float eps, float backoff,
+                         float growth, float interval, hipStream_t s) {
+  if (numel == 0) return;
+  scaler_begin_kernel<<<1, 1, 0, s>>>(sstate);
+  grad_scan_kernel<<<grid_for(numel), 256, 0, s>>>(g, numel, sstate);
+  adam_tick_kernel<<<1, 1, 0, s>>>(state, b1, b2, sstate);
+  adam_kernel<<<grid_for(numel), 256, 0, s>>>(p, g, m, v, state, numel, lr,
+                                              b1, b2, eps, 1.f, sstate);
+  scaler_update_kernel<<<1, 1, 0, s>>>(sstate, backoff, growth, interval);
 }

@@ -52,8 +52,11 @@ def build_parser():
                         help="generate a synthetic dataset in-place if processed/ is missing")
     parser.add_argument("--sync_bn", action="store_true",
                         help="exact-parity BatchNorm under DDP (statistics all-reduced)")
-    parser.add_argument("--loss_scale", type=float, default=1.0,
-                        help="static loss scaling (fp16 mode): loss*S backward, optimizer unscales")
+    parser.add_argument("--loss_scale", type=str, default="1.0",
+                        help="loss scaling (fp16 mode): a float S for static "
+                             "scaling (loss*S backward, optimizer unscales) or "
+                             "'dynamic' for GradScaler-style scaling with "
+                             "device-side overflow skip/backoff/growth")
     parser.add_argument("--precision", choices=["fp32", "bf16", "fp16"], default="fp32",
                         help="matmul compute precision on GPU")
     parser.add_argument("--hipgraph", action="store_true",
@@ -162,7 +165,12 @@ def main(argv=None):
     if torch.cuda.is_available():
         from pertgnn.ops.functional import set_gemm_precision
         set_gemm_precision(args.precision)
-    optimizer = FusedAdam(model.parameters(), lr=args.lr, grad_scale=args.loss_scale)
+    dynamic_scale = str(args.loss_scale).lower() == "dynamic"
+    static_scale = 1.0 if dynamic_scale else float(args.loss_scale)
+    args.loss_scale = static_scale  # loops use optimizer.scale_loss anyway
+    optimizer = FusedAdam(model.parameters(), lr=args.lr,
+                          grad_scale=static_scale,
+                          dynamic_scale=dynamic_scale)
     engine = FlatGradAllReduce(optimizer, comm) if comm.distributed else None
     log = JsonlLogger(args.metrics_jsonl, rank=comm.rank)
 

@@ -58,7 +58,9 @@ class GraphStepper:
                            b.batch, csr=b.csr, num_graphs=b.num_graphs)
         pred = gp.flatten()
         loss = F.quantile_loss(b.y, pred, self.tau)
-        if self.loss_scale != 1.0:
+        if hasattr(self.optimizer, "scale_loss"):
+            self.optimizer.scale_loss(loss).backward()
+        elif self.loss_scale != 1.0:
             (loss * self.loss_scale).backward()
         else:
             loss.backward()
@@ -86,6 +88,8 @@ class GraphStepper:
             "exp_avg": opt.exp_avg.detach().clone(),
             "exp_avg_sq": opt.exp_avg_sq.detach().clone(),
             "dev_state": opt.dev_state.detach().clone(),
+            "sstate": (opt.sstate.detach().clone()
+                       if getattr(opt, "sstate", None) is not None else None),
             "step_count": opt.step_count,
             "buffers": [(b, b.detach().clone()) for b in self.model.buffers()
                         if torch.is_tensor(b) and b.numel() > 0],
@@ -99,6 +103,8 @@ class GraphStepper:
             opt.exp_avg.copy_(snap["exp_avg"])
             opt.exp_avg_sq.copy_(snap["exp_avg_sq"])
             opt.dev_state.copy_(snap["dev_state"])
+            if snap["sstate"] is not None:
+                opt.sstate.copy_(snap["sstate"])
             opt.step_count = snap["step_count"]
             for b, saved in snap["buffers"]:
                 b.copy_(saved)

@@ -67,7 +67,14 @@ def train_epoch(model, loader, optimizer, tau, device, engine=None, comm=None,
             pred = global_pred.flatten()
             loss = F.quantile_loss(b.y, pred, tau)
         with _nvtx("backward"):
-            (loss * loss_scale).backward() if loss_scale != 1.0 else loss.backward()
+            # FusedAdam.scale_loss handles static AND dynamic loss scaling
+            # (the dynamic scale is a device tensor — no host sync)
+            if hasattr(optimizer, "scale_loss"):
+                optimizer.scale_loss(loss).backward()
+            elif loss_scale != 1.0:
+                (loss * loss_scale).backward()
+            else:
+                loss.backward()
         with _nvtx("allreduce"):
             if engine is not None:
                 engine.finalize()

@@ -26,7 +26,8 @@ def _trainable(params):
 
 class FusedAdam:
     def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
-                 grad_scale=1.0):
+                 grad_scale=1.0, dynamic_scale=False, init_scale=2.0 ** 16,
+                 growth_interval=2000, backoff=0.5, growth=2.0):
         self.params = _trainable(list(params))
         assert self.params, "no trainable parameters"
         assert all(p.dtype == torch.float32 for p in self.params), "fp32 master params only"
@@ -36,8 +37,22 @@ class FusedAdam:
         # static loss scaling: the loop multiplies the loss by grad_scale,
         # the optimizer divides the gradients back before the moment update
         self.grad_scale = float(grad_scale)
+        # dynamic loss scaling (GradScaler semantics, fp16 configs): the
+        # loop multiplies the loss by the DEVICE-resident scale (scale_loss),
+        # the fused step scans grads for non-finites, skips-and-backs-off on
+        # overflow and grows the scale after `growth_interval` clean steps —
+        # all device-side, so a captured step replays correctly
+        self.dynamic_scale = bool(dynamic_scale)
+        self.backoff = float(backoff)
+        self.growth = float(growth)
+        self.growth_interval = float(growth_interval)
         self.step_count = 0
         device = self.params[0].device
+        self.sstate = None
+        if self.dynamic_scale:
+            assert self.grad_scale == 1.0, "static and dynamic scaling are exclusive"
+            self.sstate = torch.tensor([float(init_scale), 0.0, 0.0],
+                                       dtype=torch.float32, device=device)
 
         total = sum(p.numel() for p in self.params)
         self.flat_param = torch.empty(total, dtype=torch.float32, device=device)
@@ -63,18 +78,48 @@ class FusedAdam:
             if p.grad is None or p.grad.data_ptr() != self.flat_grad[off:off + p.numel()].data_ptr():
                 p.grad = self.flat_grad[off:off + p.numel()].view_as(p.data)
 
+    def scale_loss(self, loss):
+        """Multiply the loss by the active scale before backward: the
+        device-resident dynamic scale (a 0-d tensor multiply, capturable),
+        the static grad_scale, or a no-op."""
+        if self.dynamic_scale:
+            return loss * self.sstate[0]
+        if self.grad_scale != 1.0:
+            return loss * self.grad_scale
+        return loss
+
     @torch.no_grad()
     def step(self):
-        self.step_count += 1
         b1, b2 = self.betas
-        inv = 1.0 / self.grad_scale
         if self.flat_param.is_cuda and has_hip():
-            ext().adam_step(self.flat_param, self.flat_grad, self.exp_avg,
-                            self.exp_avg_sq, self.dev_state, self.lr, b1, b2,
-                            self.eps, inv)
+            self.step_count += 1
+            if self.dynamic_scale:
+                ext().adam_step_dynamic(
+                    self.flat_param, self.flat_grad, self.exp_avg,
+                    self.exp_avg_sq, self.dev_state, self.sstate, self.lr,
+                    b1, b2, self.eps, self.backoff, self.growth,
+                    self.growth_interval)
+            else:
+                ext().adam_step(self.flat_param, self.flat_grad, self.exp_avg,
+                                self.exp_avg_sq, self.dev_state, self.lr, b1, b2,
+                                self.eps, 1.0 / self.grad_scale)
             return
         # eager fallback — identical formula (torch.optim.Adam)
         g = self.flat_grad
+        if self.dynamic_scale:
+            if not bool(torch.isfinite(g).all()):
+                self.sstate[0] = max(float(self.sstate[0]) * self.backoff, 1.0)
+                self.sstate[1] = 0.0
+                return  # skipped step: moments and counter untouched
+            inv = 1.0 / float(self.sstate[0])
+            self.sstate[1] += 1.0
+            if float(self.sstate[1]) >= self.growth_interval:
+                self.sstate[0] = min(float(self.sstate[0]) * self.growth,
+                                     4294967296.0)
+                self.sstate[1] = 0.0
+        else:
+            inv = 1.0 / self.grad_scale
+        self.step_count += 1
         self.exp_avg.mul_(b1).add_(g, alpha=(1 - b1) * inv)
         self.exp_avg_sq.mul_(b2).addcmul_(g, g, value=(1 - b2) * inv * inv)
         bias1 = 1 - b1 ** self.step_count
@@ -83,17 +128,16 @@ class FusedAdam:
         self.flat_param.addcdiv_(self.exp_avg, denom, value=-self.lr / bias1)
 
     def _synced_step_count(self) -> int:
-        """True step count even after hipGraph replays of the captured step
-        (replay advances only the device-side counter, not Python's)."""
-        if self.flat_param.is_cuda:
-            dev = int(self.dev_state[0].item())
-            if dev > self.step_count:
-                self.step_count = dev
+        """True step count on GPU: the device-side counter is authoritative —
+        hipGraph replays advance only it, and dynamic-scale overflow skips
+        advance only the Python counter."""
+        if self.flat_param.is_cuda and has_hip():
+            self.step_count = int(self.dev_state[0].item())
         return self.step_count
 
     # -- torch-optimizer-compatible surface ---------------------------------
     def state_dict(self):
-        return {
+        sd = {
             "step": self._synced_step_count(),
             "dev_state": self.dev_state,
             "exp_avg": self.exp_avg,
@@ -102,11 +146,16 @@ class FusedAdam:
             "betas": self.betas,
             "eps": self.eps,
         }
+        if self.sstate is not None:
+            sd["sstate"] = self.sstate
+        return sd
 
     def load_state_dict(self, sd):
         self.step_count = sd["step"]
         if "dev_state" in sd:
             self.dev_state.copy_(sd["dev_state"].to(self.dev_state.device))
+        if "sstate" in sd and self.sstate is not None:
+            self.sstate.copy_(sd["sstate"].to(self.sstate.device))
         self.exp_avg.copy_(sd["exp_avg"].to(self.exp_avg.device))
         self.exp_avg_sq.copy_(sd["exp_avg_sq"].to(self.exp_avg_sq.device))
         self.lr = sd.get("lr", self.lr)

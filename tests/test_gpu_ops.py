@@ -613,3 +613,62 @@ def test_deterministic_mode_bitwise(monkeypatch):
     # order) — loose agreement with the atomic path
     for n in ("convs.1.w4", "interface_embeds.weight"):
         assert torch.allclose(g1[n], g_atomic[n], atol=1e-2, rtol=1e-2), n
+
+
+def test_dynamic_loss_scaling_gpu():
+    """Device-side dynamic scaler: the fused kernels skip the update on a
+    non-finite grad, back the scale off, grow after clean steps, and keep
+    the trajectory identical to the unscaled optimizer — and the whole step
+    (scan + skip + scale update) replays correctly from a hipGraph."""
+    require_ext()
+    from pertgnn.train.optim import FusedAdam
+
+    torch.manual_seed(0)
+    m = torch.nn.Linear(8, 4).to(DEV)
+    opt = FusedAdam(m.parameters(), lr=1e-2, dynamic_scale=True,
+                    init_scale=8.0, growth_interval=3)
+    x = torch.randn(16, 8, device=DEV)
+    for i in range(8):
+        opt.zero_grad()
+        loss = m(x).pow(2).mean()
+        opt.scale_loss(loss).backward()
+        if i == 2:
+            opt.flat_grad[0] = float("inf")
+        before = opt.flat_param.clone()
+        sc_before = float(opt.sstate[0])
+        opt.step()
+        torch.cuda.synchronize()
+        if i == 2:
+            assert torch.equal(before, opt.flat_param), "overflow must skip"
+            assert float(opt.sstate[0]) == sc_before * 0.5
+        else:
+            assert not torch.equal(before, opt.flat_param)
+
+    torch.manual_seed(0)
+    m2 = torch.nn.Linear(8, 4).to(DEV)
+    opt2 = FusedAdam(m2.parameters(), lr=1e-2)
+    for i in range(8):
+        if i == 2:
+            continue
+        opt2.zero_grad()
+        m2(x).pow(2).mean().backward()
+        opt2.step()
+    torch.cuda.synchronize()
+    assert torch.allclose(opt.flat_param, opt2.flat_param, atol=1e-6), \
+        (opt.flat_param - opt2.flat_param).abs().max()
+    # dev_state counts only the 7 clean steps
+    assert int(opt.dev_state[0].item()) == 7
+
+    # hipGraph replay: capture one dynamic step, replay 3x, scale evolves
+    opt.zero_grad()
+    loss = m(x).pow(2).mean()
+    opt.scale_loss(loss).backward()
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        opt.step()
+    t0 = int(opt.dev_state[0].item())
+    for _ in range(3):
+        g.replay()
+    torch.cuda.synchronize()
+    assert int(opt.dev_state[0].item()) == t0 + 3

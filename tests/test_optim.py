@@ -111,3 +111,42 @@ def test_fused_adam_loss_scaling_equivalence():
 
     assert torch.allclose(p_a.data, p_b.data, atol=1e-7), \
         (p_a.data - p_b.data).abs().max()
+
+
+def test_dynamic_loss_scaling_cpu():
+    """Dynamic scaler semantics (eager path): overflow steps are skipped
+    with backoff, clean steps match the unscaled trajectory, and the scale
+    grows after growth_interval clean steps."""
+    from pertgnn.train.optim import FusedAdam
+
+    torch.manual_seed(0)
+    m = torch.nn.Linear(8, 4)
+    opt = FusedAdam(m.parameters(), lr=1e-2, dynamic_scale=True,
+                    init_scale=8.0, growth_interval=3)
+    x = torch.randn(16, 8)
+    for i in range(8):
+        opt.zero_grad()
+        loss = m(x).pow(2).mean()
+        opt.scale_loss(loss).backward()
+        if i == 2:
+            opt.flat_grad[0] = float("inf")
+        before = opt.flat_param.clone()
+        sc_before = float(opt.sstate[0])
+        opt.step()
+        if i == 2:
+            assert torch.equal(before, opt.flat_param), "overflow must skip"
+            assert float(opt.sstate[0]) == sc_before * 0.5
+        else:
+            assert not torch.equal(before, opt.flat_param)
+
+    torch.manual_seed(0)
+    m2 = torch.nn.Linear(8, 4)
+    opt2 = FusedAdam(m2.parameters(), lr=1e-2)
+    for i in range(8):
+        if i == 2:
+            continue
+        opt2.zero_grad()
+        m2(x).pow(2).mean().backward()
+        opt2.step()
+    assert torch.allclose(opt.flat_param, opt2.flat_param, atol=1e-6)
+    assert opt.sstate is not None and "sstate" in opt.state_dict()

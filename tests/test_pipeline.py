@@ -234,6 +234,70 @@ def test_cli_end_to_end_subprocess(tmp_path):
     assert all("test_mae" in r or "test" in str(r) for r in recs)
 
 
+def test_cli_hipgraph_flag_cpu(tmp_path):
+    """--hipgraph on CPU: GraphStepper degrades to eager stepping over the
+    fixed-composition resident batches; the CLI must still train and emit
+    the reference epoch lines."""
+    import subprocess
+    import sys
+
+    cmd = [sys.executable, "pert_gnn.py", "--synthetic", "--graph_type", "pert",
+           "--epochs", "2", "--num_layers", "1", "--hidden_channels", "16",
+           "--batch_size", "32", "--seed", "3", "--hipgraph",
+           "--processed_dir", str(tmp_path / "processed")]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                       cwd=str(Path(__file__).resolve().parents[1]))
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("Epoch:")]
+    assert len(lines) == 2, r.stdout[-2000:]
+    assert any("hipgraph stepping mode: eager" in l for l in r.stdout.splitlines())
+
+
+def test_graph_stepper_semantics_cpu(synthetic_workspace):
+    """GraphStepper (eager mode) must train exactly like a manual loop over
+    the same fixed batches in the same order — including the capture()
+    snapshot/restore noop on CPU."""
+    from pertgnn.models import SAGEDeterministic
+    from pertgnn.ops import functional as F
+    from pertgnn.train.capture import GraphStepper, make_resident_batches
+    from pertgnn.train.optim import FusedAdam
+    from pertgnn.data.collate import collate_native
+
+    root, (tr2data, entry2runtimes, _, runtime2pert, resource_df) = synthetic_workspace
+    data_list = build_data_list(tr2data, entry2runtimes, runtime2pert, resource_df, limit=48)
+    batches = make_resident_batches(data_list, 16, None, seed=5,
+                                    collate_fn=collate_native)
+
+    def build():
+        torch.manual_seed(9)
+        m = SAGEDeterministic(9, [64], 50, 50, 10, 16, 1, 0.0)
+        return m, FusedAdam(m.parameters(), lr=1e-3)
+
+    m1, o1 = build()
+    stepper = GraphStepper(m1, o1, None, None, 0.5, torch.device("cpu"),
+                           batches, seed=5)
+    assert stepper.capture() == "eager"
+    loss_sum, mape_sum, n = stepper.run_epoch_sums()
+    assert n == sum(b.num_graphs for b in batches)
+
+    m2, o2 = build()
+    order = torch.randperm(len(batches),
+                           generator=torch.Generator().manual_seed(5)).tolist()
+    tot = 0.0
+    for i in order:
+        b = batches[i]
+        o2.zero_grad(set_to_none=False)
+        gp, _ = m2(b.x, b.cat_X, b.edge_index, b.edge_attr,
+                   b.pattern_num_nodes, b.rt_probs, b.entry_id, b.batch,
+                   csr=b.csr, num_graphs=b.num_graphs)
+        loss = F.quantile_loss(b.y, gp.flatten(), 0.5)
+        loss.backward()
+        o2.step()
+        tot += float(loss.detach()) * b.num_graphs
+    assert torch.equal(o1.flat_param, o2.flat_param)
+    assert abs(tot - loss_sum) < 1e-3 * max(abs(tot), 1.0)
+
+
 def test_threaded_loader_equivalence(synthetic_workspace):
     """ThreadedLoader yields the same batches as iterating the base loader."""
     from pertgnn.data.prefetch import ThreadedLoader
