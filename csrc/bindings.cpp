@@ -128,8 +128,10 @@ void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   float*, long, int, int, int, int, int,
                                   hipStream_t);
 void launch_embed_grouped_scatter_bal(const float*, const int*, const int*,
-                                      const int*, const int*, float*, float*,
-                                      int, int, int, int, int, hipStream_t);
+                                      const int*, const int*, const int*,
+                                      const int*, float*, float*, float*,
+                                      int, int, int, int, int, int,
+                                      hipStream_t);
 void launch_vocab_scatter(const float*, const long*, long, float*, long, int,
                           int, int, int, hipStream_t);
 void launch_vocab_scatter_dual(const float*, const long*, int, float*, float*,
@@ -586,6 +588,8 @@ torch::Tensor embed_grouped_scatter_bal(torch::Tensor g, torch::Tensor order,
                                         torch::Tensor ptr,
                                         torch::Tensor row_map,
                                         torch::Tensor wave_start,
+                                        torch::Tensor row_map2,
+                                        torch::Tensor wave_start2,
                                         int64_t rows, int64_t h,
                                         int64_t col_off) {
   CHECK_IN(g); CHECK_IN(order); CHECK_IN(ptr);
@@ -593,12 +597,17 @@ torch::Tensor embed_grouped_scatter_bal(torch::Tensor g, torch::Tensor order,
   TORCH_CHECK(wave_start.numel() == rows + 1,
               "wave_start must have rows+1 entries");
   const int n_waves = row_map.size(0);
+  const int n_waves2 = row_map2.size(0);  // 0 => single-level fold
   auto dtable = torch::empty({rows, h}, g.options());
   auto partial = torch::empty({(long)n_waves, h}, g.options());
+  auto partial2 = torch::empty({(long)n_waves2, h}, g.options());
   launch_embed_grouped_scatter_bal(
       g.data_ptr<float>(), order.data_ptr<int>(), ptr.data_ptr<int>(),
       row_map.data_ptr<int>(), wave_start.data_ptr<int>(),
-      partial.data_ptr<float>(), dtable.data_ptr<float>(), n_waves,
+      n_waves2 ? row_map2.data_ptr<int>() : nullptr,
+      n_waves2 ? wave_start2.data_ptr<int>() : nullptr,
+      partial.data_ptr<float>(), partial2.data_ptr<float>(),
+      dtable.data_ptr<float>(), n_waves, n_waves2,
       (int)rows, (int)h, (int)g.size(1), (int)col_off, cur_stream());
   return dtable;
 }

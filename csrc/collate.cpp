@@ -13,6 +13,8 @@
 #include <ATen/Parallel.h>
 #include <cstring>
 
+#include "collate_core.h"
+
 namespace {
 
 int64_t total_rows(const std::vector<torch::Tensor>& ts, int dim = 0) {
@@ -106,12 +108,9 @@ std::vector<torch::Tensor> collate_native(
     }
   });
 
-  // stable counting sort by dst -> CSR
-  auto row_ptr = torch::zeros({n_nodes + 1}, opts(torch::kInt, pin));
-  int* rp = row_ptr.data_ptr<int>();
-  for (int64_t e = 0; e < n_edges; ++e) rp[dst0[e] + 1]++;
-  for (int64_t i = 0; i < n_nodes; ++i) rp[i + 1] += rp[i];
-  std::vector<int> cursor(rp, rp + n_nodes);
+  // stable counting sort by dst -> CSR (core shared with the sanitizer
+  // harness, csrc/collate_core.h)
+  auto row_ptr = torch::empty({n_nodes + 1}, opts(torch::kInt, pin));
   auto perm = torch::empty({n_edges}, torch::kLong);  // original -> csr slot
   long* pm = perm.data_ptr<long>();
   auto csr_src = torch::empty({n_edges}, opts(torch::kInt, pin));
@@ -119,13 +118,8 @@ std::vector<torch::Tensor> collate_native(
   auto edge_index = torch::empty({2, n_edges}, opts(torch::kLong, pin));
   long* ei_s = edge_index.data_ptr<long>();
   long* ei_d = ei_s + n_edges;
-  for (int64_t e = 0; e < n_edges; ++e) {
-    const int slot = cursor[dst0[e]]++;
-    pm[e] = slot;
-    cs[slot] = (int)src0[e];
-    ei_s[slot] = src0[e];
-    ei_d[slot] = dst0[e];
-  }
+  pertgnn_core::build_csr(src0.data(), dst0.data(), n_edges, n_nodes,
+                          row_ptr.data_ptr<int>(), pm, cs, ei_s, ei_d);
 
   // permute edge_attr into CSR order
   const int64_t acols = edge_attrs[0].size(1);
@@ -134,30 +128,18 @@ std::vector<torch::Tensor> collate_native(
   at::parallel_for(0, bsz, 1, [&](int64_t b, int64_t e) {
     for (int64_t i = b; i < e; ++i) {
       auto a = edge_attrs[i].contiguous();
-      const long* ap = a.data_ptr<long>();
-      for (int64_t j = 0; j < a.size(0); ++j) {
-        const long slot = pm[eoffs[i] + j];
-        for (int64_t cidx = 0; cidx < acols; ++cidx)
-          ea[slot * acols + cidx] = ap[j * acols + cidx];
-      }
+      pertgnn_core::permute_attrs(a.data_ptr<long>(), a.size(0), acols,
+                                  pm + eoffs[i], ea);
     }
   });
 
   // CSC over the CSR-ordered edges (stable counting sort by src)
-  auto col_ptr = torch::zeros({n_nodes + 1}, opts(torch::kInt, pin));
-  int* cp = col_ptr.data_ptr<int>();
-  for (int64_t e = 0; e < n_edges; ++e) cp[cs[e] + 1]++;
-  for (int64_t i = 0; i < n_nodes; ++i) cp[i + 1] += cp[i];
-  std::vector<int> ccursor(cp, cp + n_nodes);
+  auto col_ptr = torch::empty({n_nodes + 1}, opts(torch::kInt, pin));
   auto csc_dst = torch::empty({n_edges}, opts(torch::kInt, pin));
   auto csc_eid = torch::empty({n_edges}, opts(torch::kInt, pin));
-  int* cd = csc_dst.data_ptr<int>();
-  int* ce = csc_eid.data_ptr<int>();
-  for (int64_t e = 0; e < n_edges; ++e) {
-    const int slot = ccursor[cs[e]]++;
-    cd[slot] = (int)ei_d[e];
-    ce[slot] = (int)e;
-  }
+  pertgnn_core::build_csc(cs, ei_d, n_edges, n_nodes,
+                          col_ptr.data_ptr<int>(), csc_dst.data_ptr<int>(),
+                          csc_eid.data_ptr<int>());
 
   return {x, cat_X, edge_index, edge_attr, probs, pnn, nd,
           entry, batch, batch_ptr, y, row_ptr, csr_src, col_ptr,

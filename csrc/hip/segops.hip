@@ -282,7 +282,10 @@ __global__ void embed_grouped_bal_p1_kernel(
 #pragma unroll
   for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
   for (int p = ptr[row] + sub; p < ptr[row + 1]; p += P) {
-    const long r = order[p];
+    // order==nullptr reads rows of g directly (r = p): the SECOND reduction
+    // level folds level-1 partials this way (ptr = level-1 wave_start,
+    // each row's partials contiguous)
+    const long r = order ? order[p] : p;
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
       const int c = lane + j * PERTGNN_WAVE;
@@ -321,17 +324,19 @@ __global__ void embed_grouped_bal_p2_kernel(
   }
 }
 
-void launch_embed_grouped_scatter_bal(const float* g, const int* order,
-                                      const int* ptr, const int* row_map,
-                                      const int* wave_start, float* partial,
-                                      float* dtable, int n_waves, int rows,
-                                      int h, int gstride, int col_off,
-                                      hipStream_t s) {
+// Two-level balanced reduction: level 1 over the grouped g rows, optional
+// level 2 over the level-1 partials (needed when one group's partial count
+// itself is large — the interface-0 mega-group), final fold per row.
+void launch_embed_grouped_scatter_bal(
+    const float* g, const int* order, const int* ptr, const int* row_map,
+    const int* wave_start, const int* row_map2, const int* wave_start2,
+    float* partial, float* partial2, float* dtable, int n_waves,
+    int n_waves2, int rows, int h, int gstride, int col_off, hipStream_t s) {
   if (rows == 0) return;
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
   const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
   switch (vpt) {
-#define CASE(V)                                                                  case V:                                                                          embed_grouped_bal_p1_kernel<V>                                                     <<<dim3(ceil_div(n_waves, WAVES_PER_BLOCK)), block, 0, s>>>(                       g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride,              col_off);                                                              embed_grouped_bal_p2_kernel<V>                                                     <<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(                          partial, wave_start, dtable, rows, h);                                 break;
+#define CASE(V) case V: embed_grouped_bal_p1_kernel<V><<<dim3(ceil_div(n_waves, WAVES_PER_BLOCK)), block, 0, s>>>(g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); if (n_waves2 > 0) { embed_grouped_bal_p1_kernel<V><<<dim3(ceil_div(n_waves2, WAVES_PER_BLOCK)), block, 0, s>>>(partial, nullptr, wave_start, row_map2, wave_start2, partial2, n_waves2, h, h, 0); embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial2, wave_start2, dtable, rows, h); } else { embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial, wave_start, dtable, rows, h); } break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
     default: pertgnn_shape_fail("segops launcher", "vpt", vpt);

@@ -39,11 +39,11 @@ def _table_grad(m, g, idx, rows, h, col_off):
         g = g.float()
     if rows * h * 4 <= 160 * 1024 and not deterministic():
         return m.vocab_scatter(g, idx, rows, h, col_off)
-    order, ptr, row_map, wave_start = _group_by(idx, rows)
+    order, ptr, row_map, wave_start, row_map2, wave_start2 = _group_by(idx, rows)
     if g.dtype != torch.float32:
         g = g.float()
     return m.embed_grouped_scatter_bal(g, order, ptr, row_map, wave_start,
-                                       rows, h, col_off)
+                                       row_map2, wave_start2, rows, h, col_off)
 
 
 _GROUP_CACHE: "dict" = __import__("collections").OrderedDict()
@@ -74,7 +74,7 @@ def _group_by(idx: torch.Tensor, rows: int, iters_target: int = 32):
     hit = _GROUP_CACHE.get(key)
     if hit is not None:
         _GROUP_CACHE.move_to_end(key)
-        return hit[1], hit[2], hit[3], hit[4]
+        return hit[1], hit[2], hit[3], hit[4], hit[5], hit[6]
     idx_c = idx.contiguous()
     order = torch.argsort(idx_c, stable=True)  # ties in input order: the
     # grouped kernels' reduction order is then fully determined
@@ -82,24 +82,39 @@ def _group_by(idx: torch.Tensor, rows: int, iters_target: int = 32):
     ptr = torch.zeros(rows + 1, dtype=torch.int32, device=idx.device)
     ptr[1:] = counts.cumsum(0).to(torch.int32)
     order32 = order.to(torch.int32)
-    # wave assignment (host side; bincount above synced already): empty rows
-    # still get one wave so phase 2 writes their zeros
-    counts_h = counts.cpu()
-    waves_per_row = torch.clamp_min((counts_h + iters_target - 1) // iters_target, 1)
-    wave_start = torch.zeros(rows + 1, dtype=torch.int32)
-    wave_start[1:] = waves_per_row.cumsum(0).to(torch.int32)
-    row_map = torch.repeat_interleave(
-        torch.arange(rows, dtype=torch.int32), waves_per_row)
-    wave_start = wave_start.to(idx.device)
-    row_map = row_map.to(idx.device)
+
+    def assign(counts_h):
+        """waves per row ~ group size (empty rows still get one wave so the
+        final fold writes their zeros); returns (row_map, wave_start) CPU."""
+        wpr = torch.clamp_min((counts_h + iters_target - 1) // iters_target, 1)
+        ws = torch.zeros(counts_h.numel() + 1, dtype=torch.int32)
+        ws[1:] = wpr.cumsum(0).to(torch.int32)
+        rm = torch.repeat_interleave(
+            torch.arange(counts_h.numel(), dtype=torch.int32), wpr)
+        return rm, ws, wpr
+
+    counts_h = counts.cpu()  # bincount above synced already
+    row_map, wave_start, wpr = assign(counts_h)
+    # second reduction level when any row's partial count is itself big
+    # (the interface-0 mega-group: ~half the edges in one row)
+    if int(wpr.max()) > 64:
+        row_map2, wave_start2, _ = assign(wpr)
+    else:
+        row_map2 = torch.empty(0, dtype=torch.int32)
+        wave_start2 = torch.empty(0, dtype=torch.int32)
+    dev = idx.device
+    row_map, wave_start = row_map.to(dev), wave_start.to(dev)
+    row_map2, wave_start2 = row_map2.to(dev), wave_start2.to(dev)
     nbytes = (order32.numel() + ptr.numel() + row_map.numel()
-              + wave_start.numel()) * 4
+              + wave_start.numel() + row_map2.numel()
+              + wave_start2.numel()) * 4
     while _GROUP_CACHE and _GROUP_CACHE_BYTES[0] + nbytes > _GROUP_CACHE_CAP:
         _, old = _GROUP_CACHE.popitem(last=False)
         _GROUP_CACHE_BYTES[0] -= old[-1]
-    _GROUP_CACHE[key] = (idx, order32, ptr, row_map, wave_start, nbytes)
+    _GROUP_CACHE[key] = (idx, order32, ptr, row_map, wave_start, row_map2,
+                         wave_start2, nbytes)
     _GROUP_CACHE_BYTES[0] += nbytes
-    return order32, ptr, row_map, wave_start
+    return order32, ptr, row_map, wave_start, row_map2, wave_start2
 
 
 # ---------------------------------------------------------------------------
