@@ -117,7 +117,10 @@ def main(argv=None):
 
     cache_path = os.path.join(args.processed_dir, f"full_{args.graph_type}_data_list.pt")
     if os.path.exists(cache_path):
-        data_list = torch.load(cache_path, weights_only=False)
+        # accepts our TraceSample caches AND the reference's PyG Data
+        # pickles (no PyG needed — shim unpickler, pertgnn/data/pyg_compat)
+        from pertgnn.data.pyg_compat import load_data_list_any
+        data_list = load_data_list_any(cache_path)
     else:
         data_list = build_data_list(
             tr2data, entry2runtimes, runtime2graph, resource_df, limit=args.max_traces

@@ -330,3 +330,69 @@ def test_ingest_cached_csv_reload(synthetic_workspace, tmp_path):
         df1.reset_index(drop=True), df2.reset_index(drop=True), check_dtype=False)
     pd.testing.assert_frame_equal(
         rs1.reset_index(drop=True), rs2.reset_index(drop=True), check_dtype=False)
+
+
+def test_reference_pyg_cache_loads_without_pyg(tmp_path):
+    """full_*_data_list.pt written by the REFERENCE is a pickled list of
+    PyG Data objects (reference pert_gnn.py:317-322).  The shim unpickler
+    loads it without torch_geometric installed and converts to TraceSample
+    (rt_probs derived from the contiguous-pattern layout)."""
+    import sys
+    import types
+
+    # fabricate a minimal torch_geometric so we can WRITE a PyG-shaped
+    # pickle, then remove it so the LOAD runs without PyG (like this env)
+    tg = types.ModuleType("torch_geometric")
+    tg_data = types.ModuleType("torch_geometric.data")
+    tg_data_data = types.ModuleType("torch_geometric.data.data")
+
+    class GlobalStorage:
+        def __init__(self, mapping):
+            self._mapping = mapping
+
+    class Data:
+        def __init__(self, **kw):
+            self._store = GlobalStorage(dict(kw))
+
+    for cls in (GlobalStorage, Data):
+        cls.__module__ = "torch_geometric.data.data"
+        cls.__qualname__ = cls.__name__
+    tg_data_data.Data = Data
+    tg_data_data.GlobalStorage = GlobalStorage
+    sys.modules["torch_geometric"] = tg
+    sys.modules["torch_geometric.data"] = tg_data
+    sys.modules["torch_geometric.data.data"] = tg_data_data
+    try:
+        # two patterns: 3 nodes then 2 nodes
+        d = Data(
+            x=torch.randn(5, 9),
+            edge_index=torch.tensor([[0, 1, 3], [1, 2, 4]]),
+            edge_attr=torch.tensor([[1, 0], [0, 1], [2, 1]]),
+            cat_X=torch.tensor([[0], [1], [2], [0], [1]]),
+            node_depth=torch.tensor([[0], [1], [2], [0], [1]]),
+            pattern_num_nodes=torch.tensor([[3.], [3.], [3.], [2.], [2.]]),
+            pattern_probs=torch.tensor([[0.75], [0.25]]),
+            entry_id=torch.tensor([4]),
+            y=torch.tensor(42.0),
+        )
+        path = tmp_path / "full_pert_data_list.pt"
+        torch.save([d, d], str(path))
+    finally:
+        for m in ("torch_geometric.data.data", "torch_geometric.data",
+                  "torch_geometric"):
+            del sys.modules[m]
+
+    from pertgnn.data.pyg_compat import load_data_list_any
+
+    lst = load_data_list_any(str(path))
+    assert len(lst) == 2
+    s = lst[0]
+    assert s.x.shape == (5, 9) and float(s.y) == 42.0
+    assert int(s.entry_id) == 4
+    # rt_probs expanded per node from the contiguous pattern layout
+    assert torch.allclose(s.rt_probs.flatten(),
+                          torch.tensor([0.75, 0.75, 0.75, 0.25, 0.25]))
+    # our own format still loads through the same entry point
+    torch.save(lst, str(path))
+    lst2 = load_data_list_any(str(path))
+    assert torch.equal(lst2[1].x, s.x)
