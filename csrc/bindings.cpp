@@ -156,6 +156,11 @@ void launch_vocab_scatter_dual16(const void*, const long*, int, float*,
                                  float*, long, int, int, int, hipStream_t);
 void launch_gemm_bf16_nt_o16(const float*, const float*, const float*, void*,
                              int, int, int, hipStream_t);
+void launch_gemm_a16_glds_nt(const void*, const void*, const float*,
+                             const float*, void*, int, int, int, int, bool,
+                             hipStream_t);
+void launch_convert_w16(const float*, void*, long, hipStream_t);
+void launch_transpose_convert_w16(const float*, void*, int, int, hipStream_t);
 void launch_gemm_bf16_nn_a16(const void*, const float*, float*, int, int, int,
                              hipStream_t);
 void launch_gemm_bf16_tn_a16(const void*, const float*, float*, float*, int,
@@ -1069,12 +1074,25 @@ torch::Tensor linear_fwd_a16o16(torch::Tensor x, torch::Tensor w,
   auto y = torch::empty({m, n}, x.options());  // bf16
   const float* bias = nullptr;
   if (b.defined() && b.numel() > 0) bias = b.data_ptr<float>();
-  if (fp16c)
+  if (fp16c) {
     launch_gemm_fp16_nt_a16o16(x.data_ptr(), w.data_ptr<float>(), bias,
                                y.data_ptr(), m, n, k, cur_stream());
-  else
-    launch_gemm_bf16_nt_a16o16(x.data_ptr(), w.data_ptr<float>(), bias,
-                               y.data_ptr(), m, n, k, cur_stream());
+    return y;
+  }
+  // glds fast path (direct-to-LDS staging): needs 16-B-aligned rows, full
+  // BK tiles in k, and n a multiple of the 128-wide tile.  The bf16 weight
+  // copy costs one trivial convert kernel per call (n*k elements).
+  if (m >= 512 && n % 128 == 0 && k % 64 == 0) {
+    auto w16 = torch::empty({n, k}, x.options());  // bf16
+    launch_convert_w16(w.data_ptr<float>(), w16.data_ptr(), (long)n * k,
+                       cur_stream());
+    launch_gemm_a16_glds_nt(x.data_ptr(), w16.data_ptr(),
+                            w.data_ptr<float>(), bias, y.data_ptr(),
+                            /*c16=*/1, m, n, k, false, cur_stream());
+    return y;
+  }
+  launch_gemm_bf16_nt_a16o16(x.data_ptr(), w.data_ptr<float>(), bias,
+                             y.data_ptr(), m, n, k, cur_stream());
   return y;
 }
 
@@ -1085,12 +1103,33 @@ torch::Tensor linear_dgrad16_o16(torch::Tensor g, torch::Tensor w,
   const int n = w.size(0);
   const int k = w.size(1);
   auto dx = torch::empty({m, k}, g.options());  // bf16
-  if (fp16c)
+  if (fp16c) {
     launch_gemm_fp16_nn_a16o16(g.data_ptr(), w.data_ptr<float>(),
                                dx.data_ptr(), m, n, k, cur_stream());
-  else
-    launch_gemm_bf16_nn_a16o16(g.data_ptr(), w.data_ptr<float>(),
-                               dx.data_ptr(), m, n, k, cur_stream());
+    return dx;
+  }
+  // dgrad as glds-NT: dx[M,k] = g[M,n] . (w^T)[k,n]^T — one transposed
+  // bf16 copy of the weight turns the NN contraction into the NT fast
+  // path; the m-tail strip runs the register-staging NN kernel on the
+  // original fp32 weight.
+  if (m >= 512 && k % 128 == 0 && n % 64 == 0) {
+    auto wt16 = torch::empty({k, n}, g.options());  // bf16, transposed
+    launch_transpose_convert_w16(w.data_ptr<float>(), wt16.data_ptr(), n, k,
+                                 cur_stream());
+    launch_gemm_a16_glds_nt(g.data_ptr(), wt16.data_ptr(), nullptr, nullptr,
+                            dx.data_ptr(), /*c16=*/1, m, k, n, false,
+                            cur_stream());
+    const int m_done = m - m % 128;
+    if (m_done < m)
+      launch_gemm_bf16_nn_a16o16(
+          (const char*)g.data_ptr() + (long)m_done * n * 2,
+          w.data_ptr<float>(),
+          (char*)dx.data_ptr() + (long)m_done * k * 2, m - m_done, n, k,
+          cur_stream());
+    return dx;
+  }
+  launch_gemm_bf16_nn_a16o16(g.data_ptr(), w.data_ptr<float>(),
+                             dx.data_ptr(), m, n, k, cur_stream());
   return dx;
 }
 

@@ -860,3 +860,217 @@ void launch_gemm_fp16_tn_a16b16(const void* a_v, const void* b_v, float* c,
                                                               m, n, k2,
                                                               slices);
 }
+
+// ---------------------------------------------------------------------------
+// glds NT kernel: both operands bf16 in HBM, staged by global_load_lds
+// width-16 (direct-to-LDS DMA — no staging VGPRs, no ds_write pass; guide
+// "optimization ladder" step 3: +67% over register staging on its own).
+// LDS image is lane-linear row-major [128 rows][64 k] bf16 with the
+// st_16x32 XOR swizzle (byte ^= ((byte>>9)&1)<<5 within each 1024-B
+// subtile) applied on BOTH the per-lane glds SOURCE address and the
+// ds_read address: linear 128-B rows put every fragment lane-group on the
+// same bank row (8-way conflict); the swizzle spreads rows 4..7 of each
+// 8-row group one 32-B slot over (guide: bank-conflict 141x down).
+// One barrier per K-step: issue next tile's glds -> MFMA current ->
+// s_waitcnt vmcnt(0) -> barrier (the accepted step-3 stall).
+// Interior tiles only — the launcher routes edge tiles (m tail) to the
+// register-staging kernel via pointer offset; requires k % BK == 0 and
+// 16-B-aligned rows (lda % 8 == 0).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ unsigned glds_swz(unsigned byte_off) {
+  return byte_off ^ (((byte_off >> 9) & 1u) << 5);
+}
+
+template <int BM, int BN, int BK, typename TO, int THREADS = 256>
+__launch_bounds__(THREADS)
+__global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
+                                        const __bf16* __restrict__ b,
+                                        const float* __restrict__ bias,
+                                        TO* __restrict__ c, int m, int n,
+                                        int k, int relu) {
+  constexpr int NWAVE = THREADS / PERTGNN_WAVE;
+  constexpr int WCOL = NWAVE / 2;
+  constexpr int FM = (BM / 2) / 16, FN = (BN / WCOL) / 16;
+  constexpr int TILE_BYTES = BM * BK * 2;          // one operand K-tile
+  constexpr int GRPS = TILE_BYTES / 1024;          // 1 KiB per wave-glds
+  constexpr int GRPS_PER_WAVE = GRPS / NWAVE;
+  __shared__ __bf16 lds_a[2][BM * BK];
+  __shared__ __bf16 lds_b[2][BN * BK];
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int tiles_n = n / BN;
+  const int m0 = (bid / tiles_n) * BM;
+  const int n0 = (bid % tiles_n) * BN;
+  const int wave = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int wm = (wave / WCOL) * (BM / 2);
+  const int wn = (wave % WCOL) * (BN / WCOL);
+
+  // per-lane glds source mapping: lds slot (grp, lane*16) holds the element
+  // at LOGICAL tile offset swz(grp*1024 + lane*16) = (row, kbyte)
+  const unsigned l_off = glds_swz((unsigned)lane * 16);
+  const int src_row = (int)(l_off >> 7);           // within 8-row group
+  const int src_kb = (int)(l_off & 127);           // byte within row
+
+  const long lda = k;  // elements
+  auto stage = [&](const __bf16* g, int g0, int k0, __bf16* lds) {
+    // g0 = tile's first row; each wave stages GRPS_PER_WAVE 8-row groups
+#pragma unroll
+    for (int i = 0; i < GRPS_PER_WAVE; ++i) {
+      const int grp = wave + i * NWAVE;
+      const __bf16* src = (const __bf16*)((const char*)g +
+          (long)(g0 + grp * 8 + src_row) * lda * 2 + (long)k0 * 2 + src_kb);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lds + grp * 512),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage(a, m0, 0, lds_a[0]);
+  stage(b, n0, 0, lds_b[0]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  const int fi = lane & 15;
+  const int fk = (lane >> 4) * 8;
+  auto mma = [&](const __bf16* la, const __bf16* lb) {
+#pragma unroll
+    for (int s = 0; s < BK / 32; ++s) {
+      const int kb = (s * 32 + fk) * 2;
+      bf16x8 av[FM], bv[FN];
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi) {
+        const int row = wm + mi * 16 + fi;
+        av[mi] = *reinterpret_cast<const bf16x8*>(
+            (const char*)la + glds_swz((unsigned)(row * (BK * 2) + kb)));
+      }
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni) {
+        const int row = wn + ni * 16 + fi;
+        bv[ni] = *reinterpret_cast<const bf16x8*>(
+            (const char*)lb + glds_swz((unsigned)(row * (BK * 2) + kb)));
+      }
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              av[mi], bv[ni], acc[mi][ni], 0, 0, 0);
+    }
+  };
+
+  int buf = 0;
+  for (int k0 = BK; k0 < k; k0 += BK) {
+    stage(a, m0, k0, lds_a[buf ^ 1]);
+    stage(b, n0, k0, lds_b[buf ^ 1]);
+    mma(lds_a[buf], lds_b[buf]);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    buf ^= 1;
+  }
+  mma(lds_a[buf], lds_b[buf]);
+
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < FN; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wm + mi * 16 + frow + r;
+        const int col = n0 + wn + ni * 16 + fcol;
+        float v = acc[mi][ni][r];
+        if (bias) v += bias[col];
+        if (relu) v = fmaxf(v, 0.f);
+        c[(long)row * n + col] = (TO)v;
+      }
+}
+
+// fp32 [n][k] row-major -> bf16 [n][k] (weight operand for the glds path)
+__global__ void convert_w16_kernel(const float* __restrict__ w,
+                                   __bf16* __restrict__ o, long numel) {
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i0; t < numel; t += stride) o[t] = (__bf16)w[t];
+}
+
+// fp32 [rows][cols] -> bf16 [cols][rows] (transposed weight for dgrad-as-NT)
+__global__ void transpose_convert_w16_kernel(const float* __restrict__ w,
+                                             __bf16* __restrict__ o, int rows,
+                                             int cols) {
+  __shared__ float tile[32][33];
+  const int c0 = blockIdx.x * 32;
+  const int r0 = blockIdx.y * 32;
+  const int tx = threadIdx.x % 32;
+  const int ty = threadIdx.x / 32;
+  for (int dy = ty; dy < 32; dy += blockDim.x / 32) {
+    const int r = r0 + dy, ccol = c0 + tx;
+    tile[dy][tx] = (r < rows && ccol < cols) ? w[(long)r * cols + ccol] : 0.f;
+  }
+  __syncthreads();
+  for (int dy = ty; dy < 32; dy += blockDim.x / 32) {
+    const int ccol = c0 + dy, r = r0 + tx;
+    if (ccol < cols && r < rows) o[(long)ccol * rows + r] = (__bf16)tile[tx][dy];
+  }
+}
+
+void launch_convert_w16(const float* w, void* o, long numel, hipStream_t s) {
+  const int blocks = (int)min((numel + 255) / 256, (long)4096);
+  convert_w16_kernel<<<blocks, 256, 0, s>>>(w, (__bf16*)o, numel);
+}
+
+void launch_transpose_convert_w16(const float* w, void* o, int rows, int cols,
+                                  hipStream_t s) {
+  transpose_convert_w16_kernel<<<dim3((cols + 31) / 32, (rows + 31) / 32),
+                                 dim3(256), 0, s>>>(w, (__bf16*)o, rows, cols);
+}
+
+// A + B16 bf16, NT: full interior tiles through the glds kernel, the
+// m-tail strip through the register-staging kernel (pointer offset; it
+// stages the ORIGINAL fp32 weights, rounding to the same bf16 values the
+// convert kernel produced).
+void launch_gemm_a16_glds_nt(const void* a_v, const void* b16_v,
+                             const float* b32, const float* bias, void* c_v,
+                             int c16, int m, int n, int k, bool relu,
+                             hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  const __bf16* b = (const __bf16*)b16_v;
+  constexpr int BM = 128, BN = 128, BK = 64;
+  const int mt = m / BM;  // full row tiles
+  if (mt > 0) {
+    const int grid = mt * (n / BN);
+    if (c16)
+      gemm_a16_glds_nt_kernel<BM, BN, BK, __bf16>
+          <<<dim3(grid), dim3(256), 0, s>>>(a, b, bias, (__bf16*)c_v, m, n, k,
+                                            relu ? 1 : 0);
+    else
+      gemm_a16_glds_nt_kernel<BM, BN, BK, float>
+          <<<dim3(grid), dim3(256), 0, s>>>(a, b, bias, (float*)c_v, m, n, k,
+                                            relu ? 1 : 0);
+  }
+  const int m_done = mt * BM;
+  if (m_done < m && b32) {  // tail strip via the register-staging kernel
+    // (b32 == nullptr => caller handles the tail itself, e.g. dgrad whose
+    // fp32 weight is in the transposed layout)
+    const int ms = m - m_done;
+    const int grid = ((ms + 63) / 64) * ((n + 63) / 64);
+    if (c16)
+      gemm_bf16_nt_kernel<64, 64, 64, __bf16, __bf16, __bf16>
+          <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(
+              a + (long)m_done * k, b32, bias,
+              (__bf16*)c_v + (long)m_done * n, ms, n, k, relu ? 1 : 0);
+    else
+      gemm_bf16_nt_kernel<64, 64, 64, __bf16, __bf16, float>
+          <<<dim3(grid), dim3(BGEMM_THREADS), 0, s>>>(
+              a + (long)m_done * k, b32, bias,
+              (float*)c_v + (long)m_done * n, ms, n, k, relu ? 1 : 0);
+  }
+}
