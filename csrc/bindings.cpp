@@ -142,7 +142,7 @@ void launch_edge_attn_fused_fwd(const float*, const float*, const float*,
 void launch_edge_attn_fused_bwd(const float*, const float*, const float*,
                                 const float*, const long*, int, const float*,
                                 const int*, const int*, const int*,
-                                const int*, float*, float*, float*, float*,
+                                const int*, const int*, float*, float*,
                                 float*, int, int, long, hipStream_t);
 void launch_edge_attn_fused_fwd16(const void*, const float*, const float*,
                                   const long*, int, const int*, const int*,
@@ -150,7 +150,7 @@ void launch_edge_attn_fused_fwd16(const void*, const float*, const float*,
 void launch_edge_attn_fused_bwd16(const void*, int, const void*,
                                   const float*, const float*, const long*,
                                   int, const float*, const int*, const int*,
-                                  const int*, const int*, void*, void*, void*,
+                                  const int*, const int*, const int*, void*,
                                   void*, float*, int, int, long, hipStream_t);
 void launch_vocab_scatter_dual16(const void*, const long*, int, float*,
                                  float*, long, int, int, int, hipStream_t);
@@ -651,7 +651,7 @@ std::vector<torch::Tensor> edge_attn_fused_bwd(
     torch::Tensor g, torch::Tensor qkvs, torch::Tensor pifc,
     torch::Tensor prpc, torch::Tensor ea, torch::Tensor alpha,
     torch::Tensor row_ptr, torch::Tensor csr_src, torch::Tensor col_ptr,
-    torch::Tensor csc_eid) {
+    torch::Tensor csc_dst, torch::Tensor csc_eid) {
   CHECK_IN(g); CHECK_IN(qkvs); CHECK_IN(alpha);
   const int n = qkvs.size(0);
   const int h = qkvs.size(1) / 4;
@@ -660,29 +660,28 @@ std::vector<torch::Tensor> edge_attn_fused_bwd(
   auto dqkvs = torch::empty_like(qkvs);
   const bool b16 = qkvs.scalar_type() == torch::kBFloat16;
   auto eopt = b16 ? qkvs.options() : fopt;  // edge scratch matches qkvs dtype
+  // single-pass softmax backward: the [E,h] dek/dev scratch is gone — the
+  // col kernel regenerates the rank-1 per-edge grads from the per-edge
+  // scalars (dal = logit grad, alpha) and row gathers of q/g
   auto de = torch::empty({ne, h}, eopt);
-  auto dek = torch::empty({ne, h}, eopt);
-  auto dev = torch::empty({ne, h}, eopt);
+  auto dal = torch::empty({ne}, fopt);
   if (b16) {
-    auto dal = torch::empty({ne}, fopt);
     const int g16 = g.scalar_type() == torch::kBFloat16 ? 1 : 0;
     launch_edge_attn_fused_bwd16(
         g.data_ptr(), g16, qkvs.data_ptr(),
         pifc.data_ptr<float>(), prpc.data_ptr<float>(), ea.data_ptr<long>(),
         (int)ea.size(1), alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
         csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
-        csc_eid.data_ptr<int>(), dqkvs.data_ptr(), de.data_ptr(),
-        dek.data_ptr(), dev.data_ptr(), dal.data_ptr<float>(), n, h, ne,
-        cur_stream());
+        csc_dst.data_ptr<int>(), csc_eid.data_ptr<int>(), dqkvs.data_ptr(),
+        de.data_ptr(), dal.data_ptr<float>(), n, h, ne, cur_stream());
   } else {
-    auto dal = torch::empty({ne}, fopt);
     launch_edge_attn_fused_bwd(
         g.data_ptr<float>(), qkvs.data_ptr<float>(), pifc.data_ptr<float>(),
         prpc.data_ptr<float>(), ea.data_ptr<long>(), (int)ea.size(1),
         alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
         csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
-        csc_eid.data_ptr<int>(), dqkvs.data_ptr<float>(), de.data_ptr<float>(),
-        dek.data_ptr<float>(), dev.data_ptr<float>(), dal.data_ptr<float>(),
+        csc_dst.data_ptr<int>(), csc_eid.data_ptr<int>(),
+        dqkvs.data_ptr<float>(), de.data_ptr<float>(), dal.data_ptr<float>(),
         n, h, ne, cur_stream());
   }
   return {dqkvs, de};

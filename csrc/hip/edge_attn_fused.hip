@@ -193,16 +193,23 @@ __global__ void edge_attn_fused_fwd_kernel(
     alpha[p] = __expf(alpha[p] - m) * inv_s;
 }
 
-template <int VPT, bool VEC, typename QT = float, typename ET = float,
+template <int VPT, bool VEC, typename QT = float,
           int LPR = PERTGNN_WAVE, typename TG = float>
 __global__ void edge_attn_fused_bwd_row_kernel(
     const TG* __restrict__ g, const QT* __restrict__ qkvs,
     const float* __restrict__ pifc, const float* __restrict__ prpc,
     const long* __restrict__ ea, int astride, const float* __restrict__ alpha,
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
-    QT* __restrict__ dqkvs, ET* __restrict__ dek,
-    ET* __restrict__ dev, float* __restrict__ dal, int n, int h,
+    QT* __restrict__ dqkvs, float* __restrict__ dal, int n, int h,
     float scale) {
+  // SINGLE pass over the row's edges (the old two-pass form re-gathered
+  // ec+ke per edge after sdot was known).  Softmax backward re-expressed:
+  //   dq = scale * (Sum a*da*(k+e)  -  sdot * Sum a*(k+e))
+  // accumulates both sums alongside sdot = Sum a*da; the per-edge logit
+  // grad dl = scale*a*(da - sdot) is a SCALAR — dek = dl*q_row and
+  // dev = a*g_row are rank-1, so the [E,h] dek/dev scratch tensors are
+  // gone: the col kernel regenerates them from (dal, alpha) and row
+  // gathers of q/g.
   using S = Slice<VPT, VEC>;
   static_assert(VEC || LPR == PERTGNN_WAVE, "sub-wave rows need VEC layout");
   constexpr int RPW = PERTGNN_WAVE / LPR;
@@ -213,57 +220,54 @@ __global__ void edge_attn_fused_bwd_row_kernel(
   if (row >= n) return;
   const long ld = 4L * h;
 
-  float gr[VPT], qr[VPT], dqacc[VPT];
+  float gr[VPT], a1[VPT], a2[VPT];
   S::load(&g[(long)row * h], lane, h, gr);
-  S::load(&qkvs[row * ld], lane, h, qr);
 #pragma unroll
-  for (int j = 0; j < VPT; ++j) dqacc[j] = 0.f;
+  for (int j = 0; j < VPT; ++j) { a1[j] = 0.f; a2[j] = 0.f; }
 
   const int beg = row_ptr[row], end = row_ptr[row + 1];
   float sdot = 0.f;
   for (int p = beg; p < end; ++p) {
     const long src = csr_src[p];
-    const long a0 = ea[(long)p * astride];
-    const long a1 = ea[(long)p * astride + 1];
-    float ec[VPT], ve[VPT];
-    S::load_add(&pifc[a0 * h], &prpc[a1 * h], lane, h, ec);
+    const long e0 = ea[(long)p * astride];
+    const long e1 = ea[(long)p * astride + 1];
+    float ec[VPT], ke[VPT], ve[VPT];
+    S::load_add(&pifc[e0 * h], &prpc[e1 * h], lane, h, ec);
+    S::load(&qkvs[src * ld + h], lane, h, ke);
     S::load(&qkvs[src * ld + 2 * h], lane, h, ve);
     float part = 0.f;
 #pragma unroll
     for (int j = 0; j < VPT; ++j) part += gr[j] * (ve[j] + ec[j]);
     const float dalpha = subwave_reduce_sum<LPR>(part);
-    sdot += alpha[p] * dalpha;
-    if (lane == 0) dal[p] = dalpha;
-  }
-  for (int p = beg; p < end; ++p) {
-    const long src = csr_src[p];
-    const long a0 = ea[(long)p * astride];
-    const long a1 = ea[(long)p * astride + 1];
-    const float dalpha = dal[p];  // broadcast load (sub-group shares p)
     const float a = alpha[p];
-    const float dl = a * (dalpha - sdot) * scale;
-    float ec[VPT], ke[VPT], dekv[VPT], devv[VPT];
-    S::load_add(&pifc[a0 * h], &prpc[a1 * h], lane, h, ec);
-    S::load(&qkvs[src * ld + h], lane, h, ke);
+    sdot += a * dalpha;
+    if (lane == (p - beg) % LPR) dal[p] = dalpha;
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
-      dqacc[j] += dl * (ke[j] + ec[j]);
-      dekv[j] = dl * qr[j];
-      devv[j] = a * gr[j];
+      const float kec = ke[j] + ec[j];
+      a1[j] += a * dalpha * kec;
+      a2[j] += a * kec;
     }
-    S::store(&dek[(long)p * h], lane, h, dekv);
-    S::store(&dev[(long)p * h], lane, h, devv);
   }
-  S::store(&dqkvs[row * ld], lane, h, dqacc);          // dq
+  float dq[VPT];
+#pragma unroll
+  for (int j = 0; j < VPT; ++j) dq[j] = scale * (a1[j] - sdot * a2[j]);
+  S::store(&dqkvs[row * ld], lane, h, dq);             // dq
   S::store(&dqkvs[row * ld + 3 * h], lane, h, gr);     // dskip = g
+  // finalize per-edge logit grads: each lane rewrites exactly the slots it
+  // stored above (same-thread read-after-write)
+  for (int p = beg + lane; p < end; p += LPR)
+    dal[p] = scale * alpha[p] * (dal[p] - sdot);
 }
 
 template <int VPT, bool VEC, typename QT = float, typename ET = float,
-          int LPR = PERTGNN_WAVE>
+          int LPR = PERTGNN_WAVE, typename TG = float>
 __global__ void edge_attn_fused_bwd_col_kernel(
-    const ET* __restrict__ dek, const ET* __restrict__ dev,
-    const int* __restrict__ col_ptr, const int* __restrict__ csc_eid,
-    QT* __restrict__ dqkvs, ET* __restrict__ de, int n, int h) {
+    const TG* __restrict__ g, const QT* __restrict__ qkvs,
+    const float* __restrict__ alpha, const float* __restrict__ dal,
+    const int* __restrict__ col_ptr, const int* __restrict__ csc_dst,
+    const int* __restrict__ csc_eid, QT* __restrict__ dqkvs,
+    ET* __restrict__ de, int n, int h) {
   using S = Slice<VPT, VEC>;
   static_assert(VEC || LPR == PERTGNN_WAVE, "sub-wave rows need VEC layout");
   constexpr int RPW = PERTGNN_WAVE / LPR;
@@ -278,17 +282,21 @@ __global__ void edge_attn_fused_bwd_col_kernel(
   for (int j = 0; j < VPT; ++j) { ka[j] = 0.f; va[j] = 0.f; }
   for (int p = col_ptr[row]; p < col_ptr[row + 1]; ++p) {
     const long eid = csc_eid[p];
-    float dk1[VPT], dv1[VPT], des[VPT];
-    S::load(&dek[eid * h], lane, h, dk1);
-    S::load(&dev[eid * h], lane, h, dv1);
+    const long dst = csc_dst[p];
+    const float dl = dal[eid];
+    const float a = alpha[eid];
+    float qd[VPT], gd[VPT], des[VPT];
+    S::load(&qkvs[dst * ld], lane, h, qd);
+    S::load(&g[dst * h], lane, h, gd);
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
-      ka[j] += dk1[j];
-      va[j] += dv1[j];
-      des[j] = dk1[j] + dv1[j];
+      const float dk1 = dl * qd[j];
+      const float dv1 = a * gd[j];
+      ka[j] += dk1;
+      va[j] += dv1;
+      des[j] = dk1 + dv1;
     }
-    // every edge appears exactly once in the CSC sweep: de rides along,
-    // replacing a separate elementwise pass over dek/dev
+    // every edge appears exactly once in the CSC sweep: de rides along
     S::store(&de[eid * h], lane, h, des);
   }
   S::store(&dqkvs[(long)row * ld + h], lane, h, ka);
@@ -332,8 +340,8 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
                                 const long* ea, int astride,
                                 const float* alpha, const int* row_ptr,
                                 const int* csr_src, const int* col_ptr,
-                                const int* csc_eid, float* dqkvs, float* de,
-                                float* dek, float* dev, float* dal, int n,
+                                const int* csc_dst, const int* csc_eid,
+                                float* dqkvs, float* de, float* dal, int n,
                                 int h, long num_edges, hipStream_t stream) {
   if (n == 0) return;
   const float scale = 1.f / std::sqrt((float)h);
@@ -347,15 +355,15 @@ void launch_edge_attn_fused_bwd(const float* g, const float* qkvs,
     if (vec && (V % 4 == 0)) {                                                 \
       edge_attn_fused_bwd_row_kernel<V, true><<<grid, block, 0, stream>>>(     \
           g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,    \
-          dek, dev, dal, n, h, scale);                                         \
+          dal, n, h, scale);                                                   \
       edge_attn_fused_bwd_col_kernel<V, true><<<grid, block, 0, stream>>>(     \
-          dek, dev, col_ptr, csc_eid, dqkvs, de, n, h);                        \
+          g, qkvs, alpha, dal, col_ptr, csc_dst, csc_eid, dqkvs, de, n, h);    \
     } else {                                                                   \
       edge_attn_fused_bwd_row_kernel<V, false><<<grid, block, 0, stream>>>(    \
           g, qkvs, pifc, prpc, ea, astride, alpha, row_ptr, csr_src, dqkvs,    \
-          dek, dev, dal, n, h, scale);                                         \
+          dal, n, h, scale);                                                   \
       edge_attn_fused_bwd_col_kernel<V, false><<<grid, block, 0, stream>>>(    \
-          dek, dev, col_ptr, csc_eid, dqkvs, de, n, h);                        \
+          g, qkvs, alpha, dal, col_ptr, csc_dst, csc_eid, dqkvs, de, n, h);    \
     }                                                                          \
     break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
@@ -433,15 +441,13 @@ void launch_edge_attn_fused_bwd16(const void* g_v, int g16,
                                   const long* ea, int astride,
                                   const float* alpha, const int* row_ptr,
                                   const int* csr_src, const int* col_ptr,
-                                  const int* csc_eid, void* dqkvs_v,
-                                  void* de_v, void* dek_v, void* dev_v,
-                                  float* dal, int n, int h, long num_edges,
+                                  const int* csc_dst, const int* csc_eid,
+                                  void* dqkvs_v, void* de_v, float* dal,
+                                  int n, int h, long num_edges,
                                   hipStream_t stream) {
   const __bf16* qkvs = (const __bf16*)qkvs_v;
   __bf16* dqkvs = (__bf16*)dqkvs_v;
   __bf16* de = (__bf16*)de_v;
-  __bf16* dek = (__bf16*)dek_v;
-  __bf16* dev = (__bf16*)dev_v;
   if (n == 0) return;
   const float scale = 1.f / std::sqrt((float)h);
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
@@ -450,19 +456,25 @@ void launch_edge_attn_fused_bwd16(const void* g_v, int g16,
   const dim3 grid(ceil_div(n, rpb));
 #define BWD16(VPT, LPR)                                                        \
   do {                                                                         \
-    if (g16)                                                                   \
-      edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, __bf16, LPR, __bf16>   \
+    if (g16) {                                                                 \
+      edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, LPR, __bf16>           \
           <<<grid, block, 0, stream>>>((const __bf16*)g_v, qkvs, pifc, prpc,   \
                                        ea, astride, alpha, row_ptr, csr_src,   \
-                                       dqkvs, dek, dev, dal, n, h, scale);     \
-    else                                                                       \
-      edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, __bf16, LPR, float>    \
+                                       dqkvs, dal, n, h, scale);               \
+      edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR, __bf16>   \
+          <<<grid, block, 0, stream>>>((const __bf16*)g_v, qkvs, alpha, dal,   \
+                                       col_ptr, csc_dst, csc_eid, dqkvs, de,   \
+                                       n, h);                                  \
+    } else {                                                                   \
+      edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, LPR, float>            \
           <<<grid, block, 0, stream>>>((const float*)g_v, qkvs, pifc, prpc,    \
                                        ea, astride, alpha, row_ptr, csr_src,   \
-                                       dqkvs, dek, dev, dal, n, h, scale);     \
-    edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR>             \
-        <<<grid, block, 0, stream>>>(dek, dev, col_ptr, csc_eid, dqkvs, de, n, \
-                                     h);                                       \
+                                       dqkvs, dal, n, h, scale);               \
+      edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR, float>    \
+          <<<grid, block, 0, stream>>>((const float*)g_v, qkvs, alpha, dal,    \
+                                       col_ptr, csc_dst, csc_eid, dqkvs, de,   \
+                                       n, h);                                  \
+    }                                                                          \
   } while (0)
   if (h == 256) {
     if (lpr == 16) BWD16(16, 16);

@@ -977,21 +977,45 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
   }
   mma(lds_a[buf], lds_b[buf]);
 
+  // LDS-staged coalesced epilogue: the MFMA C fragment layout (lane owns
+  // col=l&15, rows (l>>4)*4+r) would store 2-B/4-B scalars at row stride —
+  // the measured bottleneck at [M,1024] outputs.  Each wave transposes one
+  // 16-row band at a time through its private LDS scratch, then writes
+  // 16-B lane chunks (8 lanes cover a full 128-B row of bf16).
+  __syncthreads();  // operand LDS is being re-purposed as scratch
+  constexpr int BNW = BN / WCOL;              // wave's output columns
+  constexpr int SW = BNW + 8;                 // padded scratch stride
+  TO* scratch = reinterpret_cast<TO*>(
+      reinterpret_cast<char*>(&lds_a[0][0]) + wave * 16 * SW * sizeof(TO));
   const int fcol = lane & 15;
   const int frow = (lane >> 4) * 4;
+  constexpr int EPL = 16 / sizeof(TO);        // elements per 16-B chunk
+  constexpr int CPR = BNW / EPL;              // chunks per scratch row
+  struct alignas(16) TOV { TO v[EPL]; };
 #pragma unroll
-  for (int mi = 0; mi < FM; ++mi)
+  for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
     for (int ni = 0; ni < FN; ++ni)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = m0 + wm + mi * 16 + frow + r;
-        const int col = n0 + wn + ni * 16 + fcol;
+        const int col = ni * 16 + fcol;
         float v = acc[mi][ni][r];
-        if (bias) v += bias[col];
+        if (bias) v += bias[n0 + wn + col];
         if (relu) v = fmaxf(v, 0.f);
-        c[(long)row * n + col] = (TO)v;
+        scratch[(frow + r) * SW + col] = (TO)v;
       }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int it = 0; it < 16 * CPR / PERTGNN_WAVE; ++it) {
+      const int chunk = it * PERTGNN_WAVE + lane;
+      const int r = chunk / CPR;
+      const int off = (chunk % CPR) * EPL;
+      TOV val = *reinterpret_cast<const TOV*>(&scratch[r * SW + off]);
+      *reinterpret_cast<TOV*>(
+          &c[(long)(m0 + wm + mi * 16 + r) * n + n0 + wn + off]) = val;
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  }
 }
 
 // fp32 [n][k] row-major -> bf16 [n][k] (weight operand for the glds path)

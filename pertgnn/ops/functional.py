@@ -236,20 +236,22 @@ def act16_enabled() -> bool:
 
 class _EdgeAttentionFusedFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_eid, out16=False):
+    def forward(ctx, qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr,
+                csc_dst, csc_eid, out16=False):
         m = ext()
         out, alpha = m.edge_attn_fused_fwd(qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, out16)
         ctx.save_for_backward(qkvs, pifc, prpc, edge_attr, alpha,
-                              row_ptr, csr_src, col_ptr, csc_eid)
+                              row_ptr, csr_src, col_ptr, csc_dst, csc_eid)
         return out
 
     @staticmethod
     def backward(ctx, g):
-        qkvs, pifc, prpc, edge_attr, alpha, row_ptr, csr_src, col_ptr, csc_eid = ctx.saved_tensors
+        (qkvs, pifc, prpc, edge_attr, alpha, row_ptr, csr_src, col_ptr,
+         csc_dst, csc_eid) = ctx.saved_tensors
         m = ext()
         dqkvs, de = m.edge_attn_fused_bwd(
             g.contiguous(), qkvs, pifc, prpc, edge_attr, alpha,
-            row_ptr, csr_src, col_ptr, csc_eid,
+            row_ptr, csr_src, col_ptr, csc_dst, csc_eid,
         )
         # dP tables (per-vocab segment sums of de) are independent of the
         # CSC dk/dv pass queued above — overlap them on the side stream.
@@ -262,7 +264,7 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
                 de32 = de.float() if de.dtype != torch.float32 else de
                 dpifc = _table_grad(m, de32, edge_attr[:, 0], pifc.shape[0], h, 0)
                 dprpc = _table_grad(m, de32, edge_attr[:, 1], prpc.shape[0], h, 0)
-            return dqkvs, dpifc, dprpc, None, None, None, None, None, None
+            return dqkvs, dpifc, dprpc, None, None, None, None, None, None, None
         cur = torch.cuda.current_stream()
         side = _side_stream()
         ev = torch.cuda.Event()
@@ -277,7 +279,7 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
         cur.wait_event(ev2)
         _mark_cross_stream(dpifc, cur)
         _mark_cross_stream(dprpc, cur)
-        return dqkvs, dpifc, dprpc, None, None, None, None, None, None
+        return dqkvs, dpifc, dprpc, None, None, None, None, None, None, None
 
 
 def edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr, out16=False):
@@ -286,9 +288,10 @@ def edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr, out16=False):
     P_ifc[a0] + P_rpc[a1] (exact refactoring of lin_edge(concat(ifc, rpc))).
     ``out16`` writes the aggregate bf16 (non-final layers feed BN, whose
     act16 path consumes/produces bf16 streams)."""
-    row_ptr, csr_src, col_ptr, _csc_dst, csc_eid = csr
+    row_ptr, csr_src, col_ptr, csc_dst, csc_eid = csr
     return _EdgeAttentionFusedFn.apply(
-        qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_eid, out16
+        qkvs, pifc, prpc, edge_attr, row_ptr, csr_src, col_ptr, csc_dst,
+        csc_eid, out16
     )
 
 
