@@ -882,7 +882,9 @@ __device__ __forceinline__ unsigned glds_swz(unsigned byte_off) {
   return byte_off ^ (((byte_off >> 9) & 1u) << 5);
 }
 
-template <int BM, int BN, int BK, typename TO, int THREADS = 256>
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4_t;
+
+template <int BM, int BN, int BK, typename TO, int THREADS, bool NTC>
 __launch_bounds__(THREADS)
 __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
                                         const __bf16* __restrict__ b,
@@ -895,8 +897,12 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
   constexpr int TILE_BYTES = BM * BK * 2;          // one operand K-tile
   constexpr int GRPS = TILE_BYTES / 1024;          // 1 KiB per wave-glds
   constexpr int GRPS_PER_WAVE = GRPS / NWAVE;
-  __shared__ __bf16 lds_a[2][BM * BK];
-  __shared__ __bf16 lds_b[2][BN * BK];
+  // one arena: [A0 A1 B0 B1] operand buffers; the C-store scratch reuses it
+  __shared__ char smem[2 * (BM + BN) * BK * 2];
+  const auto lds_ab = [&](int i) { return (__bf16*)(smem + i * TILE_BYTES); };
+  const auto lds_bb = [&](int i) {
+    return (__bf16*)(smem + 2 * TILE_BYTES + i * (BN * BK * 2));
+  };
   const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   const int tiles_n = n / BN;
   const int m0 = (bid / tiles_n) * BM;
@@ -914,7 +920,6 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
 
   const long lda = k;  // elements
   auto stage = [&](const __bf16* g, int g0, int k0, __bf16* lds) {
-    // g0 = tile's first row; each wave stages GRPS_PER_WAVE 8-row groups
 #pragma unroll
     for (int i = 0; i < GRPS_PER_WAVE; ++i) {
       const int grp = wave + i * NWAVE;
@@ -933,8 +938,8 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
 #pragma unroll
     for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  stage(a, m0, 0, lds_a[0]);
-  stage(b, n0, 0, lds_b[0]);
+  stage(a, m0, 0, lds_ab(0));
+  stage(b, n0, 0, lds_bb(0));
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -968,30 +973,30 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
 
   int buf = 0;
   for (int k0 = BK; k0 < k; k0 += BK) {
-    stage(a, m0, k0, lds_a[buf ^ 1]);
-    stage(b, n0, k0, lds_b[buf ^ 1]);
-    mma(lds_a[buf], lds_b[buf]);
+    stage(a, m0, k0, lds_ab(buf ^ 1));
+    stage(b, n0, k0, lds_bb(buf ^ 1));
+    mma(lds_ab(buf), lds_bb(buf));
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     buf ^= 1;
   }
-  mma(lds_a[buf], lds_b[buf]);
+  mma(lds_ab(buf), lds_bb(buf));
 
   // LDS-staged coalesced epilogue: the MFMA C fragment layout (lane owns
   // col=l&15, rows (l>>4)*4+r) would store 2-B/4-B scalars at row stride —
   // the measured bottleneck at [M,1024] outputs.  Each wave transposes one
   // 16-row band at a time through its private LDS scratch, then writes
-  // 16-B lane chunks (8 lanes cover a full 128-B row of bf16).
+  // 16-B lane chunks (8 lanes cover a full 128-B row of bf16).  NTC stores
+  // the C stream nontemporally — 370 MB of writes with zero reuse would
+  // otherwise wash the operand tiles out of L2.
   __syncthreads();  // operand LDS is being re-purposed as scratch
   constexpr int BNW = BN / WCOL;              // wave's output columns
   constexpr int SW = BNW + 8;                 // padded scratch stride
-  TO* scratch = reinterpret_cast<TO*>(
-      reinterpret_cast<char*>(&lds_a[0][0]) + wave * 16 * SW * sizeof(TO));
+  TO* scratch = reinterpret_cast<TO*>(smem + wave * 16 * SW * sizeof(TO));
   const int fcol = lane & 15;
   const int frow = (lane >> 4) * 4;
   constexpr int EPL = 16 / sizeof(TO);        // elements per 16-B chunk
   constexpr int CPR = BNW / EPL;              // chunks per scratch row
-  struct alignas(16) TOV { TO v[EPL]; };
 #pragma unroll
   for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
@@ -1010,9 +1015,13 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
       const int chunk = it * PERTGNN_WAVE + lane;
       const int r = chunk / CPR;
       const int off = (chunk % CPR) * EPL;
-      TOV val = *reinterpret_cast<const TOV*>(&scratch[r * SW + off]);
-      *reinterpret_cast<TOV*>(
-          &c[(long)(m0 + wm + mi * 16 + r) * n + n0 + wn + off]) = val;
+      const u32x4_t val = *reinterpret_cast<const u32x4_t*>(&scratch[r * SW + off]);
+      u32x4_t* dst = reinterpret_cast<u32x4_t*>(
+          &c[(long)(m0 + wm + mi * 16 + r) * n + n0 + wn + off]);
+      if constexpr (NTC)
+        __builtin_nontemporal_store(val, dst);
+      else
+        *dst = val;
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   }
@@ -1061,29 +1070,63 @@ void launch_transpose_convert_w16(const float* w, void* o, int rows, int cols,
 // m-tail strip through the register-staging kernel (pointer offset; it
 // stages the ORIGINAL fp32 weights, rounding to the same bf16 values the
 // convert kernel produced).
+static int glds_bk() {
+  static int v = [] {
+    const char* e = getenv("PERTGNN_GLDS_BK");
+    return (e && atoi(e) == 32) ? 32 : 64;
+  }();
+  return v;
+}
+static bool glds_t512() {
+  static bool v = [] {
+    const char* e = getenv("PERTGNN_GLDS_T512");
+    return e && atoi(e) == 1;
+  }();
+  return v;
+}
+static bool glds_ntc() {  // nontemporal C stores (default ON)
+  static bool v = [] {
+    const char* e = getenv("PERTGNN_GLDS_NTC");
+    return !(e && atoi(e) == 0);
+  }();
+  return v;
+}
+
 void launch_gemm_a16_glds_nt(const void* a_v, const void* b16_v,
                              const float* b32, const float* bias, void* c_v,
                              int c16, int m, int n, int k, bool relu,
                              hipStream_t s) {
   const __bf16* a = (const __bf16*)a_v;
   const __bf16* b = (const __bf16*)b16_v;
-  constexpr int BM = 128, BN = 128, BK = 64;
+  constexpr int BM = 128, BN = 128;
   const int mt = m / BM;  // full row tiles
   if (mt > 0) {
     const int grid = mt * (n / BN);
-    if (c16)
-      gemm_a16_glds_nt_kernel<BM, BN, BK, __bf16>
-          <<<dim3(grid), dim3(256), 0, s>>>(a, b, bias, (__bf16*)c_v, m, n, k,
-                                            relu ? 1 : 0);
-    else
-      gemm_a16_glds_nt_kernel<BM, BN, BK, float>
-          <<<dim3(grid), dim3(256), 0, s>>>(a, b, bias, (float*)c_v, m, n, k,
-                                            relu ? 1 : 0);
+    const int bk = glds_bk();
+    const bool t512 = glds_t512();
+    const bool ntc = glds_ntc();
+#define GLDS_DISPATCH(BKV, TH, NTCV)                                           \
+  do {                                                                         \
+    if (c16)                                                                   \
+      gemm_a16_glds_nt_kernel<BM, BN, BKV, __bf16, TH, NTCV>                   \
+          <<<dim3(grid), dim3(TH), 0, s>>>(a, b, bias, (__bf16*)c_v, m, n, k,  \
+                                           relu ? 1 : 0);                      \
+    else                                                                       \
+      gemm_a16_glds_nt_kernel<BM, BN, BKV, float, TH, NTCV>                    \
+          <<<dim3(grid), dim3(TH), 0, s>>>(a, b, bias, (float*)c_v, m, n, k,   \
+                                           relu ? 1 : 0);                      \
+  } while (0)
+    if (bk == 32 && k % 32 == 0) {
+      if (t512) { if (ntc) GLDS_DISPATCH(32, 512, true); else GLDS_DISPATCH(32, 512, false); }
+      else      { if (ntc) GLDS_DISPATCH(32, 256, true); else GLDS_DISPATCH(32, 256, false); }
+    } else {
+      if (t512) { if (ntc) GLDS_DISPATCH(64, 512, true); else GLDS_DISPATCH(64, 512, false); }
+      else      { if (ntc) GLDS_DISPATCH(64, 256, true); else GLDS_DISPATCH(64, 256, false); }
+    }
+#undef GLDS_DISPATCH
   }
   const int m_done = mt * BM;
   if (m_done < m && b32) {  // tail strip via the register-staging kernel
-    // (b32 == nullptr => caller handles the tail itself, e.g. dgrad whose
-    // fp32 weight is in the transposed layout)
     const int ms = m - m_done;
     const int grid = ((ms + 63) / 64) * ((n + 63) / 64);
     if (c16)
