@@ -913,10 +913,13 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
   const int wn = (wave % WCOL) * (BN / WCOL);
 
   // per-lane glds source mapping: lds slot (grp, lane*16) holds the element
-  // at LOGICAL tile offset swz(grp*1024 + lane*16) = (row, kbyte)
+  // at LOGICAL tile offset swz(grp*1024 + lane*16) = (row, kbyte); row
+  // width is BK*2 bytes, so one 1-KiB group covers 1024/(BK*2) rows
+  constexpr int ROW_BYTES = BK * 2;
+  constexpr int ROWS_PER_GRP = 1024 / ROW_BYTES;
   const unsigned l_off = glds_swz((unsigned)lane * 16);
-  const int src_row = (int)(l_off >> 7);           // within 8-row group
-  const int src_kb = (int)(l_off & 127);           // byte within row
+  const int src_row = (int)(l_off / ROW_BYTES);    // within the group
+  const int src_kb = (int)(l_off % ROW_BYTES);     // byte within row
 
   const long lda = k;  // elements
   auto stage = [&](const __bf16* g, int g0, int k0, __bf16* lds) {
@@ -924,7 +927,8 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
     for (int i = 0; i < GRPS_PER_WAVE; ++i) {
       const int grp = wave + i * NWAVE;
       const __bf16* src = (const __bf16*)((const char*)g +
-          (long)(g0 + grp * 8 + src_row) * lda * 2 + (long)k0 * 2 + src_kb);
+          (long)(g0 + grp * ROWS_PER_GRP + src_row) * lda * 2 +
+          (long)k0 * 2 + src_kb);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(lds + grp * 512),
@@ -1070,17 +1074,17 @@ void launch_transpose_convert_w16(const float* w, void* o, int rows, int cols,
 // m-tail strip through the register-staging kernel (pointer offset; it
 // stages the ORIGINAL fp32 weights, rounding to the same bf16 values the
 // convert kernel produced).
-static int glds_bk() {
+static int glds_bk() {  // default 32 (measured: +13-20% over 64 at model shapes)
   static int v = [] {
     const char* e = getenv("PERTGNN_GLDS_BK");
-    return (e && atoi(e) == 32) ? 32 : 64;
+    return (e && atoi(e) == 64) ? 64 : 32;
   }();
   return v;
 }
-static bool glds_t512() {
+static bool glds_t512() {  // default ON (8 waves; measured faster everywhere)
   static bool v = [] {
     const char* e = getenv("PERTGNN_GLDS_T512");
-    return e && atoi(e) == 1;
+    return !(e && atoi(e) == 0);
   }();
   return v;
 }
