@@ -69,6 +69,19 @@ def main():
               f"{td * 1e6:7.1f}us {fl / td / 1e12:5.0f}TF  "
               f"{trd * 1e6:7.1f}us {fl / trd / 1e12:5.0f}TF  relerr={rel2:.2e}")
 
+        # wgrad: dw[n,k] = g[m,n]^T x[m,k] (+ fused bias grad)
+        dw, db = C.linear_wgrad16_b16(g, x, True, False)
+        ref3 = g.float().t() @ x.float()
+        rel3 = (dw - ref3).abs().max().item() / ref3.abs().max().item()
+        reldb = (db - g.float().sum(0)).abs().max().item() / max(
+            g.float().sum(0).abs().max().item(), 1e-9)
+        tw = bench(C.linear_wgrad16_b16, args.iters, g, x, True, False)
+        trw = bench(lambda: g.t().float() @ x.float(), 5)
+        print(f"{'  wgrad':<16} {m:>7} {n:>5} {k:>4}  "
+              f"{tw * 1e6:7.1f}us {fl / tw / 1e12:5.0f}TF  "
+              f"{trw * 1e6:7.1f}us {fl / trw / 1e12:5.0f}TF  relerr={rel3:.2e}"
+              f" db={reldb:.2e}")
+
 
 if __name__ == "__main__":
     main()

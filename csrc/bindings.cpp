@@ -2,6 +2,8 @@
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
 
+#include <cstdlib>
+
 #include <c10/cuda/CUDAStream.h>
 
 namespace {
@@ -160,6 +162,8 @@ void launch_gemm_a16_glds_nt(const void*, const void*, const float*,
                              const float*, void*, int, int, int, int, bool,
                              hipStream_t);
 void launch_convert_w16(const float*, void*, long, hipStream_t);
+void launch_gemm_a16_glds_tn(const void*, const void*, float*, float*, int,
+                             int, int, hipStream_t);
 void launch_transpose_convert_w16(const float*, void*, int, int, hipStream_t);
 void launch_gemm_bf16_nn_a16(const void*, const float*, float*, int, int, int,
                              hipStream_t);
@@ -1147,14 +1151,23 @@ std::vector<torch::Tensor> linear_wgrad16_b16(torch::Tensor g, torch::Tensor x,
     db = torch::empty({n}, fopt);
     db_ptr = db.data_ptr<float>();
   }
-  if (fp16c)
+  if (fp16c) {
     launch_gemm_fp16_tn_a16b16(g.data_ptr(), x.data_ptr(),
                                dw.data_ptr<float>(), db_ptr, m, n, k,
                                cur_stream());
-  else
-    launch_gemm_bf16_tn_a16b16(g.data_ptr(), x.data_ptr(),
-                               dw.data_ptr<float>(), db_ptr, m, n, k,
-                               cur_stream());
+    return {dw, db};
+  }
+  // glds TN fast path (panel-major images + ds_read_b64_tr_b16 fragments);
+  // split-K atomics => not for the deterministic mode
+  const char* det = getenv("PERTGNN_DETERMINISTIC");
+  if (!(det && det[0] == '1') && m >= 512 && n % 128 == 0 && k % 128 == 0) {
+    launch_gemm_a16_glds_tn(g.data_ptr(), x.data_ptr(), dw.data_ptr<float>(),
+                            db_ptr, m, n, k, cur_stream());
+    return {dw, db};
+  }
+  launch_gemm_bf16_tn_a16b16(g.data_ptr(), x.data_ptr(),
+                             dw.data_ptr<float>(), db_ptr, m, n, k,
+                             cur_stream());
   return {dw, db};
 }
 

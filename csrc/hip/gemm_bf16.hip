@@ -1145,3 +1145,195 @@ void launch_gemm_a16_glds_nt(const void* a_v, const void* b16_v,
               (float*)c_v + (long)m_done * n, ms, n, k, relu ? 1 : 0);
   }
 }
+
+// ---------------------------------------------------------------------------
+// glds TN kernel (wgrad): dw[n,k2] = sum_m g[m,n]^T x[m,k2].  Both operands
+// are contract(m)-major in HBM, so the NT kernel's row-linear glds image
+// cannot feed k-minor MFMA fragments.  Instead each operand tile is staged
+// as PANEL-MAJOR images — 16-column panels of [BKM m][16 cols] row-major
+// (the per-lane glds source addresses assemble them directly) — and the
+// MFMA fragments are read with gfx950's ds_read_b64_tr_b16 hardware
+// transpose-read (lane gets 4 contract-consecutive elements at 32-B
+// stride; two reads make the 8-element k fragment).  Split-K over m with
+// atomicAdd combine, same contract as the register-staging TN kernel.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4_g;
+
+template <int BM /*n cols*/, int BN /*k2 cols*/, int BKM /*m per step*/,
+          int THREADS = 512>
+__launch_bounds__(THREADS)
+__global__ void gemm_a16_glds_tn_kernel(const __bf16* __restrict__ a,
+                                        const __bf16* __restrict__ b,
+                                        float* __restrict__ c,
+                                        float* __restrict__ dbias, int m,
+                                        int n, int k2, int slices) {
+  constexpr int NWAVE = THREADS / PERTGNN_WAVE;
+  constexpr int WCOL = NWAVE / 2;
+  constexpr int FM = (BM / 2) / 16, FN = (BN / WCOL) / 16;
+  constexpr int PANEL_BYTES = BKM * 32;            // [BKM][16] bf16
+  constexpr int TILE_BYTES = BM * BKM * 2;
+  constexpr int GRPS = TILE_BYTES / 1024;          // 1 KiB per wave-glds
+  constexpr int GRPS_PER_WAVE = GRPS / NWAVE;
+  constexpr int ROWS_PER_GRP = 1024 / 32;          // 32 m-rows per group
+  __shared__ char smem[2 * (BM + BN) * BKM * 2];
+  const auto lds_ab = [&](int i) { return smem + i * TILE_BYTES; };
+  const auto lds_bb = [&](int i) {
+    return smem + 2 * TILE_BYTES + i * (BN * BKM * 2);
+  };
+  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int tiles_k = k2 / BN;
+  const int tile_id = bid / slices;
+  const int slice = bid % slices;
+  const int n0 = (tile_id / tiles_k) * BM;
+  const int k0 = (tile_id % tiles_k) * BN;
+  const int wave = threadIdx.x / PERTGNN_WAVE;
+  const int lane = threadIdx.x % PERTGNN_WAVE;
+  const int wm = (wave / WCOL) * (BM / 2);
+  const int wn = (wave % WCOL) * (BN / WCOL);
+
+  // m range of this slice (multiples of BKM; the launcher routes the
+  // global m tail to the register-staging kernel)
+  const int m_full = m - m % BKM;
+  const int per_slice = ((m_full / BKM + slices - 1) / slices) * BKM;
+  const int c_beg = slice * per_slice;
+  const int c_end = min(m_full, c_beg + per_slice);
+  if (c_beg >= c_end) return;
+
+  // panel-major glds staging: group g covers half a panel (32 m x 16 cols);
+  // lane's 16-B chunk = 8 columns of one m-row of the panel
+  auto stage = [&](const __bf16* src_base, long ld, int col0, int mrow0,
+                   char* lds) {
+#pragma unroll
+    for (int i = 0; i < GRPS_PER_WAVE; ++i) {
+      const int grp = wave + i * NWAVE;
+      const int panel = grp / (PANEL_BYTES / 1024);
+      const int half = grp % (PANEL_BYTES / 1024);
+      const int mrow = half * ROWS_PER_GRP + (lane >> 1);
+      const int col = panel * 16 + (lane & 1) * 8;
+      const __bf16* src =
+          src_base + (long)(mrow0 + mrow) * ld + col0 + col;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lds + grp * 1024),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage(a, n, n0, c_beg, lds_ab(0));
+  stage(b, k2, k0, c_beg, lds_bb(0));
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  const bool do_bias = (dbias != nullptr) && (k0 == 0);
+  float dbsum = 0.f;
+  const int bcol = threadIdx.x;  // one thread per A column (BM <= THREADS)
+
+  // fragment read: column (l&15) of the 16-col panel, contract rows
+  // (l>>4)*8 + s*32 .. +7 via two transpose-reads (j stride = 32 B)
+  auto frag = [&](const char* img, int block16, int s) {
+    const char* p = img + (long)block16 * PANEL_BYTES +
+                    (s * 32 + ((lane >> 4) * 8)) * 32 + (lane & 15) * 2;
+    bf16x4_g lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4_g*)p);
+    bf16x4_g hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) bf16x4_g*)(p + 4 * 32));
+    union { bf16x8 v8; bf16x4_g v4[2]; } u;
+    u.v4[0] = lo;
+    u.v4[1] = hi;
+    return u.v8;
+  };
+
+  auto mma = [&](const char* la, const char* lb) {
+#pragma unroll
+    for (int s = 0; s < BKM / 32; ++s) {
+      bf16x8 av[FM], bv[FN];
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi) av[mi] = frag(la, (wm >> 4) + mi, s);
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni) bv[ni] = frag(lb, (wn >> 4) + ni, s);
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              av[mi], bv[ni], acc[mi][ni], 0, 0, 0);
+    }
+  };
+
+  auto bias_acc = [&](const char* la) {
+    if (!do_bias || bcol >= BM) return;
+    const char* p = la + (bcol / 16) * PANEL_BYTES + (bcol % 16) * 2;
+#pragma unroll
+    for (int r = 0; r < BKM; ++r)
+      dbsum += (float)*reinterpret_cast<const __bf16*>(p + r * 32);
+  };
+
+  int buf = 0;
+  for (int cc = c_beg + BKM; cc < c_end; cc += BKM) {
+    stage(a, n, n0, cc, lds_ab(buf ^ 1));
+    stage(b, k2, k0, cc, lds_bb(buf ^ 1));
+    mma(lds_ab(buf), lds_bb(buf));
+    bias_acc(lds_ab(buf));
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    buf ^= 1;
+  }
+  mma(lds_ab(buf), lds_bb(buf));
+  bias_acc(lds_ab(buf));
+  if (do_bias && bcol < BM) atomicAdd(&dbias[n0 + bcol], dbsum);
+
+  const int fcol = lane & 15;
+  const int frow = (lane >> 4) * 4;
+#pragma unroll
+  for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < FN; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = n0 + wm + mi * 16 + frow + r;
+        const int col = k0 + wn + ni * 16 + fcol;
+        if (slices == 1)
+          c[(long)row * k2 + col] = acc[mi][ni][r];
+        else
+          atomicAdd(&c[(long)row * k2 + col], acc[mi][ni][r]);
+      }
+}
+
+// Both operands bf16 contract-major (g [m][n], x [m][k2]); m tail rows
+// beyond the last BKM multiple go through the register-staging TN kernel
+// as an extra atomic slice.
+void launch_gemm_a16_glds_tn(const void* a_v, const void* b_v, float* c,
+                             float* dbias, int m, int n, int k2,
+                             hipStream_t s) {
+  const __bf16* a = (const __bf16*)a_v;
+  const __bf16* b = (const __bf16*)b_v;
+  constexpr int BM = 128, BN = 128, BKM = 64;
+  const int tiles = (n / BM) * (k2 / BN);
+  int slices = 1;
+  while (tiles * slices < 512 && slices < 64 &&
+         (long)slices * BKM * 4 < m)
+    slices *= 2;
+  HIP_CHECK(hipMemsetAsync(c, 0, (long)n * k2 * sizeof(float), s));
+  if (dbias) HIP_CHECK(hipMemsetAsync(dbias, 0, n * sizeof(float), s));
+  gemm_a16_glds_tn_kernel<BM, BN, BKM>
+      <<<dim3(tiles * slices), dim3(512), 0, s>>>(a, b, c, dbias, m, n, k2,
+                                                  slices);
+  const int m_full = m - m % BKM;
+  if (m_full < m) {
+    // register-staging kernel on the tail rows; slices=2 with a grid that
+    // only contains slice 0 makes it ATOMIC-add its contribution (its
+    // per_slice covers the whole <BKM-row strip in slice 0)
+    const int tiles_t = ((n + 127) / 128) * ((k2 + 127) / 128);
+    gemm_bf16_tn_kernel<128, 128, 32, __bf16, __bf16, __bf16>
+        <<<dim3(tiles_t * 2), dim3(BGEMM_THREADS), 0, s>>>(
+            a + (long)m_full * n, b + (long)m_full * k2, c, dbias,
+            m - m_full, n, k2, 2);
+  }
+}
