@@ -1235,15 +1235,18 @@ __global__ void gemm_a16_glds_tn_kernel(const __bf16* __restrict__ a,
   float dbsum = 0.f;
   const int bcol = threadIdx.x;  // one thread per A column (BM <= THREADS)
 
-  // fragment read: column (l&15) of the 16-col panel, contract rows
-  // (l>>4)*8 + s*32 .. +7 via two transpose-reads (j stride = 32 B)
+  // fragment read via ds_read_b64_tr_b16: each 16-lane group passes
+  // CONSECUTIVE 8-B addresses covering one 128-B [4 m][16 col] sub-tile;
+  // the hardware redistributes so lane receives COLUMN (l&15) — i.e. 4
+  // contract-consecutive elements.  Two reads (sub-tiles m+0..3, m+4..7)
+  // assemble the 8-element MFMA k fragment.
   auto frag = [&](const char* img, int block16, int s) {
     const char* p = img + (long)block16 * PANEL_BYTES +
-                    (s * 32 + ((lane >> 4) * 8)) * 32 + (lane & 15) * 2;
+                    (s * 32 + ((lane >> 4) * 8)) * 32 + (lane & 15) * 8;
     bf16x4_g lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
         (__attribute__((address_space(3))) bf16x4_g*)p);
     bf16x4_g hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-        (__attribute__((address_space(3))) bf16x4_g*)(p + 4 * 32));
+        (__attribute__((address_space(3))) bf16x4_g*)(p + 128));
     union { bf16x8 v8; bf16x4_g v4[2]; } u;
     u.v4[0] = lo;
     u.v4[1] = hi;
