@@ -197,6 +197,13 @@ def main(argv=None):
         mode = stepper.capture()
         log.print0(f"# hipgraph stepping mode: {mode} "
                    f"({len(resident)} resident batches)")
+        # eval sets collated once and held resident too (order preserved)
+        valid_loader = make_resident_batches(
+            valid_shard, args.batch_size, device, 0, collate_fn=collate_native,
+            shuffle=False)
+        test_loader = make_resident_batches(
+            test_shard, args.batch_size, device, 0, collate_fn=collate_native,
+            shuffle=False)
 
     import time as _time
 

@@ -191,11 +191,15 @@ class GraphStepper:
         return total_loss / n, mape_sum / n
 
 
-def make_resident_batches(data_list, batch_size, device, seed, collate_fn):
-    """One seeded shuffle, fixed-composition batches, collated once and
-    moved resident to ``device``."""
-    idx = torch.randperm(len(data_list),
-                         generator=torch.Generator().manual_seed(seed)).tolist()
+def make_resident_batches(data_list, batch_size, device, seed, collate_fn,
+                          shuffle=True):
+    """Fixed-composition batches (one seeded shuffle, or insertion order
+    for eval sets), collated once and moved resident to ``device``."""
+    if shuffle:
+        idx = torch.randperm(len(data_list),
+                             generator=torch.Generator().manual_seed(seed)).tolist()
+    else:
+        idx = list(range(len(data_list)))
     batches = []
     for lo in range(0, len(idx), batch_size):
         samples = [data_list[i] for i in idx[lo:lo + batch_size]]
