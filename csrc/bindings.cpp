@@ -129,10 +129,10 @@ void launch_gemm_fp16_tn(const float*, const float*, float*, float*, int, int,
 void launch_embed_grouped_scatter(const float*, const int*, const int*, float*,
                                   float*, long, int, int, int, int, int,
                                   hipStream_t);
-void launch_embed_grouped_scatter_bal(const float*, const int*, const int*,
+void launch_embed_grouped_scatter_bal(const void*, int, const int*,
                                       const int*, const int*, const int*,
-                                      const int*, float*, float*, float*,
-                                      int, int, int, int, int, int,
+                                      const int*, const int*, float*, float*,
+                                      float*, int, int, int, int, int, int,
                                       hipStream_t);
 void launch_vocab_scatter(const float*, const long*, long, float*, long, int,
                           int, int, int, hipStream_t);
@@ -607,11 +607,13 @@ torch::Tensor embed_grouped_scatter_bal(torch::Tensor g, torch::Tensor order,
               "wave_start must have rows+1 entries");
   const int n_waves = row_map.size(0);
   const int n_waves2 = row_map2.size(0);  // 0 => single-level fold
-  auto dtable = torch::empty({rows, h}, g.options());
-  auto partial = torch::empty({(long)n_waves, h}, g.options());
-  auto partial2 = torch::empty({(long)n_waves2, h}, g.options());
+  const int g16 = g.scalar_type() == torch::kBFloat16 ? 1 : 0;
+  auto fopt = g.options().dtype(torch::kFloat32);
+  auto dtable = torch::empty({rows, h}, fopt);
+  auto partial = torch::empty({(long)n_waves, h}, fopt);
+  auto partial2 = torch::empty({(long)n_waves2, h}, fopt);
   launch_embed_grouped_scatter_bal(
-      g.data_ptr<float>(), order.data_ptr<int>(), ptr.data_ptr<int>(),
+      g.data_ptr(), g16, order.data_ptr<int>(), ptr.data_ptr<int>(),
       row_map.data_ptr<int>(), wave_start.data_ptr<int>(),
       n_waves2 ? row_map2.data_ptr<int>() : nullptr,
       n_waves2 ? wave_start2.data_ptr<int>() : nullptr,

@@ -264,9 +264,9 @@ __global__ void embed_grouped_p2_kernel(const float* __restrict__ partial,
 // size (row_map[w] = row of wave w, wave_start[row] = its first wave), so
 // every wave runs a bounded iteration count regardless of skew.  Reduction
 // order per row is (sub, stride P_row) — fixed by the data, deterministic.
-template <int VPT>
+template <int VPT, typename TG = float>
 __global__ void embed_grouped_bal_p1_kernel(
-    const float* __restrict__ g, const int* __restrict__ order,
+    const TG* __restrict__ g, const int* __restrict__ order,
     const int* __restrict__ ptr, const int* __restrict__ row_map,
     const int* __restrict__ wave_start, float* __restrict__ partial,
     int n_waves, int h, int gstride, int col_off) {
@@ -289,7 +289,7 @@ __global__ void embed_grouped_bal_p1_kernel(
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
       const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) acc[j] += g[r * gstride + col_off + c];
+      if (c < h) acc[j] += (float)g[(long)r * gstride + col_off + c];
     }
   }
 #pragma unroll
@@ -328,15 +328,16 @@ __global__ void embed_grouped_bal_p2_kernel(
 // level 2 over the level-1 partials (needed when one group's partial count
 // itself is large — the interface-0 mega-group), final fold per row.
 void launch_embed_grouped_scatter_bal(
-    const float* g, const int* order, const int* ptr, const int* row_map,
-    const int* wave_start, const int* row_map2, const int* wave_start2,
-    float* partial, float* partial2, float* dtable, int n_waves,
-    int n_waves2, int rows, int h, int gstride, int col_off, hipStream_t s) {
+    const void* g, int g16, const int* order, const int* ptr,
+    const int* row_map, const int* wave_start, const int* row_map2,
+    const int* wave_start2, float* partial, float* partial2, float* dtable,
+    int n_waves, int n_waves2, int rows, int h, int gstride, int col_off,
+    hipStream_t s) {
   if (rows == 0) return;
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
   const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
   switch (vpt) {
-#define CASE(V) case V: embed_grouped_bal_p1_kernel<V><<<dim3(ceil_div(n_waves, WAVES_PER_BLOCK)), block, 0, s>>>(g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); if (n_waves2 > 0) { embed_grouped_bal_p1_kernel<V><<<dim3(ceil_div(n_waves2, WAVES_PER_BLOCK)), block, 0, s>>>(partial, nullptr, wave_start, row_map2, wave_start2, partial2, n_waves2, h, h, 0); embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial2, wave_start2, dtable, rows, h); } else { embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial, wave_start, dtable, rows, h); } break;
+#define CASE(V) case V: { if (g16) { embed_grouped_bal_p1_kernel<V, __bf16><<<dim3(ceil_div(n_waves, WAVES_PER_BLOCK)), block, 0, s>>>((const __bf16*)g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); } else { embed_grouped_bal_p1_kernel<V, float><<<dim3(ceil_div(n_waves, WAVES_PER_BLOCK)), block, 0, s>>>((const float*)g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); } if (n_waves2 > 0) { embed_grouped_bal_p1_kernel<V, float><<<dim3(ceil_div(n_waves2, WAVES_PER_BLOCK)), block, 0, s>>>(partial, nullptr, wave_start, row_map2, wave_start2, partial2, n_waves2, h, h, 0); embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial2, wave_start2, dtable, rows, h); } else { embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial, wave_start, dtable, rows, h); } } break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
     default: pertgnn_shape_fail("segops launcher", "vpt", vpt);

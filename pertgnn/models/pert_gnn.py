@@ -87,7 +87,10 @@ class TransformerConv(nn.Module):
         self.out_channels = out_channels
         self.heads = heads
         self.edge_dim = edge_dim
-        self.k_padded = (in_channels + 7) // 8 * 8
+        # pad K to a multiple of 32 so the layer-1 QKVS GEMM qualifies for
+        # the glds (direct-to-LDS) fast path (k % BK == 0); zero columns
+        # stay zero under Adam, so the math is unchanged
+        self.k_padded = (in_channels + 31) // 32 * 32
         self.w4 = nn.Parameter(torch.zeros(4 * out_channels, self.k_padded))
         self.b4 = nn.Parameter(torch.zeros(4 * out_channels))
         self.lin_query = _SegLinearView(self, 0)

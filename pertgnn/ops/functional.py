@@ -30,17 +30,17 @@ def deterministic() -> bool:
 
 def _table_grad(m, g, idx, rows, h, col_off):
     """dtable[v] = segment-sum of g[:, col_off:col_off+h] by idx — LDS vocab
-    accumulator for small tables, deterministic two-phase grouped scatter
-    otherwise (always, under PERTGNN_DETERMINISTIC=1).  bf16 g is consumed
-    directly by the wave-private kernel when its tables fit; other paths
-    upcast."""
-    if g.dtype == torch.bfloat16 and (
-            deterministic() or h % 64 != 0 or 4 * rows * 64 * 4 > 160 * 1024):
-        g = g.float()
+    accumulator for small tables, work-balanced deterministic grouped
+    scatter otherwise (always, under PERTGNN_DETERMINISTIC=1).  bf16 g is
+    consumed DIRECTLY by both paths (fp32 accumulation inside the kernels —
+    no upcast pass, half the gather traffic)."""
     if rows * h * 4 <= 160 * 1024 and not deterministic():
+        if g.dtype == torch.bfloat16 and (
+                h % 64 != 0 or 4 * rows * 64 * 4 > 160 * 1024):
+            g = g.float()
         return m.vocab_scatter(g, idx, rows, h, col_off)
     order, ptr, row_map, wave_start, row_map2, wave_start2 = _group_by(idx, rows)
-    if g.dtype != torch.float32:
+    if g.dtype not in (torch.float32, torch.bfloat16):
         g = g.float()
     return m.embed_grouped_scatter_bal(g, order, ptr, row_map, wave_start,
                                        row_map2, wave_start2, rows, h, col_off)
@@ -261,9 +261,8 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
                     and (pifc.shape[0] + prpc.shape[0]) * h * 4 <= 160 * 1024):
                 dpifc, dprpc = m.vocab_scatter_dual(de, edge_attr, pifc.shape[0], prpc.shape[0])
             else:
-                de32 = de.float() if de.dtype != torch.float32 else de
-                dpifc = _table_grad(m, de32, edge_attr[:, 0], pifc.shape[0], h, 0)
-                dprpc = _table_grad(m, de32, edge_attr[:, 1], prpc.shape[0], h, 0)
+                dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], h, 0)
+                dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], h, 0)
             return dqkvs, dpifc, dprpc, None, None, None, None, None, None, None
         cur = torch.cuda.current_stream()
         side = _side_stream()
@@ -271,9 +270,8 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
         ev.record(cur)
         side.wait_event(ev)
         with torch.cuda.stream(side):
-            de32 = de.float() if de.dtype != torch.float32 else de
-            dpifc = _table_grad(m, de32, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
-            dprpc = _table_grad(m, de32, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
+            dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
+            dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
         ev2 = torch.cuda.Event()
         ev2.record(side)
         cur.wait_event(ev2)
