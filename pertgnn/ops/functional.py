@@ -51,7 +51,24 @@ _GROUP_CACHE_BYTES = [0]
 _GROUP_CACHE_CAP = 64 * 1024 * 1024  # bytes of cached order/ptr tensors
 
 
-def _group_by(idx: torch.Tensor, rows: int, iters_target: int = 32):
+_SCATTER_ITERS = None
+
+
+def _scatter_iters() -> int:
+    """Gathers per wave in the balanced scatter (PERTGNN_SCATTER_ITERS):
+    smaller = more waves and more partial traffic, larger = longer serial
+    chains per wave."""
+    global _SCATTER_ITERS
+    if _SCATTER_ITERS is None:
+        import os
+        try:
+            _SCATTER_ITERS = max(8, int(os.environ.get("PERTGNN_SCATTER_ITERS", "32")))
+        except ValueError:
+            _SCATTER_ITERS = 32
+    return _SCATTER_ITERS
+
+
+def _group_by(idx: torch.Tensor, rows: int, iters_target: int | None = None):
     """Group positions by index value: returns (order int32, ptr
     int32[rows+1], row_map int32[n_waves], wave_start int32[rows+1]) for the
     work-balanced deterministic grouped scatter.  Cached so the sort +
@@ -70,6 +87,8 @@ def _group_by(idx: torch.Tensor, rows: int, iters_target: int = 32):
     LRU-evicted at a byte budget (the cached view reference pins its base
     storage so the allocator cannot recycle the keyed data_ptr while the
     entry is live)."""
+    if iters_target is None:
+        iters_target = _scatter_iters()
     key = (idx.data_ptr(), idx.numel(), tuple(idx.stride()), rows)
     hit = _GROUP_CACHE.get(key)
     if hit is not None:
