@@ -43,42 +43,46 @@ void launch_gather_rows(const long*, const float*, float*, long, int,
                         hipStream_t);
 void launch_bn_fwd(const float*, const float*, const float*, float*, float*,
                    float*, float*, float*, float*, long, int, float, float,
-                   bool, bool, hipStream_t);
+                   bool, bool, float, const unsigned long long*, hipStream_t);
+void launch_counter_bump(unsigned long long*, hipStream_t);
 void launch_bn_bwd(const float*, const float*, const float*, const float*,
                    const float*, const float*, float*, float*, float*, float*,
-                   long, int, bool, hipStream_t);
+                   long, int, bool, float, hipStream_t);
 void launch_bn_stats_only(const float*, long, int, float*, hipStream_t);
 void launch_bn_finalize_apply(const float*, const float*, const float*,
                               const float*, const float*, float*, float*,
                               float*, float*, float*, long, int, float, float,
-                              bool, bool, hipStream_t);
+                              bool, bool, float, const unsigned long long*,
+                              hipStream_t);
 void launch_bn_bwd_partials_only(const float*, const float*, const float*,
                                  const float*, const float*, long, int, bool,
-                                 float*, hipStream_t);
+                                 float*, float, hipStream_t);
 void launch_bn_bwd_apply_only(const float*, const float*, const float*,
                               const float*, const float*, const float*,
                               const float*, const float*, float*, long, int,
-                              bool, hipStream_t);
+                              bool, float, hipStream_t);
 void launch_bn_grad_affine(const float*, float*, float*, int, hipStream_t);
 // act16 variants (bf16 x/y/g/dx streams, fp32 statistics)
 void launch_bn_stats_only16(const void*, long, int, float*, hipStream_t);
 void launch_bn_fwd16(const void*, const float*, const float*, float*, float*,
                      float*, float*, float*, void*, long, int, float, float,
-                     bool, bool, hipStream_t);
+                     bool, bool, float, const unsigned long long*,
+                     hipStream_t);
 void launch_bn_finalize_apply16(const void*, const float*, const float*,
                                 const float*, const float*, float*, float*,
                                 float*, float*, void*, long, int, float, float,
-                                bool, bool, hipStream_t);
+                                bool, bool, float, const unsigned long long*,
+                                hipStream_t);
 void launch_bn_bwd_partials_only16(const void*, const void*, const void*,
                                    const float*, const float*, long, int, bool,
-                                   float*, hipStream_t);
+                                   float*, float, hipStream_t);
 void launch_bn_bwd_apply_only16(const void*, const void*, const void*,
                                 const float*, const float*, const float*,
                                 const float*, const float*, void*, long, int,
-                                bool, hipStream_t);
+                                bool, float, hipStream_t);
 void launch_bn_bwd16(const void*, const void*, const void*, const float*,
                      const float*, const float*, float*, void*, float*,
-                     float*, long, int, bool, hipStream_t);
+                     float*, long, int, bool, float, hipStream_t);
 void launch_seg_pool_fwd16(const void*, const float*, const float*,
                            const int*, float*, float*, int, int, int,
                            hipStream_t);
@@ -305,12 +309,22 @@ torch::Tensor gather_rows(torch::Tensor idx, torch::Tensor table) {
   return out;
 }
 
+static const unsigned long long* rng_seed_ptr(torch::Tensor& seed,
+                                              double dropout_p) {
+  if (dropout_p <= 0.0 || !seed.defined() || seed.numel() == 0) return nullptr;
+  TORCH_CHECK(seed.scalar_type() == torch::kLong && seed.is_cuda(),
+              "dropout seed must be an int64 CUDA scalar");
+  return (const unsigned long long*)seed.data_ptr<long>();
+}
+
 std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
                                        torch::Tensor beta,
                                        torch::Tensor running_mean,
                                        torch::Tensor running_var,
                                        double momentum, double eps,
-                                       bool training, bool relu) {
+                                       bool training, bool relu,
+                                       double dropout_p = 0.0,
+                                       torch::Tensor seed = {}) {
   CHECK_IN(x); CHECK_IN(gamma); CHECK_IN(beta);
   const long n = x.size(0);
   const int h = x.size(1);
@@ -318,19 +332,22 @@ std::vector<torch::Tensor> bn_relu_fwd(torch::Tensor x, torch::Tensor gamma,
   auto mean = torch::empty({h}, x.options());
   auto invstd = torch::empty({h}, x.options());
   auto partials = torch::empty({2 * h}, x.options());
+  const auto* sp = rng_seed_ptr(seed, dropout_p);
   launch_bn_fwd(x.data_ptr<float>(), gamma.data_ptr<float>(),
                 beta.data_ptr<float>(), running_mean.data_ptr<float>(),
                 running_var.data_ptr<float>(), mean.data_ptr<float>(),
                 invstd.data_ptr<float>(), partials.data_ptr<float>(),
                 y.data_ptr<float>(), n, h, (float)momentum, (float)eps,
-                training, relu, cur_stream());
+                training, relu, (float)dropout_p, sp, cur_stream());
+  if (sp) launch_counter_bump((unsigned long long*)seed.data_ptr<long>(),
+                              cur_stream());
   return {y, mean, invstd};
 }
 
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor g, torch::Tensor x,
                                        torch::Tensor gamma, torch::Tensor mean,
                                        torch::Tensor invstd, torch::Tensor y,
-                                       bool relu) {
+                                       bool relu, double keep_inv = 1.0) {
   CHECK_IN(g); CHECK_IN(x);
   const long n = x.size(0);
   const int h = x.size(1);
@@ -342,7 +359,8 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor g, torch::Tensor x,
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
                 gamma.data_ptr<float>(), partials.data_ptr<float>(),
                 dx.data_ptr<float>(), dgamma.data_ptr<float>(),
-                dbeta.data_ptr<float>(), n, h, relu, cur_stream());
+                dbeta.data_ptr<float>(), n, h, relu, (float)keep_inv,
+                cur_stream());
   return {dx, dgamma, dbeta};
 }
 
@@ -743,7 +761,7 @@ std::vector<torch::Tensor> bn_finalize_apply(
     torch::Tensor x, torch::Tensor partials,
     torch::Tensor gamma, torch::Tensor beta, torch::Tensor running_mean,
     torch::Tensor running_var, double momentum, double eps, bool training,
-    bool relu) {
+    bool relu, double dropout_p = 0.0, torch::Tensor seed = {}) {
   CHECK_IN(x); CHECK_IN(partials);
   const int h = x.size(1);
   TORCH_CHECK(partials.numel() == 2 * h + 1,
@@ -752,6 +770,7 @@ std::vector<torch::Tensor> bn_finalize_apply(
   auto y = torch::empty_like(x);
   auto mean = torch::empty({h}, x.options());
   auto invstd = torch::empty({h}, x.options());
+  const auto* sp = rng_seed_ptr(seed, dropout_p);
   launch_bn_finalize_apply(
       x.data_ptr<float>(), partials.data_ptr<float>(),
       partials.data_ptr<float>() + 2 * h,
@@ -759,20 +778,24 @@ std::vector<torch::Tensor> bn_finalize_apply(
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       mean.data_ptr<float>(), invstd.data_ptr<float>(), y.data_ptr<float>(),
       x.size(0), h, (float)momentum, (float)eps, training, relu,
-      cur_stream());
+      (float)dropout_p, sp, cur_stream());
+  if (sp) launch_counter_bump((unsigned long long*)seed.data_ptr<long>(),
+                              cur_stream());
   return {y, mean, invstd};
 }
 
 torch::Tensor bn_bwd_partials(torch::Tensor g, torch::Tensor x,
                               torch::Tensor y, torch::Tensor mean,
-                              torch::Tensor invstd, bool relu) {
+                              torch::Tensor invstd, bool relu,
+                              double keep_inv = 1.0) {
   CHECK_IN(g); CHECK_IN(x);
   const long h = x.size(1);
   auto partials = torch::empty({2 * h + 1}, x.options());
   launch_bn_bwd_partials_only(g.data_ptr<float>(), x.data_ptr<float>(),
                               y.data_ptr<float>(), mean.data_ptr<float>(),
                               invstd.data_ptr<float>(), x.size(0), h,
-                              relu, partials.data_ptr<float>(), cur_stream());
+                              relu, partials.data_ptr<float>(),
+                              (float)keep_inv, cur_stream());
   partials.narrow(0, 2 * h, 1).fill_((float)x.size(0));
   return partials;
 }
@@ -783,7 +806,7 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor g, torch::Tensor x,
                                         torch::Tensor gamma,
                                         torch::Tensor partials_global,
                                         torch::Tensor partials_local,
-                                        bool relu) {
+                                        bool relu, double keep_inv = 1.0) {
   CHECK_IN(g); CHECK_IN(x);
   const int h = x.size(1);
   TORCH_CHECK(partials_global.numel() == 2 * h + 1,
@@ -798,7 +821,7 @@ std::vector<torch::Tensor> bn_bwd_apply(torch::Tensor g, torch::Tensor x,
                            partials_global.data_ptr<float>(),
                            partials_global.data_ptr<float>() + 2 * h,
                            dx.data_ptr<float>(), x.size(0), h, relu,
-                           cur_stream());
+                           (float)keep_inv, cur_stream());
   launch_bn_grad_affine(partials_local.data_ptr<float>(),
                         dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), h,
                         cur_stream());
@@ -932,7 +955,9 @@ std::vector<torch::Tensor> bn_relu_fwd16(torch::Tensor x, torch::Tensor gamma,
                                          torch::Tensor running_mean,
                                          torch::Tensor running_var,
                                          double momentum, double eps,
-                                         bool training, bool relu) {
+                                         bool training, bool relu,
+                                         double dropout_p = 0.0,
+                                         torch::Tensor seed = {}) {
   CHECK_IN(x); CHECK_IN(gamma); CHECK_IN(beta);
   const long n = x.size(0);
   const int h = x.size(1);
@@ -941,12 +966,15 @@ std::vector<torch::Tensor> bn_relu_fwd16(torch::Tensor x, torch::Tensor gamma,
   auto mean = torch::empty({h}, fopt);
   auto invstd = torch::empty({h}, fopt);
   auto partials = torch::empty({2 * h}, fopt);
+  const auto* sp = rng_seed_ptr(seed, dropout_p);
   launch_bn_fwd16(x.data_ptr(), gamma.data_ptr<float>(),
                   beta.data_ptr<float>(), running_mean.data_ptr<float>(),
                   running_var.data_ptr<float>(), mean.data_ptr<float>(),
                   invstd.data_ptr<float>(), partials.data_ptr<float>(),
                   y.data_ptr(), n, h, (float)momentum, (float)eps, training,
-                  relu, cur_stream());
+                  relu, (float)dropout_p, sp, cur_stream());
+  if (sp) launch_counter_bump((unsigned long long*)seed.data_ptr<long>(),
+                              cur_stream());
   return {y, mean, invstd};
 }
 
@@ -954,7 +982,7 @@ std::vector<torch::Tensor> bn_relu_bwd16(torch::Tensor g, torch::Tensor x,
                                          torch::Tensor gamma,
                                          torch::Tensor mean,
                                          torch::Tensor invstd, torch::Tensor y,
-                                         bool relu) {
+                                         bool relu, double keep_inv = 1.0) {
   CHECK_IN(g); CHECK_IN(x);
   const long n = x.size(0);
   const int h = x.size(1);
@@ -967,7 +995,8 @@ std::vector<torch::Tensor> bn_relu_bwd16(torch::Tensor g, torch::Tensor x,
                   mean.data_ptr<float>(), invstd.data_ptr<float>(),
                   gamma.data_ptr<float>(), partials.data_ptr<float>(),
                   dx.data_ptr(), dgamma.data_ptr<float>(),
-                  dbeta.data_ptr<float>(), n, h, relu, cur_stream());
+                  dbeta.data_ptr<float>(), n, h, relu, (float)keep_inv,
+                  cur_stream());
   return {dx, dgamma, dbeta};
 }
 
@@ -975,7 +1004,7 @@ std::vector<torch::Tensor> bn_finalize_apply16(
     torch::Tensor x, torch::Tensor partials,
     torch::Tensor gamma, torch::Tensor beta, torch::Tensor running_mean,
     torch::Tensor running_var, double momentum, double eps, bool training,
-    bool relu) {
+    bool relu, double dropout_p = 0.0, torch::Tensor seed = {}) {
   CHECK_IN(x); CHECK_IN(partials);
   const int h = x.size(1);
   TORCH_CHECK(partials.numel() == 2 * h + 1,
@@ -985,6 +1014,7 @@ std::vector<torch::Tensor> bn_finalize_apply16(
   auto y = torch::empty({x.size(0), h}, x.options().dtype(torch::kBFloat16));
   auto mean = torch::empty({h}, fopt);
   auto invstd = torch::empty({h}, fopt);
+  const auto* sp = rng_seed_ptr(seed, dropout_p);
   launch_bn_finalize_apply16(
       x.data_ptr(), partials.data_ptr<float>(),
       partials.data_ptr<float>() + 2 * h,
@@ -992,13 +1022,16 @@ std::vector<torch::Tensor> bn_finalize_apply16(
       running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
       mean.data_ptr<float>(), invstd.data_ptr<float>(), y.data_ptr(),
       x.size(0), h, (float)momentum, (float)eps, training, relu,
-      cur_stream());
+      (float)dropout_p, sp, cur_stream());
+  if (sp) launch_counter_bump((unsigned long long*)seed.data_ptr<long>(),
+                              cur_stream());
   return {y, mean, invstd};
 }
 
 torch::Tensor bn_bwd_partials16(torch::Tensor g, torch::Tensor x,
                                 torch::Tensor y, torch::Tensor mean,
-                                torch::Tensor invstd, bool relu) {
+                                torch::Tensor invstd, bool relu,
+                                double keep_inv = 1.0) {
   CHECK_IN(g); CHECK_IN(x);
   const long h = x.size(1);
   auto partials = torch::empty({2 * h + 1},
@@ -1007,7 +1040,7 @@ torch::Tensor bn_bwd_partials16(torch::Tensor g, torch::Tensor x,
                                 y.data_ptr(), mean.data_ptr<float>(),
                                 invstd.data_ptr<float>(), x.size(0), h,
                                 relu, partials.data_ptr<float>(),
-                                cur_stream());
+                                (float)keep_inv, cur_stream());
   partials.narrow(0, 2 * h, 1).fill_((float)x.size(0));
   return partials;
 }
@@ -1015,7 +1048,7 @@ torch::Tensor bn_bwd_partials16(torch::Tensor g, torch::Tensor x,
 std::vector<torch::Tensor> bn_bwd_apply16(
     torch::Tensor g, torch::Tensor x, torch::Tensor y, torch::Tensor mean,
     torch::Tensor invstd, torch::Tensor gamma, torch::Tensor partials_global,
-    torch::Tensor partials_local, bool relu) {
+    torch::Tensor partials_local, bool relu, double keep_inv = 1.0) {
   CHECK_IN(g); CHECK_IN(x);
   const int h = x.size(1);
   TORCH_CHECK(partials_global.numel() == 2 * h + 1,
@@ -1031,7 +1064,7 @@ std::vector<torch::Tensor> bn_bwd_apply16(
                              partials_global.data_ptr<float>(),
                              partials_global.data_ptr<float>() + 2 * h,
                              dx.data_ptr(), x.size(0), h, relu,
-                             cur_stream());
+                             (float)keep_inv, cur_stream());
   launch_bn_grad_affine(partials_local.data_ptr<float>(),
                         dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), h,
                         cur_stream());
@@ -1182,9 +1215,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("linear_wgrad", &linear_wgrad);
   mod.def("embed_grouped_scatter_bal", &embed_grouped_scatter_bal);
   mod.def("bn_stats", &bn_stats);
-  mod.def("bn_finalize_apply", &bn_finalize_apply);
-  mod.def("bn_bwd_partials", &bn_bwd_partials);
-  mod.def("bn_bwd_apply", &bn_bwd_apply);
+  mod.def("bn_finalize_apply", &bn_finalize_apply, py::arg("x"),
+          py::arg("partials"), py::arg("gamma"), py::arg("beta"),
+          py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+          py::arg("eps"), py::arg("training"), py::arg("relu"),
+          py::arg("dropout_p") = 0.0, py::arg("seed") = torch::Tensor());
+  mod.def("bn_bwd_partials", &bn_bwd_partials, py::arg("g"), py::arg("x"),
+          py::arg("y"), py::arg("mean"), py::arg("invstd"), py::arg("relu"),
+          py::arg("keep_inv") = 1.0);
+  mod.def("bn_bwd_apply", &bn_bwd_apply, py::arg("g"), py::arg("x"),
+          py::arg("y"), py::arg("mean"), py::arg("invstd"), py::arg("gamma"),
+          py::arg("partials_global"), py::arg("partials_local"),
+          py::arg("relu"), py::arg("keep_inv") = 1.0);
   mod.def("vocab_scatter", &vocab_scatter);
   mod.def("collate_native", &collate_native,
           py::call_guard<py::gil_scoped_release>());
@@ -1210,11 +1252,26 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("embed_node_fwd", &embed_node_fwd, py::arg("x_raw"), py::arg("idx"), py::arg("table"), py::arg("out16") = false);
   mod.def("embed_edge_fwd", &embed_edge_fwd);
   mod.def("gather_rows", &gather_rows);
-  mod.def("bn_relu_fwd16", &bn_relu_fwd16);
-  mod.def("bn_relu_bwd16", &bn_relu_bwd16);
-  mod.def("bn_finalize_apply16", &bn_finalize_apply16);
-  mod.def("bn_bwd_partials16", &bn_bwd_partials16);
-  mod.def("bn_bwd_apply16", &bn_bwd_apply16);
+  mod.def("bn_relu_fwd16", &bn_relu_fwd16, py::arg("x"), py::arg("gamma"),
+          py::arg("beta"), py::arg("running_mean"), py::arg("running_var"),
+          py::arg("momentum"), py::arg("eps"), py::arg("training"),
+          py::arg("relu"), py::arg("dropout_p") = 0.0,
+          py::arg("seed") = torch::Tensor());
+  mod.def("bn_relu_bwd16", &bn_relu_bwd16, py::arg("g"), py::arg("x"),
+          py::arg("gamma"), py::arg("mean"), py::arg("invstd"), py::arg("y"),
+          py::arg("relu"), py::arg("keep_inv") = 1.0);
+  mod.def("bn_finalize_apply16", &bn_finalize_apply16, py::arg("x"),
+          py::arg("partials"), py::arg("gamma"), py::arg("beta"),
+          py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+          py::arg("eps"), py::arg("training"), py::arg("relu"),
+          py::arg("dropout_p") = 0.0, py::arg("seed") = torch::Tensor());
+  mod.def("bn_bwd_partials16", &bn_bwd_partials16, py::arg("g"), py::arg("x"),
+          py::arg("y"), py::arg("mean"), py::arg("invstd"), py::arg("relu"),
+          py::arg("keep_inv") = 1.0);
+  mod.def("bn_bwd_apply16", &bn_bwd_apply16, py::arg("g"), py::arg("x"),
+          py::arg("y"), py::arg("mean"), py::arg("invstd"), py::arg("gamma"),
+          py::arg("partials_global"), py::arg("partials_local"),
+          py::arg("relu"), py::arg("keep_inv") = 1.0);
   mod.def("seg_pool_fwd16", &seg_pool_fwd16);
   mod.def("seg_pool_bwd16", &seg_pool_bwd16);
   mod.def("linear_fwd_a16o16", &linear_fwd_a16o16, py::arg("x"), py::arg("w"),
@@ -1223,8 +1280,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("w"), py::arg("fp16c") = false);
   mod.def("linear_wgrad16_b16", &linear_wgrad16_b16, py::arg("g"),
           py::arg("x"), py::arg("has_bias"), py::arg("fp16c") = false);
-  mod.def("bn_relu_fwd", &bn_relu_fwd);
-  mod.def("bn_relu_bwd", &bn_relu_bwd);
+  mod.def("bn_relu_fwd", &bn_relu_fwd, py::arg("x"), py::arg("gamma"),
+          py::arg("beta"), py::arg("running_mean"), py::arg("running_var"),
+          py::arg("momentum"), py::arg("eps"), py::arg("training"),
+          py::arg("relu"), py::arg("dropout_p") = 0.0,
+          py::arg("seed") = torch::Tensor());
+  mod.def("bn_relu_bwd", &bn_relu_bwd, py::arg("g"), py::arg("x"),
+          py::arg("gamma"), py::arg("mean"), py::arg("invstd"), py::arg("y"),
+          py::arg("relu"), py::arg("keep_inv") = 1.0);
   mod.def("quantile_loss_fwd", &quantile_loss_fwd);
   mod.def("quantile_loss_bwd", &quantile_loss_bwd);
   mod.def("eval_metrics", &eval_metrics);

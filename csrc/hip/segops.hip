@@ -994,7 +994,23 @@ __global__ void bn_finalize_kernel(const float* __restrict__ partials, long n,
   }
 }
 
-// stage C: normalize + affine (+ReLU)
+// counter-based per-element uniform for fused dropout (splitmix64 mix of
+// the device-resident step counter and the element index; quality is ample
+// for dropout masks and the sequence is deterministic given the counter)
+__device__ __forceinline__ float bn_rand01(unsigned long long seed, long t) {
+  unsigned long long z = seed * 0x9E3779B97F4A7C15ull + (unsigned long long)t;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  z ^= z >> 31;
+  return (float)(z >> 40) * (1.f / 16777216.f);  // top 24 bits
+}
+
+// stage C: normalize + affine (+ReLU) (+fused dropout: reference K8 — the
+// model applies dropout right after BN+ReLU, model.py:101-103; keep_inv =
+// 1/(1-p), seed_ptr = device step counter so captured hipGraphs draw FRESH
+// masks every replay).  Post-dropout y==0 <=> dropped-or-relu-negative, so
+// the backward needs NO RNG: the existing y<=0 mask plus the keep_inv
+// scale reproduces d(dropout.relu)/dy exactly.
 // TY = output/activation dtype: fp32 in exact mode, bf16 in the act16 mode
 // (standard mixed-precision BN: statistics and normalization math stay fp32,
 // only the activation stream is 16-bit).
@@ -1005,8 +1021,12 @@ __global__ void bn_apply_kernel(const TX* __restrict__ x,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
                                 TY* __restrict__ y, long n, int h,
-                                int relu) {
+                                int relu, float dropout_p = 0.f,
+                                const unsigned long long* __restrict__
+                                    seed_ptr = nullptr) {
   typedef __attribute__((ext_vector_type(4))) float bnf4;
+  const float keep_inv = (dropout_p > 0.f) ? 1.f / (1.f - dropout_p) : 1.f;
+  const unsigned long long seed = seed_ptr ? *seed_ptr : 0ull;
   if ((h & 3) == 0) {  // 4-wide: one f32x4 load + packed store per thread
     const long numq = n * (h / 4);
     const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1023,6 +1043,8 @@ __global__ void bn_apply_kernel(const TX* __restrict__ x,
         float v = (xv[u] - mean[c + u]) * invstd[c + u] * gamma[c + u] +
                   beta[c + u];
         if (relu) v = fmaxf(v, 0.f);
+        if (dropout_p > 0.f)
+          v = (bn_rand01(seed, t + u) >= dropout_p) ? v * keep_inv : 0.f;
         o.v[u] = (TY)v;
       }
       *reinterpret_cast<TY4*>(&y[t]) = o;  // one packed 8B/16B store
@@ -1036,8 +1058,14 @@ __global__ void bn_apply_kernel(const TX* __restrict__ x,
     const int c = (int)(t % h);
     float v = ((float)x[t] - mean[c]) * invstd[c] * gamma[c] + beta[c];
     if (relu) v = fmaxf(v, 0.f);
+    if (dropout_p > 0.f)
+      v = (bn_rand01(seed, t) >= dropout_p) ? v * keep_inv : 0.f;
     y[t] = (TY)v;
   }
+}
+
+__global__ void counter_bump_kernel(unsigned long long* __restrict__ c) {
+  ++*c;
 }
 
 // backward stage A: per-channel sums of gm and gm*xhat (gm = relu-masked g)
@@ -1048,7 +1076,8 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
                                       const float* __restrict__ mean,
                                       const float* __restrict__ invstd, long n,
                                       int h, int relu,
-                                      float* __restrict__ partials) {
+                                      float* __restrict__ partials,
+                                      float keep_inv = 1.f) {
   const int c0 = threadIdx.x;
   const int c1 = threadIdx.x + 256;
   float s0 = 0.f, q0 = 0.f, s1 = 0.f, q1 = 0.f;
@@ -1071,7 +1100,7 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
       }
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        float gm = (relu && yv[u] <= 0.f) ? 0.f : gv[u];
+        float gm = (relu && yv[u] <= 0.f) ? 0.f : gv[u] * keep_inv;
         s0 += gm;
         q0 += gm * (xv[u] - m0) * i0;
       }
@@ -1086,7 +1115,7 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
       }
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        float gm = (relu && yv[u] <= 0.f) ? 0.f : gv[u];
+        float gm = (relu && yv[u] <= 0.f) ? 0.f : gv[u] * keep_inv;
         s1 += gm;
         q1 += gm * (xv[u] - m1) * i1;
       }
@@ -1094,13 +1123,13 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
   }
   for (; r < r1; ++r) {
     if (c0 < h) {
-      float gm = (float)g[r * h + c0];
+      float gm = (float)g[r * h + c0] * keep_inv;
       if (relu && (float)y[r * h + c0] <= 0.f) gm = 0.f;
       s0 += gm;
       q0 += gm * ((float)x[r * h + c0] - m0) * i0;
     }
     if (c1 < h) {
-      float gm = (float)g[r * h + c1];
+      float gm = (float)g[r * h + c1] * keep_inv;
       if (relu && (float)y[r * h + c1] <= 0.f) gm = 0.f;
       s1 += gm;
       q1 += gm * ((float)x[r * h + c1] - m1) * i1;
@@ -1127,18 +1156,18 @@ template <typename TG, typename TY, typename TX>
 static void bn_bwd_partials_dispatch(const TG* g, const TX* x, const TY* y,
                                      const float* mean, const float* invstd,
                                      long n, int h, bool relu, float* partials,
-                                     hipStream_t s) {
+                                     float keep_inv, hipStream_t s) {
   const int nblocks = (int)min((long)512, (n + 63) / 64);
   if (pertgnn_deterministic_seg() && h <= 1024) {
     float* slab = bn_det_slab();
     bn_bwd_partial_kernel<TG, TY, TX, true><<<nblocks, 256, 0, s>>>(
-        g, x, y, mean, invstd, n, h, relu ? 1 : 0, slab);
+        g, x, y, mean, invstd, n, h, relu ? 1 : 0, slab, keep_inv);
     bn_slab_reduce_kernel<<<ceil_div(2 * h, 256), 256, 0, s>>>(
         slab, nblocks, 2 * h, partials);
   } else {
     HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
     bn_bwd_partial_kernel<TG, TY, TX, false><<<nblocks, 256, 0, s>>>(
-        g, x, y, mean, invstd, n, h, relu ? 1 : 0, partials);
+        g, x, y, mean, invstd, n, h, relu ? 1 : 0, partials, keep_inv);
   }
 }
 
@@ -1152,7 +1181,7 @@ __global__ void bn_bwd_apply_kernel(
     const float* __restrict__ invstd, const float* __restrict__ gamma,
     const float* __restrict__ partials, TX* __restrict__ dx, long n,
     long count, int h, int relu,
-    const float* __restrict__ count_ptr = nullptr) {
+    const float* __restrict__ count_ptr = nullptr, float keep_inv = 1.f) {
   typedef __attribute__((ext_vector_type(4))) float bnf4;
   const float invn = 1.f / (count_ptr ? *count_ptr : (float)count);
   if ((h & 3) == 0) {
@@ -1173,7 +1202,7 @@ __global__ void bn_bwd_apply_kernel(
       TX4 o;
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
-        float gm = gv[u];
+        float gm = gv[u] * keep_inv;
         if (relu && yv[u] <= 0.f) gm = 0.f;
         const float xhat = (xv[u] - mean[c + u]) * invstd[c + u];
         o.v[u] = (TX)(gamma[c + u] * invstd[c + u] *
@@ -1189,7 +1218,7 @@ __global__ void bn_bwd_apply_kernel(
   const long stride = (long)gridDim.x * blockDim.x;
   for (long t = i0; t < numel; t += stride) {
     const int c = (int)(t % h);
-    float gm = (float)g[t];
+    float gm = (float)g[t] * keep_inv;
     if (relu && (float)y[t] <= 0.f) gm = 0.f;
     const float xhat = ((float)x[t] - mean[c]) * invstd[c];
     dx[t] = (TX)(gamma[c] * invstd[c] *
@@ -1244,6 +1273,8 @@ void launch_bn_finalize_apply(const float* x, const float* partials,
                               float* running_var, float* mean, float* invstd,
                               float* y, long n, int h, float momentum,
                               float eps, bool training, bool relu,
+                              float dropout_p,
+                              const unsigned long long* seed_ptr,
                               hipStream_t s) {
   if (training) {
     bn_finalize_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
@@ -1256,29 +1287,33 @@ void launch_bn_finalize_apply(const float* x, const float* partials,
   if (n > 0)
     bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma,
                                                     beta, y, n, h,
-                                                    relu ? 1 : 0);
+                                                    relu ? 1 : 0,
+                                                    training ? dropout_p : 0.f,
+                                                    seed_ptr);
 }
 
 void launch_bn_bwd_partials_only(const float* g, const float* x,
                                  const float* y, const float* mean,
                                  const float* invstd, long n, int h, bool relu,
-                                 float* partials, hipStream_t s) {
+                                 float* partials, float keep_inv,
+                                 hipStream_t s) {
   if (n == 0) {
     HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
     return;
   }
-  bn_bwd_partials_dispatch(g, x, y, mean, invstd, n, h, relu, partials, s);
+  bn_bwd_partials_dispatch(g, x, y, mean, invstd, n, h, relu, partials,
+                           keep_inv, s);
 }
 
 void launch_bn_bwd_apply_only(const float* g, const float* x, const float* y,
                               const float* mean, const float* invstd,
                               const float* gamma, const float* partials,
                               const float* count_ptr, float* dx, long n, int h,
-                              bool relu, hipStream_t s) {
+                              bool relu, float keep_inv, hipStream_t s) {
   if (n == 0) return;
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
       g, x, y, mean, invstd, gamma, partials, dx, n, n, h, relu ? 1 : 0,
-      count_ptr);
+      count_ptr, keep_inv);
 }
 
 void launch_bn_grad_affine(const float* partials, float* dgamma, float* dbeta,
@@ -1291,6 +1326,7 @@ void launch_bn_fwd(const float* x, const float* gamma, const float* beta,
                    float* running_mean, float* running_var, float* mean,
                    float* invstd, float* partials, float* y, long n, int h,
                    float momentum, float eps, bool training, bool relu,
+                   float dropout_p, const unsigned long long* seed_ptr,
                    hipStream_t s) {
   if (n == 0) return;
   if (training) {
@@ -1302,17 +1338,25 @@ void launch_bn_fwd(const float* x, const float* gamma, const float* beta,
     launch_bn_eval_stats(running_mean, running_var, mean, invstd, h, eps, s);
   }
   bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma, beta,
-                                                  y, n, h, relu ? 1 : 0);
+                                                  y, n, h, relu ? 1 : 0,
+                                                  training ? dropout_p : 0.f,
+                                                  seed_ptr);
+}
+
+void launch_counter_bump(unsigned long long* c, hipStream_t s) {
+  counter_bump_kernel<<<1, 1, 0, s>>>(c);
 }
 
 void launch_bn_bwd(const float* g, const float* x, const float* y,
                    const float* mean, const float* invstd, const float* gamma,
                    float* partials, float* dx, float* dgamma, float* dbeta,
-                   long n, int h, bool relu, hipStream_t s) {
+                   long n, int h, bool relu, float keep_inv, hipStream_t s) {
   if (n == 0) return;
-  bn_bwd_partials_dispatch(g, x, y, mean, invstd, n, h, relu, partials, s);
+  bn_bwd_partials_dispatch(g, x, y, mean, invstd, n, h, relu, partials,
+                           keep_inv, s);
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
-      g, x, y, mean, invstd, gamma, partials, dx, n, n, h, relu ? 1 : 0);
+      g, x, y, mean, invstd, gamma, partials, dx, n, n, h, relu ? 1 : 0,
+      nullptr, keep_inv);
   bn_grad_affine_kernel<<<ceil_div(h, 256), 256, 0, s>>>(partials, dgamma,
                                                          dbeta, h);
 }
@@ -1334,6 +1378,7 @@ void launch_bn_fwd16(const void* x, const float* gamma, const float* beta,
                      float* running_mean, float* running_var, float* mean,
                      float* invstd, float* partials, void* y, long n, int h,
                      float momentum, float eps, bool training, bool relu,
+                     float dropout_p, const unsigned long long* seed_ptr,
                      hipStream_t s) {
   if (n == 0) return;
   if (training) {
@@ -1347,7 +1392,9 @@ void launch_bn_fwd16(const void* x, const float* gamma, const float* beta,
   bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>((const __bf16*)x, mean,
                                                   invstd, gamma, beta,
                                                   (__bf16*)y, n, h,
-                                                  relu ? 1 : 0);
+                                                  relu ? 1 : 0,
+                                                  training ? dropout_p : 0.f,
+                                                  seed_ptr);
 }
 
 void launch_bn_finalize_apply16(const void* x, const float* partials,
@@ -1356,6 +1403,8 @@ void launch_bn_finalize_apply16(const void* x, const float* partials,
                                 float* running_var, float* mean, float* invstd,
                                 void* y, long n, int h, float momentum,
                                 float eps, bool training, bool relu,
+                                float dropout_p,
+                                const unsigned long long* seed_ptr,
                                 hipStream_t s) {
   if (training) {
     bn_finalize_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
@@ -1369,45 +1418,51 @@ void launch_bn_finalize_apply16(const void* x, const float* partials,
     bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>((const __bf16*)x, mean,
                                                     invstd, gamma, beta,
                                                     (__bf16*)y, n, h,
-                                                    relu ? 1 : 0);
+                                                    relu ? 1 : 0,
+                                                    training ? dropout_p : 0.f,
+                                                    seed_ptr);
 }
 
 void launch_bn_bwd_partials_only16(const void* g, const void* x,
                                    const void* y, const float* mean,
                                    const float* invstd, long n, int h,
-                                   bool relu, float* partials, hipStream_t s) {
+                                   bool relu, float* partials, float keep_inv,
+                                   hipStream_t s) {
   if (n == 0) {
     HIP_CHECK(hipMemsetAsync(partials, 0, 2 * h * sizeof(float), s));
     return;
   }
   bn_bwd_partials_dispatch((const __bf16*)g, (const __bf16*)x,
                            (const __bf16*)y, mean, invstd, n, h, relu,
-                           partials, s);
+                           partials, keep_inv, s);
 }
 
 void launch_bn_bwd_apply_only16(const void* g, const void* x, const void* y,
                                 const float* mean, const float* invstd,
                                 const float* gamma, const float* partials,
                                 const float* count_ptr, void* dx, long n,
-                                int h, bool relu, hipStream_t s) {
+                                int h, bool relu, float keep_inv,
+                                hipStream_t s) {
   if (n == 0) return;
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
       (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd,
-      gamma, partials, (__bf16*)dx, n, n, h, relu ? 1 : 0, count_ptr);
+      gamma, partials, (__bf16*)dx, n, n, h, relu ? 1 : 0, count_ptr,
+      keep_inv);
 }
 
 void launch_bn_bwd16(const void* g, const void* x, const void* y,
                      const float* mean, const float* invstd,
                      const float* gamma, float* partials, void* dx,
                      float* dgamma, float* dbeta, long n, int h, bool relu,
-                     hipStream_t s) {
+                     float keep_inv, hipStream_t s) {
   if (n == 0) return;
   bn_bwd_partials_dispatch((const __bf16*)g, (const __bf16*)x,
                            (const __bf16*)y, mean, invstd, n, h, relu,
-                           partials, s);
+                           partials, keep_inv, s);
   bn_bwd_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(
       (const __bf16*)g, (const __bf16*)x, (const __bf16*)y, mean, invstd,
-      gamma, partials, (__bf16*)dx, n, n, h, relu ? 1 : 0);
+      gamma, partials, (__bf16*)dx, n, n, h, relu ? 1 : 0, nullptr,
+      keep_inv);
   bn_grad_affine_kernel<<<ceil_div(h, 256), 256, 0, s>>>(partials, dgamma,
                                                          dbeta, h);
 }

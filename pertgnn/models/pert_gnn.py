@@ -331,14 +331,15 @@ class SAGEDeterministic(nn.Module):
         for i, conv in enumerate(self.convs[:-1]):
             x = run_conv(conv, x, out16=out16)
             bn = self.bns[i]
+            # dropout fused into the BN epilogue (K8); the CPU oracle path
+            # applies torch dropout inside batchnorm_relu
             x = ops.batchnorm_relu(
                 x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
                 bn.momentum, bn.eps, self.training, fuse_relu=True,
-                comm=self._bn_comm, out16=out16,
+                comm=self._bn_comm, out16=out16, dropout_p=self.dropout,
             )
             if self.training and bn.track_running_stats and bn.num_batches_tracked is not None:
                 bn.num_batches_tracked += 1
-            x = F.dropout(x, p=self.dropout, training=self.training)
         x = run_conv(self.convs[-1], x)
         local_predict = ops.linear(x, self.local_linear.weight, self.local_linear.bias)
         mean_x = ops.pattern_pool(x, pattern_probs, pattern_num_nodes, batch, num_graphs)

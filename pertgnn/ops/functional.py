@@ -62,9 +62,9 @@ def _scatter_iters() -> int:
     if _SCATTER_ITERS is None:
         import os
         try:
-            _SCATTER_ITERS = max(8, int(os.environ.get("PERTGNN_SCATTER_ITERS", "32")))
+            _SCATTER_ITERS = max(8, int(os.environ.get("PERTGNN_SCATTER_ITERS", "64")))
         except ValueError:
-            _SCATTER_ITERS = 32
+            _SCATTER_ITERS = 64
     return _SCATTER_ITERS
 
 
@@ -439,11 +439,33 @@ def embedding(idx, table):
 # batchnorm + relu (K7+K8)
 # ---------------------------------------------------------------------------
 
+_RNG_COUNTER = {}
+
+
+def _rng_counter(device):
+    """Device-resident dropout step counter: the fused-dropout kernels read
+    it and a 1-thread kernel bumps it after each use, so captured hipGraphs
+    draw FRESH masks on every replay.  Seeded from torch's global RNG (so
+    torch.manual_seed reproduces mask sequences)."""
+    key = str(device)
+    t = _RNG_COUNTER.get(key)
+    if t is None:
+        t = torch.randint(0, 2 ** 62, (1,), dtype=torch.int64, device=device)
+        _RNG_COUNTER[key] = t
+    return t
+
+
 class _BNReLUFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu, comm, out16=False):
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps,
+                training, fuse_relu, comm, out16=False, dropout_p=0.0):
         m = ext()
         sync = comm is not None and getattr(comm, "distributed", False) and training
+        drop = float(dropout_p) if training else 0.0
+        if drop > 0.0:
+            assert fuse_relu, "fused dropout rides the ReLU mask (model.py:102-103)"
+        seed = _rng_counter(x.device) if drop > 0.0 else torch.empty(
+            0, dtype=torch.int64, device=x.device)
         if sync:
             # sync-BN (SURVEY.md §7 hard part 4, exact-parity mode): ONE
             # all-reduce carries the partial sums AND the row count (tail
@@ -456,16 +478,20 @@ class _BNReLUFn(torch.autograd.Function):
                    else m.bn_finalize_apply)
             y, save_mean, save_invstd = fin(
                 x, partials, gamma, beta, running_mean, running_var,
-                momentum, eps, training, fuse_relu)
+                momentum, eps, training, fuse_relu, drop, seed)
         else:
             fwd = (m.bn_relu_fwd16 if x.dtype == torch.bfloat16
                    else m.bn_relu_fwd)
             y, save_mean, save_invstd = fwd(
-                x, gamma, beta, running_mean, running_var, momentum, eps, training, fuse_relu
+                x, gamma, beta, running_mean, running_var, momentum, eps,
+                training, fuse_relu, drop, seed
             )
         ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
         ctx.fuse_relu = fuse_relu
         ctx.comm = comm if sync else None
+        # post-dropout y==0 <=> dropped-or-relu-negative, so the backward
+        # only needs the keep scale on the existing y<=0 mask (no RNG)
+        ctx.keep_inv = 1.0 / (1.0 - drop) if drop > 0.0 else 1.0
         return y
 
     @staticmethod
@@ -479,18 +505,21 @@ class _BNReLUFn(torch.autograd.Function):
             bwd_apply = m.bn_bwd_apply16 if y16 else m.bn_bwd_apply
             # [2h+1] partials: sums + local count; one all-reduce gives the
             # global sums and count together (device-side, no .item()).
-            partials_local = bwd_partials(g, x, y, save_mean, save_invstd, ctx.fuse_relu)
+            partials_local = bwd_partials(g, x, y, save_mean, save_invstd,
+                                          ctx.fuse_relu, ctx.keep_inv)
             partials_global = partials_local.clone()
             ctx.comm.all_reduce_(partials_global)
             dx, dgamma, dbeta = bwd_apply(
                 g, x, y, save_mean, save_invstd, gamma,
-                partials_global, partials_local, ctx.fuse_relu)
+                partials_global, partials_local, ctx.fuse_relu, ctx.keep_inv)
         else:
             bwd = m.bn_relu_bwd16 if y16 else m.bn_relu_bwd
             dx, dgamma, dbeta = bwd(
-                g, x, gamma, save_mean, save_invstd, y, ctx.fuse_relu
+                g, x, gamma, save_mean, save_invstd, y, ctx.fuse_relu,
+                ctx.keep_inv
             )
-        return dx, dgamma, dbeta, None, None, None, None, None, None, None, None
+        return (dx, dgamma, dbeta, None, None, None, None, None, None, None,
+                None, None)
 
 
 class _EagerSyncBNFn(torch.autograd.Function):
@@ -549,23 +578,33 @@ def _eager_sync_batchnorm(x, gamma, beta, running_mean, running_var, momentum,
 
 
 def batchnorm_relu(x, gamma, beta, running_mean, running_var, momentum, eps,
-                   training, fuse_relu=True, comm=None, out16=False):
-    """BatchNorm1d over N per channel, optional fused ReLU (model.py:101-102).
-    With ``comm`` (distributed) and training=True, runs sync-BN: statistics
-    over the GLOBAL batch (exact single-process parity).  ``out16`` emits the
-    normalized activations as bf16 (act16 mode): statistics/affine math stays
-    fp32, only the activation stream narrows — the downstream QKVS GEMM and
-    its wgrad then read half the bytes."""
+                   training, fuse_relu=True, comm=None, out16=False,
+                   dropout_p=0.0):
+    """BatchNorm1d over N per channel, optional fused ReLU (model.py:101-102)
+    and fused dropout (reference K8: the model applies dropout right after
+    BN+ReLU, model.py:103; the HIP path folds mask+scale into the BN
+    epilogue, the CPU path applies torch dropout after).  With ``comm``
+    (distributed) and training=True, runs sync-BN: statistics over the
+    GLOBAL batch (exact single-process parity).  ``out16`` emits the
+    normalized activations as bf16 (act16 mode): statistics/affine math
+    stays fp32, only the activation stream narrows — the downstream QKVS
+    GEMM and its wgrad then read half the bytes."""
     if use_hip(x):
         return _BNReLUFn.apply(x, gamma, beta, running_mean, running_var,
-                               momentum, eps, training, fuse_relu, comm, out16)
+                               momentum, eps, training, fuse_relu, comm,
+                               out16, dropout_p)
     if comm is not None and getattr(comm, "distributed", False) and training:
-        return _eager_sync_batchnorm(x, gamma, beta, running_mean, running_var,
-                                     momentum, eps, fuse_relu, comm)
-    y = torch.nn.functional.batch_norm(
-        x, running_mean, running_var, gamma, beta, training, momentum, eps
-    )
-    return torch.nn.functional.relu(y) if fuse_relu else y
+        y = _eager_sync_batchnorm(x, gamma, beta, running_mean, running_var,
+                                  momentum, eps, fuse_relu, comm)
+    else:
+        y = torch.nn.functional.batch_norm(
+            x, running_mean, running_var, gamma, beta, training, momentum, eps
+        )
+        if fuse_relu:
+            y = torch.nn.functional.relu(y)
+    if dropout_p > 0.0:
+        y = torch.nn.functional.dropout(y, p=dropout_p, training=training)
+    return y
 
 
 # ---------------------------------------------------------------------------

@@ -672,3 +672,51 @@ def test_dynamic_loss_scaling_gpu():
         g.replay()
     torch.cuda.synchronize()
     assert int(opt.dev_state[0].item()) == t0 + 3
+
+
+def test_fused_dropout_semantics():
+    """BN-epilogue fused dropout (K8): kept fraction ~= 1-p, kept values are
+    exactly scaled BN+ReLU outputs, dropped are 0, the backward applies
+    keep/(1-p) through the y-mask, eval applies no dropout, and consecutive
+    calls draw different masks (device counter advances)."""
+    require_ext()
+    torch.manual_seed(123)
+    n, h, p = 4096, 256, 0.3
+    x = (torch.randn(n, h) * 2 + 0.5).to(DEV)
+    gamma = (torch.rand(h) + 0.5).to(DEV)
+    beta = torch.randn(h).to(DEV)
+    rm = torch.zeros(h, device=DEV)
+    rv = torch.ones(h, device=DEV)
+
+    def run(drop):
+        xx = x.detach().requires_grad_(True)
+        y = F.batchnorm_relu(xx, gamma, beta, rm.clone(), rv.clone(), 0.1,
+                             1e-5, True, fuse_relu=True, dropout_p=drop)
+        g = torch.ones_like(y)
+        y.backward(g)
+        return y.detach(), xx.grad.detach()
+
+    y0, _ = run(0.0)
+    y1, g1 = run(p)
+    kept = y1 != 0
+    frac = kept.float().mean().item()
+    assert abs(frac - (y0 != 0).float().mean().item() * (1 - p)) < 0.02
+    # kept values are the p=0 outputs scaled by 1/(1-p) (same statistics)
+    assert torch.allclose(y1[kept], y0[kept] / (1 - p), atol=1e-4, rtol=1e-4)
+    # different mask next call (counter bumped)
+    y2, _ = run(p)
+    assert not torch.equal(y1 == 0, y2 == 0)
+    # backward: dropped positions contribute nothing through the local term;
+    # compare against an eager BN with the SAME mask applied
+    xx = x.detach().requires_grad_(True)
+    ybn = torch.nn.functional.batch_norm(xx, None, None, gamma, beta, True,
+                                         0.1, 1e-5)
+    yref = torch.nn.functional.relu(ybn)
+    mask = kept.float() / (1 - p)
+    (yref * mask).backward(torch.ones_like(yref))
+    assert torch.allclose(g1, xx.grad, atol=2e-3, rtol=1e-3), \
+        (g1 - xx.grad).abs().max()
+    # eval mode: no dropout
+    ye = F.batchnorm_relu(x, gamma, beta, rm.clone(), rv.clone(), 0.1, 1e-5,
+                          False, fuse_relu=True, dropout_p=p)
+    assert (ye != 0).float().mean().item() > 0.9 * (y0 != 0).float().mean().item()
