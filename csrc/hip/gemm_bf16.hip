@@ -1074,184 +1074,6 @@ void launch_transpose_convert_w16(const float* w, void* o, int rows, int cols,
 // m-tail strip through the register-staging kernel (pointer offset; it
 // stages the ORIGINAL fp32 weights, rounding to the same bf16 values the
 // convert kernel produced).
-// ---------------------------------------------------------------------------
-// Deep-pipelined glds NT variant: 3-stage LDS ring with COUNTED s_waitcnt
-// vmcnt(N) and raw s_barriers instead of the vmcnt(0)-at-__syncthreads of
-// the 2-buffer kernel (the guide's documented ~20% stall: the workgroup
-// release drains the glds queue every K-step).  Prefetch keeps 2 stages in
-// flight across barriers; each wave waits only until ITS loads for the
-// stage it is about to consume have landed, then barriers so all waves
-// agree.  PERTGNN_GLDS_DEEP=1 selects it (A/B'd against the 2-buffer
-// default before any flip).
-// ---------------------------------------------------------------------------
-
-template <int N>
-__device__ __forceinline__ void glds_waitcnt_vm() {
-  asm volatile("s_waitcnt vmcnt(%0)" ::"i"(N) : "memory");
-}
-
-__device__ __forceinline__ void glds_wait_upto(int allowed) {
-  // tail path: allowed is 0.. small; map to immediates
-  switch (allowed) {
-    case 0: glds_waitcnt_vm<0>(); break;
-    case 1: glds_waitcnt_vm<1>(); break;
-    case 2: glds_waitcnt_vm<2>(); break;
-    case 3: glds_waitcnt_vm<3>(); break;
-    case 4: glds_waitcnt_vm<4>(); break;
-    case 5: glds_waitcnt_vm<5>(); break;
-    case 6: glds_waitcnt_vm<6>(); break;
-    case 7: glds_waitcnt_vm<7>(); break;
-    default: glds_waitcnt_vm<8>(); break;
-  }
-}
-
-template <int BM, int BN, int BK, typename TO, int THREADS, bool NTC>
-__launch_bounds__(THREADS)
-__global__ void gemm_a16_glds_nt_deep_kernel(const __bf16* __restrict__ a,
-                                             const __bf16* __restrict__ b,
-                                             const float* __restrict__ bias,
-                                             TO* __restrict__ c, int m, int n,
-                                             int k, int relu) {
-  constexpr int STAGES = 3;
-  constexpr int NWAVE = THREADS / PERTGNN_WAVE;
-  constexpr int WCOL = NWAVE / 2;
-  constexpr int FM = (BM / 2) / 16, FN = (BN / WCOL) / 16;
-  constexpr int TILE_BYTES = BM * BK * 2;
-  constexpr int GRPS = TILE_BYTES / 1024;
-  constexpr int GRPS_PER_WAVE = GRPS / NWAVE;
-  constexpr int OPS_PER_STAGE = 2 * GRPS_PER_WAVE;  // A + B glds per thread
-  __shared__ char smem[STAGES * (BM + BN) * BK * 2];
-  const auto lds_ab = [&](int i) { return (__bf16*)(smem + i * TILE_BYTES); };
-  const auto lds_bb = [&](int i) {
-    return (__bf16*)(smem + STAGES * TILE_BYTES + i * (BN * BK * 2));
-  };
-  const int bid = xcd_swizzle(blockIdx.x, gridDim.x);
-  const int tiles_n = n / BN;
-  const int m0 = (bid / tiles_n) * BM;
-  const int n0 = (bid % tiles_n) * BN;
-  const int wave = threadIdx.x / PERTGNN_WAVE;
-  const int lane = threadIdx.x % PERTGNN_WAVE;
-  const int wm = (wave / WCOL) * (BM / 2);
-  const int wn = (wave % WCOL) * (BN / WCOL);
-
-  constexpr int ROW_BYTES = BK * 2;
-  constexpr int ROWS_PER_GRP = 1024 / ROW_BYTES;
-  const unsigned l_off = glds_swz((unsigned)lane * 16);
-  const int src_row = (int)(l_off / ROW_BYTES);
-  const int src_kb = (int)(l_off % ROW_BYTES);
-  const long lda = k;
-  auto stage = [&](const __bf16* g, int g0, int k0, __bf16* lds) {
-#pragma unroll
-    for (int i = 0; i < GRPS_PER_WAVE; ++i) {
-      const int grp = wave + i * NWAVE;
-      const __bf16* src = (const __bf16*)((const char*)g +
-          (long)(g0 + grp * ROWS_PER_GRP + src_row) * lda * 2 +
-          (long)k0 * 2 + src_kb);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)src,
-          (__attribute__((address_space(3))) void*)(lds + grp * 512),
-          16, 0, 0);
-    }
-  };
-
-  f32x4 acc[FM][FN];
-#pragma unroll
-  for (int i = 0; i < FM; ++i)
-#pragma unroll
-    for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  const int fi = lane & 15;
-  const int fk = (lane >> 4) * 8;
-  auto mma = [&](const __bf16* la, const __bf16* lb) {
-#pragma unroll
-    for (int s = 0; s < BK / 32; ++s) {
-      const int kb = (s * 32 + fk) * 2;
-      bf16x8 av[FM], bv[FN];
-#pragma unroll
-      for (int mi = 0; mi < FM; ++mi) {
-        const int row = wm + mi * 16 + fi;
-        av[mi] = *reinterpret_cast<const bf16x8*>(
-            (const char*)la + glds_swz((unsigned)(row * (BK * 2) + kb)));
-      }
-#pragma unroll
-      for (int ni = 0; ni < FN; ++ni) {
-        const int row = wn + ni * 16 + fi;
-        bv[ni] = *reinterpret_cast<const bf16x8*>(
-            (const char*)lb + glds_swz((unsigned)(row * (BK * 2) + kb)));
-      }
-#pragma unroll
-      for (int mi = 0; mi < FM; ++mi)
-#pragma unroll
-        for (int ni = 0; ni < FN; ++ni)
-          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              av[mi], bv[ni], acc[mi][ni], 0, 0, 0);
-    }
-  };
-
-  const int nk = k / BK;
-  const int pre = min(STAGES, nk);
-  for (int t = 0; t < pre; ++t) {
-    stage(a, m0, t * BK, lds_ab(t));
-    stage(b, n0, t * BK, lds_bb(t));
-  }
-  // steady state: 2 stages stay in flight across the barriers
-  int t = 0;
-  for (; t < nk - (STAGES - 1); ++t) {
-    glds_waitcnt_vm<(STAGES - 1) * OPS_PER_STAGE>();
-    __builtin_amdgcn_s_barrier();
-    mma(lds_ab(t % STAGES), lds_bb(t % STAGES));
-    __builtin_amdgcn_s_barrier();
-    if (t + STAGES < nk) {
-      stage(a, m0, (t + STAGES) * BK, lds_ab(t % STAGES));
-      stage(b, n0, (t + STAGES) * BK, lds_bb(t % STAGES));
-    }
-  }
-  // drain
-  for (; t < nk; ++t) {
-    glds_wait_upto((nk - 1 - t) * OPS_PER_STAGE);
-    __builtin_amdgcn_s_barrier();
-    mma(lds_ab(t % STAGES), lds_bb(t % STAGES));
-    __builtin_amdgcn_s_barrier();
-  }
-
-  // epilogue identical to the 2-buffer kernel (LDS-staged coalesced stores)
-  constexpr int BNW = BN / WCOL;
-  constexpr int SW = BNW + 8;
-  TO* scratch = reinterpret_cast<TO*>(smem + wave * 16 * SW * sizeof(TO));
-  const int fcol = lane & 15;
-  const int frow = (lane >> 4) * 4;
-  constexpr int EPL = 16 / sizeof(TO);
-  constexpr int CPR = BNW / EPL;
-#pragma unroll
-  for (int mi = 0; mi < FM; ++mi) {
-#pragma unroll
-    for (int ni = 0; ni < FN; ++ni)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int col = ni * 16 + fcol;
-        float v = acc[mi][ni][r];
-        if (bias) v += bias[n0 + wn + col];
-        if (relu) v = fmaxf(v, 0.f);
-        scratch[(frow + r) * SW + col] = (TO)v;
-      }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-#pragma unroll
-    for (int it = 0; it < 16 * CPR / PERTGNN_WAVE; ++it) {
-      const int chunk = it * PERTGNN_WAVE + lane;
-      const int r = chunk / CPR;
-      const int off = (chunk % CPR) * EPL;
-      const u32x4_t val = *reinterpret_cast<const u32x4_t*>(&scratch[r * SW + off]);
-      u32x4_t* dst = reinterpret_cast<u32x4_t*>(
-          &c[(long)(m0 + wm + mi * 16 + r) * n + n0 + wn + off]);
-      if constexpr (NTC)
-        __builtin_nontemporal_store(val, dst);
-      else
-        *dst = val;
-    }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  }
-}
-
 static int glds_bk() {  // default 32 (measured: +13-20% over 64 at model shapes)
   static int v = [] {
     const char* e = getenv("PERTGNN_GLDS_BK");
@@ -1273,13 +1095,7 @@ static bool glds_ntc() {  // nontemporal C stores (default ON)
   }();
   return v;
 }
-static bool glds_deep() {  // 3-stage counted-vmcnt ring (experimental)
-  static bool v = [] {
-    const char* e = getenv("PERTGNN_GLDS_DEEP");
-    return e && atoi(e) == 1;
-  }();
-  return v;
-}
+
 
 void launch_gemm_a16_glds_nt(const void* a_v, const void* b16_v,
                              const float* b32, const float* bias, void* c_v,
@@ -1305,16 +1121,7 @@ void launch_gemm_a16_glds_nt(const void* a_v, const void* b16_v,
           <<<dim3(grid), dim3(TH), 0, s>>>(a, b, bias, (float*)c_v, m, n, k,   \
                                            relu ? 1 : 0);                      \
   } while (0)
-    if (glds_deep() && k % 32 == 0) {
-      if (c16)
-        gemm_a16_glds_nt_deep_kernel<BM, BN, 32, __bf16, 512, true>
-            <<<dim3(grid), dim3(512), 0, s>>>(a, b, bias, (__bf16*)c_v, m, n,
-                                              k, relu ? 1 : 0);
-      else
-        gemm_a16_glds_nt_deep_kernel<BM, BN, 32, float, 512, true>
-            <<<dim3(grid), dim3(512), 0, s>>>(a, b, bias, (float*)c_v, m, n,
-                                              k, relu ? 1 : 0);
-    } else if (bk == 32 && k % 32 == 0) {
+    if (bk == 32 && k % 32 == 0) {
       if (t512) { if (ntc) GLDS_DISPATCH(32, 512, true); else GLDS_DISPATCH(32, 512, false); }
       else      { if (ntc) GLDS_DISPATCH(32, 256, true); else GLDS_DISPATCH(32, 256, false); }
     } else {
