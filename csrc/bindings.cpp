@@ -150,11 +150,12 @@ void launch_edge_attn_fused_bwd(const float*, const float*, const float*,
                                 const int*, const int*, const int*,
                                 const int*, const int*, float*, float*,
                                 float*, int, int, long, hipStream_t);
-void launch_edge_attn_fused_fwd16(const void*, const float*, const float*,
-                                  const long*, int, const int*, const int*,
-                                  void*, int, float*, int, int, hipStream_t);
+void launch_edge_attn_fused_fwd16(const void*, const void*, const void*,
+                                  int, const long*, int, const int*,
+                                  const int*, void*, int, float*, int, int,
+                                  hipStream_t);
 void launch_edge_attn_fused_bwd16(const void*, int, const void*,
-                                  const float*, const float*, const long*,
+                                  const void*, const void*, int, const long*,
                                   int, const float*, const int*, const int*,
                                   const int*, const int*, const int*, void*,
                                   void*, float*, int, int, long, hipStream_t);
@@ -651,13 +652,17 @@ std::vector<torch::Tensor> edge_attn_fused_fwd(
   TORCH_CHECK(h <= 512, "H must be <= 512");
   auto fopt = qkvs.options().dtype(torch::kFloat32);
   const bool b16 = qkvs.scalar_type() == torch::kBFloat16;
+  const bool p16 = pifc.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(!out16 || b16, "out16 requires bf16 qkvs");
+  TORCH_CHECK(!p16 || b16, "bf16 P tables require bf16 qkvs");
+  TORCH_CHECK(pifc.scalar_type() == prpc.scalar_type(),
+              "P tables must share a dtype");
   auto out = torch::empty({n, h}, out16 ? qkvs.options() : fopt);
   auto alpha = torch::empty({ea.size(0)}, fopt);
   if (b16) {
     launch_edge_attn_fused_fwd16(
-        qkvs.data_ptr(), pifc.data_ptr<float>(),
-        prpc.data_ptr<float>(), ea.data_ptr<long>(), (int)ea.size(1),
+        qkvs.data_ptr(), pifc.data_ptr(), prpc.data_ptr(), p16 ? 1 : 0,
+        ea.data_ptr<long>(), (int)ea.size(1),
         row_ptr.data_ptr<int>(), csr_src.data_ptr<int>(),
         out.data_ptr(), out16 ? 1 : 0, alpha.data_ptr<float>(), n, h,
         cur_stream());
@@ -691,9 +696,10 @@ std::vector<torch::Tensor> edge_attn_fused_bwd(
   auto dal = torch::empty({ne}, fopt);
   if (b16) {
     const int g16 = g.scalar_type() == torch::kBFloat16 ? 1 : 0;
+    const int p16 = pifc.scalar_type() == torch::kBFloat16 ? 1 : 0;
     launch_edge_attn_fused_bwd16(
         g.data_ptr(), g16, qkvs.data_ptr(),
-        pifc.data_ptr<float>(), prpc.data_ptr<float>(), ea.data_ptr<long>(),
+        pifc.data_ptr(), prpc.data_ptr(), p16, ea.data_ptr<long>(),
         (int)ea.size(1), alpha.data_ptr<float>(), row_ptr.data_ptr<int>(),
         csr_src.data_ptr<int>(), col_ptr.data_ptr<int>(),
         csc_dst.data_ptr<int>(), csc_eid.data_ptr<int>(), dqkvs.data_ptr(),

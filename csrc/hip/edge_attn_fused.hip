@@ -90,6 +90,27 @@ struct Slice {
       }
     }
   }
+  static __device__ __forceinline__ void load_add(const __bf16* a,
+                                                  const __bf16* b, int lane,
+                                                  int h, float (&dst)[VPT]) {
+    if constexpr (VEC) {
+#pragma unroll
+      for (int q = 0; q < VPT; q += 4) {
+        const bf16x4_t va =
+            *reinterpret_cast<const bf16x4_t*>(&a[lane * VPT + q]);
+        const bf16x4_t vb =
+            *reinterpret_cast<const bf16x4_t*>(&b[lane * VPT + q]);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) dst[q + u] = (float)va[u] + (float)vb[u];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPT; ++j) {
+        const int c = lane + j * PERTGNN_WAVE;
+        dst[j] = (c < h) ? (float)a[c] + (float)b[c] : 0.f;
+      }
+    }
+  }
   static __device__ __forceinline__ void store(float* base, int lane, int h,
                                                const float (&src)[VPT]) {
     if constexpr (VEC) {
@@ -131,11 +152,11 @@ struct Slice {
 // owns h/LPR contiguous columns); reductions become sub-wave shfl_xor trees.
 // LPR=32 measured best at H=256 (see attn_lpr below).
 template <int VPT, bool VEC, typename QT = float, int LPR = PERTGNN_WAVE,
-          typename TO = float>
+          typename TO = float, typename PT = float>
 __global__ void edge_attn_fused_fwd_kernel(
     const QT* __restrict__ qkvs,  // [N, 4h]
-    const float* __restrict__ pifc,  // [Vi, h]
-    const float* __restrict__ prpc,  // [Vr, h]
+    const PT* __restrict__ pifc,  // [Vi, h] (bf16 in act16 mode — halves
+    const PT* __restrict__ prpc,  // [Vr, h]  the per-edge ec gathers)
     const long* __restrict__ ea, int astride,
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
     TO* __restrict__ out, float* __restrict__ alpha, int n, int h,
@@ -194,10 +215,10 @@ __global__ void edge_attn_fused_fwd_kernel(
 }
 
 template <int VPT, bool VEC, typename QT = float,
-          int LPR = PERTGNN_WAVE, typename TG = float>
+          int LPR = PERTGNN_WAVE, typename TG = float, typename PT = float>
 __global__ void edge_attn_fused_bwd_row_kernel(
     const TG* __restrict__ g, const QT* __restrict__ qkvs,
-    const float* __restrict__ pifc, const float* __restrict__ prpc,
+    const PT* __restrict__ pifc, const PT* __restrict__ prpc,
     const long* __restrict__ ea, int astride, const float* __restrict__ alpha,
     const int* __restrict__ row_ptr, const int* __restrict__ csr_src,
     QT* __restrict__ dqkvs, float* __restrict__ dal, int n, int h,
@@ -390,8 +411,8 @@ static int attn_lpr() {
   return v;
 }
 
-void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
-                                  const float* prpc, const long* ea,
+void launch_edge_attn_fused_fwd16(const void* qkvs_v, const void* pifc_v,
+                                  const void* prpc_v, int p16, const long* ea,
                                   int astride, const int* row_ptr,
                                   const int* csr_src, void* out_v, int out16,
                                   float* alpha, int n, int h,
@@ -405,16 +426,31 @@ void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
   const dim3 grid(ceil_div(n, rpb));
 #define FWD16(VPT, LPR)                                                        \
   do {                                                                         \
-    if (out16)                                                                 \
+    if (p16 && out16) {                                                        \
+      edge_attn_fused_fwd_kernel<VPT, true, __bf16, LPR, __bf16, __bf16>       \
+          <<<grid, block, 0, stream>>>(qkvs, (const __bf16*)pifc_v,            \
+                                       (const __bf16*)prpc_v, ea, astride,     \
+                                       row_ptr, csr_src, (__bf16*)out_v,       \
+                                       alpha, n, h, scale);                    \
+    } else if (p16) {                                                          \
+      edge_attn_fused_fwd_kernel<VPT, true, __bf16, LPR, float, __bf16>        \
+          <<<grid, block, 0, stream>>>(qkvs, (const __bf16*)pifc_v,            \
+                                       (const __bf16*)prpc_v, ea, astride,     \
+                                       row_ptr, csr_src, (float*)out_v,        \
+                                       alpha, n, h, scale);                    \
+    } else if (out16) {                                                        \
       edge_attn_fused_fwd_kernel<VPT, true, __bf16, LPR, __bf16>               \
-          <<<grid, block, 0, stream>>>(qkvs, pifc, prpc, ea, astride, row_ptr, \
-                                       csr_src, (__bf16*)out_v, alpha, n, h,   \
-                                       scale);                                 \
-    else                                                                       \
+          <<<grid, block, 0, stream>>>(qkvs, (const float*)pifc_v,             \
+                                       (const float*)prpc_v, ea, astride,      \
+                                       row_ptr, csr_src, (__bf16*)out_v,       \
+                                       alpha, n, h, scale);                    \
+    } else {                                                                   \
       edge_attn_fused_fwd_kernel<VPT, true, __bf16, LPR, float>                \
-          <<<grid, block, 0, stream>>>(qkvs, pifc, prpc, ea, astride, row_ptr, \
-                                       csr_src, (float*)out_v, alpha, n, h,    \
-                                       scale);                                 \
+          <<<grid, block, 0, stream>>>(qkvs, (const float*)pifc_v,             \
+                                       (const float*)prpc_v, ea, astride,      \
+                                       row_ptr, csr_src, (float*)out_v,        \
+                                       alpha, n, h, scale);                    \
+    }                                                                          \
   } while (0)
   if (h == 256) {
     if (lpr == 16) FWD16(16, 16);
@@ -437,8 +473,8 @@ void launch_edge_attn_fused_fwd16(const void* qkvs_v, const float* pifc,
 
 void launch_edge_attn_fused_bwd16(const void* g_v, int g16,
                                   const void* qkvs_v,
-                                  const float* pifc, const float* prpc,
-                                  const long* ea, int astride,
+                                  const void* pifc_v, const void* prpc_v,
+                                  int p16, const long* ea, int astride,
                                   const float* alpha, const int* row_ptr,
                                   const int* csr_src, const int* col_ptr,
                                   const int* csc_dst, const int* csc_eid,
@@ -456,9 +492,22 @@ void launch_edge_attn_fused_bwd16(const void* g_v, int g16,
   const dim3 grid(ceil_div(n, rpb));
 #define BWD16(VPT, LPR)                                                        \
   do {                                                                         \
-    if (g16) {                                                                 \
+    if (g16 && p16) {                                                          \
+      edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, LPR, __bf16, __bf16>   \
+          <<<grid, block, 0, stream>>>((const __bf16*)g_v, qkvs,               \
+                                       (const __bf16*)pifc_v,                  \
+                                       (const __bf16*)prpc_v,                  \
+                                       ea, astride, alpha, row_ptr, csr_src,   \
+                                       dqkvs, dal, n, h, scale);               \
+      edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR, __bf16>   \
+          <<<grid, block, 0, stream>>>((const __bf16*)g_v, qkvs, alpha, dal,   \
+                                       col_ptr, csc_dst, csc_eid, dqkvs, de,   \
+                                       n, h);                                  \
+    } else if (g16) {                                                          \
       edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, LPR, __bf16>           \
-          <<<grid, block, 0, stream>>>((const __bf16*)g_v, qkvs, pifc, prpc,   \
+          <<<grid, block, 0, stream>>>((const __bf16*)g_v, qkvs,               \
+                                       (const float*)pifc_v,                   \
+                                       (const float*)prpc_v,                   \
                                        ea, astride, alpha, row_ptr, csr_src,   \
                                        dqkvs, dal, n, h, scale);               \
       edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR, __bf16>   \
@@ -467,7 +516,9 @@ void launch_edge_attn_fused_bwd16(const void* g_v, int g16,
                                        n, h);                                  \
     } else {                                                                   \
       edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, LPR, float>            \
-          <<<grid, block, 0, stream>>>((const float*)g_v, qkvs, pifc, prpc,    \
+          <<<grid, block, 0, stream>>>((const float*)g_v, qkvs,                \
+                                       (const float*)pifc_v,                   \
+                                       (const float*)prpc_v,                   \
                                        ea, astride, alpha, row_ptr, csr_src,   \
                                        dqkvs, dal, n, h, scale);               \
       edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR, float>    \

@@ -203,13 +203,19 @@ class TransformerConv(nn.Module):
         # staging path
         if x.shape[1] != self.k_padded:
             x = torch.nn.functional.pad(x, (0, self.k_padded - x.shape[1]))
-        if (ops.gemm_precision() in ("bf16", "fp16") and h % 256 == 0
-                and ops.act16_enabled()):
+        act16 = (ops.gemm_precision() in ("bf16", "fp16") and h % 256 == 0
+                 and ops.act16_enabled())
+        if act16:
             qkvs = ops.linear16(x, self.w4, self.b4)
+            # bf16 P tables: halves the per-edge ec gather bytes AND the
+            # L2 footprint of the tables (2x256H fp32 = 4 MB at realistic
+            # vocab vs 4 MB L2 per XCD); logits/softmax stay fp32
+            pifc = ops.linear16(ifc_weight, self.we_ifc)
+            prpc = ops.linear16(rpc_weight, self.we_rpc)
         else:
             qkvs = ops.linear(x, self.w4, self.b4)
-        pifc = ops.linear(ifc_weight, self.we_ifc, None)
-        prpc = ops.linear(rpc_weight, self.we_rpc, None)
+            pifc = ops.linear(ifc_weight, self.we_ifc, None)
+            prpc = ops.linear(rpc_weight, self.we_rpc, None)
         return ops.edge_attention_fused(qkvs, pifc, prpc, edge_attr, csr,
                                         out16=out16)
 

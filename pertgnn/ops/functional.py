@@ -282,6 +282,9 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
             else:
                 dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], h, 0)
                 dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], h, 0)
+            if dpifc.dtype != pifc.dtype:  # bf16 P tables: match grad dtype
+                dpifc = dpifc.to(pifc.dtype)
+                dprpc = dprpc.to(prpc.dtype)
             return dqkvs, dpifc, dprpc, None, None, None, None, None, None, None
         cur = torch.cuda.current_stream()
         side = _side_stream()
@@ -291,6 +294,9 @@ class _EdgeAttentionFusedFn(torch.autograd.Function):
         with torch.cuda.stream(side):
             dpifc = _table_grad(m, de, edge_attr[:, 0], pifc.shape[0], de.shape[1], 0)
             dprpc = _table_grad(m, de, edge_attr[:, 1], prpc.shape[0], de.shape[1], 0)
+            if dpifc.dtype != pifc.dtype:
+                dpifc = dpifc.to(pifc.dtype)
+                dprpc = dprpc.to(prpc.dtype)
         ev2 = torch.cuda.Event()
         ev2.record(side)
         cur.wait_event(ev2)
