@@ -1126,7 +1126,7 @@ torch::Tensor linear_fwd_a16o16(torch::Tensor x, torch::Tensor w,
   // glds fast path (direct-to-LDS staging): needs 16-B-aligned rows, full
   // BK tiles in k, and n a multiple of the 128-wide tile.  The bf16 weight
   // copy costs one trivial convert kernel per call (n*k elements).
-  if (m >= 512 && n % 128 == 0 && k % 64 == 0) {
+  if (m >= 512 && n % 128 == 0 && k % 32 == 0) {
     auto w16 = torch::empty({n, k}, x.options());  // bf16
     launch_convert_w16(w.data_ptr<float>(), w16.data_ptr(), (long)n * k,
                        cur_stream());

@@ -514,6 +514,19 @@ void launch_edge_attn_fused_bwd16(const void* g_v, int g16,
           <<<grid, block, 0, stream>>>((const __bf16*)g_v, qkvs, alpha, dal,   \
                                        col_ptr, csc_dst, csc_eid, dqkvs, de,   \
                                        n, h);                                  \
+    } else if (p16) {                                                          \
+      /* fp32 upstream grad (the last conv's out16=false) + bf16 P tables — \
+         reading the tables at the wrong width over-reads 2x (OOB) */        \
+      edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, LPR, float, __bf16>    \
+          <<<grid, block, 0, stream>>>((const float*)g_v, qkvs,                \
+                                       (const __bf16*)pifc_v,                  \
+                                       (const __bf16*)prpc_v,                  \
+                                       ea, astride, alpha, row_ptr, csr_src,   \
+                                       dqkvs, dal, n, h, scale);               \
+      edge_attn_fused_bwd_col_kernel<VPT, true, __bf16, __bf16, LPR, float>    \
+          <<<grid, block, 0, stream>>>((const float*)g_v, qkvs, alpha, dal,    \
+                                       col_ptr, csc_dst, csc_eid, dqkvs, de,   \
+                                       n, h);                                  \
     } else {                                                                   \
       edge_attn_fused_bwd_row_kernel<VPT, true, __bf16, LPR, float>            \
           <<<grid, block, 0, stream>>>((const float*)g_v, qkvs,                \

@@ -1121,7 +1121,7 @@ void launch_gemm_a16_glds_nt(const void* a_v, const void* b16_v,
           <<<dim3(grid), dim3(TH), 0, s>>>(a, b, bias, (float*)c_v, m, n, k,   \
                                            relu ? 1 : 0);                      \
   } while (0)
-    if (bk == 32 && k % 32 == 0) {
+    if ((bk == 32 || k % 64 != 0) && k % 32 == 0) {
       if (t512) { if (ntc) GLDS_DISPATCH(32, 512, true); else GLDS_DISPATCH(32, 512, false); }
       else      { if (ntc) GLDS_DISPATCH(32, 256, true); else GLDS_DISPATCH(32, 256, false); }
     } else {
