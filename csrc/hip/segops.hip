@@ -1014,7 +1014,10 @@ __device__ __forceinline__ float bn_rand01(unsigned long long seed, long t) {
 // TY = output/activation dtype: fp32 in exact mode, bf16 in the act16 mode
 // (standard mixed-precision BN: statistics and normalization math stay fp32,
 // only the activation stream is 16-bit).
-template <typename TY, typename TX = float>
+// DROP is a TEMPLATE parameter: the hash-RNG code measurably bloats the
+// 4-wide loop even when dropout_p==0 at runtime (bn_apply 39 -> 168 us at
+// 181k x 256), so the p==0 instantiation must not contain it.
+template <typename TY, typename TX = float, bool DROP = false>
 __global__ void bn_apply_kernel(const TX* __restrict__ x,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
@@ -1025,8 +1028,9 @@ __global__ void bn_apply_kernel(const TX* __restrict__ x,
                                 const unsigned long long* __restrict__
                                     seed_ptr = nullptr) {
   typedef __attribute__((ext_vector_type(4))) float bnf4;
-  const float keep_inv = (dropout_p > 0.f) ? 1.f / (1.f - dropout_p) : 1.f;
-  const unsigned long long seed = seed_ptr ? *seed_ptr : 0ull;
+  const float keep_inv = DROP ? 1.f / (1.f - dropout_p) : 1.f;
+  unsigned long long seed = 0ull;
+  if constexpr (DROP) seed = seed_ptr ? *seed_ptr : 0ull;
   if ((h & 3) == 0) {  // 4-wide: one f32x4 load + packed store per thread
     const long numq = n * (h / 4);
     const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1043,7 +1047,7 @@ __global__ void bn_apply_kernel(const TX* __restrict__ x,
         float v = (xv[u] - mean[c + u]) * invstd[c + u] * gamma[c + u] +
                   beta[c + u];
         if (relu) v = fmaxf(v, 0.f);
-        if (dropout_p > 0.f)
+        if constexpr (DROP)
           v = (bn_rand01(seed, t + u) >= dropout_p) ? v * keep_inv : 0.f;
         o.v[u] = (TY)v;
       }
@@ -1058,7 +1062,7 @@ __global__ void bn_apply_kernel(const TX* __restrict__ x,
     const int c = (int)(t % h);
     float v = ((float)x[t] - mean[c]) * invstd[c] * gamma[c] + beta[c];
     if (relu) v = fmaxf(v, 0.f);
-    if (dropout_p > 0.f)
+    if constexpr (DROP)
       v = (bn_rand01(seed, t) >= dropout_p) ? v * keep_inv : 0.f;
     y[t] = (TY)v;
   }
@@ -1284,12 +1288,16 @@ void launch_bn_finalize_apply(const float* x, const float* partials,
     bn_eval_stats_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
         running_mean, running_var, mean, invstd, h, eps);
   }
-  if (n > 0)
-    bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma,
-                                                    beta, y, n, h,
-                                                    relu ? 1 : 0,
-                                                    training ? dropout_p : 0.f,
-                                                    seed_ptr);
+  if (n > 0) {
+    if (training && dropout_p > 0.f)
+      bn_apply_kernel<float, float, true><<<grid_for(n * h), 256, 0, s>>>(
+          x, mean, invstd, gamma, beta, y, n, h, relu ? 1 : 0, dropout_p,
+          seed_ptr);
+    else
+      bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma,
+                                                      beta, y, n, h,
+                                                      relu ? 1 : 0);
+  }
 }
 
 void launch_bn_bwd_partials_only(const float* g, const float* x,
@@ -1337,10 +1345,14 @@ void launch_bn_fwd(const float* x, const float* gamma, const float* beta,
   } else {
     launch_bn_eval_stats(running_mean, running_var, mean, invstd, h, eps, s);
   }
-  bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma, beta,
-                                                  y, n, h, relu ? 1 : 0,
-                                                  training ? dropout_p : 0.f,
-                                                  seed_ptr);
+  if (training && dropout_p > 0.f)
+    bn_apply_kernel<float, float, true><<<grid_for(n * h), 256, 0, s>>>(
+        x, mean, invstd, gamma, beta, y, n, h, relu ? 1 : 0, dropout_p,
+        seed_ptr);
+  else
+    bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>(x, mean, invstd, gamma,
+                                                    beta, y, n, h,
+                                                    relu ? 1 : 0);
 }
 
 void launch_counter_bump(unsigned long long* c, hipStream_t s) {
@@ -1389,12 +1401,15 @@ void launch_bn_fwd16(const void* x, const float* gamma, const float* beta,
   } else {
     launch_bn_eval_stats(running_mean, running_var, mean, invstd, h, eps, s);
   }
-  bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>((const __bf16*)x, mean,
-                                                  invstd, gamma, beta,
-                                                  (__bf16*)y, n, h,
-                                                  relu ? 1 : 0,
-                                                  training ? dropout_p : 0.f,
-                                                  seed_ptr);
+  if (training && dropout_p > 0.f)
+    bn_apply_kernel<__bf16, __bf16, true><<<grid_for(n * h), 256, 0, s>>>(
+        (const __bf16*)x, mean, invstd, gamma, beta, (__bf16*)y, n, h,
+        relu ? 1 : 0, dropout_p, seed_ptr);
+  else
+    bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>((const __bf16*)x, mean,
+                                                    invstd, gamma, beta,
+                                                    (__bf16*)y, n, h,
+                                                    relu ? 1 : 0);
 }
 
 void launch_bn_finalize_apply16(const void* x, const float* partials,
@@ -1414,13 +1429,17 @@ void launch_bn_finalize_apply16(const void* x, const float* partials,
     bn_eval_stats_kernel<<<ceil_div(h, 256), 256, 0, s>>>(
         running_mean, running_var, mean, invstd, h, eps);
   }
-  if (n > 0)
-    bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>((const __bf16*)x, mean,
-                                                    invstd, gamma, beta,
-                                                    (__bf16*)y, n, h,
-                                                    relu ? 1 : 0,
-                                                    training ? dropout_p : 0.f,
-                                                    seed_ptr);
+  if (n > 0) {
+    if (training && dropout_p > 0.f)
+      bn_apply_kernel<__bf16, __bf16, true><<<grid_for(n * h), 256, 0, s>>>(
+          (const __bf16*)x, mean, invstd, gamma, beta, (__bf16*)y, n, h,
+          relu ? 1 : 0, dropout_p, seed_ptr);
+    else
+      bn_apply_kernel<<<grid_for(n * h), 256, 0, s>>>((const __bf16*)x, mean,
+                                                      invstd, gamma, beta,
+                                                      (__bf16*)y, n, h,
+                                                      relu ? 1 : 0);
+  }
 }
 
 void launch_bn_bwd_partials_only16(const void* g, const void* x,
