@@ -256,15 +256,19 @@ __global__ void embed_grouped_p2_kernel(const float* __restrict__ partial,
   }
 }
 
-// Work-balanced variant: the uniform rows*P wave grid collapses under
-// skewed group sizes (quirk 6: PERT intra-ms edges all carry interface id 0,
-// so ONE table row owns ~half the edges — 8 sub-waves serialized ~12k
-// dependent gathers each, measured 39 ms vs 1 ms for an even distribution
-// of the same volume).  Here the host assigns waves PROPORTIONAL to group
-// size (row_map[w] = row of wave w, wave_start[row] = its first wave), so
-// every wave runs a bounded iteration count regardless of skew.  Reduction
-// order per row is (sub, stride P_row) — fixed by the data, deterministic.
-template <int VPT, typename TG = float>
+// order==nullptr reads g rows directly (r = p): that is how the SECOND
+// reduction level folds the level-1 partials — same balanced assignment,
+// with ptr = level-1 wave_start (each row's partials are contiguous).
+// VEC: contiguous lane*VPT column mapping with packed 8-B (bf16) / 16-B
+// (fp32) row loads — the strided lane+j*64 mapping issues VPT scalar loads
+// per gathered row, which dominates this latency-bound kernel.  Requires
+// h == VPT*64, VPT%4==0, and 4-element-aligned rows (gstride%4, col_off%4).
+// The column mapping is internal: p1, the level-2 fold and p2 all use the
+// same c_of(j), so dtable comes out identical either way.
+typedef __attribute__((ext_vector_type(4))) float segf4;
+typedef __attribute__((ext_vector_type(4))) __bf16 segb4;
+
+template <int VPT, typename TG = float, bool VEC = false>
 __global__ void embed_grouped_bal_p1_kernel(
     const TG* __restrict__ g, const int* __restrict__ order,
     const int* __restrict__ ptr, const int* __restrict__ row_map,
@@ -282,24 +286,39 @@ __global__ void embed_grouped_bal_p1_kernel(
 #pragma unroll
   for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
   for (int p = ptr[row] + sub; p < ptr[row + 1]; p += P) {
-    // order==nullptr reads rows of g directly (r = p): the SECOND reduction
-    // level folds level-1 partials this way (ptr = level-1 wave_start,
-    // each row's partials contiguous)
     const long r = order ? order[p] : p;
+    const TG* src = &g[(long)r * gstride + col_off];
+    if constexpr (VEC) {
 #pragma unroll
-    for (int j = 0; j < VPT; ++j) {
-      const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) acc[j] += (float)g[(long)r * gstride + col_off + c];
+      for (int q = 0; q < VPT; q += 4) {
+        if constexpr (sizeof(TG) == 2) {
+          const segb4 v = *reinterpret_cast<const segb4*>(
+              (const __bf16*)src + lane * VPT + q);
+#pragma unroll
+          for (int u = 0; u < 4; ++u) acc[q + u] += (float)v[u];
+        } else {
+          const segf4 v = *reinterpret_cast<const segf4*>(
+              (const float*)src + lane * VPT + q);
+#pragma unroll
+          for (int u = 0; u < 4; ++u) acc[q + u] += v[u];
+        }
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPT; ++j) {
+        const int c = lane + j * PERTGNN_WAVE;
+        if (c < h) acc[j] += (float)src[c];
+      }
     }
   }
 #pragma unroll
   for (int j = 0; j < VPT; ++j) {
-    const int c = lane + j * PERTGNN_WAVE;
+    const int c = VEC ? lane * VPT + j : lane + j * PERTGNN_WAVE;
     if (c < h) partial[(long)w * h + c] = acc[j];
   }
 }
 
-template <int VPT>
+template <int VPT, bool VEC = false>
 __global__ void embed_grouped_bal_p2_kernel(
     const float* __restrict__ partial, const int* __restrict__ wave_start,
     float* __restrict__ dtable, int rows, int h) {
@@ -311,15 +330,25 @@ __global__ void embed_grouped_bal_p2_kernel(
 #pragma unroll
   for (int j = 0; j < VPT; ++j) acc[j] = 0.f;
   for (int s = wave_start[row]; s < wave_start[row + 1]; ++s) {
+    if constexpr (VEC) {
 #pragma unroll
-    for (int j = 0; j < VPT; ++j) {
-      const int c = lane + j * PERTGNN_WAVE;
-      if (c < h) acc[j] += partial[(long)s * h + c];
+      for (int q = 0; q < VPT; q += 4) {
+        const segf4 v = *reinterpret_cast<const segf4*>(
+            &partial[(long)s * h + lane * VPT + q]);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) acc[q + u] += v[u];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VPT; ++j) {
+        const int c = lane + j * PERTGNN_WAVE;
+        if (c < h) acc[j] += partial[(long)s * h + c];
+      }
     }
   }
 #pragma unroll
   for (int j = 0; j < VPT; ++j) {
-    const int c = lane + j * PERTGNN_WAVE;
+    const int c = VEC ? lane * VPT + j : lane + j * PERTGNN_WAVE;
     if (c < h) dtable[(long)row * h + c] = acc[j];
   }
 }
@@ -336,8 +365,13 @@ void launch_embed_grouped_scatter_bal(
   if (rows == 0) return;
   const dim3 block(WAVES_PER_BLOCK * PERTGNN_WAVE);
   const int vpt = (h + PERTGNN_WAVE - 1) / PERTGNN_WAVE;
+  // packed-load column mapping when the whole row divides evenly and the
+  // g rows are 4-element aligned (the hot de/entry cases; the embed-node
+  // grad with col_off=9 keeps the strided scalar path)
+  const bool vec = (h == vpt * PERTGNN_WAVE) && (vpt % 4 == 0) &&
+                   (gstride % 4 == 0) && (col_off % 4 == 0);
   switch (vpt) {
-#define CASE(V) case V: { if (g16) { embed_grouped_bal_p1_kernel<V, __bf16><<<dim3(ceil_div(n_waves, WAVES_PER_BLOCK)), block, 0, s>>>((const __bf16*)g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); } else { embed_grouped_bal_p1_kernel<V, float><<<dim3(ceil_div(n_waves, WAVES_PER_BLOCK)), block, 0, s>>>((const float*)g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); } if (n_waves2 > 0) { embed_grouped_bal_p1_kernel<V, float><<<dim3(ceil_div(n_waves2, WAVES_PER_BLOCK)), block, 0, s>>>(partial, nullptr, wave_start, row_map2, wave_start2, partial2, n_waves2, h, h, 0); embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial2, wave_start2, dtable, rows, h); } else { embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial, wave_start, dtable, rows, h); } } break;
+#define CASE(V) case V: { const int grid1 = ceil_div(n_waves, WAVES_PER_BLOCK); if (vec) { if (g16) { embed_grouped_bal_p1_kernel<V, __bf16, true><<<dim3(grid1), block, 0, s>>>((const __bf16*)g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); } else { embed_grouped_bal_p1_kernel<V, float, true><<<dim3(grid1), block, 0, s>>>((const float*)g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); } if (n_waves2 > 0) { embed_grouped_bal_p1_kernel<V, float, true><<<dim3(ceil_div(n_waves2, WAVES_PER_BLOCK)), block, 0, s>>>(partial, nullptr, wave_start, row_map2, wave_start2, partial2, n_waves2, h, h, 0); embed_grouped_bal_p2_kernel<V, true><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial2, wave_start2, dtable, rows, h); } else { embed_grouped_bal_p2_kernel<V, true><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial, wave_start, dtable, rows, h); } } else { if (g16) { embed_grouped_bal_p1_kernel<V, __bf16><<<dim3(grid1), block, 0, s>>>((const __bf16*)g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); } else { embed_grouped_bal_p1_kernel<V, float><<<dim3(grid1), block, 0, s>>>((const float*)g, order, ptr, row_map, wave_start, partial, n_waves, h, gstride, col_off); } if (n_waves2 > 0) { embed_grouped_bal_p1_kernel<V, float><<<dim3(ceil_div(n_waves2, WAVES_PER_BLOCK)), block, 0, s>>>(partial, nullptr, wave_start, row_map2, wave_start2, partial2, n_waves2, h, h, 0); embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial2, wave_start2, dtable, rows, h); } else { embed_grouped_bal_p2_kernel<V><<<dim3(ceil_div(rows, WAVES_PER_BLOCK)), block, 0, s>>>(partial, wave_start, dtable, rows, h); } } } break;
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
 #undef CASE
     default: pertgnn_shape_fail("segops launcher", "vpt", vpt);
