@@ -213,16 +213,15 @@ class TransformerConv(nn.Module):
                  and ops.act16_enabled())
         if act16:
             qkvs = ops.linear16(x, self.w4, self.b4)
-            import os
-            if os.environ.get("PERTGNN_NO_P16", "0") == "1":
-                pifc = ops.linear(ifc_weight, self.we_ifc, None)
-                prpc = ops.linear(rpc_weight, self.we_rpc, None)
-            else:
+            if ops.p16_enabled():
                 # bf16 P tables: halves the per-edge ec gather bytes AND the
                 # L2 footprint of the tables (2x256H fp32 = 4 MB at realistic
                 # vocab vs 4 MB L2 per XCD); logits/softmax stay fp32
                 pifc = ops.linear16(ifc_weight, self.we_ifc)
                 prpc = ops.linear16(rpc_weight, self.we_rpc)
+            else:
+                pifc = ops.linear(ifc_weight, self.we_ifc, None)
+                prpc = ops.linear(rpc_weight, self.we_rpc, None)
         else:
             qkvs = ops.linear(x, self.w4, self.b4)
             pifc = ops.linear(ifc_weight, self.we_ifc, None)

@@ -237,6 +237,16 @@ def linear16(x, w, b=None):
 
 
 _ACT16 = None
+_P16 = None
+
+
+def p16_enabled() -> bool:
+    """bf16 P tables in act16 mode (PERTGNN_NO_P16=1 keeps them fp32)."""
+    global _P16
+    if _P16 is None:
+        import os
+        _P16 = os.environ.get("PERTGNN_NO_P16", "0") != "1"
+    return _P16
 
 
 def act16_enabled() -> bool:
@@ -751,6 +761,7 @@ def linear(x, w, b=None):
 
 __all__ = [
     "set_gemm_precision",
+    "p16_enabled",
     "gemm_precision",
     "linear16",
     "act16_enabled",
