@@ -644,7 +644,13 @@ static void vocab_dual_impl(const GT* g, const long* ea, int astride,
   if (n == 0) return;
   // wave-private path (7.2x the LDS-atomic kernel, see kernel comment)
   const long vrows = rows0 + rows1;
-  const int blocks = (int)min((long)128, (n + 255) / 256);
+  // block count scaled inversely with table size: each extra block
+  // costs a table-sized atomic merge, but small tables (rpctype ~5
+  // rows) need far more than 128 blocks to occupy 256 CUs (measured
+  // 1 wave/SIMD at the old cap)
+  const long merge_cap = (long)1000000 / (long)max((int)vrows * 128, 1);
+  const int blocks =
+      (int)min((n + 255) / 256, max((long)128, min((long)768, merge_cap)));
   const size_t lds128 = (size_t)WAVES_PER_BLOCK * vrows * 128 * sizeof(float);
   const size_t lds64 = (size_t)WAVES_PER_BLOCK * vrows * 64 * sizeof(float);
   if (h % 128 == 0 && lds128 <= 160 * 1024) {
@@ -757,7 +763,13 @@ void launch_vocab_scatter16(const void* g_v, const long* idx, long idx_stride,
   const __bf16* g = (const __bf16*)g_v;
   HIP_CHECK(hipMemsetAsync(dtable, 0, (long)rows * h * sizeof(float), s));
   if (n == 0) return;
-  const int blocks = (int)min((long)128, (n + 255) / 256);
+  // block count scaled inversely with table size: each extra block
+  // costs a table-sized atomic merge, but small tables (rpctype ~5
+  // rows) need far more than 128 blocks to occupy 256 CUs (measured
+  // 1 wave/SIMD at the old cap)
+  const long merge_cap = (long)1000000 / (long)max(rows * 128, 1);
+  const int blocks =
+      (int)min((n + 255) / 256, max((long)128, min((long)768, merge_cap)));
   const size_t lds128 = (size_t)WAVES_PER_BLOCK * rows * 128 * sizeof(float);
   const size_t lds64 = (size_t)WAVES_PER_BLOCK * rows * 64 * sizeof(float);
   if (h % 128 == 0 && lds128 <= 160 * 1024) {
@@ -789,7 +801,13 @@ void launch_vocab_scatter(const float* g, const long* idx, long idx_stride,
   HIP_CHECK(hipMemsetAsync(dtable, 0, (long)rows * h * sizeof(float), s));
   if (n == 0) return;
   {
-    const int blocks = (int)min((long)128, (n + 255) / 256);
+    // block count scaled inversely with table size: each extra block
+    // costs a table-sized atomic merge, but small tables (rpctype ~5
+    // rows) need far more than 128 blocks to occupy 256 CUs (measured
+    // 1 wave/SIMD at the old cap)
+    const long merge_cap = (long)1000000 / (long)max(rows * 128, 1);
+    const int blocks =
+        (int)min((n + 255) / 256, max((long)128, min((long)768, merge_cap)));
     const size_t lds128 = (size_t)WAVES_PER_BLOCK * rows * 128 * sizeof(float);
     const size_t lds64 = (size_t)WAVES_PER_BLOCK * rows * 64 * sizeof(float);
     if (h % 128 == 0 && lds128 <= 160 * 1024) {
