@@ -21,7 +21,6 @@ import numpy as np
 import pandas as pd
 import torch
 
-from .schema import NUM_RESOURCE_FEATURES
 
 
 @dataclass

@@ -20,7 +20,6 @@ container class differs.
 """
 from __future__ import annotations
 
-import io
 import pickle
 
 import torch
