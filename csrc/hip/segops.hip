@@ -921,41 +921,60 @@ void launch_gather_rows(const long* idx, const float* table, float* out,
 // stage A: per-block partial sum/sumsq over a row range; thread t owns
 // channels t and t+256 in REGISTERS (H <= 512), one atomicAdd per channel
 // per block at the end — coalesced loads, no LDS traffic.
+// adjacent-channel pair load: one 4-B (bf16) / 8-B (fp32) transaction for
+// the thread's two channels (c, c+1) — halves the load count of the
+// latency-bound per-channel reduction kernels
+__device__ __forceinline__ void bn_ld2(const float* p, float& a, float& b) {
+  typedef __attribute__((ext_vector_type(2))) float f2_t;
+  const f2_t v = *reinterpret_cast<const f2_t*>(p);
+  a = v[0]; b = v[1];
+}
+__device__ __forceinline__ void bn_ld2(const __bf16* p, float& a, float& b) {
+  typedef __attribute__((ext_vector_type(2))) __bf16 b2_t;
+  const b2_t v = *reinterpret_cast<const b2_t*>(p);
+  a = (float)v[0]; b = (float)v[1];
+}
+
 template <typename TX = float, bool SLAB = false>
 __global__ void bn_stats_partial_kernel(const TX* __restrict__ x, long n,
                                         int h, float* __restrict__ partials) {
-  const int c0 = threadIdx.x;
-  const int c1 = threadIdx.x + 256;
+  // ADJACENT channel pair per thread (c0 = 2*tid, c1 = c0+1): one paired
+  // load serves both, halving the load count of this latency-bound kernel
+  const int c0 = 2 * threadIdx.x;
+  const int c1 = c0 + 1;
   float s0 = 0.f, q0 = 0.f, s1 = 0.f, q1 = 0.f;
   const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
   const long r0 = (long)blockIdx.x * rows_per_block;
   const long r1 = min(n, r0 + rows_per_block);
   long r = r0;
-  // 4-row unroll: independent loads overlap (single-row loop is latency-bound)
-  for (; r + 4 <= r1; r += 4) {
-    if (c0 < h) {
-      float v[4];
+  if ((h & 1) == 0 && c1 < h) {
+    // 4-row unroll: independent loads overlap
+    for (; r + 4 <= r1; r += 4) {
+      float a[4], b[4];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) v[u] = x[(r + u) * h + c0];
+      for (int u = 0; u < 4; ++u) bn_ld2(&x[(r + u) * h + c0], a[u], b[u]);
 #pragma unroll
-      for (int u = 0; u < 4; ++u) { s0 += v[u]; q0 += v[u] * v[u]; }
+      for (int u = 0; u < 4; ++u) {
+        s0 += a[u]; q0 += a[u] * a[u];
+        s1 += b[u]; q1 += b[u] * b[u];
+      }
     }
-    if (c1 < h) {
-      float v[4];
-#pragma unroll
-      for (int u = 0; u < 4; ++u) v[u] = x[(r + u) * h + c1];
-#pragma unroll
-      for (int u = 0; u < 4; ++u) { s1 += v[u]; q1 += v[u] * v[u]; }
+    for (; r < r1; ++r) {
+      float a, b;
+      bn_ld2(&x[r * h + c0], a, b);
+      s0 += a; q0 += a * a;
+      s1 += b; q1 += b * b;
     }
-  }
-  for (; r < r1; ++r) {
-    if (c0 < h) {
-      const float v = x[r * h + c0];
-      s0 += v; q0 += v * v;
-    }
-    if (c1 < h) {
-      const float v = x[r * h + c1];
-      s1 += v; q1 += v * v;
+  } else {
+    for (; r < r1; ++r) {
+      if (c0 < h) {
+        const float v = x[r * h + c0];
+        s0 += v; q0 += v * v;
+      }
+      if (c1 < h) {
+        const float v = x[r * h + c1];
+        s1 += v; q1 += v * v;
+      }
     }
   }
   if (SLAB) {
@@ -1134,8 +1153,9 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
                                       int h, int relu,
                                       float* __restrict__ partials,
                                       float keep_inv = 1.f) {
-  const int c0 = threadIdx.x;
-  const int c1 = threadIdx.x + 256;
+  // adjacent channel pair per thread (see bn_stats): paired g/y/x loads
+  const int c0 = 2 * threadIdx.x;
+  const int c1 = c0 + 1;
   float s0 = 0.f, q0 = 0.f, s1 = 0.f, q1 = 0.f;
   const long rows_per_block = (n + gridDim.x - 1) / gridDim.x;
   const long r0 = (long)blockIdx.x * rows_per_block;
@@ -1145,50 +1165,47 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
   const float m1 = (c1 < h) ? mean[c1] : 0.f;
   const float i1 = (c1 < h) ? invstd[c1] : 0.f;
   long r = r0;
-  for (; r + 4 <= r1; r += 4) {
-    if (c0 < h) {
-      float gv[4], yv[4], xv[4];
+  if ((h & 1) == 0 && c1 < h) {
+    for (; r + 2 <= r1; r += 2) {
+      float ga[2], gb[2], ya[2], yb[2], xa[2], xb[2];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        gv[u] = (float)g[(r + u) * h + c0];
-        yv[u] = (float)y[(r + u) * h + c0];
-        xv[u] = (float)x[(r + u) * h + c0];
+      for (int u = 0; u < 2; ++u) {
+        bn_ld2(&g[(r + u) * h + c0], ga[u], gb[u]);
+        bn_ld2(&y[(r + u) * h + c0], ya[u], yb[u]);
+        bn_ld2(&x[(r + u) * h + c0], xa[u], xb[u]);
       }
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        float gm = (relu && yv[u] <= 0.f) ? 0.f : gv[u] * keep_inv;
+      for (int u = 0; u < 2; ++u) {
+        float gm0 = (relu && ya[u] <= 0.f) ? 0.f : ga[u] * keep_inv;
+        float gm1 = (relu && yb[u] <= 0.f) ? 0.f : gb[u] * keep_inv;
+        s0 += gm0; q0 += gm0 * (xa[u] - m0) * i0;
+        s1 += gm1; q1 += gm1 * (xb[u] - m1) * i1;
+      }
+    }
+    for (; r < r1; ++r) {
+      float ga, gb, ya, yb, xa, xb;
+      bn_ld2(&g[r * h + c0], ga, gb);
+      bn_ld2(&y[r * h + c0], ya, yb);
+      bn_ld2(&x[r * h + c0], xa, xb);
+      float gm0 = (relu && ya <= 0.f) ? 0.f : ga * keep_inv;
+      float gm1 = (relu && yb <= 0.f) ? 0.f : gb * keep_inv;
+      s0 += gm0; q0 += gm0 * (xa - m0) * i0;
+      s1 += gm1; q1 += gm1 * (xb - m1) * i1;
+    }
+  } else {
+    for (; r < r1; ++r) {
+      if (c0 < h) {
+        float gm = (float)g[r * h + c0] * keep_inv;
+        if (relu && (float)y[r * h + c0] <= 0.f) gm = 0.f;
         s0 += gm;
-        q0 += gm * (xv[u] - m0) * i0;
+        q0 += gm * ((float)x[r * h + c0] - m0) * i0;
       }
-    }
-    if (c1 < h) {
-      float gv[4], yv[4], xv[4];
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        gv[u] = (float)g[(r + u) * h + c1];
-        yv[u] = (float)y[(r + u) * h + c1];
-        xv[u] = (float)x[(r + u) * h + c1];
-      }
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        float gm = (relu && yv[u] <= 0.f) ? 0.f : gv[u] * keep_inv;
+      if (c1 < h) {
+        float gm = (float)g[r * h + c1] * keep_inv;
+        if (relu && (float)y[r * h + c1] <= 0.f) gm = 0.f;
         s1 += gm;
-        q1 += gm * (xv[u] - m1) * i1;
+        q1 += gm * ((float)x[r * h + c1] - m1) * i1;
       }
-    }
-  }
-  for (; r < r1; ++r) {
-    if (c0 < h) {
-      float gm = (float)g[r * h + c0] * keep_inv;
-      if (relu && (float)y[r * h + c0] <= 0.f) gm = 0.f;
-      s0 += gm;
-      q0 += gm * ((float)x[r * h + c0] - m0) * i0;
-    }
-    if (c1 < h) {
-      float gm = (float)g[r * h + c1] * keep_inv;
-      if (relu && (float)y[r * h + c1] <= 0.f) gm = 0.f;
-      s1 += gm;
-      q1 += gm * ((float)x[r * h + c1] - m1) * i1;
     }
   }
   if (SLAB) {
