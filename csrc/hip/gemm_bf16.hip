@@ -1222,10 +1222,13 @@ __global__ void gemm_a16_glds_tn_kernel(const __bf16* __restrict__ a,
   };
 
   f32x4 acc[FM][FN];
+  float dbacc[FM];  // fused bias grad rides the A fragments (see below)
 #pragma unroll
-  for (int i = 0; i < FM; ++i)
+  for (int i = 0; i < FM; ++i) {
+    dbacc[i] = 0.f;
 #pragma unroll
     for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+  }
 
   stage(a, n, n0, c_beg, lds_ab(0));
   stage(b, k2, k0, c_beg, lds_bb(0));
@@ -1233,8 +1236,6 @@ __global__ void gemm_a16_glds_tn_kernel(const __bf16* __restrict__ a,
   __syncthreads();
 
   const bool do_bias = (dbias != nullptr) && (k0 == 0);
-  float dbsum = 0.f;
-  const int bcol = threadIdx.x;  // one thread per A column (BM <= THREADS)
 
   // fragment read via ds_read_b64_tr_b16: each 16-lane group passes
   // CONSECUTIVE 8-B addresses covering one 128-B [4 m][16 col] sub-tile;
@@ -1254,6 +1255,13 @@ __global__ void gemm_a16_glds_tn_kernel(const __bf16* __restrict__ a,
     return u.v8;
   };
 
+  // bias grad (db = colsum of g): the A fragments already pass EVERY g
+  // element through registers, so accumulate there instead of a separate
+  // 32-B-strided LDS sweep (PMC: that sweep was 8.5x the NT kernel's LDS
+  // bank conflicts).  Each column's sum lands spread over the wave's four
+  // 16-lane groups (different k slices) — folded by shfl at the end; the
+  // WCOL waves sharing a wm row-block read identical fragments, so only
+  // wave % WCOL == 0 contributes.
   auto mma = [&](const char* la, const char* lb) {
 #pragma unroll
     for (int s = 0; s < BKM / 32; ++s) {
@@ -1262,6 +1270,11 @@ __global__ void gemm_a16_glds_tn_kernel(const __bf16* __restrict__ a,
       for (int mi = 0; mi < FM; ++mi) av[mi] = frag(la, (wm >> 4) + mi, s);
 #pragma unroll
       for (int ni = 0; ni < FN; ++ni) bv[ni] = frag(lb, (wn >> 4) + ni, s);
+      if (do_bias)
+#pragma unroll
+        for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+          for (int j = 0; j < 8; ++j) dbacc[mi] += (float)av[mi][j];
 #pragma unroll
       for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
@@ -1271,27 +1284,26 @@ __global__ void gemm_a16_glds_tn_kernel(const __bf16* __restrict__ a,
     }
   };
 
-  auto bias_acc = [&](const char* la) {
-    if (!do_bias || bcol >= BM) return;
-    const char* p = la + (bcol / 16) * PANEL_BYTES + (bcol % 16) * 2;
-#pragma unroll
-    for (int r = 0; r < BKM; ++r)
-      dbsum += (float)*reinterpret_cast<const __bf16*>(p + r * 32);
-  };
-
   int buf = 0;
   for (int cc = c_beg + BKM; cc < c_end; cc += BKM) {
     stage(a, n, n0, cc, lds_ab(buf ^ 1));
     stage(b, k2, k0, cc, lds_bb(buf ^ 1));
     mma(lds_ab(buf), lds_bb(buf));
-    bias_acc(lds_ab(buf));
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     buf ^= 1;
   }
   mma(lds_ab(buf), lds_bb(buf));
-  bias_acc(lds_ab(buf));
-  if (do_bias && bcol < BM) atomicAdd(&dbias[n0 + bcol], dbsum);
+  if (do_bias && wave % WCOL == 0) {
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi) {
+      float v = dbacc[mi];
+      v += __shfl_xor(v, 16, 64);
+      v += __shfl_xor(v, 32, 64);
+      if ((lane >> 4) == 0)
+        atomicAdd(&dbias[n0 + wm + mi * 16 + (lane & 15)], v);
+    }
+  }
 
   const int fcol = lane & 15;
   const int frow = (lane >> 4) * 4;
