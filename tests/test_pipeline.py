@@ -396,3 +396,30 @@ def test_reference_pyg_cache_loads_without_pyg(tmp_path):
     torch.save(lst, str(path))
     lst2 = load_data_list_any(str(path))
     assert torch.equal(lst2[1].x, s.x)
+
+
+def test_bench_json_contract_cpu(tmp_path):
+    """The driver parses ONE JSON line from bench.py rank 0 — guard the
+    contract keys and their semantics (CPU run, tiny config)."""
+    import json
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch-size", "16", "--n-batches", "1", "--layers", "2",
+         "--hidden", "32", "--vocab", "small", "--mae-epochs", "0"],
+        capture_output=True, text=True, timeout=420,
+        cwd=str(Path(__file__).resolve().parents[1]),
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    rec = json.loads(out.stdout.strip().splitlines()[-1])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config", "latency_mae"):
+        assert key in rec, key
+    assert rec["n_gpus"] == 1 and rec["steps"] == 2 and rec["warmup"] == 1
+    assert rec["scaling"] == "weak" and rec["higher_is_better"] is True
+    assert rec["data"] == "synthetic"
+    assert rec["config"]["global_batch"] == 16
+    assert rec["value"] > 0 and rec["ms_per_step"] > 0
