@@ -423,3 +423,20 @@ def test_bench_json_contract_cpu(tmp_path):
     assert rec["data"] == "synthetic"
     assert rec["config"]["global_batch"] == 16
     assert rec["value"] > 0 and rec["ms_per_step"] > 0
+
+
+def test_cli_dynamic_loss_scale_cpu(tmp_path):
+    """--loss_scale dynamic parses and trains on CPU (eager GradScaler
+    semantics in FusedAdam)."""
+    import subprocess
+    import sys
+
+    cmd = [sys.executable, "pert_gnn.py", "--synthetic", "--graph_type", "pert",
+           "--epochs", "2", "--num_layers", "1", "--hidden_channels", "16",
+           "--batch_size", "32", "--seed", "3", "--loss_scale", "dynamic",
+           "--processed_dir", str(tmp_path / "processed")]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                       cwd=str(Path(__file__).resolve().parents[1]))
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("Epoch:")]
+    assert len(lines) == 2
