@@ -67,19 +67,24 @@ def build_parser():
     return parser
 
 
-def load_artifacts(args):
+def load_artifacts(args, comm=None):
     import joblib
     import pandas as pd
 
     pdir = args.processed_dir
     if not os.path.isfile(os.path.join(pdir, "tr2data.pt")):
         if args.synthetic:
-            from pertgnn.data.ingest import run_ingest
-            from pertgnn.data.synthetic import SyntheticConfig, write_dataset
+            # rank 0 generates, everyone else waits at the barrier — all
+            # ranks racing to write the same processed/ dir corrupts it
+            if comm is None or comm.rank == 0:
+                from pertgnn.data.ingest import run_ingest
+                from pertgnn.data.synthetic import SyntheticConfig, write_dataset
 
-            root = os.path.dirname(pdir) or "."
-            write_dataset(root, SyntheticConfig())
-            run_ingest(data_root=os.path.join(root, "data"), processed_dir=pdir, verbose=False)
+                root = os.path.dirname(pdir) or "."
+                write_dataset(root, SyntheticConfig())
+                run_ingest(data_root=os.path.join(root, "data"), processed_dir=pdir, verbose=False)
+            if comm is not None:
+                comm.barrier()
         else:
             raise FileNotFoundError(
                 f"{pdir}/tr2data.pt not found — run `python preprocess.py` first "
@@ -113,7 +118,7 @@ def main(argv=None):
     else:
         device = torch.device("cpu")
 
-    tr2data, entry2runtimes, runtime2graph, resource_df = load_artifacts(args)
+    tr2data, entry2runtimes, runtime2graph, resource_df = load_artifacts(args, comm)
 
     cache_path = os.path.join(args.processed_dir, f"full_{args.graph_type}_data_list.pt")
     if os.path.exists(cache_path):

@@ -456,6 +456,15 @@ def test_cli_hipgraph_world2_gloo(tmp_path):
     """pert_gnn.py --hipgraph under DDP (2 gloo ranks): resident-batch
     sharding, GraphStepper's cross-rank metric reduction and the sync-BN
     interaction — the schedule an 8-GPU --hipgraph run uses, minus RCCL."""
+    # pre-generate the dataset in the parent: two ranks would otherwise
+    # race to write the same processed/ dir
+    from pertgnn.data.ingest import run_ingest
+    from pertgnn.data.synthetic import SyntheticConfig, write_dataset
+
+    write_dataset(str(tmp_path), SyntheticConfig())
+    run_ingest(data_root=str(tmp_path / "data"),
+               processed_dir=str(tmp_path / "processed"), verbose=False)
+
     world = 2
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
