@@ -170,6 +170,8 @@ void launch_convert_w16(const float*, void*, long, hipStream_t);
 void launch_gemm_a16_glds_tn(const void*, const void*, float*, float*, int,
                              int, int, hipStream_t);
 void launch_transpose_convert_w16(const float*, void*, int, int, hipStream_t);
+void launch_gemm_skinny_nn(const void*, const float*, float*, int, int, int,
+                           hipStream_t);
 void launch_gemm_bf16_nn_a16(const void*, const float*, float*, int, int, int,
                              hipStream_t);
 void launch_gemm_bf16_tn_a16(const void*, const float*, float*, float*, int,
@@ -925,6 +927,11 @@ torch::Tensor linear_dgrad16(torch::Tensor g, torch::Tensor w) {
   const int n = w.size(0);
   const int k = w.size(1);
   auto dx = torch::empty({m, k}, w.options());
+  if (m <= 16) {  // tiny-row P-table backward: see gemm_skinny_nn_kernel
+    launch_gemm_skinny_nn(g.data_ptr(), w.data_ptr<float>(),
+                          dx.data_ptr<float>(), m, n, k, cur_stream());
+    return dx;
+  }
   launch_gemm_bf16_nn_a16(g.data_ptr(),
                           w.data_ptr<float>(), dx.data_ptr<float>(), m, n, k,
                           cur_stream());

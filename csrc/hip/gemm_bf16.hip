@@ -1353,3 +1353,32 @@ void launch_gemm_a16_glds_tn(const void* a_v, const void* b_v, float* c,
             m - m_full, n, k2, 2);
   }
 }
+
+// ---------------------------------------------------------------------------
+// Skinny NN dgrad: dx[m,k2] = g[m,n] . w[n,k2] for TINY m (the P-table
+// backward: rpctype has ~5 rows).  The tiled kernel's boundary path runs
+// per-element guarded scalar staging on 8 underoccupied workgroups
+// (measured 85 us for 2.6 MFLOP at [5,512]x[512,512]); here one wave owns
+// one (row, 64-column block) output strip — w reads are lane-coalesced,
+// g[i,n] is a broadcast scalar, fully deterministic.
+// ---------------------------------------------------------------------------
+__global__ void gemm_skinny_nn_kernel(const __bf16* __restrict__ g,
+                                      const float* __restrict__ w,
+                                      float* __restrict__ dx, int m, int n,
+                                      int k2) {
+  const int i = blockIdx.x;           // output row
+  const int j0 = blockIdx.y * PERTGNN_WAVE;
+  const int j = j0 + threadIdx.x;     // output column
+  if (i >= m || j >= k2) return;
+  float acc = 0.f;
+  for (int t = 0; t < n; ++t)
+    acc += (float)g[(long)i * n + t] * w[(long)t * k2 + j];
+  dx[(long)i * k2 + j] = acc;
+}
+
+void launch_gemm_skinny_nn(const void* g_v, const float* w, float* dx, int m,
+                           int n, int k2, hipStream_t s) {
+  gemm_skinny_nn_kernel<<<dim3(m, (k2 + PERTGNN_WAVE - 1) / PERTGNN_WAVE),
+                          dim3(PERTGNN_WAVE), 0, s>>>(
+      (const __bf16*)g_v, w, dx, m, n, k2);
+}
