@@ -1370,10 +1370,25 @@ __global__ void gemm_skinny_nn_kernel(const __bf16* __restrict__ g,
   const int j0 = blockIdx.y * PERTGNN_WAVE;
   const int j = j0 + threadIdx.x;     // output column
   if (i >= m || j >= k2) return;
-  float acc = 0.f;
-  for (int t = 0; t < n; ++t)
-    acc += (float)g[(long)i * n + t] * w[(long)t * k2 + j];
-  dx[(long)i * k2 + j] = acc;
+  // 8-deep unroll with split accumulators: the w loads of a batch issue
+  // together (a bare loop serializes one memory latency per iteration —
+  // the whole kernel is a ~20-wave latency chain)
+  float a0 = 0.f, a1 = 0.f;
+  int t = 0;
+  for (; t + 8 <= n; t += 8) {
+    float gv[8], wv[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) gv[u] = (float)g[(long)i * n + t + u];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) wv[u] = w[(long)(t + u) * k2 + j];
+#pragma unroll
+    for (int u = 0; u < 8; u += 2) {
+      a0 += gv[u] * wv[u];
+      a1 += gv[u + 1] * wv[u + 1];
+    }
+  }
+  for (; t < n; ++t) a0 += (float)g[(long)i * n + t] * w[(long)t * k2 + j];
+  dx[(long)i * k2 + j] = a0 + a1;
 }
 
 void launch_gemm_skinny_nn(const void* g_v, const float* w, float* dx, int m,
