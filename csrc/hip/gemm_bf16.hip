@@ -894,9 +894,9 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
   constexpr int NWAVE = THREADS / PERTGNN_WAVE;
   constexpr int WCOL = NWAVE / 2;
   constexpr int FM = (BM / 2) / 16, FN = (BN / WCOL) / 16;
-  constexpr int TILE_BYTES = BM * BK * 2;          // one operand K-tile
-  constexpr int GRPS = TILE_BYTES / 1024;          // 1 KiB per wave-glds
-  constexpr int GRPS_PER_WAVE = GRPS / NWAVE;
+  constexpr int TILE_BYTES = BM * BK * 2;          // A-operand K-tile
+  constexpr int GRPS_A = TILE_BYTES / 1024;        // 1 KiB per wave-glds
+  constexpr int GRPS_B = (BN * BK * 2) / 1024;
   // one arena: [A0 A1 B0 B1] operand buffers; the C-store scratch reuses it
   __shared__ char smem[2 * (BM + BN) * BK * 2];
   const auto lds_ab = [&](int i) { return (__bf16*)(smem + i * TILE_BYTES); };
@@ -922,9 +922,11 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
   const int src_kb = (int)(l_off % ROW_BYTES);     // byte within row
 
   const long lda = k;  // elements
-  auto stage = [&](const __bf16* g, int g0, int k0, __bf16* lds) {
+  // per-operand group counts: A covers BM rows, B covers BN (they differ
+  // for the single-column-tile dgrad variant)
+  auto stage = [&]<int GRPS>(const __bf16* g, int g0, int k0, __bf16* lds) {
 #pragma unroll
-    for (int i = 0; i < GRPS_PER_WAVE; ++i) {
+    for (int i = 0; i < GRPS / NWAVE; ++i) {
       const int grp = wave + i * NWAVE;
       const __bf16* src = (const __bf16*)((const char*)g +
           (long)(g0 + grp * ROWS_PER_GRP + src_row) * lda * 2 +
@@ -942,8 +944,8 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
 #pragma unroll
     for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  stage(a, m0, 0, lds_ab(0));
-  stage(b, n0, 0, lds_bb(0));
+  stage.template operator()<GRPS_A>(a, m0, 0, lds_ab(0));
+  stage.template operator()<GRPS_B>(b, n0, 0, lds_bb(0));
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -977,8 +979,8 @@ __global__ void gemm_a16_glds_nt_kernel(const __bf16* __restrict__ a,
 
   int buf = 0;
   for (int k0 = BK; k0 < k; k0 += BK) {
-    stage(a, m0, k0, lds_ab(buf ^ 1));
-    stage(b, n0, k0, lds_bb(buf ^ 1));
+    stage.template operator()<GRPS_A>(a, m0, k0, lds_ab(buf ^ 1));
+    stage.template operator()<GRPS_B>(b, n0, k0, lds_bb(buf ^ 1));
     mma(lds_ab(buf), lds_bb(buf));
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
@@ -1097,6 +1099,14 @@ static bool glds_ntc() {  // nontemporal C stores (default ON)
 }
 
 
+static bool glds_n256() {  // 256-wide output tile when n == 256 (dgrad)
+  static bool v = [] {
+    const char* e = getenv("PERTGNN_GLDS_N256");
+    return e && atoi(e) == 1;
+  }();
+  return v;
+}
+
 void launch_gemm_a16_glds_nt(const void* a_v, const void* b16_v,
                              const float* b32, const float* bias, void* c_v,
                              int c16, int m, int n, int k, bool relu,
@@ -1105,7 +1115,17 @@ void launch_gemm_a16_glds_nt(const void* a_v, const void* b16_v,
   const __bf16* b = (const __bf16*)b16_v;
   constexpr int BM = 128, BN = 128;
   const int mt = m / BM;  // full row tiles
-  if (mt > 0) {
+  if (mt > 0 && n == 256 && k % 32 == 0 && glds_n256()) {
+    // experimental single-column-tile variant for the dgrad shape
+    if (c16)
+      gemm_a16_glds_nt_kernel<128, 256, 32, __bf16, 512, true>
+          <<<dim3(mt), dim3(512), 0, s>>>(a, b, bias, (__bf16*)c_v, m, n, k,
+                                          relu ? 1 : 0);
+    else
+      gemm_a16_glds_nt_kernel<128, 256, 32, float, 512, true>
+          <<<dim3(mt), dim3(512), 0, s>>>(a, b, bias, (float*)c_v, m, n, k,
+                                          relu ? 1 : 0);
+  } else if (mt > 0) {
     const int grid = mt * (n / BN);
     const int bk = glds_bk();
     const bool t512 = glds_t512();
