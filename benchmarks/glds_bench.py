@@ -34,6 +34,7 @@ def main():
     # (name, M, N(out), K): qkvs fwd + dgrad at the realistic bench scale
     shapes = [
         ("qkvs fwd", 180736, 1024, 256),
+        ("qkvs odd-m", 180186, 1024, 256),  # 26-row wgrad tail + 90-row dgrad tail
         ("qkvs fwd b256", 45056, 1024, 256),
         ("512H qkvs", 90112, 2048, 512),
     ]

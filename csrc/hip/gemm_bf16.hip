@@ -1345,9 +1345,42 @@ __global__ void gemm_a16_glds_tn_kernel(const __bf16* __restrict__ a,
       }
 }
 
+// m-tail of the TN wgrad: C[n,k2] += sum_{i<mt} g[i,n_idx]*x[i,k_idx] for the
+// mt (< BKM) rows past the last full m tile.  The work is tiny (mt*n*k2
+// MACs) but the register-staging kernel used to run it on a
+// (n/128)*(k2/128)-block grid — 16-32 workgroups of guarded scalar staging,
+// ~107 us for 0.01 GFLOP at [26,1024,256].  Here one block owns one output
+// ROW of C (grid = n, fills the chip), threads stride k2: the x reads are
+// lane-coalesced, g[i,jn] is a wave-broadcast scalar, and the contribution
+// is atomicAdd-combined with the main kernel's slices exactly like an extra
+// split-K slice.
+__global__ void gemm_tail_tn_kernel(const __bf16* __restrict__ g,
+                                    const __bf16* __restrict__ b,
+                                    float* __restrict__ c,
+                                    float* __restrict__ dbias, int mt, int n,
+                                    int k2) {
+  const int jn = blockIdx.x;
+  if (dbias && threadIdx.x == 0) {
+    float s = 0.f;
+    for (int i = 0; i < mt; ++i) s += (float)g[(long)i * n + jn];
+    atomicAdd(&dbias[jn], s);
+  }
+  for (int jk = threadIdx.x; jk < k2; jk += blockDim.x) {
+    float a0 = 0.f, a1 = 0.f;
+    int i = 0;
+    for (; i + 2 <= mt; i += 2) {
+      a0 += (float)g[(long)i * n + jn] * (float)b[(long)i * k2 + jk];
+      a1 += (float)g[(long)(i + 1) * n + jn] * (float)b[(long)(i + 1) * k2 + jk];
+    }
+    if (i < mt)
+      a0 += (float)g[(long)i * n + jn] * (float)b[(long)i * k2 + jk];
+    atomicAdd(&c[(long)jn * k2 + jk], a0 + a1);
+  }
+}
+
 // Both operands bf16 contract-major (g [m][n], x [m][k2]); m tail rows
-// beyond the last BKM multiple go through the register-staging TN kernel
-// as an extra atomic slice.
+// beyond the last BKM multiple run gemm_tail_tn_kernel as an extra atomic
+// slice.
 void launch_gemm_a16_glds_tn(const void* a_v, const void* b_v, float* c,
                              float* dbias, int m, int n, int k2,
                              hipStream_t s) {
@@ -1366,14 +1399,9 @@ void launch_gemm_a16_glds_tn(const void* a_v, const void* b_v, float* c,
                                                   slices);
   const int m_full = m - m % BKM;
   if (m_full < m) {
-    // register-staging kernel on the tail rows; slices=2 with a grid that
-    // only contains slice 0 makes it ATOMIC-add its contribution (its
-    // per_slice covers the whole <BKM-row strip in slice 0)
-    const int tiles_t = ((n + 127) / 128) * ((k2 + 127) / 128);
-    gemm_bf16_tn_kernel<128, 128, 32, __bf16, __bf16, __bf16>
-        <<<dim3(tiles_t * 2), dim3(BGEMM_THREADS), 0, s>>>(
-            a + (long)m_full * n, b + (long)m_full * k2, c, dbias,
-            m - m_full, n, k2, 2);
+    gemm_tail_tn_kernel<<<dim3(n), dim3(256), 0, s>>>(
+        a + (long)m_full * n, b + (long)m_full * k2, c, dbias, m - m_full, n,
+        k2);
   }
 }
 
@@ -1385,9 +1413,10 @@ void launch_gemm_a16_glds_tn(const void* a_v, const void* b_v, float* c,
 // one (row, 64-column block) output strip — w reads are lane-coalesced,
 // g[i,n] is a broadcast scalar, fully deterministic.
 // ---------------------------------------------------------------------------
+template <typename OT = float>
 __global__ void gemm_skinny_nn_kernel(const __bf16* __restrict__ g,
                                       const float* __restrict__ w,
-                                      float* __restrict__ dx, int m, int n,
+                                      OT* __restrict__ dx, int m, int n,
                                       int k2) {
   const int i = blockIdx.x;           // output row
   const int j0 = blockIdx.y * PERTGNN_WAVE;
@@ -1411,7 +1440,7 @@ __global__ void gemm_skinny_nn_kernel(const __bf16* __restrict__ g,
     }
   }
   for (; t < n; ++t) a0 += (float)g[(long)i * n + t] * w[(long)t * k2 + j];
-  dx[(long)i * k2 + j] = a0 + a1;
+  dx[(long)i * k2 + j] = (OT)(a0 + a1);
 }
 
 void launch_gemm_skinny_nn(const void* g_v, const float* w, float* dx, int m,
