@@ -1,5 +1,6 @@
-"""One-off diag: log the (m,n,k) of every linear16 GEMM in one flagship step
-and which routing gate each wgrad/dgrad hits (scratch; not part of the repo)."""
+"""Diagnostic: log the (m,n,k) of every linear16 GEMM in one flagship step
+and which routing gate each wgrad/dgrad call hits (run on a GPU box:
+`PYTHONPATH=. python tools/diag_wgrad_shapes.py`)."""
 import collections
 import torch
 
