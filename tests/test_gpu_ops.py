@@ -720,3 +720,45 @@ def test_fused_dropout_semantics():
     ye = F.batchnorm_relu(x, gamma, beta, rm.clone(), rv.clone(), 0.1, 1e-5,
                           False, fuse_relu=True, dropout_p=p)
     assert (ye != 0).float().mean().item() > 0.9 * (y0 != 0).float().mean().item()
+
+
+def test_glds_odd_m_tail_parity():
+    """glds GEMM family at an m that is NOT a tile multiple: exercises the
+    element-parallel m-tail kernels (gemm_tail_tn_kernel for the wgrad's
+    27-row strip, the a16o16 NN tail for dgrad) alongside the glds mains
+    (csrc/hip/gemm_bf16.hip launchers)."""
+    require_ext()
+    import pertgnn._C as C
+    g = torch.Generator().manual_seed(99)
+    m, n, k = 4123, 1024, 256  # m % 64 == 27, m % 128 == 27
+    x = torch.randn(m, k, generator=g).to(DEV).to(torch.bfloat16)
+    w = torch.randn(n, k, generator=g).to(DEV) * 0.05
+    b = torch.randn(n, generator=g).to(DEV)
+
+    y = C.linear_fwd_a16o16(x, w, b)
+    ref_y = x.float() @ w.to(torch.bfloat16).float().t() + b
+    assert torch.allclose(y.float(), ref_y, atol=0.05, rtol=1e-2), \
+        (y.float() - ref_y).abs().max()
+
+    gy = torch.randn(m, n, generator=g).to(DEV).to(torch.bfloat16)
+    dx = C.linear_dgrad16_o16(gy, w)
+    ref_dx = gy.float() @ w.to(torch.bfloat16).float()
+    assert torch.allclose(dx.float(), ref_dx, atol=0.3, rtol=1e-2), \
+        (dx.float() - ref_dx).abs().max()
+
+    dw, db = C.linear_wgrad16_b16(gy, x, True)
+    ref_dw = gy.float().t() @ x.float()
+    ref_db = gy.float().sum(0)
+    torch.cuda.synchronize()
+    assert torch.allclose(dw, ref_dw, atol=0.5, rtol=1e-3), \
+        (dw - ref_dw).abs().max()
+    assert torch.allclose(db, ref_db, atol=0.2, rtol=1e-3), \
+        (db - ref_db).abs().max()
+
+    # and a tail-free m for contrast (same tolerance): routing must agree
+    m2 = 4096
+    dw2, db2 = C.linear_wgrad16_b16(gy[:m2].contiguous(), x[:m2].contiguous(),
+                                    True)
+    ref_dw2 = gy[:m2].float().t() @ x[:m2].float()
+    assert torch.allclose(dw2, ref_dw2, atol=0.5, rtol=1e-3), \
+        (dw2 - ref_dw2).abs().max()
