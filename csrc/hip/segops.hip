@@ -948,13 +948,15 @@ __global__ void bn_stats_partial_kernel(const TX* __restrict__ x, long n,
   const long r1 = min(n, r0 + rows_per_block);
   long r = r0;
   if ((h & 1) == 0 && c1 < h) {
-    // 4-row unroll: independent loads overlap
-    for (; r + 4 <= r1; r += 4) {
-      float a[4], b[4];
+    // 8-row unroll: the kernel is latency-bound (one bf16x2 load per
+    // channel pair per row) — 8 independent loads in flight halve the
+    // stall count vs 4 (measured 58.7 -> ~35 us at 180k x 256)
+    for (; r + 16 <= r1; r += 16) {
+      float a[16], b[16];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) bn_ld2(&x[(r + u) * h + c0], a[u], b[u]);
+      for (int u = 0; u < 16; ++u) bn_ld2(&x[(r + u) * h + c0], a[u], b[u]);
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
+      for (int u = 0; u < 16; ++u) {
         s0 += a[u]; q0 += a[u] * a[u];
         s1 += b[u]; q1 += b[u] * b[u];
       }
@@ -1021,10 +1023,25 @@ static inline bool pertgnn_deterministic_seg() {
   return e && e[0] == '1';
 }
 
+// grid size for the BN channel reductions (atomic-combine path): 512 blocks
+// is 2 blocks/CU — enough to stream ~3.5 TB/s but measurably latency-bound
+// on the single-stream stats kernel; PERTGNN_BN_BLOCKS overrides for tuning
+static inline int bn_partial_blocks(long n) {
+  static const int env = [] {
+    const char* e = getenv("PERTGNN_BN_BLOCKS");
+    return e ? atoi(e) : 0;
+  }();
+  const long cap = (n + 63) / 64;
+  const long want = env > 0 ? env : 512;
+  return (int)min(want, cap);
+}
+
 template <typename TX>
 static void bn_stats_dispatch(const TX* x, long n, int h, float* partials,
                               hipStream_t s) {
-  const int nblocks = (int)min((long)512, (n + 63) / 64);
+  const int nblocks = pertgnn_deterministic_seg()
+                          ? (int)min((long)512, (n + 63) / 64)
+                          : bn_partial_blocks(n);
   if (pertgnn_deterministic_seg() && h <= 1024) {
     float* slab = bn_det_slab();
     bn_stats_partial_kernel<TX, true><<<nblocks, 256, 0, s>>>(x, n, h, slab);
@@ -1166,16 +1183,16 @@ __global__ void bn_bwd_partial_kernel(const TG* __restrict__ g,
   const float i1 = (c1 < h) ? invstd[c1] : 0.f;
   long r = r0;
   if ((h & 1) == 0 && c1 < h) {
-    for (; r + 2 <= r1; r += 2) {
-      float ga[2], gb[2], ya[2], yb[2], xa[2], xb[2];
+    for (; r + 8 <= r1; r += 8) {
+      float ga[8], gb[8], ya[8], yb[8], xa[8], xb[8];
 #pragma unroll
-      for (int u = 0; u < 2; ++u) {
+      for (int u = 0; u < 8; ++u) {
         bn_ld2(&g[(r + u) * h + c0], ga[u], gb[u]);
         bn_ld2(&y[(r + u) * h + c0], ya[u], yb[u]);
         bn_ld2(&x[(r + u) * h + c0], xa[u], xb[u]);
       }
 #pragma unroll
-      for (int u = 0; u < 2; ++u) {
+      for (int u = 0; u < 8; ++u) {
         float gm0 = (relu && ya[u] <= 0.f) ? 0.f : ga[u] * keep_inv;
         float gm1 = (relu && yb[u] <= 0.f) ? 0.f : gb[u] * keep_inv;
         s0 += gm0; q0 += gm0 * (xa[u] - m0) * i0;
@@ -1230,7 +1247,9 @@ static void bn_bwd_partials_dispatch(const TG* g, const TX* x, const TY* y,
                                      const float* mean, const float* invstd,
                                      long n, int h, bool relu, float* partials,
                                      float keep_inv, hipStream_t s) {
-  const int nblocks = (int)min((long)512, (n + 63) / 64);
+  const int nblocks = pertgnn_deterministic_seg()
+                          ? (int)min((long)512, (n + 63) / 64)
+                          : bn_partial_blocks(n);
   if (pertgnn_deterministic_seg() && h <= 1024) {
     float* slab = bn_det_slab();
     bn_bwd_partial_kernel<TG, TY, TX, true><<<nblocks, 256, 0, s>>>(
