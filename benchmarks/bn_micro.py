@@ -34,10 +34,15 @@ def main():
 
     t_stats = timeit(lambda: C.bn_stats(x))
     t_bwd = timeit(lambda: C.bn_bwd_partials16(g, x, y, mean, invstd, True, 1.0))
+    gamma = torch.ones(h, device=dev)
+    part = C.bn_bwd_partials16(g, x, y, mean, invstd, True, 1.0)
+    t_apply = timeit(lambda: C.bn_bwd_apply16(g, x, y, mean, invstd, gamma,
+                                              part, part, True, 1.0))
     gb = n * h * 2 / 1e9
     blocks = os.environ.get("PERTGNN_BN_BLOCKS", "512(default)")
     print(f"blocks={blocks:>14}  bn_stats {t_stats:7.1f} us ({gb / t_stats * 1e6 / 1e3:4.1f} TB/s)"
-          f"  bn_bwd_partials {t_bwd:7.1f} us ({3 * gb / t_bwd * 1e6 / 1e3:4.1f} TB/s)")
+          f"  bn_bwd_partials {t_bwd:7.1f} us ({3 * gb / t_bwd * 1e6 / 1e3:4.1f} TB/s)"
+          f"  bn_bwd_apply {t_apply:7.1f} us ({4 * gb / t_apply * 1e6 / 1e3:4.1f} TB/s)")
 
     # parity across block counts (vs plain torch fp32 on rounded operands)
     p = C.bn_stats(x)

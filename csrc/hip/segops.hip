@@ -1280,15 +1280,22 @@ __global__ void bn_bwd_apply_kernel(
     const long numq = n * (h / 4);
     const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
     const long stride = (long)gridDim.x * blockDim.x;
+    struct alignas(4 * sizeof(TG)) TG4 { TG v[4]; };
+    struct alignas(4 * sizeof(TY)) TY4 { TY v[4]; };
     for (long q = i0; q < numq; q += stride) {
       const long t = q * 4;
       const int c = (int)(t % h);
       float gv[4], yv[4], xv[4];
+      // vectorized g/y quad loads (4 scalar 16-bit loads left this kernel
+      // issue-bound alongside the already-vector x load)
+      const TG4 gq = *reinterpret_cast<const TG4*>(&g[t]);
 #pragma unroll
-      for (int u = 0; u < 4; ++u) gv[u] = (float)g[t + u];
-      if (relu)
+      for (int u = 0; u < 4; ++u) gv[u] = (float)gq.v[u];
+      if (relu) {
+        const TY4 yq = *reinterpret_cast<const TY4*>(&y[t]);
 #pragma unroll
-        for (int u = 0; u < 4; ++u) yv[u] = (float)y[t + u];
+        for (int u = 0; u < 4; ++u) yv[u] = (float)yq.v[u];
+      }
       vs_ld4(&x[t], xv);
       struct alignas(4 * sizeof(TX)) TX4 { TX v[4]; };
       TX4 o;
