@@ -440,3 +440,46 @@ def test_cli_dynamic_loss_scale_cpu(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     lines = [l for l in r.stdout.splitlines() if l.startswith("Epoch:")]
     assert len(lines) == 2
+
+
+def test_reference_pyg_cache_flat_layout(tmp_path):
+    """Older PyG pickles keep tensors as direct Data attributes (no
+    _store/_mapping nesting) — the shim's tensor walk must find both
+    layouts (pertgnn/data/pyg_compat.py:_find_tensors)."""
+    import sys
+    import types
+
+    tg = types.ModuleType("torch_geometric")
+    tg_data = types.ModuleType("torch_geometric.data")
+
+    class Data:
+        def __init__(self, **kw):
+            self.__dict__.update(kw)
+
+    Data.__module__ = "torch_geometric.data"
+    Data.__qualname__ = "Data"
+    tg_data.Data = Data
+    sys.modules["torch_geometric"] = tg
+    sys.modules["torch_geometric.data"] = tg_data
+    try:
+        d = Data(
+            x=torch.randn(3, 9),
+            edge_index=torch.tensor([[0, 1], [1, 2]]),
+            edge_attr=torch.tensor([[1, 0], [0, 1]]),
+            cat_X=torch.tensor([[0], [1], [2]]),
+            node_depth=torch.tensor([[0], [1], [2]]),
+            pattern_num_nodes=torch.tensor([[3.], [3.], [3.]]),
+            pattern_probs=torch.tensor([[1.0]]),
+            entry_id=torch.tensor([7]),
+            y=torch.tensor(5.0),
+        )
+        path = tmp_path / "full_span_data_list.pt"
+        torch.save([d], str(path))
+    finally:
+        del sys.modules["torch_geometric.data"], sys.modules["torch_geometric"]
+
+    from pertgnn.data.pyg_compat import load_data_list_any
+
+    lst = load_data_list_any(str(path))
+    assert len(lst) == 1 and int(lst[0].entry_id) == 7
+    assert torch.allclose(lst[0].rt_probs.flatten(), torch.ones(3))
